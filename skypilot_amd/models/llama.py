@@ -1,0 +1,192 @@
+"""Llama-3 model family, written MI355X-first.
+
+Differences from a stock HF/torchtitan implementation, driven by the
+CDNA4 kernels in :mod:`skypilot_amd.ops`:
+
+* tensors stay in ``[B, S, H, D]`` layout end-to-end — the attention
+  kernels read strided rows directly, so there are no head transposes;
+* RMSNorm / RoPE / attention / cross-entropy are the fused HIP kernels;
+* GEMMs are plain ``nn.Linear`` (hipBLASLt on ROCm);
+* RoPE cos/sin tables are precomputed fp32 buffers (on-device trig would
+  make the op VALU-bound — guide Appendix B).
+
+Reference parity: SkyPilot bundles no models (it launches user programs,
+SURVEY.md §2.11); these are the bundled train/serve entrypoint models the
+north star requires.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from skypilot_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama3-8b"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+    max_seq_len: int = 8192
+    tie_embeddings: bool = False
+
+
+CONFIGS = {
+    "llama3-8b": LlamaConfig(),
+    "llama3-70b": LlamaConfig(
+        name="llama3-70b", hidden_size=8192, intermediate_size=28672,
+        num_layers=80, num_heads=64, num_kv_heads=8),
+    # Small configs for tests / smoke.
+    "llama-debug": LlamaConfig(
+        name="llama-debug", vocab_size=512, hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=2, num_kv_heads=1,
+        head_dim=128, max_seq_len=512),
+    "llama-smoke": LlamaConfig(
+        name="llama-smoke", vocab_size=4096, hidden_size=1024,
+        intermediate_size=2816, num_layers=4, num_heads=8, num_kv_heads=4,
+        head_dim=128, max_seq_len=2048),
+}
+
+
+def rope_tables(cfg: LlamaConfig, device, dtype=torch.float32):
+    half = cfg.head_dim // 2
+    inv_freq = 1.0 / (cfg.rope_theta ** (
+        torch.arange(0, half, dtype=torch.float32, device=device) / half))
+    t = torch.arange(cfg.max_seq_len, dtype=torch.float32, device=device)
+    freqs = torch.outer(t, inv_freq)  # [S, D/2]
+    return freqs.cos().to(dtype).contiguous(), freqs.sin().to(dtype).contiguous()
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        h, d = cfg.hidden_size, cfg.head_dim
+        self.n_q, self.n_kv = cfg.num_heads, cfg.num_kv_heads
+        self.wq = nn.Linear(h, self.n_q * d, bias=False)
+        self.wk = nn.Linear(h, self.n_kv * d, bias=False)
+        self.wv = nn.Linear(h, self.n_kv * d, bias=False)
+        self.wo = nn.Linear(self.n_q * d, h, bias=False)
+        self.scale = 1.0 / math.sqrt(d)
+
+    def forward(self, x, cos, sin, positions, kv_cache=None):
+        B, S, _ = x.shape
+        d = self.cfg.head_dim
+        q = self.wq(x).view(B, S, self.n_q, d)
+        k = self.wk(x).view(B, S, self.n_kv, d)
+        v = self.wv(x).view(B, S, self.n_kv, d)
+        q = ops.rope(q.reshape(B * S, self.n_q, d), cos, sin,
+                     positions).view(B, S, self.n_q, d)
+        k = ops.rope(k.reshape(B * S, self.n_kv, d), cos, sin,
+                     positions).view(B, S, self.n_kv, d)
+        if kv_cache is not None:
+            k, v = kv_cache.update(self.layer_idx, k, v)
+            o = ops.attention(q, k, v, self.scale, causal=False) \
+                if S == 1 else ops.attention(q, k, v, self.scale, causal=True)
+        else:
+            o = ops.attention(q, k, v, self.scale, causal=True)
+        return self.wo(o.reshape(B, S, self.n_q * d))
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        h, m = cfg.hidden_size, cfg.intermediate_size
+        self.w_gate_up = nn.Linear(h, 2 * m, bias=False)  # fused gate|up GEMM
+        self.w_down = nn.Linear(m, h, bias=False)
+        self.m = m
+
+    def forward(self, x):
+        gu = self.w_gate_up(x)
+        g, u = gu.split(self.m, dim=-1)
+        return self.w_down(F.silu(g) * u)
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig, layer_idx: int):
+        super().__init__()
+        self.attn = Attention(cfg)
+        self.attn.layer_idx = layer_idx
+        self.mlp = MLP(cfg)
+        self.attn_norm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.mlp_norm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.eps = cfg.norm_eps
+
+    def forward(self, x, cos, sin, positions, kv_cache=None):
+        x = x + self.attn(ops.rmsnorm(x, self.attn_norm, self.eps), cos, sin,
+                          positions, kv_cache)
+        x = x + self.mlp(ops.rmsnorm(x, self.mlp_norm, self.eps))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.blocks = nn.ModuleList(
+            Block(cfg, i) for i in range(cfg.num_layers))
+        self.final_norm = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.embed.weight
+        self._rope = None
+
+    def _tables(self, device):
+        if self._rope is None or self._rope[0].device != device:
+            self._rope = rope_tables(self.cfg, device)
+        return self._rope
+
+    def forward(self, tokens: torch.Tensor, positions: torch.Tensor = None,
+                kv_cache=None) -> torch.Tensor:
+        B, S = tokens.shape
+        cos, sin = self._tables(tokens.device)
+        if positions is None:
+            positions = torch.arange(S, dtype=torch.int32,
+                                     device=tokens.device)
+            positions = positions.unsqueeze(0).expand(B, S).reshape(-1)
+        x = self.embed(tokens)
+        for blk in self.blocks:
+            x = blk(x, cos, sin, positions, kv_cache)
+        x = ops.rmsnorm(x, self.final_norm, self.cfg.norm_eps)
+        return self.lm_head(x)
+
+    def loss(self, tokens: torch.Tensor, targets: torch.Tensor):
+        """Forward + fused cross entropy (logit grads computed in-kernel)."""
+        logits = self.forward(tokens)
+        return ops.fused_cross_entropy(
+            logits.reshape(-1, self.cfg.vocab_size), targets.reshape(-1))
+
+    def num_params(self) -> int:
+        return sum(p.numel() for p in self.parameters())
+
+    def flops_per_token(self, seq_len: int) -> float:
+        """Training (fwd+bwd) FLOPs per token, attention included."""
+        c = self.cfg
+        return 6 * self.num_params() + \
+            6 * c.num_layers * 2 * seq_len * c.num_heads * c.head_dim
+
+
+def build_model(name: str, device="cpu", dtype=torch.bfloat16,
+                seed: int = 0) -> Llama:
+    cfg = CONFIGS[name]
+    torch.manual_seed(seed)
+    with torch.device("meta" if device != "cpu" else "cpu"):
+        model = Llama(cfg)
+    if device != "cpu":
+        model = model.to_empty(device=device)
+        with torch.no_grad():
+            for p in model.parameters():
+                p.normal_(0.0, 0.02)
+    return model.to(dtype)

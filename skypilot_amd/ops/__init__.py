@@ -1,0 +1,270 @@
+"""skypilot_amd.ops — MI355X-native fused ops for the bundled trainer.
+
+Each op routes to the hand-written CDNA4 HIP kernels (``_C.so``, built
+in-tree by :mod:`skypilot_amd.ops.build`) whenever the tensors live on a
+GPU.  On a GPU box a missing/unloadable extension is a **hard error** —
+there is deliberately no silent eager fallback (the driver verifies the
+native code actually loaded).  On CPU (the no-GPU CI container) the ops
+fall back to plain fp32 PyTorch reference implementations, which are also
+what the GPU numerics tests compare the kernels against.
+
+Reference parity note: SkyPilot itself ships no kernels (SURVEY.md
+§2.11); these ops are the MI355X-native additions the north star
+requires for the bundled Llama train/serve entrypoints.
+"""
+from __future__ import annotations
+
+import importlib.util
+import math
+from pathlib import Path
+
+import torch
+
+_C = None
+_LOAD_ERR: Exception | None = None
+
+
+def _load():
+    global _C, _LOAD_ERR
+    if _C is not None:
+        return _C
+    so = Path(__file__).resolve().parent / "_C.so"
+    try:
+        if not so.exists():
+            raise ImportError(
+                f"native extension {so} not built; run "
+                "`python -m skypilot_amd.ops.build`")
+        spec = importlib.util.spec_from_file_location("skypilot_amd.ops._C",
+                                                      so)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _C = mod
+    except Exception as e:  # noqa: BLE001
+        _LOAD_ERR = e
+        raise
+    return _C
+
+
+def native():
+    """The native module; raises loudly if unavailable."""
+    return _load()
+
+
+def native_available() -> bool:
+    try:
+        _load()
+        return True
+    except Exception:  # noqa: BLE001
+        return False
+
+
+def _require_native(opname: str):
+    try:
+        return _load()
+    except Exception as e:  # noqa: BLE001
+        raise RuntimeError(
+            f"skypilot_amd.ops.{opname}: tensor is on GPU but the native "
+            f"CDNA4 extension failed to load ({e}). Refusing to fall back "
+            "to eager PyTorch on the GPU path.") from e
+
+
+# ===========================================================================
+# Reference (CPU / verification) implementations — plain fp32 PyTorch.
+# ===========================================================================
+def rmsnorm_ref(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    xf = x.float()
+    inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * inv * w.float()).to(x.dtype)
+
+
+def rope_ref(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+             positions: torch.Tensor, backward: bool = False) -> torch.Tensor:
+    # x: [..., T, H, D]; cos/sin: [S, D/2]; positions: [T]
+    D = x.shape[-1]
+    xf = x.float()
+    c = cos[positions]  # [T, D/2]
+    s = sin[positions] * (-1.0 if backward else 1.0)
+    c = c.unsqueeze(-2)  # [T, 1, D/2]
+    s = s.unsqueeze(-2)
+    x1, x2 = xf[..., : D // 2], xf[..., D // 2:]
+    out = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+    return out.to(x.dtype)
+
+
+def attention_ref(q, k, v, scale: float, causal: bool = True):
+    # q: [B,S,Hq,D], k/v: [B,S,Hkv,D] -> [B,S,Hq,D]
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+    qf = q.float().permute(0, 2, 1, 3)  # B,Hq,S,D
+    kf = k.float().permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
+    vf = v.float().permute(0, 2, 1, 3).repeat_interleave(rep, dim=1)
+    att = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.triu(torch.ones(S, S, dtype=torch.bool,
+                                     device=q.device), 1)
+        att = att.masked_fill(mask, float("-inf"))
+    p = att.softmax(-1)
+    out = torch.matmul(p, vf)
+    return out.permute(0, 2, 1, 3).to(q.dtype)
+
+
+def cross_entropy_ref(logits: torch.Tensor, targets: torch.Tensor,
+                      ignore_index: int = -100) -> torch.Tensor:
+    return torch.nn.functional.cross_entropy(
+        logits.float(), targets.long(), ignore_index=ignore_index,
+        reduction="mean")
+
+
+# ===========================================================================
+# Autograd wrappers.
+# ===========================================================================
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        if x.is_cuda:
+            C = _require_native("rmsnorm")
+            inv_rms = torch.empty(x.numel() // x.shape[-1],
+                                  device=x.device, dtype=torch.float32)
+            y = C.rmsnorm_fwd(x.contiguous(), w.contiguous(), inv_rms, eps)
+            ctx.save_for_backward(x, w, inv_rms)
+            ctx.native = True
+            return y
+        ctx.save_for_backward(x, w)
+        ctx.eps = eps
+        ctx.native = False
+        return rmsnorm_ref(x, w, eps)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.native:
+            x, w, inv_rms = ctx.saved_tensors
+            C = native()
+            dx, dw = C.rmsnorm_bwd(x.contiguous(), w.contiguous(),
+                                   dy.contiguous(), inv_rms)
+            return dx, dw.to(w.dtype), None
+        x, w = ctx.saved_tensors
+        xf, wf, dyf = x.float(), w.float(), dy.float()
+        H = x.shape[-1]
+        inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + ctx.eps)
+        dot = (dyf * wf * xf).sum(-1, keepdim=True)
+        dx = inv * (dyf * wf - xf * inv * inv * dot / H)
+        dw = (dyf * xf * inv).reshape(-1, H).sum(0)
+        return dx.to(x.dtype), dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5):
+    return _RMSNormFn.apply(x, w, eps)
+
+
+class _RoPEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin, positions):
+        ctx.save_for_backward(cos, sin, positions)
+        if x.is_cuda:
+            C = _require_native("rope")
+            return C.rope(x.contiguous(), cos, sin, positions, False)
+        return rope_ref(x, cos, sin, positions)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin, positions = ctx.saved_tensors
+        if dy.is_cuda:
+            C = native()
+            return C.rope(dy.contiguous(), cos, sin, positions,
+                          True), None, None, None
+        return rope_ref(dy, cos, sin, positions, backward=True), \
+            None, None, None
+
+
+def rope(x, cos, sin, positions):
+    return _RoPEFn.apply(x, cos, sin, positions)
+
+
+class _AttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale, causal):
+        ctx.scale, ctx.causal = scale, causal
+        if q.is_cuda:
+            C = _require_native("attention")
+            O, lse = C.attn_fwd(q.contiguous(), k.contiguous(),
+                                v.contiguous(), scale, causal)
+            ctx.save_for_backward(q, k, v, O, lse)
+            ctx.native = True
+            return O
+        ctx.save_for_backward(q, k, v)
+        ctx.native = False
+        return attention_ref(q, k, v, scale, causal)
+
+    @staticmethod
+    def backward(ctx, dO):
+        if ctx.native:
+            q, k, v, O, lse = ctx.saved_tensors
+            C = native()
+            dq, dk, dv = C.attn_bwd(q, k, v, O, dO.contiguous(), lse,
+                                    ctx.scale, ctx.causal)
+            return dq, dk, dv, None, None
+        q, k, v = ctx.saved_tensors
+        with torch.enable_grad():
+            qd = q.detach().requires_grad_()
+            kd = k.detach().requires_grad_()
+            vd = v.detach().requires_grad_()
+            out = attention_ref(qd, kd, vd, ctx.scale, ctx.causal)
+            gq, gk, gv = torch.autograd.grad(out, (qd, kd, vd), dO)
+        return gq, gk, gv, None, None
+
+
+def attention(q, k, v, scale=None, causal=True):
+    """Flash attention, layout [B, S, H, D] with GQA (Hq multiple of Hkv)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    return _AttentionFn.apply(q, k, v, scale, causal)
+
+
+class _CrossEntropyFn(torch.autograd.Function):
+    """Fused CE: forward computes mean loss AND writes d_logits into the
+    logits buffer (scaled by 1/n_valid); backward scales by grad_output."""
+
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        if logits.is_cuda:
+            C = _require_native("cross_entropy")
+            n_valid = int((targets != ignore_index).sum().item())
+            n_valid = max(n_valid, 1)
+            logits = logits.contiguous()
+            loss = C.cross_entropy_fused(logits, targets.int(),
+                                         1.0 / n_valid, ignore_index)
+            ctx.save_for_backward(logits)  # now holds d_logits
+            ctx.native = True
+            return loss.sum() / n_valid
+        ctx.native = False
+        ctx.save_for_backward(logits, targets)
+        ctx.ignore_index = ignore_index
+        return cross_entropy_ref(logits, targets, ignore_index)
+
+    @staticmethod
+    def backward(ctx, g):
+        if ctx.native:
+            (dlogits,) = ctx.saved_tensors
+            if not (isinstance(g, torch.Tensor) and g.numel() == 1
+                    and float(g) == 1.0):
+                dlogits = dlogits * g
+            return dlogits, None, None
+        logits, targets = ctx.saved_tensors
+        with torch.enable_grad():
+            ld = logits.detach().requires_grad_()
+            loss = cross_entropy_ref(ld, targets, ctx.ignore_index)
+            (gl,) = torch.autograd.grad(loss, (ld,), g)
+        return gl, None, None
+
+
+def fused_cross_entropy(logits, targets, ignore_index: int = -100):
+    """NB: on GPU the logits buffer is overwritten with its gradient."""
+    return _CrossEntropyFn.apply(logits, targets, ignore_index)
+
+
+__all__ = [
+    "rmsnorm", "rope", "attention", "fused_cross_entropy", "native",
+    "native_available", "rmsnorm_ref", "rope_ref", "attention_ref",
+    "cross_entropy_ref",
+]

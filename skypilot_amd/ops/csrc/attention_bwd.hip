@@ -1,0 +1,404 @@
+// Flash-attention backward (causal, GQA, D=128, bf16) for CDNA4/gfx950.
+//
+// Three kernels, no atomics (FA2-style split):
+//   1. attn_bwd_pre:  Dvec = rowsum(dO * O)               [B,Hq,S] fp32
+//   2. attn_bwd_dkv:  one block per kv tile; loops grouped q-heads and
+//      q tiles >= diagonal; accumulates dK, dV in registers.
+//   3. attn_bwd_dq:   one block per q tile; loops kv tiles <= diagonal.
+// All MFMA B-fragments come from swizzled LDS tiles staged per iteration
+// (normal + transposed copies where the contraction axis demands it).
+//
+// Math (P normalized via saved lse): P = exp(scale*QK^T - lse);
+// dV = P^T dO; dP = dO V^T; dS = P*(dP - Dvec); dQ = scale*dS*K;
+// dK = scale*dS^T*Q.
+#include "common.h"
+
+#define ATT_D 128
+#define BM 64
+#define BN 64
+
+// ---------------------------------------------------------------------------
+// Dvec preprocess: one wave per (b, qh, s) row.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void attn_bwd_pre_kernel(
+    const unsigned short* __restrict__ dO, const unsigned short* __restrict__ O,
+    float* __restrict__ Dvec, long long rows, int Hq) {
+  (void)Hq;
+  const int lane = threadIdx.x & 63;
+  const long long wave_id =
+      (long long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long long stride = (long long)gridDim.x * (blockDim.x >> 6);
+  for (long long row = wave_id; row < rows; row += stride) {
+    // row = (b*S + s)*Hq + h ; layout [B,S,Hq,D] is contiguous in rows*D.
+    const unsigned short* drow = dO + row * ATT_D;
+    const unsigned short* orow = O + row * ATT_D;
+    float acc = bf2f(drow[lane]) * bf2f(orow[lane]) +
+                bf2f(drow[lane + 64]) * bf2f(orow[lane + 64]);
+    acc = wave_reduce_sum(acc);
+    // Dvec is stored [B,S,Hq] (same row order as the [B,S,Hq,D] tensors);
+    // the dkv/dq kernels index it with stride Hq accordingly.
+    if (lane == 0) Dvec[row] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dK/dV kernel. Grid: (S/BN, B*Hkv). LDS: Q,QT,dO,dOT (16KB each) + 8KB
+// shared P/dS staging = 72KB -> 2 blocks/CU.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dK, unsigned short* __restrict__ dV, int B,
+    int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ unsigned short q_lds[BM * ATT_D];
+  __shared__ unsigned short qt_lds[ATT_D * BM];
+  __shared__ unsigned short do_lds[BM * ATT_D];
+  __shared__ unsigned short dot_lds[ATT_D * BM];
+  __shared__ unsigned short p_lds[BN * BM];  // PT then reused for dST
+
+  const int kt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int kvh = bh % Hkv;
+  const int group = Hq / Hkv;
+  const int kvbase = kt * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;    // wave owns kv rows [16w, 16w+16)
+  const int lrow = lane & 15;
+  const int lgrp = lane >> 4;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  // K,V fragments for this wave's 16 kv rows (A-operand, fixed).
+  s16x8 a_k[4], a_v[4];
+  {
+    const int kvrow = kvbase + 16 * w + lrow;
+    const unsigned short* ksrc = Kb + (long long)kvrow * kv_rowstride;
+    const unsigned short* vsrc = Vb + (long long)kvrow * kv_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      a_k[ks] = *(const s16x8*)(ksrc + ks * 32 + lgrp * 8);
+      a_v[ks] = *(const s16x8*)(vsrc + ks * 32 + lgrp * 8);
+    }
+  }
+
+  f32x4 dv_acc[8], dk_acc[8];
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+    dv_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dk_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  for (int g = 0; g < group; ++g) {
+    const int qh = kvh * group + g;
+    const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+    const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+    const float* lse_b = lse + ((long long)b * Hq + qh) * S;
+    // Dvec stored [B,S,Hq] (see pre kernel).
+    const float* dvec_b = Dvec + (long long)b * S * Hq + qh;
+
+    const int qt0 = causal ? kvbase / BM : 0;
+    for (int qt = qt0; qt < S / BM; ++qt) {
+      const int qbase = qt * BM;
+      __syncthreads();
+      // Stage Q, QT, dO, dOT (swizzled).
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int idx = tid + i * 256;
+        int row = idx >> 4, ch = idx & 15;
+        s16x8 qv = *(const s16x8*)(Qb + (long long)(qbase + row) * q_rowstride + ch * 8);
+        *(s16x8*)((char*)q_lds + swz(row * 256 + ch * 16, row)) = qv;
+        s16x8 dov = *(const s16x8*)(dOb + (long long)(qbase + row) * q_rowstride + ch * 8);
+        *(s16x8*)((char*)do_lds + swz(row * 256 + ch * 16, row)) = dov;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = ch * 8 + j;
+          *(unsigned short*)((char*)qt_lds + swz(d * 128 + row * 2, d)) =
+              (unsigned short)qv[j];
+          *(unsigned short*)((char*)dot_lds + swz(d * 128 + row * 2, d)) =
+              (unsigned short)dov[j];
+        }
+      }
+      __syncthreads();
+
+      // ST = K Q^T (raw);  [kv 16][q 64] per wave.
+      f32x4 st[4], dpt[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        st[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+        dpt[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          int qrow = ct * 16 + lrow;
+          s16x8 bq = *(const s16x8*)((char*)q_lds +
+                                     swz(qrow * 256 + (ks * 32 + lgrp * 8) * 2, qrow));
+          st[ct] = MFMA_BF16(as_bf16x8(a_k[ks]), as_bf16x8(bq), st[ct]);
+          s16x8 bdo = *(const s16x8*)((char*)do_lds +
+                                      swz(qrow * 256 + (ks * 32 + lgrp * 8) * 2, qrow));
+          dpt[ct] = MFMA_BF16(as_bf16x8(a_v[ks]), as_bf16x8(bdo), dpt[ct]);
+        }
+      }
+
+      // PT = exp(scale*ST - lse[q]);  dST = PT * (dPT - Dvec[q]).
+      const int my_kvrow = kvbase + 16 * w + lgrp * 4;  // + r
+      float pt[4][4], dst[4][4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        int qcol = qbase + ct * 16 + lrow;
+        float l = lse_b[qcol];
+        float dv = dvec_b[(long long)qcol * Hq];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float sv = st[ct][r] * scale;
+          float pv;
+          if ((causal && qcol < my_kvrow + r) || l == -INFINITY)
+            pv = 0.f;
+          else
+            pv = __expf(sv - l);
+          pt[ct][r] = pv;
+          dst[ct][r] = pv * (dpt[ct][r] - dv);
+        }
+      }
+
+      // Stage PT -> p_lds [kv][q]; dV += PT * dO (via dOT).
+      __syncthreads();
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int prow = 16 * w + lgrp * 4 + r;
+          int pcol = ct * 16 + lrow;
+          *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+              f2bf(pt[ct][r]);
+        }
+      __syncthreads();
+#pragma unroll
+      for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          int prow = 16 * w + lrow;
+          s16x8 afrag = *(const s16x8*)((char*)p_lds +
+                                        swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
+          int drow = ct * 16 + lrow;
+          s16x8 bfrag = *(const s16x8*)((char*)dot_lds +
+                                        swz(drow * 128 + (ks * 32 + lgrp * 8) * 2, drow));
+          dv_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dv_acc[ct]);
+        }
+
+      // Stage dST -> p_lds [kv][q]; dK += dST * Q (via QT).
+      __syncthreads();
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int prow = 16 * w + lgrp * 4 + r;
+          int pcol = ct * 16 + lrow;
+          *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+              f2bf(dst[ct][r]);
+        }
+      __syncthreads();
+#pragma unroll
+      for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          int prow = 16 * w + lrow;
+          s16x8 afrag = *(const s16x8*)((char*)p_lds +
+                                        swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
+          int qrow2 = ct * 16 + lrow;
+          s16x8 bfrag = *(const s16x8*)((char*)qt_lds +
+                                        swz(qrow2 * 128 + (ks * 32 + lgrp * 8) * 2, qrow2));
+          dk_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dk_acc[ct]);
+        }
+    }
+  }
+
+  // Write dK, dV (bf16), applying scale to dK.
+  unsigned short* dKb = dK + ((long long)b * S * Hkv + kvh) * ATT_D;
+  unsigned short* dVb = dV + ((long long)b * S * Hkv + kvh) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kvrow = kvbase + 16 * w + lgrp * 4 + r;
+    unsigned short* krow = dKb + (long long)kvrow * kv_rowstride;
+    unsigned short* vrow = dVb + (long long)kvrow * kv_rowstride;
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct) {
+      krow[ct * 16 + lrow] = f2bf(dk_acc[ct][r] * scale);
+      vrow[ct * 16 + lrow] = f2bf(dv_acc[ct][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dQ kernel. Grid: (S/BM, B*Hq). LDS: K + KT + V + dS = 56KB.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dQ, int B, int S, int Hq, int Hkv,
+    float scale, int causal) {
+  __shared__ unsigned short k_lds[BN * ATT_D];
+  __shared__ unsigned short kt_lds[ATT_D * BN];
+  __shared__ unsigned short v_lds[BN * ATT_D];
+  __shared__ unsigned short ds_lds[BM * BN];
+
+  const int qt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * BM;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int lrow = lane & 15;
+  const int lgrp = lane >> 4;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+  const float* lse_b = lse + ((long long)b * Hq + qh) * S;
+  const float* dvec_b = Dvec + (long long)b * S * Hq + qh;
+
+  // Fixed A fragments: Q rows and dO rows for this wave.
+  s16x8 a_q[4], a_do[4];
+  float my_lse[4], my_dvec[4];
+  {
+    const int qrow = qbase + 16 * w + lrow;
+    const unsigned short* qsrc = Qb + (long long)qrow * q_rowstride;
+    const unsigned short* dsrc = dOb + (long long)qrow * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      a_q[ks] = *(const s16x8*)(qsrc + ks * 32 + lgrp * 8);
+      a_do[ks] = *(const s16x8*)(dsrc + ks * 32 + lgrp * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int rr = qbase + 16 * w + lgrp * 4 + r;
+      my_lse[r] = lse_b[rr];
+      my_dvec[r] = dvec_b[(long long)rr * Hq];
+    }
+  }
+
+  f32x4 dq_acc[8];
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) dq_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_kv_tiles = causal ? (qbase + BM + BN - 1) / BN : S / BN;
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kvbase = kt * BN;
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int idx = tid + i * 256;
+      int row = idx >> 4, ch = idx & 15;
+      s16x8 kv8 = *(const s16x8*)(Kb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+      *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
+      s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+      *(s16x8*)((char*)v_lds + swz(row * 256 + ch * 16, row)) = vv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = ch * 8 + j;
+        *(unsigned short*)((char*)kt_lds + swz(d * 128 + row * 2, d)) =
+            (unsigned short)kv8[j];
+      }
+    }
+    __syncthreads();
+
+    // S = Q K^T (raw), dP = dO V^T; both [q 16][kv 64] per wave.
+    f32x4 s_acc[4], dp[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+      s_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+      dp[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int krow = ct * 16 + lrow;
+        s16x8 bk = *(const s16x8*)((char*)k_lds +
+                                   swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
+        s_acc[ct] = MFMA_BF16(as_bf16x8(a_q[ks]), as_bf16x8(bk), s_acc[ct]);
+        s16x8 bv = *(const s16x8*)((char*)v_lds +
+                                   swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
+        dp[ct] = MFMA_BF16(as_bf16x8(a_do[ks]), as_bf16x8(bv), dp[ct]);
+      }
+    }
+
+    // dS = P * (dP - Dvec);  P = exp(scale*S - lse).
+    const int my_qrow = qbase + 16 * w + lgrp * 4;
+    __syncthreads();
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int col = kvbase + ct * 16 + lrow;
+        float pv;
+        if ((causal && col > my_qrow + r) || my_lse[r] == -INFINITY)
+          pv = 0.f;
+        else
+          pv = __expf(s_acc[ct][r] * scale - my_lse[r]);
+        float ds = pv * (dp[ct][r] - my_dvec[r]);
+        int prow = 16 * w + lgrp * 4 + r;
+        int pcol = ct * 16 + lrow;
+        *(unsigned short*)((char*)ds_lds + swz(prow * 128 + pcol * 2, prow)) =
+            f2bf(ds);
+      }
+    __syncthreads();
+
+    // dQ += dS K (via KT).
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        int prow = 16 * w + lrow;
+        s16x8 afrag = *(const s16x8*)((char*)ds_lds +
+                                      swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
+        int krow = ct * 16 + lrow;
+        s16x8 bfrag = *(const s16x8*)((char*)kt_lds +
+                                      swz(krow * 128 + (ks * 32 + lgrp * 8) * 2, krow));
+        dq_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dq_acc[ct]);
+      }
+  }
+
+  unsigned short* dQb = dQ + ((long long)b * S * Hq + qh) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = qbase + 16 * w + lgrp * 4 + r;
+    unsigned short* orow = dQb + (long long)qrow * q_rowstride;
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+      orow[ct * 16 + lrow] = f2bf(dq_acc[ct][r] * scale);
+  }
+}
+
+extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
+                                const void* O, const void* dO,
+                                const float* lse, float* Dvec, void* dQ,
+                                void* dK, void* dV, int B, int S, int Hq,
+                                int Hkv, float scale, bool causal,
+                                hipStream_t stream) {
+  long long rows = (long long)B * S * Hq;
+  hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3(membound_grid(rows, 4)),
+                     dim3(256), 0, stream, (const unsigned short*)dO,
+                     (const unsigned short*)O, Dvec, rows, Hq);
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(S / BN, B * Hkv), dim3(256), 0,
+                     stream, (const unsigned short*)Q,
+                     (const unsigned short*)K, (const unsigned short*)V,
+                     (const unsigned short*)dO, lse, Dvec,
+                     (unsigned short*)dK, (unsigned short*)dV, B, S, Hq, Hkv,
+                     scale, causal ? 1 : 0);
+  hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(S / BM, B * Hq), dim3(256), 0,
+                     stream, (const unsigned short*)Q,
+                     (const unsigned short*)K, (const unsigned short*)V,
+                     (const unsigned short*)dO, lse, Dvec,
+                     (unsigned short*)dQ, B, S, Hq, Hkv, scale, causal ? 1 : 0);
+}

@@ -1,0 +1,203 @@
+// Flash-attention forward (causal, GQA, D=128, bf16) for CDNA4/gfx950.
+//
+// MI355X-native design (no reference counterpart — SkyPilot ships no
+// kernels): 64 q-rows per 256-thread block (4 waves,
+// wave w owns q rows [16w,16w+16)), kv tiles of 64, MFMA 16x16x32 bf16.
+// K tile and transposed-V tile live in XOR-swizzled LDS so every MFMA
+// B-fragment is one conflict-free ds_read_b128; P round-trips through a
+// swizzled LDS tile to move from MFMA C-layout to A-layout.  Online
+// softmax (running max/sum) is held in registers; log-sum-exp is written
+// for the backward pass.
+//
+// Layouts: Q,O [B,S,Hq,D]; K,V [B,S,Hkv,D]; lse [B,Hq,S] fp32.
+#include "common.h"
+
+#define ATT_D 128
+#define BM 64  // q rows per block
+#define BN 64  // kv rows per tile
+#define NW 4   // waves per block
+
+extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, unsigned short* __restrict__ O,
+    float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, float scale,
+    int causal) {
+  __shared__ unsigned short k_lds[BN * ATT_D];    // [kv][d], swizzled, 16KB
+  __shared__ unsigned short vt_lds[ATT_D * BN];   // [d][kv], swizzled, 16KB
+  __shared__ unsigned short p_lds[BM * BN];       // [q][kv], swizzled, 8KB
+
+  const int qt = blockIdx.x;            // q tile index
+  const int bh = blockIdx.y;            // b * Hq + qh
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * BM;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;               // wave id: q rows [16w, 16w+16)
+  const int lrow = lane & 15;           // fragment row/col index
+  const int lgrp = lane >> 4;           // fragment k-group
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  // ---- Q fragments for this wave: rows qbase+16w+lrow, 4 k-subtiles.
+  s16x8 a_q[4];
+  {
+    const int qrow = qbase + 16 * w + lrow;
+    const unsigned short* src = Qb + (long long)qrow * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks)
+      a_q[ks] = *(const s16x8*)(src + ks * 32 + lgrp * 8);
+  }
+
+  // ---- online-softmax state: 4 rows per lane (rows lgrp*4 + r).
+  float m_run[4], l_run[4];
+  f32x4 o_acc[8];  // 8 d col-tiles
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -INFINITY;
+    l_run[r] = 0.f;
+  }
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) o_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_kv_tiles = causal ? (qbase + BM + BN - 1) / BN : (S + BN - 1) / BN;
+
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kvbase = kt * BN;
+    // ---- stage K tile [kv][d] (swizzled) + V tile transposed [d][kv].
+    __syncthreads();
+    {
+      // K: thread t loads 16B chunk (row t/16, chunk t%16), 4 rounds.
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int idx = tid + i * 256;          // 1024 chunks total
+        int row = idx >> 4, ch = idx & 15;
+        s16x8 kv8 = *(const s16x8*)(Kb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+        *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
+        // V: same chunk, scattered into transposed layout.
+        s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = ch * 8 + j;
+          *(unsigned short*)((char*)vt_lds + swz(d * 128 + row * 2, d)) =
+              (unsigned short)vv8[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S = scale * Q K^T for this wave's 16 rows x 64 cols.
+    f32x4 s_acc[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+      s_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int krow = ct * 16 + lrow;
+        s16x8 bfrag = *(const s16x8*)((char*)k_lds +
+                                      swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
+        s_acc[ct] = MFMA_BF16(as_bf16x8(a_q[ks]), as_bf16x8(bfrag), s_acc[ct]);
+      }
+    }
+
+    // ---- mask + online softmax.
+    const int my_qrow = qbase + 16 * w + lgrp * 4;  // + r
+    float p[4][4];  // [ct][r]
+    float m_new[4], corr[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = -INFINITY;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        float s = s_acc[ct][r] * scale;
+        int col = kvbase + ct * 16 + lrow;
+        if ((causal && col > my_qrow + r) || col >= S) s = -INFINITY;
+        p[ct][r] = s;
+        mx = fmaxf(mx, s);
+      }
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+      m_new[r] = fmaxf(m_run[r], mx);
+      // All -inf row (above diagonal): keep m at -inf, contribute nothing.
+      float m_safe = (m_new[r] == -INFINITY) ? 0.f : m_new[r];
+      corr[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_safe);
+      float rs = 0.f;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        float e = (p[ct][r] == -INFINITY) ? 0.f : __expf(p[ct][r] - m_safe);
+        p[ct][r] = e;
+        rs += e;
+      }
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) rs += __shfl_xor(rs, off, 64);
+      l_run[r] = l_run[r] * corr[r] + rs;
+      m_run[r] = m_new[r];
+    }
+    // rescale existing O.
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[ct][r] *= corr[r];
+
+    // ---- P -> LDS (bf16, swizzled [q 64][kv 64]).
+    __syncthreads();  // all waves done reading k_lds? (p_lds separate; sync
+                      // protects p_lds reuse across kv iterations)
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int prow = 16 * w + lgrp * 4 + r;
+        int pcol = ct * 16 + lrow;
+        *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+            f2bf(p[ct][r]);
+      }
+    __syncthreads();
+
+    // ---- O += P V : A from p_lds, B from vt_lds.
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        int prow = 16 * w + lrow;
+        s16x8 afrag = *(const s16x8*)((char*)p_lds +
+                                      swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
+        int vrow = ct * 16 + lrow;
+        s16x8 bfrag = *(const s16x8*)((char*)vt_lds +
+                                      swz(vrow * 128 + (ks * 32 + lgrp * 8) * 2, vrow));
+        o_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), o_acc[ct]);
+      }
+    }
+  }
+
+  // ---- epilogue: O /= l, write bf16; lse = m + log(l).
+  unsigned short* Ob = O + ((long long)b * S * Hq + qh) * ATT_D;
+  float* lse_b = lse_out + ((long long)b * Hq + qh) * S;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = qbase + 16 * w + lgrp * 4 + r;
+    float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+    unsigned short* orow = Ob + (long long)qrow * q_rowstride;
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+      orow[ct * 16 + lrow] = f2bf(o_acc[ct][r] * inv_l);
+    if (lrow == 0)
+      lse_b[qrow] = (l_run[r] > 0.f) ? m_run[r] + __logf(l_run[r]) : -INFINITY;
+  }
+}
+
+extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
+                                void* O, float* lse, int B, int S, int Hq,
+                                int Hkv, float scale, bool causal,
+                                hipStream_t stream) {
+  dim3 grid(S / BM, B * Hq);
+  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
+                     (const unsigned short*)Q, (const unsigned short*)K,
+                     (const unsigned short*)V, (unsigned short*)O, lse, B, S,
+                     Hq, Hkv, scale, causal ? 1 : 0);
+}

@@ -1,0 +1,182 @@
+// Python bindings for the skypilot_amd CDNA4 kernels.
+//
+// Compiled with hipcc against libtorch; kernels live in the .hip TUs and
+// are reached through extern "C" launch functions so this TU only needs
+// the HIP runtime + torch headers.
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+#define CHECK_BF16(x) \
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, #x " must be bf16")
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+// ---- extern "C" launchers from the .hip files -----------------------------
+extern "C" {
+void rmsnorm_fwd_launch(const void*, const void*, void*, float*, long long,
+                        int, float, hipStream_t);
+void rmsnorm_bwd_launch(const void*, const void*, const void*, const float*,
+                        void*, float*, long long, int, hipStream_t);
+void rope_launch(const void*, void*, const float*, const float*, const int*,
+                 long long, int, int, bool, hipStream_t);
+void adamw_launch(void*, float*, const void*, float*, float*, long long,
+                  float, float, float, float, float, int, float, hipStream_t);
+void cross_entropy_launch(void*, const int*, float*, long long, int, float,
+                          int, hipStream_t);
+void attn_fwd_launch(const void*, const void*, const void*, void*, float*,
+                     int, int, int, int, float, bool, hipStream_t);
+void attn_bwd_launch(const void*, const void*, const void*, const void*,
+                     const void*, const float*, float*, void*, void*, void*,
+                     int, int, int, int, float, bool, hipStream_t);
+void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
+}
+
+// ---- RMSNorm --------------------------------------------------------------
+torch::Tensor rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                          torch::Tensor inv_rms, double eps) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  CHECK_GPU(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  const int H = (int)x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden dim must be a multiple of 8");
+  long long rows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  rmsnorm_fwd_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     inv_rms.data_ptr<float>(), rows, H, (float)eps,
+                     cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor dy,
+                                       torch::Tensor inv_rms) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(dy);
+  const int H = (int)x.size(-1);
+  long long rows = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dw = torch::zeros({(long)H},
+                         x.options().dtype(at::kFloat));
+  rmsnorm_bwd_launch(x.data_ptr(), w.data_ptr(), dy.data_ptr(),
+                     inv_rms.data_ptr<float>(), dx.data_ptr(),
+                     dw.data_ptr<float>(), rows, H, cur_stream());
+  return {dx, dw};
+}
+
+// ---- RoPE -----------------------------------------------------------------
+torch::Tensor rope(torch::Tensor x, torch::Tensor cos_tab,
+                   torch::Tensor sin_tab, torch::Tensor positions,
+                   bool backward) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  TORCH_CHECK(positions.scalar_type() == at::kInt);
+  const int D = (int)x.size(-1);
+  const int H = (int)x.size(-2);
+  long long n_tokens = x.numel() / ((long long)H * D);
+  auto y = torch::empty_like(x);
+  rope_launch(x.data_ptr(), y.data_ptr(), cos_tab.data_ptr<float>(),
+              sin_tab.data_ptr<float>(), positions.data_ptr<int>(), n_tokens,
+              H, D, backward, cur_stream());
+  return y;
+}
+
+// ---- AdamW ----------------------------------------------------------------
+void adamw_step(std::vector<torch::Tensor> params_bf16,
+                std::vector<torch::Tensor> params_master,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> exp_avg,
+                std::vector<torch::Tensor> exp_avg_sq, double lr, double beta1,
+                double beta2, double eps, double weight_decay, int64_t step,
+                double grad_scale, std::vector<bool> decay_mask) {
+  auto stream = cur_stream();
+  for (size_t i = 0; i < params_bf16.size(); ++i) {
+    auto& p = params_bf16[i];
+    long long n = p.numel();
+    float wd = decay_mask.empty() || decay_mask[i] ? (float)weight_decay : 0.f;
+    adamw_launch(p.data_ptr(), params_master[i].data_ptr<float>(),
+                 grads[i].data_ptr(), exp_avg[i].data_ptr<float>(),
+                 exp_avg_sq[i].data_ptr<float>(), n, (float)lr, (float)beta1,
+                 (float)beta2, (float)eps, wd, (int)step, (float)grad_scale,
+                 stream);
+  }
+}
+
+// ---- Cross entropy --------------------------------------------------------
+torch::Tensor cross_entropy_fused(torch::Tensor logits, torch::Tensor targets,
+                                  double grad_scale, int64_t ignore_index) {
+  CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+  TORCH_CHECK(targets.scalar_type() == at::kInt);
+  const int V = (int)logits.size(-1);
+  long long N = logits.numel() / V;
+  auto loss = torch::empty({(long)N}, logits.options().dtype(at::kFloat));
+  cross_entropy_launch(logits.data_ptr(), targets.data_ptr<int>(),
+                       loss.data_ptr<float>(), N, V, (float)grad_scale,
+                       (int)ignore_index, cur_stream());
+  return loss;
+}
+
+// ---- Attention ------------------------------------------------------------
+std::vector<torch::Tensor> attn_fwd(torch::Tensor Q, torch::Tensor K,
+                                    torch::Tensor V, double scale,
+                                    bool causal) {
+  CHECK_GPU(Q); CHECK_CONTIG(Q); CHECK_CONTIG(K); CHECK_CONTIG(V);
+  CHECK_BF16(Q);
+  const int B = (int)Q.size(0), S = (int)Q.size(1), Hq = (int)Q.size(2),
+            D = (int)Q.size(3);
+  const int Hkv = (int)K.size(2);
+  TORCH_CHECK(D == 128, "attention kernel requires head_dim=128");
+  TORCH_CHECK(S % 64 == 0, "seq len must be a multiple of 64");
+  TORCH_CHECK(Hq % Hkv == 0);
+  auto O = torch::empty_like(Q);
+  auto lse = torch::empty({B, Hq, S}, Q.options().dtype(at::kFloat));
+  attn_fwd_launch(Q.data_ptr(), K.data_ptr(), V.data_ptr(), O.data_ptr(),
+                  lse.data_ptr<float>(), B, S, Hq, Hkv, (float)scale, causal,
+                  cur_stream());
+  return {O, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor Q, torch::Tensor K,
+                                    torch::Tensor V, torch::Tensor O,
+                                    torch::Tensor dO, torch::Tensor lse,
+                                    double scale, bool causal) {
+  CHECK_GPU(Q); CHECK_CONTIG(dO);
+  const int B = (int)Q.size(0), S = (int)Q.size(1), Hq = (int)Q.size(2);
+  const int Hkv = (int)K.size(2);
+  auto dQ = torch::empty_like(Q);
+  auto dK = torch::empty_like(K);
+  auto dV = torch::empty_like(V);
+  auto Dvec = torch::empty({(long)B * S * Hq}, Q.options().dtype(at::kFloat));
+  attn_bwd_launch(Q.data_ptr(), K.data_ptr(), V.data_ptr(), O.data_ptr(),
+                  dO.data_ptr(), lse.data_ptr<float>(),
+                  Dvec.data_ptr<float>(), dQ.data_ptr(), dK.data_ptr(),
+                  dV.data_ptr(), B, S, Hq, Hkv, (float)scale, causal,
+                  cur_stream());
+  return {dQ, dK, dV};
+}
+
+// ---- MFMA layout probe (used by tests/test_gpu_mfma.py) -------------------
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  CHECK_GPU(A); CHECK_BF16(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
+  TORCH_CHECK(A.size(0) == 16 && A.size(1) == 32);
+  TORCH_CHECK(B.size(0) == 32 && B.size(1) == 16);
+  auto C = torch::zeros({16, 16}, A.options().dtype(at::kFloat));
+  mfma_probe_launch(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(),
+                    cur_stream());
+  return C;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope", &rope);
+  m.def("adamw_step", &adamw_step);
+  m.def("cross_entropy_fused", &cross_entropy_fused);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("mfma_probe", &mfma_probe);
+}

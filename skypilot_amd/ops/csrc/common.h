@@ -1,0 +1,144 @@
+// Common device helpers for skypilot_amd CDNA4 (gfx950) kernels.
+//
+// All kernels in this tree are written directly for MI355X: wave64,
+// 32-bank LDS, MFMA 16x16x32 bf16 tiles. No CUDA compatibility paths.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define WAVE 64
+
+// ---------------------------------------------------------------------------
+// Vector types (ext_vector_type so hipcc maps them onto VGPR tuples).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(2))) float  f32x2;
+typedef __attribute__((ext_vector_type(4))) float  f32x4;
+typedef __attribute__((ext_vector_type(8))) float  f32x8;
+typedef __attribute__((ext_vector_type(4))) short  s16x4;
+typedef __attribute__((ext_vector_type(8))) short  s16x8;
+typedef __attribute__((ext_vector_type(2))) short  s16x2;
+typedef __attribute__((ext_vector_type(4))) int    i32x4;
+typedef __attribute__((ext_vector_type(2))) int    i32x2;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+union bf8_cast {
+  s16x8 s;
+  bf16x8 b;
+};
+__device__ __forceinline__ bf16x8 as_bf16x8(s16x8 v) {
+  bf8_cast c;
+  c.s = v;
+  return c.b;
+}
+
+// MFMA 16x16x32 bf16 fragment layouts on gfx950 (verified on-device by
+// tests/test_gpu_mfma.py probe):
+//   A[16x32]: lane l holds A[l&15][(l>>4)*8 + j]        j=0..7
+//   B[32x16]: lane l holds B[(l>>4)*8 + j][l&15]
+//   C/D[16x16]: lane l reg r holds C[(l>>4)*4 + r][l&15]
+#define MFMA_BF16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
+
+// XOR swizzle for row-major LDS tiles with 128B/256B row stride: spreads a
+// column slice across 8 16B slots so wave-wide ds_read_b128 at fixed column
+// is ~conflict-free (guide §6 Guideline 4).
+__device__ __forceinline__ int swz(int byte_off, int row) {
+  return byte_off ^ ((row & 7) << 4);
+}
+
+// bf16 <-> f32 via bit ops (we deliberately avoid __hip_bfloat16 so these
+// headers stay independent of HIP half/bf16 operator macros).
+__device__ __forceinline__ float bf2f(unsigned short h) {
+  union { unsigned int u; float f; } v;
+  v.u = ((unsigned int)h) << 16;
+  return v.f;
+}
+
+__device__ __forceinline__ unsigned short f2bf(float f) {
+  union { float f; unsigned int u; } v;
+  v.f = f;
+  unsigned int u = v.u;
+  // NaN -> canonical bf16 NaN
+  if ((u & 0x7fffffffu) > 0x7f800000u) return 0x7fc0;
+  // round-to-nearest-even
+  unsigned int round = 0x7fffu + ((u >> 16) & 1u);
+  return (unsigned short)((u + round) >> 16);
+}
+
+// Convert 8 bf16 (as s16x8) to 8 floats.
+__device__ __forceinline__ void bf8_to_f32(const s16x8 v, float* out) {
+#pragma unroll
+  for (int i = 0; i < 8; ++i) out[i] = bf2f((unsigned short)v[i]);
+}
+
+__device__ __forceinline__ s16x8 f32_to_bf8(const float* in) {
+  s16x8 v;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) v[i] = (short)f2bf(in[i]);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// Wave-level reductions (64-wide).
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+
+// Block-level reduce for blocks of up to 1024 threads (multiple of 64).
+// Returns the reduced value on every thread. `lds` must hold >= 16 floats.
+__device__ __forceinline__ float block_reduce_sum(float x, float* lds) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int nwaves = blockDim.x / WAVE;
+  x = wave_reduce_sum(x);
+  if (lane == 0) lds[wid] = x;
+  __syncthreads();
+  float r = (lane < nwaves) ? lds[lane] : 0.f;
+  r = wave_reduce_sum(r);  // cheap: only first nwaves lanes nonzero
+  __syncthreads();
+  return r;
+}
+
+__device__ __forceinline__ float block_reduce_max(float x, float* lds) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int nwaves = blockDim.x / WAVE;
+  x = wave_reduce_max(x);
+  if (lane == 0) lds[wid] = x;
+  __syncthreads();
+  float r = (lane < nwaves) ? lds[lane] : -INFINITY;
+  r = wave_reduce_max(r);
+  __syncthreads();
+  return r;
+}
+
+// ---------------------------------------------------------------------------
+// Grid sizing helper: memory-bound kernels cap the grid and grid-stride.
+// 256 CUs x 8 blocks/CU (see MI355X guide, Guideline 11).
+// ---------------------------------------------------------------------------
+#define MAX_MEMBOUND_BLOCKS 2048
+
+__host__ __forceinline__ int membound_grid(long long work_items, int block) {
+  long long g = (work_items + block - 1) / block;
+  if (g > MAX_MEMBOUND_BLOCKS) g = MAX_MEMBOUND_BLOCKS;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess) {                                                 \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e), __FILE__,    \
+             __LINE__);                                                     \
+    }                                                                       \
+  } while (0)

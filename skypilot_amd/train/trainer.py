@@ -1,0 +1,117 @@
+"""Bundled Llama trainer — the flagship training entrypoint.
+
+This is what `bench.py` and the `sky launch` training task YAML run: one
+process per GPU, bf16 Llama on the fused CDNA4 kernels, bucketed
+all-reduce DDP over RCCL/xGMI, fused AdamW with fp32 master state.
+Synthetic data (there is no network on the pool): random tokens of the
+benchmark shape, random-init weights.
+"""
+from __future__ import annotations
+
+import os
+import time
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+from skypilot_amd.models.llama import build_model
+from skypilot_amd.parallel.ddp import BucketedDDP
+from skypilot_amd.train.optim import FusedAdamW
+
+
+@dataclass
+class TrainConfig:
+    model: str = "llama3-8b"
+    micro_batch: int = 4
+    seq_len: int = 4096
+    lr: float = 3e-4
+    weight_decay: float = 0.1
+    seed: int = 1234
+    bucket_mb: int = 64
+    device: str = "cuda"
+
+
+def dist_env():
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    return rank, world, local_rank
+
+
+def setup_distributed(backend: str | None = None):
+    rank, world, local_rank = dist_env()
+    if world > 1 and not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(backend, rank=rank, world_size=world)
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+    return rank, world, local_rank
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig):
+        self.cfg = cfg
+        self.rank, self.world, self.local_rank = dist_env()
+        dev = cfg.device
+        if dev == "cuda":
+            dev = f"cuda:{self.local_rank}"
+        self.device = torch.device(dev)
+        self.model = build_model(cfg.model, device=str(self.device),
+                                 dtype=torch.bfloat16, seed=cfg.seed)
+        self.ddp = BucketedDDP(self.model,
+                               bucket_bytes=cfg.bucket_mb << 20)
+        self.opt = FusedAdamW(self.model.parameters(), lr=cfg.lr,
+                              weight_decay=cfg.weight_decay)
+        self.step_count = 0
+        g = torch.Generator(device="cpu").manual_seed(cfg.seed + self.rank)
+        self._gen = g
+
+    def synthetic_batch(self):
+        c = self.cfg
+        vocab = self.model.cfg.vocab_size
+        tok = torch.randint(0, vocab, (c.micro_batch, c.seq_len + 1),
+                            generator=self._gen)
+        tok = tok.to(self.device, non_blocking=True)
+        return tok[:, :-1].contiguous(), tok[:, 1:].contiguous()
+
+    def train_step(self, batch=None) -> float:
+        if batch is None:
+            batch = self.synthetic_batch()
+        tokens, targets = batch
+        self.ddp.zero_grad()
+        self.ddp.mark_step_start()
+        loss = self.model.loss(tokens, targets)
+        loss.backward()
+        self.ddp.finish()
+        self.opt.step(grad_scale=self.ddp.grad_scale)
+        self.step_count += 1
+        return float(loss.detach())
+
+    def tokens_per_step(self) -> int:
+        return self.cfg.micro_batch * self.cfg.seq_len * self.world
+
+
+def run_training(cfg: TrainConfig, steps: int, warmup: int = 2,
+                 log_every: int = 1):
+    """Used by the bundled train task; prints throughput per step."""
+    rank, world, _ = setup_distributed()
+    tr = Trainer(cfg)
+    for i in range(warmup):
+        tr.train_step()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(steps):
+        loss = tr.train_step()
+        if rank == 0 and (i + 1) % log_every == 0:
+            dt = time.perf_counter() - t0
+            tps = tr.tokens_per_step() * (i + 1) / dt
+            print(f"step {i+1}/{steps} loss {loss:.4f} "
+                  f"tokens/s {tps:,.0f}", flush=True)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    return tr

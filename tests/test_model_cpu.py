@@ -1,0 +1,96 @@
+"""CPU tests for the model / trainer / DDP stack (reference op paths)."""
+import os
+
+import pytest
+import torch
+
+from skypilot_amd.models.llama import CONFIGS, build_model
+from skypilot_amd.train.optim import FusedAdamW
+
+
+def test_model_forward_shapes():
+    m = build_model("llama-debug", dtype=torch.float32)
+    tok = torch.randint(0, 512, (2, 64))
+    logits = m(tok)
+    assert logits.shape == (2, 64, 512)
+
+
+def test_model_loss_decreases_cpu():
+    torch.manual_seed(0)
+    m = build_model("llama-debug", dtype=torch.float32)
+    opt = FusedAdamW(m.parameters(), lr=1e-3)
+    tok = torch.randint(0, 512, (2, 65))
+    inputs, targets = tok[:, :-1], tok[:, 1:].contiguous()
+    losses = []
+    for _ in range(6):
+        opt.zero_grad()
+        loss = m.loss(inputs, targets)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+
+
+def test_flops_per_token_sane():
+    m = build_model("llama-debug", dtype=torch.float32)
+    f = m.flops_per_token(128)
+    assert f > 6 * m.num_params()
+
+
+def test_config_inventory():
+    assert "llama3-8b" in CONFIGS and "llama3-70b" in CONFIGS
+    c = CONFIGS["llama3-8b"]
+    assert c.hidden_size == 4096 and c.num_layers == 32
+    assert c.num_kv_heads == 8 and c.vocab_size == 128256
+
+
+def _ddp_worker(rank, world, port):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(100 + rank)  # different init per rank
+        from skypilot_amd.parallel.ddp import BucketedDDP
+        m = build_model("llama-debug", dtype=torch.float32, seed=rank)
+        ddp = BucketedDDP(m, bucket_bytes=1 << 20)
+        torch.manual_seed(7 + rank)
+        tok = torch.randint(0, 512, (1, 33))
+        ddp.zero_grad()
+        ddp.mark_step_start()
+        loss = m.loss(tok[:, :-1], tok[:, 1:].contiguous())
+        loss.backward()
+        ddp.finish()
+        # After broadcast at init, params identical; grads summed: verify a
+        # grad is identical across ranks by all-gathering a slice.
+        g = m.lm_head.weight._sky_grad.flatten()[:128].clone()
+        gather = [torch.zeros_like(g) for _ in range(world)]
+        dist.all_gather(gather, g)
+        assert torch.allclose(gather[0], gather[1], atol=1e-6)
+        opt = FusedAdamW(m.parameters(), lr=1e-3)
+        opt.step(grad_scale=ddp.grad_scale)
+        p = m.lm_head.weight.flatten()[:128].clone()
+        gather2 = [torch.zeros_like(p) for _ in range(world)]
+        dist.all_gather(gather2, p)
+        assert torch.allclose(gather2[0], gather2[1], atol=1e-6)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_ddp_two_process_gloo():
+    import multiprocessing as mp
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_ddp_worker, args=(r, 2, port))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    for p in procs:
+        assert p.exitcode == 0
