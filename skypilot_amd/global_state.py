@@ -1,0 +1,184 @@
+"""Global user state — SQLite-backed cluster/request/event tables.
+
+Reference: sky/global_user_state.py (tables clusters, cluster_history,
+cluster_events; handle pickled into the cluster row :1456).  We store the
+handle as JSON instead of pickle (no cross-version pickle compat burden)
+and keep the event-history table that powers job-start-latency
+measurement (reference: global_user_state.py:1001 add_cluster_event).
+"""
+from __future__ import annotations
+
+import json
+import os
+import sqlite3
+import threading
+import time
+from pathlib import Path
+from typing import Any, Dict, List, Optional
+
+_DB_LOCK = threading.Lock()
+
+
+def root_dir() -> Path:
+    d = Path(os.environ.get("SKY_AMD_HOME", "~/.sky_amd")).expanduser()
+    d.mkdir(parents=True, exist_ok=True)
+    return d
+
+
+def _db_path() -> Path:
+    return root_dir() / "state.db"
+
+
+_SCHEMA = """
+CREATE TABLE IF NOT EXISTS clusters (
+    name TEXT PRIMARY KEY,
+    status TEXT NOT NULL,
+    handle TEXT NOT NULL,
+    resources TEXT NOT NULL,
+    launched_at REAL,
+    last_use TEXT,
+    autostop_idle_minutes INTEGER DEFAULT -1,
+    autostop_down INTEGER DEFAULT 0,
+    to_down INTEGER DEFAULT 0
+);
+CREATE TABLE IF NOT EXISTS cluster_events (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    cluster TEXT NOT NULL,
+    ts REAL NOT NULL,
+    event TEXT NOT NULL,
+    detail TEXT
+);
+CREATE TABLE IF NOT EXISTS cluster_history (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    name TEXT NOT NULL,
+    launched_at REAL,
+    torn_down_at REAL,
+    resources TEXT
+);
+CREATE TABLE IF NOT EXISTS storage (
+    name TEXT PRIMARY KEY,
+    source TEXT,
+    store_type TEXT,
+    created_at REAL
+);
+CREATE TABLE IF NOT EXISTS config_kv (
+    key TEXT PRIMARY KEY,
+    value TEXT
+);
+"""
+
+# Cluster status values (reference: sky/utils/status_lib.ClusterStatus).
+INIT = "INIT"
+UP = "UP"
+STOPPED = "STOPPED"
+
+
+def _conn() -> sqlite3.Connection:
+    conn = sqlite3.connect(_db_path(), timeout=30)
+    conn.execute("PRAGMA journal_mode=WAL")
+    conn.executescript(_SCHEMA)
+    return conn
+
+
+def add_or_update_cluster(name: str, status: str, handle: Dict[str, Any],
+                          resources: Dict[str, Any],
+                          launched_at: Optional[float] = None) -> None:
+    with _DB_LOCK, _conn() as c:
+        existing = c.execute("SELECT launched_at FROM clusters WHERE name=?",
+                             (name,)).fetchone()
+        if launched_at is None:
+            launched_at = existing[0] if existing else time.time()
+        c.execute(
+            "INSERT INTO clusters (name,status,handle,resources,launched_at)"
+            " VALUES (?,?,?,?,?) ON CONFLICT(name) DO UPDATE SET "
+            "status=excluded.status, handle=excluded.handle, "
+            "resources=excluded.resources, launched_at=excluded.launched_at",
+            (name, status, json.dumps(handle), json.dumps(resources),
+             launched_at))
+
+
+def set_cluster_status(name: str, status: str) -> None:
+    with _DB_LOCK, _conn() as c:
+        c.execute("UPDATE clusters SET status=? WHERE name=?", (status, name))
+
+
+def set_cluster_autostop(name: str, idle_minutes: int, down: bool) -> None:
+    with _DB_LOCK, _conn() as c:
+        c.execute(
+            "UPDATE clusters SET autostop_idle_minutes=?, autostop_down=? "
+            "WHERE name=?", (idle_minutes, int(down), name))
+
+
+def get_cluster(name: str) -> Optional[Dict[str, Any]]:
+    with _DB_LOCK, _conn() as c:
+        row = c.execute(
+            "SELECT name,status,handle,resources,launched_at,"
+            "autostop_idle_minutes,autostop_down FROM clusters WHERE name=?",
+            (name,)).fetchone()
+    if row is None:
+        return None
+    return _row_to_cluster(row)
+
+
+def _row_to_cluster(row) -> Dict[str, Any]:
+    return {
+        "name": row[0],
+        "status": row[1],
+        "handle": json.loads(row[2]),
+        "resources": json.loads(row[3]),
+        "launched_at": row[4],
+        "autostop_idle_minutes": row[5],
+        "autostop_down": bool(row[6]),
+    }
+
+
+def list_clusters() -> List[Dict[str, Any]]:
+    with _DB_LOCK, _conn() as c:
+        rows = c.execute(
+            "SELECT name,status,handle,resources,launched_at,"
+            "autostop_idle_minutes,autostop_down FROM clusters "
+            "ORDER BY launched_at DESC").fetchall()
+    return [_row_to_cluster(r) for r in rows]
+
+
+def remove_cluster(name: str) -> None:
+    with _DB_LOCK, _conn() as c:
+        row = c.execute(
+            "SELECT launched_at, resources FROM clusters WHERE name=?",
+            (name,)).fetchone()
+        if row:
+            c.execute(
+                "INSERT INTO cluster_history "
+                "(name,launched_at,torn_down_at,resources) VALUES (?,?,?,?)",
+                (name, row[0], time.time(), row[1]))
+        c.execute("DELETE FROM clusters WHERE name=?", (name,))
+
+
+def add_cluster_event(cluster: str, event: str, detail: str = "") -> None:
+    with _DB_LOCK, _conn() as c:
+        c.execute(
+            "INSERT INTO cluster_events (cluster,ts,event,detail) "
+            "VALUES (?,?,?,?)", (cluster, time.time(), event, detail))
+
+
+def get_cluster_events(cluster: str) -> List[Dict[str, Any]]:
+    with _DB_LOCK, _conn() as c:
+        rows = c.execute(
+            "SELECT ts,event,detail FROM cluster_events WHERE cluster=? "
+            "ORDER BY ts", (cluster,)).fetchall()
+    return [{"ts": r[0], "event": r[1], "detail": r[2]} for r in rows]
+
+
+def set_config(key: str, value: Any) -> None:
+    with _DB_LOCK, _conn() as c:
+        c.execute(
+            "INSERT INTO config_kv (key,value) VALUES (?,?) "
+            "ON CONFLICT(key) DO UPDATE SET value=excluded.value",
+            (key, json.dumps(value)))
+
+
+def get_config(key: str, default: Any = None) -> Any:
+    with _DB_LOCK, _conn() as c:
+        row = c.execute("SELECT value FROM config_kv WHERE key=?",
+                        (key,)).fetchone()
+    return json.loads(row[0]) if row else default
