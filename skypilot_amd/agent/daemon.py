@@ -1,0 +1,212 @@
+"""Node agent daemon (skylet equivalent) — HTTP control plane per cluster.
+
+Reference: sky/skylet/skylet.py (gRPC daemon, port 46590) + services.py.
+The RPC surface mirrors the skylet protos (SURVEY.md Appendix A:
+AddJob/QueueJob/GetJobQueue/CancelJobs/TailLogs/GetJobStatus +
+SetAutostop/IsAutostopping) as JSON-over-HTTP on a per-cluster local
+port; SSH pools tunnel the same port.  A scheduler thread runs the
+GPU-aware FIFO queue (reference: job_lib.py:278 JobScheduler) and an
+autostop event loop mirrors skylet/events.py:268 StopEvent.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import subprocess
+import sys
+import threading
+import time
+from pathlib import Path
+
+import uvicorn
+from fastapi import FastAPI
+from fastapi.responses import StreamingResponse
+
+from skypilot_amd.agent import job_lib
+
+SCHEDULER_INTERVAL = 1.0
+AUTOSTOP_INTERVAL = 30.0
+
+
+def create_app(cluster_dir: str, gpu_ids: list[int]) -> FastAPI:
+    app = FastAPI()
+    table = job_lib.JobTable(cluster_dir)
+    state = {"autostop_idle_minutes": -1, "autostop_down": False,
+             "last_active": time.time(), "autostopping": False,
+             "gpu_ids": gpu_ids}
+
+    # ---- scheduler thread (FIFO + GPU accounting) -------------------------
+    def scheduler_loop():
+        while True:
+            try:
+                table.reconcile()
+                free = [g for g in state["gpu_ids"]
+                        if g not in table.allocated_gpus()]
+                for job in table.pending_jobs():
+                    spec = job["spec"]
+                    need = int(spec.get("num_nodes", 1)) * \
+                        int(spec.get("gpus_per_node", 0))
+                    if need > len(free):
+                        break  # strict FIFO (reference FIFOScheduler)
+                    gpu_ids_alloc = free[:need]
+                    free = free[need:]
+                    spec["gpu_ids"] = gpu_ids_alloc
+                    with table._conn() as c:
+                        c.execute("UPDATE jobs SET spec=?, status=? "
+                                  "WHERE job_id=? AND status='PENDING'",
+                                  (json.dumps(spec), job_lib.INIT,
+                                   job["job_id"]))
+                    dlog = open(Path(cluster_dir) / "driver.log", "ab")
+                    subprocess.Popen(
+                        [sys.executable, "-m", "skypilot_amd.agent.driver",
+                         cluster_dir, str(job["job_id"])],
+                        stdout=dlog, stderr=subprocess.STDOUT,
+                        start_new_session=True)
+                    dlog.close()
+                    state["last_active"] = time.time()
+            except Exception as e:  # noqa: BLE001
+                print(f"scheduler error: {e}", file=sys.stderr)
+            time.sleep(SCHEDULER_INTERVAL)
+
+    threading.Thread(target=scheduler_loop, daemon=True).start()
+
+    # ---- autostop loop (reference: skylet/events.py StopEvent:268) --------
+    def autostop_loop():
+        while True:
+            time.sleep(AUTOSTOP_INTERVAL)
+            try:
+                idle_min = state["autostop_idle_minutes"]
+                if idle_min < 0:
+                    continue
+                if not table.is_idle():
+                    state["last_active"] = time.time()
+                    continue
+                idle_for = (time.time() - state["last_active"]) / 60.0
+                if idle_for >= idle_min:
+                    state["autostopping"] = True
+                    flag = Path(cluster_dir) / "autostop_triggered"
+                    flag.write_text(
+                        json.dumps({"down": state["autostop_down"],
+                                    "ts": time.time()}))
+            except Exception as e:  # noqa: BLE001
+                print(f"autostop error: {e}", file=sys.stderr)
+
+    threading.Thread(target=autostop_loop, daemon=True).start()
+
+    # ---- RPC surface ------------------------------------------------------
+    @app.get("/health")
+    def health():
+        return {"ok": True, "cluster_dir": cluster_dir,
+                "gpu_ids": state["gpu_ids"], "pid": os.getpid()}
+
+    @app.post("/jobs/queue")
+    def queue_job(body: dict):
+        job_id = table.add_job(body.get("name"), body["spec"])
+        state["last_active"] = time.time()
+        return {"job_id": job_id}
+
+    @app.get("/jobs")
+    def get_job_queue():
+        return {"jobs": table.list()}
+
+    @app.get("/jobs/{job_id}")
+    def get_job(job_id: int):
+        j = table.get(job_id)
+        return {"job": j}
+
+    @app.post("/jobs/{job_id}/cancel")
+    def cancel(job_id: int):
+        return {"cancelled": table.cancel(job_id)}
+
+    @app.post("/jobs/cancel_all")
+    def cancel_all():
+        n = 0
+        for j in table.list():
+            if j["status"] in job_lib.NONTERMINAL:
+                table.cancel(j["job_id"])
+                n += 1
+        return {"cancelled": n}
+
+    @app.get("/jobs/{job_id}/logs")
+    def tail_logs(job_id: int, follow: bool = True, tail: int = 0):
+        j = table.get(job_id)
+        if j is None:
+            return StreamingResponse(iter(()), media_type="text/plain")
+
+        def stream():
+            log_dir = None
+            # wait for driver to create the log dir
+            for _ in range(300):
+                jj = table.get(job_id)
+                log_dir = jj.get("log_dir")
+                if log_dir or jj["status"] in job_lib.TERMINAL:
+                    break
+                time.sleep(0.2)
+            if not log_dir:
+                yield b"(no logs)\n"
+                return
+            def log_files():
+                d = Path(log_dir)
+                files = [d / "setup.log", d / "run.log"]
+                files += sorted(d.glob("*-node.log"))
+                return [f for f in files if f.exists()]
+
+            for _ in range(300):
+                if log_files():
+                    break
+                jj = table.get(job_id)
+                if jj["status"] in job_lib.TERMINAL:
+                    break
+                time.sleep(0.2)
+            for lf in log_files():
+                with open(lf, "rb") as f:
+                    while True:
+                        chunk = f.read(65536)
+                        if chunk:
+                            yield chunk
+                            continue
+                        jj = table.get(job_id)
+                        if not follow or jj["status"] in job_lib.TERMINAL:
+                            break
+                        time.sleep(0.3)
+
+        return StreamingResponse(stream(), media_type="text/plain")
+
+    @app.post("/autostop")
+    def set_autostop(body: dict):
+        state["autostop_idle_minutes"] = int(body.get("idle_minutes", -1))
+        state["autostop_down"] = bool(body.get("down", False))
+        state["last_active"] = time.time()
+        return {"ok": True}
+
+    @app.get("/autostop")
+    def is_autostopping():
+        return {"autostopping": state["autostopping"],
+                "idle_minutes": state["autostop_idle_minutes"],
+                "down": state["autostop_down"]}
+
+    @app.get("/idle")
+    def idle():
+        return {"idle": table.is_idle()}
+
+    return app
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--cluster-dir", required=True)
+    ap.add_argument("--port", type=int, required=True)
+    ap.add_argument("--gpu-ids", default="")
+    args = ap.parse_args()
+    gpu_ids = [int(g) for g in args.gpu_ids.split(",") if g != ""]
+    app = create_app(args.cluster_dir, gpu_ids)
+    # Record readiness for the provisioner.
+    Path(args.cluster_dir).mkdir(parents=True, exist_ok=True)
+    (Path(args.cluster_dir) / "agent.json").write_text(
+        json.dumps({"port": args.port, "pid": os.getpid()}))
+    uvicorn.run(app, host="127.0.0.1", port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

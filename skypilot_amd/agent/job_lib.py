@@ -140,6 +140,9 @@ class JobTable:
         j = self.get(job_id)
         if j is None or j["status"] in TERMINAL:
             return False
+        # Mark first so the driver's final-status write sees CANCELLED
+        # (kill-then-mark would race the driver's own FAILED write).
+        self.set_status(job_id, CANCELLED)
         pid = j.get("driver_pid")
         if pid:
             try:
@@ -149,7 +152,6 @@ class JobTable:
                     os.kill(pid, signal.SIGTERM)
                 except ProcessLookupError:
                     pass
-        self.set_status(job_id, CANCELLED)
         return True
 
     def reconcile(self):

@@ -1,0 +1,310 @@
+"""`sky` CLI (reference: sky/client/cli/command.py — launch:1318,
+exec:1563, status:2090, jobs_launch:5787, serve_up:7282, api_start:7912).
+
+Run as `python -m skypilot_amd.cli ...` or via the repo's `bin/sky`.
+"""
+from __future__ import annotations
+
+import json
+import sys
+from typing import Optional
+
+import click
+
+from skypilot_amd.client import sdk
+from skypilot_amd.task import Task
+
+
+def _load_task(entrypoint: str, env: tuple, overrides: dict) -> Task:
+    """YAML file -> Task, with CLI overrides merged (reference:
+    _make_task_or_dag_from_entrypoint_with_overrides, command.py:945)."""
+    import yaml as _yaml
+    if entrypoint.endswith((".yaml", ".yml")):
+        with open(entrypoint) as f:
+            cfg = _yaml.safe_load(f) or {}
+    else:
+        cfg = {"run": entrypoint}
+    env_overrides = dict(kv.split("=", 1) for kv in env)
+    res = cfg.setdefault("resources", {}) or {}
+    for k, v in overrides.items():
+        if v is not None:
+            res[k] = v
+    cfg["resources"] = res
+    if overrides.get("_num_nodes"):
+        cfg["num_nodes"] = overrides["_num_nodes"]
+    cfg.pop("_num_nodes", None)
+    res.pop("_num_nodes", None)
+    return Task.from_yaml_config(cfg, env_overrides)
+
+
+def _print_result(result):
+    click.echo(json.dumps(result, indent=2, default=str))
+
+
+@click.group()
+def cli():
+    """skypilot-amd: run tasks on an MI355X pool."""
+
+
+# ---------------------------------------------------------------------------
+@cli.command()
+@click.argument("entrypoint")
+@click.option("--cluster", "-c", default=None)
+@click.option("--gpus", default=None, help='e.g. "MI355X:8"')
+@click.option("--num-nodes", type=int, default=None)
+@click.option("--env", multiple=True, help="KEY=VALUE")
+@click.option("--down", is_flag=True, help="tear down after the job")
+@click.option("--idle-minutes-to-autostop", "-i", type=int, default=None)
+@click.option("--detach-run", "-d", is_flag=True,
+              help="don't stream logs after submit")
+@click.option("--async", "async_", is_flag=True,
+              help="print request id and return")
+def launch(entrypoint, cluster, gpus, num_nodes, env, down,
+           idle_minutes_to_autostop, detach_run, async_):
+    """Launch a task (provision + sync + setup + run)."""
+    task = _load_task(entrypoint, env,
+                      {"accelerators": gpus, "_num_nodes": num_nodes})
+    rid = sdk.launch(task, cluster, down=down,
+                     idle_minutes_to_autostop=idle_minutes_to_autostop)
+    if async_:
+        click.echo(rid)
+        return
+    result = sdk.stream_and_get(rid)
+    click.echo(f"Job submitted: cluster={result.get('cluster_name')} "
+               f"job_id={result.get('job_id')}")
+    if not detach_run and result.get("job_id") is not None:
+        sdk.tail_logs(result.get("cluster_name") or cluster,
+                      result["job_id"])
+
+
+@cli.command("exec")
+@click.argument("cluster")
+@click.argument("entrypoint")
+@click.option("--env", multiple=True)
+@click.option("--gpus", default=None)
+@click.option("--async", "async_", is_flag=True)
+def exec_cmd(cluster, entrypoint, env, gpus, async_):
+    """Run a task on an existing cluster (no provision/setup)."""
+    task = _load_task(entrypoint, env, {"accelerators": gpus})
+    rid = sdk.exec(task, cluster)
+    if async_:
+        click.echo(rid)
+        return
+    result = sdk.get(rid)
+    click.echo(f"Job submitted: job_id={result.get('job_id')}")
+    sdk.tail_logs(cluster, result["job_id"])
+
+
+@cli.command()
+@click.option("--refresh", "-r", is_flag=True)
+def status(refresh):
+    """Show clusters."""
+    records = sdk.get(sdk.status(refresh=refresh))
+    if not records:
+        click.echo("No existing clusters.")
+        return
+    fmt = "{:<16} {:<9} {:<22} {:<8}"
+    click.echo(fmt.format("NAME", "STATUS", "RESOURCES", "GPUS"))
+    for r in records:
+        h = r["handle"]
+        res = f"{h.get('num_nodes', 1)}x local"
+        if h.get("gpus_per_node"):
+            res += f" MI355X:{h['gpus_per_node']}"
+        click.echo(fmt.format(r["name"], r["status"], res,
+                              ",".join(map(str, h.get("gpu_ids", [])))))
+
+
+@cli.command()
+@click.argument("clusters", nargs=-1, required=True)
+def start(clusters):
+    for cl in clusters:
+        sdk.get(sdk.start(cl))
+        click.echo(f"Cluster {cl} started.")
+
+
+@cli.command()
+@click.argument("clusters", nargs=-1, required=True)
+def stop(clusters):
+    for cl in clusters:
+        sdk.get(sdk.stop(cl))
+        click.echo(f"Cluster {cl} stopped.")
+
+
+@cli.command()
+@click.argument("clusters", nargs=-1, required=True)
+@click.option("--yes", "-y", is_flag=True)
+def down(clusters, yes):
+    for cl in clusters:
+        sdk.get(sdk.down(cl))
+        click.echo(f"Cluster {cl} terminated.")
+
+
+@cli.command()
+@click.argument("cluster")
+@click.option("--idle-minutes", "-i", type=int, required=True)
+@click.option("--down", is_flag=True)
+def autostop(cluster, idle_minutes, down):
+    sdk.get(sdk.autostop(cluster, idle_minutes, down))
+    click.echo(f"Autostop set on {cluster}: {idle_minutes}m "
+               f"({'down' if down else 'stop'})")
+
+
+@cli.command()
+@click.argument("cluster")
+def queue(cluster):
+    jobs = sdk.get(sdk.queue(cluster))
+    fmt = "{:<6} {:<18} {:<12} {:<10}"
+    click.echo(fmt.format("ID", "NAME", "STATUS", "GPUS"))
+    for j in jobs:
+        click.echo(fmt.format(j["job_id"], str(j.get("name") or "-"),
+                              j["status"],
+                              ",".join(map(str, j.get("gpus", [])))))
+
+
+@cli.command()
+@click.argument("cluster")
+@click.argument("job_id", type=int, required=False)
+@click.option("--no-follow", is_flag=True)
+def logs(cluster, job_id, no_follow):
+    sdk.tail_logs(cluster, job_id, follow=not no_follow)
+
+
+@cli.command()
+@click.argument("cluster")
+@click.argument("job_ids", nargs=-1, type=int)
+@click.option("--all", "all_jobs", is_flag=True)
+def cancel(cluster, job_ids, all_jobs):
+    n = sdk.get(sdk.cancel(cluster, list(job_ids) or None, all_jobs))
+    click.echo(f"Cancelled {n} job(s).")
+
+
+@cli.command()
+def check():
+    _print_result(sdk.get(sdk.check()))
+
+
+@cli.command("show-gpus")
+def show_gpus():
+    gpus = sdk.get(sdk.show_gpus())
+    fmt = "{:<6} {:<10} {:<10} {:<6} {:<12}"
+    click.echo(fmt.format("GPU", "NAME", "MEM_GB", "NUMA", "USED_BY"))
+    for g in gpus:
+        click.echo(fmt.format(g["index"], g["name"], g["memory_gb"],
+                              g["numa_node"], str(g.get("used_by") or "-")))
+
+
+# ---- api ------------------------------------------------------------------
+@cli.group()
+def api():
+    """API server management."""
+
+
+@api.command("start")
+def api_start():
+    started = sdk.api_start()
+    click.echo("API server started." if started
+               else "API server already running.")
+
+
+@api.command("stop")
+def api_stop():
+    click.echo("Stopped." if sdk.api_stop() else "Not running.")
+
+
+@api.command("status")
+def api_status():
+    click.echo("healthy" if sdk.api_healthy() else "not running")
+
+
+# ---- jobs -----------------------------------------------------------------
+@cli.group()
+def jobs():
+    """Managed jobs (auto-recovery)."""
+
+
+@jobs.command("launch")
+@click.argument("entrypoint")
+@click.option("--name", "-n", default=None)
+@click.option("--env", multiple=True)
+@click.option("--gpus", default=None)
+@click.option("--async", "async_", is_flag=True)
+def jobs_launch(entrypoint, name, env, gpus, async_):
+    task = _load_task(entrypoint, env, {"accelerators": gpus})
+    rid = sdk.jobs_launch(task, name)
+    if async_:
+        click.echo(rid)
+        return
+    _print_result(sdk.stream_and_get(rid))
+
+
+@jobs.command("queue")
+def jobs_queue():
+    rows = sdk.get(sdk.jobs_queue())
+    fmt = "{:<6} {:<18} {:<12} {:<8} {:<10}"
+    click.echo(fmt.format("ID", "NAME", "STATUS", "RECOV", "CLUSTER"))
+    for j in rows:
+        click.echo(fmt.format(j["job_id"], str(j.get("name") or "-"),
+                              j["status"], j.get("recovery_count", 0),
+                              str(j.get("cluster_name") or "-")))
+
+
+@jobs.command("cancel")
+@click.argument("job_ids", nargs=-1, type=int)
+@click.option("--all", "all_jobs", is_flag=True)
+def jobs_cancel(job_ids, all_jobs):
+    n = sdk.get(sdk.jobs_cancel(list(job_ids) or None, all_jobs))
+    click.echo(f"Cancelled {n} managed job(s).")
+
+
+# ---- serve ----------------------------------------------------------------
+@cli.group()
+def serve():
+    """Sky Serve: replicated serving with autoscaling."""
+
+
+@serve.command("up")
+@click.argument("entrypoint")
+@click.option("--service-name", "-n", required=True)
+@click.option("--env", multiple=True)
+def serve_up(entrypoint, service_name, env):
+    task = _load_task(entrypoint, env, {})
+    _print_result(sdk.stream_and_get(sdk.serve_up(task, service_name)))
+
+
+@serve.command("down")
+@click.argument("service_name")
+def serve_down(service_name):
+    sdk.get(sdk.serve_down(service_name))
+    click.echo(f"Service {service_name} torn down.")
+
+
+@serve.command("status")
+@click.argument("service_name", required=False)
+def serve_status(service_name):
+    _print_result(sdk.get(sdk.serve_status(service_name)))
+
+
+# ---- storage --------------------------------------------------------------
+@cli.group()
+def storage():
+    """Storage management."""
+
+
+@storage.command("ls")
+def storage_ls():
+    _print_result(sdk.get(sdk.storage_list()))
+
+
+@storage.command("delete")
+@click.argument("name")
+def storage_delete(name):
+    sdk.get(sdk.storage_delete(name))
+    click.echo(f"Storage {name} deleted.")
+
+
+def main():
+    cli()
+
+
+if __name__ == "__main__":
+    main()

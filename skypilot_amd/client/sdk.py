@@ -1,0 +1,271 @@
+"""Client SDK — one function per REST endpoint, async request IDs.
+
+Reference: sky/client/sdk.py (launch:694, get:2409, stream_and_get).
+Every call POSTs to the API server and returns a request_id; results are
+fetched with get()/stream_and_get().  If no server is running on
+localhost it is auto-started (reference: `sky api start` implicit).
+"""
+from __future__ import annotations
+
+import json
+import os
+import subprocess
+import sys
+import time
+from typing import Any, Dict, List, Optional
+
+import httpx
+
+from skypilot_amd.exceptions import ApiServerError, SkyAmdError
+from skypilot_amd.server.app import server_url
+from skypilot_amd.task import Task
+
+_TEST_CLIENT = None  # set by tests to route in-process (TestClient)
+
+
+def use_test_client(client) -> None:
+    global _TEST_CLIENT
+    _TEST_CLIENT = client
+
+
+from contextlib import contextmanager
+
+
+@contextmanager
+def _client():
+    if _TEST_CLIENT is not None:
+        yield _TEST_CLIENT  # never closed here; tests own its lifecycle
+    else:
+        c = httpx.Client(base_url=server_url(), timeout=30.0)
+        try:
+            yield c
+        finally:
+            c.close()
+
+
+def api_start(wait: float = 15.0) -> bool:
+    """Start the local API server if not running."""
+    if api_healthy():
+        return False
+    env = dict(os.environ)
+    log = open(os.path.expanduser("~/.sky_amd_api.log"), "ab")
+    subprocess.Popen(
+        [sys.executable, "-m", "skypilot_amd.server.app"],
+        stdout=log, stderr=subprocess.STDOUT, start_new_session=True,
+        env=env)
+    log.close()
+    deadline = time.time() + wait
+    while time.time() < deadline:
+        if api_healthy():
+            return True
+        time.sleep(0.3)
+    raise ApiServerError("API server failed to start; see ~/.sky_amd_api.log")
+
+
+def api_healthy() -> bool:
+    try:
+        with _client() as c:
+            return c.get("/health").status_code == 200
+    except (httpx.HTTPError, OSError):
+        return False
+
+
+def api_stop() -> bool:
+    from skypilot_amd import global_state
+    meta = global_state.root_dir() / "api" / "server.json"
+    if not meta.exists():
+        return False
+    try:
+        pid = json.loads(meta.read_text()).get("pid")
+        if pid:
+            import signal
+            os.kill(pid, signal.SIGTERM)
+            return True
+    except (OSError, ValueError, ProcessLookupError):
+        pass
+    return False
+
+
+def _ensure_server():
+    if _TEST_CLIENT is None and not api_healthy():
+        api_start()
+
+
+def _submit(name: str, body: Dict[str, Any]) -> str:
+    _ensure_server()
+    with _client() as c:
+        r = c.post(f"/api/v1/{name}", json=body)
+        if r.status_code != 200:
+            raise ApiServerError(f"{name}: {r.status_code} {r.text[:400]}")
+        return r.json()["request_id"]
+
+
+def get(request_id: str, timeout: float = 3600.0, poll: float = 0.3) -> Any:
+    """Block until the request finishes; return its result
+    (reference: sdk.py:2409)."""
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        with _client() as c:
+            r = c.get("/api/get", params={"request_id": request_id})
+            if r.status_code != 200:
+                raise ApiServerError(r.text[:400])
+            data = r.json()
+        if data["status"] == "SUCCEEDED":
+            return data["result"]
+        if data["status"] == "FAILED":
+            raise SkyAmdError(data["error"] or "request failed")
+        if data["status"] == "CANCELLED":
+            raise SkyAmdError("request cancelled")
+        time.sleep(poll)
+    raise TimeoutError(f"request {request_id} timed out")
+
+
+def stream_and_get(request_id: str, out=None) -> Any:
+    out = out or sys.stdout
+    if _TEST_CLIENT is not None:
+        r = _TEST_CLIENT.get("/api/stream",
+                             params={"request_id": request_id})
+        out.write(r.text)
+        return get(request_id)
+    with httpx.Client(base_url=server_url(), timeout=None) as c:
+        with c.stream("GET", "/api/stream",
+                      params={"request_id": request_id}) as r:
+            for chunk in r.iter_text():
+                out.write(chunk)
+                out.flush()
+    return get(request_id)
+
+
+def cancel_request(request_id: str) -> bool:
+    with _client() as c:
+        r = c.post("/api/cancel", json={"request_id": request_id})
+        return r.json().get("cancelled", False)
+
+
+def _task_body(task) -> Dict[str, Any]:
+    if isinstance(task, Task):
+        return task.to_yaml_config()
+    return dict(task)
+
+
+# ---- public API (mirrors reference sdk surface) ---------------------------
+def launch(task, cluster_name: Optional[str] = None, *, down: bool = False,
+           idle_minutes_to_autostop: Optional[int] = None,
+           detach_run: bool = True) -> str:
+    return _submit("launch", {
+        "task": _task_body(task), "cluster_name": cluster_name,
+        "down": down,
+        "idle_minutes_to_autostop": idle_minutes_to_autostop,
+        "detach_run": detach_run})
+
+
+def exec(task, cluster_name: str, *, detach_run: bool = True) -> str:  # noqa: A001
+    return _submit("exec", {"task": _task_body(task),
+                            "cluster_name": cluster_name,
+                            "detach_run": detach_run})
+
+
+def status(cluster_names: Optional[List[str]] = None,
+           refresh: bool = False) -> str:
+    return _submit("status", {"cluster_names": cluster_names,
+                              "refresh": refresh})
+
+
+def start(cluster_name: str) -> str:
+    return _submit("start", {"cluster_name": cluster_name})
+
+
+def stop(cluster_name: str) -> str:
+    return _submit("stop", {"cluster_name": cluster_name})
+
+
+def down(cluster_name: str) -> str:
+    return _submit("down", {"cluster_name": cluster_name})
+
+
+def autostop(cluster_name: str, idle_minutes: int, down: bool = False) -> str:
+    return _submit("autostop", {"cluster_name": cluster_name,
+                                "idle_minutes": idle_minutes, "down": down})
+
+
+def queue(cluster_name: str) -> str:
+    return _submit("queue", {"cluster_name": cluster_name})
+
+
+def cancel(cluster_name: str, job_ids: Optional[List[int]] = None,
+           all_jobs: bool = False) -> str:
+    return _submit("cancel", {"cluster_name": cluster_name,
+                              "job_ids": job_ids, "all_jobs": all_jobs})
+
+
+def job_status(cluster_name: str, job_id: int) -> str:
+    return _submit("job_status", {"cluster_name": cluster_name,
+                                  "job_id": job_id})
+
+
+def check() -> str:
+    return _submit("check", {})
+
+
+def show_gpus() -> str:
+    return _submit("show_gpus", {})
+
+
+def cluster_events(cluster_name: str) -> str:
+    return _submit("cluster_events", {"cluster_name": cluster_name})
+
+
+def tail_logs(cluster_name: str, job_id: Optional[int] = None,
+              follow: bool = True, out=None):
+    """Stream job logs to `out` (direct streaming route, not a request)."""
+    out = out or sys.stdout
+    _ensure_server()
+    if _TEST_CLIENT is not None:
+        r = _TEST_CLIENT.get(f"/api/v1-logs/{cluster_name}",
+                             params={"job_id": job_id, "follow": follow})
+        out.write(r.text)
+        return
+    with httpx.Client(base_url=server_url(), timeout=None) as c:
+        params = {"follow": follow}
+        if job_id is not None:
+            params["job_id"] = job_id
+        with c.stream("GET", f"/api/v1-logs/{cluster_name}",
+                      params=params) as r:
+            for chunk in r.iter_text():
+                out.write(chunk)
+                out.flush()
+
+
+# managed jobs / serve ------------------------------------------------------
+def jobs_launch(task, name: Optional[str] = None) -> str:
+    return _submit("jobs_launch", {"task": _task_body(task), "name": name})
+
+
+def jobs_queue() -> str:
+    return _submit("jobs_queue", {})
+
+
+def jobs_cancel(job_ids: Optional[List[int]] = None,
+                all_jobs: bool = False) -> str:
+    return _submit("jobs_cancel", {"job_ids": job_ids, "all_jobs": all_jobs})
+
+
+def serve_up(task, service_name: str) -> str:
+    return _submit("serve_up", {"task": _task_body(task),
+                                "service_name": service_name})
+
+
+def serve_down(service_name: str) -> str:
+    return _submit("serve_down", {"service_name": service_name})
+
+
+def serve_status(service_name: Optional[str] = None) -> str:
+    return _submit("serve_status", {"service_name": service_name})
+
+
+def storage_list() -> str:
+    return _submit("storage_list", {})
+
+
+def storage_delete(name: str) -> str:
+    return _submit("storage_delete", {"name": name})

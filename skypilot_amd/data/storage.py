@@ -1,0 +1,122 @@
+"""Storage + file mounts for the MI355X pool.
+
+Reference: sky/data/storage.py (Storage/AbstractStore, modes
+MOUNT/COPY/MOUNT_CACHED :336-430) and sky/data/mounting_utils.py.  On a
+single-node pool the durable store is a local directory under
+`~/.sky_amd/storage/<name>` (S3-compatible remotes can layer on via
+rclone later); the three modes map to:
+
+  COPY         — materialize a copy at the mount path
+  MOUNT        — symlink the mount path to the store dir (write-through,
+                 survives cluster teardown: this is the managed-jobs
+                 checkpoint contract, reference SURVEY.md §2.7)
+  MOUNT_CACHED — same as MOUNT on a local FS (writeback cache is a no-op)
+"""
+from __future__ import annotations
+
+import os
+import shutil
+import time
+from pathlib import Path
+from typing import Any, Dict
+
+from skypilot_amd import global_state
+from skypilot_amd.exceptions import TaskValidationError
+from skypilot_amd.utils.command_runner import LocalProcessCommandRunner
+
+MODE_COPY = "COPY"
+MODE_MOUNT = "MOUNT"
+MODE_MOUNT_CACHED = "MOUNT_CACHED"
+
+
+def storage_root() -> Path:
+    d = global_state.root_dir() / "storage"
+    d.mkdir(parents=True, exist_ok=True)
+    return d
+
+
+def get_or_create_store(name: str, source: str | None = None) -> Path:
+    d = storage_root() / name
+    first = not d.exists()
+    d.mkdir(parents=True, exist_ok=True)
+    if first:
+        with global_state._DB_LOCK, global_state._conn() as c:
+            c.execute(
+                "INSERT OR IGNORE INTO storage "
+                "(name,source,store_type,created_at) VALUES (?,?,?,?)",
+                (name, source or "", "local", time.time()))
+        if source:
+            src = os.path.expanduser(source)
+            if os.path.isdir(src):
+                LocalProcessCommandRunner().rsync(src.rstrip("/") + "/",
+                                                  str(d))
+            elif os.path.exists(src):
+                shutil.copy2(src, d)
+    return d
+
+
+def list_storage():
+    with global_state._DB_LOCK, global_state._conn() as c:
+        rows = c.execute(
+            "SELECT name,source,store_type,created_at FROM storage").fetchall()
+    return [{"name": r[0], "source": r[1], "store_type": r[2],
+             "created_at": r[3]} for r in rows]
+
+
+def delete_storage(name: str) -> bool:
+    d = storage_root() / name
+    existed = d.exists()
+    shutil.rmtree(d, ignore_errors=True)
+    with global_state._DB_LOCK, global_state._conn() as c:
+        c.execute("DELETE FROM storage WHERE name=?", (name,))
+    return existed
+
+
+def _resolve_dst(handle: Dict[str, Any], dst: str) -> Path:
+    """Absolute mount paths are real paths (single-node pool shares the
+    FS); relative paths land in the cluster workdir."""
+    if os.path.isabs(dst):
+        return Path(dst)
+    return Path(handle["cluster_dir"]) / "workdir" / dst
+
+
+def execute_file_mounts(handle: Dict[str, Any],
+                        file_mounts: Dict[str, Any]) -> None:
+    runner = LocalProcessCommandRunner()
+    for dst, src in (file_mounts or {}).items():
+        target = _resolve_dst(handle, dst)
+        target.parent.mkdir(parents=True, exist_ok=True)
+        if isinstance(src, str):
+            path = os.path.expanduser(src)
+            if not os.path.exists(path):
+                raise TaskValidationError(
+                    f"file_mount source not found: {src}")
+            if os.path.isdir(path):
+                runner.rsync(path.rstrip("/") + "/", str(target))
+            else:
+                target.parent.mkdir(parents=True, exist_ok=True)
+                shutil.copy2(path, target)
+        elif isinstance(src, dict):
+            name = src.get("name")
+            if not name:
+                raise TaskValidationError(
+                    f"storage mount for {dst} needs a name")
+            mode = str(src.get("mode", MODE_MOUNT)).upper()
+            store = get_or_create_store(name, src.get("source"))
+            if mode == MODE_COPY:
+                if target.is_symlink():
+                    target.unlink()
+                runner.rsync(str(store) + "/", str(target))
+            elif mode in (MODE_MOUNT, MODE_MOUNT_CACHED):
+                if target.is_symlink() or target.exists():
+                    if target.is_symlink():
+                        target.unlink()
+                    elif target.is_dir() and not any(target.iterdir()):
+                        target.rmdir()
+                    else:
+                        continue  # real data already at mount point
+                target.symlink_to(store)
+            else:
+                raise TaskValidationError(f"unknown mount mode {mode}")
+        else:
+            raise TaskValidationError(f"bad file_mount value for {dst}")

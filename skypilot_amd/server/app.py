@@ -1,0 +1,142 @@
+"""API server — FastAPI app + async request store.
+
+Reference: sky/server/server.py (app :1176, routes :2034-2555, request
+store /api/get /api/stream /api/cancel :2597-2911).  Every mutating
+route schedules a request on the executor and returns a request_id; the
+client polls /api/get or streams /api/stream.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import time
+from pathlib import Path
+from typing import Any, Dict, Optional
+
+from fastapi import FastAPI, HTTPException
+from fastapi.responses import StreamingResponse
+
+from skypilot_amd import global_state
+from skypilot_amd.server import executor
+from skypilot_amd.server import handlers  # noqa: F401  (registers handlers)
+from skypilot_amd.server import requests_db as rdb
+
+DEFAULT_PORT = 46580
+API_PREFIX = "/api/v1"
+
+
+def create_app(start_workers: bool = True) -> FastAPI:
+    app = FastAPI(title="skypilot-amd API server")
+
+    @app.on_event("startup")
+    def _startup():
+        if start_workers:
+            executor.start_workers()
+        # serve/jobs background refreshers (reference: server/daemons.py)
+        from skypilot_amd.server import daemons
+        daemons.start()
+
+    @app.get("/health")
+    def health():
+        return {"ok": True, "version": "0.1.0", "pid": os.getpid()}
+
+    # ---- generic async request plumbing -----------------------------------
+    @app.post(API_PREFIX + "/{name}")
+    def submit(name: str, body: Dict[str, Any] = None):
+        try:
+            rid = executor.schedule(name, body or {})
+        except KeyError:
+            raise HTTPException(404, f"unknown request type {name!r}")
+        return {"request_id": rid}
+
+    @app.get("/api/get")
+    def api_get(request_id: str):
+        req = rdb.get(request_id)
+        if req is None:
+            raise HTTPException(404, "no such request")
+        return {
+            "request_id": request_id,
+            "name": req["name"],
+            "status": req["status"],
+            "result": req["result"],
+            "error": req["error"],
+        }
+
+    @app.get("/api/stream")
+    async def api_stream(request_id: str):
+        req = rdb.get(request_id)
+        if req is None:
+            raise HTTPException(404, "no such request")
+
+        async def gen():
+            pos = 0
+            path = Path(req["log_path"])
+            while True:
+                if path.exists():
+                    with open(path, "rb") as f:
+                        f.seek(pos)
+                        chunk = f.read()
+                    if chunk:
+                        pos += len(chunk)
+                        yield chunk
+                r = rdb.get(request_id)
+                if r["status"] in rdb.TERMINAL:
+                    if path.exists():
+                        with open(path, "rb") as f:
+                            f.seek(pos)
+                            tailc = f.read()
+                        if tailc:
+                            yield tailc
+                    break
+                await asyncio.sleep(0.2)
+
+        return StreamingResponse(gen(), media_type="text/plain")
+
+    @app.post("/api/cancel")
+    def api_cancel(body: Dict[str, Any]):
+        rid = body["request_id"]
+        return {"cancelled": executor.cancel_request(rid)}
+
+    @app.get("/api/requests")
+    def api_requests(limit: int = 100):
+        reqs = rdb.list_requests(limit)
+        return [{k: r[k] for k in
+                 ("request_id", "name", "status", "created_at",
+                  "finished_at", "error")} for r in reqs]
+
+    # ---- log streaming for cluster jobs (proxied to the node agent) -------
+    @app.get(API_PREFIX + "-logs/{cluster_name}")
+    def cluster_logs(cluster_name: str, job_id: Optional[int] = None,
+                     follow: bool = True):
+        from skypilot_amd import core
+        gen = core.tail_logs(cluster_name, job_id, follow)
+        return StreamingResponse(gen, media_type="text/plain")
+
+    return app
+
+
+def server_url() -> str:
+    port = int(os.environ.get("SKY_AMD_API_PORT", DEFAULT_PORT))
+    return os.environ.get("SKY_AMD_API_SERVER", f"http://127.0.0.1:{port}")
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int,
+                    default=int(os.environ.get("SKY_AMD_API_PORT",
+                                               DEFAULT_PORT)))
+    args = ap.parse_args()
+    import uvicorn
+    (global_state.root_dir() / "api").mkdir(parents=True, exist_ok=True)
+    (global_state.root_dir() / "api" / "server.json").write_text(
+        json.dumps({"port": args.port, "pid": os.getpid(),
+                    "started_at": time.time()}))
+    uvicorn.run(create_app(), host=args.host, port=args.port,
+                log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

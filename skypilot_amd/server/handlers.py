@@ -1,0 +1,143 @@
+"""Request handlers — the server-side bodies of each API call.
+
+Reference: sky/server/server.py routes (:2034-2555) delegating into
+sky/core.py + sky/execution.py.  Each handler runs inside a per-request
+process (see executor.py); returns must be JSON-serializable.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+from skypilot_amd import core, execution
+from skypilot_amd.server.executor import LONG, SHORT, register
+from skypilot_amd.task import Task
+
+
+@register("launch", LONG)
+def launch(task: Dict[str, Any], cluster_name: Optional[str] = None,
+           down: bool = False, idle_minutes_to_autostop: Optional[int] = None,
+           detach_run: bool = True) -> Dict[str, Any]:
+    t = Task.from_yaml_config(task)
+    job_id, handle = execution.launch(
+        t, cluster_name, detach_run=detach_run, down=down,
+        idle_minutes_to_autostop=idle_minutes_to_autostop)
+    return {"job_id": job_id, "cluster_name": cluster_name,
+            "handle": handle}
+
+
+@register("exec", LONG)
+def exec_(task: Dict[str, Any], cluster_name: str,
+          detach_run: bool = True) -> Dict[str, Any]:
+    t = Task.from_yaml_config(task)
+    job_id, handle = execution.exec_(t, cluster_name, detach_run=detach_run)
+    return {"job_id": job_id, "cluster_name": cluster_name}
+
+
+@register("status", SHORT)
+def status(cluster_names: Optional[List[str]] = None,
+           refresh: bool = False) -> List[Dict[str, Any]]:
+    return core.status(cluster_names, refresh)
+
+
+@register("start", LONG)
+def start(cluster_name: str) -> Dict[str, Any]:
+    return core.start(cluster_name)
+
+
+@register("stop", LONG)
+def stop(cluster_name: str) -> None:
+    core.stop(cluster_name)
+
+
+@register("down", LONG)
+def down(cluster_name: str) -> None:
+    core.down(cluster_name)
+
+
+@register("autostop", SHORT)
+def autostop(cluster_name: str, idle_minutes: int, down: bool = False
+             ) -> None:
+    core.autostop(cluster_name, idle_minutes, down)
+
+
+@register("queue", SHORT)
+def queue(cluster_name: str) -> List[Dict[str, Any]]:
+    return core.queue(cluster_name)
+
+
+@register("cancel", SHORT)
+def cancel(cluster_name: str, job_ids: Optional[List[int]] = None,
+           all_jobs: bool = False) -> int:
+    return core.cancel(cluster_name, job_ids, all_jobs)
+
+
+@register("job_status", SHORT)
+def job_status(cluster_name: str, job_id: int):
+    return core.job_status(cluster_name, job_id)
+
+
+@register("check", SHORT)
+def check() -> Dict[str, Any]:
+    return core.check()
+
+
+@register("show_gpus", SHORT)
+def show_gpus() -> List[Dict[str, Any]]:
+    return core.show_gpus()
+
+
+@register("cluster_events", SHORT)
+def cluster_events(cluster_name: str) -> List[Dict[str, Any]]:
+    return core.cluster_events(cluster_name)
+
+
+@register("storage_list", SHORT)
+def storage_list():
+    from skypilot_amd.data import storage as st
+    return st.list_storage()
+
+
+@register("storage_delete", SHORT)
+def storage_delete(name: str) -> bool:
+    from skypilot_amd.data import storage as st
+    return st.delete_storage(name)
+
+
+# ---- managed jobs (controller recursion; see jobs/) -----------------------
+@register("jobs_launch", LONG)
+def jobs_launch(task: Dict[str, Any], name: Optional[str] = None
+                ) -> Dict[str, Any]:
+    from skypilot_amd.jobs import server as jobs_server
+    return jobs_server.launch(task, name)
+
+
+@register("jobs_queue", SHORT)
+def jobs_queue() -> List[Dict[str, Any]]:
+    from skypilot_amd.jobs import server as jobs_server
+    return jobs_server.queue()
+
+
+@register("jobs_cancel", SHORT)
+def jobs_cancel(job_ids: Optional[List[int]] = None,
+                all_jobs: bool = False) -> int:
+    from skypilot_amd.jobs import server as jobs_server
+    return jobs_server.cancel(job_ids, all_jobs)
+
+
+# ---- serve ----------------------------------------------------------------
+@register("serve_up", LONG)
+def serve_up(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:
+    from skypilot_amd.serve import server as serve_server
+    return serve_server.up(task, service_name)
+
+
+@register("serve_down", LONG)
+def serve_down(service_name: str) -> None:
+    from skypilot_amd.serve import server as serve_server
+    serve_server.down(service_name)
+
+
+@register("serve_status", SHORT)
+def serve_status(service_name: Optional[str] = None):
+    from skypilot_amd.serve import server as serve_server
+    return serve_server.status(service_name)

@@ -1,0 +1,137 @@
+"""Persistent async-request table (reference: sky/server/requests/
+requests.py — id, name, status, pickled return value, logs path)."""
+from __future__ import annotations
+
+import json
+import sqlite3
+import time
+import uuid
+from pathlib import Path
+from typing import Any, Dict, List, Optional
+
+from skypilot_amd import global_state
+
+PENDING = "PENDING"
+RUNNING = "RUNNING"
+SUCCEEDED = "SUCCEEDED"
+FAILED = "FAILED"
+CANCELLED = "CANCELLED"
+TERMINAL = {SUCCEEDED, FAILED, CANCELLED}
+
+_SCHEMA = """
+CREATE TABLE IF NOT EXISTS requests (
+    request_id TEXT PRIMARY KEY,
+    name TEXT NOT NULL,
+    queue TEXT NOT NULL,
+    status TEXT NOT NULL,
+    created_at REAL,
+    started_at REAL,
+    finished_at REAL,
+    body TEXT,
+    result TEXT,
+    error TEXT,
+    worker_pid INTEGER,
+    log_path TEXT
+);
+"""
+
+
+def api_dir() -> Path:
+    d = global_state.root_dir() / "api"
+    (d / "logs").mkdir(parents=True, exist_ok=True)
+    return d
+
+
+def _conn():
+    conn = sqlite3.connect(api_dir() / "requests.db", timeout=30)
+    conn.execute("PRAGMA journal_mode=WAL")
+    conn.executescript(_SCHEMA)
+    return conn
+
+
+def create(name: str, body: Dict[str, Any], queue: str) -> str:
+    rid = uuid.uuid4().hex[:16]
+    log_path = str(api_dir() / "logs" / f"{rid}.log")
+    with _conn() as c:
+        c.execute(
+            "INSERT INTO requests (request_id,name,queue,status,created_at,"
+            "body,log_path) VALUES (?,?,?,?,?,?,?)",
+            (rid, name, queue, PENDING, time.time(), json.dumps(body),
+             log_path))
+    return rid
+
+
+def claim_next(queue: str, worker_pid: int) -> Optional[Dict[str, Any]]:
+    with _conn() as c:
+        c.execute("BEGIN IMMEDIATE")
+        row = c.execute(
+            "SELECT request_id FROM requests WHERE status=? AND queue=? "
+            "ORDER BY created_at LIMIT 1", (PENDING, queue)).fetchone()
+        if row is None:
+            c.execute("COMMIT")
+            return None
+        rid = row[0]
+        c.execute(
+            "UPDATE requests SET status=?, started_at=?, worker_pid=? "
+            "WHERE request_id=?", (RUNNING, time.time(), worker_pid, rid))
+        c.execute("COMMIT")
+    return get(rid)
+
+
+def get(rid: str) -> Optional[Dict[str, Any]]:
+    with _conn() as c:
+        cols = [d[0] for d in
+                c.execute("SELECT * FROM requests LIMIT 0").description]
+        row = c.execute("SELECT * FROM requests WHERE request_id=?",
+                        (rid,)).fetchone()
+    if row is None:
+        return None
+    d = dict(zip(cols, row))
+    d["body"] = json.loads(d["body"] or "{}")
+    d["result"] = json.loads(d["result"]) if d["result"] else None
+    return d
+
+
+def finish(rid: str, status: str, result: Any = None,
+           error: Optional[str] = None) -> None:
+    with _conn() as c:
+        c.execute(
+            "UPDATE requests SET status=?, finished_at=?, result=?, error=? "
+            "WHERE request_id=?",
+            (status, time.time(), json.dumps(result), error, rid))
+
+
+def set_pid(rid: str, pid: int) -> None:
+    with _conn() as c:
+        c.execute("UPDATE requests SET worker_pid=? WHERE request_id=?",
+                  (pid, rid))
+
+
+def list_requests(limit: int = 100) -> List[Dict[str, Any]]:
+    with _conn() as c:
+        cols = [d[0] for d in
+                c.execute("SELECT * FROM requests LIMIT 0").description]
+        rows = c.execute(
+            "SELECT * FROM requests ORDER BY created_at DESC LIMIT ?",
+            (limit,)).fetchall()
+    out = []
+    for r in rows:
+        d = dict(zip(cols, r))
+        d["body"] = json.loads(d["body"] or "{}")
+        d["result"] = json.loads(d["result"]) if d["result"] else None
+        out.append(d)
+    return out
+
+
+def mark_cancelled(rid: str) -> Optional[int]:
+    """Mark cancelled; returns worker pid if it was running."""
+    with _conn() as c:
+        row = c.execute(
+            "SELECT status, worker_pid FROM requests WHERE request_id=?",
+            (rid,)).fetchone()
+        if row is None or row[0] in TERMINAL:
+            return None
+        c.execute(
+            "UPDATE requests SET status=?, finished_at=? WHERE request_id=?",
+            (CANCELLED, time.time(), rid))
+    return row[1]

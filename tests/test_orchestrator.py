@@ -1,0 +1,189 @@
+"""End-to-end orchestrator tests — in-process client->server->executor
+harness on a fake local pool (reference test strategy: SURVEY.md §4,
+tests/common_test_fixtures.py mock_client_requests)."""
+import json
+import os
+import time
+from pathlib import Path
+
+import pytest
+
+pytestmark = pytest.mark.orchestrator
+
+
+@pytest.fixture()
+def sky_env(tmp_path, monkeypatch):
+    home = tmp_path / "sky_home"
+    monkeypatch.setenv("SKY_AMD_HOME", str(home))
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "8")
+    # Fresh module state: the executor registry is global, state DB paths
+    # are computed from env at call time.
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    yield home
+    # Teardown: kill any agents started under this home.
+    for agent_meta in home.glob("clusters/*/agent.json"):
+        try:
+            pid = json.loads(agent_meta.read_text()).get("pid")
+            if pid:
+                os.kill(pid, 15)
+        except (OSError, ValueError):
+            pass
+
+
+@pytest.fixture()
+def client(sky_env):
+    from fastapi.testclient import TestClient
+    from skypilot_amd.client import sdk
+    from skypilot_amd.server import executor
+    from skypilot_amd.server.app import create_app
+    app = create_app(start_workers=True)
+    with TestClient(app) as c:
+        sdk.use_test_client(c)
+        yield c
+    sdk.use_test_client(None)
+    executor.stop_workers()
+
+
+def _wait_job_done(cluster, job_id, timeout=60):
+    from skypilot_amd.client import sdk
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        j = sdk.get(sdk.job_status(cluster, job_id))
+        if j and j["status"] in ("SUCCEEDED", "FAILED", "FAILED_SETUP",
+                                 "CANCELLED", "FAILED_DRIVER"):
+            return j
+        time.sleep(0.3)
+    raise TimeoutError("job did not finish")
+
+
+def test_launch_hello_world(client, tmp_path):
+    """BASELINE config 1: sky launch hello-world on the local pool."""
+    from skypilot_amd.client import sdk
+    out_file = tmp_path / "out.txt"
+    task = {
+        "name": "hello",
+        "run": f"echo hello-from-sky > {out_file}; "
+               "echo rank=$SKYPILOT_NODE_RANK nodes=$SKYPILOT_NUM_NODES",
+    }
+    rid = sdk.launch(task, "t-hello")
+    result = sdk.get(rid, timeout=60)
+    assert result["job_id"] == 1
+    job = _wait_job_done("t-hello", 1)
+    assert job["status"] == "SUCCEEDED"
+    assert out_file.read_text().strip() == "hello-from-sky"
+    # logs
+    import io
+    buf = io.StringIO()
+    sdk.tail_logs("t-hello", 1, follow=False, out=buf)
+    assert "rank=0 nodes=1" in buf.getvalue()
+
+
+def test_env_contract_and_gpu_slices(client):
+    """The gang launcher must inject the SKYPILOT_* env contract and
+    disjoint HIP_VISIBLE_DEVICES per node slice."""
+    from skypilot_amd.client import sdk
+    task = {
+        "name": "envtest",
+        "num_nodes": 2,
+        "resources": {"accelerators": "MI355X:2"},
+        "run": "echo R$SKYPILOT_NODE_RANK/$SKYPILOT_NUM_NODES/"
+               "$SKYPILOT_NUM_GPUS_PER_NODE/GPUS=$HIP_VISIBLE_DEVICES",
+    }
+    result = sdk.get(sdk.launch(task, "t-env"), timeout=60)
+    job = _wait_job_done("t-env", result["job_id"])
+    assert job["status"] == "SUCCEEDED", job
+    import io
+    buf = io.StringIO()
+    sdk.tail_logs("t-env", result["job_id"], follow=False, out=buf)
+    text = buf.getvalue()
+    assert "R0/2/2/GPUS=0,1" in text
+    assert "R1/2/2/GPUS=2,3" in text
+
+
+def test_status_queue_cancel_down(client):
+    from skypilot_amd.client import sdk
+    task = {"name": "sleeper", "run": "sleep 600"}
+    result = sdk.get(sdk.launch(task, "t-q"), timeout=60)
+    jid = result["job_id"]
+    records = sdk.get(sdk.status())
+    assert any(r["name"] == "t-q" and r["status"] == "UP" for r in records)
+    # queue shows the running/pending job
+    jobs = sdk.get(sdk.queue("t-q"))
+    assert any(j["job_id"] == jid for j in jobs)
+    n = sdk.get(sdk.cancel("t-q", [jid]))
+    assert n == 1
+    job = _wait_job_done("t-q", jid)
+    assert job["status"] == "CANCELLED"
+    sdk.get(sdk.down("t-q"))
+    records = sdk.get(sdk.status())
+    assert not any(r["name"] == "t-q" for r in records)
+
+
+def test_exec_on_existing_cluster(client):
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"name": "base", "run": "true"}, "t-exec"))
+    result = sdk.get(sdk.exec({"run": "echo execced"}, "t-exec"))
+    job = _wait_job_done("t-exec", result["job_id"])
+    assert job["status"] == "SUCCEEDED"
+    sdk.get(sdk.down("t-exec"))
+
+
+def test_setup_failure_is_failed_setup(client):
+    from skypilot_amd.client import sdk
+    task = {"name": "badsetup", "setup": "exit 3", "run": "echo never"}
+    result = sdk.get(sdk.launch(task, "t-bad"), timeout=60)
+    job = _wait_job_done("t-bad", result["job_id"])
+    assert job["status"] == "FAILED_SETUP"
+    sdk.get(sdk.down("t-bad"))
+
+
+def test_file_mounts_and_storage_mount(client, tmp_path):
+    from skypilot_amd.client import sdk
+    src = tmp_path / "data.txt"
+    src.write_text("mounted-data")
+    dst = tmp_path / "mnt" / "data.txt"
+    ckpt_dst = tmp_path / "ckpt"
+    task = {
+        "name": "mounts",
+        "file_mounts": {
+            str(dst): str(src),
+            str(ckpt_dst): {"name": "test-bucket", "mode": "MOUNT"},
+        },
+        "run": f"cat {dst}; echo persisted > {ckpt_dst}/state.txt",
+    }
+    result = sdk.get(sdk.launch(task, "t-mounts"), timeout=60)
+    job = _wait_job_done("t-mounts", result["job_id"])
+    assert job["status"] == "SUCCEEDED"
+    # MOUNT mode: data written through the mount survives in the store.
+    home = Path(os.environ["SKY_AMD_HOME"])
+    assert (home / "storage" / "test-bucket" / "state.txt").exists()
+    stores = sdk.get(sdk.storage_list())
+    assert any(s["name"] == "test-bucket" for s in stores)
+    sdk.get(sdk.down("t-mounts"))
+
+
+def test_gpu_oversubscription_rejected(client):
+    from skypilot_amd.client import sdk
+    from skypilot_amd.exceptions import SkyAmdError
+    task = {"run": "true", "resources": {"accelerators": "MI355X:16"}}
+    rid = sdk.launch(task, "t-big")
+    with pytest.raises(SkyAmdError, match="pool has"):
+        sdk.get(rid, timeout=60)
+
+
+def test_job_start_latency_metric(client):
+    """Job-start latency: submit->RUNNING on a warm cluster
+    should be well under the reference's cloud EXEC path."""
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"name": "warm", "run": "true"}, "t-lat"))
+    _wait_job_done("t-lat", 1)
+    t0 = time.time()
+    result = sdk.get(sdk.exec({"run": "true"}, "t-lat"))
+    job = _wait_job_done("t-lat", result["job_id"])
+    latency = time.time() - t0
+    assert job["status"] == "SUCCEEDED"
+    assert latency < 30, f"job start latency {latency:.1f}s"
+    events = sdk.get(sdk.cluster_events("t-lat"))
+    assert any(e["event"] == "JOB_SUBMIT" for e in events)
+    sdk.get(sdk.down("t-lat"))
