@@ -54,10 +54,20 @@ class JobTable:
         self.db_path = self.cluster_dir / "jobs.db"
 
     def _conn(self):
-        conn = sqlite3.connect(self.db_path, timeout=30)
-        conn.execute("PRAGMA journal_mode=WAL")
-        conn.executescript(_SCHEMA)
-        return conn
+        import contextlib
+
+        @contextlib.contextmanager
+        def cm():
+            conn = sqlite3.connect(self.db_path, timeout=30)
+            try:
+                conn.execute("PRAGMA journal_mode=WAL")
+                conn.execute("PRAGMA busy_timeout=30000")
+                conn.executescript(_SCHEMA)
+                with conn:
+                    yield conn
+            finally:
+                conn.close()
+        return cm()
 
     def add_job(self, name: Optional[str], spec: Dict[str, Any]) -> int:
         with self._conn() as c:

@@ -73,11 +73,20 @@ UP = "UP"
 STOPPED = "STOPPED"
 
 
-def _conn() -> sqlite3.Connection:
+import contextlib
+
+
+@contextlib.contextmanager
+def _conn():
     conn = sqlite3.connect(_db_path(), timeout=30)
-    conn.execute("PRAGMA journal_mode=WAL")
-    conn.executescript(_SCHEMA)
-    return conn
+    try:
+        conn.execute("PRAGMA journal_mode=WAL")
+        conn.execute("PRAGMA busy_timeout=30000")
+        conn.executescript(_SCHEMA)
+        with conn:
+            yield conn
+    finally:
+        conn.close()
 
 
 def add_or_update_cluster(name: str, status: str, handle: Dict[str, Any],

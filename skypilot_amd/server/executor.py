@@ -83,6 +83,14 @@ class RequestWorker(threading.Thread):
     def run(self):
         import subprocess
         while not self._stop_evt.is_set():
+            try:
+                self._tick(subprocess)
+            except Exception:  # noqa: BLE001 — a worker must never die
+                traceback.print_exc()
+                time.sleep(0.5)
+
+    def _tick(self, subprocess):
+        if True:
             alive = []
             for p, rid in self._children:
                 if p.poll() is None:
@@ -98,11 +106,11 @@ class RequestWorker(threading.Thread):
             busy = len(self._children) + len(self._inline)
             if busy >= self.parallelism:
                 time.sleep(0.05)
-                continue
+                return
             req = rdb.claim_next(self.queue, os.getpid())
             if req is None:
                 time.sleep(0.05)
-                continue
+                return
             if self.queue == SHORT:
                 t = threading.Thread(target=_run_request_inline, args=(req,),
                                      daemon=True)

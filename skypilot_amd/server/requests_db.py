@@ -42,11 +42,20 @@ def api_dir() -> Path:
     return d
 
 
+import contextlib
+
+
+@contextlib.contextmanager
 def _conn():
     conn = sqlite3.connect(api_dir() / "requests.db", timeout=30)
-    conn.execute("PRAGMA journal_mode=WAL")
-    conn.executescript(_SCHEMA)
-    return conn
+    try:
+        conn.execute("PRAGMA journal_mode=WAL")
+        conn.execute("PRAGMA busy_timeout=30000")
+        conn.executescript(_SCHEMA)
+        with conn:
+            yield conn
+    finally:
+        conn.close()
 
 
 def create(name: str, body: Dict[str, Any], queue: str) -> str:
