@@ -1,0 +1,139 @@
+"""Managed-job controller — one process per managed job.
+
+Reference: sky/jobs/controller.py (JobController:193, _run_one_task:557,
+monitor loop :998-1120 detecting preemption and entering RECOVERING).
+The controller re-enters the ordinary execution path (`execution.launch`)
+to run the user's cluster — the signature controller-as-task recursion
+of the reference (SURVEY.md §1 layer 10).  Consolidation mode: the
+controller runs on the API-server host (reference:
+utils/controller_utils.py:1422), the right default for a one-node pool.
+
+Usage: python -m skypilot_amd.jobs.controller <managed_job_id>
+"""
+from __future__ import annotations
+
+import os
+import sys
+import time
+import traceback
+
+from skypilot_amd import execution, global_state
+from skypilot_amd.agent import job_lib
+from skypilot_amd.backends.pool_backend import PoolBackend
+from skypilot_amd.jobs import recovery, state
+from skypilot_amd.task import Task
+
+STATUS_CHECK_GAP_SECONDS = float(
+    os.environ.get("SKY_AMD_JOBS_POLL_SECONDS", "2.0"))
+
+
+class JobController:
+    def __init__(self, managed_job_id: int):
+        self.job_id = managed_job_id
+        record = state.get(managed_job_id)
+        if record is None:
+            raise RuntimeError(f"managed job {managed_job_id} not found")
+        self.record = record
+        self.task = Task.from_yaml_config(record["task"])
+        jr = self.task.resources.job_recovery
+        self.strategy = recovery.make(
+            jr.strategy if jr else None,
+            jr.max_restarts_on_errors if jr else 0)
+        self.cluster_name = f"sky-jobs-{managed_job_id}"
+        self.backend = PoolBackend()
+
+    # ------------------------------------------------------------------
+    def run(self) -> None:
+        state.update(self.job_id, controller_pid=os.getpid(),
+                     cluster_name=self.cluster_name)
+        try:
+            final = self._run_with_recovery()
+            state.set_status(self.job_id, final)
+        except BaseException as e:  # noqa: BLE001
+            traceback.print_exc()
+            state.set_status(self.job_id, state.FAILED_CONTROLLER, str(e))
+        finally:
+            self._teardown_cluster()
+
+    def _launch_cluster(self):
+        state.set_status(self.job_id, state.STARTING)
+        job_id, handle = execution.launch(
+            self.task, self.cluster_name, detach_run=True,
+            managed_job_id=self.job_id)
+        return job_id, handle
+
+    def _teardown_cluster(self):
+        record = global_state.get_cluster(self.cluster_name)
+        if record is not None:
+            try:
+                self.backend.teardown(record["handle"], terminate=True)
+            except Exception:  # noqa: BLE001
+                pass
+
+    def _run_with_recovery(self) -> str:
+        while True:
+            try:
+                cluster_job_id, handle = self._launch_cluster()
+            except Exception as e:  # noqa: BLE001
+                traceback.print_exc()
+                return state.FAILED_SETUP if "setup" in str(e).lower() \
+                    else state.FAILED
+            state.set_status(self.job_id, state.RUNNING)
+            outcome = self._monitor(cluster_job_id, handle)
+            if outcome == "succeeded":
+                return state.SUCCEEDED
+            if outcome == "cancelled":
+                return state.CANCELLED
+            if outcome == "failed_setup":
+                return state.FAILED_SETUP
+            if outcome == "failed_user":
+                # User-code failure: bounded restarts
+                # (resources.job_recovery.max_restarts_on_errors).
+                if not self.strategy.should_restart_on_failure():
+                    return state.FAILED
+            # preemption / infra failure / bounded user-failure restart:
+            state.set_status(self.job_id, state.RECOVERING)
+            state.bump_recovery(self.job_id)
+            if not self.strategy.keep_placement_first():
+                self._teardown_cluster()
+            self.strategy.wait_before_retry()
+
+    def _monitor(self, cluster_job_id: int, handle) -> str:
+        """Poll the cluster job; classify its end state
+        (reference: controller.py:998-1120)."""
+        while True:
+            time.sleep(STATUS_CHECK_GAP_SECONDS)
+            me = state.get(self.job_id)
+            if me and me["status"] == state.CANCELLED:
+                try:
+                    self.backend.cancel_jobs(handle, [cluster_job_id])
+                except Exception:  # noqa: BLE001
+                    pass
+                return "cancelled"
+            try:
+                agent = self.backend._agent(handle)
+                job = agent.get_job(cluster_job_id)
+            except Exception:  # noqa: BLE001 — cluster gone: preemption
+                return "preempted"
+            if job is None:
+                return "preempted"
+            st = job["status"]
+            if st == job_lib.SUCCEEDED:
+                return "succeeded"
+            if st == job_lib.FAILED_SETUP:
+                return "failed_setup"
+            if st == job_lib.CANCELLED:
+                # Cancelled underneath us (not by sky jobs cancel):
+                # treat as preemption and recover.
+                return "preempted"
+            if st in (job_lib.FAILED, job_lib.FAILED_DRIVER):
+                return "failed_user" if st == job_lib.FAILED else "preempted"
+
+
+def main():
+    job_id = int(sys.argv[1])
+    JobController(job_id).run()
+
+
+if __name__ == "__main__":
+    main()

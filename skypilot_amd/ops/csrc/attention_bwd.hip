@@ -45,7 +45,7 @@ extern "C" __global__ void attn_bwd_pre_kernel(
 // dK/dV kernel. Grid: (S/BN, B*Hkv). LDS: Q,QT,dO,dOT (16KB each) + 8KB
 // shared P/dS staging = 72KB -> 2 blocks/CU.
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
     const float* __restrict__ lse, const float* __restrict__ Dvec,
@@ -55,7 +55,9 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
   __shared__ unsigned short qt_lds[ATT_D * BM];
   __shared__ unsigned short do_lds[BM * ATT_D];
   __shared__ unsigned short dot_lds[ATT_D * BM];
-  __shared__ unsigned short p_lds[BN * BM];  // PT then reused for dST
+  __shared__ unsigned short p_lds[BN * BM];   // PT
+  __shared__ unsigned short ds_lds[BN * BM];  // dST (separate so both
+                                              // mfma passes share 1 barrier)
 
   const int kt = blockIdx.x;
   const int bh = blockIdx.y;
@@ -145,9 +147,11 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
         }
       }
 
-      // PT = exp(scale*ST - lse[q]);  dST = PT * (dPT - Dvec[q]).
+      // PT = exp(scale*ST - lse[q]); dST = PT * (dPT - Dvec[q]).
+      // Streamed straight to LDS (no pt/dst register arrays — keeping
+      // them cost 32 VGPRs and dropped occupancy to 1 wave/SIMD).
       const int my_kvrow = kvbase + 16 * w + lgrp * 4;  // + r
-      float pt[4][4], dst[4][4];
+      __syncthreads();  // prior-iteration mfma reads of p_lds/ds_lds done
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
         int qcol = qbase + ct * 16 + lrow;
@@ -161,59 +165,33 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
             pv = 0.f;
           else
             pv = __expf(sv - l);
-          pt[ct][r] = pv;
-          dst[ct][r] = pv * (dpt[ct][r] - dv);
+          int prow = 16 * w + lgrp * 4 + r;
+          int pcol = ct * 16 + lrow;
+          int off = swz(prow * 128 + pcol * 2, prow);
+          *(unsigned short*)((char*)p_lds + off) = f2bf(pv);
+          *(unsigned short*)((char*)ds_lds + off) =
+              f2bf(pv * (dpt[ct][r] - dv));
         }
       }
+      __syncthreads();
 
-      // Stage PT -> p_lds [kv][q]; dV += PT * dO (via dOT).
-      __syncthreads();
-#pragma unroll
-      for (int ct = 0; ct < 4; ++ct)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int prow = 16 * w + lgrp * 4 + r;
-          int pcol = ct * 16 + lrow;
-          *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
-              f2bf(pt[ct][r]);
-        }
-      __syncthreads();
+      // dV += PT * dO (via dOT);  dK += dST * Q (via QT).
 #pragma unroll
       for (int ct = 0; ct < 8; ++ct)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
           int prow = 16 * w + lrow;
-          s16x8 afrag = *(const s16x8*)((char*)p_lds +
-                                        swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
-          int drow = ct * 16 + lrow;
-          s16x8 bfrag = *(const s16x8*)((char*)dot_lds +
-                                        swz(drow * 128 + (ks * 32 + lgrp * 8) * 2, drow));
-          dv_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dv_acc[ct]);
-        }
-
-      // Stage dST -> p_lds [kv][q]; dK += dST * Q (via QT).
-      __syncthreads();
-#pragma unroll
-      for (int ct = 0; ct < 4; ++ct)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int prow = 16 * w + lgrp * 4 + r;
-          int pcol = ct * 16 + lrow;
-          *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
-              f2bf(dst[ct][r]);
-        }
-      __syncthreads();
-#pragma unroll
-      for (int ct = 0; ct < 8; ++ct)
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          int prow = 16 * w + lrow;
-          s16x8 afrag = *(const s16x8*)((char*)p_lds +
-                                        swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
-          int qrow2 = ct * 16 + lrow;
-          s16x8 bfrag = *(const s16x8*)((char*)qt_lds +
-                                        swz(qrow2 * 128 + (ks * 32 + lgrp * 8) * 2, qrow2));
-          dk_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dk_acc[ct]);
+          int a_off = swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow);
+          int brow = ct * 16 + lrow;
+          int b_off = swz(brow * 128 + (ks * 32 + lgrp * 8) * 2, brow);
+          s16x8 pfrag = *(const s16x8*)((char*)p_lds + a_off);
+          s16x8 dofrag = *(const s16x8*)((char*)dot_lds + b_off);
+          dv_acc[ct] = MFMA_BF16(as_bf16x8(pfrag), as_bf16x8(dofrag),
+                                 dv_acc[ct]);
+          s16x8 dsfrag = *(const s16x8*)((char*)ds_lds + a_off);
+          s16x8 qfrag = *(const s16x8*)((char*)qt_lds + b_off);
+          dk_acc[ct] = MFMA_BF16(as_bf16x8(dsfrag), as_bf16x8(qfrag),
+                                 dk_acc[ct]);
         }
     }
   }

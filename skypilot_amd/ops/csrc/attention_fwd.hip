@@ -68,29 +68,49 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 
   const int n_kv_tiles = causal ? (qbase + BM + BN - 1) / BN : (S + BN - 1) / BN;
 
+  // Async-STAGE split (guide §6 G15 / T14): each tile's K/V global loads
+  // are issued one iteration early into registers, hiding HBM latency
+  // under the previous tile's QK^T + softmax; the LDS writes happen at
+  // the top of the owning iteration.  Thread t owns chunks
+  // (row t/16 + 16i, 16B-chunk t%16), i = 0..3.
+  s16x8 kpre[4], vpre[4];
+  const int pre_row = tid >> 4, pre_ch = tid & 15;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    long long r = (long long)(pre_row + 16 * i) * kv_rowstride + pre_ch * 8;
+    kpre[i] = *(const s16x8*)(Kb + r);
+    vpre[i] = *(const s16x8*)(Vb + r);
+  }
+
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    const int kvbase = kt * BN;
-    // ---- stage K tile [kv][d] (swizzled) + V tile transposed [d][kv].
+    // ---- stage K tile [kv][d] (swizzled) + V tile transposed [d][kv]
+    // from the prefetch registers.
     __syncthreads();
-    {
-      // K: thread t loads 16B chunk (row t/16, chunk t%16), 4 rounds.
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        int idx = tid + i * 256;          // 1024 chunks total
-        int row = idx >> 4, ch = idx & 15;
-        s16x8 kv8 = *(const s16x8*)(Kb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
-        *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
-        // V: same chunk, scattered into transposed layout.
-        s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+    for (int i = 0; i < 4; ++i) {
+      int row = pre_row + 16 * i;
+      *(s16x8*)((char*)k_lds + swz(row * 256 + pre_ch * 16, row)) = kpre[i];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int d = ch * 8 + j;
-          *(unsigned short*)((char*)vt_lds + swz(d * 128 + row * 2, d)) =
-              (unsigned short)vv8[j];
-        }
+      for (int j = 0; j < 8; ++j) {
+        int d = pre_ch * 8 + j;
+        *(unsigned short*)((char*)vt_lds + swz(d * 128 + row * 2, d)) =
+            (unsigned short)vpre[i][j];
       }
     }
     __syncthreads();
+    if (kt + 1 < n_kv_tiles) {
+      // Issue next tile's loads now; they stay in flight through QK^T +
+      // softmax (the sync before PV drains vmcnt).
+      const long long base = (long long)(kt + 1) * BN * kv_rowstride;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        long long r = base + (long long)(pre_row + 16 * i) * kv_rowstride +
+                      pre_ch * 8;
+        kpre[i] = *(const s16x8*)(Kb + r);
+        vpre[i] = *(const s16x8*)(Vb + r);
+      }
+    }
+    const int kvbase = kt * BN;
 
     // ---- S = scale * Q K^T for this wave's 16 rows x 64 cols.
     f32x4 s_acc[4];
@@ -106,10 +126,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       }
     }
 
-    // ---- mask + online softmax.
+    // ---- mask + online softmax.  P is streamed straight to LDS (no
+    // p[4][4] register tile — saves 16 VGPRs of occupancy).
     const int my_qrow = qbase + 16 * w + lgrp * 4;  // + r
-    float p[4][4];  // [ct][r]
-    float m_new[4], corr[4];
+    float corr[4], m_safe[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float mx = -INFINITY;
@@ -118,45 +138,44 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         float s = s_acc[ct][r] * scale;
         int col = kvbase + ct * 16 + lrow;
         if ((causal && col > my_qrow + r) || col >= S) s = -INFINITY;
-        p[ct][r] = s;
+        s_acc[ct][r] = s;
         mx = fmaxf(mx, s);
       }
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-      m_new[r] = fmaxf(m_run[r], mx);
+      float m_new = fmaxf(m_run[r], mx);
       // All -inf row (above diagonal): keep m at -inf, contribute nothing.
-      float m_safe = (m_new[r] == -INFINITY) ? 0.f : m_new[r];
-      corr[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_safe);
-      float rs = 0.f;
+      m_safe[r] = (m_new == -INFINITY) ? 0.f : m_new;
+      corr[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_safe[r]);
+      m_run[r] = m_new;
+    }
+    // p_lds has been free since the top-of-loop barrier (prior PV read it
+    // before that); no extra barrier needed before writing P.
+    float rs[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int ct = 0; ct < 4; ++ct) {
-        float e = (p[ct][r] == -INFINITY) ? 0.f : __expf(p[ct][r] - m_safe);
-        p[ct][r] = e;
-        rs += e;
+    for (int ct = 0; ct < 4; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float e = (s_acc[ct][r] == -INFINITY)
+                      ? 0.f : __expf(s_acc[ct][r] - m_safe[r]);
+        rs[r] += e;
+        int prow = 16 * w + lgrp * 4 + r;
+        int pcol = ct * 16 + lrow;
+        *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+            f2bf(e);
       }
 #pragma unroll
-      for (int off = 8; off > 0; off >>= 1) rs += __shfl_xor(rs, off, 64);
-      l_run[r] = l_run[r] * corr[r] + rs;
-      m_run[r] = m_new[r];
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        rs[r] += __shfl_xor(rs[r], off, 64);
+      l_run[r] = l_run[r] * corr[r] + rs[r];
     }
     // rescale existing O.
 #pragma unroll
     for (int ct = 0; ct < 8; ++ct)
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[ct][r] *= corr[r];
-
-    // ---- P -> LDS (bf16, swizzled [q 64][kv 64]).
-    __syncthreads();  // all waves done reading k_lds? (p_lds separate; sync
-                      // protects p_lds reuse across kv iterations)
-#pragma unroll
-    for (int ct = 0; ct < 4; ++ct)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int prow = 16 * w + lgrp * 4 + r;
-        int pcol = ct * 16 + lrow;
-        *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
-            f2bf(p[ct][r]);
-      }
     __syncthreads();
 
     // ---- O += P V : A from p_lds, B from vt_lds.

@@ -1,0 +1,109 @@
+"""Managed-jobs tests: controller recursion, recovery, cancellation,
+checkpoint-backed resume contract (reference: SURVEY.md §2.7/§3.2)."""
+import json
+import os
+import time
+from pathlib import Path
+
+import pytest
+
+from tests.test_orchestrator import client, sky_env  # fixtures  # noqa: F401
+
+
+def _wait_managed(job_id, statuses, timeout=90):
+    from skypilot_amd.client import sdk
+    deadline = time.time() + timeout
+    last = None
+    while time.time() < deadline:
+        rows = sdk.get(sdk.jobs_queue())
+        for r in rows:
+            if r["job_id"] == job_id:
+                last = r
+                if r["status"] in statuses:
+                    return r
+        time.sleep(0.5)
+    raise TimeoutError(f"managed job {job_id} last={last}")
+
+
+def test_managed_job_success(client):
+    from skypilot_amd.client import sdk
+    task = {"name": "mj-ok", "run": "echo managed-ok"}
+    res = sdk.get(sdk.jobs_launch(task, "mj-ok"))
+    job = _wait_managed(res["job_id"], {"SUCCEEDED", "FAILED"})
+    assert job["status"] == "SUCCEEDED"
+    # Controller tore the job cluster down.
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        records = sdk.get(sdk.status())
+        if not any(r["name"] == f"sky-jobs-{res['job_id']}"
+                   for r in records):
+            break
+        time.sleep(0.5)
+    else:
+        pytest.fail("managed-job cluster not cleaned up")
+
+
+def test_managed_job_restart_on_failure(client, tmp_path):
+    """max_restarts_on_errors: first attempt exits 1, retry succeeds."""
+    from skypilot_amd.client import sdk
+    marker = tmp_path / "attempted"
+    task = {
+        "name": "mj-flaky",
+        "resources": {"job_recovery": {"strategy": "FAILOVER",
+                                       "max_restarts_on_errors": 2}},
+        "run": f"if [ ! -f {marker} ]; then touch {marker}; exit 1; "
+               "else echo recovered-ok; fi",
+    }
+    res = sdk.get(sdk.jobs_launch(task, "mj-flaky"))
+    job = _wait_managed(res["job_id"], {"SUCCEEDED", "FAILED"}, timeout=120)
+    assert job["status"] == "SUCCEEDED"
+    assert job["recovery_count"] >= 1
+
+
+def test_managed_job_failure_budget_exhausted(client):
+    from skypilot_amd.client import sdk
+    task = {"name": "mj-bad", "run": "exit 7"}  # no recovery budget
+    res = sdk.get(sdk.jobs_launch(task, "mj-bad"))
+    job = _wait_managed(res["job_id"], {"SUCCEEDED", "FAILED"}, timeout=120)
+    assert job["status"] == "FAILED"
+
+
+def test_managed_job_cancel(client):
+    from skypilot_amd.client import sdk
+    task = {"name": "mj-sleep", "run": "sleep 600"}
+    res = sdk.get(sdk.jobs_launch(task, "mj-sleep"))
+    _wait_managed(res["job_id"], {"RUNNING"})
+    n = sdk.get(sdk.jobs_cancel([res["job_id"]]))
+    assert n == 1
+    job = _wait_managed(res["job_id"], {"CANCELLED"})
+    assert job["status"] == "CANCELLED"
+
+
+def test_managed_job_preemption_recovery(client, tmp_path):
+    """Kill the job's cluster under the controller: it must detect the
+    preemption, relaunch, and the task resumes from its mounted state
+    (the checkpoint contract)."""
+    from skypilot_amd.client import sdk
+    ckpt = tmp_path / "ckpt"
+    task = {
+        "name": "mj-preempt",
+        "file_mounts": {str(ckpt): {"name": "mj-preempt-ckpt",
+                                    "mode": "MOUNT"}},
+        # First incarnation sleeps (gets preempted); after recovery the
+        # state file exists and it finishes immediately.
+        "run": f"if [ -f {ckpt}/state ]; then echo resumed-from-ckpt; "
+               f"else echo phase1 > {ckpt}/state; sleep 600; fi",
+    }
+    res = sdk.get(sdk.jobs_launch(task, "mj-preempt"))
+    job_id = res["job_id"]
+    _wait_managed(job_id, {"RUNNING"})
+    time.sleep(2)  # let phase1 write its state
+    # Preempt: tear the managed cluster down out from under the controller.
+    cluster = f"sky-jobs-{job_id}"
+    sdk.get(sdk.down(cluster))
+    job = _wait_managed(job_id, {"SUCCEEDED", "FAILED"}, timeout=120)
+    assert job["status"] == "SUCCEEDED"
+    assert job["recovery_count"] >= 1
+    # The mounted store kept phase1's state across the preemption.
+    home = Path(os.environ["SKY_AMD_HOME"])
+    assert (home / "storage" / "mj-preempt-ckpt" / "state").exists()
