@@ -1,2 +1,149 @@
-def reconcile():
-    pass
+"""Serve controller state DB (reference: sky/serve/serve_state.py)."""
+from __future__ import annotations
+
+import contextlib
+import json
+import sqlite3
+import time
+from typing import Any, Dict, List, Optional
+
+from skypilot_amd import global_state
+
+# Service statuses (reference: serve_state.ServiceStatus)
+CONTROLLER_INIT = "CONTROLLER_INIT"
+REPLICA_INIT = "REPLICA_INIT"
+READY = "READY"
+SHUTTING_DOWN = "SHUTTING_DOWN"
+FAILED = "FAILED"
+SHUTDOWN = "SHUTDOWN"
+
+# Replica statuses (reference: serve_state.ReplicaStatus)
+R_PROVISIONING = "PROVISIONING"
+R_STARTING = "STARTING"
+R_READY = "READY"
+R_NOT_READY = "NOT_READY"
+R_FAILED = "FAILED"
+R_SHUTTING_DOWN = "SHUTTING_DOWN"
+R_SHUTDOWN = "SHUTDOWN"
+
+_SCHEMA = """
+CREATE TABLE IF NOT EXISTS services (
+    name TEXT PRIMARY KEY,
+    status TEXT NOT NULL,
+    task TEXT NOT NULL,
+    spec TEXT NOT NULL,
+    controller_pid INTEGER,
+    lb_port INTEGER,
+    created_at REAL
+);
+CREATE TABLE IF NOT EXISTS replicas (
+    service TEXT NOT NULL,
+    replica_id INTEGER NOT NULL,
+    status TEXT NOT NULL,
+    cluster_name TEXT,
+    endpoint TEXT,
+    launched_at REAL,
+    PRIMARY KEY (service, replica_id)
+);
+"""
+
+
+@contextlib.contextmanager
+def _conn():
+    conn = sqlite3.connect(global_state.root_dir() / "serve.db", timeout=30)
+    try:
+        conn.execute("PRAGMA journal_mode=WAL")
+        conn.execute("PRAGMA busy_timeout=30000")
+        conn.executescript(_SCHEMA)
+        with conn:
+            yield conn
+    finally:
+        conn.close()
+
+
+def add_service(name: str, task: Dict[str, Any], spec: Dict[str, Any],
+                lb_port: int) -> None:
+    with _conn() as c:
+        c.execute(
+            "INSERT OR REPLACE INTO services "
+            "(name,status,task,spec,lb_port,created_at) VALUES (?,?,?,?,?,?)",
+            (name, CONTROLLER_INIT, json.dumps(task), json.dumps(spec),
+             lb_port, time.time()))
+
+
+def update_service(name: str, **fields) -> None:
+    if not fields:
+        return
+    cols = ", ".join(f"{k}=?" for k in fields)
+    with _conn() as c:
+        c.execute(f"UPDATE services SET {cols} WHERE name=?",
+                  (*fields.values(), name))
+
+
+def get_service(name: str) -> Optional[Dict[str, Any]]:
+    with _conn() as c:
+        cols = [d[0] for d in
+                c.execute("SELECT * FROM services LIMIT 0").description]
+        row = c.execute("SELECT * FROM services WHERE name=?",
+                        (name,)).fetchone()
+    if row is None:
+        return None
+    d = dict(zip(cols, row))
+    d["task"] = json.loads(d["task"])
+    d["spec"] = json.loads(d["spec"])
+    return d
+
+
+def list_services() -> List[Dict[str, Any]]:
+    with _conn() as c:
+        rows = c.execute("SELECT name FROM services").fetchall()
+    return [get_service(r[0]) for r in rows]
+
+
+def remove_service(name: str) -> None:
+    with _conn() as c:
+        c.execute("DELETE FROM services WHERE name=?", (name,))
+        c.execute("DELETE FROM replicas WHERE service=?", (name,))
+
+
+def upsert_replica(service: str, replica_id: int, **fields) -> None:
+    with _conn() as c:
+        c.execute(
+            "INSERT INTO replicas (service,replica_id,status) "
+            "VALUES (?,?,?) ON CONFLICT(service,replica_id) DO NOTHING",
+            (service, replica_id, fields.get("status", R_PROVISIONING)))
+        if fields:
+            cols = ", ".join(f"{k}=?" for k in fields)
+            c.execute(
+                f"UPDATE replicas SET {cols} WHERE service=? AND "
+                "replica_id=?", (*fields.values(), service, replica_id))
+
+
+def remove_replica(service: str, replica_id: int) -> None:
+    with _conn() as c:
+        c.execute("DELETE FROM replicas WHERE service=? AND replica_id=?",
+                  (service, replica_id))
+
+
+def list_replicas(service: str) -> List[Dict[str, Any]]:
+    with _conn() as c:
+        cols = [d[0] for d in
+                c.execute("SELECT * FROM replicas LIMIT 0").description]
+        rows = c.execute("SELECT * FROM replicas WHERE service=? "
+                         "ORDER BY replica_id", (service,)).fetchall()
+    return [dict(zip(cols, r)) for r in rows]
+
+
+def reconcile() -> None:
+    import os
+    for s in list_services():
+        if s["status"] in (SHUTDOWN, FAILED):
+            continue
+        pid = s.get("controller_pid")
+        if pid:
+            try:
+                os.kill(pid, 0)
+            except ProcessLookupError:
+                update_service(s["name"], status=FAILED)
+            except PermissionError:
+                pass
