@@ -80,7 +80,7 @@ class Attention(nn.Module):
         self.wo = nn.Linear(self.n_q * d, h, bias=False)
         self.scale = 1.0 / math.sqrt(d)
 
-    def forward(self, x, cos, sin, positions, kv_cache=None):
+    def forward(self, x, cos, sin, positions, infer_ctx=None):
         B, S, _ = x.shape
         d = self.cfg.head_dim
         q = self.wq(x).view(B, S, self.n_q, d)
@@ -90,12 +90,22 @@ class Attention(nn.Module):
                      positions).view(B, S, self.n_q, d)
         k = ops.rope(k.reshape(B * S, self.n_kv, d), cos, sin,
                      positions).view(B, S, self.n_kv, d)
-        if kv_cache is not None:
-            k, v = kv_cache.update(self.layer_idx, k, v)
-            o = ops.attention(q, k, v, self.scale, causal=False) \
-                if S == 1 else ops.attention(q, k, v, self.scale, causal=True)
-        else:
+        if infer_ctx is None:
             o = ops.attention(q, k, v, self.scale, causal=True)
+        elif infer_ctx.mode == "prefill":
+            # New sequence: plain causal attention over the prompt; the
+            # (unpadded) K/V rows land in the cache for decode.
+            infer_ctx.cache.write_prefill(self.layer_idx,
+                                          infer_ctx.prefill_slot, k, v,
+                                          infer_ctx.prefill_len)
+            o = ops.attention(q, k, v, self.scale, causal=True)
+        else:  # decode: one new token per active sequence
+            infer_ctx.cache.write_decode(self.layer_idx, infer_ctx.slots,
+                                         infer_ctx.pos, k, v)
+            o = ops.attn_decode(
+                q.view(B, self.n_q, d), infer_ctx.cache.k[self.layer_idx],
+                infer_ctx.cache.v[self.layer_idx], infer_ctx.kv_lens,
+                infer_ctx.slot_ids_i32, self.scale).view(B, 1, self.n_q, d)
         return self.wo(o.reshape(B, S, self.n_q * d))
 
 
@@ -123,9 +133,9 @@ class Block(nn.Module):
         self.mlp_norm = nn.Parameter(torch.ones(cfg.hidden_size))
         self.eps = cfg.norm_eps
 
-    def forward(self, x, cos, sin, positions, kv_cache=None):
+    def forward(self, x, cos, sin, positions, infer_ctx=None):
         x = x + self.attn(ops.rmsnorm(x, self.attn_norm, self.eps), cos, sin,
-                          positions, kv_cache)
+                          positions, infer_ctx)
         x = x + self.mlp(ops.rmsnorm(x, self.mlp_norm, self.eps))
         return x
 
@@ -149,7 +159,7 @@ class Llama(nn.Module):
         return self._rope
 
     def forward(self, tokens: torch.Tensor, positions: torch.Tensor = None,
-                kv_cache=None) -> torch.Tensor:
+                infer_ctx=None) -> torch.Tensor:
         B, S = tokens.shape
         cos, sin = self._tables(tokens.device)
         if positions is None:
@@ -158,7 +168,7 @@ class Llama(nn.Module):
             positions = positions.unsqueeze(0).expand(B, S).reshape(-1)
         x = self.embed(tokens)
         for blk in self.blocks:
-            x = blk(x, cos, sin, positions, kv_cache)
+            x = blk(x, cos, sin, positions, infer_ctx)
         x = ops.rmsnorm(x, self.final_norm, self.cfg.norm_eps)
         return self.lm_head(x)
 

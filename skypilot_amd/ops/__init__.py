@@ -263,8 +263,36 @@ def fused_cross_entropy(logits, targets, ignore_index: int = -100):
     return _CrossEntropyFn.apply(logits, targets, ignore_index)
 
 
+def attn_decode_ref(q, kc, vc, kv_lens, slot_ids, scale):
+    """q: [B, Hq, D]; kc/vc: [slots, S_max, Hkv, D]."""
+    B, Hq, D = q.shape
+    Hkv = kc.shape[2]
+    rep = Hq // Hkv
+    out = torch.empty_like(q)
+    for i in range(B):
+        slot = int(slot_ids[i])
+        L = int(kv_lens[i])
+        k = kc[slot, :L].float()  # [L, Hkv, D]
+        v = vc[slot, :L].float()
+        kf = k.repeat_interleave(rep, dim=1)  # [L, Hq, D]
+        vf = v.repeat_interleave(rep, dim=1)
+        att = torch.einsum("hd,lhd->hl", q[i].float(), kf) * scale
+        p = att.softmax(-1)
+        out[i] = torch.einsum("hl,lhd->hd", p, vf).to(q.dtype)
+    return out
+
+
+def attn_decode(q, kc, vc, kv_lens, slot_ids, scale):
+    """Decode attention over the KV cache (no autograd)."""
+    if q.is_cuda:
+        C = _require_native("attn_decode")
+        return C.attn_decode(q.contiguous(), kc, vc, kv_lens.int(),
+                             slot_ids.int(), scale)
+    return attn_decode_ref(q, kc, vc, kv_lens, slot_ids, scale)
+
+
 __all__ = [
-    "rmsnorm", "rope", "attention", "fused_cross_entropy", "native",
-    "native_available", "rmsnorm_ref", "rope_ref", "attention_ref",
-    "cross_entropy_ref",
+    "rmsnorm", "rope", "attention", "fused_cross_entropy", "attn_decode",
+    "native", "native_available", "rmsnorm_ref", "rope_ref",
+    "attention_ref", "cross_entropy_ref", "attn_decode_ref",
 ]

@@ -25,6 +25,7 @@ HIP_SOURCES = [
     "cross_entropy.hip",
     "attention_fwd.hip",
     "attention_bwd.hip",
+    "attention_decode.hip",
     "mfma_probe.hip",
 ]
 CPP_SOURCES = ["bindings.cpp"]

@@ -37,6 +37,9 @@ void attn_bwd_launch(const void*, const void*, const void*, const void*,
                      const void*, const float*, float*, void*, void*, void*,
                      int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
+void attn_decode_launch(const void*, const void*, const void*, void*,
+                        const int*, const int*, int, int, int, int, float,
+                        hipStream_t);
 }
 
 // ---- RMSNorm --------------------------------------------------------------
@@ -159,6 +162,23 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor Q, torch::Tensor K,
   return {dQ, dK, dV};
 }
 
+torch::Tensor attn_decode(torch::Tensor Q, torch::Tensor Kc,
+                          torch::Tensor Vc, torch::Tensor kv_lens,
+                          torch::Tensor slot_ids, double scale) {
+  CHECK_GPU(Q); CHECK_CONTIG(Q); CHECK_BF16(Q);
+  TORCH_CHECK(kv_lens.scalar_type() == at::kInt);
+  TORCH_CHECK(slot_ids.scalar_type() == at::kInt);
+  const int B = (int)Q.size(0), Hq = (int)Q.size(1), D = (int)Q.size(2);
+  const int S_max = (int)Kc.size(1), Hkv = (int)Kc.size(2);
+  TORCH_CHECK(D == 128);
+  auto O = torch::empty_like(Q);
+  attn_decode_launch(Q.data_ptr(), Kc.data_ptr(), Vc.data_ptr(),
+                     O.data_ptr(), kv_lens.data_ptr<int>(),
+                     slot_ids.data_ptr<int>(), B, S_max, Hq,
+                     Hkv, (float)scale, cur_stream());
+  return O;
+}
+
 // ---- MFMA layout probe (used by tests/test_gpu_mfma.py) -------------------
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   CHECK_GPU(A); CHECK_BF16(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
@@ -178,5 +198,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fused", &cross_entropy_fused);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("attn_decode", &attn_decode);
   m.def("mfma_probe", &mfma_probe);
 }

@@ -1,0 +1,191 @@
+"""Inference engine — continuous batching over a slot-based KV cache.
+
+MI355X-native serving core for the bundled OpenAI-compatible entrypoint
+(no reference counterpart; SkyPilot points users at vLLM images,
+SURVEY.md §2.11).  Prefill runs the causal flash kernel (prompt padded
+to the 64-row tile); decode steps batch every active sequence into one
+forward through the decode-attention kernel.  max_batch is sized to the
+288 GB HBM budget (KVCache.sized_for_memory).
+"""
+from __future__ import annotations
+
+import queue
+import threading
+import time
+import uuid
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from skypilot_amd.models.llama import build_model
+from skypilot_amd.serve.kv_cache import KVCache
+
+
+@dataclass
+class InferenceContext:
+    cache: KVCache
+    mode: str  # "prefill" | "decode"
+    prefill_slot: int = -1
+    prefill_len: int = 0
+    slots: Optional[torch.Tensor] = None        # int64 [n] (decode)
+    pos: Optional[torch.Tensor] = None          # int64 [n] write positions
+    kv_lens: Optional[torch.Tensor] = None      # int32 [n] incl. new token
+    slot_ids_i32: Optional[torch.Tensor] = None  # int32 [n]
+
+
+@dataclass
+class Request:
+    prompt_ids: List[int]
+    max_tokens: int = 64
+    temperature: float = 0.0
+    stop_id: Optional[int] = None
+    id: str = field(default_factory=lambda: uuid.uuid4().hex[:12])
+    out_ids: List[int] = field(default_factory=list)
+    done: threading.Event = field(default_factory=threading.Event)
+    created: float = field(default_factory=time.time)
+    first_token_at: Optional[float] = None
+    finished_at: Optional[float] = None
+
+
+class Engine:
+    def __init__(self, model_name: str, device: Optional[str] = None,
+                 max_seq: int = 4096, max_batch: Optional[int] = None,
+                 hbm_budget_gb: Optional[float] = None):
+        self.device = torch.device(device or (
+            "cuda" if torch.cuda.is_available() else "cpu"))
+        dtype = torch.bfloat16
+        self.model = build_model(model_name, device=str(self.device),
+                                 dtype=dtype)
+        self.model.eval()
+        self.cfg = self.model.cfg
+        self.max_seq = min(max_seq, self.cfg.max_seq_len)
+        if max_batch is None:
+            if hbm_budget_gb is None:
+                if self.device.type == "cuda":
+                    free, total = torch.cuda.mem_get_info(self.device)
+                    hbm_budget_gb = (free / 1e9) * 0.8
+                else:
+                    hbm_budget_gb = 0.5
+            self.cache = KVCache.sized_for_memory(
+                self.cfg, self.max_seq, int(hbm_budget_gb * 1e9),
+                self.device)
+        else:
+            self.cache = KVCache(self.cfg, max_batch, self.max_seq,
+                                 self.device)
+        self.max_batch = self.cache.max_batch
+        self.free_slots = list(range(self.max_batch))
+        self.active: Dict[int, Request] = {}  # slot -> request
+        self.pending: "queue.Queue[Request]" = queue.Queue()
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self.stats = {"requests": 0, "tokens_generated": 0,
+                      "prefill_tokens": 0}
+
+    # ------------------------------------------------------------------
+    def submit(self, req: Request) -> Request:
+        if len(req.prompt_ids) >= self.max_seq:
+            req.prompt_ids = req.prompt_ids[-(self.max_seq - 1 -
+                                              req.max_tokens):]
+        self.stats["requests"] += 1
+        self.pending.put(req)
+        return req
+
+    def generate(self, prompt_ids: List[int], max_tokens: int = 64,
+                 temperature: float = 0.0,
+                 timeout: float = 300.0) -> List[int]:
+        req = self.submit(Request(prompt_ids=prompt_ids,
+                                  max_tokens=max_tokens,
+                                  temperature=temperature))
+        if not req.done.wait(timeout):
+            raise TimeoutError("generation timed out")
+        return req.out_ids
+
+    def start(self):
+        if self._thread is None:
+            self._thread = threading.Thread(target=self._loop, daemon=True)
+            self._thread.start()
+
+    def stop(self):
+        self._stop.set()
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def _prefill(self, req: Request) -> None:
+        slot = self.free_slots.pop()
+        L = len(req.prompt_ids)
+        pad = (L + 63) // 64 * 64
+        toks = torch.zeros(1, pad, dtype=torch.long, device=self.device)
+        toks[0, :L] = torch.tensor(req.prompt_ids, device=self.device)
+        positions = torch.arange(pad, dtype=torch.int32,
+                                 device=self.device)
+        ctx = InferenceContext(cache=self.cache, mode="prefill",
+                               prefill_slot=slot, prefill_len=L)
+        logits = self.model(toks, positions, ctx)
+        next_id = self._sample(logits[0, L - 1], req.temperature)
+        self.cache.lens[slot] = L
+        req.out_ids.append(next_id)
+        req.first_token_at = time.time()
+        self.active[slot] = req
+        self.stats["prefill_tokens"] += L
+        self._maybe_finish(slot, req, next_id)
+
+    @torch.no_grad()
+    def _decode_step(self) -> None:
+        slots = sorted(self.active.keys())
+        if not slots:
+            return
+        reqs = [self.active[s] for s in slots]
+        toks = torch.tensor([[r.out_ids[-1]] for r in reqs],
+                            dtype=torch.long, device=self.device)
+        lens = [self.cache.lens[s] for s in slots]
+        slots_t = torch.tensor(slots, dtype=torch.long, device=self.device)
+        pos_t = torch.tensor(lens, dtype=torch.long, device=self.device)
+        positions = torch.tensor(lens, dtype=torch.int32,
+                                 device=self.device)
+        kv_lens = torch.tensor([l + 1 for l in lens], dtype=torch.int32,
+                               device=self.device)
+        ctx = InferenceContext(
+            cache=self.cache, mode="decode", slots=slots_t, pos=pos_t,
+            kv_lens=kv_lens, slot_ids_i32=slots_t.int())
+        logits = self.model(toks, positions, ctx)  # [n, 1, V]
+        for i, slot in enumerate(slots):
+            self.cache.lens[slot] += 1
+            req = self.active[slot]
+            next_id = self._sample(logits[i, 0], req.temperature)
+            req.out_ids.append(next_id)
+            self.stats["tokens_generated"] += 1
+            self._maybe_finish(slot, req, next_id)
+
+    def _sample(self, logits: torch.Tensor, temperature: float) -> int:
+        if temperature and temperature > 0:
+            probs = (logits.float() / temperature).softmax(-1)
+            return int(torch.multinomial(probs, 1).item())
+        return int(logits.argmax().item())
+
+    def _maybe_finish(self, slot: int, req: Request, last_id: int) -> None:
+        if (len(req.out_ids) >= req.max_tokens or
+                (req.stop_id is not None and last_id == req.stop_id) or
+                self.cache.lens[slot] + 1 >= self.max_seq):
+            del self.active[slot]
+            self.cache.free(slot)
+            self.free_slots.append(slot)
+            req.finished_at = time.time()
+            req.done.set()
+
+    def _loop(self):
+        while not self._stop.is_set():
+            did = False
+            # Admit pending requests while slots are free.
+            while self.free_slots and not self.pending.empty():
+                try:
+                    req = self.pending.get_nowait()
+                except queue.Empty:
+                    break
+                self._prefill(req)
+                did = True
+            if self.active:
+                self._decode_step()
+                did = True
+            if not did:
+                time.sleep(0.005)

