@@ -121,9 +121,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int d = ch * 8 + j;
-          *(unsigned short*)((char*)qt_lds + swz(d * 128 + row * 2, d)) =
+          *(unsigned short*)((char*)qt_lds + swzT(d * 128 + row * 2, d)) =
               (unsigned short)qv[j];
-          *(unsigned short*)((char*)dot_lds + swz(d * 128 + row * 2, d)) =
+          *(unsigned short*)((char*)dot_lds + swzT(d * 128 + row * 2, d)) =
               (unsigned short)dov[j];
         }
       }
@@ -183,7 +183,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
           int prow = 16 * w + lrow;
           int a_off = swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow);
           int brow = ct * 16 + lrow;
-          int b_off = swz(brow * 128 + (ks * 32 + lgrp * 8) * 2, brow);
+          int b_off = swzT(brow * 128 + (ks * 32 + lgrp * 8) * 2, brow);
           s16x8 pfrag = *(const s16x8*)((char*)p_lds + a_off);
           s16x8 dofrag = *(const s16x8*)((char*)dot_lds + b_off);
           dv_acc[ct] = MFMA_BF16(as_bf16x8(pfrag), as_bf16x8(dofrag),
@@ -287,7 +287,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int d = ch * 8 + j;
-        *(unsigned short*)((char*)kt_lds + swz(d * 128 + row * 2, d)) =
+        *(unsigned short*)((char*)kt_lds + swzT(d * 128 + row * 2, d)) =
             (unsigned short)kv8[j];
       }
     }
@@ -342,7 +342,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
                                       swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
         int krow = ct * 16 + lrow;
         s16x8 bfrag = *(const s16x8*)((char*)kt_lds +
-                                      swz(krow * 128 + (ks * 32 + lgrp * 8) * 2, krow));
+                                      swzT(krow * 128 + (ks * 32 + lgrp * 8) * 2, krow));
         dq_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dq_acc[ct]);
       }
   }

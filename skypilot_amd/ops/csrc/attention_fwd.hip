@@ -93,7 +93,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int d = pre_ch * 8 + j;
-        *(unsigned short*)((char*)vt_lds + swz(d * 128 + row * 2, d)) =
+        *(unsigned short*)((char*)vt_lds + swzT(d * 128 + row * 2, d)) =
             (unsigned short)vpre[i][j];
       }
     }
@@ -188,7 +188,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
                                       swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
         int vrow = ct * 16 + lrow;
         s16x8 bfrag = *(const s16x8*)((char*)vt_lds +
-                                      swz(vrow * 128 + (ks * 32 + lgrp * 8) * 2, vrow));
+                                      swzT(vrow * 128 + (ks * 32 + lgrp * 8) * 2, vrow));
         o_acc[ct] = MFMA_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), o_acc[ct]);
       }
     }

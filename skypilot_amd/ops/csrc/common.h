@@ -46,6 +46,17 @@ __device__ __forceinline__ int swz(int byte_off, int row) {
   return byte_off ^ ((row & 7) << 4);
 }
 
+// Swizzle for TRANSPOSED tiles ([d][seq] layouts): their scatter-writes
+// walk d in strides of 8 (one 16B source chunk covers d = 8c..8c+7), so
+// (row&7) is constant per write instruction — keying on it left every
+// write a 32-way bank conflict (measured: ~1-2e9 SQ_LDS_BANK_CONFLICT
+// per attention dispatch).  Key on (row>>3) instead: varies per lane on
+// the write side, spans 2 values per 16-row read slice on the read side
+// (reads stay balanced: the chunk index itself varies with lane group).
+__device__ __forceinline__ int swzT(int byte_off, int row) {
+  return byte_off ^ (((row >> 3) & 7) << 4);
+}
+
 // bf16 <-> f32 via bit ops (we deliberately avoid __hip_bfloat16 so these
 // headers stay independent of HIP half/bf16 operator macros).
 __device__ __forceinline__ float bf2f(unsigned short h) {

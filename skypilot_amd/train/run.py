@@ -34,13 +34,15 @@ def main() -> int:
     ap.add_argument("--checkpoint-dir", default=os.environ.get(
         "SKY_AMD_CHECKPOINT_DIR"))
     ap.add_argument("--checkpoint-every", type=int, default=20)
+    ap.add_argument("--tp", type=int, default=1)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
     rank, world, _ = setup_distributed()
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     cfg = TrainConfig(model=args.model, micro_batch=args.micro_batch,
-                      seq_len=args.seq_len, lr=args.lr, device=device)
+                      seq_len=args.seq_len, lr=args.lr, device=device,
+                      tp=args.tp)
     tr = Trainer(cfg)
 
     snap = None

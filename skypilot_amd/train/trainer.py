@@ -30,6 +30,7 @@ class TrainConfig:
     seed: int = 1234
     bucket_mb: int = 64
     device: str = "cuda"
+    tp: int = 1  # tensor-parallel degree (divides world size)
 
 
 def dist_env():
@@ -60,14 +61,38 @@ class Trainer:
         if dev == "cuda":
             dev = f"cuda:{self.local_rank}"
         self.device = torch.device(dev)
-        self.model = build_model(cfg.model, device=str(self.device),
-                                 dtype=torch.bfloat16, seed=cfg.seed)
+        self.tp = max(1, cfg.tp)
+        self.tp_group = self.dp_group = None
+        if self.tp > 1:
+            assert self.world % self.tp == 0, "tp must divide world size"
+            # Consecutive ranks share a TP group (same xGMI neighborhood);
+            # DP groups stride across TP groups.
+            for i in range(self.world // self.tp):
+                g = dist.new_group(list(range(i * self.tp,
+                                              (i + 1) * self.tp)))
+                if self.rank // self.tp == i:
+                    self.tp_group = g
+            for j in range(self.tp):
+                g = dist.new_group(list(range(j, self.world, self.tp)))
+                if self.rank % self.tp == j:
+                    self.dp_group = g
+            from skypilot_amd.parallel.tp import build_tp_model
+            self.model = build_tp_model(
+                cfg.model, self.tp, self.rank % self.tp,
+                device=str(self.device), dtype=torch.bfloat16,
+                group=self.tp_group, seed=cfg.seed)
+        else:
+            self.model = build_model(cfg.model, device=str(self.device),
+                                     dtype=torch.bfloat16, seed=cfg.seed)
         self.ddp = BucketedDDP(self.model,
-                               bucket_bytes=cfg.bucket_mb << 20)
+                               bucket_bytes=cfg.bucket_mb << 20,
+                               process_group=self.dp_group)
         self.opt = FusedAdamW(self.model.parameters(), lr=cfg.lr,
                               weight_decay=cfg.weight_decay)
         self.step_count = 0
-        g = torch.Generator(device="cpu").manual_seed(cfg.seed + self.rank)
+        # Data-parallel peers draw different data; TP peers the same.
+        dp_rank = self.rank // self.tp if self.tp > 1 else self.rank
+        g = torch.Generator(device="cpu").manual_seed(cfg.seed + dp_rank)
         self._gen = g
 
     def synthetic_batch(self):
@@ -92,7 +117,8 @@ class Trainer:
         return float(loss.detach())
 
     def tokens_per_step(self) -> int:
-        return self.cfg.micro_batch * self.cfg.seq_len * self.world
+        dp = self.world // self.tp
+        return self.cfg.micro_batch * self.cfg.seq_len * dp
 
 
 def run_training(cfg: TrainConfig, steps: int, warmup: int = 2,
