@@ -37,8 +37,8 @@ REMOTE_ROOT = "~/.sky_amd_node"
 AGENT_PORT = 46590  # fixed remote port (skylet's, reference constants:184)
 
 
-def load_pools(path: str = POOLS_PATH) -> Dict[str, Any]:
-    p = Path(os.path.expanduser(path))
+def load_pools(path: Optional[str] = None) -> Dict[str, Any]:
+    p = Path(os.path.expanduser(path or POOLS_PATH))
     if not p.exists():
         return {}
     with open(p) as f:
