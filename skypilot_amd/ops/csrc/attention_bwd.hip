@@ -168,7 +168,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
           int prow = 16 * w + lgrp * 4 + r;
           int pcol = ct * 16 + lrow;
           int off = swz(prow * 128 + pcol * 2, prow);
-          *(unsigned short*)((char*)p_lds + off) = f2bf(pv);
+          *(unsigned short*)((char*)p_lds + off) = f2bf_trunc(pv);
           *(unsigned short*)((char*)ds_lds + off) =
               f2bf(pv * (dpt[ct][r] - dv));
         }
@@ -328,7 +328,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
         int prow = 16 * w + lgrp * 4 + r;
         int pcol = ct * 16 + lrow;
         *(unsigned short*)((char*)ds_lds + swz(prow * 128 + pcol * 2, prow)) =
-            f2bf(ds);
+            f2bf(ds);  // dS can be negative/large: keep rounded convert
       }
     __syncthreads();
 

@@ -129,6 +129,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     // ---- mask + online softmax.  P is streamed straight to LDS (no
     // p[4][4] register tile — saves 16 VGPRs of occupancy).
     const int my_qrow = qbase + 16 * w + lgrp * 4;  // + r
+    // Interior tiles (fully below the diagonal, fully in range) skip the
+    // per-element mask — wave-uniform branch, most tiles qualify.
+    const bool need_mask =
+        (causal && kvbase + BN - 1 > qbase + 16 * w) || (kvbase + BN > S);
     float corr[4], m_safe[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -136,8 +140,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
         float s = s_acc[ct][r] * scale;
-        int col = kvbase + ct * 16 + lrow;
-        if ((causal && col > my_qrow + r) || col >= S) s = -INFINITY;
+        if (need_mask) {
+          int col = kvbase + ct * 16 + lrow;
+          if ((causal && col > my_qrow + r) || col >= S) s = -INFINITY;
+        }
         s_acc[ct][r] = s;
         mx = fmaxf(mx, s);
       }
@@ -162,7 +168,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         int prow = 16 * w + lgrp * 4 + r;
         int pcol = ct * 16 + lrow;
         *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
-            f2bf(e);
+            f2bf_trunc(e);
       }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {

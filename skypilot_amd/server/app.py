@@ -99,6 +99,23 @@ def create_app(start_workers: bool = True) -> FastAPI:
         rid = body["request_id"]
         return {"cancelled": executor.cancel_request(rid)}
 
+    @app.get("/metrics")
+    def metrics():
+        # Prometheus metrics (reference: sky/server/metrics.py).
+        from fastapi.responses import PlainTextResponse
+        from skypilot_amd.server import requests_db as _rdb
+        lines = []
+        by_status: Dict[str, int] = {}
+        for r in _rdb.list_requests(1000):
+            by_status[r["status"]] = by_status.get(r["status"], 0) + 1
+        for st_, n in by_status.items():
+            lines.append(
+                f'sky_amd_requests_total{{status="{st_.lower()}"}} {n}')
+        from skypilot_amd import global_state as _gs
+        ups = sum(1 for c in _gs.list_clusters() if c["status"] == "UP")
+        lines.append(f"sky_amd_clusters_up {ups}")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.get("/api/requests")
     def api_requests(limit: int = 100):
         reqs = rdb.list_requests(limit)

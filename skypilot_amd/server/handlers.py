@@ -17,6 +17,8 @@ from skypilot_amd.task import Task
 def launch(task: Dict[str, Any], cluster_name: Optional[str] = None,
            down: bool = False, idle_minutes_to_autostop: Optional[int] = None,
            detach_run: bool = True) -> Dict[str, Any]:
+    from skypilot_amd import admin_policy
+    task = admin_policy.apply(task, cluster_name, "launch")
     t = Task.from_yaml_config(task)
     job_id, handle = execution.launch(
         t, cluster_name, detach_run=detach_run, down=down,

@@ -44,3 +44,20 @@ def test_timeline_records(tmp_path, monkeypatch):
     with timeline.Event("unit-test"):
         pass
     assert any(e["name"] == "unit-test" for e in timeline._events)
+
+
+def test_admin_policy_apply(monkeypatch):
+    from skypilot_amd import admin_policy, config
+    # no policy configured -> identity
+    cfg = {"run": "echo hi"}
+    assert admin_policy.apply(cfg) == cfg
+
+    class AddLabel(admin_policy.AdminPolicy):
+        def validate_and_mutate(self, request):
+            tc = dict(request.task_config)
+            tc.setdefault("envs", {})["POLICY"] = "1"
+            return admin_policy.MutatedUserRequest(task_config=tc)
+
+    monkeypatch.setattr(admin_policy, "load_policy", lambda: AddLabel())
+    out = admin_policy.apply(cfg)
+    assert out["envs"]["POLICY"] == "1"
