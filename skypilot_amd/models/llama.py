@@ -118,9 +118,7 @@ class MLP(nn.Module):
         self.m = m
 
     def forward(self, x):
-        gu = self.w_gate_up(x)
-        g, u = gu.split(self.m, dim=-1)
-        return self.w_down(F.silu(g) * u)
+        return self.w_down(ops.swiglu(self.w_gate_up(x)))
 
 
 class Block(nn.Module):

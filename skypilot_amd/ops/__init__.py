@@ -263,6 +263,38 @@ def fused_cross_entropy(logits, targets, ignore_index: int = -100):
     return _CrossEntropyFn.apply(logits, targets, ignore_index)
 
 
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        ctx.save_for_backward(gu)
+        if gu.is_cuda:
+            C = _require_native("swiglu")
+            return C.swiglu_fwd(gu.contiguous())
+        M = gu.shape[-1] // 2
+        g, u = gu.float().split(M, dim=-1)
+        return (torch.nn.functional.silu(g) * u).to(gu.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (gu,) = ctx.saved_tensors
+        if gu.is_cuda:
+            C = native()
+            return C.swiglu_bwd(gu, dy.contiguous())
+        M = gu.shape[-1] // 2
+        g, u = gu.float().split(M, dim=-1)
+        dyf = dy.float()
+        sg = torch.sigmoid(g)
+        silu = g * sg
+        dg = dyf * u * (sg + silu * (1 - sg))
+        du = dyf * silu
+        return torch.cat([dg, du], dim=-1).to(gu.dtype)
+
+
+def swiglu(gu):
+    """Fused silu(gate) * up over a fused [.., 2M] gate|up tensor."""
+    return _SwiGLUFn.apply(gu)
+
+
 def attn_decode_ref(q, kc, vc, kv_lens, slot_ids, scale):
     """q: [B, Hq, D]; kc/vc: [slots, S_max, Hkv, D]."""
     B, Hq, D = q.shape
@@ -293,6 +325,7 @@ def attn_decode(q, kc, vc, kv_lens, slot_ids, scale):
 
 __all__ = [
     "rmsnorm", "rope", "attention", "fused_cross_entropy", "attn_decode",
+    "swiglu",
     "native", "native_available", "rmsnorm_ref", "rope_ref",
     "attention_ref", "cross_entropy_ref", "attn_decode_ref",
 ]

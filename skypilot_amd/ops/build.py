@@ -26,6 +26,7 @@ HIP_SOURCES = [
     "attention_fwd.hip",
     "attention_bwd.hip",
     "attention_decode.hip",
+    "swiglu.hip",
     "mfma_probe.hip",
 ]
 CPP_SOURCES = ["bindings.cpp"]

@@ -226,7 +226,8 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
   __shared__ unsigned short v_lds[BN * ATT_D];
   __shared__ unsigned short ds_lds[BM * BN];
 
-  const int qt = blockIdx.x;
+  // Heavy blocks first (see attn_fwd).
+  const int qt = gridDim.x - 1 - blockIdx.x;
   const int bh = blockIdx.y;
   const int b = bh / Hq;
   const int qh = bh % Hq;

@@ -26,7 +26,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   __shared__ unsigned short vt_lds[ATT_D * BN];   // [d][kv], swizzled, 16KB
   __shared__ unsigned short p_lds[BM * BN];       // [q][kv], swizzled, 8KB
 
-  const int qt = blockIdx.x;            // q tile index
+  // Heavy blocks (large qt => many causal kv tiles) launch first so the
+  // dispatch tail is short blocks (load balance across 256 CUs).
+  const int qt = gridDim.x - 1 - blockIdx.x;  // q tile index
   const int bh = blockIdx.y;            // b * Hq + qh
   const int b = bh / Hq;
   const int qh = bh % Hq;

@@ -37,6 +37,9 @@ void attn_bwd_launch(const void*, const void*, const void*, const void*,
                      const void*, const float*, float*, void*, void*, void*,
                      int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
+void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
+void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
+                       hipStream_t);
 void attn_decode_launch(const void*, const void*, const void*, void*,
                         const int*, const int*, int, int, int, int, float,
                         hipStream_t);
@@ -179,6 +182,28 @@ torch::Tensor attn_decode(torch::Tensor Q, torch::Tensor Kc,
   return O;
 }
 
+torch::Tensor swiglu_fwd(torch::Tensor gu) {
+  CHECK_GPU(gu); CHECK_CONTIG(gu); CHECK_BF16(gu);
+  const int M2 = (int)gu.size(-1);
+  long long rows = gu.numel() / M2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = M2 / 2;
+  auto y = torch::empty(sizes, gu.options());
+  swiglu_fwd_launch(gu.data_ptr(), y.data_ptr(), rows, M2 / 2,
+                    cur_stream());
+  return y;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor gu, torch::Tensor dy) {
+  CHECK_GPU(gu); CHECK_CONTIG(dy);
+  const int M2 = (int)gu.size(-1);
+  long long rows = gu.numel() / M2;
+  auto dgu = torch::empty_like(gu);
+  swiglu_bwd_launch(gu.data_ptr(), dy.data_ptr(), dgu.data_ptr(), rows,
+                    M2 / 2, cur_stream());
+  return dgu;
+}
+
 // ---- MFMA layout probe (used by tests/test_gpu_mfma.py) -------------------
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   CHECK_GPU(A); CHECK_BF16(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
@@ -199,5 +224,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_decode", &attn_decode);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);
 }

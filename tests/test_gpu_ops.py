@@ -196,3 +196,20 @@ def test_engine_cached_decode_matches_gpu():
         assert matches >= 6, (out, ids[len(prompt):])
     finally:
         eng.stop()
+
+
+def test_swiglu_fwd_bwd():
+    torch.manual_seed(8)
+    rows, M = 2048, 14336
+    gu0 = torch.randn(rows, 2 * M)
+    dy0 = torch.randn(rows, M)
+    gu = gu0.bfloat16().to(dev()).requires_grad_()
+    y = ops.swiglu(gu)
+    y.backward(dy0.bfloat16().to(dev()))
+
+    gu_ref = gu0.bfloat16().float().requires_grad_()
+    g, u = gu_ref.split(M, dim=-1)
+    y_ref = torch.nn.functional.silu(g) * u
+    y_ref.backward(dy0)
+    assert rel_err(y.cpu(), y_ref) < 2e-2
+    assert rel_err(gu.grad.cpu(), gu_ref.grad) < 2e-2

@@ -116,3 +116,14 @@ def test_native_extension_loads():
     # The .so is cross-compiled for gfx950 in this container; it must at
     # least load (GPU calls are tested under -m gpu).
     assert ops.native_available(), "in-tree _C.so missing or unloadable"
+
+
+def test_swiglu_cpu_matches_manual():
+    torch.manual_seed(9)
+    gu = torch.randn(4, 64, requires_grad=True)
+    y = ops.swiglu(gu)
+    g, u = gu.detach().split(32, dim=-1)
+    expect = torch.nn.functional.silu(g) * u
+    assert torch.allclose(y, expect, atol=1e-5)
+    y.sum().backward()
+    assert gu.grad is not None
