@@ -302,6 +302,41 @@ def storage_delete(name):
     click.echo(f"Storage {name} deleted.")
 
 
+# ---- volumes --------------------------------------------------------------
+@cli.group()
+def volumes():
+    """Persistent volumes."""
+
+
+@volumes.command("ls")
+def volumes_ls():
+    _print_result(sdk.get(sdk.volumes_list()))
+
+
+@volumes.command("create")
+@click.argument("name")
+@click.option("--size-gb", type=int, default=None)
+def volumes_create(name, size_gb):
+    _print_result(sdk.get(sdk.volumes_create(name, size_gb)))
+
+
+@volumes.command("delete")
+@click.argument("name")
+def volumes_delete(name):
+    sdk.get(sdk.volumes_delete(name))
+    click.echo(f"Volume {name} deleted.")
+
+
+@cli.command("recipes")
+def recipes_cmd():
+    """List bundled task recipes (examples/*.yaml)."""
+    rows = sdk.get(sdk.recipes_list())
+    fmt = "{:<28} {:<22} {:<12}"
+    click.echo(fmt.format("NAME", "TASK", "GPUS"))
+    for r in rows:
+        click.echo(fmt.format(r["name"], r["task_name"], r["accelerators"]))
+
+
 def main():
     cli()
 

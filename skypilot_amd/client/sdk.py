@@ -263,6 +263,22 @@ def serve_status(service_name: Optional[str] = None) -> str:
     return _submit("serve_status", {"service_name": service_name})
 
 
+def volumes_list() -> str:
+    return _submit("volumes_list", {})
+
+
+def volumes_create(name: str, size_gb: Optional[int] = None) -> str:
+    return _submit("volumes_create", {"name": name, "size_gb": size_gb})
+
+
+def volumes_delete(name: str) -> str:
+    return _submit("volumes_delete", {"name": name})
+
+
+def recipes_list() -> str:
+    return _submit("recipes_list", {})
+
+
 def storage_list() -> str:
     return _submit("storage_list", {})
 

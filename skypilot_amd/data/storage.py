@@ -97,6 +97,14 @@ def execute_file_mounts(handle: Dict[str, Any],
                 target.parent.mkdir(parents=True, exist_ok=True)
                 shutil.copy2(path, target)
         elif isinstance(src, dict):
+            if "volume" in src:
+                from skypilot_amd.data import volumes
+                vpath = volumes.mount_path(src["volume"])
+                if target.is_symlink():
+                    target.unlink()
+                if not target.exists():
+                    target.symlink_to(vpath)
+                continue
             name = src.get("name")
             if not name:
                 raise TaskValidationError(

@@ -99,6 +99,12 @@ def create_app(start_workers: bool = True) -> FastAPI:
         rid = body["request_id"]
         return {"cancelled": executor.cancel_request(rid)}
 
+    @app.get("/dashboard")
+    def dashboard():
+        from fastapi.responses import HTMLResponse
+        from skypilot_amd.server.dashboard import render
+        return HTMLResponse(render())
+
     @app.get("/metrics")
     def metrics():
         # Prometheus metrics (reference: sky/server/metrics.py).

@@ -1,0 +1,78 @@
+"""Dashboard — single-page cluster/jobs/services view served by the API
+server (reference: sky/dashboard/, a 47k-LoC Next.js app; here a
+server-rendered page with auto-refresh keeps the surface without the
+node toolchain)."""
+from __future__ import annotations
+
+import html
+import time
+from typing import Any, Dict, List
+
+
+def _table(title: str, headers: List[str], rows: List[List[Any]]) -> str:
+    if not rows:
+        body = f"<tr><td colspan={len(headers)} class=empty>none</td></tr>"
+    else:
+        body = "".join(
+            "<tr>" + "".join(f"<td>{html.escape(str(c))}</td>" for c in row)
+            + "</tr>" for row in rows)
+    head = "".join(f"<th>{h}</th>" for h in headers)
+    return (f"<h2>{title}</h2><table><thead><tr>{head}</tr></thead>"
+            f"<tbody>{body}</tbody></table>")
+
+
+def render() -> str:
+    from skypilot_amd import global_state
+    from skypilot_amd.jobs import state as jobs_state
+    from skypilot_amd.serve import serve_state
+    from skypilot_amd.utils.gpu_topology import detect_gpus
+
+    clusters = []
+    for c in global_state.list_clusters():
+        h = c["handle"]
+        clusters.append([
+            c["name"], c["status"],
+            f"{h.get('num_nodes', 1)}x{h.get('gpus_per_node', 0)} GPU",
+            ",".join(map(str, h.get("gpu_ids", []))) or "-",
+            time.strftime("%m-%d %H:%M",
+                          time.localtime(c.get("launched_at") or 0)),
+        ])
+    jobs = [[j["job_id"], j.get("name") or "-", j["status"],
+             j.get("recovery_count", 0), j.get("cluster_name") or "-"]
+            for j in jobs_state.list_jobs()]
+    services = []
+    for s in serve_state.list_services():
+        if s is None:
+            continue
+        reps = serve_state.list_replicas(s["name"])
+        ready = sum(1 for r in reps if r["status"] == "READY")
+        services.append([s["name"], s["status"], f"{ready}/{len(reps)}",
+                         f"http://127.0.0.1:{s['lb_port']}"])
+    gpus_taken: Dict[int, str] = {}
+    for c in global_state.list_clusters():
+        if c["status"] == "UP":
+            for g in c["handle"].get("gpu_ids", []):
+                gpus_taken[g] = c["name"]
+    gpus = [[g.index, g.name, f"{g.memory_gb} GB", g.numa_node,
+             gpus_taken.get(g.index, "-")] for g in detect_gpus()]
+
+    return f"""<!doctype html><html><head><title>skypilot-amd</title>
+<meta http-equiv="refresh" content="5">
+<style>
+body {{ font-family: ui-monospace, monospace; margin: 2em; background:#111;
+        color:#ddd; }}
+h1 {{ color:#e8443a; }} h2 {{ color:#ccc; margin-top:1.4em; }}
+table {{ border-collapse: collapse; min-width: 48em; }}
+th, td {{ border:1px solid #333; padding:4px 10px; text-align:left; }}
+th {{ background:#222; color:#e8443a; }}
+.empty {{ color:#666; }}
+</style></head><body>
+<h1>skypilot-amd <small style="color:#666">MI355X pool</small></h1>
+{_table("Clusters", ["name", "status", "shape", "gpus", "launched"],
+        clusters)}
+{_table("Managed jobs", ["id", "name", "status", "recoveries", "cluster"],
+        jobs)}
+{_table("Services", ["name", "status", "ready", "endpoint"], services)}
+{_table("Pool GPUs", ["idx", "model", "HBM", "numa", "used by"], gpus)}
+<p style="color:#555">auto-refreshes every 5s · {time.strftime("%H:%M:%S")}
+</p></body></html>"""

@@ -105,6 +105,30 @@ def storage_delete(name: str) -> bool:
     return st.delete_storage(name)
 
 
+@register("volumes_list", SHORT)
+def volumes_list():
+    from skypilot_amd.data import volumes
+    return volumes.list_volumes()
+
+
+@register("volumes_create", SHORT)
+def volumes_create(name: str, size_gb: Optional[int] = None):
+    from skypilot_amd.data import volumes
+    return volumes.create(name, size_gb)
+
+
+@register("volumes_delete", SHORT)
+def volumes_delete(name: str) -> bool:
+    from skypilot_amd.data import volumes
+    return volumes.delete(name)
+
+
+@register("recipes_list", SHORT)
+def recipes_list():
+    from skypilot_amd import recipes
+    return recipes.list_recipes()
+
+
 # ---- managed jobs (controller recursion; see jobs/) -----------------------
 @register("jobs_launch", LONG)
 def jobs_launch(task: Dict[str, Any], name: Optional[str] = None

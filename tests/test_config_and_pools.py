@@ -61,3 +61,23 @@ def test_admin_policy_apply(monkeypatch):
     monkeypatch.setattr(admin_policy, "load_policy", lambda: AddLabel())
     out = admin_policy.apply(cfg)
     assert out["envs"]["POLICY"] == "1"
+
+
+def test_volumes_crud(tmp_path, monkeypatch):
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path / "home"))
+    from skypilot_amd.data import volumes
+    v = volumes.create("vol1", size_gb=10)
+    assert v["name"] == "vol1"
+    assert any(x["name"] == "vol1" for x in volumes.list_volumes())
+    p = volumes.mount_path("vol1")
+    (p / "data.txt").write_text("x")
+    assert volumes.delete("vol1")
+    assert not any(x["name"] == "vol1" for x in volumes.list_volumes())
+
+
+def test_recipes_list():
+    from skypilot_amd import recipes
+    rs = recipes.list_recipes()
+    names = {r["name"] for r in rs}
+    assert "hello" in names and "train_llama3_8b" in names
+    assert recipes.get_recipe_path("hello").endswith("hello.yaml")
