@@ -66,6 +66,11 @@ class Snapshotter:
                 for h, d in zip(self._host[key], st[key]):
                     h.copy_(d, non_blocking=True)
             self._event.record(self._stream)
+        # The next optimizer step mutates these tensors in place on the
+        # main stream; hand the trainer our event so it orders the next
+        # AdamW after the copies (stream-level wait, no host sync —
+        # forward/backward still overlap the D2H traffic).
+        self.trainer._snapshot_event = self._event
 
     def commit(self, blocking: bool = False) -> None:
         """Wait for the async copies, then write to disk in a background

@@ -112,6 +112,12 @@ class Trainer:
         loss = self.model.loss(tokens, targets)
         loss.backward()
         self.ddp.finish()
+        # An in-flight checkpoint snapshot reads optimizer state on a side
+        # stream; order this step's in-place AdamW after those copies.
+        ev = getattr(self, "_snapshot_event", None)
+        if ev is not None and self.device.type == "cuda":
+            torch.cuda.current_stream().wait_event(ev)
+            self._snapshot_event = None
         self.opt.step(grad_scale=self.ddp.grad_scale)
         self.step_count += 1
         return float(loss.detach())
