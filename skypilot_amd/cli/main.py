@@ -248,6 +248,18 @@ def jobs_queue():
                               str(j.get("cluster_name") or "-")))
 
 
+@jobs.command("logs")
+@click.argument("job_id", type=int)
+def jobs_logs(job_id):
+    result = sdk.get(sdk.jobs_logs(job_id))
+    if result.get("controller_log"):
+        click.echo("=== controller log ===")
+        click.echo(result["controller_log"])
+    if result.get("task_log"):
+        click.echo("=== task log ===")
+        click.echo(result["task_log"])
+
+
 @jobs.command("cancel")
 @click.argument("job_ids", nargs=-1, type=int)
 @click.option("--all", "all_jobs", is_flag=True)

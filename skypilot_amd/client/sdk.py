@@ -250,6 +250,10 @@ def jobs_cancel(job_ids: Optional[List[int]] = None,
     return _submit("jobs_cancel", {"job_ids": job_ids, "all_jobs": all_jobs})
 
 
+def jobs_logs(job_id: int) -> str:
+    return _submit("jobs_logs", {"job_id": job_id})
+
+
 def serve_up(task, service_name: str) -> str:
     return _submit("serve_up", {"task": _task_body(task),
                                 "service_name": service_name})

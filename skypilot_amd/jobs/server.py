@@ -42,6 +42,40 @@ def queue() -> List[Dict[str, Any]]:
     return jobs
 
 
+def controller_log_path(job_id: int):
+    return state.global_state.root_dir() / f"jobs-controller-{job_id}.log"
+
+
+def logs(job_id: int, tail_lines: int = 200) -> Dict[str, Any]:
+    """Controller log + the current cluster job's log tail."""
+    out: Dict[str, Any] = {"job_id": job_id}
+    p = controller_log_path(job_id)
+    if p.exists():
+        lines = p.read_text(errors="replace").splitlines()[-tail_lines:]
+        out["controller_log"] = "\n".join(lines)
+    j = state.get(job_id)
+    if j and j.get("cluster_name"):
+        from skypilot_amd import global_state
+        rec = global_state.get_cluster(j["cluster_name"])
+        if rec:
+            try:
+                from skypilot_amd.backends.pool_backend import PoolBackend
+                agent = PoolBackend()._agent(rec["handle"])
+                jobs_on_cluster = agent.get_job_queue()
+                if jobs_on_cluster:
+                    jid = jobs_on_cluster[0]["job_id"]
+                    chunks = []
+                    for c in agent.tail_logs(jid, follow=False):
+                        chunks.append(c)
+                        if sum(len(x) for x in chunks) > 1 << 20:
+                            break
+                    out["task_log"] = b"".join(chunks).decode(
+                        errors="replace")[-65536:]
+            except Exception as e:  # noqa: BLE001
+                out["task_log_error"] = str(e)
+    return out
+
+
 def cancel(job_ids: Optional[List[int]] = None,
            all_jobs: bool = False) -> int:
     jobs = state.list_jobs()

@@ -150,6 +150,12 @@ def jobs_cancel(job_ids: Optional[List[int]] = None,
     return jobs_server.cancel(job_ids, all_jobs)
 
 
+@register("jobs_logs", SHORT)
+def jobs_logs(job_id: int):
+    from skypilot_amd.jobs import server as jobs_server
+    return jobs_server.logs(job_id)
+
+
 # ---- serve ----------------------------------------------------------------
 @register("serve_up", LONG)
 def serve_up(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:

@@ -1,0 +1,55 @@
+"""CLI surface tests via click's CliRunner + the in-process harness."""
+import time
+
+from click.testing import CliRunner
+
+from tests.test_orchestrator import client, sky_env  # noqa: F401
+
+
+def test_cli_launch_status_queue_down(client, tmp_path):
+    from skypilot_amd.cli.main import cli
+    r = CliRunner()
+    yaml_path = tmp_path / "t.yaml"
+    yaml_path.write_text("name: cli-test\nrun: echo cli-ok\n")
+    res = r.invoke(cli, ["launch", str(yaml_path), "-c", "t-cli",
+                         "--detach-run"])
+    assert res.exit_code == 0, res.output
+    assert "job_id=1" in res.output
+
+    res = r.invoke(cli, ["status"])
+    assert res.exit_code == 0 and "t-cli" in res.output
+
+    res = r.invoke(cli, ["queue", "t-cli"])
+    assert res.exit_code == 0 and "cli-test" in res.output
+
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        res = r.invoke(cli, ["queue", "t-cli"])
+        if "SUCCEEDED" in res.output:
+            break
+        time.sleep(0.5)
+    assert "SUCCEEDED" in res.output
+
+    res = r.invoke(cli, ["logs", "t-cli", "1", "--no-follow"])
+    assert "cli-ok" in res.output
+
+    res = r.invoke(cli, ["show-gpus"])
+    assert res.exit_code == 0 and "MI355X" in res.output
+
+    res = r.invoke(cli, ["check"])
+    assert res.exit_code == 0 and "local" in res.output
+
+    res = r.invoke(cli, ["recipes"])
+    assert res.exit_code == 0 and "hello" in res.output
+
+    res = r.invoke(cli, ["down", "t-cli"])
+    assert res.exit_code == 0
+
+
+def test_cli_inline_command(client):
+    from skypilot_amd.cli.main import cli
+    r = CliRunner()
+    res = r.invoke(cli, ["launch", "echo inline-run", "-c", "t-inline",
+                         "--detach-run"])
+    assert res.exit_code == 0, res.output
+    r.invoke(cli, ["down", "t-inline"])
