@@ -187,3 +187,44 @@ def test_job_start_latency_metric(client):
     events = sdk.get(sdk.cluster_events("t-lat"))
     assert any(e["event"] == "JOB_SUBMIT" for e in events)
     sdk.get(sdk.down("t-lat"))
+
+
+def test_cancel_long_request(client):
+    """LONG requests run in their own process and must be killable via
+    /api/cancel (reference: per-request process, executor.py:302)."""
+    import time as _t
+    from skypilot_amd.client import sdk
+    from skypilot_amd.server import requests_db as rdb
+    # A launch whose job would run forever; cancel the REQUEST while the
+    # runner process is provisioning/submitting.
+    rid = sdk.launch({"name": "c", "run": "sleep 5"}, "t-cancelreq")
+    # wait until it is RUNNING in a child process
+    deadline = _t.time() + 30
+    while _t.time() < deadline:
+        req = rdb.get(rid)
+        if req["status"] == "RUNNING" and req.get("worker_pid"):
+            break
+        if req["status"] in ("SUCCEEDED", "FAILED"):
+            break  # too fast to cancel; fine
+        _t.sleep(0.05)
+    if req["status"] == "RUNNING":
+        assert sdk.cancel_request(rid)
+        req = rdb.get(rid)
+        assert req["status"] == "CANCELLED"
+    # server request table lists it either way
+    r = client.get("/api/requests")
+    assert any(x["request_id"] == rid for x in r.json())
+    # metrics endpoint exposes counts
+    m = client.get("/metrics").text
+    assert "sky_amd_requests_total" in m
+    # cleanup if the cluster came up
+    try:
+        sdk.get(sdk.down("t-cancelreq"), timeout=60)
+    except Exception:
+        pass
+
+
+def test_dashboard_renders(client):
+    r = client.get("/dashboard")
+    assert r.status_code == 200
+    assert "skypilot-amd" in r.text and "Clusters" in r.text
