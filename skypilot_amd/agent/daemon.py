@@ -37,9 +37,17 @@ def create_app(cluster_dir: str, gpu_ids: list[int]) -> FastAPI:
              "gpu_ids": gpu_ids}
 
     # ---- scheduler thread (FIFO + GPU accounting) -------------------------
+    driver_procs: list = []
+
     def scheduler_loop():
         while True:
             try:
+                # Reap finished/killed driver processes so PID liveness
+                # checks don't see zombies (a SIGKILLed driver must be
+                # detected as dead by reconcile()).
+                for pr in driver_procs[:]:
+                    if pr.poll() is not None:
+                        driver_procs.remove(pr)
                 table.reconcile()
                 free = [g for g in state["gpu_ids"]
                         if g not in table.allocated_gpus()]
@@ -58,11 +66,11 @@ def create_app(cluster_dir: str, gpu_ids: list[int]) -> FastAPI:
                                   (json.dumps(spec), job_lib.INIT,
                                    job["job_id"]))
                     dlog = open(Path(cluster_dir) / "driver.log", "ab")
-                    subprocess.Popen(
+                    driver_procs.append(subprocess.Popen(
                         [sys.executable, "-m", "skypilot_amd.agent.driver",
                          cluster_dir, str(job["job_id"])],
                         stdout=dlog, stderr=subprocess.STDOUT,
-                        start_new_session=True)
+                        start_new_session=True))
                     dlog.close()
                     state["last_active"] = time.time()
             except Exception as e:  # noqa: BLE001

@@ -1,0 +1,56 @@
+"""Fault-injection tests (reference: tests/chaos/chaos_proxy.py — here we
+kill framework processes directly and assert the state machines
+reconcile)."""
+import json
+import os
+import signal
+import time
+from pathlib import Path
+
+from tests.test_orchestrator import client, sky_env, _wait_job_done  # noqa: F401
+
+
+def test_driver_killed_marks_failed_driver(client):
+    """Kill the job driver process: reconcile must mark FAILED_DRIVER
+    (reference: job_lib.py:833 liveness by driver PID)."""
+    from skypilot_amd.client import sdk
+    res = sdk.get(sdk.launch({"name": "victim", "run": "sleep 600"},
+                             "t-chaos1"), timeout=60)
+    jid = res["job_id"]
+    deadline = time.time() + 30
+    driver_pid = None
+    while time.time() < deadline:
+        j = sdk.get(sdk.job_status("t-chaos1", jid))
+        if j and j["status"] == "RUNNING" and j.get("driver_pid"):
+            driver_pid = j["driver_pid"]
+            break
+        time.sleep(0.3)
+    assert driver_pid, "job never started"
+    os.kill(driver_pid, signal.SIGKILL)  # exact pid, never a pattern
+    j = _wait_job_done("t-chaos1", jid, timeout=60)
+    assert j["status"] == "FAILED_DRIVER"
+    sdk.get(sdk.down("t-chaos1"))
+
+
+def test_agent_killed_detected_by_refresh(client):
+    """Kill the node agent: `sky status -r` must reconcile the cluster to
+    STOPPED; `sky start` restarts the agent on the same GPU lease."""
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"name": "al", "run": "true",
+                        "resources": {"accelerators": "MI355X:2"}},
+                       "t-chaos2"), timeout=60)
+    home = Path(os.environ["SKY_AMD_HOME"])
+    meta = home / "clusters" / "t-chaos2" / "agent.json"
+    pid = json.loads(meta.read_text())["pid"]
+    os.kill(pid, signal.SIGKILL)
+    time.sleep(0.5)
+    records = sdk.get(sdk.status(refresh=True))
+    rec = next(r for r in records if r["name"] == "t-chaos2")
+    assert rec["status"] == "STOPPED"
+    # restart reuses the recorded GPU lease
+    handle = sdk.get(sdk.start("t-chaos2"))
+    assert handle["gpu_ids"] == rec["handle"]["gpu_ids"]
+    records = sdk.get(sdk.status())
+    rec = next(r for r in records if r["name"] == "t-chaos2")
+    assert rec["status"] == "UP"
+    sdk.get(sdk.down("t-chaos2"))
