@@ -113,6 +113,13 @@ def run_job(cluster_dir: str, job_id: int) -> int:
         p._logf = f  # keep ref
         procs.append(p)
 
+    # Record node-process pgids so the agent can reap orphans if this
+    # driver dies (reference: skylet/subprocess_daemon.py).
+    spec["node_pids"] = [p.pid for p in procs]
+    with table._conn() as c:
+        c.execute("UPDATE jobs SET spec=? WHERE job_id=?",
+                  (json.dumps(spec), job_id))
+
     def forward_term(signum, frame):
         for p in procs:
             try:

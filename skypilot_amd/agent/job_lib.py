@@ -166,11 +166,17 @@ class JobTable:
 
     def reconcile(self):
         """Mark jobs whose driver died without reporting as FAILED_DRIVER
-        (reference: job_lib.py:833/:850 update_job_status)."""
+        and reap their orphaned node processes (reference:
+        job_lib.py:833/:850 + skylet/subprocess_daemon.py)."""
         for j in self.active_jobs():
             pid = j.get("driver_pid")
             if pid and not _pid_alive(pid):
                 self.set_status(j["job_id"], FAILED_DRIVER, exit_code=-1)
+                for npid in j["spec"].get("node_pids", []):
+                    try:
+                        os.killpg(npid, signal.SIGTERM)
+                    except (ProcessLookupError, PermissionError):
+                        pass
 
 
 def _pid_alive(pid: int) -> bool:
