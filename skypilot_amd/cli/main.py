@@ -283,6 +283,14 @@ def serve_up(entrypoint, service_name, env):
     _print_result(sdk.stream_and_get(sdk.serve_up(task, service_name)))
 
 
+@serve.command("update")
+@click.argument("entrypoint")
+@click.option("--service-name", "-n", required=True)
+def serve_update(entrypoint, service_name):
+    task = _load_task(entrypoint, (), {})
+    _print_result(sdk.get(sdk.serve_update(task, service_name)))
+
+
 @serve.command("down")
 @click.argument("service_name")
 def serve_down(service_name):

@@ -259,6 +259,11 @@ def serve_up(task, service_name: str) -> str:
                                 "service_name": service_name})
 
 
+def serve_update(task, service_name: str) -> str:
+    return _submit("serve_update", {"task": _task_body(task),
+                                    "service_name": service_name})
+
+
 def serve_down(service_name: str) -> str:
     return _submit("serve_down", {"service_name": service_name})
 

@@ -52,7 +52,25 @@ def _free_port() -> int:
 def run_instances(cluster_name: str, num_nodes: int, accelerator: str | None,
                   acc_count: int, existing_handle: Optional[Dict] = None
                   ) -> Dict[str, Any]:
-    """Allocate GPUs + start the agent. Returns the cluster handle."""
+    """Allocate GPUs + start the agent. Returns the cluster handle.
+    Guarded by a per-cluster file lock (reference: utils/locks.py used at
+    cloud_vm_ray_backend.py:3433) so concurrent launches can't
+    double-allocate GPUs or double-start the agent."""
+    import fcntl
+    lock_dir = global_state.root_dir() / "locks"
+    lock_dir.mkdir(parents=True, exist_ok=True)
+    lock_f = open(lock_dir / f"provision-{cluster_name}.lock", "w")
+    fcntl.flock(lock_f, fcntl.LOCK_EX)
+    try:
+        return _run_instances_locked(cluster_name, num_nodes, accelerator,
+                                     acc_count, existing_handle)
+    finally:
+        fcntl.flock(lock_f, fcntl.LOCK_UN)
+        lock_f.close()
+
+
+def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
+                          existing_handle=None) -> Dict[str, Any]:
     gpus = detect_gpus()
     need = num_nodes * acc_count
     if existing_handle and existing_handle.get("gpu_ids") is not None:

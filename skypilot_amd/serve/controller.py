@@ -74,12 +74,19 @@ class ServiceRuntime:
         return Task.from_yaml_config(cfg)
 
     def _launch_replica(self):
+        svc = st.get_service(self.name) or {}
+        version = svc.get("version", 1)
+        # Pick up the latest task config (rolling updates).
+        if svc.get("task"):
+            self.task_cfg = dict(svc["task"])
+            self.spec = ServiceSpec.from_config(self.task_cfg.get("service")
+                                                or {})
         rid = self._next_rid
         self._next_rid += 1
         port = _free_port()
         cluster = f"sky-serve-{self.name}-{rid}"
         st.upsert_replica(self.name, rid, status=st.R_PROVISIONING,
-                          cluster_name=cluster,
+                          cluster_name=cluster, version=version,
                           endpoint=f"http://127.0.0.1:{port}",
                           launched_at=time.time())
         try:
@@ -163,6 +170,21 @@ class ServiceRuntime:
         for r in replicas:
             if r["status"] in (st.R_FAILED, st.R_NOT_READY):
                 self._terminate_replica(r["replica_id"], r["cluster_name"])
+        # Rolling update: when the service version moved past a READY
+        # replica's version, replace stale replicas one at a time, only
+        # while a newer or equal-count READY capacity exists.
+        cur_version = (svc or {}).get("version", 1)
+        replicas = st.list_replicas(self.name)
+        ready_now = [r for r in replicas if r["status"] == st.R_READY]
+        stale_ready = [r for r in ready_now
+                       if r.get("version", 1) < cur_version]
+        fresh_ready = [r for r in ready_now
+                       if r.get("version", 1) >= cur_version]
+        if stale_ready and (fresh_ready or len(ready_now) > 1 or
+                            self.spec.policy.min_replicas == 1):
+            victim = stale_ready[0]
+            self._terminate_replica(victim["replica_id"],
+                                    victim["cluster_name"])
         replicas = st.list_replicas(self.name)
         alive = [r for r in replicas if r["status"] in
                  (st.R_PROVISIONING, st.R_STARTING, st.R_READY)]

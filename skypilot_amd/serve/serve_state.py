@@ -34,6 +34,7 @@ CREATE TABLE IF NOT EXISTS services (
     spec TEXT NOT NULL,
     controller_pid INTEGER,
     lb_port INTEGER,
+    version INTEGER DEFAULT 1,
     created_at REAL
 );
 CREATE TABLE IF NOT EXISTS replicas (
@@ -42,6 +43,7 @@ CREATE TABLE IF NOT EXISTS replicas (
     status TEXT NOT NULL,
     cluster_name TEXT,
     endpoint TEXT,
+    version INTEGER DEFAULT 1,
     launched_at REAL,
     PRIMARY KEY (service, replica_id)
 );
@@ -66,9 +68,22 @@ def add_service(name: str, task: Dict[str, Any], spec: Dict[str, Any],
     with _conn() as c:
         c.execute(
             "INSERT OR REPLACE INTO services "
-            "(name,status,task,spec,lb_port,created_at) VALUES (?,?,?,?,?,?)",
+            "(name,status,task,spec,lb_port,version,created_at) "
+            "VALUES (?,?,?,?,?,1,?)",
             (name, CONTROLLER_INIT, json.dumps(task), json.dumps(spec),
              lb_port, time.time()))
+
+
+def bump_version(name: str, task: Dict[str, Any],
+                 spec: Dict[str, Any]) -> int:
+    """Rolling update: record the new task/spec and bump the service
+    version (reference: serve replica_managers version tracking)."""
+    with _conn() as c:
+        c.execute("UPDATE services SET task=?, spec=?, version=version+1 "
+                  "WHERE name=?", (json.dumps(task), json.dumps(spec), name))
+        row = c.execute("SELECT version FROM services WHERE name=?",
+                        (name,)).fetchone()
+    return row[0] if row else 0
 
 
 def update_service(name: str, **fields) -> None:

@@ -51,6 +51,17 @@ def up(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:
             "min_replicas": spec.policy.min_replicas}
 
 
+def update(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:
+    svc = st.get_service(service_name)
+    if svc is None:
+        raise ServeError(f"service {service_name!r} not found")
+    if not task.get("service"):
+        raise ServeError("task YAML needs a `service:` section")
+    ServiceSpec.from_config(task["service"])  # validate
+    version = st.bump_version(service_name, task, task["service"])
+    return {"service_name": service_name, "version": version}
+
+
 def down(service_name: str) -> None:
     svc = st.get_service(service_name)
     if svc is None:

@@ -163,6 +163,12 @@ def serve_up(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:
     return serve_server.up(task, service_name)
 
 
+@register("serve_update", LONG)
+def serve_update(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:
+    from skypilot_amd.serve import server as serve_server
+    return serve_server.update(task, service_name)
+
+
 @register("serve_down", LONG)
 def serve_down(service_name: str) -> None:
     from skypilot_amd.serve import server as serve_server
