@@ -109,10 +109,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     for (int qt = qt0; qt < S / BM; ++qt) {
       const int qbase = qt * BM;
       __syncthreads();
-      // Stage Q/dO row-wise + their transposes via transposed global
-      // reads (coalesced 128B per instruction) -> vector LDS writes; the
-      // previous 8-way scalar scatter was the dominant bank-conflict
-      // source (see attn_fwd staging comment).
+      // Stage Q, QT, dO, dOT (swizzled).
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         int idx = tid + i * 256;
@@ -121,18 +118,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         *(s16x8*)((char*)q_lds + swz(row * 256 + ch * 16, row)) = qv;
         s16x8 dov = *(const s16x8*)(dOb + (long long)(qbase + row) * q_rowstride + ch * 8);
         *(s16x8*)((char*)do_lds + swz(row * 256 + ch * 16, row)) = dov;
-        int d = lane + 64 * (i & 1);
-        int tch = w + 4 * (i >> 1);
-        s16x8 qt, dot;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          long long src = (long long)(qbase + tch * 8 + j) * q_rowstride + d;
-          qt[j] = (short)Qb[src];
-          dot[j] = (short)dOb[src];
+          int d = ch * 8 + j;
+          *(unsigned short*)((char*)qt_lds + swzT(d * 128 + row * 2, d)) =
+              (unsigned short)qv[j];
+          *(unsigned short*)((char*)dot_lds + swzT(d * 128 + row * 2, d)) =
+              (unsigned short)dov[j];
         }
-        int toff = swzT(d * 128 + tch * 16, d);
-        *(s16x8*)((char*)qt_lds + toff) = qt;
-        *(s16x8*)((char*)dot_lds + toff) = dot;
       }
       __syncthreads();
 
@@ -292,15 +285,12 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
       *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
       s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
       *(s16x8*)((char*)v_lds + swz(row * 256 + ch * 16, row)) = vv8;
-      // K^T via transposed global reads (see attn_fwd staging comment).
-      int d = lane + 64 * (i & 1);
-      int tch = w + 4 * (i >> 1);
-      s16x8 ktv;
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        ktv[j] = (short)Kb[(long long)(kvbase + tch * 8 + j) *
-                           kv_rowstride + d];
-      *(s16x8*)((char*)kt_lds + swzT(d * 128 + tch * 16, d)) = ktv;
+      for (int j = 0; j < 8; ++j) {
+        int d = ch * 8 + j;
+        *(unsigned short*)((char*)kt_lds + swzT(d * 128 + row * 2, d)) =
+            (unsigned short)kv8[j];
+      }
     }
     __syncthreads();
 

@@ -73,52 +73,44 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   // Async-STAGE split (guide §6 G15 / T14): each tile's K/V global loads
   // are issued one iteration early into registers, hiding HBM latency
   // under the previous tile's QK^T + softmax; the LDS writes happen at
-  // the top of the owning iteration.
-  //
-  // K is fetched row-wise (thread t owns chunks (row t/16 + 16i, 16B
-  // chunk t%16)).  V is fetched TRANSPOSED from global — per
-  // instruction every lane reads 2B of one kv-row at consecutive d, a
-  // fully-coalesced 128B segment — so the V^T tile is built with vector
-  // b128 LDS writes instead of 8-way scalar scatter (the scatter's bank
-  // conflicts were the #1 cost in the r01 PMC profile).
-  //   V^T mapping: iter i: d = lane + 64*(i&1), chunk = wave + 4*(i>>1);
-  //   vpre[i][j] = V[kvbase + chunk*8 + j][d].
+  // the top of the owning iteration.  Thread t owns chunks
+  // (row t/16 + 16i, 16B-chunk t%16), i = 0..3.
   s16x8 kpre[4], vpre[4];
   const int pre_row = tid >> 4, pre_ch = tid & 15;
-  const int vt_d0 = lane;            // + 64*(i&1)
-  const int vt_ch0 = w;              // + 4*(i>>1)
-
-#define LOAD_TILE(kvb)                                                      \
-  do {                                                                      \
-    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                         \
-      kpre[i] = *(const s16x8*)(Kb + (long long)((kvb) + pre_row + 16 * i) *\
-                                    kv_rowstride + pre_ch * 8);             \
-      int d = vt_d0 + 64 * (i & 1);                                         \
-      int ch = vt_ch0 + 4 * (i >> 1);                                       \
-      _Pragma("unroll") for (int j = 0; j < 8; ++j)                         \
-        vpre[i][j] = (short)Vb[(long long)((kvb) + ch * 8 + j) *            \
-                               kv_rowstride + d];                           \
-    }                                                                       \
-  } while (0)
-
-  LOAD_TILE(0);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    long long r = (long long)(pre_row + 16 * i) * kv_rowstride + pre_ch * 8;
+    kpre[i] = *(const s16x8*)(Kb + r);
+    vpre[i] = *(const s16x8*)(Vb + r);
+  }
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    // ---- stage K tile [kv][d] + transposed V tile [d][kv] (swizzled).
+    // ---- stage K tile [kv][d] (swizzled) + V tile transposed [d][kv]
+    // from the prefetch registers.
     __syncthreads();
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int row = pre_row + 16 * i;
       *(s16x8*)((char*)k_lds + swz(row * 256 + pre_ch * 16, row)) = kpre[i];
-      int d = vt_d0 + 64 * (i & 1);
-      int ch = vt_ch0 + 4 * (i >> 1);
-      *(s16x8*)((char*)vt_lds + swzT(d * 128 + ch * 16, d)) = vpre[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = pre_ch * 8 + j;
+        *(unsigned short*)((char*)vt_lds + swzT(d * 128 + row * 2, d)) =
+            (unsigned short)vpre[i][j];
+      }
     }
     __syncthreads();
     if (kt + 1 < n_kv_tiles) {
       // Issue next tile's loads now; they stay in flight through QK^T +
       // softmax (the sync before PV drains vmcnt).
-      LOAD_TILE((kt + 1) * BN);
+      const long long base = (long long)(kt + 1) * BN * kv_rowstride;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        long long r = base + (long long)(pre_row + 16 * i) * kv_rowstride +
+                      pre_ch * 8;
+        kpre[i] = *(const s16x8*)(Kb + r);
+        vpre[i] = *(const s16x8*)(Vb + r);
+      }
     }
     const int kvbase = kt * BN;
 
