@@ -74,6 +74,10 @@ def main() -> int:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    peak_gb = torch.cuda.max_memory_allocated() / 1e9
+    print(f"[bench] rank {rank}: peak GPU memory {peak_gb:.1f} GB",
+          file=sys.stderr, flush=True)
+
     tokens = args.steps * cfg.micro_batch * cfg.seq_len * world
     value = tokens / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
