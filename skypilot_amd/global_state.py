@@ -37,9 +37,14 @@ CREATE TABLE IF NOT EXISTS clusters (
     resources TEXT NOT NULL,
     launched_at REAL,
     last_use TEXT,
+    user TEXT,
     autostop_idle_minutes INTEGER DEFAULT -1,
     autostop_down INTEGER DEFAULT 0,
     to_down INTEGER DEFAULT 0
+);
+CREATE TABLE IF NOT EXISTS users (
+    name TEXT PRIMARY KEY,
+    created_at REAL
 );
 CREATE TABLE IF NOT EXISTS cluster_events (
     id INTEGER PRIMARY KEY AUTOINCREMENT,
@@ -89,21 +94,30 @@ def _conn():
         conn.close()
 
 
+def current_user() -> str:
+    """reference: sky/models.py User — here: env override or OS user."""
+    return os.environ.get("SKY_AMD_USER") or os.environ.get("USER", "root")
+
+
 def add_or_update_cluster(name: str, status: str, handle: Dict[str, Any],
                           resources: Dict[str, Any],
                           launched_at: Optional[float] = None) -> None:
+    user = current_user()
     with _DB_LOCK, _conn() as c:
         existing = c.execute("SELECT launched_at FROM clusters WHERE name=?",
                              (name,)).fetchone()
         if launched_at is None:
             launched_at = existing[0] if existing else time.time()
+        c.execute("INSERT OR IGNORE INTO users (name, created_at) "
+                  "VALUES (?,?)", (user, time.time()))
         c.execute(
-            "INSERT INTO clusters (name,status,handle,resources,launched_at)"
-            " VALUES (?,?,?,?,?) ON CONFLICT(name) DO UPDATE SET "
+            "INSERT INTO clusters "
+            "(name,status,handle,resources,launched_at,user)"
+            " VALUES (?,?,?,?,?,?) ON CONFLICT(name) DO UPDATE SET "
             "status=excluded.status, handle=excluded.handle, "
             "resources=excluded.resources, launched_at=excluded.launched_at",
             (name, status, json.dumps(handle), json.dumps(resources),
-             launched_at))
+             launched_at, user))
 
 
 def set_cluster_status(name: str, status: str) -> None:
@@ -122,7 +136,8 @@ def get_cluster(name: str) -> Optional[Dict[str, Any]]:
     with _DB_LOCK, _conn() as c:
         row = c.execute(
             "SELECT name,status,handle,resources,launched_at,"
-            "autostop_idle_minutes,autostop_down FROM clusters WHERE name=?",
+            "autostop_idle_minutes,autostop_down,user "
+            "FROM clusters WHERE name=?",
             (name,)).fetchone()
     if row is None:
         return None
@@ -138,6 +153,7 @@ def _row_to_cluster(row) -> Dict[str, Any]:
         "launched_at": row[4],
         "autostop_idle_minutes": row[5],
         "autostop_down": bool(row[6]),
+        "user": row[7],
     }
 
 
@@ -145,7 +161,7 @@ def list_clusters() -> List[Dict[str, Any]]:
     with _DB_LOCK, _conn() as c:
         rows = c.execute(
             "SELECT name,status,handle,resources,launched_at,"
-            "autostop_idle_minutes,autostop_down FROM clusters "
+            "autostop_idle_minutes,autostop_down,user FROM clusters "
             "ORDER BY launched_at DESC").fetchall()
     return [_row_to_cluster(r) for r in rows]
 
@@ -176,6 +192,12 @@ def get_cluster_events(cluster: str) -> List[Dict[str, Any]]:
             "SELECT ts,event,detail FROM cluster_events WHERE cluster=? "
             "ORDER BY ts", (cluster,)).fetchall()
     return [{"ts": r[0], "event": r[1], "detail": r[2]} for r in rows]
+
+
+def list_users() -> List[Dict[str, Any]]:
+    with _DB_LOCK, _conn() as c:
+        rows = c.execute("SELECT name, created_at FROM users").fetchall()
+    return [{"name": r[0], "created_at": r[1]} for r in rows]
 
 
 def set_config(key: str, value: Any) -> None:
