@@ -30,7 +30,7 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--micro-batch", type=int, default=4)
+    ap.add_argument("--micro-batch", type=int, default=6)
     ap.add_argument("--seq-len", type=int, default=4096)
     args = ap.parse_args()
 
