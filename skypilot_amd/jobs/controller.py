@@ -34,7 +34,14 @@ class JobController:
         if record is None:
             raise RuntimeError(f"managed job {managed_job_id} not found")
         self.record = record
-        self.task = Task.from_yaml_config(record["task"])
+        cfg = record["task"]
+        # Pipelines (reference: jobs support chained DAGs): a task config
+        # may carry `tasks: [...]` — run sequentially, each with recovery.
+        if isinstance(cfg, dict) and "tasks" in cfg:
+            self.pipeline = [Task.from_yaml_config(t) for t in cfg["tasks"]]
+        else:
+            self.pipeline = [Task.from_yaml_config(cfg)]
+        self.task = self.pipeline[0]
         jr = self.task.resources.job_recovery
         self.strategy = recovery.make(
             jr.strategy if jr else None,
@@ -47,7 +54,16 @@ class JobController:
         state.update(self.job_id, controller_pid=os.getpid(),
                      cluster_name=self.cluster_name)
         try:
-            final = self._run_with_recovery()
+            final = state.SUCCEEDED
+            for i, task in enumerate(self.pipeline):
+                self.task = task
+                jr = task.resources.job_recovery
+                self.strategy = recovery.make(
+                    jr.strategy if jr else None,
+                    jr.max_restarts_on_errors if jr else 0)
+                final = self._run_with_recovery()
+                if final != state.SUCCEEDED:
+                    break
             state.set_status(self.job_id, final)
         except BaseException as e:  # noqa: BLE001
             traceback.print_exc()

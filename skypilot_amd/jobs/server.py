@@ -16,7 +16,11 @@ from skypilot_amd.jobs import state
 def launch(task: Dict[str, Any], name: Optional[str] = None
            ) -> Dict[str, Any]:
     from skypilot_amd.task import Task
-    Task.from_yaml_config(dict(task))  # validate before persisting
+    if "tasks" in task:  # pipeline: validate each stage
+        for t in task["tasks"]:
+            Task.from_yaml_config(dict(t))
+    else:
+        Task.from_yaml_config(dict(task))  # validate before persisting
     job_id = state.create(name or task.get("name"), task)
     pkg_root = os.path.dirname(os.path.dirname(
         os.path.dirname(os.path.abspath(__file__))))

@@ -218,10 +218,206 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// 32x32x16 variant: 128 q rows per block (wave owns 32), kv tiles of 64.
+// Doubles FLOPs per MFMA and per staged byte vs the 16x16 kernel; used
+// whenever S % 128 == 0 (the flagship shapes).  Layouts verified by
+// tests/test_gpu_ops.py::test_mfma32_layout_probe.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define MFMA32_BF16(a, b, c) \
+  __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+#define BM32 128
+
+extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, unsigned short* __restrict__ O,
+    float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, float scale,
+    int causal) {
+  __shared__ unsigned short k_lds[BN * ATT_D];     // [kv][d]   16KB
+  __shared__ unsigned short vt_lds[ATT_D * BN];    // [d][kv]   16KB
+  __shared__ unsigned short p_lds[BM32 * BN];      // [q][kv]   16KB
+  // Online-softmax running max/sum per q row live in LDS: keeping 2x16
+  // floats per lane in VGPRs spilled to scratch (all 32 lanes of a row
+  // write identical post-reduce values, so the shared write is benign).
+  __shared__ float m_lds[BM32], l_lds[BM32];
+
+  const int qt = gridDim.x - 1 - blockIdx.x;  // heavy blocks first
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * BM32;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;        // wave owns q rows [32w, 32w+32)
+  const int col = lane & 31;
+  const int half = lane >> 5;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  // Q fragments: rows qbase+32w+col, 8 k-subtiles of 16.
+  s16x8 a_q[8];
+  {
+    const unsigned short* src =
+        Qb + (long long)(qbase + 32 * w + col) * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+      a_q[ks] = *(const s16x8*)(src + ks * 16 + half * 8);
+  }
+
+  f32x16 o_acc[4];
+#pragma unroll
+  for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) o_acc[nt][r] = 0.f;
+  for (int i = tid; i < BM32; i += 256) {
+    m_lds[i] = -INFINITY;
+    l_lds[i] = 0.f;
+  }
+
+  const int n_kv_tiles =
+      causal ? (qbase + BM32 + BN - 1) / BN : (S + BN - 1) / BN;
+
+  // No T14 reg-prefetch here: the 32x32 accumulators already fill the
+  // register budget (prefetch regs caused scratch spills).
+  const int pre_row = tid >> 4, pre_ch = tid & 15;
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kvbase = kt * BN;
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = pre_row + 16 * i;
+      long long src = (long long)(kvbase + row) * kv_rowstride + pre_ch * 8;
+      s16x8 kv8 = *(const s16x8*)(Kb + src);
+      *(s16x8*)((char*)k_lds + swz(row * 256 + pre_ch * 16, row)) = kv8;
+      s16x8 vv8 = *(const s16x8*)(Vb + src);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = pre_ch * 8 + j;
+        *(unsigned short*)((char*)vt_lds + swzT(d * 128 + row * 2, d)) =
+            (unsigned short)vv8[j];
+      }
+    }
+    __syncthreads();
+
+    // S = scale * Q K^T : wave's 32 rows x 64 cols (2 col-tiles of 32).
+    f32x16 s_acc[2];
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s_acc[nt][r] = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < 8; ++ks) {
+        int krow = nt * 32 + col;
+        s16x8 bfrag = *(const s16x8*)(
+            (char*)k_lds + swz(krow * 256 + (ks * 16 + half * 8) * 2, krow));
+        s_acc[nt] = MFMA32_BF16(as_bf16x8(a_q[ks]), as_bf16x8(bfrag),
+                                s_acc[nt]);
+      }
+    }
+
+    // Mask + online softmax; P streamed to LDS.
+    const bool need_mask =
+        (causal && kvbase + BN - 1 > qbase + 32 * w) || (kvbase + BN > S);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int my_row = qbase + 32 * w + (r & 3) + 8 * (r >> 2) + 4 * half;
+      float mx = -INFINITY;
+#pragma unroll
+      for (int nt = 0; nt < 2; ++nt) {
+        float s = s_acc[nt][r] * scale;
+        if (need_mask) {
+          int c2 = kvbase + nt * 32 + col;
+          if ((causal && c2 > my_row) || c2 >= S) s = -INFINITY;
+        }
+        s_acc[nt][r] = s;
+        mx = fmaxf(mx, s);
+      }
+#pragma unroll
+      for (int off = 16; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+      const int prow = 32 * w + (r & 3) + 8 * (r >> 2) + 4 * half;
+      float m_old = m_lds[prow];
+      float m_new = fmaxf(m_old, mx);
+      float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
+      float corr = (m_old == -INFINITY) ? 0.f : __expf(m_old - m_safe);
+      m_lds[prow] = m_new;
+      float rs = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < 2; ++nt) {
+        float e = (s_acc[nt][r] == -INFINITY)
+                      ? 0.f : __expf(s_acc[nt][r] - m_safe);
+        rs += e;
+        int pcol = nt * 32 + col;
+        *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+            f2bf_trunc(e);
+      }
+#pragma unroll
+      for (int off = 16; off > 0; off >>= 1) rs += __shfl_xor(rs, off, 64);
+      l_lds[prow] = l_lds[prow] * corr + rs;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) o_acc[nt][r] *= corr;
+    }
+    __syncthreads();
+
+    // O += P V : A-frags from p_lds (4, reused), B from vt_lds.
+    s16x8 pa[4];
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      int prow = 32 * w + col;
+      pa[ks] = *(const s16x8*)(
+          (char*)p_lds + swz(prow * 128 + (ks * 16 + half * 8) * 2, prow));
+    }
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int vrow = nt * 32 + col;
+        s16x8 bfrag = *(const s16x8*)(
+            (char*)vt_lds + swzT(vrow * 128 + (ks * 16 + half * 8) * 2,
+                                 vrow));
+        o_acc[nt] = MFMA32_BF16(as_bf16x8(pa[ks]), as_bf16x8(bfrag),
+                                o_acc[nt]);
+      }
+  }
+
+  // Epilogue.
+  unsigned short* Ob = O + ((long long)b * S * Hq + qh) * ATT_D;
+  float* lse_b = lse_out + ((long long)b * Hq + qh) * S;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int prow = 32 * w + (r & 3) + 8 * (r >> 2) + 4 * half;
+    const int qrow = qbase + prow;
+    float lv = l_lds[prow];
+    float inv_l = (lv > 0.f) ? 1.f / lv : 0.f;
+    unsigned short* orow = Ob + (long long)qrow * q_rowstride;
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+      orow[nt * 32 + col] = f2bf(o_acc[nt][r] * inv_l);
+    if (col == 0)
+      lse_b[qrow] = (lv > 0.f) ? m_lds[prow] + __logf(lv) : -INFINITY;
+  }
+}
+
 extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
                                 void* O, float* lse, int B, int S, int Hq,
                                 int Hkv, float scale, bool causal,
                                 hipStream_t stream) {
+  if (S % BM32 == 0) {
+    dim3 grid(S / BM32, B * Hq);
+    hipLaunchKernelGGL(attn_fwd_kernel32, grid, dim3(256), 0, stream,
+                       (const unsigned short*)Q, (const unsigned short*)K,
+                       (const unsigned short*)V, (unsigned short*)O, lse, B,
+                       S, Hq, Hkv, scale, causal ? 1 : 0);
+    return;
+  }
   dim3 grid(S / BM, B * Hq);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
                      (const unsigned short*)Q, (const unsigned short*)K,

@@ -37,6 +37,7 @@ void attn_bwd_launch(const void*, const void*, const void*, const void*,
                      const void*, const float*, float*, void*, void*, void*,
                      int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
+void mfma32_probe_launch(const void*, const void*, float*, hipStream_t);
 void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
                        hipStream_t);
@@ -215,6 +216,16 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+torch::Tensor mfma32_probe(torch::Tensor A, torch::Tensor B) {
+  CHECK_GPU(A); CHECK_BF16(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
+  TORCH_CHECK(A.size(0) == 32 && A.size(1) == 16);
+  TORCH_CHECK(B.size(0) == 16 && B.size(1) == 32);
+  auto C = torch::zeros({32, 32}, A.options().dtype(at::kFloat));
+  mfma32_probe_launch(A.data_ptr(), B.data_ptr(), C.data_ptr<float>(),
+                      cur_stream());
+  return C;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
@@ -227,4 +238,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("mfma32_probe", &mfma32_probe);
 }

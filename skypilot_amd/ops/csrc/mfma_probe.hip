@@ -26,3 +26,38 @@ extern "C" void mfma_probe_launch(const void* A, const void* B, float* C,
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
                      (const unsigned short*)A, (const unsigned short*)B, C);
 }
+
+// 32x32x16 bf16 probe — layouts (verified by tests/test_gpu_ops.py):
+//   A[32x16]: lane l holds A[l&31][(l>>5)*8 + j]          j=0..7
+//   B[16x32]: lane l holds B[(l>>5)*8 + j][l&31]
+//   C/D[32x32]: lane l reg r holds C[(r&3)+8*(r>>2)+4*(l>>5)][l&31]
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define MFMA32_BF16(a, b, c) \
+  __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+
+extern "C" __global__ void mfma32_probe_kernel(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
+    float* __restrict__ C) {
+  const int lane = threadIdx.x & 63;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+  s16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = (short)A[col * 16 + half * 8 + j];       // A[l&31][(l>>5)*8+j]
+    b[j] = (short)B[(half * 8 + j) * 32 + col];     // B[(l>>5)*8+j][l&31]
+  }
+  f32x16 c;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) c[r] = 0.f;
+  c = MFMA32_BF16(as_bf16x8(a), as_bf16x8(b), c);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    C[((r & 3) + 8 * (r >> 2) + 4 * half) * 32 + col] = c[r];
+}
+
+extern "C" void mfma32_probe_launch(const void* A, const void* B, float* C,
+                                    hipStream_t stream) {
+  hipLaunchKernelGGL(mfma32_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const unsigned short*)A, (const unsigned short*)B, C);
+}

@@ -213,3 +213,12 @@ def test_swiglu_fwd_bwd():
     y_ref.backward(dy0)
     assert rel_err(y.cpu(), y_ref) < 2e-2
     assert rel_err(gu.grad.cpu(), gu_ref.grad) < 2e-2
+
+
+def test_mfma32_layout_probe():
+    torch.manual_seed(10)
+    A = (torch.randn(32, 16) * 0.5).bfloat16().to(dev())
+    B = (torch.randn(16, 32) * 0.5).bfloat16().to(dev())
+    C = ops.native().mfma32_probe(A, B)
+    expect = A.float() @ B.float()
+    assert rel_err(C, expect) < 2e-2, (C[:3, :3], expect[:3, :3])

@@ -107,3 +107,21 @@ def test_managed_job_preemption_recovery(client, tmp_path):
     # The mounted store kept phase1's state across the preemption.
     home = Path(os.environ["SKY_AMD_HOME"])
     assert (home / "storage" / "mj-preempt-ckpt" / "state").exists()
+
+
+def test_managed_job_pipeline(client, tmp_path):
+    """Chained tasks run sequentially on the controller (reference:
+    pipelines via chained DAGs)."""
+    from skypilot_amd.client import sdk
+    out = tmp_path / "pipe.txt"
+    task = {
+        "name": "mj-pipe",
+        "tasks": [
+            {"name": "stage1", "run": f"echo one >> {out}"},
+            {"name": "stage2", "run": f"echo two >> {out}"},
+        ],
+    }
+    res = sdk.get(sdk.jobs_launch(task, "mj-pipe"))
+    job = _wait_managed(res["job_id"], {"SUCCEEDED", "FAILED"}, timeout=120)
+    assert job["status"] == "SUCCEEDED"
+    assert out.read_text().split() == ["one", "two"]
