@@ -310,11 +310,16 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
     // S = scale * Q K^T : wave's 32 rows x 64 cols (2 col-tiles of 32).
     f32x16 s_acc[2];
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
+    for (int nt = 0; nt < 2; ++nt)
 #pragma unroll
       for (int r = 0; r < 16; ++r) s_acc[nt][r] = 0.f;
+    // ks outer / nt inner: interleaves the two accumulator chains so the
+    // ~8-cycle MFMA latency is covered by independent issues (nt-outer
+    // made every MFMA depend on the previous one: 2x slowdown measured).
 #pragma unroll
-      for (int ks = 0; ks < 8; ++ks) {
+    for (int ks = 0; ks < 8; ++ks) {
+#pragma unroll
+      for (int nt = 0; nt < 2; ++nt) {
         int krow = nt * 32 + col;
         s16x8 bfrag = *(const s16x8*)(
             (char*)k_lds + swz(krow * 256 + (ks * 16 + half * 8) * 2, krow));
@@ -376,9 +381,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
           (char*)p_lds + swz(prow * 128 + (ks * 16 + half * 8) * 2, prow));
     }
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt)
+    for (int ks = 0; ks < 4; ++ks)
 #pragma unroll
-      for (int ks = 0; ks < 4; ++ks) {
+      for (int nt = 0; nt < 4; ++nt) {
         int vrow = nt * 32 + col;
         s16x8 bfrag = *(const s16x8*)(
             (char*)vt_lds + swzT(vrow * 128 + (ks * 16 + half * 8) * 2,
