@@ -415,7 +415,13 @@ extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
                                 void* O, float* lse, int B, int S, int Hq,
                                 int Hkv, float scale, bool causal,
                                 hipStream_t stream) {
-  if (S % BM32 == 0) {
+  // NOTE: attn_fwd_kernel32 (32x32x16 MFMA, 128-row blocks) measured
+  // SLOWER than this 16x16 kernel (97 vs 207 TF/s at the bench shape)
+  // despite 2x the arithmetic intensity — the 16-row serialized softmax
+  // and deeper accumulator chains dominate at 2 waves/SIMD.  Kept
+  // compiled (and layout-verified by tests) as the starting point for a
+  // wave-specialized rewrite; dispatch stays on the 16x16 kernel.
+  if (false && S % BM32 == 0) {
     dim3 grid(S / BM32, B * Hq);
     hipLaunchKernelGGL(attn_fwd_kernel32, grid, dim3(256), 0, stream,
                        (const unsigned short*)Q, (const unsigned short*)K,
