@@ -268,6 +268,33 @@ def jobs_cancel(job_ids, all_jobs):
     click.echo(f"Cancelled {n} managed job(s).")
 
 
+@jobs.group("pool")
+def jobs_pool():
+    """Worker pools for managed jobs."""
+
+
+@jobs_pool.command("apply")
+@click.argument("entrypoint")
+@click.option("--pool", "-p", "name", required=True)
+@click.option("--workers", type=int, default=2)
+def jobs_pool_apply(entrypoint, name, workers):
+    task = _load_task(entrypoint, (), {})
+    _print_result(sdk.get(sdk.jobs_pool_apply(name, task, workers)))
+
+
+@jobs_pool.command("status")
+@click.argument("name", required=False)
+def jobs_pool_status(name):
+    _print_result(sdk.get(sdk.jobs_pool_status(name)))
+
+
+@jobs_pool.command("down")
+@click.argument("name")
+def jobs_pool_down(name):
+    n = sdk.get(sdk.jobs_pool_down(name))
+    click.echo(f"Tore down {n} pool worker(s).")
+
+
 # ---- serve ----------------------------------------------------------------
 @cli.group()
 def serve():

@@ -254,6 +254,20 @@ def jobs_logs(job_id: int) -> str:
     return _submit("jobs_logs", {"job_id": job_id})
 
 
+def jobs_pool_apply(name: str, template, num_workers: int = 2) -> str:
+    return _submit("jobs_pool_apply", {"name": name,
+                                       "template": _task_body(template),
+                                       "num_workers": num_workers})
+
+
+def jobs_pool_status(name: Optional[str] = None) -> str:
+    return _submit("jobs_pool_status", {"name": name})
+
+
+def jobs_pool_down(name: str) -> str:
+    return _submit("jobs_pool_down", {"name": name})
+
+
 def serve_up(task, service_name: str) -> str:
     return _submit("serve_up", {"task": _task_body(task),
                                 "service_name": service_name})

@@ -35,6 +35,8 @@ class JobController:
             raise RuntimeError(f"managed job {managed_job_id} not found")
         self.record = record
         cfg = record["task"]
+        self.pool = cfg.get("pool") if isinstance(cfg, dict) else None
+        self.worker_cluster = None
         # Pipelines (reference: jobs support chained DAGs): a task config
         # may carry `tasks: [...]` — run sequentially, each with recovery.
         if isinstance(cfg, dict) and "tasks" in cfg:
@@ -73,12 +75,34 @@ class JobController:
 
     def _launch_cluster(self):
         state.set_status(self.job_id, state.STARTING)
+        if self.pool:
+            # Pool mode (reference: `sky jobs pool`): claim a warm worker
+            # and just exec — no provisioning on the job path.
+            from skypilot_amd.jobs import pools
+            while True:
+                cluster = pools.acquire(self.pool, self.job_id)
+                if cluster is not None:
+                    break
+                me = state.get(self.job_id)
+                if me and me["status"] == state.CANCELLED:
+                    raise RuntimeError("cancelled while queued for pool")
+                time.sleep(1.0)
+            self.worker_cluster = cluster
+            state.update(self.job_id, cluster_name=cluster)
+            job_id, handle = execution.exec_(self.task, cluster,
+                                             detach_run=True)
+            return job_id, handle
         job_id, handle = execution.launch(
             self.task, self.cluster_name, detach_run=True,
             managed_job_id=self.job_id)
         return job_id, handle
 
     def _teardown_cluster(self):
+        if self.pool:
+            from skypilot_amd.jobs import pools
+            pools.release(self.pool, self.job_id)
+            self.worker_cluster = None
+            return
         record = global_state.get_cluster(self.cluster_name)
         if record is not None:
             try:

@@ -16,6 +16,11 @@ from skypilot_amd.jobs import state
 def launch(task: Dict[str, Any], name: Optional[str] = None
            ) -> Dict[str, Any]:
     from skypilot_amd.task import Task
+    if task.get("pool"):
+        from skypilot_amd.jobs import pools
+        if not pools.exists(task["pool"]):
+            raise ValueError(f"pool {task['pool']!r} does not exist; "
+                             "create it with `sky jobs pool apply`")
     if "tasks" in task:  # pipeline: validate each stage
         for t in task["tasks"]:
             Task.from_yaml_config(dict(t))

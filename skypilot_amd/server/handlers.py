@@ -156,6 +156,25 @@ def jobs_logs(job_id: int):
     return jobs_server.logs(job_id)
 
 
+@register("jobs_pool_apply", LONG)
+def jobs_pool_apply(name: str, template: Dict[str, Any],
+                    num_workers: int = 2):
+    from skypilot_amd.jobs import pools
+    return pools.apply(name, template, num_workers)
+
+
+@register("jobs_pool_status", SHORT)
+def jobs_pool_status(name: Optional[str] = None):
+    from skypilot_amd.jobs import pools
+    return pools.status(name)
+
+
+@register("jobs_pool_down", LONG)
+def jobs_pool_down(name: str) -> int:
+    from skypilot_amd.jobs import pools
+    return pools.down(name)
+
+
 # ---- serve ----------------------------------------------------------------
 @register("serve_up", LONG)
 def serve_up(task: Dict[str, Any], service_name: str) -> Dict[str, Any]:

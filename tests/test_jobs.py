@@ -125,3 +125,32 @@ def test_managed_job_pipeline(client, tmp_path):
     job = _wait_managed(res["job_id"], {"SUCCEEDED", "FAILED"}, timeout=120)
     assert job["status"] == "SUCCEEDED"
     assert out.read_text().split() == ["one", "two"]
+
+
+def test_jobs_pool(client, tmp_path):
+    """Pool workers: warm clusters; pooled jobs exec on free workers and
+    queue when the pool is busy (reference: sky jobs pool, App. C)."""
+    from skypilot_amd.client import sdk
+    template = {"name": "pool-tmpl", "resources": {}}
+    res = sdk.get(sdk.jobs_pool_apply("p1", template, 1), timeout=90)
+    assert len(res["workers"]) == 1
+
+    out = tmp_path / "pool-out.txt"
+    j1 = sdk.get(sdk.jobs_launch(
+        {"name": "pj1", "pool": "p1",
+         "run": f"echo first >> {out}; sleep 3"}))
+    j2 = sdk.get(sdk.jobs_launch(
+        {"name": "pj2", "pool": "p1", "run": f"echo second >> {out}"}))
+    r1 = _wait_managed(j1["job_id"], {"SUCCEEDED", "FAILED"}, timeout=120)
+    r2 = _wait_managed(j2["job_id"], {"SUCCEEDED", "FAILED"}, timeout=120)
+    assert r1["status"] == "SUCCEEDED" and r2["status"] == "SUCCEEDED"
+    # One worker => serialized: first finished before second started.
+    assert out.read_text().split() == ["first", "second"]
+    st = sdk.get(sdk.jobs_pool_status("p1"))
+    assert st[0]["workers"][0]["status"] == "READY"
+    # Worker cluster survives the jobs (warm pool).
+    records = sdk.get(sdk.status())
+    assert any(r["name"] == "sky-pool-p1-0" for r in records)
+    sdk.get(sdk.jobs_pool_down("p1"))
+    records = sdk.get(sdk.status())
+    assert not any(r["name"].startswith("sky-pool-p1") for r in records)
