@@ -274,44 +274,25 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
   for (int ct = 0; ct < 8; ++ct) dq_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int n_kv_tiles = causal ? (qbase + BM + BN - 1) / BN : S / BN;
-  // T14 async-stage split: K/V tiles for iteration kt+1 are loaded into
-  // registers while iteration kt computes (guide §6 G15; VGPR headroom
-  // exists here: 162 regs before this change).
-  s16x8 kpre[4], vpre[4];
-  const int pre_row = tid >> 4, pre_ch = tid & 15;
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    long long src = (long long)(pre_row + 16 * i) * kv_rowstride +
-                    pre_ch * 8;
-    kpre[i] = *(const s16x8*)(Kb + src);
-    vpre[i] = *(const s16x8*)(Vb + src);
-  }
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int kvbase = kt * BN;
     __syncthreads();
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int row = pre_row + 16 * i;
-      *(s16x8*)((char*)k_lds + swz(row * 256 + pre_ch * 16, row)) = kpre[i];
-      *(s16x8*)((char*)v_lds + swz(row * 256 + pre_ch * 16, row)) = vpre[i];
+      int idx = tid + i * 256;
+      int row = idx >> 4, ch = idx & 15;
+      s16x8 kv8 = *(const s16x8*)(Kb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+      *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
+      s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+      *(s16x8*)((char*)v_lds + swz(row * 256 + ch * 16, row)) = vv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int d = pre_ch * 8 + j;
+        int d = ch * 8 + j;
         *(unsigned short*)((char*)kt_lds + swzT(d * 128 + row * 2, d)) =
-            (unsigned short)kpre[i][j];
+            (unsigned short)kv8[j];
       }
     }
     __syncthreads();
-    if (kt + 1 < n_kv_tiles) {
-      const long long base = (long long)(kt + 1) * BN * kv_rowstride;
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        long long src = base + (long long)(pre_row + 16 * i) *
-                        kv_rowstride + pre_ch * 8;
-        kpre[i] = *(const s16x8*)(Kb + src);
-        vpre[i] = *(const s16x8*)(Vb + src);
-      }
-    }
 
     // S = Q K^T (raw), dP = dO V^T; both [q 16][kv 64] per wave.
     f32x4 s_acc[4], dp[4];
