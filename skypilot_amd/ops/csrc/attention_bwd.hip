@@ -112,8 +112,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       // Stage Q, QT, dO, dOT (swizzled).
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        int idx = tid + i * 256;
-        int row = idx >> 4, ch = idx & 15;
+        // 8-row x 8-chunk staging map: scatter writes hit all 32 banks
+        // (see attn_fwd staging comment).
+        int ch = (tid & 7) | ((i & 1) << 3);
+        int row = (tid >> 3) | ((i >> 1) << 5);
         s16x8 qv = *(const s16x8*)(Qb + (long long)(qbase + row) * q_rowstride + ch * 8);
         *(s16x8*)((char*)q_lds + swz(row * 256 + ch * 16, row)) = qv;
         s16x8 dov = *(const s16x8*)(dOb + (long long)(qbase + row) * q_rowstride + ch * 8);
@@ -279,8 +281,8 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
     __syncthreads();
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int idx = tid + i * 256;
-      int row = idx >> 4, ch = idx & 15;
+      int ch = (tid & 7) | ((i & 1) << 3);
+      int row = (tid >> 3) | ((i >> 1) << 5);
       s16x8 kv8 = *(const s16x8*)(Kb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
       *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
       s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);

@@ -75,11 +75,17 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   // under the previous tile's QK^T + softmax; the LDS writes happen at
   // the top of the owning iteration.  Thread t owns chunks
   // (row t/16 + 16i, 16B-chunk t%16), i = 0..3.
+  // Staging map: chunk = (tid&7)|((i&1)<<3), row = (tid>>3)|((i>>1)<<5).
+  // A wave then covers 8 rows x 8 chunks per instruction: the transposed
+  // scatter writes spread over 4(row-words) x 8(d-groups) = 32 banks
+  // (the old 4-row map peaked at 16 banks = 4-way conflicts), while the
+  // global reads stay 128B-segment coalesced.
   s16x8 kpre[4], vpre[4];
-  const int pre_row = tid >> 4, pre_ch = tid & 15;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
-    long long r = (long long)(pre_row + 16 * i) * kv_rowstride + pre_ch * 8;
+    int s_ch = (tid & 7) | ((i & 1) << 3);
+    int s_row = (tid >> 3) | ((i >> 1) << 5);
+    long long r = (long long)s_row * kv_rowstride + s_ch * 8;
     kpre[i] = *(const s16x8*)(Kb + r);
     vpre[i] = *(const s16x8*)(Vb + r);
   }
@@ -90,12 +96,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     __syncthreads();
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      int row = pre_row + 16 * i;
-      *(s16x8*)((char*)k_lds + swz(row * 256 + pre_ch * 16, row)) = kpre[i];
+      int s_ch = (tid & 7) | ((i & 1) << 3);
+      int s_row = (tid >> 3) | ((i >> 1) << 5);
+      *(s16x8*)((char*)k_lds + swz(s_row * 256 + s_ch * 16, s_row)) =
+          kpre[i];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int d = pre_ch * 8 + j;
-        *(unsigned short*)((char*)vt_lds + swzT(d * 128 + row * 2, d)) =
+        int d = s_ch * 8 + j;
+        *(unsigned short*)((char*)vt_lds + swzT(d * 128 + s_row * 2, d)) =
             (unsigned short)vpre[i][j];
       }
     }
@@ -106,8 +114,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       const long long base = (long long)(kt + 1) * BN * kv_rowstride;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        long long r = base + (long long)(pre_row + 16 * i) * kv_rowstride +
-                      pre_ch * 8;
+        int s_ch = (tid & 7) | ((i & 1) << 3);
+        int s_row = (tid >> 3) | ((i >> 1) << 5);
+        long long r = base + (long long)s_row * kv_rowstride + s_ch * 8;
         kpre[i] = *(const s16x8*)(Kb + r);
         vpre[i] = *(const s16x8*)(Vb + r);
       }
