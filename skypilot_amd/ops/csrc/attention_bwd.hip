@@ -169,7 +169,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
             pv = __expf(sv - l);
           int prow = 16 * w + lgrp * 4 + r;
           int pcol = ct * 16 + lrow;
-          int off = swz(prow * 128 + pcol * 2, prow);
+          int off = swzP(prow * 128 + pcol * 2, prow);
           *(unsigned short*)((char*)p_lds + off) = f2bf_trunc(pv);
           *(unsigned short*)((char*)ds_lds + off) =
               f2bf(pv * (dpt[ct][r] - dv));
@@ -183,7 +183,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
           int prow = 16 * w + lrow;
-          int a_off = swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow);
+          int a_off = swzP(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow);
           int brow = ct * 16 + lrow;
           int b_off = swzT(brow * 128 + (ks * 32 + lgrp * 8) * 2, brow);
           s16x8 pfrag = *(const s16x8*)((char*)p_lds + a_off);
@@ -330,7 +330,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
         float ds = pv * (dp[ct][r] - my_dvec[r]);
         int prow = 16 * w + lgrp * 4 + r;
         int pcol = ct * 16 + lrow;
-        *(unsigned short*)((char*)ds_lds + swz(prow * 128 + pcol * 2, prow)) =
+        *(unsigned short*)((char*)ds_lds + swzP(prow * 128 + pcol * 2, prow)) =
             f2bf(ds);  // dS can be negative/large: keep rounded convert
       }
     __syncthreads();

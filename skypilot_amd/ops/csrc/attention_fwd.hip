@@ -178,7 +178,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         rs[r] += e;
         int prow = 16 * w + lgrp * 4 + r;
         int pcol = ct * 16 + lrow;
-        *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+        *(unsigned short*)((char*)p_lds + swzP(prow * 128 + pcol * 2, prow)) =
             f2bf_trunc(e);
       }
 #pragma unroll
@@ -370,7 +370,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
                       ? 0.f : __expf(s_acc[nt][r] - m_safe);
         rs += e;
         int pcol = nt * 32 + col;
-        *(unsigned short*)((char*)p_lds + swz(prow * 128 + pcol * 2, prow)) =
+        *(unsigned short*)((char*)p_lds + swzP(prow * 128 + pcol * 2, prow)) =
             f2bf_trunc(e);
       }
 #pragma unroll
@@ -387,7 +387,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
     for (int ks = 0; ks < 4; ++ks) {
       int prow = 32 * w + col;
       pa[ks] = *(const s16x8*)(
-          (char*)p_lds + swz(prow * 128 + (ks * 16 + half * 8) * 2, prow));
+          (char*)p_lds + swzP(prow * 128 + (ks * 16 + half * 8) * 2, prow));
     }
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks)

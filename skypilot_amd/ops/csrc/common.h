@@ -57,6 +57,15 @@ __device__ __forceinline__ int swzT(int byte_off, int row) {
   return byte_off ^ (((row >> 3) & 7) << 4);
 }
 
+// Swizzle for the P / dS staging tiles ([q or kv rows][64] C-layout
+// scatter): writes vary the row as lgrp*4+r (only 2 distinct (row&7)
+// per instruction) but cols span 16 — keying on (row>>1) gives 4 row
+// classes x 8 col-words = all 32 banks on the write side while the
+// 16-consecutive-row fragment reads stay balanced.
+__device__ __forceinline__ int swzP(int byte_off, int row) {
+  return byte_off ^ (((row >> 1) & 7) << 4);
+}
+
 // bf16 <-> f32 via bit ops (we deliberately avoid __hip_bfloat16 so these
 // headers stay independent of HIP half/bf16 operator macros).
 __device__ __forceinline__ float bf2f(unsigned short h) {
