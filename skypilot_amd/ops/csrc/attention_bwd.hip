@@ -342,7 +342,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
       for (int ks = 0; ks < 2; ++ks) {
         int prow = 16 * w + lrow;
         s16x8 afrag = *(const s16x8*)((char*)ds_lds +
-                                      swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
+                                      swzP(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
         int krow = ct * 16 + lrow;
         s16x8 bfrag = *(const s16x8*)((char*)kt_lds +
                                       swzT(krow * 128 + (ks * 32 + lgrp * 8) * 2, krow));

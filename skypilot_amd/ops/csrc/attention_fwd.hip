@@ -202,7 +202,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       for (int ks = 0; ks < 2; ++ks) {
         int prow = 16 * w + lrow;
         s16x8 afrag = *(const s16x8*)((char*)p_lds +
-                                      swz(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
+                                      swzP(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow));
         int vrow = ct * 16 + lrow;
         s16x8 bfrag = *(const s16x8*)((char*)vt_lds +
                                       swzT(vrow * 128 + (ks * 32 + lgrp * 8) * 2, vrow));
