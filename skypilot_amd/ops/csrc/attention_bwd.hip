@@ -153,7 +153,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       // Streamed straight to LDS (no pt/dst register arrays — keeping
       // them cost 32 VGPRs and dropped occupancy to 1 wave/SIMD).
       const int my_kvrow = kvbase + 16 * w + lgrp * 4;  // + r
-      __syncthreads();  // prior-iteration mfma reads of p_lds/ds_lds done
+      // No barrier needed before the P/dS writes: the prior iteration's
+      // dV/dK reads of these buffers completed before this iteration's
+      // top-of-loop barrier (two barriers ago).
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct) {
         int qcol = qbase + ct * 16 + lrow;
@@ -315,8 +317,9 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
     }
 
     // dS = P * (dP - Dvec);  P = exp(scale*S - lse).
+    // (No barrier: prior-iteration dQ reads of ds_lds finished before the
+    // top-of-loop barrier.)
     const int my_qrow = qbase + 16 * w + lgrp * 4;
-    __syncthreads();
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct)
 #pragma unroll
