@@ -325,6 +325,19 @@ def serve_down(service_name):
     click.echo(f"Service {service_name} torn down.")
 
 
+@serve.command("logs")
+@click.argument("service_name")
+@click.option("--replica", type=int, default=None)
+def serve_logs(service_name, replica):
+    result = sdk.get(sdk.serve_logs(service_name, replica))
+    if result.get("controller_log"):
+        click.echo("=== controller ===")
+        click.echo(result["controller_log"])
+    for rid, log in (result.get("replica_logs") or {}).items():
+        click.echo(f"=== replica {rid} ===")
+        click.echo(log)
+
+
 @serve.command("status")
 @click.argument("service_name", required=False)
 def serve_status(service_name):

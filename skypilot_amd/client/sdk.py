@@ -282,6 +282,11 @@ def serve_down(service_name: str) -> str:
     return _submit("serve_down", {"service_name": service_name})
 
 
+def serve_logs(service_name: str, replica_id: Optional[int] = None) -> str:
+    return _submit("serve_logs", {"service_name": service_name,
+                                  "replica_id": replica_id})
+
+
 def serve_status(service_name: Optional[str] = None) -> str:
     return _submit("serve_status", {"service_name": service_name})
 

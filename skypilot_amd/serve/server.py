@@ -81,6 +81,42 @@ def down(service_name: str) -> None:
     st.remove_service(service_name)
 
 
+def logs(service_name: str, replica_id: int = None,
+         tail_lines: int = 200):
+    svc = st.get_service(service_name)
+    if svc is None:
+        raise ServeError(f"service {service_name!r} not found")
+    out = {"service": service_name}
+    log_path = global_state.root_dir() / f"serve-{service_name}.log"
+    if log_path.exists():
+        out["controller_log"] = "\n".join(
+            log_path.read_text(errors="replace").splitlines()[-tail_lines:])
+    reps = st.list_replicas(service_name)
+    if replica_id is not None:
+        reps = [r for r in reps if r["replica_id"] == replica_id]
+    rep_logs = {}
+    for r in reps:
+        rec = global_state.get_cluster(r["cluster_name"]) if             r.get("cluster_name") else None
+        if not rec:
+            continue
+        try:
+            from skypilot_amd.backends.pool_backend import PoolBackend
+            agent = PoolBackend()._agent(rec["handle"])
+            jobs = agent.get_job_queue()
+            if jobs:
+                chunks = []
+                for c in agent.tail_logs(jobs[0]["job_id"], follow=False):
+                    chunks.append(c)
+                    if sum(len(x) for x in chunks) > 1 << 20:
+                        break
+                rep_logs[r["replica_id"]] = b"".join(chunks).decode(
+                    errors="replace")[-32768:]
+        except Exception as e:  # noqa: BLE001
+            rep_logs[r["replica_id"]] = f"(error: {e})"
+    out["replica_logs"] = rep_logs
+    return out
+
+
 def status(service_name: Optional[str] = None) -> List[Dict[str, Any]]:
     st.reconcile()
     services = ([st.get_service(service_name)] if service_name

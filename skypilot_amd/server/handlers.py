@@ -194,6 +194,12 @@ def serve_down(service_name: str) -> None:
     serve_server.down(service_name)
 
 
+@register("serve_logs", SHORT)
+def serve_logs(service_name: str, replica_id: Optional[int] = None):
+    from skypilot_amd.serve import server as serve_server
+    return serve_server.logs(service_name, replica_id)
+
+
 @register("serve_status", SHORT)
 def serve_status(service_name: Optional[str] = None):
     from skypilot_amd.serve import server as serve_server

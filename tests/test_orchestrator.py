@@ -45,7 +45,7 @@ def client(sky_env):
     executor.stop_workers()
 
 
-def _wait_job_done(cluster, job_id, timeout=60):
+def _wait_job_done(cluster, job_id, timeout=120):
     from skypilot_amd.client import sdk
     deadline = time.time() + timeout
     while time.time() < deadline:
