@@ -35,6 +35,9 @@ def main() -> int:
         "SKY_AMD_CHECKPOINT_DIR"))
     ap.add_argument("--checkpoint-every", type=int, default=20)
     ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--grad-accum", type=int, default=1)
+    ap.add_argument("--warmup-steps", type=int, default=0)
+    ap.add_argument("--lr-decay-steps", type=int, default=0)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -42,7 +45,9 @@ def main() -> int:
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     cfg = TrainConfig(model=args.model, micro_batch=args.micro_batch,
                       seq_len=args.seq_len, lr=args.lr, device=device,
-                      tp=args.tp)
+                      tp=args.tp, grad_accum=args.grad_accum,
+                      warmup_steps=args.warmup_steps,
+                      lr_decay_steps=args.lr_decay_steps)
     tr = Trainer(cfg)
 
     snap = None

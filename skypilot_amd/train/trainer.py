@@ -31,6 +31,10 @@ class TrainConfig:
     bucket_mb: int = 64
     device: str = "cuda"
     tp: int = 1  # tensor-parallel degree (divides world size)
+    grad_accum: int = 1          # micro-steps per optimizer step
+    warmup_steps: int = 0        # linear LR warmup
+    lr_decay_steps: int = 0      # cosine decay horizon (0 = constant)
+    min_lr_ratio: float = 0.1
 
 
 def dist_env():
@@ -103,15 +107,38 @@ class Trainer:
         tok = tok.to(self.device, non_blocking=True)
         return tok[:, :-1].contiguous(), tok[:, 1:].contiguous()
 
+    def current_lr(self) -> float:
+        """Linear warmup + cosine decay (step_count is 0-based here)."""
+        import math as _math
+        c = self.cfg
+        step = self.step_count
+        lr = c.lr
+        if c.warmup_steps and step < c.warmup_steps:
+            return lr * (step + 1) / c.warmup_steps
+        if c.lr_decay_steps:
+            t = min(1.0, (step - c.warmup_steps) /
+                    max(1, c.lr_decay_steps - c.warmup_steps))
+            floor = c.lr * c.min_lr_ratio
+            return floor + 0.5 * (lr - floor) * (1 + _math.cos(_math.pi * t))
+        return lr
+
     def train_step(self, batch=None) -> float:
-        if batch is None:
-            batch = self.synthetic_batch()
-        tokens, targets = batch
+        """One optimizer step = cfg.grad_accum micro-steps (gradients
+        accumulate in the DDP buckets; the all-reduce fires on the final
+        micro-step only)."""
+        c = self.cfg
         self.ddp.zero_grad()
-        self.ddp.mark_step_start()
-        loss = self.model.loss(tokens, targets)
-        loss.backward()
-        self.ddp.finish()
+        loss = 0.0
+        for micro in range(c.grad_accum):
+            b = batch if batch is not None else self.synthetic_batch()
+            tokens, targets = b
+            self.ddp.mark_step_start(
+                accumulating=(micro < c.grad_accum - 1))
+            l = self.model.loss(tokens, targets)
+            (l / c.grad_accum if c.grad_accum > 1 else l).backward()
+            self.ddp.finish()
+            loss = float(l.detach())
+        self.opt.lr = self.current_lr()
         # An in-flight checkpoint snapshot reads optimizer state on a side
         # stream; order this step's in-place AdamW after those copies.
         ev = getattr(self, "_snapshot_event", None)
@@ -120,11 +147,12 @@ class Trainer:
             self._snapshot_event = None
         self.opt.step(grad_scale=self.ddp.grad_scale)
         self.step_count += 1
-        return float(loss.detach())
+        return loss
 
     def tokens_per_step(self) -> int:
         dp = self.world // self.tp
-        return self.cfg.micro_batch * self.cfg.seq_len * dp
+        return self.cfg.micro_batch * self.cfg.seq_len * dp * \
+            self.cfg.grad_accum
 
 
 def run_training(cfg: TrainConfig, steps: int, warmup: int = 2,

@@ -94,3 +94,47 @@ def test_ddp_two_process_gloo():
         p.join(timeout=300)
     for p in procs:
         assert p.exitcode == 0
+
+
+def test_grad_accum_matches_big_batch():
+    """grad_accum=2 with batch B must match one step at batch 2B."""
+    from skypilot_amd.train.trainer import TrainConfig, Trainer
+    torch.manual_seed(0)
+    cfg1 = TrainConfig(model="llama-debug", micro_batch=2, seq_len=64,
+                       device="cpu", seed=11, grad_accum=1, lr=1e-3)
+    cfg2 = TrainConfig(model="llama-debug", micro_batch=1, seq_len=64,
+                       device="cpu", seed=11, grad_accum=2, lr=1e-3)
+    tr1, tr2 = Trainer(cfg1), Trainer(cfg2)
+    tok = torch.randint(0, 512, (2, 65))
+    batch_big = (tok[:, :-1], tok[:, 1:].contiguous())
+    tr1.train_step(batch_big)
+    # same data split into two micro-batches
+    b1 = (tok[:1, :-1], tok[:1, 1:].contiguous())
+    b2 = (tok[1:, :-1], tok[1:, 1:].contiguous())
+    c = tr2.cfg
+    tr2.ddp.zero_grad()
+    tr2.ddp.mark_step_start(accumulating=True)
+    (tr2.model.loss(*b1) / 2).backward()
+    tr2.ddp.finish()
+    tr2.ddp.mark_step_start(accumulating=False)
+    (tr2.model.loss(*b2) / 2).backward()
+    tr2.ddp.finish()
+    tr2.opt.lr = tr2.current_lr()
+    tr2.opt.step(grad_scale=tr2.ddp.grad_scale)
+    p1 = tr1.model.lm_head.weight.detach()
+    p2 = tr2.model.lm_head.weight.detach()
+    assert torch.allclose(p1.float(), p2.float(), atol=2e-3, rtol=2e-3)
+
+
+def test_lr_schedule():
+    from skypilot_amd.train.trainer import TrainConfig, Trainer
+    cfg = TrainConfig(model="llama-debug", micro_batch=1, seq_len=64,
+                      device="cpu", lr=1e-3, warmup_steps=10,
+                      lr_decay_steps=100)
+    tr = Trainer(cfg)
+    tr.step_count = 0
+    assert abs(tr.current_lr() - 1e-4) < 1e-9       # warmup start
+    tr.step_count = 9
+    assert abs(tr.current_lr() - 1e-3) < 1e-9       # warmup end
+    tr.step_count = 100
+    assert abs(tr.current_lr() - 1e-4) < 1e-6       # cosine floor
