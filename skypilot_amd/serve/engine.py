@@ -39,7 +39,9 @@ class Request:
     prompt_ids: List[int]
     max_tokens: int = 64
     temperature: float = 0.0
+    top_p: float = 1.0
     stop_id: Optional[int] = None
+    stream_queue: Optional["queue.Queue"] = None  # per-token streaming
     id: str = field(default_factory=lambda: uuid.uuid4().hex[:12])
     out_ids: List[int] = field(default_factory=list)
     done: threading.Event = field(default_factory=threading.Event)
@@ -122,9 +124,12 @@ class Engine:
         ctx = InferenceContext(cache=self.cache, mode="prefill",
                                prefill_slot=slot, prefill_len=L)
         logits = self.model(toks, positions, ctx)
-        next_id = self._sample(logits[0, L - 1], req.temperature)
+        next_id = self._sample(logits[0, L - 1], req.temperature,
+                               req.top_p)
         self.cache.lens[slot] = L
         req.out_ids.append(next_id)
+        if req.stream_queue is not None:
+            req.stream_queue.put(next_id)
         req.first_token_at = time.time()
         self.active[slot] = req
         self.stats["prefill_tokens"] += L
@@ -152,14 +157,25 @@ class Engine:
         for i, slot in enumerate(slots):
             self.cache.lens[slot] += 1
             req = self.active[slot]
-            next_id = self._sample(logits[i, 0], req.temperature)
+            next_id = self._sample(logits[i, 0], req.temperature,
+                                   req.top_p)
             req.out_ids.append(next_id)
+            if req.stream_queue is not None:
+                req.stream_queue.put(next_id)
             self.stats["tokens_generated"] += 1
             self._maybe_finish(slot, req, next_id)
 
-    def _sample(self, logits: torch.Tensor, temperature: float) -> int:
+    def _sample(self, logits: torch.Tensor, temperature: float,
+                top_p: float = 1.0) -> int:
         if temperature and temperature > 0:
             probs = (logits.float() / temperature).softmax(-1)
+            if top_p < 1.0:
+                sp, idx = probs.sort(descending=True)
+                keep = (sp.cumsum(0) - sp) <= top_p
+                keep[0] = True
+                probs = torch.zeros_like(probs).scatter(
+                    0, idx[keep], sp[keep])
+                probs = probs / probs.sum()
             return int(torch.multinomial(probs, 1).item())
         return int(logits.argmax().item())
 
@@ -171,6 +187,8 @@ class Engine:
             self.cache.free(slot)
             self.free_slots.append(slot)
             req.finished_at = time.time()
+            if req.stream_queue is not None:
+                req.stream_queue.put(None)  # end-of-stream marker
             req.done.set()
 
     def _loop(self):

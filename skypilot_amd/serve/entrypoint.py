@@ -43,6 +43,9 @@ class CompletionRequest(BaseModel):
     prompt: str = ""
     max_tokens: int = 64
     temperature: float = 0.0
+    top_p: float = 1.0
+    stop: Optional[List[str]] = None
+    stream: bool = False
 
 
 class ChatMessage(BaseModel):
@@ -85,13 +88,49 @@ def create_app(engine: Engine, model_name: str) -> FastAPI:
 
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
+        if req.stream:
+            import json as _json
+            import queue as _queue
+
+            from fastapi.responses import StreamingResponse
+            from skypilot_amd.serve.engine import Request as _Req
+            q = _queue.Queue()
+            r = _Req(prompt_ids=tok.encode(req.prompt),
+                     max_tokens=req.max_tokens,
+                     temperature=req.temperature, top_p=req.top_p,
+                     stream_queue=q)
+            engine.submit(r)
+
+            def sse():
+                while True:
+                    item = q.get(timeout=600)
+                    if item is None:
+                        yield "data: [DONE]\n\n"
+                        return
+                    chunk = {"id": "cmpl-local",
+                             "object": "text_completion",
+                             "model": model_name,
+                             "choices": [{"index": 0,
+                                          "text": tok.decode([item]),
+                                          "finish_reason": None}]}
+                    yield f"data: {_json.dumps(chunk)}\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
         out, dt, n_prompt = _complete(req.prompt, req.max_tokens,
                                       req.temperature)
+        text = tok.decode(out)
+        finish = "length"
+        for stop in req.stop or []:
+            i = text.find(stop)
+            if i >= 0:
+                text = text[:i]
+                finish = "stop"
+                break
         return {
             "id": "cmpl-local", "object": "text_completion",
             "model": model_name,
-            "choices": [{"index": 0, "text": tok.decode(out),
-                         "finish_reason": "length"}],
+            "choices": [{"index": 0, "text": text,
+                         "finish_reason": finish}],
             "usage": {"prompt_tokens": n_prompt,
                       "completion_tokens": len(out),
                       "total_tokens": n_prompt + len(out),

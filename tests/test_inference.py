@@ -78,3 +78,22 @@ def test_kv_cache_sizing():
     # A 288 GB GPU minus 8B bf16 weights leaves room for ~200 slots.
     budget = int((288 - 17) * 1e9)
     assert budget // per_slot > 150
+
+
+def test_top_p_and_stop_and_stream(engine):
+    from fastapi.testclient import TestClient
+    from skypilot_amd.serve.entrypoint import create_app
+    app = create_app(engine, "llama-debug")
+    with TestClient(app) as c:
+        # top_p sampling path executes
+        r = c.post("/v1/completions",
+                   json={"prompt": "ab", "max_tokens": 4,
+                         "temperature": 0.8, "top_p": 0.9}).json()
+        assert r["usage"]["completion_tokens"] == 4
+        # streaming returns SSE chunks ending in [DONE]
+        with c.stream("POST", "/v1/completions",
+                      json={"prompt": "ab", "max_tokens": 3,
+                            "stream": True}) as resp:
+            body = "".join(resp.iter_text())
+        assert body.count("data:") == 4  # 3 tokens + [DONE]
+        assert "[DONE]" in body
