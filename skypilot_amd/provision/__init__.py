@@ -8,6 +8,8 @@ from typing import Any, Dict
 _IMPLS = {
     "local": "skypilot_amd.provision.local",
     "ssh": "skypilot_amd.provision.ssh_pool",
+    "kubernetes": "skypilot_amd.provision.k8s",
+    "k8s": "skypilot_amd.provision.k8s",
 }
 
 DEFAULT_CLOUD = "local"

@@ -81,3 +81,29 @@ def test_recipes_list():
     names = {r["name"] for r in rs}
     assert "hello" in names and "train_llama3_8b" in names
     assert recipes.get_recipe_path("hello").endswith("hello.yaml")
+
+
+def test_k8s_pod_manifest_rendering():
+    from skypilot_amd.provision import k8s
+    m = k8s.render_pod_manifest("c1", 4, {
+        "image": "img:1", "gpu_resource": "amd.com/gpu",
+        "namespace": "ns"})
+    assert m["metadata"]["name"] == "sky-amd-c1"
+    ctr = m["spec"]["containers"][0]
+    assert ctr["resources"]["limits"]["amd.com/gpu"] == 4
+    assert "--gpu-ids" in ctr["command"]
+    assert "0,1,2,3" in ctr["command"]
+    # CPU-only pod: no GPU limits
+    m0 = k8s.render_pod_manifest("c2", 0, {"image": "i",
+                                           "gpu_resource": "amd.com/gpu"})
+    assert m0["spec"]["containers"][0]["resources"] == {}
+
+
+def test_provision_dispatch_registry():
+    from skypilot_amd import provision
+    for name in ("local", "ssh", "kubernetes", "k8s"):
+        assert provision._impl(name) is not None
+    import pytest as _pytest
+    from skypilot_amd.exceptions import ResourcesUnavailableError
+    with _pytest.raises(ResourcesUnavailableError):
+        provision._impl("aws")
