@@ -364,6 +364,206 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// 32x32x16 dK/dV kernel: 512 threads (8 waves), role-split per q-tile —
+// waves 0-3 compute S^T tiles (+exp -> p_lds), waves 4-7 compute dP^T
+// tiles (+dS -> ds_lds), then all 8 accumulate their own (kv-half,
+// d-quarter) dV and dK 32x32 output tiles.  Same FLOPs as the 16x16
+// kernel at half the instruction count and half the per-FLOP barrier
+// cost; no online softmax (lse precomputed) so the 32x32 C-layout costs
+// nothing here.  Layouts verified by test_mfma32_layout_probe.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define MFMA32_BF16(a, b, c) \
+  __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+
+extern "C" __global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel32(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dK, unsigned short* __restrict__ dV, int B,
+    int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ unsigned short q_lds[BM * ATT_D];
+  __shared__ unsigned short qt_lds[ATT_D * BM];
+  __shared__ unsigned short do_lds[BM * ATT_D];
+  __shared__ unsigned short dot_lds[ATT_D * BM];
+  __shared__ unsigned short p_lds[BN * BM];
+  __shared__ unsigned short ds_lds[BN * BM];
+
+  const int kt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int kvh_head = bh % Hkv;
+  const int group = Hq / Hkv;
+  const int kvbase = kt * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;                 // 8 waves
+  const int col = lane & 31;
+  const int half = lane >> 5;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh_head) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh_head) * ATT_D;
+
+  // Role assignment.
+  const bool is_st = (w < 4);             // S^T waves vs dP^T waves
+  const int t_kvh = is_st ? (w >> 1) : ((w - 4) >> 1);  // kv half (ST/dPT)
+  const int t_qh = is_st ? (w & 1) : ((w - 4) & 1);     // q half (ST/dPT)
+  const int o_kvh = w >> 2;               // output tile: kv half
+  const int o_dq = w & 3;                 // output tile: d quarter
+
+  // A-operand fragments held in registers: K rows (ST waves) or V rows
+  // (dPT waves) for the wave's ST/dPT tile — fixed for the whole block.
+  s16x8 a_kv[8];
+  {
+    const unsigned short* src =
+        (is_st ? Kb : Vb) +
+        (long long)(kvbase + t_kvh * 32 + col) * kv_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+      a_kv[ks] = *(const s16x8*)(src + ks * 16 + half * 8);
+  }
+
+  f32x16 dv_acc, dk_acc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    dv_acc[r] = 0.f;
+    dk_acc[r] = 0.f;
+  }
+
+  for (int g = 0; g < group; ++g) {
+    const int qh_head = kvh_head * group + g;
+    const unsigned short* Qb = Q + ((long long)b * S * Hq + qh_head) * ATT_D;
+    const unsigned short* dOb =
+        dO + ((long long)b * S * Hq + qh_head) * ATT_D;
+    const float* lse_b = lse + ((long long)b * Hq + qh_head) * S;
+    const float* dvec_b = Dvec + (long long)b * S * Hq + qh_head;
+
+    const int qt0 = causal ? kvbase / BM : 0;
+    for (int qt = qt0; qt < S / BM; ++qt) {
+      const int qbase = qt * BM;
+      __syncthreads();
+      // Stage Q/QT/dO/dOT — 512 threads, 2 chunks each (same map as the
+      // 16x16 kernel: ch=(t&7)|((i&1)<<3), row=(t>>3)|((i>>1)<<5) with
+      // t = tid%256 and the high thread bit folded into i).
+#pragma unroll
+      for (int i2 = 0; i2 < 2; ++i2) {
+        int i = i2 * 2 + (tid >> 8);      // 0..3
+        int t = tid & 255;
+        int ch = (t & 7) | ((i & 1) << 3);
+        int row = (t >> 3) | ((i >> 1) << 5);
+        s16x8 qv = *(const s16x8*)(Qb + (long long)(qbase + row) *
+                                       q_rowstride + ch * 8);
+        *(s16x8*)((char*)q_lds + swz(row * 256 + ch * 16, row)) = qv;
+        s16x8 dov = *(const s16x8*)(dOb + (long long)(qbase + row) *
+                                        q_rowstride + ch * 8);
+        *(s16x8*)((char*)do_lds + swz(row * 256 + ch * 16, row)) = dov;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = ch * 8 + j;
+          *(unsigned short*)((char*)qt_lds + swzT(d * 128 + row * 2, d)) =
+              (unsigned short)qv[j];
+          *(unsigned short*)((char*)dot_lds + swzT(d * 128 + row * 2, d)) =
+              (unsigned short)dov[j];
+        }
+      }
+      __syncthreads();
+
+      // ST waves: st = K Q^T tile [kv 32][q 32]; dPT waves: dpt = V dO^T.
+      const unsigned short* b_src = is_st ? q_lds : do_lds;
+      f32x16 acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+#pragma unroll
+      for (int ks = 0; ks < 8; ++ks) {
+        int qrow = t_qh * 32 + col;
+        s16x8 bfrag = *(const s16x8*)(
+            (char*)b_src + swz(qrow * 256 + (ks * 16 + half * 8) * 2, qrow));
+        acc = MFMA32_BF16(as_bf16x8(a_kv[ks]), as_bf16x8(bfrag), acc);
+      }
+
+      if (is_st) {
+        // PT = exp(scale*ST - lse[q]) with causal mask -> p_lds.
+        const int qcol = qbase + t_qh * 32 + col;
+        const float l = lse_b[qcol];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kvrow = t_kvh * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          float pv;
+          if ((causal && qcol < kvbase + kvrow) || l == -INFINITY)
+            pv = 0.f;
+          else
+            pv = __expf(acc[r] * scale - l);
+          int prow = kvrow;
+          int pcol = t_qh * 32 + col;
+          *(unsigned short*)((char*)p_lds +
+                             swzP(prow * 128 + pcol * 2, prow)) =
+              f2bf_trunc(pv);
+        }
+      }
+      __syncthreads();
+
+      if (!is_st) {
+        // dST = PT * (dPT - Dvec[q]) -> ds_lds (reads PT written above).
+        const int qcol = qbase + t_qh * 32 + col;
+        const float dvq = dvec_b[(long long)qcol * Hq];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kvrow = t_kvh * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          int off = swzP(kvrow * 128 + (t_qh * 32 + col) * 2, kvrow);
+          float pv = bf2f(*(const unsigned short*)((char*)p_lds + off));
+          *(unsigned short*)((char*)ds_lds + off) =
+              f2bf(pv * (acc[r] - dvq));
+        }
+      }
+      // dV += PT x dO (A from p_lds, B from dot_lds) — PT is ready for
+      // all waves after the barrier above.
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int arow = o_kvh * 32 + col;
+        s16x8 afrag = *(const s16x8*)(
+            (char*)p_lds +
+            swzP(arow * 128 + (ks * 16 + half * 8) * 2, arow));
+        int brow = o_dq * 32 + col;
+        s16x8 bfrag = *(const s16x8*)(
+            (char*)dot_lds +
+            swzT(brow * 128 + (ks * 16 + half * 8) * 2, brow));
+        dv_acc = MFMA32_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dv_acc);
+      }
+      __syncthreads();
+      // dK += dST x Q (A from ds_lds, B from qt_lds).
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int arow = o_kvh * 32 + col;
+        s16x8 afrag = *(const s16x8*)(
+            (char*)ds_lds +
+            swzP(arow * 128 + (ks * 16 + half * 8) * 2, arow));
+        int brow = o_dq * 32 + col;
+        s16x8 bfrag = *(const s16x8*)(
+            (char*)qt_lds +
+            swzT(brow * 128 + (ks * 16 + half * 8) * 2, brow));
+        dk_acc = MFMA32_BF16(as_bf16x8(afrag), as_bf16x8(bfrag), dk_acc);
+      }
+    }
+  }
+
+  // Write this wave's dK/dV 32x32 tiles (bf16; scale folded into dK).
+  unsigned short* dKb = dK + ((long long)b * S * Hkv + kvh_head) * ATT_D;
+  unsigned short* dVb = dV + ((long long)b * S * Hkv + kvh_head) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kvrow =
+        kvbase + o_kvh * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+    const int d = o_dq * 32 + col;
+    dKb[(long long)kvrow * kv_rowstride + d] = f2bf(dk_acc[r] * scale);
+    dVb[(long long)kvrow * kv_rowstride + d] = f2bf(dv_acc[r]);
+  }
+}
+
 extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
                                 const void* O, const void* dO,
                                 const float* lse, float* Dvec, void* dQ,
@@ -374,8 +574,8 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
   hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3(membound_grid(rows, 4)),
                      dim3(256), 0, stream, (const unsigned short*)dO,
                      (const unsigned short*)O, Dvec, rows, Hq);
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(S / BN, B * Hkv), dim3(256), 0,
-                     stream, (const unsigned short*)Q,
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel32, dim3(S / BN, B * Hkv),
+                     dim3(512), 0, stream, (const unsigned short*)Q,
                      (const unsigned short*)K, (const unsigned short*)V,
                      (const unsigned short*)dO, lse, Dvec,
                      (unsigned short*)dK, (unsigned short*)dV, B, S, Hq, Hkv,
