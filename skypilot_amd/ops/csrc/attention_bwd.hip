@@ -574,8 +574,13 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
   hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3(membound_grid(rows, 4)),
                      dim3(256), 0, stream, (const unsigned short*)dO,
                      (const unsigned short*)O, Dvec, rows, Hq);
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel32, dim3(S / BN, B * Hkv),
-                     dim3(512), 0, stream, (const unsigned short*)Q,
+  // NOTE: attn_bwd_dkv_kernel32 (8-wave 32x32, half the instructions)
+  // measured 148 vs 163 TF/s — its 512-thread block leaves 1 block/CU,
+  // losing the cross-block barrier/compute overlap that two independent
+  // 256-thread blocks provide.  Kept for the round-2 deep-pipeline
+  // rewrite; dispatch stays on the 16x16 kernel.
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(S / BN, B * Hkv),
+                     dim3(256), 0, stream, (const unsigned short*)Q,
                      (const unsigned short*)K, (const unsigned short*)V,
                      (const unsigned short*)dO, lse, Dvec,
                      (unsigned short*)dK, (unsigned short*)dV, B, S, Hq, Hkv,
