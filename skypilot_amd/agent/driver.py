@@ -146,9 +146,31 @@ def run_job(cluster_dir: str, job_id: int) -> int:
     final = job_lib.SUCCEEDED if all(r == 0 for r in rcs) else job_lib.FAILED
     cur = table.get(job_id)
     if cur and cur["status"] == job_lib.CANCELLED:
-        return 0
-    table.set_status(job_id, final, exit_code=worst)
-    return worst
+        final = job_lib.CANCELLED
+    else:
+        table.set_status(job_id, final, exit_code=worst)
+    _run_event_callback(spec, task_id, job_id, final, workdir, log_dir)
+    return 0 if final == job_lib.CANCELLED else worst
+
+
+def _run_event_callback(spec, task_id, job_id, status, workdir, log_dir):
+    """reference: jobs/utils.py:1090 event callbacks with SKYPILOT_TASK_ID
+    / JOB_STATUS env."""
+    cb = spec.get("event_callback")
+    if not cb:
+        return
+    env = dict(os.environ)
+    env.update({
+        "SKYPILOT_TASK_ID": task_id,
+        "SKYPILOT_INTERNAL_JOB_ID": str(job_id),
+        "JOB_STATUS": status,
+    })
+    try:
+        with open(Path(log_dir) / "event_callback.log", "ab") as f:
+            subprocess.run(["bash", "-c", cb], cwd=workdir, env=env,
+                           stdout=f, stderr=subprocess.STDOUT, timeout=120)
+    except (OSError, subprocess.TimeoutExpired):
+        pass
 
 
 def main():

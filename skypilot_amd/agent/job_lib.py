@@ -130,7 +130,11 @@ class JobTable:
                 (pid, json.dumps(gpus), log_dir, job_id))
 
     def pending_jobs(self) -> List[Dict[str, Any]]:
-        return [j for j in reversed(self.list()) if j["status"] == PENDING]
+        """FIFO within priority class (reference: resources.priority)."""
+        out = [j for j in reversed(self.list()) if j["status"] == PENDING]
+        out.sort(key=lambda j: (-(j["spec"].get("priority") or 0),
+                                j["job_id"]))
+        return out
 
     def active_jobs(self) -> List[Dict[str, Any]]:
         return [j for j in self.list() if j["status"] in

@@ -88,6 +88,8 @@ class PoolBackend(Backend):
             "node_ips": handle.get("node_ips"),
             "task_id": task_id,
             "managed_job_id": managed_job_id,
+            "event_callback": task.event_callback,
+            "priority": task.resources.priority or 0,
         }
         job_id = agent.queue_job(spec, name=task.name)
         cluster = _cluster_of(handle)

@@ -107,19 +107,35 @@ def job_status(cluster_name: str, job_id: int) -> Optional[Dict[str, Any]]:
 
 
 def check() -> Dict[str, Any]:
-    """reference: sky/check.py — pool capability check."""
+    """reference: sky/check.py — per-pool capability check."""
     gpus = detect_gpus()
-    return {
-        "pools": {
-            "local": {
-                "enabled": True,
-                "gpus": [{"index": g.index, "name": g.name,
-                          "memory_gb": g.memory_gb,
-                          "numa_node": g.numa_node} for g in gpus],
-            },
+    pools: Dict[str, Any] = {
+        "local": {
+            "enabled": True,
+            "gpus": [{"index": g.index, "name": g.name,
+                      "memory_gb": g.memory_gb,
+                      "numa_node": g.numa_node} for g in gpus],
         },
-        "timestamp": time.time(),
     }
+    try:
+        from skypilot_amd.provision import ssh_pool
+        hosts = ssh_pool.parse_hosts()
+        pools["ssh"] = {"enabled": bool(hosts),
+                        "hosts": [h["ip"] for h in hosts]}
+    except Exception as e:  # noqa: BLE001
+        pools["ssh"] = {"enabled": False, "error": str(e)}
+    try:
+        import shutil as _sh
+        from skypilot_amd.provision import k8s
+        st = k8s.k8s_settings()
+        pools["kubernetes"] = {
+            "enabled": bool(_sh.which("kubectl") and st.get("image")),
+            "namespace": st.get("namespace"),
+            "image": st.get("image"),
+        }
+    except Exception as e:  # noqa: BLE001
+        pools["kubernetes"] = {"enabled": False, "error": str(e)}
+    return {"pools": pools, "timestamp": time.time()}
 
 
 def show_gpus() -> List[Dict[str, Any]]:

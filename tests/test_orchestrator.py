@@ -228,3 +228,31 @@ def test_dashboard_renders(client):
     r = client.get("/dashboard")
     assert r.status_code == 200
     assert "skypilot-amd" in r.text and "Clusters" in r.text
+
+
+def test_event_callback_and_priority(client, tmp_path):
+    from skypilot_amd.client import sdk
+    cb_out = tmp_path / "events.txt"
+    # Occupy the only schedulable slot order with a low-priority sleeper
+    # and a high-priority job: with 1-GPU tasks on the 8-GPU pool both
+    # run; instead verify priority ORDERING via the pending queue on a
+    # busy cluster (single 8-GPU allocation).
+    task_lo = {"name": "lo", "resources": {"accelerators": "MI355X:8"},
+               "run": "sleep 4"}
+    res = sdk.get(sdk.launch(task_lo, "t-prio"), timeout=90)
+    # Queue two more on the same cluster: low prio first, high prio second.
+    sdk.get(sdk.exec({"name": "late-lo", "run": f"echo lo >> {cb_out}",
+                      "resources": {"accelerators": "MI355X:8"}}, "t-prio"))
+    t_hi = {"name": "hi", "resources": {"accelerators": "MI355X:8",
+                                        "priority": 10},
+            "run": f"echo hi >> {cb_out}",
+            "event_callback": f"echo cb-$JOB_STATUS >> {cb_out}"}
+    sdk.get(sdk.exec(t_hi, "t-prio"))
+    for jid in (2, 3):
+        _wait_job_done("t-prio", jid, timeout=120)
+    lines = cb_out.read_text().split()
+    # high-priority job ran before the earlier-submitted low-prio one,
+    # and its event callback fired with the final status.
+    assert lines.index("hi") < lines.index("lo")
+    assert "cb-SUCCEEDED" in lines
+    sdk.get(sdk.down("t-prio"))
