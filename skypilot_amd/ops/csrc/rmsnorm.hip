@@ -124,11 +124,11 @@ extern "C" void rmsnorm_bwd_launch(const void* x, const void* w,
                                    const void* dy, const float* inv_rms,
                                    void* dx, float* dw, long long rows, int H,
                                    hipStream_t stream) {
-  // Grid sized for occupancy (256 CUs x ~8 blocks); dw atomic traffic is
-  // one add per column per block at the very end — 2048 adds/column over
-  // the kernel lifetime is well within fp32 atomic throughput (the old
-  // 512-block cap left the dx pass at 2 blocks/CU and ~1.9 TB/s).
-  int grid = rows < 2048 ? (int)rows : 2048;
+  // Grid capped at 512: measured A/B — 2048 blocks drop the kernel to
+  // 816 GB/s (the per-column atomicAdd tail serializes at 2048
+  // adds/address), 512 blocks reach 1.9 TB/s. A partial-buffer + second
+  // reduce kernel is the round-2 fix if this op ever matters more.
+  int grid = rows < 512 ? (int)rows : 512;
   const int nvec = H / 8;
   dim3 b(256);
 #define LAUNCH(NV)                                                          \
