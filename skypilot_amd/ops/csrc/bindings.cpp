@@ -24,8 +24,7 @@ extern "C" {
 void rmsnorm_fwd_launch(const void*, const void*, void*, float*, long long,
                         int, float, hipStream_t);
 void rmsnorm_bwd_launch(const void*, const void*, const void*, const float*,
-                        void*, float*, float*, int, long long, int,
-                        hipStream_t);
+                        void*, float*, long long, int, hipStream_t);
 void rope_launch(const void*, void*, const float*, const float*, const int*,
                  long long, int, int, bool, hipStream_t);
 void adamw_launch(void*, float*, const void*, float*, float*, long long,
@@ -69,14 +68,11 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
   const int H = (int)x.size(-1);
   long long rows = x.numel() / H;
   auto dx = torch::empty_like(x);
-  int grid = rows < 2048 ? (int)rows : 2048;
-  auto dw_partial = torch::empty({(long)grid, (long)H},
-                                 x.options().dtype(at::kFloat));
-  auto dw = torch::empty({(long)H}, x.options().dtype(at::kFloat));
+  auto dw = torch::zeros({(long)H},
+                         x.options().dtype(at::kFloat));
   rmsnorm_bwd_launch(x.data_ptr(), w.data_ptr(), dy.data_ptr(),
                      inv_rms.data_ptr<float>(), dx.data_ptr(),
-                     dw_partial.data_ptr<float>(), dw.data_ptr<float>(),
-                     grid, rows, H, cur_stream());
+                     dw.data_ptr<float>(), rows, H, cur_stream());
   return {dx, dw};
 }
 
