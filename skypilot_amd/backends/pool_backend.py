@@ -18,7 +18,7 @@ from skypilot_amd import global_state, provision
 from skypilot_amd.agent.client import AgentClient
 from skypilot_amd.backends.backend import Backend
 from skypilot_amd.data import storage as storage_lib
-from skypilot_amd.exceptions import ClusterDoesNotExist, ClusterNotUpError
+from skypilot_amd.exceptions import ClusterNotUpError
 from skypilot_amd.task import Task
 from skypilot_amd.utils.command_runner import LocalProcessCommandRunner
 

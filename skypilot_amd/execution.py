@@ -16,7 +16,6 @@ from skypilot_amd.backends.pool_backend import PoolBackend
 from skypilot_amd.dag import to_dag
 from skypilot_amd.exceptions import ClusterDoesNotExist, ClusterNotUpError
 from skypilot_amd.optimizer import Optimizer
-from skypilot_amd.task import Task
 from skypilot_amd.utils.timeline import event as timeline_event
 
 

@@ -17,7 +17,6 @@ from typing import Any, Dict, List, Optional
 
 from skypilot_amd import execution, global_state
 from skypilot_amd.backends.pool_backend import PoolBackend
-from skypilot_amd.exceptions import ManagedJobError
 from skypilot_amd.task import Task
 
 _SCHEMA = """

@@ -5,7 +5,6 @@ spawned on the API-server host (reference: utils/controller_utils.py:1422)
 from __future__ import annotations
 
 import os
-import signal
 import subprocess
 import sys
 from typing import Any, Dict, List, Optional

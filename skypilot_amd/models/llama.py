@@ -21,7 +21,6 @@ from dataclasses import dataclass
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from skypilot_amd import ops
 

@@ -15,7 +15,6 @@ Pool config: ~/.sky_amd/ssh_node_pools.yaml
 """
 from __future__ import annotations
 
-import json
 import os
 import shlex
 import socket

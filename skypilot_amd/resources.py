@@ -9,7 +9,6 @@ autostop, job_recovery, image_id, disk_size.  Multi-cloud catalog fields
 """
 from __future__ import annotations
 
-import re
 from dataclasses import dataclass, field
 from typing import Any, Dict, Optional
 

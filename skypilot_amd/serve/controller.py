@@ -11,7 +11,6 @@ Usage: python -m skypilot_amd.serve.controller <service_name>
 """
 from __future__ import annotations
 
-import asyncio
 import collections
 import os
 import socket
