@@ -127,3 +127,40 @@ def test_swiglu_cpu_matches_manual():
     assert torch.allclose(y, expect, atol=1e-5)
     y.sum().backward()
     assert gu.grad is not None
+
+
+def test_all_example_yamls_parse():
+    """Every shipped example must parse into a valid Task (or pipeline)."""
+    from pathlib import Path
+
+    import yaml as _yaml
+
+    from skypilot_amd.task import Task
+    ex = Path(__file__).parent.parent / "examples"
+    files = sorted(ex.glob("*.yaml"))
+    assert len(files) >= 7
+    for f in files:
+        cfg = _yaml.safe_load(f.read_text())
+        t = Task.from_yaml_config(cfg)
+        assert t.run or t.service, f
+
+
+def test_task_yaml_round_trip():
+    from skypilot_amd.task import Task
+    cfg = {
+        "name": "rt", "num_nodes": 2,
+        "resources": {"accelerators": "MI355X:4", "cpus": "8+",
+                      "job_recovery": {"strategy": "FAILOVER",
+                                       "max_restarts_on_errors": 2},
+                      "labels": {"team": "ml"}},
+        "envs": {"A": "1"},
+        "run": "echo hi",
+    }
+    t = Task.from_yaml_config(cfg)
+    out = t.to_yaml_config()
+    t2 = Task.from_yaml_config(out)
+    assert t2.num_nodes == 2
+    assert t2.resources.accelerator_count == 4
+    assert t2.resources.cpus == 8 and t2.resources.cpus_is_min
+    assert t2.resources.job_recovery.max_restarts_on_errors == 2
+    assert t2.envs == {"A": "1"}
