@@ -36,7 +36,7 @@ echo "=== job-start latency (warm exec) ===" >> $LOG
 T0=$(date +%s.%N)
 timeout 60 python -m skypilot_amd.cli exec e2e 'echo warm-exec-done' >> $LOG 2>&1
 T1=$(date +%s.%N)
-echo "warm exec wall: $(echo "$T1 - $T0" | bc)s" >> $LOG
+python3 -c "print(f'warm exec wall: {$T1 - $T0:.2f}s')" >> $LOG
 timeout 60 python -m skypilot_amd.cli down e2e >> $LOG 2>&1
 
 echo "=== sky serve GPU inference ===" >> $LOG
