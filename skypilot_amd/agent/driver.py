@@ -43,6 +43,7 @@ def run_job(cluster_dir: str, job_id: int) -> int:
     gpus_per_node = int(spec.get("gpus_per_node", 0))
     gpu_ids = spec.get("gpu_ids") or []
     workdir = spec.get("workdir") or str(Path(cluster_dir) / "workdir")
+    Path(workdir).mkdir(parents=True, exist_ok=True)
     log_dir = Path(cluster_dir) / "logs" / str(job_id)
     log_dir.mkdir(parents=True, exist_ok=True)
     run_cmd = spec.get("run") or ""

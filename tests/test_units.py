@@ -1,0 +1,85 @@
+"""Unit tests for under-covered pieces: agent HTTP surface (in-process),
+Dag, Resources multi-candidate, optimizer feasibility."""
+import time
+
+import pytest
+
+
+def test_agent_app_surface(tmp_path, monkeypatch):
+    """The skylet-equivalent RPC surface, exercised in-process
+    (reference surface: SURVEY.md Appendix A)."""
+    from fastapi.testclient import TestClient
+
+    from skypilot_amd.agent.daemon import create_app
+    app = create_app(str(tmp_path / "c1"), [0, 1])
+    with TestClient(app) as c:
+        h = c.get("/health").json()
+        assert h["ok"] and h["gpu_ids"] == [0, 1]
+        # queue a job that needs no GPU and runs instantly
+        r = c.post("/jobs/queue", json={
+            "name": "t", "spec": {"run": "echo unit-agent",
+                                  "num_nodes": 1, "gpus_per_node": 0}})
+        jid = r.json()["job_id"]
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            j = c.get(f"/jobs/{jid}").json()["job"]
+            if j["status"] in ("SUCCEEDED", "FAILED", "FAILED_DRIVER"):
+                break
+            time.sleep(0.3)
+        assert j["status"] == "SUCCEEDED", j
+        assert any(x["job_id"] == jid for x in c.get("/jobs").json()["jobs"])
+        # logs endpoint returns the output
+        text = c.get(f"/jobs/{jid}/logs",
+                     params={"follow": False}).text
+        assert "unit-agent" in text
+        # autostop state round-trips
+        c.post("/autostop", json={"idle_minutes": 7, "down": True})
+        a = c.get("/autostop").json()
+        assert a["idle_minutes"] == 7 and a["down"]
+        assert c.get("/idle").json()["idle"] is True
+
+
+def test_dag_context_and_chain():
+    import skypilot_amd.dag as dag_mod
+    from skypilot_amd.task import Task
+    with dag_mod.Dag("d") as d:
+        assert dag_mod.get_current_dag() is d
+        t1, t2 = Task("a", run="x"), Task("b", run="y")
+        d.add(t1)
+        d.add(t2)
+        d.add_edge(t1, t2)
+    assert dag_mod.get_current_dag() is None
+    assert d.is_chain() and len(d) == 2
+    d2 = dag_mod.to_dag(Task("solo", run="z"))
+    assert len(d2) == 1
+
+
+def test_resources_any_of_and_aliases():
+    from skypilot_amd.resources import Resources, canonical_accelerator
+    r = Resources.from_yaml_config({
+        "any_of": [{"accelerators": "MI355:4"},
+                   {"accelerators": "MI355X:8"}]})
+    assert r.accelerators == "MI355X" and r.accelerator_count == 4
+    assert canonical_accelerator("mi355") == "MI355X"
+    from skypilot_amd.exceptions import TaskValidationError
+    with pytest.raises(TaskValidationError):
+        Resources.from_yaml_config({"bogus": 1})
+
+
+def test_optimizer_feasibility(monkeypatch):
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "8")
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    from skypilot_amd.exceptions import ResourcesUnavailableError
+    from skypilot_amd.optimizer import Optimizer
+    from skypilot_amd.task import Task
+    Optimizer.optimize(Task("ok", run="x") .set_resources(
+        __import__("skypilot_amd.resources",
+                   fromlist=["Resources"]).Resources.from_yaml_config(
+            {"accelerators": "MI355X:8"})))
+    bad = Task("bad", run="x")
+    bad.resources = bad.resources.copy(accelerators="H100",
+                                       accelerator_count=8)
+    with pytest.raises(ResourcesUnavailableError):
+        Optimizer.optimize(bad)
+    gpu_topology.detect_gpus.cache_clear()
