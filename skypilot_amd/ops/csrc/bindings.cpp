@@ -38,6 +38,8 @@ void attn_bwd_launch(const void*, const void*, const void*, const void*,
                      int, int, int, int, float, bool, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
 void mfma32_probe_launch(const void*, const void*, float*, hipStream_t);
+void trb16_probe_launch(float*, int, hipStream_t);
+void permlane_probe_launch(float*, hipStream_t);
 void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
                        hipStream_t);
@@ -226,6 +228,18 @@ torch::Tensor mfma32_probe(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+torch::Tensor trb16_probe(int64_t mode, torch::Tensor ref_dev) {
+  auto out = torch::zeros({64, 4}, ref_dev.options().dtype(at::kFloat));
+  trb16_probe_launch(out.data_ptr<float>(), (int)mode, cur_stream());
+  return out;
+}
+
+torch::Tensor permlane_probe(torch::Tensor ref_dev) {
+  auto out = torch::zeros({64, 2}, ref_dev.options().dtype(at::kFloat));
+  permlane_probe_launch(out.data_ptr<float>(), cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
@@ -239,4 +253,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma32_probe", &mfma32_probe);
+  m.def("trb16_probe", &trb16_probe);
+  m.def("permlane_probe", &permlane_probe);
 }

@@ -66,6 +66,32 @@ __device__ __forceinline__ int swzP(int byte_off, int row) {
   return byte_off ^ (((row >> 1) & 7) << 4);
 }
 
+// Paired ds_read_b64_tr_b16: two transpose-reads + one waitcnt.
+// Semantics (HW-verified by trb16 probe, 2026-09-12): per 16-lane
+// group, out[lw][j] = in_lane[(lw>>2)+4j][lw&3] where in_lane[r] is the
+// 4 bf16 at lane r's address.  With per-lane addresses
+//   q(r) = qbase + (r>>2),  d(r) = dbase + (r&3)*4
+// over a row-major [seq][D] tile, the two reads assemble exactly the
+// MFMA B-fragment B[k=q][n=d] — no transposed staging copy needed.
+// The sched_barrier is required (guide rule #18): hipcc hoists
+// register-only MFMA past inline-asm lgkmcnt.
+__device__ __forceinline__ s16x8 ds_tr_b16_pair(unsigned int a0,
+                                                unsigned int a1) {
+  unsigned long long lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(lo), "=v"(hi)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  union { unsigned long long u[2]; s16x8 v; } cvt;
+  cvt.u[0] = lo;
+  cvt.u[1] = hi;
+  return cvt.v;
+}
+
 // bf16 <-> f32 via bit ops (we deliberately avoid __hip_bfloat16 so these
 // headers stay independent of HIP half/bf16 operator macros).
 __device__ __forceinline__ float bf2f(unsigned short h) {

@@ -61,3 +61,59 @@ extern "C" void mfma32_probe_launch(const void* A, const void* B, float* C,
   hipLaunchKernelGGL(mfma32_probe_kernel, dim3(1), dim3(64), 0, stream,
                      (const unsigned short*)A, (const unsigned short*)B, C);
 }
+
+// ds_read_b64_tr_b16 semantics probe (round-2 groundwork, guide T10):
+// LDS is filled with bf16 value == element index; each lane issues one
+// tr-read at addr = base + lane*8 (contiguous per-lane, as a plain b64
+// read would) and we record which 4 elements each lane received.
+extern "C" __global__ void trb16_probe_kernel(float* __restrict__ out,
+                                              int mode) {
+  __shared__ unsigned short buf[1024];
+  const int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x)
+    buf[i] = f2bf((float)(i + 100));  // distinguishable from zero
+  __syncthreads();
+  unsigned int base = (unsigned int)(unsigned long long)&buf[0];
+  unsigned int addr;
+  if (mode == 0)
+    addr = base + lane * 8;            // contiguous 8B per lane
+  else if (mode == 1)
+    addr = base;                       // uniform base
+  else
+    addr = base + ((lane & 15) + (lane >> 4) * 64) * 2;
+  unsigned long long dst;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %1\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(dst)
+      : "v"(addr)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  unsigned short e[4] = {(unsigned short)(dst & 0xffff),
+                         (unsigned short)((dst >> 16) & 0xffff),
+                         (unsigned short)((dst >> 32) & 0xffff),
+                         (unsigned short)((dst >> 48) & 0xffff)};
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = bf2f(e[j]);
+}
+
+extern "C" void trb16_probe_launch(float* out, int mode,
+                                   hipStream_t stream) {
+  hipLaunchKernelGGL(trb16_probe_kernel, dim3(1), dim3(64), 0, stream, out,
+                     mode);
+}
+
+// permlane32_swap semantics probe (round-2 swapped-QK^T groundwork).
+extern "C" __global__ void permlane_probe_kernel(float* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  unsigned int a = 1000 + lane;   // marker values
+  unsigned int b = 2000 + lane;
+  auto pair = __builtin_amdgcn_permlane32_swap(a, b, false, false);
+  out[lane * 2 + 0] = (float)pair[0];
+  out[lane * 2 + 1] = (float)pair[1];
+}
+
+extern "C" void permlane_probe_launch(float* out, hipStream_t stream) {
+  hipLaunchKernelGGL(permlane_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     out);
+}
