@@ -49,6 +49,8 @@ def main():
               f"{toks/dt:8.1f} tok/s decode "
               f"(ttft incl. prefill; prompt {args.prompt_len})",
               flush=True)
+    print(f"graph buckets captured: {eng.stats['graph_buckets']} "
+          f"(use_graphs={eng.use_graphs})")
     eng.stop()
 
 

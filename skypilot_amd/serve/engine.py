@@ -53,7 +53,8 @@ class Request:
 class Engine:
     def __init__(self, model_name: str, device: Optional[str] = None,
                  max_seq: int = 4096, max_batch: Optional[int] = None,
-                 hbm_budget_gb: Optional[float] = None):
+                 hbm_budget_gb: Optional[float] = None,
+                 use_graphs: bool = True):
         self.device = torch.device(device or (
             "cuda" if torch.cuda.is_available() else "cpu"))
         dtype = torch.bfloat16
@@ -73,7 +74,10 @@ class Engine:
                 self.cfg, self.max_seq, int(hbm_budget_gb * 1e9),
                 self.device)
         else:
-            self.cache = KVCache(self.cfg, max_batch, self.max_seq,
+            # +1 slot when graphs are on: the graph-padding scratch slot
+            # must not cost a unit of serving concurrency.
+            extra = 1 if (use_graphs and self.device.type == "cuda") else 0
+            self.cache = KVCache(self.cfg, max_batch + extra, self.max_seq,
                                  self.device)
         self.max_batch = self.cache.max_batch
         self.free_slots = list(range(self.max_batch))
@@ -82,7 +86,20 @@ class Engine:
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self.stats = {"requests": 0, "tokens_generated": 0,
-                      "prefill_tokens": 0}
+                      "prefill_tokens": 0, "graph_buckets": []}
+        # hipGraph decode: the whole decode forward (~7 kernels/layer +
+        # head = ~230 launches for 8B) replays as ONE graph launch per
+        # power-of-two batch bucket.  Low-concurrency decode is
+        # launch-bound, so this is the serving latency lever.  Padded
+        # rows point at a reserved scratch slot so their K/V writes
+        # never touch a live sequence.
+        self.use_graphs = (use_graphs and self.device.type == "cuda"
+                           and self.max_batch >= 2)
+        self._graphs: Dict[int, tuple] = {}
+        self._scratch_slot = -1
+        if self.use_graphs:
+            self._scratch_slot = self.free_slots.pop()
+            self.max_batch -= 1
 
     # ------------------------------------------------------------------
     def submit(self, req: Request) -> Request:
@@ -110,6 +127,9 @@ class Engine:
 
     def stop(self):
         self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=10)
+            self._thread = None
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -135,30 +155,108 @@ class Engine:
         self.stats["prefill_tokens"] += L
         self._maybe_finish(slot, req, next_id)
 
+    # -------------------------- hipGraph decode ------------------------
+    def _bucket(self, n: int) -> int:
+        b = 1
+        while b < n:
+            b <<= 1
+        return b
+
+    def _get_graph(self, b: int):
+        """Lazily capture the decode forward for batch bucket `b`."""
+        if b in self._graphs:
+            return self._graphs[b]
+        dev = self.device
+        st = {
+            "toks": torch.zeros(b, 1, dtype=torch.long, device=dev),
+            "positions": torch.zeros(b, dtype=torch.int32, device=dev),
+            "slots": torch.full((b,), self._scratch_slot,
+                                dtype=torch.long, device=dev),
+            "pos": torch.zeros(b, dtype=torch.long, device=dev),
+            "kv_lens": torch.ones(b, dtype=torch.int32, device=dev),
+            "slot_i32": torch.full((b,), self._scratch_slot,
+                                   dtype=torch.int32, device=dev),
+        }
+        ctx = InferenceContext(
+            cache=self.cache, mode="decode", slots=st["slots"],
+            pos=st["pos"], kv_lens=st["kv_lens"],
+            slot_ids_i32=st["slot_i32"])
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warmup on a side stream (autotune etc.)
+                self.model(st["toks"], st["positions"], ctx)
+        torch.cuda.current_stream().wait_stream(side)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            st["out"] = self.model(st["toks"], st["positions"], ctx)
+        self._graphs[b] = (g, st)
+        self.stats["graph_buckets"] = sorted(self._graphs)
+        return self._graphs[b]
+
     @torch.no_grad()
     def _decode_step(self) -> None:
         slots = sorted(self.active.keys())
         if not slots:
             return
         reqs = [self.active[s] for s in slots]
-        toks = torch.tensor([[r.out_ids[-1]] for r in reqs],
-                            dtype=torch.long, device=self.device)
+        n = len(slots)
         lens = [self.cache.lens[s] for s in slots]
-        slots_t = torch.tensor(slots, dtype=torch.long, device=self.device)
-        pos_t = torch.tensor(lens, dtype=torch.long, device=self.device)
-        positions = torch.tensor(lens, dtype=torch.int32,
+        tok_l = [r.out_ids[-1] for r in reqs]
+        if self.use_graphs:
+            b = self._bucket(n)
+            try:
+                g, st = self._get_graph(b)
+            except Exception as e:  # capture unsupported -> eager forever
+                import sys
+                print(f"[engine] hipGraph capture failed, falling back "
+                      f"to eager decode: {e!r}", file=sys.stderr)
+                self.use_graphs = False
+                return self._decode_step()
+            pad = b - n
+            scr = self._scratch_slot
+            st["toks"].copy_(torch.tensor(
+                tok_l + [0] * pad, dtype=torch.long).view(b, 1))
+            st["positions"].copy_(torch.tensor(
+                lens + [0] * pad, dtype=torch.int32))
+            st["slots"].copy_(torch.tensor(
+                slots + [scr] * pad, dtype=torch.long))
+            st["pos"].copy_(torch.tensor(
+                lens + [0] * pad, dtype=torch.long))
+            st["kv_lens"].copy_(torch.tensor(
+                [l + 1 for l in lens] + [1] * pad, dtype=torch.int32))
+            st["slot_i32"].copy_(torch.tensor(
+                slots + [scr] * pad, dtype=torch.int32))
+            g.replay()
+            logits = st["out"]  # [b, 1, V]
+        else:
+            toks = torch.tensor([[t] for t in tok_l], dtype=torch.long,
+                                device=self.device)
+            slots_t = torch.tensor(slots, dtype=torch.long,
+                                   device=self.device)
+            pos_t = torch.tensor(lens, dtype=torch.long,
                                  device=self.device)
-        kv_lens = torch.tensor([l + 1 for l in lens], dtype=torch.int32,
-                               device=self.device)
-        ctx = InferenceContext(
-            cache=self.cache, mode="decode", slots=slots_t, pos=pos_t,
-            kv_lens=kv_lens, slot_ids_i32=slots_t.int())
-        logits = self.model(toks, positions, ctx)  # [n, 1, V]
+            positions = torch.tensor(lens, dtype=torch.int32,
+                                     device=self.device)
+            kv_lens = torch.tensor([l + 1 for l in lens],
+                                   dtype=torch.int32, device=self.device)
+            ctx = InferenceContext(
+                cache=self.cache, mode="decode", slots=slots_t, pos=pos_t,
+                kv_lens=kv_lens, slot_ids_i32=slots_t.int())
+            logits = self.model(toks, positions, ctx)  # [n, 1, V]
+        # Greedy rows batch into one argmax + one sync; sampled rows
+        # (temperature > 0) go through _sample individually.
+        greedy_ids = None
+        if all(r.temperature == 0 for r in reqs):
+            greedy_ids = logits[:n, 0].argmax(-1).tolist()
         for i, slot in enumerate(slots):
             self.cache.lens[slot] += 1
             req = self.active[slot]
-            next_id = self._sample(logits[i, 0], req.temperature,
-                                   req.top_p)
+            if greedy_ids is not None:
+                next_id = greedy_ids[i]
+            else:
+                next_id = self._sample(logits[i, 0], req.temperature,
+                                       req.top_p)
             req.out_ids.append(next_id)
             if req.stream_queue is not None:
                 req.stream_queue.put(next_id)

@@ -97,3 +97,28 @@ def test_top_p_and_stop_and_stream(engine):
             body = "".join(resp.iter_text())
         assert body.count("data:") == 4  # 3 tokens + [DONE]
         assert "[DONE]" in body
+
+
+@pytest.mark.gpu
+def test_graph_decode_matches_eager():
+    """hipGraph decode replay must produce the same greedy tokens as the
+    eager decode path (same model seed, same prompts)."""
+    import torch
+    assert torch.cuda.is_available()
+    from skypilot_amd.serve.engine import Engine
+    prompts = [[1, 5, 9, 13, 2], [7, 7, 3], [2, 4, 6, 8, 10, 12, 1]]
+    outs = {}
+    for graphs in (False, True):
+        eng = Engine("llama-debug", device="cuda", max_seq=256,
+                     max_batch=8, use_graphs=graphs)
+        assert eng.use_graphs == graphs
+        eng.start()
+        try:
+            reqs = [eng.submit(Request(prompt_ids=list(p), max_tokens=12))
+                    for p in prompts]
+            for r in reqs:
+                assert r.done.wait(60)
+            outs[graphs] = [r.out_ids for r in reqs]
+        finally:
+            eng.stop()
+    assert outs[True] == outs[False]
