@@ -82,9 +82,13 @@ class Attention(nn.Module):
     def forward(self, x, cos, sin, positions, infer_ctx=None):
         B, S, _ = x.shape
         d = self.cfg.head_dim
-        q = self.wq(x).view(B, S, self.n_q, d)
-        k = self.wk(x).view(B, S, self.n_kv, d)
-        v = self.wv(x).view(B, S, self.n_kv, d)
+        decode = infer_ctx is not None and infer_ctx.mode == "decode"
+        # Decode-step GEMMs are [n<=8, h] GEMVs: route to the skinny
+        # kernel (weights-BW-bound) instead of hipBLASLt.
+        lin = ops.decode_linear if decode else torch.nn.functional.linear
+        q = lin(x, self.wq.weight).view(B, S, self.n_q, d)
+        k = lin(x, self.wk.weight).view(B, S, self.n_kv, d)
+        v = lin(x, self.wv.weight).view(B, S, self.n_kv, d)
         q = ops.rope(q.reshape(B * S, self.n_q, d), cos, sin,
                      positions).view(B, S, self.n_q, d)
         k = ops.rope(k.reshape(B * S, self.n_kv, d), cos, sin,
@@ -105,7 +109,7 @@ class Attention(nn.Module):
                 q.view(B, self.n_q, d), infer_ctx.cache.k[self.layer_idx],
                 infer_ctx.cache.v[self.layer_idx], infer_ctx.kv_lens,
                 infer_ctx.slot_ids_i32, self.scale).view(B, 1, self.n_q, d)
-        return self.wo(o.reshape(B, S, self.n_q * d))
+        return lin(o.reshape(B, S, self.n_q * d), self.wo.weight)
 
 
 class MLP(nn.Module):
@@ -116,8 +120,11 @@ class MLP(nn.Module):
         self.w_down = nn.Linear(m, h, bias=False)
         self.m = m
 
-    def forward(self, x):
-        return self.w_down(ops.swiglu(self.w_gate_up(x)))
+    def forward(self, x, infer_ctx=None):
+        decode = infer_ctx is not None and infer_ctx.mode == "decode"
+        lin = ops.decode_linear if decode else torch.nn.functional.linear
+        return lin(ops.swiglu(lin(x, self.w_gate_up.weight)),
+                   self.w_down.weight)
 
 
 class Block(nn.Module):
@@ -133,7 +140,8 @@ class Block(nn.Module):
     def forward(self, x, cos, sin, positions, infer_ctx=None):
         x = x + self.attn(ops.rmsnorm(x, self.attn_norm, self.eps), cos, sin,
                           positions, infer_ctx)
-        x = x + self.mlp(ops.rmsnorm(x, self.mlp_norm, self.eps))
+        x = x + self.mlp(ops.rmsnorm(x, self.mlp_norm, self.eps),
+                         infer_ctx)
         return x
 
 
@@ -167,6 +175,8 @@ class Llama(nn.Module):
         for blk in self.blocks:
             x = blk(x, cos, sin, positions, infer_ctx)
         x = ops.rmsnorm(x, self.final_norm, self.cfg.norm_eps)
+        if infer_ctx is not None and infer_ctx.mode == "decode":
+            return ops.decode_linear(x, self.lm_head.weight)
         return self.lm_head(x)
 
     def loss(self, tokens: torch.Tensor, targets: torch.Tensor):

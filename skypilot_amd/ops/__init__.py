@@ -323,9 +323,35 @@ def attn_decode(q, kc, vc, kv_lens, slot_ids, scale):
     return attn_decode_ref(q, kc, vc, kv_lens, slot_ids, scale)
 
 
+def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """F.linear for the decode step: y = x @ W^T with x [.., n, in],
+    n <= 8.  On GPU this runs the weight-bandwidth-bound skinny GEMV
+    (skinny_gemm.hip, ~3x hipBLASLt on these shapes); otherwise (CPU,
+    n > 8, odd inner dim) it falls back to F.linear.  No autograd —
+    inference only."""
+    lead = x.shape[:-1]
+    n = 1
+    for d in lead:
+        n *= d
+    i, o = x.shape[-1], weight.shape[0]
+    wbytes = 2 * i * o
+    # Measured routing (bench_gemv.py on MI355X, docs/BENCHMARKS.md):
+    # n=1 the GEMV streams W at 6.5-7.3 TB/s and always wins (hipBLASLt
+    # floor is ~19 us/GEMM); at larger n it goes VALU-bound on big
+    # shapes and only wins where hipBLASLt is launch-bound.
+    use_native = (n == 1 or (n <= 4 and wbytes <= 34_000_000)
+                  or wbytes <= 10_000_000)
+    if (use_native and x.is_cuda and x.dtype == torch.bfloat16
+            and 1 <= n <= 8 and i % 512 == 0):
+        C = _require_native("skinny_gemm")
+        y = C.skinny_gemm(x.reshape(n, i).contiguous(), weight)
+        return y.view(*lead, o)
+    return torch.nn.functional.linear(x, weight)
+
+
 __all__ = [
     "rmsnorm", "rope", "attention", "fused_cross_entropy", "attn_decode",
-    "swiglu",
+    "swiglu", "decode_linear",
     "native", "native_available", "rmsnorm_ref", "rope_ref",
     "attention_ref", "cross_entropy_ref", "attn_decode_ref",
 ]

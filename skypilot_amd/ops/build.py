@@ -27,6 +27,7 @@ HIP_SOURCES = [
     "attention_bwd.hip",
     "attention_decode.hip",
     "swiglu.hip",
+    "skinny_gemm.hip",
     "mfma_probe.hip",
 ]
 CPP_SOURCES = ["bindings.cpp"]

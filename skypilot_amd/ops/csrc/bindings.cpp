@@ -40,6 +40,8 @@ void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
 void mfma32_probe_launch(const void*, const void*, float*, hipStream_t);
 void trb16_probe_launch(float*, int, hipStream_t);
 void permlane_probe_launch(float*, hipStream_t);
+void skinny_gemm_launch(const void*, const void*, void*, int, int, int,
+                        hipStream_t);
 void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
                        hipStream_t);
@@ -185,6 +187,20 @@ torch::Tensor attn_decode(torch::Tensor Q, torch::Tensor Kc,
   return O;
 }
 
+torch::Tensor skinny_gemm(torch::Tensor X, torch::Tensor W) {
+  // Y[N,O] = X[N,I] @ W[O,I]^T (decode GEMV; see skinny_gemm.hip)
+  CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
+  CHECK_GPU(W); CHECK_CONTIG(W); CHECK_BF16(W);
+  const int N = (int)X.size(0), I = (int)X.size(1), O = (int)W.size(0);
+  TORCH_CHECK(W.size(1) == I, "skinny_gemm: inner dims mismatch");
+  TORCH_CHECK(N >= 1 && N <= 8, "skinny_gemm: N must be 1..8");
+  TORCH_CHECK(I % 512 == 0, "skinny_gemm: I must be a multiple of 512");
+  auto y = torch::empty({N, O}, X.options());
+  skinny_gemm_launch(W.data_ptr(), X.data_ptr(), y.data_ptr(), N, I, O,
+                     cur_stream());
+  return y;
+}
+
 torch::Tensor swiglu_fwd(torch::Tensor gu) {
   CHECK_GPU(gu); CHECK_CONTIG(gu); CHECK_BF16(gu);
   const int M2 = (int)gu.size(-1);
@@ -249,6 +265,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_decode", &attn_decode);
+  m.def("skinny_gemm", &skinny_gemm);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);

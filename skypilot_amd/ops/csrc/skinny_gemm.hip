@@ -1,0 +1,90 @@
+// Decode GEMV / skinny GEMM: Y[N,O] = X[N,I] @ W[O,I]^T, bf16 in/out,
+// fp32 accumulate.  N = decode batch (1..8 after hipGraph bucketing).
+//
+// Rationale (measured, docs/BENCHMARKS.md): hipBLASLt moves ~2.1 TB/s
+// effective on the decode-step [n,h]x[h,m] shapes, leaving ~3x of the
+// 6.3 TB/s HBM roofline on the table.  The op is purely weight-
+// bandwidth-bound: X is at most 8 x 14336 bf16 = 229 KB and stays L2
+// resident while W (up to 1 GB for lm_head) streams through once.
+// One wave owns one output row (a contiguous 2I-byte run of W), four
+// rows per 256-thread block; each lane strides b128 loads across the
+// row and keeps N fp32 partial sums; a wave-wide shuffle reduction
+// finishes each dot product.  No LDS, no barriers.
+//
+// Reference parity note: the reference framework has no serving GEMV
+// (it delegates serving to external images);
+// this backs skypilot_amd.serve's single-stream latency path.
+#include "common.h"
+
+template <int N>
+__global__ __launch_bounds__(256) void skinny_gemm_kernel(
+    const unsigned short* __restrict__ W,
+    const unsigned short* __restrict__ X,
+    unsigned short* __restrict__ Y, int I, int O) {
+  int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  int o = blockIdx.x * 4 + wave;
+  if (o >= O) return;
+  const unsigned short* wrow = W + (long long)o * I;
+  float acc[N];
+#pragma unroll
+  for (int b = 0; b < N; ++b) acc[b] = 0.f;
+  // 4 independent b128 W loads in flight per lane (the loop-carried
+  // FMA chain alone leaves one load outstanding and stalls on vmcnt);
+  // W streams through once -> nontemporal, so X stays L2-resident.
+  int i = lane * 8;
+  for (; i + 3 * 512 + 8 <= I; i += 4 * 512) {
+    s16x8 w0 = __builtin_nontemporal_load((const s16x8*)(wrow + i));
+    s16x8 w1 = __builtin_nontemporal_load((const s16x8*)(wrow + i + 512));
+    s16x8 w2 = __builtin_nontemporal_load((const s16x8*)(wrow + i + 1024));
+    s16x8 w3 = __builtin_nontemporal_load((const s16x8*)(wrow + i + 1536));
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      const unsigned short* xb = X + (long long)b * I + i;
+      s16x8 x0 = *(const s16x8*)(xb);
+      s16x8 x1 = *(const s16x8*)(xb + 512);
+      s16x8 x2 = *(const s16x8*)(xb + 1024);
+      s16x8 x3 = *(const s16x8*)(xb + 1536);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        acc[b] += bf2f(w0[j]) * bf2f(x0[j]);
+        acc[b] += bf2f(w1[j]) * bf2f(x1[j]);
+        acc[b] += bf2f(w2[j]) * bf2f(x2[j]);
+        acc[b] += bf2f(w3[j]) * bf2f(x3[j]);
+      }
+    }
+  }
+  for (; i < I; i += 512) {
+    s16x8 wv = __builtin_nontemporal_load((const s16x8*)(wrow + i));
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      s16x8 xv = *(const s16x8*)(X + (long long)b * I + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[b] += bf2f(wv[j]) * bf2f(xv[j]);
+    }
+  }
+#pragma unroll
+  for (int b = 0; b < N; ++b) {
+    float r = wave_reduce_sum(acc[b]);
+    if (lane == 0) Y[(long long)b * O + o] = f2bf(r);
+  }
+}
+
+extern "C" void skinny_gemm_launch(const void* W, const void* X, void* Y,
+                                   int N, int I, int O,
+                                   hipStream_t stream) {
+  dim3 grid((O + 3) / 4), block(256);
+  const unsigned short* w = (const unsigned short*)W;
+  const unsigned short* x = (const unsigned short*)X;
+  unsigned short* y = (unsigned short*)Y;
+#define CASE(n)                                                        \
+  case n:                                                              \
+    hipLaunchKernelGGL(skinny_gemm_kernel<n>, grid, block, 0, stream,  \
+                       w, x, y, I, O);                                 \
+    break;
+  switch (N) {
+    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+    default:
+      break;  // host wrapper guarantees 1 <= N <= 8
+  }
+#undef CASE
+}

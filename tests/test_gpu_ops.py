@@ -222,3 +222,26 @@ def test_mfma32_layout_probe():
     C = ops.native().mfma32_probe(A, B)
     expect = A.float() @ B.float()
     assert rel_err(C, expect) < 2e-2, (C[:3, :3], expect[:3, :3])
+
+
+@pytest.mark.gpu
+def test_skinny_gemm_matches_fp32_linear():
+    """Decode GEMV (skinny_gemm.hip) vs plain fp32 F.linear."""
+    C = ops.native()
+    torch.manual_seed(3)
+    for n, i, o in [(1, 4096, 4096), (3, 4096, 1024), (8, 14336, 4096),
+                    (2, 4096, 128256), (5, 512, 512)]:
+        x = (torch.randn(n, i, device="cuda") * 0.5).bfloat16()
+        w = (torch.randn(o, i, device="cuda") * 0.02).bfloat16()
+        y = C.skinny_gemm(x, w)
+        ref = torch.nn.functional.linear(x.float(), w.float())
+        assert y.shape == (n, o)
+        torch.testing.assert_close(y.float(), ref, atol=0.02, rtol=0.02)
+    # dispatcher: lead-dim flattening + CPU fallback parity
+    x3 = (torch.randn(2, 1, 4096, device="cuda") * 0.5).bfloat16()
+    w = (torch.randn(1024, 4096, device="cuda") * 0.02).bfloat16()
+    out = ops.decode_linear(x3, w)
+    assert out.shape == (2, 1, 1024)
+    torch.testing.assert_close(
+        out.float(), torch.nn.functional.linear(x3.float(), w.float()),
+        atol=0.02, rtol=0.02)
