@@ -78,14 +78,33 @@ class Attention(nn.Module):
         self.wv = nn.Linear(h, self.n_kv * d, bias=False)
         self.wo = nn.Linear(self.n_q * d, h, bias=False)
         self.scale = 1.0 / math.sqrt(d)
+        self._wqkv = None  # lazy fused [q|k|v] weight for the decode GEMV
 
     def forward(self, x, cos, sin, positions, infer_ctx=None):
         B, S, _ = x.shape
         d = self.cfg.head_dim
         decode = infer_ctx is not None and infer_ctx.mode == "decode"
-        # Decode-step GEMMs are [n<=8, h] GEMVs: route to the skinny
-        # kernel (weights-BW-bound) instead of hipBLASLt.
-        lin = ops.decode_linear if decode else torch.nn.functional.linear
+        if decode:
+            # One packed qkv GEMV, then one fused rope+cache-write
+            # kernel (decode_fused.hip) — replaces 3 GEMVs + rope x2 +
+            # scatter x2.  Decode GEMMs route via ops.decode_linear
+            # (skinny GEMV vs hipBLASLt per measured thresholds).
+            if self._wqkv is None:
+                self._wqkv = torch.cat(
+                    [self.wq.weight, self.wk.weight, self.wv.weight],
+                    0).contiguous()
+            cache = infer_ctx.cache
+            qkv = ops.decode_linear(x.reshape(B, -1), self._wqkv)
+            q = ops.rope_kvwrite(
+                qkv, cache.k[self.layer_idx], cache.v[self.layer_idx],
+                cos, sin, positions, infer_ctx.slot_ids_i32, self.n_q,
+                self.n_kv)
+            o = ops.attn_decode(
+                q, cache.k[self.layer_idx], cache.v[self.layer_idx],
+                infer_ctx.kv_lens, infer_ctx.slot_ids_i32, self.scale)
+            return ops.decode_linear(o.reshape(B, self.n_q * d),
+                                     self.wo.weight).view(B, S, -1)
+        lin = torch.nn.functional.linear
         q = lin(x, self.wq.weight).view(B, S, self.n_q, d)
         k = lin(x, self.wk.weight).view(B, S, self.n_kv, d)
         v = lin(x, self.wv.weight).view(B, S, self.n_kv, d)
@@ -95,20 +114,12 @@ class Attention(nn.Module):
                      positions).view(B, S, self.n_kv, d)
         if infer_ctx is None:
             o = ops.attention(q, k, v, self.scale, causal=True)
-        elif infer_ctx.mode == "prefill":
-            # New sequence: plain causal attention over the prompt; the
+        else:  # prefill: plain causal attention over the prompt; the
             # (unpadded) K/V rows land in the cache for decode.
             infer_ctx.cache.write_prefill(self.layer_idx,
                                           infer_ctx.prefill_slot, k, v,
                                           infer_ctx.prefill_len)
             o = ops.attention(q, k, v, self.scale, causal=True)
-        else:  # decode: one new token per active sequence
-            infer_ctx.cache.write_decode(self.layer_idx, infer_ctx.slots,
-                                         infer_ctx.pos, k, v)
-            o = ops.attn_decode(
-                q.view(B, self.n_q, d), infer_ctx.cache.k[self.layer_idx],
-                infer_ctx.cache.v[self.layer_idx], infer_ctx.kv_lens,
-                infer_ctx.slot_ids_i32, self.scale).view(B, 1, self.n_q, d)
         return lin(o.reshape(B, S, self.n_q * d), self.wo.weight)
 
 
@@ -125,6 +136,8 @@ class MLP(nn.Module):
         lin = ops.decode_linear if decode else torch.nn.functional.linear
         return lin(ops.swiglu(lin(x, self.w_gate_up.weight)),
                    self.w_down.weight)
+
+    # (decode path is driven from Llama.forward's fused-residual loop)
 
 
 class Block(nn.Module):
@@ -172,11 +185,25 @@ class Llama(nn.Module):
                                      device=tokens.device)
             positions = positions.unsqueeze(0).expand(B, S).reshape(-1)
         x = self.embed(tokens)
+        if infer_ctx is not None and infer_ctx.mode == "decode":
+            # Fused-residual decode loop: every residual add rides in
+            # the next rmsnorm_res kernel (one launch instead of two),
+            # including the final-norm + lm-head hand-off.
+            res = None
+            for blk in self.blocks:
+                if res is None:
+                    h = ops.rmsnorm(x, blk.attn_norm, blk.eps)
+                else:
+                    x, h = ops.rmsnorm_res(x, res, blk.attn_norm, blk.eps)
+                a = blk.attn(h, cos, sin, positions, infer_ctx)
+                x, h2 = ops.rmsnorm_res(x, a, blk.mlp_norm, blk.eps)
+                res = blk.mlp(h2, infer_ctx)
+            _, hf = ops.rmsnorm_res(x, res, self.final_norm,
+                                    self.cfg.norm_eps)
+            return ops.decode_linear(hf, self.lm_head.weight)
         for blk in self.blocks:
             x = blk(x, cos, sin, positions, infer_ctx)
         x = ops.rmsnorm(x, self.final_norm, self.cfg.norm_eps)
-        if infer_ctx is not None and infer_ctx.mode == "decode":
-            return ops.decode_linear(x, self.lm_head.weight)
         return self.lm_head(x)
 
     def loss(self, tokens: torch.Tensor, targets: torch.Tensor):

@@ -349,9 +349,41 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, weight)
 
 
+def rmsnorm_res(x, res, weight, eps: float = 1e-5):
+    """Fused (x + res, rmsnorm(x + res) * w) for the decode step (one
+    kernel instead of add + norm; decode_fused.hip).  No autograd."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 256 == 0:
+        C = _require_native("rmsnorm_res")
+        x2, h = C.rmsnorm_res(x.contiguous(), res.contiguous(), weight,
+                              eps)
+        return x2, h
+    x2 = (x + res)
+    return x2, rmsnorm(x2, weight, eps)
+
+
+def rope_kvwrite(qkv, kc, vc, cos, sin, positions, slot_ids, Hq, Hkv):
+    """Packed-qkv rope + KV-cache scatter (decode_fused.hip): applies
+    rope to the q and k segments of qkv [n, (Hq+2*Hkv)*D], writes the
+    roped k and raw v rows into the caches at (slot_ids, positions),
+    and returns contiguous roped q [n, Hq, D].  No autograd."""
+    D = kc.shape[3]
+    n = qkv.shape[0]
+    if qkv.is_cuda and qkv.dtype == torch.bfloat16 and D == 128:
+        C = _require_native("rope_kvwrite")
+        return C.rope_kvwrite(qkv.contiguous(), kc, vc, cos, sin,
+                              positions.int(), slot_ids.int(), Hq, Hkv)
+    q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    qr = rope(q.reshape(n, Hq, D), cos, sin, positions)
+    kr = rope(k.reshape(n, Hkv, D), cos, sin, positions)
+    pos = positions.long()
+    kc[slot_ids.long(), pos] = kr
+    vc[slot_ids.long(), pos] = v.reshape(n, Hkv, D)
+    return qr
+
+
 __all__ = [
     "rmsnorm", "rope", "attention", "fused_cross_entropy", "attn_decode",
-    "swiglu", "decode_linear",
+    "swiglu", "decode_linear", "rmsnorm_res", "rope_kvwrite",
     "native", "native_available", "rmsnorm_ref", "rope_ref",
     "attention_ref", "cross_entropy_ref", "attn_decode_ref",
 ]

@@ -28,6 +28,7 @@ HIP_SOURCES = [
     "attention_decode.hip",
     "swiglu.hip",
     "skinny_gemm.hip",
+    "decode_fused.hip",
     "mfma_probe.hip",
 ]
 CPP_SOURCES = ["bindings.cpp"]

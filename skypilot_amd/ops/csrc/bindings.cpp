@@ -42,6 +42,11 @@ void trb16_probe_launch(float*, int, hipStream_t);
 void permlane_probe_launch(float*, hipStream_t);
 void skinny_gemm_launch(const void*, const void*, void*, int, int, int,
                         hipStream_t);
+void rmsnorm_res_launch(const void*, const void*, const void*, void*, void*,
+                        int, int, float, hipStream_t);
+void rope_kvwrite_launch(const void*, void*, void*, void*, const void*,
+                         const void*, const void*, const void*, int, int,
+                         int, int, int, hipStream_t);
 void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
                        hipStream_t);
@@ -201,6 +206,45 @@ torch::Tensor skinny_gemm(torch::Tensor X, torch::Tensor W) {
   return y;
 }
 
+std::vector<torch::Tensor> rmsnorm_res(torch::Tensor x, torch::Tensor res,
+                                       torch::Tensor w, double eps) {
+  // (x + res, rmsnorm(x + res) * w) in one kernel (decode_fused.hip)
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  CHECK_GPU(res); CHECK_CONTIG(res); CHECK_BF16(res);
+  const int H = (int)x.size(-1);
+  long long rows = x.numel() / H;
+  TORCH_CHECK(res.sizes() == x.sizes(), "rmsnorm_res: shape mismatch");
+  TORCH_CHECK(H % 256 == 0 && H <= 8192, "rmsnorm_res: H must be a "
+              "multiple of 256 and <= 8192");
+  auto x_out = torch::empty_like(x);
+  auto h_out = torch::empty_like(x);
+  rmsnorm_res_launch(x.data_ptr(), res.data_ptr(), w.data_ptr(),
+                     x_out.data_ptr(), h_out.data_ptr(), (int)rows, H,
+                     (float)eps, cur_stream());
+  return {x_out, h_out};
+}
+
+torch::Tensor rope_kvwrite(torch::Tensor qkv, torch::Tensor kc,
+                           torch::Tensor vc, torch::Tensor cos_tab,
+                           torch::Tensor sin_tab, torch::Tensor positions,
+                           torch::Tensor slot_ids, int64_t Hq,
+                           int64_t Hkv) {
+  // packed-qkv rope + KV-cache scatter (decode_fused.hip)
+  CHECK_GPU(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
+  const int D = (int)kc.size(3), S_max = (int)kc.size(1);
+  const int n = (int)qkv.size(0);
+  TORCH_CHECK(D == 128, "rope_kvwrite: head_dim must be 128");
+  TORCH_CHECK(qkv.size(1) == (Hq + 2 * Hkv) * D,
+              "rope_kvwrite: packed width mismatch");
+  auto q = torch::empty({n, (long)Hq, (long)D}, qkv.options());
+  rope_kvwrite_launch(qkv.data_ptr(), q.data_ptr(), kc.data_ptr(),
+                      vc.data_ptr(), cos_tab.data_ptr(),
+                      sin_tab.data_ptr(), positions.data_ptr(),
+                      slot_ids.data_ptr(), n, (int)Hq, (int)Hkv, D, S_max,
+                      cur_stream());
+  return q;
+}
+
 torch::Tensor swiglu_fwd(torch::Tensor gu) {
   CHECK_GPU(gu); CHECK_CONTIG(gu); CHECK_BF16(gu);
   const int M2 = (int)gu.size(-1);
@@ -266,6 +310,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_decode", &attn_decode);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("rmsnorm_res", &rmsnorm_res);
+  m.def("rope_kvwrite", &rope_kvwrite);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);

@@ -16,6 +16,24 @@
 // this backs skypilot_amd.serve's single-stream latency path.
 #include "common.h"
 
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+union b128 {
+  s16x8 v;
+  bf16x2 h[4];
+};
+// 2-way bf16 dot with f32 accumulate: one v_dot2c_f32_bf16 per 2 MACs
+// (4 VALU ops per b128 chunk instead of 16 cvt+fma) — keeps n<=8
+// weight-bandwidth-bound instead of VALU-bound.
+__device__ __forceinline__ float dot2(s16x8 a, s16x8 b, float acc) {
+  union b128 ua, ub;
+  ua.v = a;
+  ub.v = b;
+#pragma unroll
+  for (int j = 0; j < 4; ++j)
+    acc = __builtin_amdgcn_fdot2_f32_bf16(ua.h[j], ub.h[j], acc, false);
+  return acc;
+}
+
 template <int N>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     const unsigned short* __restrict__ W,
@@ -25,12 +43,13 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   int o = blockIdx.x * 4 + wave;
   if (o >= O) return;
   const unsigned short* wrow = W + (long long)o * I;
-  float acc[N];
+  // 4 independent b128 W loads in flight per lane (a single
+  // loop-carried chain stalls on vmcnt), W nontemporal so X stays
+  // L2-resident, and 4 accumulators per batch row so the dot2c
+  // dependency chains interleave.
+  float a0[N], a1[N], a2[N], a3[N];
 #pragma unroll
-  for (int b = 0; b < N; ++b) acc[b] = 0.f;
-  // 4 independent b128 W loads in flight per lane (the loop-carried
-  // FMA chain alone leaves one load outstanding and stalls on vmcnt);
-  // W streams through once -> nontemporal, so X stays L2-resident.
+  for (int b = 0; b < N; ++b) a0[b] = a1[b] = a2[b] = a3[b] = 0.f;
   int i = lane * 8;
   for (; i + 3 * 512 + 8 <= I; i += 4 * 512) {
     s16x8 w0 = __builtin_nontemporal_load((const s16x8*)(wrow + i));
@@ -40,31 +59,21 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 #pragma unroll
     for (int b = 0; b < N; ++b) {
       const unsigned short* xb = X + (long long)b * I + i;
-      s16x8 x0 = *(const s16x8*)(xb);
-      s16x8 x1 = *(const s16x8*)(xb + 512);
-      s16x8 x2 = *(const s16x8*)(xb + 1024);
-      s16x8 x3 = *(const s16x8*)(xb + 1536);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        acc[b] += bf2f(w0[j]) * bf2f(x0[j]);
-        acc[b] += bf2f(w1[j]) * bf2f(x1[j]);
-        acc[b] += bf2f(w2[j]) * bf2f(x2[j]);
-        acc[b] += bf2f(w3[j]) * bf2f(x3[j]);
-      }
+      a0[b] = dot2(w0, *(const s16x8*)(xb), a0[b]);
+      a1[b] = dot2(w1, *(const s16x8*)(xb + 512), a1[b]);
+      a2[b] = dot2(w2, *(const s16x8*)(xb + 1024), a2[b]);
+      a3[b] = dot2(w3, *(const s16x8*)(xb + 1536), a3[b]);
     }
   }
   for (; i < I; i += 512) {
     s16x8 wv = __builtin_nontemporal_load((const s16x8*)(wrow + i));
 #pragma unroll
-    for (int b = 0; b < N; ++b) {
-      s16x8 xv = *(const s16x8*)(X + (long long)b * I + i);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[b] += bf2f(wv[j]) * bf2f(xv[j]);
-    }
+    for (int b = 0; b < N; ++b)
+      a0[b] = dot2(wv, *(const s16x8*)(X + (long long)b * I + i), a0[b]);
   }
 #pragma unroll
   for (int b = 0; b < N; ++b) {
-    float r = wave_reduce_sum(acc[b]);
+    float r = wave_reduce_sum(a0[b] + a1[b] + a2[b] + a3[b]);
     if (lane == 0) Y[(long long)b * O + o] = f2bf(r);
   }
 }

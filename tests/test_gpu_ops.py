@@ -245,3 +245,64 @@ def test_skinny_gemm_matches_fp32_linear():
     torch.testing.assert_close(
         out.float(), torch.nn.functional.linear(x3.float(), w.float()),
         atol=0.02, rtol=0.02)
+
+
+@pytest.mark.gpu
+def test_rmsnorm_res_matches_ref():
+    """Fused residual-add + rmsnorm (decode_fused.hip) vs fp32 ref."""
+    C = ops.native()
+    torch.manual_seed(4)
+    for n, h in [(1, 4096), (8, 4096), (3, 256), (2, 8192)]:
+        x = (torch.randn(n, h, device="cuda")).bfloat16()
+        r = (torch.randn(n, h, device="cuda")).bfloat16()
+        w = (torch.randn(h, device="cuda") * 0.1 + 1).bfloat16()
+        x2, out = C.rmsnorm_res(x, r, w, 1e-5)
+        ref_x2 = (x + r)
+        torch.testing.assert_close(x2, ref_x2)
+        rf = ref_x2.float()
+        ref = rf * torch.rsqrt(rf.pow(2).mean(-1, keepdim=True) + 1e-5) \
+            * w.float()
+        torch.testing.assert_close(out.float(), ref, atol=0.05, rtol=0.05)
+
+
+@pytest.mark.gpu
+def test_rope_kvwrite_matches_ref():
+    """Packed rope + cache scatter (decode_fused.hip) vs composed ref."""
+    C = ops.native()
+    torch.manual_seed(5)
+    n, Hq, Hkv, D, slots, S = 4, 8, 2, 128, 6, 64
+    qkv = (torch.randn(n, (Hq + 2 * Hkv) * D, device="cuda")).bfloat16()
+    kc = torch.zeros(slots, S, Hkv, D, device="cuda",
+                     dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    kc_ref, vc_ref = kc.clone(), vc.clone()
+    half = D // 2
+    t = torch.arange(S, device="cuda", dtype=torch.float32)
+    inv = 1.0 / (10000 ** (torch.arange(half, device="cuda").float()
+                           / half))
+    freqs = torch.outer(t, inv)
+    cos, sin = freqs.cos().contiguous(), freqs.sin().contiguous()
+    positions = torch.tensor([0, 5, 63, 17], device="cuda",
+                             dtype=torch.int32)
+    slot_ids = torch.tensor([5, 0, 3, 2], device="cuda",
+                            dtype=torch.int32)
+    q = C.rope_kvwrite(qkv, kc, vc, cos, sin, positions, slot_ids,
+                       Hq, Hkv)
+    # composed reference (same ops the CPU fallback uses)
+    qs, ks, vs = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+
+    def rope_ref(xx, H):
+        xx = xx.reshape(n, H, D).float()
+        x1, x2 = xx[..., :half], xx[..., half:]
+        c = cos[positions.long()].unsqueeze(1)
+        s = sin[positions.long()].unsqueeze(1)
+        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1)
+
+    torch.testing.assert_close(q.float(), rope_ref(qs, Hq), atol=0.02,
+                               rtol=0.02)
+    kr = rope_ref(ks, Hkv).bfloat16()
+    kc_ref[slot_ids.long(), positions.long()] = kr
+    vc_ref[slot_ids.long(), positions.long()] = vs.reshape(n, Hkv, D)
+    torch.testing.assert_close(kc.float(), kc_ref.float(), atol=0.02,
+                               rtol=0.02)
+    torch.testing.assert_close(vc, vc_ref)

@@ -114,8 +114,10 @@ def test_graph_decode_matches_eager():
         assert eng.use_graphs == graphs
         eng.start()
         try:
-            reqs = [eng.submit(Request(prompt_ids=list(p), max_tokens=12))
-                    for p in prompts]
+            # Differing max_tokens exercise per-request finish inside a
+            # chained-replay chunk (engine.CHUNK = 8).
+            reqs = [eng.submit(Request(prompt_ids=list(p), max_tokens=mt))
+                    for p, mt in zip(prompts, (5, 12, 9))]
             for r in reqs:
                 assert r.done.wait(60)
             outs[graphs] = [r.out_ids for r in reqs]
