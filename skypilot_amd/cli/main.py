@@ -387,6 +387,55 @@ def volumes_delete(name):
     click.echo(f"Volume {name} deleted.")
 
 
+@cli.group()
+def users():
+    """User, role and service-account management (admin)."""
+
+
+@users.command("list")
+def users_list():
+    with sdk._client() as c:
+        _print_result(c.get("/api/users").json())
+
+
+@users.command("set-role")
+@click.argument("name")
+@click.argument("role", type=click.Choice(["admin", "user", "viewer"]))
+def users_set_role(name, role):
+    with sdk._client() as c:
+        r = c.post("/api/users/role", json={"name": name, "role": role})
+        if r.status_code != 200:
+            raise click.ClickException(r.text)
+    click.echo(f"{name} -> {role}")
+
+
+@users.command("token")
+@click.argument("name")
+@click.option("--role", default="user",
+              type=click.Choice(["admin", "user", "viewer"]))
+def users_token(name, role):
+    """Mint a service-account token (shown once)."""
+    with sdk._client() as c:
+        r = c.post("/api/users/token", json={"name": name, "role": role})
+        if r.status_code != 200:
+            raise click.ClickException(r.text)
+        click.echo(r.json()["token"])
+
+
+@users.command("tokens")
+def users_tokens():
+    with sdk._client() as c:
+        _print_result(c.get("/api/users/tokens").json())
+
+
+@users.command("revoke")
+@click.argument("name")
+def users_revoke(name):
+    with sdk._client() as c:
+        r = c.post("/api/users/token/revoke", json={"name": name})
+        _print_result(r.json())
+
+
 @cli.command("recipes")
 def recipes_cmd():
     """List bundled task recipes (examples/*.yaml)."""

@@ -31,12 +31,24 @@ def use_test_client(client) -> None:
 from contextlib import contextmanager
 
 
+def _auth_headers() -> dict:
+    """Identity for RBAC: a service-account bearer token
+    (SKY_AMD_API_TOKEN) wins; otherwise the local username travels in
+    X-Skypilot-User (reference: sky client auth headers)."""
+    tok = os.environ.get("SKY_AMD_API_TOKEN")
+    if tok:
+        return {"Authorization": f"Bearer {tok}"}
+    user = os.environ.get("SKY_AMD_USER") or os.environ.get("USER")
+    return {"X-Skypilot-User": user} if user else {}
+
+
 @contextmanager
 def _client():
     if _TEST_CLIENT is not None:
         yield _TEST_CLIENT  # never closed here; tests own its lifecycle
     else:
-        c = httpx.Client(base_url=server_url(), timeout=30.0)
+        c = httpx.Client(base_url=server_url(), timeout=30.0,
+                         headers=_auth_headers())
         try:
             yield c
         finally:

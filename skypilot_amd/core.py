@@ -8,7 +8,7 @@ import time
 from pathlib import Path
 from typing import Any, Dict, List, Optional
 
-from skypilot_amd import global_state, provision
+from skypilot_amd import global_state, users, provision
 from skypilot_amd.backends.pool_backend import PoolBackend
 from skypilot_amd.exceptions import ClusterDoesNotExist, ClusterNotUpError
 from skypilot_amd.utils.gpu_topology import detect_gpus, xgmi_topology
@@ -49,6 +49,7 @@ def status(cluster_names: Optional[List[str]] = None,
 
 
 def start(cluster_name: str) -> Dict[str, Any]:
+    users.check_cluster_owner(cluster_name)
     record = _get_record(cluster_name)
     handle = record["handle"]
     flag = Path(handle["cluster_dir"]) / "autostop_triggered"
@@ -64,17 +65,20 @@ def start(cluster_name: str) -> Dict[str, Any]:
 
 
 def stop(cluster_name: str) -> None:
+    users.check_cluster_owner(cluster_name)
     record = _get_record(cluster_name)
     PoolBackend().teardown(record["handle"], terminate=False)
 
 
 def down(cluster_name: str) -> None:
+    users.check_cluster_owner(cluster_name)
     record = _get_record(cluster_name)
     PoolBackend().teardown(record["handle"], terminate=True)
 
 
 def autostop(cluster_name: str, idle_minutes: int, down_: bool = False
              ) -> None:
+    users.check_cluster_owner(cluster_name)
     record = _get_record(cluster_name)
     if record["status"] != global_state.UP:
         raise ClusterNotUpError(f"cluster {cluster_name!r} is not UP")
@@ -90,6 +94,7 @@ def queue(cluster_name: str) -> List[Dict[str, Any]]:
 
 def cancel(cluster_name: str, job_ids: Optional[List[int]] = None,
            all_jobs: bool = False) -> int:
+    users.check_cluster_owner(cluster_name)
     record = _get_record(cluster_name)
     return PoolBackend().cancel_jobs(record["handle"],
                                      None if all_jobs else job_ids)

@@ -6,6 +6,10 @@ class SkyAmdError(Exception):
     """Base class for all framework errors."""
 
 
+class PermissionDeniedError(SkyAmdError):
+    """RBAC: the requesting user's role does not allow this operation."""
+
+
 class ClusterNotUpError(SkyAmdError):
     pass
 

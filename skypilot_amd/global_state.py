@@ -94,9 +94,21 @@ def _conn():
         conn.close()
 
 
+_request_user = threading.local()
+
+
+def set_request_user(name):
+    """Thread-scoped identity for SHORT in-process request handlers
+    (LONG handlers run in their own process and use the env var)."""
+    _request_user.name = name
+
+
 def current_user() -> str:
-    """reference: sky/models.py User — here: env override or OS user."""
-    return os.environ.get("SKY_AMD_USER") or os.environ.get("USER", "root")
+    """reference: sky/models.py User — request identity (thread-local,
+    set by the API executor), else env override, else OS user."""
+    return (getattr(_request_user, "name", None)
+            or os.environ.get("SKY_AMD_USER")
+            or os.environ.get("USER", "root"))
 
 
 def add_or_update_cluster(name: str, status: str, handle: Dict[str, Any],

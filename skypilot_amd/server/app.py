@@ -15,10 +15,10 @@ import time
 from pathlib import Path
 from typing import Any, Dict, Optional
 
-from fastapi import FastAPI, HTTPException
+from fastapi import FastAPI, HTTPException, Request
 from fastapi.responses import StreamingResponse
 
-from skypilot_amd import global_state
+from skypilot_amd import global_state, users
 from skypilot_amd.server import executor
 from skypilot_amd.server import handlers  # noqa: F401  (registers handlers)
 from skypilot_amd.server import requests_db as rdb
@@ -32,6 +32,8 @@ def create_app(start_workers: bool = True) -> FastAPI:
 
     @app.on_event("startup")
     def _startup():
+        # the server's own OS identity bootstraps as admin
+        users.ensure_user(global_state.current_user())
         if start_workers:
             executor.start_workers()
         # serve/jobs background refreshers (reference: server/daemons.py)
@@ -42,11 +44,73 @@ def create_app(start_workers: bool = True) -> FastAPI:
     def health():
         return {"ok": True, "version": "0.1.0", "pid": os.getpid()}
 
+    # ---- identity & RBAC (reference: sky/users/rbac.py + server auth) ----
+    def _identity(request: Request) -> Dict[str, str]:
+        """Resolve (user, role) from a bearer service-account token, the
+        X-Skypilot-User header, or the server's own OS user."""
+        auth = request.headers.get("authorization", "")
+        if auth.lower().startswith("bearer "):
+            sa = users.resolve_token(auth[7:].strip())
+            if sa is None:
+                raise HTTPException(401, "invalid service-account token")
+            return {"user": sa["name"], "role": sa["role"]}
+        name = request.headers.get("x-skypilot-user")
+        if name:
+            return {"user": name, "role": users.ensure_user(name)}
+        me = global_state.current_user()
+        return {"user": me, "role": users.ensure_user(me)}
+
+    def _require_admin(request: Request) -> Dict[str, str]:
+        ident = _identity(request)
+        if ident["role"] != "admin":
+            raise HTTPException(
+                403, f"admin role required (you are {ident['role']})")
+        return ident
+
+    @app.get("/api/users")
+    def api_users(request: Request):
+        _identity(request)
+        return users.list_users()
+
+    @app.post("/api/users/role")
+    def api_users_role(request: Request, body: Dict[str, Any]):
+        _require_admin(request)
+        try:
+            users.set_role(body["name"], body["role"])
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+        return {"ok": True}
+
+    @app.post("/api/users/token")
+    def api_users_token(request: Request, body: Dict[str, Any]):
+        ident = _require_admin(request)
+        try:
+            token = users.create_token(body["name"], ident["user"],
+                                       body.get("role", "user"))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+        return {"token": token}
+
+    @app.get("/api/users/tokens")
+    def api_users_tokens(request: Request):
+        _require_admin(request)
+        return users.list_tokens()
+
+    @app.post("/api/users/token/revoke")
+    def api_users_token_revoke(request: Request, body: Dict[str, Any]):
+        _require_admin(request)
+        return {"revoked": users.revoke_token(body["name"])}
+
     # ---- generic async request plumbing -----------------------------------
     @app.post(API_PREFIX + "/{name}")
-    def submit(name: str, body: Dict[str, Any] = None):
+    def submit(name: str, request: Request, body: Dict[str, Any] = None):
+        ident = _identity(request)
         try:
-            rid = executor.schedule(name, body or {})
+            users.authorize(ident["role"], name)
+        except Exception as e:
+            raise HTTPException(403, str(e))
+        try:
+            rid = executor.schedule(name, body or {}, user=ident["user"])
         except KeyError:
             raise HTTPException(404, f"unknown request type {name!r}")
         return {"request_id": rid}

@@ -51,6 +51,8 @@ def handler(name: str) -> Callable:
 def _run_request_inline(req: Dict[str, Any]) -> None:
     rid = req["request_id"]
     buf = io.StringIO()
+    from skypilot_amd import global_state
+    global_state.set_request_user(req.get("user"))  # thread-scoped RBAC id
     try:
         fn = handler(req["name"])
         with contextlib.redirect_stdout(buf), contextlib.redirect_stderr(buf):
@@ -60,6 +62,7 @@ def _run_request_inline(req: Dict[str, Any]) -> None:
         buf.write(traceback.format_exc())
         rdb.finish(rid, rdb.FAILED, error=f"{type(e).__name__}: {e}")
     finally:
+        global_state.set_request_user(None)
         out = buf.getvalue()
         if out:
             try:
@@ -156,10 +159,11 @@ def stop_workers():
     _workers = []
 
 
-def schedule(name: str, body: Dict[str, Any]) -> str:
+def schedule(name: str, body: Dict[str, Any],
+             user: "str | None" = None) -> str:
     if name not in _REGISTRY:
         raise KeyError(f"unknown request {name!r}")
-    return rdb.create(name, body, queue_of(name))
+    return rdb.create(name, body, queue_of(name), user=user)
 
 
 def cancel_request(rid: str) -> bool:

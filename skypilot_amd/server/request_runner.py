@@ -20,6 +20,8 @@ def main() -> int:
     with open(req["log_path"], "ab", buffering=0) as logf:
         os.dup2(logf.fileno(), 1)
         os.dup2(logf.fileno(), 2)
+    if req.get("user"):
+        os.environ["SKY_AMD_USER"] = req["user"]
     try:
         from skypilot_amd.server import executor
         import skypilot_amd.server.handlers  # noqa: F401 (registry)

@@ -31,7 +31,8 @@ CREATE TABLE IF NOT EXISTS requests (
     result TEXT,
     error TEXT,
     worker_pid INTEGER,
-    log_path TEXT
+    log_path TEXT,
+    user TEXT
 );
 """
 
@@ -52,21 +53,25 @@ def _conn():
         conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
         conn.executescript(_SCHEMA)
+        cols = [r[1] for r in conn.execute("PRAGMA table_info(requests)")]
+        if "user" not in cols:  # pre-RBAC databases
+            conn.execute("ALTER TABLE requests ADD COLUMN user TEXT")
         with conn:
             yield conn
     finally:
         conn.close()
 
 
-def create(name: str, body: Dict[str, Any], queue: str) -> str:
+def create(name: str, body: Dict[str, Any], queue: str,
+           user: Optional[str] = None) -> str:
     rid = uuid.uuid4().hex[:16]
     log_path = str(api_dir() / "logs" / f"{rid}.log")
     with _conn() as c:
         c.execute(
             "INSERT INTO requests (request_id,name,queue,status,created_at,"
-            "body,log_path) VALUES (?,?,?,?,?,?,?)",
+            "body,log_path,user) VALUES (?,?,?,?,?,?,?,?)",
             (rid, name, queue, PENDING, time.time(), json.dumps(body),
-             log_path))
+             log_path, user))
     return rid
 
 

@@ -1,0 +1,96 @@
+"""RBAC: roles, service-account tokens, per-request authorization and
+cluster-ownership enforcement (reference: sky/users/rbac.py,
+sky/users/server.py token auth)."""
+import pytest
+
+from tests.test_orchestrator import sky_env, client  # noqa: F401 (fixtures)
+
+
+def test_bootstrap_admin_and_roles(sky_env, client):
+    # server's own identity bootstrapped as admin
+    me = client.get("/api/users").json()
+    assert any(u["role"] == "admin" for u in me)
+    # a fresh header identity is auto-created as plain user
+    r = client.get("/api/users", headers={"X-Skypilot-User": "alice"})
+    assert r.status_code == 200
+    roles = {u["name"]: u["role"] for u in client.get("/api/users").json()}
+    assert roles["alice"] == "user"
+
+
+def test_viewer_denied_mutations(sky_env, client):
+    client.post("/api/users/role",
+                json={"name": "bob", "role": "viewer"},
+                headers=None)
+    # bob must exist first (auto-create), then demote
+    client.get("/api/users", headers={"X-Skypilot-User": "bob"})
+    assert client.post("/api/users/role",
+                       json={"name": "bob", "role": "viewer"}
+                       ).status_code == 200
+    hdr = {"X-Skypilot-User": "bob"}
+    r = client.post("/api/v1/launch",
+                    json={"task": {"run": "true"}}, headers=hdr)
+    assert r.status_code == 403
+    # read-only still fine
+    assert client.post("/api/v1/status", json={}, headers=hdr
+                       ).status_code == 200
+
+
+def test_non_admin_cannot_manage_users(sky_env, client):
+    hdr = {"X-Skypilot-User": "carol"}
+    client.get("/api/users", headers=hdr)  # auto-create as 'user'
+    r = client.post("/api/users/token",
+                    json={"name": "ci"}, headers=hdr)
+    assert r.status_code == 403
+    r = client.post("/api/users/role",
+                    json={"name": "carol", "role": "admin"}, headers=hdr)
+    assert r.status_code == 403
+
+
+def test_service_account_token_auth(sky_env, client):
+    r = client.post("/api/users/token",
+                    json={"name": "ci-bot", "role": "user"})
+    assert r.status_code == 200
+    tok = r.json()["token"]
+    assert tok.startswith("sky_")
+    hdr = {"Authorization": f"Bearer {tok}"}
+    assert client.post("/api/v1/status", json={},
+                       headers=hdr).status_code == 200
+    toks = client.get("/api/users/tokens").json()
+    assert any(t["name"] == "ci-bot" for t in toks)
+    # bad token rejected
+    bad = {"Authorization": "Bearer sky_deadbeef"}
+    assert client.post("/api/v1/status", json={},
+                       headers=bad).status_code == 401
+    # revoke kills it
+    client.post("/api/users/token/revoke", json={"name": "ci-bot"})
+    assert client.post("/api/v1/status", json={},
+                       headers=hdr).status_code == 401
+
+
+def test_cluster_ownership_enforced(sky_env, client):
+    import time
+    # admin (default identity) launches a cluster
+    r = client.post("/api/v1/launch",
+                    json={"task": {"run": "sleep 0.1"},
+                          "cluster_name": "owned-c"})
+    rid = r.json()["request_id"]
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        st = client.get("/api/get", params={"request_id": rid}).json()
+        if st["status"] in ("SUCCEEDED", "FAILED", "CANCELLED"):
+            break
+        time.sleep(0.5)
+    assert st["status"] == "SUCCEEDED", st
+    # another plain user may not down it
+    hdr = {"X-Skypilot-User": "mallory"}
+    r = client.post("/api/v1/down",
+                    json={"cluster_name": "owned-c"}, headers=hdr)
+    rid = r.json()["request_id"]
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        st = client.get("/api/get", params={"request_id": rid}).json()
+        if st["status"] in ("SUCCEEDED", "FAILED", "CANCELLED"):
+            break
+        time.sleep(0.5)
+    assert st["status"] == "FAILED"
+    assert "PermissionDenied" in (st.get("error") or "")
