@@ -14,7 +14,9 @@ from typing import Any, Dict, Optional, Tuple
 from skypilot_amd import global_state
 from skypilot_amd.backends.pool_backend import PoolBackend
 from skypilot_amd.dag import to_dag
-from skypilot_amd.exceptions import ClusterDoesNotExist, ClusterNotUpError
+from skypilot_amd.exceptions import (ClusterDoesNotExist,
+                                     ClusterNotUpError,
+                                     ResourcesUnavailableError)
 from skypilot_amd.optimizer import Optimizer
 from skypilot_amd.utils.timeline import event as timeline_event
 
@@ -53,7 +55,25 @@ def _execute(task_or_dag, cluster_name: Optional[str], stages,
 
     handle: Optional[Dict[str, Any]] = None
     if Stage.PROVISION in stages:
-        handle = backend.provision(task, cluster_name)
+        # Candidate failover (reference: RetryingVmProvisioner over
+        # any_of/ordered resources): try each candidate in order until
+        # one provisions.
+        cands = task.resources.candidates or (task.resources,)
+        last_err: Optional[Exception] = None
+        for cand in cands:
+            task.resources = cand
+            try:
+                handle = backend.provision(task, cluster_name)
+                last_err = None
+                break
+            except ResourcesUnavailableError as e:
+                last_err = e
+                global_state.add_cluster_event(
+                    cluster_name, "PROVISION_FAILOVER",
+                    f"{cand.accelerators or 'cpu'}:"
+                    f"{cand.accelerator_count}: {e}")
+        if last_err is not None:
+            raise last_err
     else:
         record = global_state.get_cluster(cluster_name)
         if record is None:

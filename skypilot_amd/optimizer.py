@@ -18,12 +18,25 @@ class Optimizer:
     def optimize(cls, dag: Dag, quiet: bool = True) -> Dag:
         dag = to_dag(dag)
         for task in dag.tasks:
-            cls._check_feasible(task)
+            cands = task.resources.candidates or (task.resources,)
+            feasible, errs = [], []
+            for cand in cands:
+                try:
+                    cls._check_feasible(task, cand)
+                    feasible.append(cand)
+                except ResourcesUnavailableError as e:
+                    errs.append(str(e))
+            if not feasible:
+                raise ResourcesUnavailableError("; ".join(errs))
+            # keep the feasible candidates (in order) for provision-time
+            # failover in execution.py
+            task.resources = feasible[0]
+            task.resources.candidates = tuple(feasible)
         return dag
 
     @staticmethod
-    def _check_feasible(task) -> None:
-        res = task.resources
+    def _check_feasible(task, res=None) -> None:
+        res = res if res is not None else task.resources
         need = task.num_nodes * res.accelerator_count
         if need == 0:
             return

@@ -114,6 +114,11 @@ class Resources:
     labels: Dict[str, str] = field(default_factory=dict)
     autostop: Optional[AutostopConfig] = None
     priority: Optional[int] = None
+    # Failover candidates (reference: any_of/ordered resources,
+    # sky/resources.py multi-candidate sets).  candidates[0] is self's
+    # config; execution tries each in order on
+    # ResourcesUnavailableError.
+    candidates: tuple = ()
     _raw: Dict[str, Any] = field(default_factory=dict, repr=False)
 
     KNOWN_KEYS = {
@@ -135,14 +140,22 @@ class Resources:
             raise TaskValidationError(
                 f"unknown resources keys: {sorted(unknown)}")
         if "any_of" in cfg or "ordered" in cfg:
-            # Multi-candidate resources: on a single pool we take the first
-            # feasible candidate (reference semantics preserved shallowly).
-            cands = cfg.get("any_of") or cfg.get("ordered")
+            # Multi-candidate resources (reference: sky/resources.py
+            # any_of/ordered): `ordered` keeps the given preference
+            # order; `any_of` is sorted cheapest-first (fewest
+            # accelerators, then cpus).  execution.py retries down the
+            # list on ResourcesUnavailableError.
+            cand_cfgs = cfg.get("ordered") or cfg.get("any_of")
             base = {k: v for k, v in cfg.items() if k not in ("any_of",
                                                               "ordered")}
-            merged = dict(base)
-            merged.update(cands[0])
-            return cls.from_yaml_config(merged)
+            cands = [cls.from_yaml_config({**base, **c})
+                     for c in cand_cfgs]
+            if "ordered" not in cfg:
+                cands.sort(key=lambda r: (r.accelerator_count,
+                                          r.cpus or 0))
+            primary = cands[0]
+            primary.candidates = tuple(cands)
+            return primary
         acc, n = parse_accelerators(cfg.get("accelerators"))
         cpus, cpus_min = _parse_plus(cfg.get("cpus"))
         mem, mem_min = _parse_plus(cfg.get("memory"))

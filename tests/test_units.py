@@ -83,3 +83,68 @@ def test_optimizer_feasibility(monkeypatch):
     with pytest.raises(ResourcesUnavailableError):
         Optimizer.optimize(bad)
     gpu_topology.detect_gpus.cache_clear()
+
+
+def test_any_of_candidates_sorted_and_ordered_kept():
+    from skypilot_amd.resources import Resources
+    r = Resources.from_yaml_config({
+        "any_of": [{"accelerators": "MI355X:8"},
+                   {"accelerators": "MI355X:1"}]})
+    assert [c.accelerator_count for c in r.candidates] == [1, 8]
+    r = Resources.from_yaml_config({
+        "ordered": [{"accelerators": "MI355X:8"},
+                    {"accelerators": "MI355X:1"}]})
+    assert [c.accelerator_count for c in r.candidates] == [8, 1]
+
+
+def test_optimizer_prunes_infeasible_candidate(tmp_path, monkeypatch):
+    """First candidate infeasible (999 GPUs on a 0-GPU box) -> the
+    optimizer prunes it and launch uses the cpu-only candidate
+    (reference: optimizer candidate enumeration)."""
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    from skypilot_amd import execution
+    from skypilot_amd.task import Task
+    task = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"ordered": [{"accelerators": "MI355X:999"},
+                                  {"cpus": 1}]},
+    })
+    job_id, handle = execution.launch(task, "failover-c",
+                                      detach_run=True)
+    assert handle is not None
+    assert task.resources.accelerator_count == 0  # fell to candidate 2
+    from skypilot_amd import core
+    core.down("failover-c")
+
+
+def test_provision_time_failover(tmp_path, monkeypatch):
+    """Feasible-looking candidate fails at provision time (lease race)
+    -> execution retries the next candidate
+    (reference: RetryingVmProvisioner)."""
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    from skypilot_amd import execution, global_state
+    from skypilot_amd.backends.pool_backend import PoolBackend
+    from skypilot_amd.exceptions import ResourcesUnavailableError
+    from skypilot_amd.task import Task
+    real = PoolBackend.provision
+    calls = {"n": 0}
+
+    def flaky(self, task, cluster_name, **kw):
+        calls["n"] += 1
+        if calls["n"] == 1:
+            raise ResourcesUnavailableError("lease lost")
+        return real(self, task, cluster_name, **kw)
+
+    monkeypatch.setattr(PoolBackend, "provision", flaky)
+    task = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"ordered": [{"cpus": 1}, {"cpus": 2}]},
+    })
+    job_id, handle = execution.launch(task, "failover-d",
+                                      detach_run=True)
+    assert handle is not None and calls["n"] == 2
+    ev = [e["event"] for e in
+          global_state.get_cluster_events("failover-d")]
+    assert "PROVISION_FAILOVER" in ev
+    from skypilot_amd import core
+    core.down("failover-d")
