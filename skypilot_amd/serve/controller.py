@@ -243,8 +243,16 @@ class ServiceRuntime:
         st.update_service(self.name, controller_pid=os.getpid(),
                           status=st.REPLICA_INIT)
         threading.Thread(target=self.controller_loop, daemon=True).start()
+        ssl_kw = {}
+        if self.spec.tls is not None:
+            tls = self.spec.tls.ensure_materialized(
+                st.service_dir(self.name) / "tls")
+            ssl_kw = {"ssl_certfile": tls.certfile,
+                      "ssl_keyfile": tls.keyfile}
+            st.update_service(self.name, tls=True)
         config = uvicorn.Config(self.lb_app(), host="127.0.0.1",
-                                port=self.lb_port, log_level="warning")
+                                port=self.lb_port, log_level="warning",
+                                **ssl_kw)
         server = uvicorn.Server(config)
 
         def watch_shutdown():

@@ -57,6 +57,10 @@ def _conn():
         conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
         conn.executescript(_SCHEMA)
+        cols = [r[1] for r in conn.execute("PRAGMA table_info(services)")]
+        if "tls" not in cols:
+            conn.execute("ALTER TABLE services ADD COLUMN tls INTEGER "
+                         "DEFAULT 0")
         with conn:
             yield conn
     finally:
@@ -84,6 +88,13 @@ def bump_version(name: str, task: Dict[str, Any],
         row = c.execute("SELECT version FROM services WHERE name=?",
                         (name,)).fetchone()
     return row[0] if row else 0
+
+
+def service_dir(name: str):
+    from pathlib import Path
+    d = Path(global_state.root_dir()) / "serve" / name
+    d.mkdir(parents=True, exist_ok=True)
+    return d
 
 
 def update_service(name: str, **fields) -> None:

@@ -58,11 +58,51 @@ class ReplicaPolicy:
 
 
 @dataclass
+class TLSConfig:
+    """LB TLS termination (reference: sky/serve service schema `tls:`
+    keyfile/certfile).  `tls: true` auto-generates a self-signed pair
+    under the service dir (openssl), for pools without a CA."""
+    certfile: Optional[str] = None
+    keyfile: Optional[str] = None
+    auto: bool = False
+
+    @classmethod
+    def from_config(cls, cfg) -> Optional["TLSConfig"]:
+        if not cfg:
+            return None
+        if cfg is True:
+            return cls(auto=True)
+        if isinstance(cfg, dict):
+            return cls(certfile=cfg.get("certfile"),
+                       keyfile=cfg.get("keyfile"),
+                       auto=bool(cfg.get("auto", False)))
+        raise TaskValidationError(f"bad tls config: {cfg!r}")
+
+    def ensure_materialized(self, out_dir) -> "TLSConfig":
+        """Generate a self-signed pair if auto and none given."""
+        import pathlib
+        import subprocess
+        if self.certfile and self.keyfile:
+            return self
+        d = pathlib.Path(out_dir)
+        d.mkdir(parents=True, exist_ok=True)
+        cert, key = d / "lb.crt", d / "lb.key"
+        if not (cert.exists() and key.exists()):
+            subprocess.run(
+                ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+                 "-keyout", str(key), "-out", str(cert), "-days", "3650",
+                 "-nodes", "-subj", "/CN=skypilot-amd-lb"],
+                check=True, capture_output=True)
+        return TLSConfig(certfile=str(cert), keyfile=str(key))
+
+
+@dataclass
 class ServiceSpec:
     readiness_probe: ReadinessProbe = field(default_factory=ReadinessProbe)
     policy: ReplicaPolicy = field(default_factory=ReplicaPolicy)
     load_balancing_policy: str = "least_load"
     port: Optional[int] = None  # replica port (task may also use $PORT)
+    tls: Optional[TLSConfig] = None
 
     KNOWN_KEYS = {"readiness_probe", "replica_policy", "replicas",
                   "load_balancing_policy", "load_balancer", "ports", "tls",
@@ -92,4 +132,5 @@ class ServiceSpec:
             load_balancing_policy=cfg.get("load_balancing_policy",
                                           "least_load"),
             port=int(port) if port is not None else None,
+            tls=TLSConfig.from_config(cfg.get("tls")),
         )
