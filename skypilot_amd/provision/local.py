@@ -102,14 +102,41 @@ def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
     return handle
 
 
-def _agent_alive(cdir: Path) -> Optional[int]:
+def _agent_meta(cdir: Path) -> Optional[Dict]:
     meta = cdir / "agent.json"
     if not meta.exists():
         return None
     try:
-        info = json.loads(meta.read_text())
-        client = AgentClient(info["port"])
-        if client.healthy():
+        return json.loads(meta.read_text())
+    except Exception:  # noqa: BLE001
+        return None
+
+
+def _agent_pid_alive(cdir: Path) -> Optional[Dict]:
+    """Process-level liveness only (the pid exists)."""
+    info = _agent_meta(cdir)
+    if info is None:
+        return None
+    try:
+        pid = info["pid"]
+        os.kill(pid, 0)
+        # a SIGKILLed-but-unreaped agent is a zombie, not a live agent
+        with open(f"/proc/{pid}/stat") as f:
+            if f.read().rsplit(")", 1)[1].split()[0] == "Z":
+                return None
+        return info
+    except (ProcessLookupError, PermissionError, KeyError, OSError,
+            IndexError):
+        return None
+
+
+def _agent_alive(cdir: Path) -> Optional[int]:
+    """Full liveness: pid exists AND the HTTP endpoint answers."""
+    info = _agent_pid_alive(cdir)
+    if info is None:
+        return None
+    try:
+        if AgentClient(info["port"], timeout=5.0).healthy():
             return info["port"]
     except Exception:  # noqa: BLE001
         pass
@@ -169,10 +196,22 @@ def _kill_agent(handle: Dict[str, Any]) -> None:
 
 
 def query_instances(cluster_name: str, handle: Dict[str, Any]) -> str:
+    """UP = agent process alive AND answering HTTP.  A live pid with a
+    dead/hung HTTP endpoint (partition, wedged agent) reports INIT so
+    `sky status -r` surfaces the degradation without hanging
+    (reference: abnormal clusters map to INIT)."""
     cdir = handle.get("cluster_dir")
-    if cdir and _agent_alive(Path(cdir)) is not None:
-        return global_state.UP
-    return global_state.STOPPED
+    if not cdir:
+        return global_state.STOPPED
+    info = _agent_pid_alive(Path(cdir))
+    if info is None:
+        return global_state.STOPPED
+    try:
+        if AgentClient(info["port"], timeout=2.0).healthy():
+            return global_state.UP
+    except Exception:  # noqa: BLE001
+        pass
+    return global_state.INIT
 
 
 def get_cluster_info(handle: Dict[str, Any]) -> Dict[str, Any]:

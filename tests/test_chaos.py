@@ -54,3 +54,30 @@ def test_agent_killed_detected_by_refresh(client):
     rec = next(r for r in records if r["name"] == "t-chaos2")
     assert rec["status"] == "UP"
     sdk.get(sdk.down("t-chaos2"))
+
+
+def test_network_partition_reports_init_and_recovers(client):
+    """SIGSTOP the agent (alive pid, dead HTTP = partition/wedge): the
+    status refresh must complete promptly and report INIT, then UP again
+    after SIGCONT (reference: abnormal clusters -> INIT)."""
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"name": "ap", "run": "true",
+                        "resources": {"cpus": 1}}, "t-chaos3"),
+            timeout=60)
+    home = Path(os.environ["SKY_AMD_HOME"])
+    meta = home / "clusters" / "t-chaos3" / "agent.json"
+    pid = json.loads(meta.read_text())["pid"]
+    os.kill(pid, signal.SIGSTOP)
+    try:
+        t0 = time.time()
+        records = sdk.get(sdk.status(refresh=True), timeout=60)
+        elapsed = time.time() - t0
+        rec = next(r for r in records if r["name"] == "t-chaos3")
+        assert rec["status"] == "INIT", rec
+        assert elapsed < 30, f"refresh hung {elapsed:.0f}s on partition"
+    finally:
+        os.kill(pid, signal.SIGCONT)
+    records = sdk.get(sdk.status(refresh=True), timeout=60)
+    rec = next(r for r in records if r["name"] == "t-chaos3")
+    assert rec["status"] == "UP"
+    sdk.get(sdk.down("t-chaos3"))
