@@ -191,6 +191,17 @@ def remove_cluster(name: str) -> None:
         c.execute("DELETE FROM clusters WHERE name=?", (name,))
 
 
+def list_cluster_history(limit: int = 50):
+    with _conn() as c:
+        rows = c.execute(
+            "SELECT name, launched_at, torn_down_at, resources FROM "
+            "cluster_history ORDER BY torn_down_at DESC LIMIT ?",
+            (limit,)).fetchall()
+    return [{"name": r[0], "launched_at": r[1], "torn_down_at": r[2],
+             "resources": json.loads(r[3]) if r[3] else {}}
+            for r in rows]
+
+
 def add_cluster_event(cluster: str, event: str, detail: str = "") -> None:
     with _DB_LOCK, _conn() as c:
         c.execute(

@@ -25,6 +25,7 @@ def render() -> str:
     from skypilot_amd import global_state
     from skypilot_amd.jobs import state as jobs_state
     from skypilot_amd.serve import serve_state
+    from skypilot_amd.server import requests_db as rdb
     from skypilot_amd.utils.gpu_topology import detect_gpus
 
     clusters = []
@@ -56,6 +57,27 @@ def render() -> str:
     gpus = [[g.index, g.name, f"{g.memory_gb} GB", g.numa_node,
              gpus_taken.get(g.index, "-")] for g in detect_gpus()]
 
+    # recent API requests (reference dashboard: requests table)
+    reqs = []
+    for r in rdb.list_requests(limit=15):
+        dur = ""
+        if r.get("started_at"):
+            end = r.get("finished_at") or time.time()
+            dur = f"{end - r['started_at']:.1f}s"
+        reqs.append([r["request_id"][:8], r["name"],
+                     r.get("user") or "-", r["status"], dur,
+                     time.strftime("%H:%M:%S",
+                                   time.localtime(r["created_at"]))])
+
+    # teardown history (reference dashboard: cluster history view)
+    hist = [[h["name"],
+             time.strftime("%m-%d %H:%M",
+                           time.localtime(h.get("launched_at") or 0)),
+             time.strftime("%m-%d %H:%M",
+                           time.localtime(h.get("torn_down_at") or 0)),
+             f"{((h.get('torn_down_at') or 0) - (h.get('launched_at') or 0)) / 60:.0f} min"]
+            for h in global_state.list_cluster_history(limit=10)]
+
     return f"""<!doctype html><html><head><title>skypilot-amd</title>
 <meta http-equiv="refresh" content="5">
 <style>
@@ -74,5 +96,9 @@ th {{ background:#222; color:#e8443a; }}
         jobs)}
 {_table("Services", ["name", "status", "ready", "endpoint"], services)}
 {_table("Pool GPUs", ["idx", "model", "HBM", "numa", "used by"], gpus)}
+{_table("Recent API requests",
+        ["id", "type", "user", "status", "duration", "at"], reqs)}
+{_table("Cluster history", ["name", "launched", "torn down", "lifetime"],
+        hist)}
 <p style="color:#555">auto-refreshes every 5s · {time.strftime("%H:%M:%S")}
 </p></body></html>"""
