@@ -436,6 +436,19 @@ def users_revoke(name):
         _print_result(r.json())
 
 
+@cli.command("cost-report")
+def cost_report_cmd():
+    """GPU-hour usage per cluster (live + torn down)."""
+    rows = sdk.get(sdk.cost_report())
+    fmt = "{:<20} {:<10} {:<11} {:>5} {:>10} {:>10}"
+    click.echo(fmt.format("NAME", "USER", "STATUS", "GPUS", "HOURS",
+                          "GPU-HOURS"))
+    for r in rows:
+        click.echo(fmt.format(r["name"], str(r.get("user") or "-"),
+                              r["status"], r["gpus"],
+                              r["duration_hours"], r["gpu_hours"]))
+
+
 @cli.command("recipes")
 def recipes_cmd():
     """List bundled task recipes (examples/*.yaml)."""

@@ -74,6 +74,36 @@ def api_start(wait: float = 15.0) -> bool:
     raise ApiServerError("API server failed to start; see ~/.sky_amd_api.log")
 
 
+CLIENT_API_VERSION = 2
+_version_checked = False
+
+
+def check_server_compat() -> None:
+    """One-shot client/server API-version handshake (reference:
+    sky/server versions compat).  Old server + new client: warn and
+    continue; server that no longer serves this client: hard error."""
+    global _version_checked
+    if _version_checked:
+        return
+    try:
+        with _client() as c:
+            info = c.get("/health").json()
+    except Exception:  # noqa: BLE001
+        return  # health failures surface elsewhere
+    _version_checked = True
+    srv = info.get("api_version", 1)
+    min_client = info.get("min_client_api_version", 1)
+    if CLIENT_API_VERSION < min_client:
+        raise ApiServerError(
+            f"this client speaks API v{CLIENT_API_VERSION} but the "
+            f"server requires >= v{min_client}; upgrade the client")
+    if srv < CLIENT_API_VERSION:
+        import sys
+        print(f"[sky] note: server API v{srv} is older than client "
+              f"v{CLIENT_API_VERSION}; some flags may be ignored",
+              file=sys.stderr)
+
+
 def api_healthy() -> bool:
     try:
         with _client() as c:
@@ -105,6 +135,7 @@ def _ensure_server():
 
 def _submit(name: str, body: Dict[str, Any]) -> str:
     _ensure_server()
+    check_server_compat()
     with _client() as c:
         r = c.post(f"/api/v1/{name}", json=body)
         if r.status_code != 200:
@@ -213,6 +244,10 @@ def cancel(cluster_name: str, job_ids: Optional[List[int]] = None,
 def job_status(cluster_name: str, job_id: int) -> str:
     return _submit("job_status", {"cluster_name": cluster_name,
                                   "job_id": job_id})
+
+
+def cost_report():
+    return _submit("cost_report", {})
 
 
 def check() -> str:

@@ -163,3 +163,37 @@ def cluster_events(cluster_name: str) -> List[Dict[str, Any]]:
 def topology() -> Dict[str, Any]:
     return {"xgmi": xgmi_topology(), "gpus": [g.__dict__ for g in
                                               detect_gpus()]}
+
+
+def cost_report() -> List[Dict[str, Any]]:
+    """GPU-hour accounting per cluster, live and historical
+    (reference: sky/client/cli cost-report backed by cluster_history;
+    a local pool has no $ prices, so the unit is GPU-hours)."""
+    now = time.time()
+    rows: List[Dict[str, Any]] = []
+    for r in global_state.list_clusters():
+        h = r["handle"]
+        ngpu = h.get("gpus_per_node", 0) * h.get("num_nodes", 1)
+        dur_h = (now - (r.get("launched_at") or now)) / 3600
+        rows.append({
+            "name": r["name"], "user": r.get("user"), "status":
+            r["status"], "gpus": ngpu,
+            "duration_hours": round(dur_h, 3),
+            "gpu_hours": round(ngpu * dur_h, 3), "live": True,
+        })
+    for hrec in global_state.list_cluster_history(limit=200):
+        res = hrec.get("resources") or {}
+        acc = str(res.get("accelerators") or "")
+        ngpu = 0
+        if ":" in acc:
+            ngpu = int(acc.split(":")[1])
+        elif acc:
+            ngpu = 1
+        dur_h = max(0.0, ((hrec.get("torn_down_at") or 0) -
+                          (hrec.get("launched_at") or 0)) / 3600)
+        rows.append({
+            "name": hrec["name"], "user": None, "status": "TERMINATED",
+            "gpus": ngpu, "duration_hours": round(dur_h, 3),
+            "gpu_hours": round(ngpu * dur_h, 3), "live": False,
+        })
+    return rows

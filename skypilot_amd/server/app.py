@@ -25,6 +25,10 @@ from skypilot_amd.server import requests_db as rdb
 
 DEFAULT_PORT = 46580
 API_PREFIX = "/api/v1"
+# Bump API_VERSION on wire-format changes; raise MIN_CLIENT_API_VERSION
+# only when old clients can no longer be served.
+API_VERSION = 2
+MIN_CLIENT_API_VERSION = 1
 
 
 def create_app(start_workers: bool = True) -> FastAPI:
@@ -42,7 +46,11 @@ def create_app(start_workers: bool = True) -> FastAPI:
 
     @app.get("/health")
     def health():
-        return {"ok": True, "version": "0.1.0", "pid": os.getpid()}
+        # api_version gates client compatibility (reference:
+        # sky/server/constants.py API_VERSION + version-mismatch errors)
+        return {"ok": True, "version": "0.1.0", "api_version": API_VERSION,
+                "min_client_api_version": MIN_CLIENT_API_VERSION,
+                "pid": os.getpid()}
 
     # ---- identity & RBAC (reference: sky/users/rbac.py + server auth) ----
     def _identity(request: Request) -> Dict[str, str]:

@@ -78,6 +78,12 @@ def job_status(cluster_name: str, job_id: int):
     return core.job_status(cluster_name, job_id)
 
 
+@register("cost_report", SHORT)
+def cost_report():
+    """reference: sky cost-report (GPU-hours on a local pool)."""
+    return core.cost_report()
+
+
 @register("check", SHORT)
 def check() -> Dict[str, Any]:
     return core.check()

@@ -33,7 +33,7 @@ READONLY_REQUESTS = {
     "status", "queue", "job_status", "check", "show_gpus",
     "cluster_events", "storage_list", "volumes_list", "recipes_list",
     "jobs_queue", "jobs_logs", "jobs_pool_status", "serve_status",
-    "serve_logs",
+    "serve_logs", "cost_report",
 }
 
 _SCHEMA = """

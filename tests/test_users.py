@@ -94,3 +94,42 @@ def test_cluster_ownership_enforced(sky_env, client):
         time.sleep(0.5)
     assert st["status"] == "FAILED"
     assert "PermissionDenied" in (st.get("error") or "")
+
+
+def test_api_version_handshake(sky_env, client, monkeypatch, capsys):
+    """Old-server warning and too-old-client rejection (reference:
+    sky/server API version compat)."""
+    from skypilot_amd.client import sdk
+    from skypilot_amd.exceptions import ApiServerError
+    h = client.get("/health").json()
+    assert h["api_version"] >= h["min_client_api_version"]
+    # same version: silent
+    monkeypatch.setattr(sdk, "_version_checked", False)
+    sdk.check_server_compat()
+    # client older than server's minimum: hard error
+    monkeypatch.setattr(sdk, "_version_checked", False)
+    monkeypatch.setattr(sdk, "CLIENT_API_VERSION", 0)
+    import pytest as _pytest
+    with _pytest.raises(ApiServerError):
+        sdk.check_server_compat()
+    # server older than client: stderr note, no error
+    monkeypatch.setattr(sdk, "_version_checked", False)
+    monkeypatch.setattr(sdk, "CLIENT_API_VERSION", 99)
+    sdk.check_server_compat()
+    assert "older than client" in capsys.readouterr().err
+
+
+def test_cost_report(sky_env, client):
+    """GPU-hour accounting includes live and historical clusters."""
+    import time
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"run": "true", "resources": {"cpus": 1}},
+                       "cost-c"), timeout=60)
+    rows = sdk.get(sdk.cost_report())
+    live = next(r for r in rows if r["name"] == "cost-c")
+    assert live["live"] and live["status"] == "UP"
+    sdk.get(sdk.down("cost-c"))
+    rows = sdk.get(sdk.cost_report())
+    hist = [r for r in rows if r["name"] == "cost-c" and not r["live"]]
+    assert hist and hist[0]["status"] == "TERMINATED"
+    assert hist[0]["duration_hours"] >= 0
