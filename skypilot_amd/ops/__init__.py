@@ -336,11 +336,12 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     i, o = x.shape[-1], weight.shape[0]
     wbytes = 2 * i * o
     # Measured routing (bench_gemv.py on MI355X, docs/BENCHMARKS.md):
-    # n=1 the GEMV streams W at 6.5-7.3 TB/s and always wins (hipBLASLt
-    # floor is ~19 us/GEMM); at larger n it goes VALU-bound on big
-    # shapes and only wins where hipBLASLt is launch-bound.
-    use_native = (n == 1 or (n <= 4 and wbytes <= 34_000_000)
-                  or wbytes <= 10_000_000)
+    # n=1 always wins (W streams at 6.5-7.3 TB/s; hipBLASLt floors at
+    # ~19 us/GEMM); n>=2 wins on small-W shapes (launch-bound for
+    # hipBLASLt) and, via the R=4 multi-row variant, on vocab-sized
+    # projections (O >= 100k) up to n<=4.
+    use_native = (n == 1 or wbytes <= 34_000_000
+                  or (o >= 100_000 and n <= 4))
     if (use_native and x.is_cuda and x.dtype == torch.bfloat16
             and 1 <= n <= 8 and i % 512 == 0):
         C = _require_native("skinny_gemm")

@@ -34,6 +34,48 @@ __device__ __forceinline__ float dot2(s16x8 a, s16x8 b, float acc) {
   return acc;
 }
 
+// Multi-row variant for N >= 2 on large O (>= 16384 keeps the grid
+// full at R=4): each wave owns R=4 consecutive output
+// rows, so every X chunk loaded from L2 feeds 4 dot products — N=8's
+// X re-read traffic through L2 (8x the W bytes at R=1) drops 4x, which
+// is what capped the R=1 kernel at ~2.6 TB/s on the big shapes.
+template <int N, int R>
+__global__ __launch_bounds__(256) void skinny_gemm_multirow_kernel(
+    const unsigned short* __restrict__ W,
+    const unsigned short* __restrict__ X,
+    unsigned short* __restrict__ Y, int I, int O) {
+  int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  int o0 = (blockIdx.x * 4 + wave) * R;
+  if (o0 >= O) return;
+  float acc[R][N];
+#pragma unroll
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int b = 0; b < N; ++b) acc[r][b] = 0.f;
+  const unsigned short* wrow = W + (long long)o0 * I;
+  for (int i = lane * 8; i < I; i += 512) {
+    s16x8 wv[R];
+#pragma unroll
+    for (int r = 0; r < R; ++r)
+      wv[r] = __builtin_nontemporal_load(
+          (const s16x8*)(wrow + (long long)r * I + i));
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      s16x8 xv = *(const s16x8*)(X + (long long)b * I + i);
+#pragma unroll
+      for (int r = 0; r < R; ++r) acc[r][b] = dot2(wv[r], xv, acc[r][b]);
+    }
+  }
+#pragma unroll
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      float v = wave_reduce_sum(acc[r][b]);
+      if (lane == 0 && o0 + r < O)
+        Y[(long long)b * O + o0 + r] = f2bf(v);
+    }
+}
+
 template <int N>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     const unsigned short* __restrict__ W,
@@ -85,10 +127,15 @@ extern "C" void skinny_gemm_launch(const void* W, const void* X, void* Y,
   const unsigned short* w = (const unsigned short*)W;
   const unsigned short* x = (const unsigned short*)X;
   unsigned short* y = (unsigned short*)Y;
+  dim3 grid_mr((O + 15) / 16);
 #define CASE(n)                                                        \
   case n:                                                              \
-    hipLaunchKernelGGL(skinny_gemm_kernel<n>, grid, block, 0, stream,  \
-                       w, x, y, I, O);                                 \
+    if (n >= 2 && O >= 16384 && O % 4 == 0)                            \
+      hipLaunchKernelGGL((skinny_gemm_multirow_kernel<n, 4>), grid_mr, \
+                         block, 0, stream, w, x, y, I, O);             \
+    else                                                               \
+      hipLaunchKernelGGL(skinny_gemm_kernel<n>, grid, block, 0,        \
+                         stream, w, x, y, I, O);                       \
     break;
   switch (N) {
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
