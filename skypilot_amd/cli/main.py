@@ -355,6 +355,13 @@ def storage_ls():
     _print_result(sdk.get(sdk.storage_list()))
 
 
+@storage.command("sync")
+@click.argument("name")
+def storage_sync_cmd(name):
+    """Push a store to its S3-compatible remote (rclone)."""
+    _print_result(sdk.get(sdk.storage_sync(name)))
+
+
 @storage.command("delete")
 @click.argument("name")
 def storage_delete(name):

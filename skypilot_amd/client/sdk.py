@@ -246,6 +246,10 @@ def job_status(cluster_name: str, job_id: int) -> str:
                                   "job_id": job_id})
 
 
+def storage_sync(name: str):
+    return _submit("storage_sync", {"name": name})
+
+
 def cost_report():
     return _submit("cost_report", {})
 

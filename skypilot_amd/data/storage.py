@@ -28,6 +28,51 @@ MODE_COPY = "COPY"
 MODE_MOUNT = "MOUNT"
 MODE_MOUNT_CACHED = "MOUNT_CACHED"
 
+# Remote (S3-compatible) stores layer on via rclone (reference:
+# sky/data/storage.py S3Store/GcsStore/R2Store + data_transfer).  A
+# source like s3://bucket/path pulls into the local store dir at create
+# time and pushes back with sync_to_remote (sky storage sync) — the
+# local dir stays the mount contract, so COPY/MOUNT modes are unchanged.
+REMOTE_SCHEMES = ("s3://", "gs://", "r2://", "b2://", "minio://")
+
+
+def _is_remote(source: str | None) -> bool:
+    return bool(source) and source.startswith(REMOTE_SCHEMES)
+
+
+def _rclone_target(source: str) -> str:
+    """s3://bucket/path -> s3:bucket/path (rclone remote syntax; the
+    remote must be configured in ~/.config/rclone/rclone.conf)."""
+    scheme, rest = source.split("://", 1)
+    return f"{scheme}:{rest}"
+
+
+def _rclone(*args: str) -> None:
+    import shutil as _sh
+    import subprocess
+    if _sh.which("rclone") is None:
+        raise TaskValidationError(
+            "remote storage requires rclone on PATH (not bundled in "
+            "this offline image); configure ~/.config/rclone/rclone.conf")
+    proc = subprocess.run(["rclone", *args], capture_output=True,
+                          text=True)
+    if proc.returncode != 0:
+        raise TaskValidationError(
+            f"rclone {' '.join(args[:2])} failed: {proc.stderr[:400]}")
+
+
+def sync_to_remote(name: str) -> str:
+    """Push a store's local contents back to its remote source."""
+    rec = next((r for r in list_storage() if r["name"] == name), None)
+    if rec is None:
+        raise TaskValidationError(f"no such storage {name!r}")
+    if not _is_remote(rec["source"]):
+        raise TaskValidationError(
+            f"storage {name!r} has no remote source to sync to")
+    d = storage_root() / name
+    _rclone("sync", str(d), _rclone_target(rec["source"]))
+    return rec["source"]
+
 
 def storage_root() -> Path:
     d = global_state.root_dir() / "storage"
@@ -40,12 +85,15 @@ def get_or_create_store(name: str, source: str | None = None) -> Path:
     first = not d.exists()
     d.mkdir(parents=True, exist_ok=True)
     if first:
+        store_type = ("s3" if _is_remote(source) else "local")
         with global_state._DB_LOCK, global_state._conn() as c:
             c.execute(
                 "INSERT OR IGNORE INTO storage "
                 "(name,source,store_type,created_at) VALUES (?,?,?,?)",
-                (name, source or "", "local", time.time()))
-        if source:
+                (name, source or "", store_type, time.time()))
+        if _is_remote(source):
+            _rclone("sync", _rclone_target(source), str(d))
+        elif source:
             src = os.path.expanduser(source)
             if os.path.isdir(src):
                 LocalProcessCommandRunner().rsync(src.rstrip("/") + "/",

@@ -105,6 +105,13 @@ def storage_list():
     return st.list_storage()
 
 
+@register("storage_sync", SHORT)
+def storage_sync(name: str) -> Dict[str, Any]:
+    """Push a store to its S3-compatible remote (rclone)."""
+    from skypilot_amd.data import storage as storage_mod
+    return {"synced_to": storage_mod.sync_to_remote(name)}
+
+
 @register("storage_delete", SHORT)
 def storage_delete(name: str) -> bool:
     from skypilot_amd.data import storage as st
