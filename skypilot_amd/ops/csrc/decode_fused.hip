@@ -27,8 +27,45 @@ extern "C" __global__ void rmsnorm_res_kernel(
   const unsigned short* xr = x + row * H;
   const unsigned short* rr = res + row * H;
   float vals[32];  // H <= 8192, 256 threads
-  const int nchunk = H / 256;
   float ss = 0.f;
+  if (H % 2048 == 0) {
+    // vector path: b128 loads/stores (scalar bf16 costs 2-2.5x, G13)
+    const int nv = H / 2048;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      if (c >= nv) break;
+      int i = c * 2048 + threadIdx.x * 8;
+      s16x8 xv = *(const s16x8*)(xr + i);
+      s16x8 rv = *(const s16x8*)(rr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = bf2f((unsigned short)xv[j]) +
+                  bf2f((unsigned short)rv[j]);
+        vals[c * 8 + j] = v;
+        ss += v * v;
+      }
+    }
+    const float inv = rsqrtf(block_reduce_sum(ss, red) / H + eps);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      if (c >= nv) break;
+      int i = c * 2048 + threadIdx.x * 8;
+      s16x8 wv = *(const s16x8*)(w + i);
+      s16x8 xo, ho;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned short xb = f2bf(vals[c * 8 + j]);
+        xo[j] = (short)xb;
+        // match rmsnorm.hip: normalize the bf16-rounded sum
+        ho[j] = (short)f2bf(bf2f(xb) * inv *
+                            bf2f((unsigned short)wv[j]));
+      }
+      *(s16x8*)(x_out + row * H + i) = xo;
+      *(s16x8*)(h_out + row * H + i) = ho;
+    }
+    return;
+  }
+  const int nchunk = H / 256;
   // constant trip bound so `vals` fully promotes to registers
 #pragma unroll
   for (int c = 0; c < 32; ++c) {
@@ -45,7 +82,6 @@ extern "C" __global__ void rmsnorm_res_kernel(
     int i = c * 256 + threadIdx.x;
     unsigned short xv = f2bf(vals[c]);
     x_out[row * H + i] = xv;
-    // match rmsnorm.hip: normalize the bf16-rounded sum
     h_out[row * H + i] = f2bf(bf2f(xv) * inv * bf2f(w[i]));
   }
 }
