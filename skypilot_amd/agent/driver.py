@@ -74,9 +74,20 @@ def run_job(cluster_dir: str, job_id: int) -> int:
         return 0
 
     # -- gang launch --------------------------------------------------------
+    # Two dispatch modes per node rank:
+    #   local slice  — a process on this box (single-node pool, SSH head)
+    #   peer agent   — a leaf job POSTed to another node's agent over
+    #                  HTTP (multi-pod k8s gangs: every pod runs an
+    #                  agent; rank 0 is this pod, ranks 1.. dispatch to
+    #                  peer_agents[i-1] = "ip:port").  The gang env
+    #                  travels in the leaf spec's gang_env.
     node_ips = spec.get("node_ips") or ["127.0.0.1"] * num_nodes
-    master_port = _free_port()
+    peer_agents = spec.get("peer_agents") or []
+    master_addr = spec.get("master_addr") or "127.0.0.1"
+    gang_env_override = spec.get("gang_env")  # set on leaf jobs
+    master_port = int(spec.get("master_port") or _free_port())
     procs = []
+    remote = []  # (AgentClient, remote_job_id, node_rank)
     for node_rank in range(num_nodes):
         node_gpus = gpu_ids[node_rank * gpus_per_node:
                             (node_rank + 1) * gpus_per_node]
@@ -92,12 +103,29 @@ def run_job(cluster_dir: str, job_id: int) -> int:
             # Convenience for torchrun on the one-box pool: a unique
             # rendezvous port per job (reference leaves this to the user).
             "SKYPILOT_MASTER_PORT": str(master_port),
-            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_ADDR": master_addr,
             "MASTER_PORT": str(master_port),
             "HSA_ENABLE_IPC_MODE_LEGACY": "0",
         })
+        if gang_env_override:
+            env.update(gang_env_override)
         if spec.get("managed_job_id"):
             env["SKYPILOT_MANAGED_JOB_ID"] = str(spec["managed_job_id"])
+        if node_rank > 0 and node_rank <= len(peer_agents):
+            from skypilot_amd.agent.client import AgentClient
+            host, _, port = peer_agents[node_rank - 1].rpartition(":")
+            peer = AgentClient(int(port), host=host)
+            leaf_env = {k: v for k, v in env.items()
+                        if k.startswith(("SKYPILOT_", "MASTER_"))}
+            leaf_env.update(envs)
+            rjid = peer.queue_job({
+                "run": run_cmd, "setup": spec.get("setup"),
+                "envs": {}, "num_nodes": 1,
+                "gpus_per_node": gpus_per_node,
+                "gang_env": leaf_env, "task_id": task_id,
+            }, name=f"{task_id}-rank{node_rank}")
+            remote.append((peer, rjid, node_rank))
+            continue
         if node_gpus:
             env["HIP_VISIBLE_DEVICES"] = ",".join(str(g) for g in node_gpus)
             env["CUDA_VISIBLE_DEVICES"] = env["HIP_VISIBLE_DEVICES"]
@@ -127,13 +155,64 @@ def run_job(cluster_dir: str, job_id: int) -> int:
                 os.killpg(p.pid, signal.SIGTERM)
             except ProcessLookupError:
                 pass
+        for peer, rjid, _ in remote:
+            try:
+                peer.cancel_job(rjid)
+            except Exception:  # noqa: BLE001
+                pass
 
     signal.signal(signal.SIGTERM, forward_term)
 
+    def _pull_peer_log(peer, rjid, node_rank):
+        try:
+            with open(log_dir / f"{node_rank}-node.log", "ab") as lf:
+                for chunk in peer.tail_logs(rjid, follow=False):
+                    lf.write(chunk if isinstance(chunk, bytes)
+                             else chunk.encode())
+        except Exception:  # noqa: BLE001
+            pass
+
+    # Gang semantics (reference: a Ray placement-group task error fails
+    # the gang): first non-zero rank kills local slices and cancels
+    # peer leaf jobs instead of waiting out the survivors.
     rcs = []
     try:
-        for p in procs:
-            rcs.append(p.wait())
+        pending_p = list(procs)
+        pending_r = list(remote)
+        failing = False
+        while pending_p or pending_r:
+            progressed = False
+            for p in list(pending_p):
+                rc = p.poll()
+                if rc is not None:
+                    rcs.append(rc)
+                    pending_p.remove(p)
+                    progressed = True
+                    failing = failing or rc != 0
+            for item in list(pending_r):
+                peer, rjid, node_rank = item
+                j = peer.get_job(rjid)
+                if j and j["status"] in job_lib.TERMINAL:
+                    rc = (0 if j["status"] == job_lib.SUCCEEDED
+                          else (j.get("exit_code") or 1))
+                    rcs.append(rc)
+                    pending_r.remove(item)
+                    progressed = True
+                    failing = failing or rc != 0
+                    _pull_peer_log(peer, rjid, node_rank)
+            if failing and (pending_p or pending_r):
+                for p in pending_p:
+                    try:
+                        os.killpg(p.pid, signal.SIGTERM)
+                    except ProcessLookupError:
+                        pass
+                for peer, rjid, _ in pending_r:
+                    try:
+                        peer.cancel_job(rjid)
+                    except Exception:  # noqa: BLE001
+                        pass
+            if not progressed:
+                time.sleep(0.5)
     finally:
         for p in procs:
             if p.poll() is None:
@@ -142,6 +221,13 @@ def run_job(cluster_dir: str, job_id: int) -> int:
                 except ProcessLookupError:
                     pass
             p._logf.close()
+        for peer, rjid, _ in remote:
+            try:
+                j = peer.get_job(rjid)
+                if j and j["status"] not in job_lib.TERMINAL:
+                    peer.cancel_job(rjid)
+            except Exception:  # noqa: BLE001
+                pass
 
     worst = max((abs(r) for r in rcs), default=0)
     final = job_lib.SUCCEEDED if all(r == 0 for r in rcs) else job_lib.FAILED

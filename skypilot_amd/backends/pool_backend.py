@@ -86,6 +86,8 @@ class PoolBackend(Backend):
             "gpus_per_node": task.resources.accelerator_count,
             "workdir": str(Path(handle["cluster_dir"]) / "workdir"),
             "node_ips": handle.get("node_ips"),
+            "peer_agents": handle.get("peer_agents"),
+            "master_addr": handle.get("master_addr"),
             "task_id": task_id,
             "managed_job_id": managed_job_id,
             "event_callback": task.event_callback,

@@ -94,50 +94,67 @@ def _run_kubectl(args: List[str], settings, input_text: str = None,
                           timeout=timeout)
 
 
-def run_instances(cluster_name: str, num_nodes: int, accelerator,
-                  acc_count: int, existing_handle: Optional[Dict] = None
-                  ) -> Dict[str, Any]:
-    settings = k8s_settings()
-    if num_nodes != 1:
-        raise ResourcesUnavailableError(
-            "kubernetes pool: one pod per cluster (use num_nodes=1; "
-            "multi-pod gangs land in round 2)")
-    manifest = render_pod_manifest(cluster_name, acc_count, settings)
-    proc = _run_kubectl(["apply", "-f", "-"], settings,
-                        input_text=yaml.safe_dump(manifest))
-    if proc.returncode != 0:
-        raise ResourcesUnavailableError(
-            f"kubectl apply failed: {proc.stderr[:400]}")
-    # Wait for the pod to be Running.
-    pod = manifest["metadata"]["name"]
-    deadline = time.time() + 300
+def _pod_name(cluster_name: str, rank: int) -> str:
+    return (f"sky-amd-{cluster_name}" if rank == 0
+            else f"sky-amd-{cluster_name}-{rank}")
+
+
+def _wait_running(pod: str, settings, deadline: float) -> str:
+    """Wait for Running; returns the pod IP."""
     while time.time() < deadline:
         out = _run_kubectl(["get", "pod", pod, "-o", "json"], settings)
         if out.returncode == 0:
-            phase = json.loads(out.stdout).get("status", {}).get("phase")
-            if phase == "Running":
-                break
-            if phase in ("Failed", "Unknown"):
-                raise ResourcesUnavailableError(f"pod {pod} phase {phase}")
+            st = json.loads(out.stdout).get("status", {})
+            if st.get("phase") == "Running":
+                return st.get("podIP", "")
+            if st.get("phase") in ("Failed", "Unknown"):
+                raise ResourcesUnavailableError(
+                    f"pod {pod} phase {st.get('phase')}")
         time.sleep(2)
-    else:
-        raise ResourcesUnavailableError(f"pod {pod} never became Running")
+    raise ResourcesUnavailableError(f"pod {pod} never became Running")
 
-    local_port = _port_forward(pod, settings)
+
+def run_instances(cluster_name: str, num_nodes: int, accelerator,
+                  acc_count: int, existing_handle: Optional[Dict] = None
+                  ) -> Dict[str, Any]:
+    """Multi-pod gang: one agent pod per node.  Pod 0 is the head (the
+    control plane reaches its agent through a kubectl port-forward
+    tunnel); the head's job driver dispatches rank>0 leaf jobs to the
+    peer pods' agents at podIP:AGENT_PORT (agent/driver.py peer_agents
+    path) — no Ray, no ssh, just the agent HTTP surface
+    (reference: multi-node pods via Ray in
+    sky/provision/kubernetes/instance.py)."""
+    settings = k8s_settings()
+    pods = [_pod_name(cluster_name, i) for i in range(num_nodes)]
+    for pod in pods:
+        manifest = render_pod_manifest(cluster_name, acc_count, settings)
+        manifest["metadata"]["name"] = pod
+        proc = _run_kubectl(["apply", "-f", "-"], settings,
+                            input_text=yaml.safe_dump(manifest))
+        if proc.returncode != 0:
+            raise ResourcesUnavailableError(
+                f"kubectl apply failed: {proc.stderr[:400]}")
+    deadline = time.time() + 300
+    ips = [_wait_running(pod, settings, deadline) for pod in pods]
+
+    local_port = _port_forward(pods[0], settings)
     AgentClient(local_port).wait_ready(timeout=60)
     cdir = global_state.root_dir() / "clusters" / cluster_name
     cdir.mkdir(parents=True, exist_ok=True)
     return {
         "cloud": CLOUD_NAME,
         "cluster_dir": str(cdir),
-        "pod": pod,
+        "pod": pods[0],
+        "pods": pods,
         "namespace": settings.get("namespace", "default"),
-        "gpu_ids": list(range(acc_count)),
-        "num_nodes": 1,
+        "gpu_ids": list(range(acc_count * num_nodes)),
+        "num_nodes": num_nodes,
         "gpus_per_node": acc_count,
-        "head_ip": "127.0.0.1",
-        "node_ips": ["127.0.0.1"],
+        "head_ip": ips[0] or "127.0.0.1",
+        "node_ips": ips,
         "agent_port": local_port,
+        "master_addr": ips[0] or "127.0.0.1",
+        "peer_agents": [f"{ip}:{AGENT_PORT}" for ip in ips[1:]],
     }
 
 
@@ -166,8 +183,8 @@ def stop_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
 def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
     stop_instances(cluster_name, handle)
     settings = k8s_settings()
-    pod = handle.get("pod")
-    if pod:
+    for pod in handle.get("pods") or ([handle["pod"]]
+                                      if handle.get("pod") else []):
         try:
             _run_kubectl(["delete", "pod", pod, "--wait=false"], settings)
         except (OSError, subprocess.TimeoutExpired):
