@@ -41,7 +41,8 @@ class PoolBackend(Backend):
         t0 = time.time()
         handle = provision.run_instances(
             cloud, cluster_name, task.num_nodes, res.accelerators,
-            res.accelerator_count, existing_handle)
+            res.accelerator_count, existing_handle,
+            use_spot=res.use_spot)
         global_state.add_or_update_cluster(
             cluster_name, global_state.UP, handle, res.to_yaml_config())
         global_state.add_cluster_event(

@@ -115,8 +115,8 @@ def _wait_running(pod: str, settings, deadline: float) -> str:
 
 
 def run_instances(cluster_name: str, num_nodes: int, accelerator,
-                  acc_count: int, existing_handle: Optional[Dict] = None
-                  ) -> Dict[str, Any]:
+                  acc_count: int, existing_handle: Optional[Dict] = None,
+                  use_spot: bool = False) -> Dict[str, Any]:
     """Multi-pod gang: one agent pod per node.  Pod 0 is the head (the
     control plane reaches its agent through a kubectl port-forward
     tunnel); the head's job driver dispatches rank>0 leaf jobs to the

@@ -50,8 +50,8 @@ def _free_port() -> int:
 
 
 def run_instances(cluster_name: str, num_nodes: int, accelerator: str | None,
-                  acc_count: int, existing_handle: Optional[Dict] = None
-                  ) -> Dict[str, Any]:
+                  acc_count: int, existing_handle: Optional[Dict] = None,
+                  use_spot: bool = False) -> Dict[str, Any]:
     """Allocate GPUs + start the agent. Returns the cluster handle.
     Guarded by a per-cluster file lock (reference: utils/locks.py used at
     cloud_vm_ray_backend.py:3433) so concurrent launches can't
@@ -63,14 +63,44 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator: str | None,
     fcntl.flock(lock_f, fcntl.LOCK_EX)
     try:
         return _run_instances_locked(cluster_name, num_nodes, accelerator,
-                                     acc_count, existing_handle)
+                                     acc_count, existing_handle, use_spot)
     finally:
         fcntl.flock(lock_f, fcntl.LOCK_UN)
         lock_f.close()
 
 
+def _spot_victims(requester: str) -> list:
+    """Live spot clusters, newest first (cheapest to preempt by
+    seniority: reference spot semantics kill the market's choice; on a
+    pool we take the most recently launched)."""
+    out = []
+    for c in global_state.list_clusters():
+        if c["name"] == requester or c["status"] != global_state.UP:
+            continue
+        h = c["handle"]
+        if h.get("cloud") == CLOUD_NAME and h.get("use_spot")                 and h.get("gpu_ids"):
+            out.append(c)
+    out.sort(key=lambda c: c.get("launched_at") or 0, reverse=True)
+    return out
+
+
+def _preempt(victim: Dict) -> None:
+    """Kill a spot cluster like a revoked spot instance: agent + record
+    gone (the managed-jobs monitor then sees the cluster missing and
+    enters RECOVERING; reference: jobs/controller.py preemption path)."""
+    name = victim["name"]
+    global_state.add_cluster_event(name, "PREEMPTED",
+                                   "spot capacity reclaimed")
+    try:
+        terminate_instances(name, victim["handle"])
+    except Exception:  # noqa: BLE001
+        pass
+    global_state.remove_cluster(name)
+
+
 def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
-                          existing_handle=None) -> Dict[str, Any]:
+                          existing_handle=None, use_spot=False
+                          ) -> Dict[str, Any]:
     gpus = detect_gpus()
     need = num_nodes * acc_count
     if existing_handle and existing_handle.get("gpu_ids") is not None:
@@ -78,6 +108,15 @@ def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
     elif need > 0:
         taken = set(_allocated_gpus_elsewhere(cluster_name))
         free = [g.index for g in gpus if g.index not in taken]
+        if len(free) < need and not use_spot:
+            # On-demand requests reclaim spot capacity (reference:
+            # spot preemption; spot requests never preempt anyone).
+            for victim in _spot_victims(cluster_name):
+                _preempt(victim)
+                taken = set(_allocated_gpus_elsewhere(cluster_name))
+                free = [g.index for g in gpus if g.index not in taken]
+                if len(free) >= need:
+                    break
         if len(free) < need:
             raise ResourcesUnavailableError(
                 f"need {need}x{accelerator or 'GPU'}, pool has "
@@ -94,6 +133,7 @@ def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
         "gpu_ids": gpu_ids,
         "num_nodes": num_nodes,
         "gpus_per_node": acc_count,
+        "use_spot": bool(use_spot),
         "head_ip": "127.0.0.1",
         "node_ips": ["127.0.0.1"] * num_nodes,
     }

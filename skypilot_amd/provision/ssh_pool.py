@@ -79,8 +79,8 @@ def _hosts_in_use(except_cluster: str) -> List[str]:
 
 
 def run_instances(cluster_name: str, num_nodes: int, accelerator,
-                  acc_count: int, existing_handle: Optional[Dict] = None
-                  ) -> Dict[str, Any]:
+                  acc_count: int, existing_handle: Optional[Dict] = None,
+                  use_spot: bool = False) -> Dict[str, Any]:
     hosts = parse_hosts()
     if not hosts:
         raise ResourcesUnavailableError(

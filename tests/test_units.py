@@ -148,3 +148,43 @@ def test_provision_time_failover(tmp_path, monkeypatch):
     assert "PROVISION_FAILOVER" in ev
     from skypilot_amd import core
     core.down("failover-d")
+
+
+def test_spot_preemption(tmp_path, monkeypatch):
+    """On-demand launches reclaim spot GPUs; spot never preempts
+    (reference: spot instance revocation -> managed-job RECOVERING)."""
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "8")
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    from skypilot_amd import core, execution, global_state
+    from skypilot_amd.exceptions import ResourcesUnavailableError
+    from skypilot_amd.task import Task
+
+    spot = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"accelerators": "MI355X:6", "use_spot": True}})
+    execution.launch(spot, "spot-c", detach_run=True)
+    assert global_state.get_cluster("spot-c")["handle"]["use_spot"]
+
+    # a second spot task must NOT preempt the first
+    spot2 = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"accelerators": "MI355X:4", "use_spot": True}})
+    import pytest as _pytest
+    with _pytest.raises(ResourcesUnavailableError):
+        execution.launch(spot2, "spot-d", detach_run=True)
+    assert global_state.get_cluster("spot-c") is not None
+
+    # an on-demand task reclaims the spot capacity
+    od = Task.from_yaml_config({
+        "run": "true", "resources": {"accelerators": "MI355X:4"}})
+    execution.launch(od, "od-c", detach_run=True)
+    assert global_state.get_cluster("spot-c") is None  # preempted
+    ev = [e["event"] for e in global_state.get_cluster_events("spot-c")]
+    assert "PREEMPTED" in ev
+    rec = global_state.get_cluster("od-c")
+    assert rec["status"] == "UP"
+    assert len(rec["handle"]["gpu_ids"]) == 4
+    core.down("od-c")
+    gpu_topology.detect_gpus.cache_clear()
