@@ -154,3 +154,32 @@ def test_jobs_pool(client, tmp_path):
     sdk.get(sdk.jobs_pool_down("p1"))
     records = sdk.get(sdk.status())
     assert not any(r["name"].startswith("sky-pool-p1") for r in records)
+
+
+def test_managed_spot_job_recovers_after_preemption(client, tmp_path):
+    """Spot end-to-end: a use_spot managed job loses its GPUs to an
+    on-demand launch (true capacity preemption, not a manual kill),
+    enters RECOVERING, and finishes once capacity frees up."""
+    from skypilot_amd.client import sdk
+    ckpt = tmp_path / "spot-ckpt"
+    task = {
+        "name": "mj-spot",
+        "resources": {"accelerators": "MI355X:6", "use_spot": True},
+        "file_mounts": {str(ckpt): {"name": "mj-spot-ckpt",
+                                    "mode": "MOUNT"}},
+        "run": f"if [ -f {ckpt}/state ]; then echo resumed; "
+               f"else echo phase1 > {ckpt}/state; sleep 600; fi",
+    }
+    res = sdk.get(sdk.jobs_launch(task, "mj-spot"))
+    job_id = res["job_id"]
+    _wait_managed(job_id, {"RUNNING"})
+    time.sleep(2)
+    # On-demand launch takes the GPUs -> the spot cluster is reclaimed.
+    sdk.get(sdk.launch({"run": "true",
+                        "resources": {"accelerators": "MI355X:4"}},
+                       "od-grab"), timeout=60)
+    # Spot job must notice and recover; capacity frees when od-grab downs.
+    sdk.get(sdk.down("od-grab"))
+    job = _wait_managed(job_id, {"SUCCEEDED", "FAILED"}, timeout=180)
+    assert job["status"] == "SUCCEEDED", job
+    assert job["recovery_count"] >= 1
