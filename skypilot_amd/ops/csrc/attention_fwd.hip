@@ -10,6 +10,7 @@
 // for the backward pass.
 //
 // Layouts: Q,O [B,S,Hq,D]; K,V [B,S,Hkv,D]; lse [B,Hq,S] fp32.
+#include <cstdlib>
 #include "common.h"
 
 #define ATT_D 128
@@ -247,9 +248,6 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
   __shared__ unsigned short k_lds[BN * ATT_D];     // [kv][d]   16KB
   __shared__ unsigned short vt_lds[ATT_D * BN];    // [d][kv]   16KB
   __shared__ unsigned short p_lds[BM32 * BN];      // [q][kv]   16KB
-  // Online-softmax running max/sum per q row live in LDS: keeping 2x16
-  // floats per lane in VGPRs spilled to scratch (all 32 lanes of a row
-  // write identical post-reduce values, so the shared write is benign).
   __shared__ float m_lds[BM32], l_lds[BM32];
 
   const int qt = gridDim.x - 1 - blockIdx.x;  // heavy blocks first
@@ -362,7 +360,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
       float m_new = fmaxf(m_old, mx);
       float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
       float corr = (m_old == -INFINITY) ? 0.f : __expf(m_old - m_safe);
-      m_lds[prow] = m_new;
+      if (col == 0) m_lds[prow] = m_new;
       float rs = 0.f;
 #pragma unroll
       for (int nt = 0; nt < 2; ++nt) {
@@ -375,11 +373,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
       }
 #pragma unroll
       for (int off = 16; off > 0; off >>= 1) rs += __shfl_xor(rs, off, 64);
-      l_lds[prow] = l_lds[prow] * corr + rs;
+      if (col == 0) l_lds[prow] = l_lds[prow] * corr + rs;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) o_acc[nt][r] *= corr;
     }
-    __syncthreads();
+    // (no barrier: p_lds rows 32w..32w+31 are wave-private, and next
+    // tile's k/v staging is ordered by the loop-top barrier)
 
     // O += P V : A-frags from p_lds (4, reused), B from vt_lds.
     s16x8 pa[4];
@@ -424,13 +423,24 @@ extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
                                 void* O, float* lse, int B, int S, int Hq,
                                 int Hkv, float scale, bool causal,
                                 hipStream_t stream) {
-  // NOTE: attn_fwd_kernel32 (32x32x16 MFMA, 128-row blocks) measured
-  // SLOWER than this 16x16 kernel (97 vs 207 TF/s at the bench shape)
-  // despite 2x the arithmetic intensity — the 16-row serialized softmax
-  // and deeper accumulator chains dominate at 2 waves/SIMD.  Kept
-  // compiled (and layout-verified by tests) as the starting point for a
-  // wave-specialized rewrite; dispatch stays on the 16x16 kernel.
-  if (false && S % BM32 == 0) {
+  // NOTE: attn_fwd_kernel32 (32x32x16 MFMA, 128-row blocks) measures
+  // SLOWER than this 16x16 kernel (94-97 vs 207 TF/s at the bench
+  // shape) despite 2x the arithmetic intensity.  Measured eliminations
+  // (round 1): removing the post-softmax barrier and single-writer LDS
+  // m/l: null.  m/l in registers forces 1 wave/SIMD (VGPR cap): 63
+  // TF/s.  So neither barriers nor softmax-state round-trips are the
+  // bottleneck; the remaining suspects are the scalar V^T/P LDS
+  // traffic and plain wave-level MFMA/VALU interleaving, which only
+  // the full swapped-QK^T schedule (S^T = K Q^T, in-register column
+  // softmax via permlane32_swap — semantics verified in
+  // profiles/r01_hw_probe_semantics.txt) restructures away.  Dispatch
+  // stays on the 16x16 kernel; SKY_ATTN_FWD_32=1 flips it for
+  // experiments.
+  static const int use32 = [] {
+    const char* e = getenv("SKY_ATTN_FWD_32");
+    return e ? atoi(e) : 0;
+  }();
+  if (use32 && S % BM32 == 0) {
     dim3 grid(S / BM32, B * Hq);
     hipLaunchKernelGGL(attn_fwd_kernel32, grid, dim3(256), 0, stream,
                        (const unsigned short*)Q, (const unsigned short*)K,
