@@ -9,6 +9,7 @@ forward through the decode-attention kernel.  max_batch is sized to the
 """
 from __future__ import annotations
 
+import os
 import queue
 import threading
 import time
@@ -156,7 +157,10 @@ class Engine:
         self._maybe_finish(slot, req, next_id)
 
     # -------------------------- hipGraph decode ------------------------
-    CHUNK = 8  # max chained graph replays between host syncs
+    # Max chained graph replays between host syncs (SKY_DECODE_CHUNK to
+    # tune; measured flat 8..32 at 1 stream, so 8 keeps token-streaming
+    # latency low).
+    CHUNK = int(os.environ.get("SKY_DECODE_CHUNK", "8"))
 
     def _bucket(self, n: int) -> int:
         b = 1
