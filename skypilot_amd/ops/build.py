@@ -79,6 +79,18 @@ def build(force: bool = False, verbose: bool = True) -> Path:
         "-fPIC",
         f"-D_GLIBCXX_USE_CXX11_ABI={tp['cxx11_abi']}",
     ]
+    # SKY_AMD_SANITIZE=1: host AddressSanitizer on the binding/runtime
+    # code (SURVEY.md §5.2 — the reference is pure Python and needs
+    # none; the native layer here does).  Device code stays unsanitized
+    # (use SKY_AMD_SANITIZE=xnack + HSA_XNACK=1 on a box that supports
+    # it for amdgpu ASAN).
+    san = os.environ.get("SKY_AMD_SANITIZE")
+    if san == "1":
+        common_flags += ["-fsanitize=address", "-shared-libasan",
+                         "-g", "-fno-omit-frame-pointer"]
+    elif san == "xnack":
+        common_flags += ["-fsanitize=address", "-shared-libasan", "-g",
+                         f"--offload-arch={ARCH}:xnack+"]
     objs = []
     # Device TUs: pure HIP, no torch headers.
     for src in HIP_SOURCES:
@@ -104,7 +116,10 @@ def build(force: bool = False, verbose: bool = True) -> Path:
              + common_flags + torch_flags)
         objs.append(str(obj))
 
-    link = [hipcc, "-shared", "-fPIC", "-o", str(SO_PATH)] + objs + [
+    link = [hipcc, "-shared", "-fPIC", "-o", str(SO_PATH)] + objs
+    if os.environ.get("SKY_AMD_SANITIZE"):
+        link += ["-fsanitize=address", "-shared-libasan"]
+    link += [
         f"-L{tp['lib_dir']}",
         "-ltorch", "-ltorch_cpu", "-ltorch_hip", "-lc10", "-lc10_hip",
         "-ltorch_python",

@@ -192,6 +192,18 @@ def create_app(start_workers: bool = True) -> FastAPI:
         from skypilot_amd import global_state as _gs
         ups = sum(1 for c in _gs.list_clusters() if c["status"] == "UP")
         lines.append(f"sky_amd_clusters_up {ups}")
+        # executor free-slot gauges (reference: executor.py:369-404)
+        from skypilot_amd.server import daemons as _d
+        from skypilot_amd.server import executor as _ex
+        for q, free in _ex.free_slots().items():
+            lines.append(
+                f'sky_amd_executor_free_slots{{queue="{q.lower()}"}} '
+                f"{free}")
+        lines.append(
+            f"sky_amd_loop_stall_max_seconds "
+            f"{_d.loop_stall_max_seconds():.4f}")
+        lines.append(
+            f"sky_amd_daemons_leader {1 if _d.is_leader() else 0}")
         return PlainTextResponse("\n".join(lines) + "\n")
 
     @app.get("/api/requests")

@@ -141,6 +141,16 @@ class RequestWorker(threading.Thread):
 _workers: list = []
 
 
+def free_slots() -> Dict[str, int]:
+    """Per-queue free worker slots (Prometheus gauge; reference:
+    executor.py:369-404 free-slot accounting)."""
+    out: Dict[str, int] = {}
+    for w in _workers:
+        busy = len(w._children) + len(w._inline)
+        out[w.queue] = max(0, w.parallelism - busy)
+    return out
+
+
 def start_workers(long_parallelism: int = 4, short_parallelism: int = 16):
     """reference: executor.py:1317 (start) — one LONG + one SHORT worker."""
     global _workers

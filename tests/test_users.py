@@ -133,3 +133,34 @@ def test_cost_report(sky_env, client):
     hist = [r for r in rows if r["name"] == "cost-c" and not r["live"]]
     assert hist and hist[0]["status"] == "TERMINATED"
     assert hist[0]["duration_hours"] >= 0
+
+
+def test_metrics_gauges_and_leader(sky_env, client):
+    """Executor free-slot gauges, loop-stall metric and leader flag on
+    /metrics; leader election is exclusive across 'processes'."""
+    body = client.get("/metrics").text
+    assert 'sky_amd_executor_free_slots{queue="long"}' in body
+    assert 'sky_amd_executor_free_slots{queue="short"}' in body
+    assert "sky_amd_loop_stall_max_seconds" in body
+    assert "sky_amd_daemons_leader 1" in body
+    from skypilot_amd.server import daemons
+    assert daemons.is_leader()
+    assert daemons.try_acquire_leadership()  # idempotent for the holder
+    # a second server process must NOT win the election while this one
+    # holds the flock
+    import subprocess
+    import sys
+    code = "\n".join([
+        "import fcntl, sys",
+        "f = open(sys.argv[1] + '/daemons-leader.lock', 'w')",
+        "try:",
+        "    fcntl.flock(f, fcntl.LOCK_EX | fcntl.LOCK_NB)",
+        "    print('WON')",
+        "except OSError:",
+        "    print('LOST')",
+    ])
+    # point the child at the lock dir the current leader actually used
+    actual = daemons._leader_lock_file.name.rsplit("/", 1)[0]
+    out = subprocess.run([sys.executable, "-c", code, actual],
+                         capture_output=True, text=True)
+    assert out.stdout.strip() == "LOST", out
