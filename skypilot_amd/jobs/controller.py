@@ -127,16 +127,24 @@ class JobController:
             if outcome == "failed_setup":
                 return state.FAILED_SETUP
             if outcome == "failed_user":
-                # User-code failure: bounded restarts
-                # (resources.job_recovery.max_restarts_on_errors).
-                if not self.strategy.should_restart_on_failure():
-                    return state.FAILED
+                # recover_on_exit_codes: these codes always recover and
+                # do not consume the restart budget (reference:
+                # job_recovery.recover_on_exit_codes).
+                jr = self.task.resources.job_recovery
+                codes = jr.recover_on_exit_codes if jr else ()
+                if self._last_exit_code not in codes:
+                    # Other user-code failure: bounded restarts
+                    # (resources.job_recovery.max_restarts_on_errors).
+                    if not self.strategy.should_restart_on_failure():
+                        return state.FAILED
             # preemption / infra failure / bounded user-failure restart:
             state.set_status(self.job_id, state.RECOVERING)
             state.bump_recovery(self.job_id)
             if not self.strategy.keep_placement_first():
                 self._teardown_cluster()
             self.strategy.wait_before_retry()
+
+    _last_exit_code = None
 
     def _monitor(self, cluster_job_id: int, handle) -> str:
         """Poll the cluster job; classify its end state
@@ -167,6 +175,7 @@ class JobController:
                 # treat as preemption and recover.
                 return "preempted"
             if st in (job_lib.FAILED, job_lib.FAILED_DRIVER):
+                self._last_exit_code = job.get("exit_code")
                 return "failed_user" if st == job_lib.FAILED else "preempted"
 
 

@@ -50,6 +50,9 @@ class JobRecovery:
     """reference: sky/utils/schemas.py job_recovery subschema."""
     strategy: str = "FAILOVER"
     max_restarts_on_errors: int = 0
+    # Exit codes that always recover (not billed to the restart budget)
+    # — reference: job_recovery.recover_on_exit_codes.
+    recover_on_exit_codes: tuple = ()
 
     @classmethod
     def from_yaml_config(cls, cfg) -> Optional["JobRecovery"]:
@@ -60,7 +63,9 @@ class JobRecovery:
         if isinstance(cfg, dict):
             return cls(
                 strategy=str(cfg.get("strategy", "FAILOVER") or "FAILOVER").upper(),
-                max_restarts_on_errors=int(cfg.get("max_restarts_on_errors", 0)))
+                max_restarts_on_errors=int(cfg.get("max_restarts_on_errors", 0)),
+                recover_on_exit_codes=tuple(
+                    int(c) for c in cfg.get("recover_on_exit_codes") or ()))
         raise TaskValidationError(f"bad job_recovery: {cfg!r}")
 
 

@@ -38,7 +38,7 @@ class RequestRateAutoscaler(Autoscaler):
     def _raw_target(self, qps: float) -> int:
         p = self.policy
         want = math.ceil(qps / p.target_qps_per_replica) if qps > 0 else 0
-        want = max(want, p.min_replicas)
+        want = max(want, p.min_replicas) + p.num_overprovision
         if p.max_replicas is not None:
             want = min(want, p.max_replicas)
         return want

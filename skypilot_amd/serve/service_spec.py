@@ -38,6 +38,7 @@ class ReplicaPolicy:
     min_replicas: int = 1
     max_replicas: Optional[int] = None
     target_qps_per_replica: Optional[float] = None
+    num_overprovision: int = 0  # extra replicas above the computed target
     upscale_delay_seconds: int = 300
     downscale_delay_seconds: int = 1200
 
@@ -51,6 +52,7 @@ class ReplicaPolicy:
             min_replicas=mn,
             max_replicas=int(mx) if mx is not None else None,
             target_qps_per_replica=cfg.get("target_qps_per_replica"),
+            num_overprovision=int(cfg.get("num_overprovision", 0)),
             upscale_delay_seconds=int(cfg.get("upscale_delay_seconds", 300)),
             downscale_delay_seconds=int(
                 cfg.get("downscale_delay_seconds", 1200)),

@@ -151,11 +151,19 @@ def free_slots() -> Dict[str, int]:
     return out
 
 
-def start_workers(long_parallelism: int = 4, short_parallelism: int = 16):
-    """reference: executor.py:1317 (start) — one LONG + one SHORT worker."""
+def start_workers(long_parallelism: int = None,
+                  short_parallelism: int = None):
+    """reference: executor.py:1317 (start) + server/config.py:89
+    (compute_server_config): LONG workers scale with CPU (x2, capped at
+    4 for local deployments), SHORT workers keep ample idle slots."""
     global _workers
     if _workers:
         return
+    ncpu = os.cpu_count() or 2
+    if long_parallelism is None:
+        long_parallelism = max(1, min(4, ncpu * 2))
+    if short_parallelism is None:
+        short_parallelism = max(8, min(32, ncpu * 4))
     for q, par in ((LONG, long_parallelism), (SHORT, short_parallelism)):
         w = RequestWorker(q, par)
         w.start()

@@ -183,3 +183,28 @@ def test_managed_spot_job_recovers_after_preemption(client, tmp_path):
     job = _wait_managed(job_id, {"SUCCEEDED", "FAILED"}, timeout=180)
     assert job["status"] == "SUCCEEDED", job
     assert job["recovery_count"] >= 1
+
+
+def test_recover_on_exit_codes(client):
+    """Exit codes listed in job_recovery.recover_on_exit_codes recover
+    without consuming the restart budget (reference: schema field)."""
+    from skypilot_amd.client import sdk
+    # exit 42 twice (recover_on_exit_codes, no budget needed), then 0.
+    marker = "/tmp/sky-roec-count"
+    import os as _os
+    if _os.path.exists(marker):
+        _os.unlink(marker)
+    task = {
+        "name": "mj-roec",
+        "resources": {"job_recovery": {"strategy": "EAGER",
+                                       "recover_on_exit_codes": [42],
+                                       "max_restarts_on_errors": 0}},
+        "run": f"n=$(cat {marker} 2>/dev/null || echo 0); "
+               f"echo $((n+1)) > {marker}; "
+               f"if [ $n -lt 2 ]; then exit 42; fi",
+    }
+    res = sdk.get(sdk.jobs_launch(task, "mj-roec"))
+    job = _wait_managed(res["job_id"], {"SUCCEEDED", "FAILED"},
+                        timeout=180)
+    assert job["status"] == "SUCCEEDED", job
+    assert job["recovery_count"] >= 2
