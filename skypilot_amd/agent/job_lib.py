@@ -60,8 +60,8 @@ class JobTable:
         def cm():
             conn = sqlite3.connect(self.db_path, timeout=30)
             try:
-                conn.execute("PRAGMA journal_mode=WAL")
                 conn.execute("PRAGMA busy_timeout=30000")
+                conn.execute("PRAGMA journal_mode=WAL")
                 conn.executescript(_SCHEMA)
                 with conn:
                     yield conn

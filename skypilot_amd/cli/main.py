@@ -273,6 +273,35 @@ def jobs_pool():
     """Worker pools for managed jobs."""
 
 
+@jobs.group("group")
+def jobs_group():
+    """Job groups: co-located concurrent tasks (one shared cluster)."""
+
+
+@jobs_group.command("launch")
+@click.argument("entrypoint")
+@click.option("--name", "-n", required=True)
+def jobs_group_launch_cmd(entrypoint, name):
+    """Launch a job group from a YAML with a `tasks:` list."""
+    import yaml as _yaml
+    with open(entrypoint) as f:
+        cfg = _yaml.safe_load(f) or {}
+    tasks = cfg.get("tasks") or []
+    _print_result(sdk.get(sdk.jobs_group_launch(name, tasks)))
+
+
+@jobs_group.command("status")
+@click.argument("name")
+def jobs_group_status_cmd(name):
+    _print_result(sdk.get(sdk.jobs_group_status(name)))
+
+
+@jobs_group.command("down")
+@click.argument("name")
+def jobs_group_down_cmd(name):
+    _print_result(sdk.get(sdk.jobs_group_down(name)))
+
+
 @jobs_pool.command("apply")
 @click.argument("entrypoint")
 @click.option("--pool", "-p", "name", required=True)

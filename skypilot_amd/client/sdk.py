@@ -246,6 +246,18 @@ def job_status(cluster_name: str, job_id: int) -> str:
                                   "job_id": job_id})
 
 
+def jobs_group_launch(name: str, tasks):
+    return _submit("jobs_group_launch", {"name": name, "tasks": tasks})
+
+
+def jobs_group_status(name: str):
+    return _submit("jobs_group_status", {"name": name})
+
+
+def jobs_group_down(name: str):
+    return _submit("jobs_group_down", {"name": name})
+
+
 def storage_sync(name: str):
     return _submit("storage_sync", {"name": name})
 

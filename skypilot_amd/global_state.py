@@ -85,8 +85,8 @@ import contextlib
 def _conn():
     conn = sqlite3.connect(_db_path(), timeout=30)
     try:
-        conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
+        conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
         with conn:
             yield conn

@@ -41,8 +41,8 @@ CREATE TABLE IF NOT EXISTS pool_workers (
 def _conn():
     conn = sqlite3.connect(global_state.root_dir() / "pools.db", timeout=30)
     try:
-        conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
+        conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
         with conn:
             yield conn

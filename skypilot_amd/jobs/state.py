@@ -47,8 +47,8 @@ def _conn():
     conn = sqlite3.connect(global_state.root_dir() / "managed_jobs.db",
                            timeout=30)
     try:
-        conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
+        conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
         with conn:
             yield conn

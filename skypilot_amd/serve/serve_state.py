@@ -54,8 +54,8 @@ CREATE TABLE IF NOT EXISTS replicas (
 def _conn():
     conn = sqlite3.connect(global_state.root_dir() / "serve.db", timeout=30)
     try:
-        conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
+        conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
         cols = [r[1] for r in conn.execute("PRAGMA table_info(services)")]
         if "tls" not in cols:

@@ -169,6 +169,26 @@ def jobs_logs(job_id: int):
     return jobs_server.logs(job_id)
 
 
+@register("jobs_group_launch", LONG)
+def jobs_group_launch(name: str, tasks: List[Dict[str, Any]]
+                      ) -> Dict[str, Any]:
+    """reference: JobGroup (jobs/job_group_networking.py)."""
+    from skypilot_amd.jobs import groups
+    return groups.launch(name, tasks)
+
+
+@register("jobs_group_status", SHORT)
+def jobs_group_status(name: str) -> Dict[str, Any]:
+    from skypilot_amd.jobs import groups
+    return groups.status(name)
+
+
+@register("jobs_group_down", LONG)
+def jobs_group_down(name: str) -> Dict[str, Any]:
+    from skypilot_amd.jobs import groups
+    return groups.down(name)
+
+
 @register("jobs_pool_apply", LONG)
 def jobs_pool_apply(name: str, template: Dict[str, Any],
                     num_workers: int = 2):

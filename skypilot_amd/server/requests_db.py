@@ -50,8 +50,8 @@ import contextlib
 def _conn():
     conn = sqlite3.connect(api_dir() / "requests.db", timeout=30)
     try:
-        conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
+        conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
         cols = [r[1] for r in conn.execute("PRAGMA table_info(requests)")]
         if "user" not in cols:  # pre-RBAC databases

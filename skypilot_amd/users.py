@@ -32,8 +32,8 @@ ROLES = ("admin", "user", "viewer")
 READONLY_REQUESTS = {
     "status", "queue", "job_status", "check", "show_gpus",
     "cluster_events", "storage_list", "volumes_list", "recipes_list",
-    "jobs_queue", "jobs_logs", "jobs_pool_status", "serve_status",
-    "serve_logs", "cost_report",
+    "jobs_queue", "jobs_logs", "jobs_pool_status", "jobs_group_status",
+    "serve_status", "serve_logs", "cost_report",
 }
 
 _SCHEMA = """
@@ -53,8 +53,8 @@ def _conn():
     conn = sqlite3.connect(global_state.root_dir() / "users.db",
                            timeout=30)
     try:
-        conn.execute("PRAGMA journal_mode=WAL")
         conn.execute("PRAGMA busy_timeout=30000")
+        conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
         with conn:
             yield conn

@@ -208,3 +208,40 @@ def test_recover_on_exit_codes(client):
                         timeout=180)
     assert job["status"] == "SUCCEEDED", job
     assert job["recovery_count"] >= 2
+
+
+def test_job_group_concurrent_colocated(client, tmp_path):
+    """JobGroup (reference: jobs/job_group_networking.py): members run
+    CONCURRENTLY on one shared cluster with group addressing env."""
+    from skypilot_amd.client import sdk
+    d = tmp_path / "grp"
+    d.mkdir()
+    # Each member writes its group env, then waits for the OTHER's file
+    # — only concurrent execution on one cluster lets both finish.
+    def run(me, other):
+        return (f"env | grep SKYPILOT_JOBGROUP > {d}/{me}.env; "
+                f"for i in $(seq 1 100); do "
+                f"[ -f {d}/{other}.env ] && exit 0; sleep 0.2; done; "
+                f"exit 1")
+    res = sdk.get(sdk.jobs_group_launch("g1", [
+        {"name": "ps", "run": run("ps", "worker"),
+         "resources": {"accelerators": "MI355X:2"}},
+        {"name": "worker", "run": run("worker", "ps"),
+         "resources": {"accelerators": "MI355X:2"}},
+    ]), timeout=120)
+    assert res["cluster"] == "sky-group-g1"
+    assert len(res["members"]) == 2
+    deadline = time.time() + 90
+    while time.time() < deadline:
+        st = sdk.get(sdk.jobs_group_status("g1"))
+        if all(m["status"] in ("SUCCEEDED", "FAILED", "CANCELLED")
+               for m in st["members"]):
+            break
+        time.sleep(1)
+    assert all(m["status"] == "SUCCEEDED" for m in st["members"]), st
+    ps_env = (d / "ps.env").read_text()
+    assert "SKYPILOT_JOBGROUP_NAME=g1" in ps_env
+    assert "SKYPILOT_JOBGROUP_TASK=ps" in ps_env
+    assert "SKYPILOT_JOBGROUP_HOST=127.0.0.1" in ps_env
+    assert "SKYPILOT_JOBGROUP_TASKS=ps,worker" in ps_env
+    sdk.get(sdk.jobs_group_down("g1"))
