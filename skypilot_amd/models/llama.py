@@ -115,10 +115,17 @@ class Attention(nn.Module):
         if infer_ctx is None:
             o = ops.attention(q, k, v, self.scale, causal=True)
         else:  # prefill: plain causal attention over the prompt; the
-            # (unpadded) K/V rows land in the cache for decode.
-            infer_ctx.cache.write_prefill(self.layer_idx,
-                                          infer_ctx.prefill_slot, k, v,
-                                          infer_ctx.prefill_len)
+            # (unpadded) K/V rows land in the cache for decode.  Rows of
+            # a batched prefill are independent sequences.
+            slots = (infer_ctx.prefill_slots
+                     if infer_ctx.prefill_slots is not None
+                     else [infer_ctx.prefill_slot])
+            lens = (infer_ctx.prefill_lens
+                    if infer_ctx.prefill_lens is not None
+                    else [infer_ctx.prefill_len])
+            for i, (slot, ln) in enumerate(zip(slots, lens)):
+                infer_ctx.cache.write_prefill(self.layer_idx, slot,
+                                              k[i:i + 1], v[i:i + 1], ln)
             o = ops.attention(q, k, v, self.scale, causal=True)
         return lin(o.reshape(B, S, self.n_q * d), self.wo.weight)
 
