@@ -55,16 +55,18 @@ def cli():
 @click.option("--env", multiple=True, help="KEY=VALUE")
 @click.option("--down", is_flag=True, help="tear down after the job")
 @click.option("--idle-minutes-to-autostop", "-i", type=int, default=None)
+@click.option("--retry-until-up", "-r", is_flag=True,
+              help="retry provisioning with backoff until capacity frees")
 @click.option("--detach-run", "-d", is_flag=True,
               help="don't stream logs after submit")
 @click.option("--async", "async_", is_flag=True,
               help="print request id and return")
 def launch(entrypoint, cluster, gpus, num_nodes, env, down,
-           idle_minutes_to_autostop, detach_run, async_):
+           idle_minutes_to_autostop, retry_until_up, detach_run, async_):
     """Launch a task (provision + sync + setup + run)."""
     task = _load_task(entrypoint, env,
                       {"accelerators": gpus, "_num_nodes": num_nodes})
-    rid = sdk.launch(task, cluster, down=down,
+    rid = sdk.launch(task, cluster, down=down, retry_until_up=retry_until_up,
                      idle_minutes_to_autostop=idle_minutes_to_autostop)
     if async_:
         click.echo(rid)

@@ -42,7 +42,8 @@ def _generate_cluster_name() -> str:
 def _execute(task_or_dag, cluster_name: Optional[str], stages,
              detach_run: bool = True, down: bool = False,
              idle_minutes_to_autostop: Optional[int] = None,
-             managed_job_id: Optional[int] = None
+             managed_job_id: Optional[int] = None,
+             retry_until_up: bool = False
              ) -> Tuple[Optional[int], Dict[str, Any]]:
     dag = to_dag(task_or_dag)
     assert len(dag.tasks) == 1, "chained DAGs execute task-by-task"
@@ -73,7 +74,24 @@ def _execute(task_or_dag, cluster_name: Optional[str], stages,
                     f"{cand.accelerators or 'cpu'}:"
                     f"{cand.accelerator_count}: {e}")
         if last_err is not None:
-            raise last_err
+            if retry_until_up:
+                # reference: `sky launch --retry-until-up` — loop the
+                # whole candidate list with backoff until capacity
+                # appears.
+                delay = 5.0
+                while last_err is not None:
+                    time.sleep(delay)
+                    delay = min(delay * 2, 60.0)
+                    for cand in cands:
+                        task.resources = cand
+                        try:
+                            handle = backend.provision(task, cluster_name)
+                            last_err = None
+                            break
+                        except ResourcesUnavailableError as e:
+                            last_err = e
+            if last_err is not None:
+                raise last_err
     else:
         record = global_state.get_cluster(cluster_name)
         if record is None:
@@ -114,10 +132,12 @@ def _execute(task_or_dag, cluster_name: Optional[str], stages,
 def launch(task, cluster_name: Optional[str] = None, *,
            detach_run: bool = True, down: bool = False,
            idle_minutes_to_autostop: Optional[int] = None,
-           managed_job_id: Optional[int] = None):
+           managed_job_id: Optional[int] = None,
+           retry_until_up: bool = False):
     """reference: sky/execution.py:688 (launch)."""
     return _execute(task, cluster_name, ALL_STAGES, detach_run, down,
-                    idle_minutes_to_autostop, managed_job_id)
+                    idle_minutes_to_autostop, managed_job_id,
+                    retry_until_up)
 
 
 def exec_(task, cluster_name: str, *, detach_run: bool = True):

@@ -188,3 +188,36 @@ def test_spot_preemption(tmp_path, monkeypatch):
     assert len(rec["handle"]["gpu_ids"]) == 4
     core.down("od-c")
     gpu_topology.detect_gpus.cache_clear()
+
+
+def test_retry_until_up(tmp_path, monkeypatch):
+    """--retry-until-up loops provisioning with backoff until capacity
+    frees (reference: sky launch --retry-until-up)."""
+    import threading
+    import time as _time
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "4")
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    from skypilot_amd import core, execution
+    from skypilot_amd.task import Task
+    hog = Task.from_yaml_config(
+        {"run": "true", "resources": {"accelerators": "MI355X:4"}})
+    execution.launch(hog, "hog-c", detach_run=True)
+
+    def free_later():
+        _time.sleep(6)
+        core.down("hog-c")
+
+    t = threading.Thread(target=free_later)
+    t.start()
+    want = Task.from_yaml_config(
+        {"run": "true", "resources": {"accelerators": "MI355X:2"}})
+    t0 = _time.time()
+    _, handle = execution.launch(want, "want-c", detach_run=True,
+                                 retry_until_up=True)
+    assert handle is not None
+    assert _time.time() - t0 >= 5  # actually waited for capacity
+    t.join()
+    core.down("want-c")
+    gpu_topology.detect_gpus.cache_clear()
