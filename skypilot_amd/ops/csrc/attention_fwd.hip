@@ -419,6 +419,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel32(
   }
 }
 
+extern "C" __global__ void attn_fwd_swapped_kernel(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    unsigned short*, float*, int, int, int, int, float, int);
+
 extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
                                 void* O, float* lse, int B, int S, int Hq,
                                 int Hkv, float scale, bool causal,
@@ -436,6 +440,21 @@ extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
   // profiles/r01_hw_probe_semantics.txt) restructures away.  Dispatch
   // stays on the 16x16 kernel; SKY_ATTN_FWD_32=1 flips it for
   // experiments.
+  // Swapped-operand kernel is the default (238 vs 207 TF/s measured at
+  // the bench shape); SKY_ATTN_FWD_SWAPPED=0 falls back for A/B tests.
+  static const int use_swapped = [] {
+    const char* e = getenv("SKY_ATTN_FWD_SWAPPED");
+    return e ? atoi(e) : 1;
+  }();
+  if (use_swapped && S % BM == 0) {
+    dim3 grid(S / BM, B * Hq);
+    hipLaunchKernelGGL(attn_fwd_swapped_kernel, grid, dim3(256), 0,
+                       stream, (const unsigned short*)Q,
+                       (const unsigned short*)K, (const unsigned short*)V,
+                       (unsigned short*)O, lse, B, S, Hq, Hkv, scale,
+                       causal ? 1 : 0);
+    return;
+  }
   static const int use32 = [] {
     const char* e = getenv("SKY_ATTN_FWD_32");
     return e ? atoi(e) : 0;
@@ -453,4 +472,203 @@ extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
                      (const unsigned short*)Q, (const unsigned short*)K,
                      (const unsigned short*)V, (unsigned short*)O, lse, B, S,
                      Hq, Hkv, scale, causal ? 1 : 0);
+}
+
+// ---------------------------------------------------------------------------
+// Swapped-operand forward: S^T = K Q^T, in-register column softmax,
+// P^T redistributed to B-fragments with cross-lane shuffles, O^T
+// accumulated and transposed on store.  Eliminates the P LDS roundtrip
+// (32 swzP writes + 16 reads per wave-tile) and all softmax LDS state.
+// Same MFMA count and the SAME k_lds/vt_lds reads as attn_fwd_kernel —
+// only the operand roles change (K becomes A, Q becomes B), which is
+// what makes the softmax land on lanes instead of register rows.
+// permlane/tr probe groundwork: profiles/r01_hw_probe_semantics.txt.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_swapped_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, unsigned short* __restrict__ O,
+    float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, float scale,
+    int causal) {
+  __shared__ unsigned short k_lds[BN * ATT_D];    // [kv][d], swizzled, 16KB
+  __shared__ unsigned short vt_lds[ATT_D * BN];   // [d][kv], swizzled, 16KB
+
+  const int qt = gridDim.x - 1 - blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * BM;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;               // wave id: q rows [16w, 16w+16)
+  const int lrow = lane & 15;           // q column owned by this lane
+  const int lgrp = lane >> 4;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  // Q as B-fragments: B[k=d][n=q]: lane holds Q[q=lrow][d=ks*32+lgrp*8+j]
+  // (byte-identical to the A-fragment loads of the unswapped kernel).
+  s16x8 q_b[4];
+  {
+    const int qrow = qbase + 16 * w + lrow;
+    const unsigned short* src = Qb + (long long)qrow * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks)
+      q_b[ks] = *(const s16x8*)(src + ks * 32 + lgrp * 8);
+  }
+
+  // Per-lane softmax state for q column lrow (replicated across lgrp).
+  float m_run = -INFINITY, l_run = 0.f;
+  f32x4 o_t[8];  // O^T: C[row=d=ct*16+lgrp*4+r][col=q=lrow]
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) o_t[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_kv_tiles = causal ? (qbase + BM + BN - 1) / BN : (S + BN - 1) / BN;
+
+  s16x8 kpre[4], vpre[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int s_ch = (tid & 7) | ((i & 1) << 3);
+    int s_row = (tid >> 3) | ((i >> 1) << 5);
+    long long r = (long long)s_row * kv_rowstride + s_ch * 8;
+    kpre[i] = *(const s16x8*)(Kb + r);
+    vpre[i] = *(const s16x8*)(Vb + r);
+  }
+
+  const int my_q = qbase + 16 * w + lrow;
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int s_ch = (tid & 7) | ((i & 1) << 3);
+      int s_row = (tid >> 3) | ((i >> 1) << 5);
+      *(s16x8*)((char*)k_lds + swz(s_row * 256 + s_ch * 16, s_row)) =
+          kpre[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = s_ch * 8 + j;
+        *(unsigned short*)((char*)vt_lds + swzT(d * 128 + s_row * 2, d)) =
+            (unsigned short)vpre[i][j];
+      }
+    }
+    __syncthreads();
+    if (kt + 1 < n_kv_tiles) {
+      const long long base = (long long)(kt + 1) * BN * kv_rowstride;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int s_ch = (tid & 7) | ((i & 1) << 3);
+        int s_row = (tid >> 3) | ((i >> 1) << 5);
+        long long r = base + (long long)s_row * kv_rowstride + s_ch * 8;
+        kpre[i] = *(const s16x8*)(Kb + r);
+        vpre[i] = *(const s16x8*)(Vb + r);
+      }
+    }
+    const int kvbase = kt * BN;
+
+    // ---- S^T = scale * K Q^T : 64 kv rows x 16 q cols for this wave.
+    f32x4 st_acc[4];
+#pragma unroll
+    for (int kv4 = 0; kv4 < 4; ++kv4) {
+      st_acc[kv4] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int krow = kv4 * 16 + lrow;
+        s16x8 a_k = *(const s16x8*)((char*)k_lds +
+            swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
+        st_acc[kv4] = MFMA_BF16(as_bf16x8(a_k), as_bf16x8(q_b[ks]),
+                                st_acc[kv4]);
+      }
+    }
+
+    // ---- mask + in-register column softmax (q = lrow on every lane).
+    const bool need_mask =
+        (causal && kvbase + BN - 1 > qbase + 16 * w) || (kvbase + BN > S);
+    float mx = -INFINITY;
+#pragma unroll
+    for (int kv4 = 0; kv4 < 4; ++kv4)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = st_acc[kv4][r] * scale;
+        if (need_mask) {
+          int kv = kvbase + kv4 * 16 + lgrp * 4 + r;
+          if ((causal && kv > my_q) || kv >= S) s = -INFINITY;
+        }
+        st_acc[kv4][r] = s;
+        mx = fmaxf(mx, s);
+      }
+    mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
+    mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+    float m_new = fmaxf(m_run, mx);
+    float m_safe = (m_new == -INFINITY) ? 0.f : m_new;
+    float corr = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_safe);
+    m_run = m_new;
+    float rs = 0.f;
+#pragma unroll
+    for (int kv4 = 0; kv4 < 4; ++kv4)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float e = (st_acc[kv4][r] == -INFINITY)
+                      ? 0.f : __expf(st_acc[kv4][r] - m_safe);
+        st_acc[kv4][r] = e;  // P^T stays in the S registers
+        rs += e;
+      }
+    rs += __shfl_xor(rs, 16, 64);
+    rs += __shfl_xor(rs, 32, 64);
+    l_run = l_run * corr + rs;
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_t[ct][r] *= corr;
+
+    // ---- P^T -> B-fragments via cross-lane shuffles.
+    // Dest (group g, elem j) needs P^T[kv = ks*32 + g*8 + j][q=lrow],
+    // held by lane (2*(g&1) + (j>>2))*16 + lrow in sub-register
+    // [kvsub = 2ks + (g>>1)][r = j&3].  kvsub depends on the dest group
+    // (runtime), so shuffle both candidates (compile-time indices) and
+    // select — 2 shfl + 1 cndmask per element.
+    s16x8 p_b[2];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int src = (2 * (lgrp & 1) + (j >> 2)) * 16 + lrow;
+        float v0 = __shfl(st_acc[2 * ks][j & 3], src, 64);
+        float v1 = __shfl(st_acc[2 * ks + 1][j & 3], src, 64);
+        p_b[ks][j] = (short)f2bf_trunc((lgrp >> 1) ? v1 : v0);
+      }
+    }
+
+    // ---- O^T += V^T P^T : A from vt_lds (same reads as unswapped).
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        int vrow = ct * 16 + lrow;
+        s16x8 a_vt = *(const s16x8*)((char*)vt_lds +
+            swzT(vrow * 128 + (ks * 32 + lgrp * 8) * 2, vrow));
+        o_t[ct] = MFMA_BF16(as_bf16x8(a_vt), as_bf16x8(p_b[ks]), o_t[ct]);
+      }
+    }
+  }
+
+  // ---- epilogue: O[q][d] = O^T[d][q] / l (s16x4 stores, d block
+  // ct*16 + lgrp*4), lse once per q column.
+  unsigned short* Ob = O + ((long long)b * S * Hq + qh) * ATT_D;
+  float* lse_b = lse_out + ((long long)b * Hq + qh) * S;
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  unsigned short* orow = Ob + (long long)my_q * q_rowstride;
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+    s16x4 ov;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) ov[r] = (short)f2bf(o_t[ct][r] * inv_l);
+    *(s16x4*)(orow + ct * 16 + lgrp * 4) = ov;
+  }
+  if (lgrp == 0)
+    lse_b[my_q] = (l_run > 0.f) ? m_run + __logf(l_run) : -INFINITY;
 }
