@@ -571,19 +571,22 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_swapped_kernel(
     const int kvbase = kt * BN;
 
     // ---- S^T = scale * K Q^T : 64 kv rows x 16 q cols for this wave.
+    // ks outer / kv4 inner: 4 independent accumulator chains in flight
+    // per MFMA latency window (kv4-outer serializes each 4-deep chain).
     f32x4 st_acc[4];
 #pragma unroll
-    for (int kv4 = 0; kv4 < 4; ++kv4) {
+    for (int kv4 = 0; kv4 < 4; ++kv4)
       st_acc[kv4] = f32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int ks = 0; ks < 4; ++ks) {
+    for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+      for (int kv4 = 0; kv4 < 4; ++kv4) {
         int krow = kv4 * 16 + lrow;
         s16x8 a_k = *(const s16x8*)((char*)k_lds +
             swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
         st_acc[kv4] = MFMA_BF16(as_bf16x8(a_k), as_bf16x8(q_b[ks]),
                                 st_acc[kv4]);
       }
-    }
 
     // ---- mask + in-register column softmax (q = lrow on every lane).
     const bool need_mask =
@@ -645,7 +648,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_swapped_kernel(
 
     // ---- O^T += V^T P^T : A from vt_lds (same reads as unswapped).
 #pragma unroll
-    for (int ct = 0; ct < 8; ++ct) {
+    for (int ct = 0; ct < 8; ++ct)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
         int vrow = ct * 16 + lrow;
@@ -653,7 +656,6 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_swapped_kernel(
             swzT(vrow * 128 + (ks * 32 + lgrp * 8) * 2, vrow));
         o_t[ct] = MFMA_BF16(as_bf16x8(a_vt), as_bf16x8(p_b[ks]), o_t[ct]);
       }
-    }
   }
 
   // ---- epilogue: O[q][d] = O^T[d][q] / l (s16x4 stores, d block

@@ -57,6 +57,16 @@ __device__ __forceinline__ int swzT(int byte_off, int row) {
   return byte_off ^ (((row >> 3) & 7) << 4);
 }
 
+// Read-friendly transposed-tile swizzle: keyed on the LOW d bits so the
+// B/A-fragment reads (lanes sweep d = ct*16 + lrow, identical kv
+// offset) spread across banks (swzT keyed d>>3 leaves them ~4-way
+// conflicted — the dominant LDS stall measured in attention).  Writes
+// must then be the vectorized transpose-staging path (s16x4 at d-pairs
+// spanning ch and quad), which this keying also spreads.
+__device__ __forceinline__ int swzT2(int byte_off, int row) {
+  return byte_off ^ ((row & 7) << 4);
+}
+
 // Swizzle for the P / dS staging tiles ([q or kv rows][64] C-layout
 // scatter): writes vary the row as lgrp*4+r (only 2 distinct (row&7)
 // per instruction) but cols span 16 — keying on (row>>1) gives 4 row
