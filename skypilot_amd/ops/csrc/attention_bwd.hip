@@ -11,6 +11,7 @@
 // Math (P normalized via saved lse): P = exp(scale*QK^T - lse);
 // dV = P^T dO; dP = dO V^T; dS = P*(dP - Dvec); dQ = scale*dS*K;
 // dK = scale*dS^T*Q.
+#include <cstdlib>
 #include "common.h"
 
 #define ATT_D 128
@@ -564,6 +565,11 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel32(
   }
 }
 
+extern "C" __global__ void attn_bwd_dq_swapped_kernel(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const float*, const float*, unsigned short*,
+    int, int, int, int, float, int);
+
 extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
                                 const void* O, const void* dO,
                                 const float* lse, float* Dvec, void* dQ,
@@ -585,9 +591,176 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
                      (const unsigned short*)dO, lse, Dvec,
                      (unsigned short*)dK, (unsigned short*)dV, B, S, Hq, Hkv,
                      scale, causal ? 1 : 0);
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(S / BM, B * Hq), dim3(256), 0,
-                     stream, (const unsigned short*)Q,
-                     (const unsigned short*)K, (const unsigned short*)V,
-                     (const unsigned short*)dO, lse, Dvec,
-                     (unsigned short*)dQ, B, S, Hq, Hkv, scale, causal ? 1 : 0);
+  // Swapped dq is the default (bwd 163 -> 169 TF/s measured);
+  // SKY_ATTN_DQ_SWAPPED=0 falls back for A/B tests.
+  static const int dq_swapped = [] {
+    const char* e = getenv("SKY_ATTN_DQ_SWAPPED");
+    return e ? atoi(e) : 1;
+  }();
+  if (dq_swapped)
+    hipLaunchKernelGGL(attn_bwd_dq_swapped_kernel, dim3(S / BM, B * Hq),
+                       dim3(256), 0, stream, (const unsigned short*)Q,
+                       (const unsigned short*)K, (const unsigned short*)V,
+                       (const unsigned short*)dO, lse, Dvec,
+                       (unsigned short*)dQ, B, S, Hq, Hkv, scale,
+                       causal ? 1 : 0);
+  else
+    hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(S / BM, B * Hq), dim3(256),
+                       0, stream, (const unsigned short*)Q,
+                       (const unsigned short*)K, (const unsigned short*)V,
+                       (const unsigned short*)dO, lse, Dvec,
+                       (unsigned short*)dQ, B, S, Hq, Hkv, scale,
+                       causal ? 1 : 0);
+}
+
+// ---------------------------------------------------------------------------
+// Swapped-operand dQ kernel (same trick as attn_fwd_swapped_kernel):
+// S^T = K Q^T and dP^T = V dO^T put q on lanes, so P^T/dS^T stay in
+// registers (lse/Dvec become per-lane scalars) and dQ^T = K^T dS^T
+// consumes dS^T as shuffled B-fragments — the ds_lds roundtrip and one
+// of the two barriers per tile disappear.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_swapped_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dQ, int B, int S, int Hq, int Hkv,
+    float scale, int causal) {
+  __shared__ unsigned short k_lds[BN * ATT_D];
+  __shared__ unsigned short kt_lds[ATT_D * BN];
+  __shared__ unsigned short v_lds[BN * ATT_D];
+
+  const int qt = gridDim.x - 1 - blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * BM;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int lrow = lane & 15;
+  const int lgrp = lane >> 4;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+  const float* lse_b = lse + ((long long)b * Hq + qh) * S;
+  const float* dvec_b = Dvec + (long long)b * S * Hq + qh;
+
+  // Q / dO as B-fragments (byte-identical loads to the A-fragments of
+  // the unswapped kernel); lse/D collapse to per-lane scalars.
+  s16x8 q_b[4], do_b[4];
+  const int my_q = qbase + 16 * w + lrow;
+  {
+    const unsigned short* qsrc = Qb + (long long)my_q * q_rowstride;
+    const unsigned short* dsrc = dOb + (long long)my_q * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      q_b[ks] = *(const s16x8*)(qsrc + ks * 32 + lgrp * 8);
+      do_b[ks] = *(const s16x8*)(dsrc + ks * 32 + lgrp * 8);
+    }
+  }
+  const float my_lse = lse_b[my_q];
+  const float my_dvec = dvec_b[(long long)my_q * Hq];
+
+  f32x4 dq_t[8];  // dQ^T: C[row=d=ct*16+lgrp*4+r][col=q=lrow]
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) dq_t[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_kv_tiles = causal ? (qbase + BM + BN - 1) / BN : S / BN;
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kvbase = kt * BN;
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int ch = (tid & 7) | ((i & 1) << 3);
+      int row = (tid >> 3) | ((i >> 1) << 5);
+      s16x8 kv8 = *(const s16x8*)(Kb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+      *(s16x8*)((char*)k_lds + swz(row * 256 + ch * 16, row)) = kv8;
+      s16x8 vv8 = *(const s16x8*)(Vb + (long long)(kvbase + row) * kv_rowstride + ch * 8);
+      *(s16x8*)((char*)v_lds + swz(row * 256 + ch * 16, row)) = vv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int d = ch * 8 + j;
+        *(unsigned short*)((char*)kt_lds + swzT(d * 128 + row * 2, d)) =
+            (unsigned short)kv8[j];
+      }
+    }
+    __syncthreads();
+
+    // S^T = K Q^T, dP^T = V dO^T (interleaved independent chains).
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int kv4 = 0; kv4 < 4; ++kv4) {
+      st[kv4] = f32x4{0.f, 0.f, 0.f, 0.f};
+      dpt[kv4] = f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+      for (int kv4 = 0; kv4 < 4; ++kv4) {
+        int krow = kv4 * 16 + lrow;
+        s16x8 a_k = *(const s16x8*)((char*)k_lds +
+            swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
+        st[kv4] = MFMA_BF16(as_bf16x8(a_k), as_bf16x8(q_b[ks]), st[kv4]);
+        s16x8 a_v = *(const s16x8*)((char*)v_lds +
+            swz(krow * 256 + (ks * 32 + lgrp * 8) * 2, krow));
+        dpt[kv4] = MFMA_BF16(as_bf16x8(a_v), as_bf16x8(do_b[ks]),
+                             dpt[kv4]);
+      }
+
+    // dS^T = P^T * (dP^T - D);  P^T = exp(scale*S^T - lse).
+#pragma unroll
+    for (int kv4 = 0; kv4 < 4; ++kv4)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int kv = kvbase + kv4 * 16 + lgrp * 4 + r;
+        float pv;
+        if ((causal && kv > my_q) || my_lse == -INFINITY)
+          pv = 0.f;
+        else
+          pv = __expf(st[kv4][r] * scale - my_lse);
+        st[kv4][r] = pv * (dpt[kv4][r] - my_dvec);
+      }
+
+    // dS^T -> B-fragments (same shuffle recipe as attn_fwd_swapped).
+    s16x8 ds_b[2];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int src = (2 * (lgrp & 1) + (j >> 2)) * 16 + lrow;
+        float v0 = __shfl(st[2 * ks][j & 3], src, 64);
+        float v1 = __shfl(st[2 * ks + 1][j & 3], src, 64);
+        ds_b[ks][j] = (short)f2bf((lgrp >> 1) ? v1 : v0);
+      }
+
+    // dQ^T += K^T dS^T : A from kt_lds (same reads as unswapped).
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        int krow = ct * 16 + lrow;
+        s16x8 a_kt = *(const s16x8*)((char*)kt_lds +
+            swzT(krow * 128 + (ks * 32 + lgrp * 8) * 2, krow));
+        dq_t[ct] = MFMA_BF16(as_bf16x8(a_kt), as_bf16x8(ds_b[ks]),
+                             dq_t[ct]);
+      }
+  }
+
+  unsigned short* dQb = dQ + ((long long)b * S * Hq + qh) * ATT_D;
+  unsigned short* orow = dQb + (long long)my_q * q_rowstride;
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+    s16x4 ov;
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      ov[r] = (short)f2bf(dq_t[ct][r] * scale);
+    *(s16x4*)(orow + ct * 16 + lgrp * 4) = ov;
+  }
 }
