@@ -565,6 +565,11 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel32(
   }
 }
 
+extern "C" __global__ void attn_bwd_dkv_swapped_kernel(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const float*, const float*, unsigned short*,
+    unsigned short*, int, int, int, int, float, int);
+
 extern "C" __global__ void attn_bwd_dq_swapped_kernel(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const unsigned short*, const float*, const float*, unsigned short*,
@@ -585,12 +590,24 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
   // losing the cross-block barrier/compute overlap that two independent
   // 256-thread blocks provide.  Kept for the round-2 deep-pipeline
   // rewrite; dispatch stays on the 16x16 kernel.
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(S / BN, B * Hkv),
-                     dim3(256), 0, stream, (const unsigned short*)Q,
-                     (const unsigned short*)K, (const unsigned short*)V,
-                     (const unsigned short*)dO, lse, Dvec,
-                     (unsigned short*)dK, (unsigned short*)dV, B, S, Hq, Hkv,
-                     scale, causal ? 1 : 0);
+  static const int dkv_swapped = [] {
+    const char* e = getenv("SKY_ATTN_DKV_SWAPPED");
+    return e ? atoi(e) : 0;
+  }();
+  if (dkv_swapped)
+    hipLaunchKernelGGL(attn_bwd_dkv_swapped_kernel, dim3(S / BN, B * Hkv),
+                       dim3(256), 0, stream, (const unsigned short*)Q,
+                       (const unsigned short*)K, (const unsigned short*)V,
+                       (const unsigned short*)dO, lse, Dvec,
+                       (unsigned short*)dK, (unsigned short*)dV, B, S, Hq,
+                       Hkv, scale, causal ? 1 : 0);
+  else
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(S / BN, B * Hkv),
+                       dim3(256), 0, stream, (const unsigned short*)Q,
+                       (const unsigned short*)K, (const unsigned short*)V,
+                       (const unsigned short*)dO, lse, Dvec,
+                       (unsigned short*)dK, (unsigned short*)dV, B, S, Hq,
+                       Hkv, scale, causal ? 1 : 0);
   // Swapped dq is the default (bwd 163 -> 169 TF/s measured);
   // SKY_ATTN_DQ_SWAPPED=0 falls back for A/B tests.
   static const int dq_swapped = [] {
@@ -762,5 +779,191 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_swapped_kernel(
     for (int r = 0; r < 4; ++r)
       ov[r] = (short)f2bf(dq_t[ct][r] * scale);
     *(s16x4*)(orow + ct * 16 + lgrp * 4) = ov;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Swapped-operand dK/dV kernel: S = Q K^T / dP = dO V^T keep q in
+// register rows and kv on lanes, so P and dS stay in registers and feed
+// dV^T = dO^T P and dK^T = Q^T dS as shuffled B-fragments — the
+// p_lds/ds_lds roundtrips and one barrier per q-tile disappear (LDS
+// 80 -> 64 KB).  Q^T/dO^T staging is unchanged.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_swapped_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dK, unsigned short* __restrict__ dV, int B,
+    int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ unsigned short q_lds[BM * ATT_D];
+  __shared__ unsigned short qt_lds[ATT_D * BM];
+  __shared__ unsigned short do_lds[BM * ATT_D];
+  __shared__ unsigned short dot_lds[ATT_D * BM];
+
+  const int kt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int kvh = bh % Hkv;
+  const int group = Hq / Hkv;
+  const int kvbase = kt * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;    // wave owns kv rows [16w, 16w+16)
+  const int lrow = lane & 15;
+  const int lgrp = lane >> 4;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  // K,V as B-fragments for this wave's 16 kv columns (byte-identical
+  // loads to the unswapped kernel's A-fragments).
+  s16x8 k_b[4], v_b[4];
+  const int my_kv = kvbase + 16 * w + lrow;
+  {
+    const unsigned short* ksrc = Kb + (long long)my_kv * kv_rowstride;
+    const unsigned short* vsrc = Vb + (long long)my_kv * kv_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      k_b[ks] = *(const s16x8*)(ksrc + ks * 32 + lgrp * 8);
+      v_b[ks] = *(const s16x8*)(vsrc + ks * 32 + lgrp * 8);
+    }
+  }
+
+  f32x4 dv_t[8], dk_t[8];  // C[row=d=ct*16+lgrp*4+r][col=kv=lrow]
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+    dv_t[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dk_t[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  for (int g = 0; g < group; ++g) {
+    const int qh = kvh * group + g;
+    const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+    const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+    const float* lse_b = lse + ((long long)b * Hq + qh) * S;
+    const float* dvec_b = Dvec + (long long)b * S * Hq + qh;
+
+    const int qt0 = causal ? kvbase / BM : 0;
+    for (int qt = qt0; qt < S / BM; ++qt) {
+      const int qbase = qt * BM;
+      __syncthreads();
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int ch = (tid & 7) | ((i & 1) << 3);
+        int row = (tid >> 3) | ((i >> 1) << 5);
+        s16x8 qv = *(const s16x8*)(Qb + (long long)(qbase + row) * q_rowstride + ch * 8);
+        *(s16x8*)((char*)q_lds + swz(row * 256 + ch * 16, row)) = qv;
+        s16x8 dov = *(const s16x8*)(dOb + (long long)(qbase + row) * q_rowstride + ch * 8);
+        *(s16x8*)((char*)do_lds + swz(row * 256 + ch * 16, row)) = dov;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = ch * 8 + j;
+          *(unsigned short*)((char*)qt_lds + swzT(d * 128 + row * 2, d)) =
+              (unsigned short)qv[j];
+          *(unsigned short*)((char*)dot_lds + swzT(d * 128 + row * 2, d)) =
+              (unsigned short)dov[j];
+        }
+      }
+      __syncthreads();
+
+      // S = Q K^T, dP = dO V^T : [q 64 rows][kv 16 lanes] per wave.
+      f32x4 sp[4], dp[4];
+#pragma unroll
+      for (int qsub = 0; qsub < 4; ++qsub) {
+        sp[qsub] = f32x4{0.f, 0.f, 0.f, 0.f};
+        dp[qsub] = f32x4{0.f, 0.f, 0.f, 0.f};
+      }
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+        for (int qsub = 0; qsub < 4; ++qsub) {
+          int qrow = qsub * 16 + lrow;
+          s16x8 a_q = *(const s16x8*)((char*)q_lds +
+              swz(qrow * 256 + (ks * 32 + lgrp * 8) * 2, qrow));
+          sp[qsub] = MFMA_BF16(as_bf16x8(a_q), as_bf16x8(k_b[ks]),
+                               sp[qsub]);
+          s16x8 a_do = *(const s16x8*)((char*)do_lds +
+              swz(qrow * 256 + (ks * 32 + lgrp * 8) * 2, qrow));
+          dp[qsub] = MFMA_BF16(as_bf16x8(a_do), as_bf16x8(v_b[ks]),
+                               dp[qsub]);
+        }
+
+      // P = exp(scale*S - lse[q]); dS = P * (dP - D[q]); in registers.
+      // lse/D loaded once per lane column (4 each) and redistributed
+      // to register rows by shuffle — 16 scalar loads would stall.
+      float lse_l[4], dv_l[4];
+#pragma unroll
+      for (int qsub = 0; qsub < 4; ++qsub) {
+        int qc = qbase + qsub * 16 + lrow;
+        lse_l[qsub] = lse_b[qc];
+        dv_l[qsub] = dvec_b[(long long)qc * Hq];
+      }
+#pragma unroll
+      for (int qsub = 0; qsub < 4; ++qsub)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int qrow = qbase + qsub * 16 + lgrp * 4 + r;
+          float l = __shfl(lse_l[qsub], lgrp * 4 + r, 64);
+          float pv;
+          if ((causal && qrow < my_kv) || l == -INFINITY)
+            pv = 0.f;
+          else
+            pv = __expf(sp[qsub][r] * scale - l);
+          sp[qsub][r] = pv;
+          dp[qsub][r] = pv * (dp[qsub][r] -
+                              __shfl(dv_l[qsub], lgrp * 4 + r, 64));
+        }
+
+      // P / dS -> B-fragments (same shuffle recipe as the fwd kernel).
+      s16x8 p_b[2], ds_b[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int src = (2 * (lgrp & 1) + (j >> 2)) * 16 + lrow;
+          float p0 = __shfl(sp[2 * ks][j & 3], src, 64);
+          float p1 = __shfl(sp[2 * ks + 1][j & 3], src, 64);
+          p_b[ks][j] = (short)f2bf_trunc((lgrp >> 1) ? p1 : p0);
+          float d0 = __shfl(dp[2 * ks][j & 3], src, 64);
+          float d1 = __shfl(dp[2 * ks + 1][j & 3], src, 64);
+          ds_b[ks][j] = (short)f2bf((lgrp >> 1) ? d1 : d0);
+        }
+
+      // dV^T += dO^T P ; dK^T += Q^T dS  (A reads identical bytes to
+      // the unswapped kernel's dOT/QT B reads).
+#pragma unroll
+      for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          int drow = ct * 16 + lrow;
+          int b_off = swzT(drow * 128 + (ks * 32 + lgrp * 8) * 2, drow);
+          s16x8 a_dot = *(const s16x8*)((char*)dot_lds + b_off);
+          dv_t[ct] = MFMA_BF16(as_bf16x8(a_dot), as_bf16x8(p_b[ks]),
+                               dv_t[ct]);
+          s16x8 a_qt = *(const s16x8*)((char*)qt_lds + b_off);
+          dk_t[ct] = MFMA_BF16(as_bf16x8(a_qt), as_bf16x8(ds_b[ks]),
+                               dk_t[ct]);
+        }
+    }
+  }
+
+  // dK/dV transpose on store (s16x4 per d block), dK scaled.
+  unsigned short* dKb = dK + ((long long)b * S * Hkv + kvh) * ATT_D;
+  unsigned short* dVb = dV + ((long long)b * S * Hkv + kvh) * ATT_D;
+  unsigned short* krow_p = dKb + (long long)my_kv * kv_rowstride;
+  unsigned short* vrow_p = dVb + (long long)my_kv * kv_rowstride;
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+    s16x4 kv4, vv4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      kv4[r] = (short)f2bf(dk_t[ct][r] * scale);
+      vv4[r] = (short)f2bf(dv_t[ct][r]);
+    }
+    *(s16x4*)(krow_p + ct * 16 + lgrp * 4) = kv4;
+    *(s16x4*)(vrow_p + ct * 16 + lgrp * 4) = vv4;
   }
 }
