@@ -29,14 +29,19 @@ def main():
     # warmup
     eng.generate(prompt, max_tokens=8)
 
+    from skypilot_amd.serve.engine import Request
     for nb in [int(x) for x in args.batches.split(",")]:
         results = []
+        ttfts = []
         t0 = time.perf_counter()
 
         def worker():
-            out = eng.generate(prompt, max_tokens=args.gen_len,
-                               timeout=1200)
-            results.append(len(out))
+            req = eng.submit(Request(prompt_ids=list(prompt),
+                                     max_tokens=args.gen_len))
+            req.done.wait(1200)
+            results.append(len(req.out_ids))
+            if req.first_token_at:
+                ttfts.append(req.first_token_at - req.created)
 
         threads = [threading.Thread(target=worker) for _ in range(nb)]
         for t in threads:
@@ -45,9 +50,12 @@ def main():
             t.join()
         dt = time.perf_counter() - t0
         toks = sum(results)
+        ttfts.sort()
+        p50 = ttfts[len(ttfts) // 2] * 1e3 if ttfts else 0
+        p95 = ttfts[int(len(ttfts) * 0.95)] * 1e3 if ttfts else 0
         print(f"concurrency {nb:3d}: {toks} tokens in {dt:6.2f}s = "
-              f"{toks/dt:8.1f} tok/s decode "
-              f"(ttft incl. prefill; prompt {args.prompt_len})",
+              f"{toks/dt:8.1f} tok/s decode | ttft p50 {p50:6.1f} ms "
+              f"p95 {p95:6.1f} ms (prompt {args.prompt_len})",
               flush=True)
     print(f"graph buckets captured: {eng.stats['graph_buckets']} "
           f"(use_graphs={eng.use_graphs})")
