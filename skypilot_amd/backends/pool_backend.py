@@ -51,8 +51,25 @@ class PoolBackend(Backend):
         return handle
 
     # ---- file sync --------------------------------------------------------
-    def sync_workdir(self, handle: Dict[str, Any], workdir: str) -> None:
+    def sync_workdir(self, handle: Dict[str, Any], workdir) -> None:
         dst = Path(handle["cluster_dir"]) / "workdir"
+        if isinstance(workdir, dict):
+            # git-source workdir (reference: schemas.py {url, ref}):
+            # clone/fetch into the cluster workdir and check out `ref`.
+            import subprocess
+            url = workdir["url"]
+            ref = workdir.get("ref")
+            if (dst / ".git").exists():
+                subprocess.run(["git", "-C", str(dst), "fetch", "origin"],
+                               check=True, capture_output=True)
+            else:
+                dst.mkdir(parents=True, exist_ok=True)
+                subprocess.run(["git", "clone", url, str(dst)],
+                               check=True, capture_output=True)
+            if ref:
+                subprocess.run(["git", "-C", str(dst), "checkout", ref],
+                               check=True, capture_output=True)
+            return
         runner = LocalProcessCommandRunner()
         src = os.path.expanduser(workdir)
         if not src.endswith("/"):

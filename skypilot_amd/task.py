@@ -78,8 +78,14 @@ class Task:
         self.validate_run()
         if self.num_nodes < 1:
             raise TaskValidationError("num_nodes must be >= 1")
-        if self.workdir is not None and not isinstance(self.workdir, str):
-            raise TaskValidationError("workdir must be a string path")
+        if self.workdir is not None and not isinstance(self.workdir,
+                                                       (str, dict)):
+            raise TaskValidationError(
+                "workdir must be a path or {url, ref} git source")
+        if isinstance(self.workdir, dict):
+            # reference: schemas.py workdir {url, ref} git-source form
+            if "url" not in self.workdir:
+                raise TaskValidationError("git workdir needs a url")
         for k in self.envs:
             if not isinstance(k, str):
                 raise TaskValidationError(f"env name must be str: {k!r}")

@@ -256,3 +256,31 @@ def test_event_callback_and_priority(client, tmp_path):
     assert lines.index("hi") < lines.index("lo")
     assert "cb-SUCCEEDED" in lines
     sdk.get(sdk.down("t-prio"))
+
+
+def test_git_workdir(sky_env, client, tmp_path):
+    """workdir as {url, ref} clones a git source into the cluster
+    workdir (reference: schemas.py git-source workdir)."""
+    import subprocess
+    from skypilot_amd.client import sdk
+    repo = tmp_path / "src-repo"
+    repo.mkdir()
+    subprocess.run(["git", "init", "-q", str(repo)], check=True)
+    (repo / "hello.txt").write_text("v1")
+    env = {"GIT_AUTHOR_NAME": "t", "GIT_AUTHOR_EMAIL": "t@t",
+           "GIT_COMMITTER_NAME": "t", "GIT_COMMITTER_EMAIL": "t@t",
+           "HOME": str(tmp_path), "PATH": os.environ["PATH"]}
+    subprocess.run(["git", "-C", str(repo), "add", "-A"], check=True,
+                   env=env)
+    subprocess.run(["git", "-C", str(repo), "commit", "-qm", "c1"],
+                   check=True, env=env)
+    subprocess.run(["git", "-C", str(repo), "branch", "feat"],
+                   check=True, env=env)
+    sdk.get(sdk.launch({"workdir": {"url": str(repo), "ref": "feat"},
+                        "run": "cat hello.txt"},
+                       "git-wd"), timeout=60)
+    home = Path(os.environ["SKY_AMD_HOME"])
+    wd = home / "clusters" / "git-wd" / "workdir"
+    assert (wd / "hello.txt").read_text() == "v1"
+    assert (wd / ".git").exists()
+    sdk.get(sdk.down("git-wd"))
