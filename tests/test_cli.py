@@ -53,3 +53,38 @@ def test_cli_inline_command(client):
                          "--detach-run"])
     assert res.exit_code == 0, res.output
     r.invoke(cli, ["down", "t-inline"])
+
+
+def test_cli_users_cost_group_storage(client, tmp_path):
+    """Surface checks for the round-1 additions: sky users / cost-report
+    / jobs group / storage sync."""
+    from skypilot_amd.cli.main import cli
+    r = CliRunner()
+
+    res = r.invoke(cli, ["users", "list"])
+    assert res.exit_code == 0 and "admin" in res.output
+
+    res = r.invoke(cli, ["users", "token", "cli-bot", "--role", "viewer"])
+    assert res.exit_code == 0 and res.output.strip().startswith("sky_")
+    res = r.invoke(cli, ["users", "tokens"])
+    assert "cli-bot" in res.output
+    res = r.invoke(cli, ["users", "revoke", "cli-bot"])
+    assert res.exit_code == 0
+
+    res = r.invoke(cli, ["cost-report"])
+    assert res.exit_code == 0 and "GPU-HOURS" in res.output
+
+    gy = tmp_path / "grp.yaml"
+    gy.write_text(
+        "tasks:\n"
+        "  - name: a\n    run: echo A\n"
+        "  - name: b\n    run: echo B\n")
+    res = r.invoke(cli, ["jobs", "group", "launch", str(gy), "-n", "cg"])
+    assert res.exit_code == 0, res.output
+    res = r.invoke(cli, ["jobs", "group", "status", "cg"])
+    assert res.exit_code == 0 and '"cluster": "sky-group-cg"' in res.output
+    res = r.invoke(cli, ["jobs", "group", "down", "cg"])
+    assert res.exit_code == 0
+
+    res = r.invoke(cli, ["storage", "ls"])
+    assert res.exit_code == 0
