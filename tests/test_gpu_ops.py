@@ -306,3 +306,37 @@ def test_rope_kvwrite_matches_ref():
     torch.testing.assert_close(kc.float(), kc_ref.float(), atol=0.02,
                                rtol=0.02)
     torch.testing.assert_close(vc, vc_ref)
+
+
+@pytest.mark.gpu
+def test_attention_shape_fuzz():
+    """Swapped fwd/bwd kernels across shapes: batch, GQA ratio, causal,
+    seq length — all against the fp32 reference."""
+    torch.manual_seed(7)
+    for (B, S, Hq, Hkv, causal) in [
+            (1, 64, 8, 8, True), (3, 128, 16, 2, True),
+            (2, 256, 32, 8, False), (1, 1024, 8, 2, True),
+            (2, 192, 4, 4, False)]:
+        q = (torch.randn(B, S, Hq, 128, device="cuda") * 0.5).bfloat16()
+        k = (torch.randn(B, S, Hkv, 128, device="cuda") * 0.5).bfloat16()
+        v = (torch.randn(B, S, Hkv, 128, device="cuda") * 0.5).bfloat16()
+        q.requires_grad_(True)
+        k.requires_grad_(True)
+        v.requires_grad_(True)
+        out = ops.attention(q, k, v, 128 ** -0.5, causal=causal)
+        g = torch.randn_like(out) * 0.5
+        out.backward(g)
+        qf = q.detach().float().requires_grad_(True)
+        kf = k.detach().float().requires_grad_(True)
+        vf = v.detach().float().requires_grad_(True)
+        ref = ops.attention_ref(qf, kf, vf, 128 ** -0.5, causal=causal)
+        ref.backward(g.float())
+        cfg = (B, S, Hq, Hkv, causal)
+        torch.testing.assert_close(out.float(), ref, atol=0.05,
+                                   rtol=0.05, msg=str(cfg))
+        torch.testing.assert_close(q.grad.float(), qf.grad, atol=0.08,
+                                   rtol=0.08, msg=str(cfg))
+        torch.testing.assert_close(k.grad.float(), kf.grad, atol=0.08,
+                                   rtol=0.08, msg=str(cfg))
+        torch.testing.assert_close(v.grad.float(), vf.grad, atol=0.08,
+                                   rtol=0.08, msg=str(cfg))
