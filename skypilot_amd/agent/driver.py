@@ -237,7 +237,35 @@ def run_job(cluster_dir: str, job_id: int) -> int:
     else:
         table.set_status(job_id, final, exit_code=worst)
     _run_event_callback(spec, task_id, job_id, final, workdir, log_dir)
+    _ship_logs(task_id, job_id, final, log_dir)
     return 0 if final == job_lib.CANCELLED else worst
+
+
+def _ship_logs(task_id, job_id, status, log_dir):
+    """External log shipping (reference: sky/logs/agent.py
+    FluentbitAgent installed at provision time; here a plain HTTP sink
+    configured as `logs: {endpoint: <url>}` in ~/.sky_amd/config.yaml).
+    Ships each completed job's log files as JSON lines; failures never
+    affect the job."""
+    try:
+        from skypilot_amd import config as sky_config
+        endpoint = sky_config.get_nested(["logs", "endpoint"])
+        if not endpoint:
+            return
+        import json as _json
+        import urllib.request
+        for lf in sorted(Path(log_dir).glob("*.log")):
+            body = _json.dumps({
+                "task_id": task_id, "job_id": job_id, "status": status,
+                "file": lf.name,
+                "content": lf.read_text(errors="replace")[-65536:],
+            }).encode()
+            req = urllib.request.Request(
+                endpoint, data=body,
+                headers={"Content-Type": "application/json"})
+            urllib.request.urlopen(req, timeout=10).read()
+    except Exception:  # noqa: BLE001 — best-effort shipping
+        pass
 
 
 def _run_event_callback(spec, task_id, job_id, status, workdir, log_dir):

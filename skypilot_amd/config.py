@@ -14,7 +14,17 @@ import yaml
 _lock = threading.Lock()
 _cache: Optional[Dict[str, Any]] = None
 
-USER_CONFIG_PATH = "~/.sky_amd/config.yaml"
+# The user config lives under the state root so SKY_AMD_HOME relocates
+# it together with the rest of the deployment (and tests isolate it).
+def _user_config_path() -> str:
+    import os as _os
+    if USER_CONFIG_PATH != "~/.sky_amd/config.yaml":
+        return USER_CONFIG_PATH  # explicitly overridden (tests/tools)
+    return _os.path.join(
+        _os.environ.get("SKY_AMD_HOME", "~/.sky_amd"), "config.yaml")
+
+
+USER_CONFIG_PATH = "~/.sky_amd/config.yaml"  # legacy fallback
 PROJECT_CONFIG_PATH = "./sky_amd_config.yaml"
 
 DEFAULTS: Dict[str, Any] = {
@@ -56,7 +66,7 @@ def load(refresh: bool = False) -> Dict[str, Any]:
     with _lock:
         if _cache is None or refresh:
             cfg = copy.deepcopy(DEFAULTS)
-            cfg = _merge(cfg, _load_file(USER_CONFIG_PATH))
+            cfg = _merge(cfg, _load_file(_user_config_path()))
             cfg = _merge(cfg, _load_file(PROJECT_CONFIG_PATH))
             _cache = cfg
         return _cache

@@ -81,3 +81,47 @@ def test_network_partition_reports_init_and_recovers(client):
     rec = next(r for r in records if r["name"] == "t-chaos3")
     assert rec["status"] == "UP"
     sdk.get(sdk.down("t-chaos3"))
+
+
+def test_log_shipping_to_http_sink(client, tmp_path):
+    """`logs: {endpoint}` config ships completed-job logs to an external
+    HTTP sink (reference: sky/logs/agent.py FluentbitAgent)."""
+    import http.server
+    import threading
+
+    received = []
+
+    class Sink(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            n = int(self.headers.get("Content-Length", 0))
+            received.append(json.loads(self.rfile.read(n)))
+            self.send_response(200)
+            self.end_headers()
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), Sink)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        home = Path(os.environ["SKY_AMD_HOME"])
+        (home / "config.yaml").write_text(
+            f"logs:\n  endpoint: http://127.0.0.1:{srv.server_port}/\n")
+        from skypilot_amd import config as sky_config
+        sky_config.load(refresh=True)
+        from skypilot_amd.client import sdk
+        sdk.get(sdk.launch({"run": "echo shipped-$SKYPILOT_TASK_ID"},
+                           "ship-c"), timeout=60)
+        deadline = time.time() + 30
+        while time.time() < deadline and not received:
+            time.sleep(0.3)
+        assert received, "no logs shipped"
+        rec = next(r for r in received if "shipped-" in r["content"])
+        assert rec["status"] == "SUCCEEDED"
+        sdk.get(sdk.down("ship-c"))
+    finally:
+        srv.shutdown()
+        (home / "config.yaml").unlink(missing_ok=True)
+        from skypilot_amd import config as sky_config
+        sky_config.load(refresh=True)
