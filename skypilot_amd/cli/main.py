@@ -99,21 +99,25 @@ def exec_cmd(cluster, entrypoint, env, gpus, async_):
 
 @cli.command()
 @click.option("--refresh", "-r", is_flag=True)
-def status(refresh):
-    """Show clusters."""
-    records = sdk.get(sdk.status(refresh=refresh))
+@click.option("--all-workspaces", "-u", is_flag=True)
+def status(refresh, all_workspaces):
+    """Show clusters (active workspace; -u for every workspace)."""
+    records = sdk.get(sdk.status(refresh=refresh,
+                                 all_workspaces=all_workspaces))
     if not records:
         click.echo("No existing clusters.")
         return
-    fmt = "{:<16} {:<9} {:<22} {:<8}"
-    click.echo(fmt.format("NAME", "STATUS", "RESOURCES", "GPUS"))
+    fmt = "{:<16} {:<9} {:<22} {:<8} {:<10}"
+    click.echo(fmt.format("NAME", "STATUS", "RESOURCES", "GPUS",
+                          "WORKSPACE"))
     for r in records:
         h = r["handle"]
         res = f"{h.get('num_nodes', 1)}x local"
         if h.get("gpus_per_node"):
             res += f" MI355X:{h['gpus_per_node']}"
         click.echo(fmt.format(r["name"], r["status"], res,
-                              ",".join(map(str, h.get("gpu_ids", [])))))
+                              ",".join(map(str, h.get("gpu_ids", []))),
+                              r.get("workspace", "default")))
 
 
 @cli.command()
@@ -423,6 +427,14 @@ def volumes_create(name, size_gb):
 def volumes_delete(name):
     sdk.get(sdk.volumes_delete(name))
     click.echo(f"Volume {name} deleted.")
+
+
+@cli.command("workspace")
+def workspace_cmd():
+    """Show the active workspace (SKY_AMD_WORKSPACE env or config
+    `workspace:`)."""
+    from skypilot_amd import global_state
+    click.echo(global_state.current_workspace())
 
 
 @cli.group()

@@ -209,9 +209,10 @@ def exec(task, cluster_name: str, *, detach_run: bool = True) -> str:  # noqa: A
 
 
 def status(cluster_names: Optional[List[str]] = None,
-           refresh: bool = False) -> str:
+           refresh: bool = False, all_workspaces: bool = False) -> str:
     return _submit("status", {"cluster_names": cluster_names,
-                              "refresh": refresh})
+                              "refresh": refresh,
+                              "all_workspaces": all_workspaces})
 
 
 def start(cluster_name: str) -> str:

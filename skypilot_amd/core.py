@@ -22,8 +22,9 @@ def _get_record(cluster_name: str) -> Dict[str, Any]:
 
 
 def status(cluster_names: Optional[List[str]] = None,
-           refresh: bool = False) -> List[Dict[str, Any]]:
-    records = global_state.list_clusters()
+           refresh: bool = False,
+           all_workspaces: bool = False) -> List[Dict[str, Any]]:
+    records = global_state.list_clusters(all_workspaces=all_workspaces)
     if cluster_names:
         records = [r for r in records if r["name"] in cluster_names]
     if refresh:

@@ -40,7 +40,8 @@ CREATE TABLE IF NOT EXISTS clusters (
     user TEXT,
     autostop_idle_minutes INTEGER DEFAULT -1,
     autostop_down INTEGER DEFAULT 0,
-    to_down INTEGER DEFAULT 0
+    to_down INTEGER DEFAULT 0,
+    workspace TEXT DEFAULT 'default'
 );
 CREATE TABLE IF NOT EXISTS users (
     name TEXT PRIMARY KEY,
@@ -88,6 +89,10 @@ def _conn():
         conn.execute("PRAGMA busy_timeout=30000")
         conn.execute("PRAGMA journal_mode=WAL")
         conn.executescript(_SCHEMA)
+        cols = [r[1] for r in conn.execute("PRAGMA table_info(clusters)")]
+        if "workspace" not in cols:  # pre-workspace databases
+            conn.execute("ALTER TABLE clusters ADD COLUMN workspace TEXT "
+                         "DEFAULT 'default'")
         with conn:
             yield conn
     finally:
@@ -101,6 +106,20 @@ def set_request_user(name):
     """Thread-scoped identity for SHORT in-process request handlers
     (LONG handlers run in their own process and use the env var)."""
     _request_user.name = name
+
+
+def current_workspace() -> str:
+    """Active workspace (reference: sky workspaces — named scopes that
+    partition clusters/jobs): SKY_AMD_WORKSPACE env, else config
+    `workspace:`, else "default"."""
+    ws = os.environ.get("SKY_AMD_WORKSPACE")
+    if ws:
+        return ws
+    try:
+        from skypilot_amd import config as sky_config
+        return sky_config.get_nested(["workspace"], "default") or "default"
+    except Exception:  # noqa: BLE001
+        return "default"
 
 
 def current_user() -> str:
@@ -124,12 +143,12 @@ def add_or_update_cluster(name: str, status: str, handle: Dict[str, Any],
                   "VALUES (?,?)", (user, time.time()))
         c.execute(
             "INSERT INTO clusters "
-            "(name,status,handle,resources,launched_at,user)"
-            " VALUES (?,?,?,?,?,?) ON CONFLICT(name) DO UPDATE SET "
+            "(name,status,handle,resources,launched_at,user,workspace)"
+            " VALUES (?,?,?,?,?,?,?) ON CONFLICT(name) DO UPDATE SET "
             "status=excluded.status, handle=excluded.handle, "
             "resources=excluded.resources, launched_at=excluded.launched_at",
             (name, status, json.dumps(handle), json.dumps(resources),
-             launched_at, user))
+             launched_at, user, current_workspace()))
 
 
 def set_cluster_status(name: str, status: str) -> None:
@@ -148,7 +167,7 @@ def get_cluster(name: str) -> Optional[Dict[str, Any]]:
     with _DB_LOCK, _conn() as c:
         row = c.execute(
             "SELECT name,status,handle,resources,launched_at,"
-            "autostop_idle_minutes,autostop_down,user "
+            "autostop_idle_minutes,autostop_down,user,workspace "
             "FROM clusters WHERE name=?",
             (name,)).fetchone()
     if row is None:
@@ -166,15 +185,22 @@ def _row_to_cluster(row) -> Dict[str, Any]:
         "autostop_idle_minutes": row[5],
         "autostop_down": bool(row[6]),
         "user": row[7],
+        "workspace": row[8] if len(row) > 8 else "default",
     }
 
 
-def list_clusters() -> List[Dict[str, Any]]:
+def list_clusters(all_workspaces: bool = False) -> List[Dict[str, Any]]:
+    """Clusters in the active workspace (all with all_workspaces)."""
+    q = ("SELECT name,status,handle,resources,launched_at,"
+         "autostop_idle_minutes,autostop_down,user,workspace FROM "
+         "clusters ")
     with _DB_LOCK, _conn() as c:
-        rows = c.execute(
-            "SELECT name,status,handle,resources,launched_at,"
-            "autostop_idle_minutes,autostop_down,user FROM clusters "
-            "ORDER BY launched_at DESC").fetchall()
+        if all_workspaces:
+            rows = c.execute(q + "ORDER BY launched_at DESC").fetchall()
+        else:
+            rows = c.execute(
+                q + "WHERE workspace=? ORDER BY launched_at DESC",
+                (current_workspace(),)).fetchall()
     return [_row_to_cluster(r) for r in rows]
 
 

@@ -39,8 +39,10 @@ def exec_(task: Dict[str, Any], cluster_name: str,
 
 @register("status", SHORT)
 def status(cluster_names: Optional[List[str]] = None,
+           all_workspaces: bool = False,
            refresh: bool = False) -> List[Dict[str, Any]]:
-    return core.status(cluster_names, refresh)
+    return core.status(cluster_names, refresh,
+                       all_workspaces=all_workspaces)
 
 
 @register("start", LONG)

@@ -164,3 +164,26 @@ def test_metrics_gauges_and_leader(sky_env, client):
     out = subprocess.run([sys.executable, "-c", code, actual],
                          capture_output=True, text=True)
     assert out.stdout.strip() == "LOST", out
+
+
+def test_workspaces_scope_clusters(sky_env, client, monkeypatch):
+    """Clusters are scoped to the active workspace (reference: sky
+    workspaces); --all-workspaces sees everything."""
+    from skypilot_amd.client import sdk
+    monkeypatch.setenv("SKY_AMD_WORKSPACE", "team-a")
+    sdk.get(sdk.launch({"run": "true", "resources": {"cpus": 1}},
+                       "ws-a"), timeout=60)
+    monkeypatch.setenv("SKY_AMD_WORKSPACE", "team-b")
+    sdk.get(sdk.launch({"run": "true", "resources": {"cpus": 1}},
+                       "ws-b"), timeout=60)
+    names = {r["name"] for r in sdk.get(sdk.status())}
+    assert "ws-b" in names and "ws-a" not in names
+    monkeypatch.setenv("SKY_AMD_WORKSPACE", "team-a")
+    names = {r["name"] for r in sdk.get(sdk.status())}
+    assert "ws-a" in names and "ws-b" not in names
+    allr = sdk.get(sdk.status(all_workspaces=True))
+    ws = {r["name"]: r["workspace"] for r in allr}
+    assert ws["ws-a"] == "team-a" and ws["ws-b"] == "team-b"
+    sdk.get(sdk.down("ws-a"))
+    monkeypatch.setenv("SKY_AMD_WORKSPACE", "team-b")
+    sdk.get(sdk.down("ws-b"))
