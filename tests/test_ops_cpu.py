@@ -141,6 +141,11 @@ def test_all_example_yamls_parse():
     assert len(files) >= 7
     for f in files:
         cfg = _yaml.safe_load(f.read_text())
+        if "tasks" in cfg:  # pipeline / job-group form
+            for sub in cfg["tasks"]:
+                t = Task.from_yaml_config(sub)
+                assert t.run, f
+            continue
         t = Task.from_yaml_config(cfg)
         assert t.run or t.service, f
 
