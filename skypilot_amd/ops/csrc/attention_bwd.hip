@@ -565,6 +565,11 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_bwd_dkv_kernel32(
   }
 }
 
+extern "C" __global__ void attn_bwd_dkv_tr_kernel(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const float*, const float*, unsigned short*,
+    unsigned short*, int, int, int, int, float, int);
+
 extern "C" __global__ void attn_bwd_dkv_swapped_kernel(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const unsigned short*, const float*, const float*, unsigned short*,
@@ -594,7 +599,18 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
     const char* e = getenv("SKY_ATTN_DKV_SWAPPED");
     return e ? atoi(e) : 0;
   }();
-  if (dkv_swapped)
+  static const int dkv_tr = [] {
+    const char* e = getenv("SKY_ATTN_DKV_TR");
+    return e ? atoi(e) : 0;
+  }();
+  if (dkv_tr)
+    hipLaunchKernelGGL(attn_bwd_dkv_tr_kernel, dim3(S / BN, B * Hkv),
+                       dim3(256), 0, stream, (const unsigned short*)Q,
+                       (const unsigned short*)K, (const unsigned short*)V,
+                       (const unsigned short*)dO, lse, Dvec,
+                       (unsigned short*)dK, (unsigned short*)dV, B, S, Hq,
+                       Hkv, scale, causal ? 1 : 0);
+  else if (dkv_swapped)
     hipLaunchKernelGGL(attn_bwd_dkv_swapped_kernel, dim3(S / BN, B * Hkv),
                        dim3(256), 0, stream, (const unsigned short*)Q,
                        (const unsigned short*)K, (const unsigned short*)V,
@@ -965,5 +981,205 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_swapped_kernel
     }
     *(s16x4*)(krow_p + ct * 16 + lgrp * 4) = kv4;
     *(s16x4*)(vrow_p + ct * 16 + lgrp * 4) = vv4;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dK/dV with counted-wait transpose reads (guide §5.5 T3/T10): the
+// Q^T/dO^T staging (64 scalar scatter writes/thread/tile + 32 KB LDS)
+// is deleted; the dV/dK B-fragments come straight from the row-major
+// q_lds/do_lds via ds_read_b64_tr_b16 in a depth-2 software pipeline —
+// the i+1 fragment's four tr reads are in flight while fragment i's
+// MFMAs run, with an explicit s_waitcnt lgkmcnt(4) boundary (the naive
+// per-fragment lgkmcnt(0) version measured 8.6 -> 10.1 ms).  All LDS
+// ops inside the pipelined region go through inline asm so the
+// compiler cannot insert its own (miscounting) waits.
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_tr_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dK, unsigned short* __restrict__ dV, int B,
+    int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ unsigned short q_lds[BM * ATT_D];
+  __shared__ unsigned short do_lds[BM * ATT_D];
+  __shared__ unsigned short p_lds[BN * BM];   // PT
+  __shared__ unsigned short ds_lds[BN * BM];  // dST
+
+  const int kt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int kvh = bh % Hkv;
+  const int group = Hq / Hkv;
+  const int kvbase = kt * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int lrow = lane & 15;
+  const int lgrp = lane >> 4;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  s16x8 a_k[4], a_v[4];
+  {
+    const int kvrow = kvbase + 16 * w + lrow;
+    const unsigned short* ksrc = Kb + (long long)kvrow * kv_rowstride;
+    const unsigned short* vsrc = Vb + (long long)kvrow * kv_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      a_k[ks] = *(const s16x8*)(ksrc + ks * 32 + lgrp * 8);
+      a_v[ks] = *(const s16x8*)(vsrc + ks * 32 + lgrp * 8);
+    }
+  }
+
+  f32x4 dv_acc[8], dk_acc[8];
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+    dv_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dk_acc[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  const unsigned int q_base_a = (unsigned int)(unsigned long long)q_lds;
+  const unsigned int do_base_a = (unsigned int)(unsigned long long)do_lds;
+
+  for (int g = 0; g < group; ++g) {
+    const int qh = kvh * group + g;
+    const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+    const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+    const float* lse_b = lse + ((long long)b * Hq + qh) * S;
+    const float* dvec_b = Dvec + (long long)b * S * Hq + qh;
+
+    const int qt0 = causal ? kvbase / BM : 0;
+    for (int qt = qt0; qt < S / BM; ++qt) {
+      const int qbase = qt * BM;
+      __syncthreads();
+      // Stage Q, dO (vector writes only — no transposed scatter).
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int idx = tid + i * 256;
+        int row = idx >> 4, ch = idx & 15;
+        s16x8 qv = *(const s16x8*)(Qb + (long long)(qbase + row) * q_rowstride + ch * 8);
+        *(s16x8*)((char*)q_lds + swz(row * 256 + ch * 16, row)) = qv;
+        s16x8 dov = *(const s16x8*)(dOb + (long long)(qbase + row) * q_rowstride + ch * 8);
+        *(s16x8*)((char*)do_lds + swz(row * 256 + ch * 16, row)) = dov;
+      }
+      __syncthreads();
+
+      // ST = K Q^T, dPT = V dO^T (unchanged from the classic kernel).
+      f32x4 st[4], dpt[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        st[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+        dpt[ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          int qrow = ct * 16 + lrow;
+          s16x8 bq = *(const s16x8*)((char*)q_lds +
+                                     swz(qrow * 256 + (ks * 32 + lgrp * 8) * 2, qrow));
+          st[ct] = MFMA_BF16(as_bf16x8(a_k[ks]), as_bf16x8(bq), st[ct]);
+          s16x8 bdo = *(const s16x8*)((char*)do_lds +
+                                      swz(qrow * 256 + (ks * 32 + lgrp * 8) * 2, qrow));
+          dpt[ct] = MFMA_BF16(as_bf16x8(a_v[ks]), as_bf16x8(bdo), dpt[ct]);
+        }
+      }
+
+      const int my_kvrow = kvbase + 16 * w + lgrp * 4;  // + r
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        int qcol = qbase + ct * 16 + lrow;
+        float l = lse_b[qcol];
+        float dvv = dvec_b[(long long)qcol * Hq];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float sv = st[ct][r] * scale;
+          float pv;
+          if ((causal && qcol < my_kvrow + r) || l == -INFINITY)
+            pv = 0.f;
+          else
+            pv = __expf(sv - l);
+          int prow = 16 * w + lgrp * 4 + r;
+          int pcol = ct * 16 + lrow;
+          int off = swzP(prow * 128 + pcol * 2, prow);
+          *(unsigned short*)((char*)p_lds + off) = f2bf_trunc(pv);
+          *(unsigned short*)((char*)ds_lds + off) =
+              f2bf(pv * (dpt[ct][r] - dvv));
+        }
+      }
+      __syncthreads();
+
+      // A-fragments (PT/dST) via synchronous asm b128 reads so the
+      // compiler tracks zero outstanding ds ops in the counted region.
+      s16x8 pfrag[2], dsfrag[2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        int prow = 16 * w + lrow;
+        unsigned a_off = (unsigned)(unsigned long long)(char*)p_lds +
+            (unsigned)swzP(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow);
+        pfrag[ks] = ds_read_b128_sync(a_off);
+        unsigned d_off = (unsigned)(unsigned long long)(char*)ds_lds +
+            (unsigned)swzP(prow * 128 + (ks * 32 + lgrp * 8) * 2, prow);
+        dsfrag[ks] = ds_read_b128_sync(d_off);
+      }
+
+      // Counted-wait pipeline over the 16 (ct, ks) fragments: issue
+      // fragment i+1's four tr reads, wait until only those four remain
+      // outstanding, MFMA fragment i.
+      tr4 buf[2];
+      int q0 = 0 * 32 + lgrp * 8 + (lrow >> 2);  // ks=0 of ct=0
+      int din0 = 0 * 32 + (lrow & 3) * 8;
+      {
+        int ch0 = din0 >> 4;
+        int b0 = q0 * 256 + ((ch0 ^ (q0 & 7)) << 4) + (din0 & 15);
+        int q1 = q0 + 4;
+        int b1 = q1 * 256 + ((ch0 ^ (q1 & 7)) << 4) + (din0 & 15);
+        ds_tr4_issue(&buf[0], do_base_a + b0, do_base_a + b1,
+                     q_base_a + b0, q_base_a + b1);
+      }
+#pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const int ct = i >> 1, ks = i & 1;
+        if (i + 1 < 16) {
+          const int nct = (i + 1) >> 1, nks = (i + 1) & 1;
+          int nq0 = nks * 32 + lgrp * 8 + (lrow >> 2);
+          int ndin = nct * 32 + (lrow & 3) * 8;
+          int nch = ndin >> 4;
+          int nb0 = nq0 * 256 + ((nch ^ (nq0 & 7)) << 4) + (ndin & 15);
+          int nq1 = nq0 + 4;
+          int nb1 = nq1 * 256 + ((nch ^ (nq1 & 7)) << 4) + (ndin & 15);
+          ds_tr4_issue(&buf[(i + 1) & 1], do_base_a + nb0,
+                       do_base_a + nb1, q_base_a + nb0, q_base_a + nb1);
+          lgkm_wait4_bind(&buf[i & 1]);
+        } else {
+          lgkm_wait0_bind(&buf[i & 1]);
+        }
+        union { unsigned long long u[2]; s16x8 v; } dof, qf;
+        dof.u[0] = buf[i & 1].d[0];
+        dof.u[1] = buf[i & 1].d[1];
+        qf.u[0] = buf[i & 1].d[2];
+        qf.u[1] = buf[i & 1].d[3];
+        dv_acc[ct] = MFMA_BF16(as_bf16x8(pfrag[ks]), as_bf16x8(dof.v),
+                               dv_acc[ct]);
+        dk_acc[ct] = MFMA_BF16(as_bf16x8(dsfrag[ks]), as_bf16x8(qf.v),
+                               dk_acc[ct]);
+      }
+    }
+  }
+
+  unsigned short* dKb = dK + ((long long)b * S * Hkv + kvh) * ATT_D;
+  unsigned short* dVb = dV + ((long long)b * S * Hkv + kvh) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kvrow = kvbase + 16 * w + lgrp * 4 + r;
+    unsigned short* krow = dKb + (long long)kvrow * kv_rowstride;
+    unsigned short* vrow = dVb + (long long)kvrow * kv_rowstride;
+#pragma unroll
+    for (int ct = 0; ct < 8; ++ct) {
+      krow[ct * 16 + lrow] = f2bf(dk_acc[ct][r] * scale);
+      vrow[ct * 16 + lrow] = f2bf(dv_acc[ct][r]);
+    }
   }
 }

@@ -102,6 +102,61 @@ __device__ __forceinline__ s16x8 ds_tr_b16_pair(unsigned int a0,
   return cvt.v;
 }
 
+// Split-phase tr reads for counted-wait pipelines (guide §5.5 T3):
+// ds_tr4_issue fires FOUR ds_read_b64_tr_b16 without waiting;
+// lgkm_wait4_bind blocks until only the most recent 4 lgkm ops remain
+// outstanding and binds the dependency to the destination registers so
+// the compiler cannot hoist consumers above the wait.  All LDS traffic
+// in the pipelined region must go through these helpers (a stray
+// compiler ds op breaks the count).
+struct tr4 {
+  unsigned long long d[4];
+};
+__device__ __forceinline__ void ds_tr4_issue(tr4* t, unsigned a0,
+                                             unsigned a1, unsigned a2,
+                                             unsigned a3) {
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n\t"
+      "ds_read_b64_tr_b16 %1, %5\n\t"
+      "ds_read_b64_tr_b16 %2, %6\n\t"
+      "ds_read_b64_tr_b16 %3, %7"
+      : "=v"(t->d[0]), "=v"(t->d[1]), "=v"(t->d[2]), "=v"(t->d[3])
+      : "v"(a0), "v"(a1), "v"(a2), "v"(a3)
+      : "memory");
+}
+__device__ __forceinline__ void lgkm_wait4_bind(tr4* t) {
+  asm volatile("s_waitcnt lgkmcnt(4)"
+               : "+v"(t->d[0]), "+v"(t->d[1]), "+v"(t->d[2]),
+                 "+v"(t->d[3])
+               :
+               : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+}
+__device__ __forceinline__ void lgkm_wait0_bind(tr4* t) {
+  asm volatile("s_waitcnt lgkmcnt(0)"
+               : "+v"(t->d[0]), "+v"(t->d[1]), "+v"(t->d[2]),
+                 "+v"(t->d[3])
+               :
+               : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+}
+
+// Synchronous b128 LDS read through inline asm: used before a
+// counted-wait region so NO compiler-tracked ds op remains outstanding
+// (the compiler would otherwise insert its own s_waitcnt mid-region and
+// count our in-flight tr reads).
+__device__ __forceinline__ s16x8 ds_read_b128_sync(unsigned int addr) {
+  f32x4 d;
+  asm volatile("ds_read_b128 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(d)
+               : "v"(addr)
+               : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  union { f32x4 f; s16x8 s; } u;
+  u.f = d;
+  return u.s;
+}
+
 // bf16 <-> f32 via bit ops (we deliberately avoid __hip_bfloat16 so these
 // headers stay independent of HIP half/bf16 operator macros).
 __device__ __forceinline__ float bf2f(unsigned short h) {
