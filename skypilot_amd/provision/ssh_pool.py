@@ -44,6 +44,24 @@ def load_pools(path: Optional[str] = None) -> Dict[str, Any]:
         return yaml.safe_load(f) or {}
 
 
+def ensure_keypair() -> str:
+    """Framework SSH keypair (reference: sky/authentication.py — an
+    ed25519 pair generated once under the state dir; its pubkey is what
+    gets installed on pool hosts).  Used for any host that does not set
+    identity_file."""
+    import subprocess
+    d = global_state.root_dir() / "ssh"
+    d.mkdir(parents=True, exist_ok=True)
+    key = d / "sky-key"
+    if not key.exists():
+        subprocess.run(
+            ["ssh-keygen", "-t", "ed25519", "-N", "", "-q", "-f",
+             str(key), "-C", "skypilot-amd"],
+            check=True, capture_output=True)
+        key.chmod(0o600)
+    return str(key)
+
+
 def parse_hosts(pool: Optional[str] = None) -> List[Dict[str, Any]]:
     pools = load_pools()
     if not pools:
@@ -53,10 +71,12 @@ def parse_hosts(pool: Optional[str] = None) -> List[Dict[str, Any]]:
     hosts = []
     for h in cfg.get("hosts", []):
         if isinstance(h, str):
-            hosts.append({"ip": h, "user": "root", "gpus": 8})
+            hosts.append({"ip": h, "user": "root", "gpus": 8,
+                          "identity_file": ensure_keypair()})
         else:
             hosts.append({"ip": h["ip"], "user": h.get("user", "root"),
-                          "identity_file": h.get("identity_file"),
+                          "identity_file": (h.get("identity_file")
+                                            or ensure_keypair()),
                           "port": int(h.get("port", 22)),
                           "gpus": int(h.get("gpus", 8))})
     return hosts
