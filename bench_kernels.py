@@ -98,6 +98,20 @@ def main():
     gb = 2 * x.numel() * 2 / 1e9
     print(f"rope         {t*1e3:8.3f} ms  {gb/t:7.0f} GB/s")
 
+    # Decode GEMV (n=1, wq shape) — weight-BW roofline kernel
+    xg = torch.randn(1, 4096, device=dev).bfloat16()
+    wg = (torch.randn(4096, 4096, device=dev) * 0.02).bfloat16()
+    t = timeit(lambda: C.skinny_gemm(xg, wg))
+    print(f"skinny_gemv  {t*1e6:8.1f} us  {wg.numel()*2/t/1e9:7.0f} GB/s "
+          f"(n=1 4096x4096)")
+
+    # Fused residual+rmsnorm (decode shape)
+    xr = torch.randn(1, 4096, device=dev).bfloat16()
+    rr = torch.randn(1, 4096, device=dev).bfloat16()
+    wr = torch.randn(4096, device=dev).bfloat16()
+    t = timeit(lambda: C.rmsnorm_res(xr, rr, wr, 1e-5))
+    print(f"rmsnorm_res  {t*1e6:8.1f} us  (decode n=1)")
+
 
 if __name__ == "__main__":
     main()
