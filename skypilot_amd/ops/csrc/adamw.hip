@@ -66,3 +66,84 @@ extern "C" void adamw_launch(void* p_bf16, float* p_master, const void* g_bf16,
                      (const unsigned short*)g_bf16, m, v, n, lr, beta1, beta2,
                      eps, wd, bc1, bc2, grad_scale);
 }
+
+// Multi-tensor variant: ONE launch per optimizer step (the per-tensor
+// loop above costs ~250 launches/step for an 8B model = 41 ms vs the
+// ~16 ms state-traffic bound).  The host packs per-tensor pointers +
+// a chunk->(tensor, offset) map into device buffers once (pointers are
+// stable across steps); each workgroup owns one 64 Ki-element chunk.
+#define ADAMW_CHUNK 65536LL
+
+extern "C" __global__ __launch_bounds__(256) void adamw_mt_kernel(
+    const unsigned long long* __restrict__ ptrs,  // [T][5] p,mp,g,m,v
+    const float* __restrict__ wd_arr,             // [T]
+    const long long* __restrict__ sizes,          // [T]
+    const int* __restrict__ chunk_tensor,         // [C]
+    const long long* __restrict__ chunk_start,    // [C]
+    float lr, float beta1, float beta2, float eps, float bc1, float bc2,
+    float grad_scale) {
+  const int t = chunk_tensor[blockIdx.x];
+  const long long base = chunk_start[blockIdx.x];
+  const long long n = sizes[t];
+  const long long end =
+      (base + ADAMW_CHUNK < n) ? base + ADAMW_CHUNK : n;
+  unsigned short* p_bf16 = (unsigned short*)ptrs[5 * t];
+  float* p_master = (float*)ptrs[5 * t + 1];
+  const unsigned short* g_bf16 = (const unsigned short*)ptrs[5 * t + 2];
+  float* m = (float*)ptrs[5 * t + 3];
+  float* v = (float*)ptrs[5 * t + 4];
+  const float wd = wd_arr[t];
+  for (long long i = base + (long long)threadIdx.x * 4; i < end;
+       i += 256 * 4) {
+    if (i + 4 <= end) {
+      f32x4 mv = *(f32x4*)(m + i);
+      f32x4 vv = *(f32x4*)(v + i);
+      f32x4 pv = *(f32x4*)(p_master + i);
+      s16x4 gv = *(s16x4*)(g_bf16 + i);
+      s16x4 pb;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float g = bf2f((unsigned short)gv[k]) * grad_scale;
+        float mm = beta1 * mv[k] + (1.f - beta1) * g;
+        float vvk = beta2 * vv[k] + (1.f - beta2) * g * g;
+        float p = pv[k];
+        p -= lr * ((mm / bc1) / (sqrtf(vvk / bc2) + eps) + wd * p);
+        mv[k] = mm;
+        vv[k] = vvk;
+        pv[k] = p;
+        pb[k] = (short)f2bf(p);
+      }
+      *(f32x4*)(m + i) = mv;
+      *(f32x4*)(v + i) = vv;
+      *(f32x4*)(p_master + i) = pv;
+      *(s16x4*)(p_bf16 + i) = pb;
+    } else {
+      for (long long j = i; j < end; ++j) {
+        float g = bf2f(g_bf16[j]) * grad_scale;
+        float mm = beta1 * m[j] + (1.f - beta1) * g;
+        float vvk = beta2 * v[j] + (1.f - beta2) * g * g;
+        float p = p_master[j];
+        p -= lr * ((mm / bc1) / (sqrtf(vvk / bc2) + eps) + wd * p);
+        m[j] = mm;
+        v[j] = vvk;
+        p_master[j] = p;
+        p_bf16[j] = f2bf(p);
+      }
+    }
+  }
+}
+
+extern "C" void adamw_mt_launch(const void* ptrs, const void* wd_arr,
+                                const void* sizes, const void* chunk_tensor,
+                                const void* chunk_start, long long n_chunks,
+                                float lr, float beta1, float beta2,
+                                float eps, int step, float grad_scale,
+                                hipStream_t stream) {
+  float bc1 = 1.f - powf(beta1, (float)step);
+  float bc2 = 1.f - powf(beta2, (float)step);
+  hipLaunchKernelGGL(adamw_mt_kernel, dim3((unsigned)n_chunks), dim3(256),
+                     0, stream, (const unsigned long long*)ptrs,
+                     (const float*)wd_arr, (const long long*)sizes,
+                     (const int*)chunk_tensor, (const long long*)chunk_start,
+                     lr, beta1, beta2, eps, bc1, bc2, grad_scale);
+}

@@ -42,6 +42,9 @@ void trb16_probe_launch(float*, int, hipStream_t);
 void permlane_probe_launch(float*, hipStream_t);
 void skinny_gemm_launch(const void*, const void*, void*, int, int, int,
                         hipStream_t);
+void adamw_mt_launch(const void*, const void*, const void*, const void*,
+                     const void*, long long, float, float, float, float,
+                     int, float, hipStream_t);
 void rmsnorm_res_launch(const void*, const void*, const void*, void*, void*,
                         int, int, float, hipStream_t);
 void rope_kvwrite_launch(const void*, void*, void*, void*, const void*,
@@ -192,6 +195,20 @@ torch::Tensor attn_decode(torch::Tensor Q, torch::Tensor Kc,
   return O;
 }
 
+void adamw_step_mt(torch::Tensor ptrs, torch::Tensor wd_arr,
+                   torch::Tensor sizes, torch::Tensor chunk_tensor,
+                   torch::Tensor chunk_start, double lr, double beta1,
+                   double beta2, double eps, int64_t step,
+                   double grad_scale) {
+  // One launch for the whole parameter list (adamw.hip multi-tensor).
+  CHECK_GPU(ptrs); CHECK_GPU(chunk_tensor);
+  adamw_mt_launch(ptrs.data_ptr(), wd_arr.data_ptr(), sizes.data_ptr(),
+                  chunk_tensor.data_ptr(), chunk_start.data_ptr(),
+                  (long long)chunk_tensor.numel(), (float)lr, (float)beta1,
+                  (float)beta2, (float)eps, (int)step, (float)grad_scale,
+                  cur_stream());
+}
+
 torch::Tensor skinny_gemm(torch::Tensor X, torch::Tensor W) {
   // Y[N,O] = X[N,I] @ W[O,I]^T (decode GEMV; see skinny_gemm.hip)
   CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
@@ -310,6 +327,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_decode", &attn_decode);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("adamw_step_mt", &adamw_step_mt);
   m.def("rmsnorm_res", &rmsnorm_res);
   m.def("rope_kvwrite", &rope_kvwrite);
   m.def("swiglu_fwd", &swiglu_fwd);

@@ -1,7 +1,11 @@
 """Fused AdamW with fp32 master weights for bf16 training on MI355X.
 
 The update runs entirely in the hand-written CDNA4 kernel
-(`ops/csrc/adamw.hip`) — one launch per tensor, looped in C++.  Master
+(`ops/csrc/adamw.hip`) — ONE launch per step via the multi-tensor
+chunk kernel (per-tensor launches measured 41 ms/step vs the ~16 ms
+state-traffic bound for 8B params); device-side metadata (pointers,
+sizes, chunk map) is built once and reused while pointers are
+stable.  Master
 weights and both moments are fp32; model params stay bf16 (refreshed
 from the master copy each step inside the same kernel).  Gradient
 averaging for DDP is folded in via ``grad_scale``.
@@ -32,6 +36,7 @@ class FusedAdamW:
         self.decay_mask = [
             (p.dim() >= 2) if decay_2d_only else True for p in self.params
         ]
+        self._mt = None  # cached device metadata for the one-launch path
 
     def _grads(self):
         gs = []
@@ -44,12 +49,57 @@ class FusedAdamW:
             gs.append(g)
         return gs
 
+    _CHUNK = 65536
+
+    def _mt_meta(self, grads):
+        key = tuple(g.data_ptr() for g in grads)
+        if self._mt is not None and self._mt["key"] == key:
+            return self._mt
+        dev = self.params[0].device
+        T = len(self.params)
+        ptrs = torch.empty(T, 5, dtype=torch.int64)
+        sizes = torch.empty(T, dtype=torch.int64)
+        wd = torch.empty(T, dtype=torch.float32)
+        ct, cs = [], []
+        aligned = True
+        for i, (p, mp, g, m, v, dm) in enumerate(
+                zip(self.params, self.master, grads, self.exp_avg,
+                    self.exp_avg_sq, self.decay_mask)):
+            ptrs[i] = torch.tensor([p.data_ptr(), mp.data_ptr(),
+                                    g.data_ptr(), m.data_ptr(),
+                                    v.data_ptr()], dtype=torch.int64)
+            n = p.numel()
+            sizes[i] = n
+            wd[i] = self.wd if dm else 0.0
+            aligned &= (g.data_ptr() % 8 == 0 and p.data_ptr() % 8 == 0)
+            for off in range(0, n, self._CHUNK):
+                ct.append(i)
+                cs.append(off)
+        if not aligned:
+            return None  # odd bucket view: per-tensor fallback
+        self._mt = {
+            "key": key,
+            "ptrs": ptrs.to(dev),
+            "wd": wd.to(dev),
+            "sizes": sizes.to(dev),
+            "ct": torch.tensor(ct, dtype=torch.int32, device=dev),
+            "cs": torch.tensor(cs, dtype=torch.int64, device=dev),
+        }
+        return self._mt
+
     @torch.no_grad()
     def step(self, grad_scale: float = 1.0):
         self.step_count += 1
         grads = self._grads()
         if self.params[0].is_cuda:
             C = ops.native()
+            meta = self._mt_meta(grads)
+            if meta is not None:
+                C.adamw_step_mt(meta["ptrs"], meta["wd"], meta["sizes"],
+                                meta["ct"], meta["cs"], self.lr,
+                                self.betas[0], self.betas[1], self.eps,
+                                self.step_count, grad_scale)
+                return
             C.adamw_step([p.data for p in self.params], self.master, grads,
                          self.exp_avg, self.exp_avg_sq, self.lr,
                          self.betas[0], self.betas[1], self.eps, self.wd,

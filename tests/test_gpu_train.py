@@ -36,3 +36,40 @@ def test_train_loss_decreases_with_all_kernels():
     tok, tgt = tr.synthetic_batch()
     losses = [tr.train_step((tok, tgt)) for _ in range(10)]
     assert losses[-1] < losses[0], losses  # monotone-ish decrease on random data
+
+
+@pytest.mark.gpu
+def test_adamw_mt_matches_per_tensor():
+    """One-launch multi-tensor AdamW must match the per-tensor kernel
+    bitwise (identical per-element expression)."""
+    import torch
+    from skypilot_amd import ops
+    from skypilot_amd.train.optim import FusedAdamW
+    torch.manual_seed(11)
+    shapes = [(4096, 256), (333,), (128, 64), (70000,)]
+    pa = [torch.randn(s, device="cuda").bfloat16().requires_grad_(True)
+          for s in shapes]
+    pb = [p.detach().clone().requires_grad_(True) for p in pa]
+    grads = [torch.randn(s, device="cuda").bfloat16() * 0.1
+             for s in shapes]
+    oa = FusedAdamW(pa, lr=1e-3)
+    ob = FusedAdamW(pb, lr=1e-3)
+    C = ops.native()
+    for it in range(3):
+        for p, g in zip(pa, grads):
+            p.grad = g
+        for p, g in zip(pb, grads):
+            p.grad = g
+        oa.step()  # multi-tensor path
+        # force per-tensor path for b
+        ob.step_count += 1
+        C.adamw_step([p.data for p in ob.params], ob.master,
+                     [p.grad for p in ob.params], ob.exp_avg,
+                     ob.exp_avg_sq, ob.lr, ob.betas[0], ob.betas[1],
+                     ob.eps, ob.wd, ob.step_count, 1.0, ob.decay_mask)
+    for a, b in zip(pa, pb):
+        assert torch.equal(a, b)
+    for a, b in zip(oa.master, ob.master):
+        assert torch.equal(a, b)
+    for a, b in zip(oa.exp_avg_sq, ob.exp_avg_sq):
+        assert torch.equal(a, b)
