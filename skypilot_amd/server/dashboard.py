@@ -21,6 +21,26 @@ def _table(title: str, headers: List[str], rows: List[List[Any]]) -> str:
             f"<tbody>{body}</tbody></table>")
 
 
+def _sparkline(values, width=560, height=60) -> str:
+    """Inline SVG sparkline (no JS, no chart deps)."""
+    if not values or max(values) == 0:
+        return "<p class=empty>no activity yet</p>"
+    mx = max(values)
+    n = len(values)
+    bar_w = max(2, width // n - 1)
+    bars = []
+    for i, v in enumerate(values):
+        h = int(v / mx * (height - 4))
+        bars.append(
+            f'<rect x="{i * (bar_w + 1)}" y="{height - h}" '
+            f'width="{bar_w}" height="{h}" fill="#e8443a"/>')
+    return (f'<svg width="{width}" height="{height + 14}">'
+            + "".join(bars)
+            + f'<text x="0" y="{height + 12}" fill="#666" '
+              f'font-size="10">last {n} minutes; peak {mx}/min</text>'
+            "</svg>")
+
+
 def render() -> str:
     from skypilot_amd import global_state
     from skypilot_amd.jobs import state as jobs_state
@@ -69,6 +89,15 @@ def render() -> str:
                      time.strftime("%H:%M:%S",
                                    time.localtime(r["created_at"]))])
 
+    # requests-per-minute time series (reference dashboard charts)
+    now = time.time()
+    buckets = [0] * 30
+    for r in rdb.list_requests(limit=1000):
+        age_min = int((now - (r["created_at"] or now)) / 60)
+        if 0 <= age_min < 30:
+            buckets[29 - age_min] += 1
+    spark = _sparkline(buckets)
+
     # teardown history (reference dashboard: cluster history view)
     hist = [[h["name"],
              time.strftime("%m-%d %H:%M",
@@ -96,6 +125,7 @@ th {{ background:#222; color:#e8443a; }}
         jobs)}
 {_table("Services", ["name", "status", "ready", "endpoint"], services)}
 {_table("Pool GPUs", ["idx", "model", "HBM", "numa", "used by"], gpus)}
+<h2>Request activity</h2>{spark}
 {_table("Recent API requests",
         ["id", "type", "user", "status", "duration", "at"], reqs)}
 {_table("Cluster history", ["name", "launched", "torn down", "lifetime"],
