@@ -36,21 +36,64 @@ extern "C" __global__ __launch_bounds__(256) void attn_decode_kernel(
   const unsigned short* Kb = Kc + ((long long)b * S_max * Hkv + kvh) * DEC_D;
   const unsigned short* Vb = Vc + ((long long)b * S_max * Hkv + kvh) * DEC_D;
 
-  // Online softmax over this wave's KV slice; o accumulator: 2 d per lane.
-  float m = -INFINITY, l = 0.f, o0 = 0.f, o1 = 0.f;
-  for (int s = w; s < kv_len; s += 4) {
+  // Online softmax over this wave's KV slice; o accumulator: 2 d per
+  // lane.  TWO independent chains per wave (even/odd rows of the
+  // slice, merged below): the per-row dot -> wave-reduce -> exp ->
+  // rescale dependency chain is the latency bound at long kv, and the
+  // second chain fills its stalls.
+  float m_c[2] = {-INFINITY, -INFINITY};
+  float l_c[2] = {0.f, 0.f}, o0_c[2] = {0.f, 0.f}, o1_c[2] = {0.f, 0.f};
+  int s = w;
+  for (; s + 4 < kv_len; s += 8) {
+    const unsigned short* krow0 = Kb + s * kv_rowstride;
+    const unsigned short* krow1 = Kb + (s + 4) * kv_rowstride;
+    float d0 = q0 * bf2f(krow0[lane]) + q1 * bf2f(krow0[lane + 64]);
+    float d1 = q0 * bf2f(krow1[lane]) + q1 * bf2f(krow1[lane + 64]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      d0 += __shfl_xor(d0, off, 64);
+      d1 += __shfl_xor(d1, off, 64);
+    }
+    const unsigned short* vrow0 = Vb + s * kv_rowstride;
+    const unsigned short* vrow1 = Vb + (s + 4) * kv_rowstride;
+    float mn0 = fmaxf(m_c[0], d0);
+    float mn1 = fmaxf(m_c[1], d1);
+    float c0 = (m_c[0] == -INFINITY) ? 0.f : __expf(m_c[0] - mn0);
+    float c1 = (m_c[1] == -INFINITY) ? 0.f : __expf(m_c[1] - mn1);
+    float e0 = __expf(d0 - mn0);
+    float e1 = __expf(d1 - mn1);
+    o0_c[0] = o0_c[0] * c0 + e0 * bf2f(vrow0[lane]);
+    o1_c[0] = o1_c[0] * c0 + e0 * bf2f(vrow0[lane + 64]);
+    o0_c[1] = o0_c[1] * c1 + e1 * bf2f(vrow1[lane]);
+    o1_c[1] = o1_c[1] * c1 + e1 * bf2f(vrow1[lane + 64]);
+    l_c[0] = l_c[0] * c0 + e0;
+    l_c[1] = l_c[1] * c1 + e1;
+    m_c[0] = mn0;
+    m_c[1] = mn1;
+  }
+  for (; s < kv_len; s += 4) {
     const unsigned short* krow = Kb + s * kv_rowstride;
-    // dot(q, k): each lane contributes 2 elements, wave-reduce.
     float dot = q0 * bf2f(krow[lane]) + q1 * bf2f(krow[lane + 64]);
     dot = wave_reduce_sum(dot);
-    float m_new = fmaxf(m, dot);
-    float corr = (m == -INFINITY) ? 0.f : __expf(m - m_new);
+    float m_new = fmaxf(m_c[0], dot);
+    float corr = (m_c[0] == -INFINITY) ? 0.f : __expf(m_c[0] - m_new);
     float e = __expf(dot - m_new);
     const unsigned short* vrow = Vb + s * kv_rowstride;
-    o0 = o0 * corr + e * bf2f(vrow[lane]);
-    o1 = o1 * corr + e * bf2f(vrow[lane + 64]);
-    l = l * corr + e;
-    m = m_new;
+    o0_c[0] = o0_c[0] * corr + e * bf2f(vrow[lane]);
+    o1_c[0] = o1_c[0] * corr + e * bf2f(vrow[lane + 64]);
+    l_c[0] = l_c[0] * corr + e;
+    m_c[0] = m_new;
+  }
+  // merge the two chains
+  float m = fmaxf(m_c[0], m_c[1]);
+  float cc0 = (m_c[0] == -INFINITY) ? 0.f : __expf(m_c[0] - m);
+  float cc1 = (m_c[1] == -INFINITY) ? 0.f : __expf(m_c[1] - m);
+  float l = l_c[0] * cc0 + l_c[1] * cc1;
+  float o0 = o0_c[0] * cc0 + o0_c[1] * cc1;
+  float o1 = o1_c[0] * cc0 + o1_c[1] * cc1;
+  if (m == -INFINITY) {
+    l = 0.f;
+    o0 = o1 = 0.f;
   }
 
   // Merge the 4 wave partials via LDS.
