@@ -41,35 +41,40 @@ extern "C" __global__ __launch_bounds__(256) void attn_decode_kernel(
   // slice, merged below): the per-row dot -> wave-reduce -> exp ->
   // rescale dependency chain is the latency bound at long kv, and the
   // second chain fills its stalls.
-  float m_c[2] = {-INFINITY, -INFINITY};
-  float l_c[2] = {0.f, 0.f}, o0_c[2] = {0.f, 0.f}, o1_c[2] = {0.f, 0.f};
-  int s = w;
-  for (; s + 4 < kv_len; s += 8) {
-    const unsigned short* krow0 = Kb + s * kv_rowstride;
-    const unsigned short* krow1 = Kb + (s + 4) * kv_rowstride;
-    float d0 = q0 * bf2f(krow0[lane]) + q1 * bf2f(krow0[lane + 64]);
-    float d1 = q0 * bf2f(krow1[lane]) + q1 * bf2f(krow1[lane + 64]);
+#define DEC_CHAINS 4
+  float m_c[DEC_CHAINS], l_c[DEC_CHAINS];
+  float o0_c[DEC_CHAINS], o1_c[DEC_CHAINS];
 #pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      d0 += __shfl_xor(d0, off, 64);
-      d1 += __shfl_xor(d1, off, 64);
+  for (int c = 0; c < DEC_CHAINS; ++c) {
+    m_c[c] = -INFINITY;
+    l_c[c] = 0.f;
+    o0_c[c] = 0.f;
+    o1_c[c] = 0.f;
+  }
+  int s = w;
+  for (; s + 4 * (DEC_CHAINS - 1) < kv_len; s += 4 * DEC_CHAINS) {
+    float d[DEC_CHAINS];
+#pragma unroll
+    for (int c = 0; c < DEC_CHAINS; ++c) {
+      const unsigned short* krow = Kb + (s + 4 * c) * kv_rowstride;
+      d[c] = q0 * bf2f(krow[lane]) + q1 * bf2f(krow[lane + 64]);
     }
-    const unsigned short* vrow0 = Vb + s * kv_rowstride;
-    const unsigned short* vrow1 = Vb + (s + 4) * kv_rowstride;
-    float mn0 = fmaxf(m_c[0], d0);
-    float mn1 = fmaxf(m_c[1], d1);
-    float c0 = (m_c[0] == -INFINITY) ? 0.f : __expf(m_c[0] - mn0);
-    float c1 = (m_c[1] == -INFINITY) ? 0.f : __expf(m_c[1] - mn1);
-    float e0 = __expf(d0 - mn0);
-    float e1 = __expf(d1 - mn1);
-    o0_c[0] = o0_c[0] * c0 + e0 * bf2f(vrow0[lane]);
-    o1_c[0] = o1_c[0] * c0 + e0 * bf2f(vrow0[lane + 64]);
-    o0_c[1] = o0_c[1] * c1 + e1 * bf2f(vrow1[lane]);
-    o1_c[1] = o1_c[1] * c1 + e1 * bf2f(vrow1[lane + 64]);
-    l_c[0] = l_c[0] * c0 + e0;
-    l_c[1] = l_c[1] * c1 + e1;
-    m_c[0] = mn0;
-    m_c[1] = mn1;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+#pragma unroll
+      for (int c = 0; c < DEC_CHAINS; ++c)
+        d[c] += __shfl_xor(d[c], off, 64);
+#pragma unroll
+    for (int c = 0; c < DEC_CHAINS; ++c) {
+      const unsigned short* vrow = Vb + (s + 4 * c) * kv_rowstride;
+      float mn = fmaxf(m_c[c], d[c]);
+      float cr = (m_c[c] == -INFINITY) ? 0.f : __expf(m_c[c] - mn);
+      float e = __expf(d[c] - mn);
+      o0_c[c] = o0_c[c] * cr + e * bf2f(vrow[lane]);
+      o1_c[c] = o1_c[c] * cr + e * bf2f(vrow[lane + 64]);
+      l_c[c] = l_c[c] * cr + e;
+      m_c[c] = mn;
+    }
   }
   for (; s < kv_len; s += 4) {
     const unsigned short* krow = Kb + s * kv_rowstride;
@@ -84,16 +89,17 @@ extern "C" __global__ __launch_bounds__(256) void attn_decode_kernel(
     l_c[0] = l_c[0] * corr + e;
     m_c[0] = m_new;
   }
-  // merge the two chains
-  float m = fmaxf(m_c[0], m_c[1]);
-  float cc0 = (m_c[0] == -INFINITY) ? 0.f : __expf(m_c[0] - m);
-  float cc1 = (m_c[1] == -INFINITY) ? 0.f : __expf(m_c[1] - m);
-  float l = l_c[0] * cc0 + l_c[1] * cc1;
-  float o0 = o0_c[0] * cc0 + o0_c[1] * cc1;
-  float o1 = o1_c[0] * cc0 + o1_c[1] * cc1;
-  if (m == -INFINITY) {
-    l = 0.f;
-    o0 = o1 = 0.f;
+  // merge the chains
+  float m = -INFINITY;
+#pragma unroll
+  for (int c = 0; c < DEC_CHAINS; ++c) m = fmaxf(m, m_c[c]);
+  float l = 0.f, o0 = 0.f, o1 = 0.f;
+#pragma unroll
+  for (int c = 0; c < DEC_CHAINS; ++c) {
+    float cc = (m_c[c] == -INFINITY) ? 0.f : __expf(m_c[c] - m);
+    l += l_c[c] * cc;
+    o0 += o0_c[c] * cc;
+    o1 += o1_c[c] * cc;
   }
 
   // Merge the 4 wave partials via LDS.
