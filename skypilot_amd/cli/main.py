@@ -123,6 +123,7 @@ def status(refresh, all_workspaces):
 @cli.command()
 @click.argument("clusters", nargs=-1, required=True)
 def start(clusters):
+    """Restart a stopped cluster (same GPU lease)."""
     for cl in clusters:
         sdk.get(sdk.start(cl))
         click.echo(f"Cluster {cl} started.")
@@ -131,6 +132,7 @@ def start(clusters):
 @cli.command()
 @click.argument("clusters", nargs=-1, required=True)
 def stop(clusters):
+    """Stop a cluster (keep its state dir for sky start)."""
     for cl in clusters:
         sdk.get(sdk.stop(cl))
         click.echo(f"Cluster {cl} stopped.")
@@ -140,6 +142,7 @@ def stop(clusters):
 @click.argument("clusters", nargs=-1, required=True)
 @click.option("--yes", "-y", is_flag=True)
 def down(clusters, yes):
+    """Tear the cluster down and release its GPUs."""
     for cl in clusters:
         sdk.get(sdk.down(cl))
         click.echo(f"Cluster {cl} terminated.")
@@ -150,6 +153,7 @@ def down(clusters, yes):
 @click.option("--idle-minutes", "-i", type=int, required=True)
 @click.option("--down", is_flag=True)
 def autostop(cluster, idle_minutes, down):
+    """Stop the cluster after N idle minutes (-i)."""
     sdk.get(sdk.autostop(cluster, idle_minutes, down))
     click.echo(f"Autostop set on {cluster}: {idle_minutes}m "
                f"({'down' if down else 'stop'})")
@@ -158,6 +162,7 @@ def autostop(cluster, idle_minutes, down):
 @cli.command()
 @click.argument("cluster")
 def queue(cluster):
+    """Show a cluster's job queue."""
     jobs = sdk.get(sdk.queue(cluster))
     fmt = "{:<6} {:<18} {:<12} {:<10}"
     click.echo(fmt.format("ID", "NAME", "STATUS", "GPUS"))
@@ -172,6 +177,7 @@ def queue(cluster):
 @click.argument("job_id", type=int, required=False)
 @click.option("--no-follow", is_flag=True)
 def logs(cluster, job_id, no_follow):
+    """Stream a job's logs."""
     sdk.tail_logs(cluster, job_id, follow=not no_follow)
 
 
@@ -180,17 +186,20 @@ def logs(cluster, job_id, no_follow):
 @click.argument("job_ids", nargs=-1, type=int)
 @click.option("--all", "all_jobs", is_flag=True)
 def cancel(cluster, job_ids, all_jobs):
+    """Cancel queued/running jobs on a cluster."""
     n = sdk.get(sdk.cancel(cluster, list(job_ids) or None, all_jobs))
     click.echo(f"Cancelled {n} job(s).")
 
 
 @cli.command()
 def check():
+    """Verify pool credentials/capabilities (local, ssh, k8s)."""
     _print_result(sdk.get(sdk.check()))
 
 
 @cli.command("show-gpus")
 def show_gpus():
+    """List pool GPUs, topology and current leases."""
     gpus = sdk.get(sdk.show_gpus())
     fmt = "{:<6} {:<10} {:<10} {:<6} {:<12}"
     click.echo(fmt.format("GPU", "NAME", "MEM_GB", "NUMA", "USED_BY"))
