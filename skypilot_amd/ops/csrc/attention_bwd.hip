@@ -653,6 +653,9 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
 // consumes dS^T as shuffled B-fragments — the ds_lds roundtrip and one
 // of the two barriers per tile disappear.
 // ---------------------------------------------------------------------------
+// (launch_bounds(256,3) forces 168 VGPRs + 100 B/lane scratch for 3
+// waves/SIMD: measured 158 vs 169 TF/s — spills cost more than the
+// extra wave hides.)
 extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dq_swapped_kernel(
     const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
