@@ -42,6 +42,8 @@ void trb16_probe_launch(float*, int, hipStream_t);
 void permlane_probe_launch(float*, hipStream_t);
 void skinny_gemm_launch(const void*, const void*, void*, int, int, int,
                         hipStream_t);
+void skinny_gemm_swiglu_launch(const void*, const void*, void*, int, int,
+                               int, hipStream_t);
 void adamw_mt_launch(const void*, const void*, const void*, const void*,
                      const void*, long long, float, float, float, float,
                      int, float, hipStream_t);
@@ -209,6 +211,21 @@ void adamw_step_mt(torch::Tensor ptrs, torch::Tensor wd_arr,
                   cur_stream());
 }
 
+torch::Tensor skinny_gemm_swiglu(torch::Tensor GU, torch::Tensor W) {
+  // Y[N,O] = silu(G)*U @ W^T where GU = [N, 2M] packed gate|up.
+  CHECK_GPU(GU); CHECK_CONTIG(GU); CHECK_BF16(GU);
+  CHECK_GPU(W); CHECK_CONTIG(W); CHECK_BF16(W);
+  const int N = (int)GU.size(0), M = (int)GU.size(1) / 2;
+  const int O = (int)W.size(0);
+  TORCH_CHECK(W.size(1) == M, "skinny_gemm_swiglu: dims mismatch");
+  TORCH_CHECK(N >= 1 && N <= 4, "skinny_gemm_swiglu: N must be 1..4");
+  TORCH_CHECK(M % 512 == 0, "skinny_gemm_swiglu: M %% 512 != 0");
+  auto y = torch::empty({N, O}, GU.options());
+  skinny_gemm_swiglu_launch(W.data_ptr(), GU.data_ptr(), y.data_ptr(), N,
+                            M, O, cur_stream());
+  return y;
+}
+
 torch::Tensor skinny_gemm(torch::Tensor X, torch::Tensor W) {
   // Y[N,O] = X[N,I] @ W[O,I]^T (decode GEMV; see skinny_gemm.hip)
   CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
@@ -328,6 +345,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("adamw_step_mt", &adamw_step_mt);
+  m.def("skinny_gemm_swiglu", &skinny_gemm_swiglu);
   m.def("rmsnorm_res", &rmsnorm_res);
   m.def("rope_kvwrite", &rope_kvwrite);
   m.def("swiglu_fwd", &swiglu_fwd);

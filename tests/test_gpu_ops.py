@@ -340,3 +340,18 @@ def test_attention_shape_fuzz():
                                    rtol=0.08, msg=str(cfg))
         torch.testing.assert_close(v.grad.float(), vf.grad, atol=0.08,
                                    rtol=0.08, msg=str(cfg))
+
+
+@pytest.mark.gpu
+def test_skinny_gemm_swiglu_matches_ref():
+    """Fused silu(g)*u + down GEMV vs composed fp32 reference."""
+    C = ops.native()
+    torch.manual_seed(13)
+    for n, m, o in [(1, 14336, 4096), (2, 512, 1024)]:
+        gu = (torch.randn(n, 2 * m, device="cuda") * 0.5).bfloat16()
+        w = (torch.randn(o, m, device="cuda") * 0.02).bfloat16()
+        y = C.skinny_gemm_swiglu(gu, w)
+        g, u = gu.float().split(m, dim=-1)
+        x = torch.nn.functional.silu(g) * u
+        ref = torch.nn.functional.linear(x, w.float())
+        torch.testing.assert_close(y.float(), ref, atol=0.05, rtol=0.05)
