@@ -183,3 +183,61 @@ def test_tls_lb_serves_https(tmp_path):
         raise AssertionError(f"https never came up: {last}")
     server.should_exit = True
     t.join(10)
+
+
+def test_lb_under_replica_churn(client):
+    """Requests keep flowing through the LB while a replica dies and is
+    replaced: error rate stays low (survivor takes traffic) and the
+    service returns to full readiness (round-2 verification-debt
+    item)."""
+    import threading
+
+    import httpx
+
+    from skypilot_amd.client import sdk
+    res = sdk.get(sdk.serve_up(_service_task(2), "churn"), timeout=180)
+    endpoint = res["endpoint"]
+    stat = _wait_ready("churn", 2)
+    stop = threading.Event()
+    results = {"ok": 0, "fail": 0}
+
+    def hammer():
+        while not stop.is_set():
+            try:
+                r = httpx.get(endpoint + "/", timeout=5)
+                if r.status_code == 200:
+                    results["ok"] += 1
+                else:
+                    results["fail"] += 1
+            except Exception:  # noqa: BLE001
+                results["fail"] += 1
+            time.sleep(0.05)
+
+    t = threading.Thread(target=hammer)
+    t.start()
+    try:
+        time.sleep(2)
+        victim = stat["replicas"][0]
+        sdk.get(sdk.down(victim["cluster_name"]))
+        # keep hammering through detection + replacement
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            stats = sdk.get(sdk.serve_status("churn"))
+            ready = [r for r in stats[0]["replicas"]
+                     if r["status"] == "READY"]
+            if len(ready) >= 2 and all(
+                    r["replica_id"] != victim["replica_id"]
+                    or r["status"] == "READY" for r in ready):
+                break
+            time.sleep(1)
+        time.sleep(2)
+    finally:
+        stop.set()
+        t.join(10)
+    total = results["ok"] + results["fail"]
+    assert total > 20, results
+    # the LB may lose a few in-flight requests at the kill instant but
+    # must keep the service usable throughout
+    assert results["fail"] / total < 0.3, results
+    assert results["ok"] > 0
+    sdk.get(sdk.serve_down("churn"), timeout=120)
