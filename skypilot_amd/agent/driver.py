@@ -40,7 +40,12 @@ def run_job(cluster_dir: str, job_id: int) -> int:
         return 1
     spec = job["spec"]
     num_nodes = int(spec.get("num_nodes", 1))
-    gpus_per_node = int(spec.get("gpus_per_node", 0))
+    gpn_raw = float(spec.get("gpus_per_node", 0) or 0)
+    gpu_fraction = gpn_raw if 0 < gpn_raw < 1 else None
+    # fractional share: the one shared device is fully visible; the
+    # fraction travels in SKYPILOT_GPU_FRACTION (best-effort sharing,
+    # reference: fractional k8s accelerators)
+    gpus_per_node = 1 if gpu_fraction else int(gpn_raw)
     gpu_ids = spec.get("gpu_ids") or []
     workdir = spec.get("workdir") or str(Path(cluster_dir) / "workdir")
     Path(workdir).mkdir(parents=True, exist_ok=True)
@@ -98,6 +103,8 @@ def run_job(cluster_dir: str, job_id: int) -> int:
             "SKYPILOT_NODE_RANK": str(node_rank),
             "SKYPILOT_NUM_NODES": str(num_nodes),
             "SKYPILOT_NUM_GPUS_PER_NODE": str(gpus_per_node),
+            **({"SKYPILOT_GPU_FRACTION": str(gpu_fraction)}
+               if gpu_fraction else {}),
             "SKYPILOT_TASK_ID": task_id,
             "SKYPILOT_INTERNAL_JOB_ID": str(job_id),
             # Convenience for torchrun on the one-box pool: a unique

@@ -40,6 +40,16 @@ class Optimizer:
         need = task.num_nodes * res.accelerator_count
         if need == 0:
             return
+        if 0 < res.accelerator_count < 1:
+            # fractional share of one GPU (reference: fractional
+            # accelerators on k8s) — single node only; feasible iff the
+            # pool has any GPU at all (load is checked at provision).
+            if task.num_nodes != 1:
+                raise ResourcesUnavailableError(
+                    "fractional accelerators require num_nodes=1")
+            if not detect_gpus():
+                raise ResourcesUnavailableError("pool has no GPUs")
+            return
         if res.accelerators not in (None, "MI355X"):
             raise ResourcesUnavailableError(
                 f"pool has MI355X only, requested {res.accelerators}")

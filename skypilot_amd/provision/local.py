@@ -98,15 +98,54 @@ def _preempt(victim: Dict) -> None:
     global_state.remove_cluster(name)
 
 
+def _fractional_load(except_cluster: str):
+    """Per-GPU fractional usage + the set of fully-leased GPUs."""
+    frac = {}
+    full = set()
+    for c in global_state.list_clusters(all_workspaces=True):
+        if c["name"] == except_cluster or c["status"] not in (
+                global_state.UP, global_state.INIT):
+            continue
+        h = c["handle"]
+        if h.get("cloud") != CLOUD_NAME:
+            continue
+        f = h.get("gpu_fraction")
+        for g in h.get("gpu_ids", []):
+            if f:
+                frac[g] = frac.get(g, 0.0) + f
+            else:
+                full.add(g)
+    return frac, full
+
+
 def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
                           existing_handle=None, use_spot=False
                           ) -> Dict[str, Any]:
     gpus = detect_gpus()
     need = num_nodes * acc_count
+    gpu_fraction = None
     if existing_handle and existing_handle.get("gpu_ids") is not None:
         gpu_ids = existing_handle["gpu_ids"]
+        gpu_fraction = existing_handle.get("gpu_fraction")
+    elif 0 < acc_count < 1:
+        # Fractional share of one GPU (reference: fractional
+        # accelerators): bin-packing — pick the MOST-loaded GPU that
+        # still fits, so whole GPUs stay free for whole-GPU leases.
+        frac, full = _fractional_load(cluster_name)
+        cand = [(frac.get(g.index, 0.0), g.index) for g in gpus
+                if g.index not in full
+                and frac.get(g.index, 0.0) + acc_count <= 1.0 + 1e-6]
+        if not cand:
+            raise ResourcesUnavailableError(
+                f"no GPU has {acc_count:g} capacity free "
+                f"(fractional shares)")
+        cand.sort(reverse=True)
+        gpu_ids = [cand[0][1]]
+        gpu_fraction = float(acc_count)
     elif need > 0:
+        frac_used, full_taken = _fractional_load(cluster_name)
         taken = set(_allocated_gpus_elsewhere(cluster_name))
+        taken |= set(frac_used)  # fractionally-shared GPUs are not whole
         free = [g.index for g in gpus if g.index not in taken]
         if len(free) < need and not use_spot:
             # On-demand requests reclaim spot capacity (reference:
@@ -133,6 +172,7 @@ def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
         "gpu_ids": gpu_ids,
         "num_nodes": num_nodes,
         "gpus_per_node": acc_count,
+        "gpu_fraction": gpu_fraction,
         "use_spot": bool(use_spot),
         "head_ip": "127.0.0.1",
         "node_ips": ["127.0.0.1"] * num_nodes,

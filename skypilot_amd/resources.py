@@ -81,16 +81,30 @@ def _parse_plus(v) -> tuple[Optional[float], bool]:
     return float(s), False
 
 
-def parse_accelerators(spec) -> tuple[Optional[str], int]:
+def _acc_count(v) -> float:
+    """Counts may be fractional (reference schema: e.g. "A100:0.5" on
+    k8s); fractions must be in (0, 1) — whole GPUs stay ints."""
+    c = float(v)
+    if c <= 0:
+        raise TaskValidationError(f"accelerator count must be > 0: {v}")
+    if c >= 1:
+        if c != int(c):
+            raise TaskValidationError(
+                f"fractional accelerator counts must be < 1: {v}")
+        return int(c)
+    return c
+
+
+def parse_accelerators(spec) -> tuple[Optional[str], float]:
     if spec is None:
         return None, 0
     if isinstance(spec, dict):
         (name, count), = spec.items()
-        return canonical_accelerator(name), int(count)
+        return canonical_accelerator(name), _acc_count(count)
     s = str(spec)
     if ":" in s:
         name, count = s.split(":", 1)
-        return canonical_accelerator(name), int(count)
+        return canonical_accelerator(name), _acc_count(count)
     return canonical_accelerator(s), 1
 
 
@@ -106,7 +120,7 @@ class Resources:
     infra: Optional[str] = None            # local | ssh | k8s pool name
     cloud: Optional[str] = None
     accelerators: Optional[str] = None     # canonical name
-    accelerator_count: int = 0
+    accelerator_count: float = 0           # int, or a fraction in (0,1)
     cpus: Optional[float] = None
     cpus_is_min: bool = False
     memory: Optional[float] = None

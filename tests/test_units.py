@@ -221,3 +221,39 @@ def test_retry_until_up(tmp_path, monkeypatch):
     t.join()
     core.down("want-c")
     gpu_topology.detect_gpus.cache_clear()
+
+
+def test_fractional_gpu_shares(tmp_path, monkeypatch):
+    """MI355X:0.5 shares one GPU between clusters (reference schema:
+    fractional accelerators); whole-GPU leases never land on a shared
+    device, and over-subscription is rejected."""
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "2")
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    from skypilot_amd import core, execution, global_state
+    from skypilot_amd.exceptions import ResourcesUnavailableError
+    from skypilot_amd.task import Task
+
+    def t(acc):
+        return Task.from_yaml_config(
+            {"run": "true", "resources": {"accelerators": acc}})
+
+    execution.launch(t("MI355X:0.5"), "fa", detach_run=True)
+    execution.launch(t("MI355X:0.5"), "fb", detach_run=True)
+    ha = global_state.get_cluster("fa")["handle"]
+    hb = global_state.get_cluster("fb")["handle"]
+    assert ha["gpu_fraction"] == 0.5 and hb["gpu_fraction"] == 0.5
+    assert ha["gpu_ids"] == hb["gpu_ids"]  # packed onto one GPU
+    # whole-GPU lease avoids the shared device
+    execution.launch(t("MI355X:1"), "fw", detach_run=True)
+    hw = global_state.get_cluster("fw")["handle"]
+    assert hw["gpu_ids"][0] != ha["gpu_ids"][0]
+    # no capacity left: another 0.5 fits nowhere (gpu0 full by shares,
+    # gpu1 fully leased)
+    import pytest as _pytest
+    with _pytest.raises(ResourcesUnavailableError):
+        execution.launch(t("MI355X:0.75"), "fc", detach_run=True)
+    for name in ("fa", "fb", "fw"):
+        core.down(name)
+    gpu_topology.detect_gpus.cache_clear()
