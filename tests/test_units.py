@@ -257,3 +257,42 @@ def test_fractional_gpu_shares(tmp_path, monkeypatch):
     for name in ("fa", "fb", "fw"):
         core.down(name)
     gpu_topology.detect_gpus.cache_clear()
+
+
+def test_authorize_matrix():
+    from skypilot_amd import users
+    from skypilot_amd.exceptions import PermissionDeniedError
+    users.authorize("admin", "launch")
+    users.authorize("user", "launch")
+    users.authorize("viewer", "status")
+    users.authorize("viewer", "cost_report")
+    import pytest as _pytest
+    for req in ("launch", "down", "jobs_launch", "serve_up",
+                "storage_delete"):
+        with _pytest.raises(PermissionDeniedError):
+            users.authorize("viewer", req)
+
+
+def test_accelerator_count_validation():
+    from skypilot_amd.exceptions import TaskValidationError
+    from skypilot_amd.resources import parse_accelerators
+    assert parse_accelerators("MI355X:8") == ("MI355X", 8)
+    assert parse_accelerators("MI355X:0.5") == ("MI355X", 0.5)
+    assert parse_accelerators({"MI355X": 0.25}) == ("MI355X", 0.25)
+    import pytest as _pytest
+    with _pytest.raises(TaskValidationError):
+        parse_accelerators("MI355X:1.5")  # fractions must be < 1
+    with _pytest.raises(TaskValidationError):
+        parse_accelerators("MI355X:0")
+
+
+def test_decode_bucket():
+    from skypilot_amd.serve.engine import Engine
+    b = Engine._bucket
+    class _E:  # unbound helper needs no engine state
+        pass
+    e = _E()
+    assert Engine._bucket(e, 1) == 1
+    assert Engine._bucket(e, 3) == 4
+    assert Engine._bucket(e, 16) == 16
+    assert Engine._bucket(e, 17) == 32
