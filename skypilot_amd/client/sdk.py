@@ -35,11 +35,18 @@ def _auth_headers() -> dict:
     """Identity for RBAC: a service-account bearer token
     (SKY_AMD_API_TOKEN) wins; otherwise the local username travels in
     X-Skypilot-User (reference: sky client auth headers)."""
+    h = {}
+    ws = os.environ.get("SKY_AMD_WORKSPACE")
+    if ws:
+        h["X-Skypilot-Workspace"] = ws
     tok = os.environ.get("SKY_AMD_API_TOKEN")
     if tok:
-        return {"Authorization": f"Bearer {tok}"}
+        h["Authorization"] = f"Bearer {tok}"
+        return h
     user = os.environ.get("SKY_AMD_USER") or os.environ.get("USER")
-    return {"X-Skypilot-User": user} if user else {}
+    if user:
+        h["X-Skypilot-User"] = user
+    return h
 
 
 @contextmanager

@@ -108,11 +108,19 @@ def set_request_user(name):
     _request_user.name = name
 
 
+def set_request_workspace(ws):
+    """Thread-scoped workspace for SHORT in-process request handlers
+    (LONG handlers run in their own process and use the env var)."""
+    _request_user.workspace = ws
+
+
 def current_workspace() -> str:
     """Active workspace (reference: sky workspaces — named scopes that
-    partition clusters/jobs): SKY_AMD_WORKSPACE env, else config
+    partition clusters/jobs): request header (thread-local/env, set by
+    the API executor), else SKY_AMD_WORKSPACE env, else config
     `workspace:`, else "default"."""
-    ws = os.environ.get("SKY_AMD_WORKSPACE")
+    ws = (getattr(_request_user, "workspace", None)
+          or os.environ.get("SKY_AMD_WORKSPACE"))
     if ws:
         return ws
     try:

@@ -118,7 +118,9 @@ def create_app(start_workers: bool = True) -> FastAPI:
         except Exception as e:
             raise HTTPException(403, str(e))
         try:
-            rid = executor.schedule(name, body or {}, user=ident["user"])
+            rid = executor.schedule(
+                name, body or {}, user=ident["user"],
+                workspace=request.headers.get("x-skypilot-workspace"))
         except KeyError:
             raise HTTPException(404, f"unknown request type {name!r}")
         return {"request_id": rid}

@@ -53,6 +53,7 @@ def _run_request_inline(req: Dict[str, Any]) -> None:
     buf = io.StringIO()
     from skypilot_amd import global_state
     global_state.set_request_user(req.get("user"))  # thread-scoped RBAC id
+    global_state.set_request_workspace(req.get("workspace"))
     try:
         fn = handler(req["name"])
         with contextlib.redirect_stdout(buf), contextlib.redirect_stderr(buf):
@@ -63,6 +64,7 @@ def _run_request_inline(req: Dict[str, Any]) -> None:
         rdb.finish(rid, rdb.FAILED, error=f"{type(e).__name__}: {e}")
     finally:
         global_state.set_request_user(None)
+        global_state.set_request_workspace(None)
         out = buf.getvalue()
         if out:
             try:
@@ -178,10 +180,12 @@ def stop_workers():
 
 
 def schedule(name: str, body: Dict[str, Any],
-             user: "str | None" = None) -> str:
+             user: "str | None" = None,
+             workspace: "str | None" = None) -> str:
     if name not in _REGISTRY:
         raise KeyError(f"unknown request {name!r}")
-    return rdb.create(name, body, queue_of(name), user=user)
+    return rdb.create(name, body, queue_of(name), user=user,
+                      workspace=workspace)
 
 
 def cancel_request(rid: str) -> bool:

@@ -22,6 +22,8 @@ def main() -> int:
         os.dup2(logf.fileno(), 2)
     if req.get("user"):
         os.environ["SKY_AMD_USER"] = req["user"]
+    if req.get("workspace"):
+        os.environ["SKY_AMD_WORKSPACE"] = req["workspace"]
     try:
         from skypilot_amd.server import executor
         import skypilot_amd.server.handlers  # noqa: F401 (registry)

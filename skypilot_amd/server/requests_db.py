@@ -32,7 +32,8 @@ CREATE TABLE IF NOT EXISTS requests (
     error TEXT,
     worker_pid INTEGER,
     log_path TEXT,
-    user TEXT
+    user TEXT,
+    workspace TEXT
 );
 """
 
@@ -56,6 +57,8 @@ def _conn():
         cols = [r[1] for r in conn.execute("PRAGMA table_info(requests)")]
         if "user" not in cols:  # pre-RBAC databases
             conn.execute("ALTER TABLE requests ADD COLUMN user TEXT")
+        if "workspace" not in cols:
+            conn.execute("ALTER TABLE requests ADD COLUMN workspace TEXT")
         with conn:
             yield conn
     finally:
@@ -63,15 +66,16 @@ def _conn():
 
 
 def create(name: str, body: Dict[str, Any], queue: str,
-           user: Optional[str] = None) -> str:
+           user: Optional[str] = None,
+           workspace: Optional[str] = None) -> str:
     rid = uuid.uuid4().hex[:16]
     log_path = str(api_dir() / "logs" / f"{rid}.log")
     with _conn() as c:
         c.execute(
             "INSERT INTO requests (request_id,name,queue,status,created_at,"
-            "body,log_path,user) VALUES (?,?,?,?,?,?,?,?)",
+            "body,log_path,user,workspace) VALUES (?,?,?,?,?,?,?,?,?)",
             (rid, name, queue, PENDING, time.time(), json.dumps(body),
-             log_path, user))
+             log_path, user, workspace))
     return rid
 
 

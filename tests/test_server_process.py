@@ -83,8 +83,19 @@ def test_real_server_launch_roundtrip(real_server):
     # metrics endpoint live
     m = httpx.get(base + "/metrics", timeout=5).text
     assert "sky_amd_clusters_up" in m
-    # down
-    r = httpx.post(base + "/api/v1/down",
-                   json={"cluster_name": "real-c"}, timeout=10)
+    # workspace header travels with the request (LONG subprocess path)
+    r = httpx.post(base + "/api/v1/launch",
+                   json={"task": {"run": "true"}, "cluster_name": "ws-c"},
+                   headers={"X-Skypilot-Workspace": "hdr-ws"}, timeout=10)
     st = _wait_req(base, r.json()["request_id"])
-    assert st["status"] == "SUCCEEDED"
+    assert st["status"] == "SUCCEEDED", st
+    r = httpx.post(base + "/api/v1/status", json={"all_workspaces": True},
+                   timeout=10)
+    st = _wait_req(base, r.json()["request_id"])
+    ws = {c["name"]: c.get("workspace") for c in st["result"]}
+    assert ws.get("ws-c") == "hdr-ws", ws
+    for name in ("real-c", "ws-c"):
+        r = httpx.post(base + "/api/v1/down",
+                       json={"cluster_name": name}, timeout=10)
+        st = _wait_req(base, r.json()["request_id"])
+        assert st["status"] == "SUCCEEDED"
