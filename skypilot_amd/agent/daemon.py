@@ -206,6 +206,9 @@ def main():
     ap.add_argument("--cluster-dir", required=True)
     ap.add_argument("--port", type=int, required=True)
     ap.add_argument("--gpu-ids", default="")
+    ap.add_argument("--host", default="127.0.0.1",
+                    help="bind address (k8s gang pods bind 0.0.0.0 so "
+                         "peer agents are reachable at podIP)")
     args = ap.parse_args()
     gpu_ids = [int(g) for g in args.gpu_ids.split(",") if g != ""]
     app = create_app(args.cluster_dir, gpu_ids)
@@ -213,7 +216,7 @@ def main():
     Path(args.cluster_dir).mkdir(parents=True, exist_ok=True)
     (Path(args.cluster_dir) / "agent.json").write_text(
         json.dumps({"port": args.port, "pid": os.getpid()}))
-    uvicorn.run(app, host="127.0.0.1", port=args.port, log_level="warning")
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 
 
 if __name__ == "__main__":

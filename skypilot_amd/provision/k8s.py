@@ -76,6 +76,7 @@ def render_pod_manifest(cluster_name: str, acc_count: int,
                 "command": ["python3", "-m", "skypilot_amd.agent.daemon",
                             "--cluster-dir", "/tmp/sky_amd_cluster",
                             "--port", str(AGENT_PORT),
+                            "--host", "0.0.0.0",  # peers dial podIP
                             "--gpu-ids",
                             ",".join(str(i) for i in range(acc_count))],
                 "workingDir": REMOTE_REPO,
