@@ -268,7 +268,10 @@ class ServiceRuntime:
             ssl_kw = {"ssl_certfile": tls.certfile,
                       "ssl_keyfile": tls.keyfile}
             st.update_service(self.name, tls=True)
-        config = uvicorn.Config(self.lb_app(), host="127.0.0.1",
+        from skypilot_amd import config as sky_config
+        lb_host = sky_config.get_nested(["serve", "lb_host"],
+                                        "127.0.0.1") or "127.0.0.1"
+        config = uvicorn.Config(self.lb_app(), host=lb_host,
                                 port=self.lb_port, log_level="warning",
                                 **ssl_kw)
         server = uvicorn.Server(config)
