@@ -79,6 +79,18 @@ def down(service_name: str) -> None:
         except PermissionError:
             break
     st.remove_service(service_name)
+    # Belt-and-braces: if the controller exited before finishing its
+    # teardown (or was killed), reap any replica clusters it left —
+    # otherwise `sky status` shows orphaned sky-serve-* clusters.
+    from skypilot_amd import global_state
+    from skypilot_amd.backends.pool_backend import PoolBackend
+    prefix = f"sky-serve-{service_name}-"
+    for c in global_state.list_clusters(all_workspaces=True):
+        if c["name"].startswith(prefix):
+            try:
+                PoolBackend().teardown(c["handle"], terminate=True)
+            except Exception:  # noqa: BLE001
+                global_state.remove_cluster(c["name"])
 
 
 def logs(service_name: str, replica_id: int = None,
