@@ -50,6 +50,12 @@ class Snapshotter:
         """Launch D2H copies on the side stream; returns immediately.
         Must be called after the optimizer step (main-stream work is
         ordered before the copies via an event)."""
+        # A previous non-blocking commit may still be torch.save-ing the
+        # SAME pinned host buffers; overwriting them mid-write would let a
+        # torn checkpoint atomically replace the last good one. Join it
+        # before launching new copies into the shared staging buffers.
+        if self._writer is not None and self._writer.is_alive():
+            self._writer.join()
         st = self.trainer.opt.state_tensors()
         self._step = st["step_count"]
         self._train_step = self.trainer.step_count

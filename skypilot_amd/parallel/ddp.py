@@ -85,11 +85,12 @@ class BucketedDDP:
 
     def _hook(self, p):
         b = self._p2b[p]
-        view = b.views[p]
-        if self._accumulating:
-            view.add_(p.grad)
-        else:
-            view.copy_(p.grad)
+        # Always accumulate: zero_grad() clears the flat buffers at the
+        # start of each optimizer step, so add_ is correct for both the
+        # single-micro-step case and every micro-step of grad accumulation.
+        # (copy_ on the final micro-step would overwrite the accumulated
+        # sum from earlier micro-steps.)
+        b.views[p].add_(p.grad)
         p.grad = None  # free eagerly; flat buffer is the only grad storage
         b.pending -= 1
         if b.pending == 0 and self.world > 1 and not self._accumulating:

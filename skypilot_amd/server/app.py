@@ -53,20 +53,47 @@ def create_app(start_workers: bool = True) -> FastAPI:
                 "pid": os.getpid()}
 
     # ---- identity & RBAC (reference: sky/users/rbac.py + server auth) ----
+    # Auth modes (reference: sky server only trusts identity headers set
+    # by an authenticating proxy — X-Auth-Request-Email behind
+    # oauth2-proxy; otherwise identity comes from tokens):
+    #   local — single-user/localhost deployment: the client-supplied
+    #           X-Skypilot-User header is trusted (anyone who can reach
+    #           the loopback socket IS the operator).  Default for
+    #           loopback binds.
+    #   token — non-local deployment: identity REQUIRES a bearer
+    #           service-account token, except when an authenticating
+    #           proxy is declared trusted (SKY_AMD_TRUST_PROXY_AUTH=1),
+    #           in which case X-Auth-Request-Email is accepted.
+    def _auth_mode() -> str:
+        return os.environ.get("SKY_AMD_AUTH_MODE", "local")
+
     def _identity(request: Request) -> Dict[str, str]:
-        """Resolve (user, role) from a bearer service-account token, the
-        X-Skypilot-User header, or the server's own OS user."""
+        """Resolve (user, role) per the active auth mode."""
         auth = request.headers.get("authorization", "")
         if auth.lower().startswith("bearer "):
             sa = users.resolve_token(auth[7:].strip())
             if sa is None:
                 raise HTTPException(401, "invalid service-account token")
             return {"user": sa["name"], "role": sa["role"]}
+        if _auth_mode() == "token":
+            if os.environ.get("SKY_AMD_TRUST_PROXY_AUTH") == "1":
+                email = request.headers.get("x-auth-request-email")
+                if email:
+                    return {"user": email, "role": users.ensure_user(email)}
+            raise HTTPException(
+                401, "authentication required: pass a service-account "
+                     "bearer token (server runs in token auth mode)")
         name = request.headers.get("x-skypilot-user")
         if name:
             return {"user": name, "role": users.ensure_user(name)}
         me = global_state.current_user()
         return {"user": me, "role": users.ensure_user(me)}
+
+    def _may_access_request(ident: Dict[str, str], req) -> bool:
+        """Request rows are visible/cancellable by their owner or admin."""
+        if ident["role"] == "admin":
+            return True
+        return req.get("user") in (None, ident["user"])
 
     def _require_admin(request: Request) -> Dict[str, str]:
         ident = _identity(request)
@@ -126,10 +153,14 @@ def create_app(start_workers: bool = True) -> FastAPI:
         return {"request_id": rid}
 
     @app.get("/api/get")
-    def api_get(request_id: str):
+    def api_get(request_id: str, request: Request = None):
+        ident = _identity(request)
         req = rdb.get(request_id)
         if req is None:
             raise HTTPException(404, "no such request")
+        if not _may_access_request(ident, req):
+            raise HTTPException(
+                403, f"request {request_id} belongs to {req.get('user')!r}")
         return {
             "request_id": request_id,
             "name": req["name"],
@@ -139,10 +170,14 @@ def create_app(start_workers: bool = True) -> FastAPI:
         }
 
     @app.get("/api/stream")
-    async def api_stream(request_id: str):
+    async def api_stream(request_id: str, request: Request = None):
+        ident = _identity(request)
         req = rdb.get(request_id)
         if req is None:
             raise HTTPException(404, "no such request")
+        if not _may_access_request(ident, req):
+            raise HTTPException(
+                403, f"request {request_id} belongs to {req.get('user')!r}")
 
         async def gen():
             pos = 0
@@ -169,8 +204,13 @@ def create_app(start_workers: bool = True) -> FastAPI:
         return StreamingResponse(gen(), media_type="text/plain")
 
     @app.post("/api/cancel")
-    def api_cancel(body: Dict[str, Any]):
+    def api_cancel(body: Dict[str, Any], request: Request = None):
+        ident = _identity(request)
         rid = body["request_id"]
+        req = rdb.get(rid)
+        if req is not None and not _may_access_request(ident, req):
+            raise HTTPException(
+                403, f"request {rid} belongs to {req.get('user')!r}")
         return {"cancelled": executor.cancel_request(rid)}
 
     @app.get("/dashboard")
@@ -209,8 +249,10 @@ def create_app(start_workers: bool = True) -> FastAPI:
         return PlainTextResponse("\n".join(lines) + "\n")
 
     @app.get("/api/requests")
-    def api_requests(limit: int = 100):
-        reqs = rdb.list_requests(limit)
+    def api_requests(limit: int = 100, request: Request = None):
+        ident = _identity(request)
+        reqs = [r for r in rdb.list_requests(limit)
+                if _may_access_request(ident, r)]
         return [{k: r[k] for k in
                  ("request_id", "name", "status", "created_at",
                   "finished_at", "error")} for r in reqs]
@@ -238,6 +280,14 @@ def main():
                     default=int(os.environ.get("SKY_AMD_API_PORT",
                                                DEFAULT_PORT)))
     args = ap.parse_args()
+    if (args.host not in ("127.0.0.1", "localhost", "::1")
+            and "SKY_AMD_AUTH_MODE" not in os.environ):
+        # Non-local bind: never trust client-supplied identity headers
+        # by default (header spoofing → admin escalation).
+        os.environ["SKY_AMD_AUTH_MODE"] = "token"
+        print(f"[server] binding {args.host}: auth mode set to 'token' "
+              "(set SKY_AMD_AUTH_MODE=local to trust identity headers, "
+              "e.g. behind an authenticating proxy)")
     import uvicorn
     (global_state.root_dir() / "api").mkdir(parents=True, exist_ok=True)
     (global_state.root_dir() / "api" / "server.json").write_text(

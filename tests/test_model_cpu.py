@@ -121,11 +121,54 @@ def test_grad_accum_matches_big_batch():
     tr2.ddp.mark_step_start(accumulating=False)
     (tr2.model.loss(*b2) / 2).backward()
     tr2.ddp.finish()
+    # Compare the RAW accumulated gradients (not post-Adam weights, which
+    # tiny-lr can mask): the accumulated sum over both micro-batches must
+    # match the big-batch gradient, not just the last micro-batch's.
+    for pa, pb in zip(tr1.model.parameters(), tr2.model.parameters()):
+        ga, gb = pa._sky_grad.float(), pb._sky_grad.float()
+        # bf16 buckets: allow a few ulp (2^-8 relative) of rounding noise
+        assert torch.allclose(ga, gb, rtol=0.05, atol=1e-4), \
+            "accumulated grad != big-batch grad"
     tr2.opt.lr = tr2.current_lr()
     tr2.opt.step(grad_scale=tr2.ddp.grad_scale)
     p1 = tr1.model.lm_head.weight.detach()
     p2 = tr2.model.lm_head.weight.detach()
     assert torch.allclose(p1.float(), p2.float(), atol=5e-4, rtol=0)
+
+
+def test_grad_accum_sums_not_overwrites():
+    """The final micro-step must ADD into the bucket, not overwrite it.
+
+    Regression test for the round-1 bug where mark_step_start(False) made
+    the hook copy_ the last micro-batch's grad over the accumulated sum.
+    Uses two different micro-batches so sum != last.
+    """
+    from skypilot_amd.train.trainer import TrainConfig, Trainer
+    torch.manual_seed(3)
+    cfg = TrainConfig(model="llama-debug", micro_batch=1, seq_len=32,
+                      device="cpu", seed=7, grad_accum=2)
+    tr = Trainer(cfg)
+    t1 = torch.randint(0, 512, (1, 33))
+    t2 = torch.randint(0, 512, (1, 33))
+    b1 = (t1[:, :-1], t1[:, 1:].contiguous())
+    b2 = (t2[:, :-1], t2[:, 1:].contiguous())
+    # grad of b2 alone
+    tr.ddp.zero_grad()
+    tr.ddp.mark_step_start(accumulating=False)
+    tr.model.loss(*b2).backward()
+    tr.ddp.finish()
+    g_last = tr.model.lm_head.weight._sky_grad.float().clone()
+    # accumulated grad of b1 then b2
+    tr.ddp.zero_grad()
+    tr.ddp.mark_step_start(accumulating=True)
+    tr.model.loss(*b1).backward()
+    tr.ddp.finish()
+    tr.ddp.mark_step_start(accumulating=False)
+    tr.model.loss(*b2).backward()
+    tr.ddp.finish()
+    g_sum = tr.model.lm_head.weight._sky_grad.float()
+    rel = float((g_sum - g_last).abs().max() / g_last.abs().max())
+    assert rel > 0.05, "accumulated grad == last micro-batch grad (overwrite bug)"
 
 
 def test_lr_schedule():

@@ -67,6 +67,54 @@ def test_service_account_token_auth(sky_env, client):
                        headers=hdr).status_code == 401
 
 
+def test_token_auth_mode_rejects_header_identity(sky_env, client,
+                                                 monkeypatch):
+    """In token auth mode (non-local binds) the X-Skypilot-User header
+    must NOT grant identity — only bearer tokens."""
+    r = client.post("/api/users/token", json={"name": "tok-mode",
+                                              "role": "user"})
+    tok = r.json()["token"]
+    monkeypatch.setenv("SKY_AMD_AUTH_MODE", "token")
+    # header identity rejected
+    assert client.post("/api/v1/status", json={},
+                       headers={"X-Skypilot-User": "imposter"}
+                       ).status_code == 401
+    # no identity at all rejected too
+    assert client.post("/api/v1/status", json={}).status_code == 401
+    assert client.get("/api/requests").status_code == 401
+    # bearer token still works
+    assert client.post("/api/v1/status", json={},
+                       headers={"Authorization": f"Bearer {tok}"}
+                       ).status_code == 200
+    # trusted-proxy header accepted only when explicitly configured
+    hdr = {"X-Auth-Request-Email": "eve@corp"}
+    assert client.post("/api/v1/status", json={},
+                       headers=hdr).status_code == 401
+    monkeypatch.setenv("SKY_AMD_TRUST_PROXY_AUTH", "1")
+    assert client.post("/api/v1/status", json={},
+                       headers=hdr).status_code == 200
+
+
+def test_request_access_scoped_to_owner(sky_env, client):
+    """/api/get, /api/cancel and /api/requests are owner-or-admin only."""
+    hdr_a = {"X-Skypilot-User": "req-owner"}
+    hdr_b = {"X-Skypilot-User": "req-snoop"}
+    rid = client.post("/api/v1/status", json={},
+                      headers=hdr_a).json()["request_id"]
+    # another plain user cannot read, cancel, or list it
+    assert client.get("/api/get", params={"request_id": rid},
+                      headers=hdr_b).status_code == 403
+    assert client.post("/api/cancel", json={"request_id": rid},
+                       headers=hdr_b).status_code == 403
+    listed = client.get("/api/requests", headers=hdr_b).json()
+    assert all(r["request_id"] != rid for r in listed)
+    # the owner and the admin can
+    assert client.get("/api/get", params={"request_id": rid},
+                      headers=hdr_a).status_code == 200
+    assert client.get("/api/get", params={"request_id": rid}
+                      ).status_code == 200
+
+
 def test_cluster_ownership_enforced(sky_env, client):
     import time
     # admin (default identity) launches a cluster
