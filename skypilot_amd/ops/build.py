@@ -24,6 +24,7 @@ HIP_SOURCES = [
     "adamw.hip",
     "cross_entropy.hip",
     "attention_fwd.hip",
+    "attention_fwd_v3.hip",
     "attention_bwd.hip",
     "attention_decode.hip",
     "swiglu.hip",

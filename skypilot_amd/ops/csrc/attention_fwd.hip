@@ -423,6 +423,11 @@ extern "C" __global__ void attn_fwd_swapped_kernel(
     const unsigned short*, const unsigned short*, const unsigned short*,
     unsigned short*, float*, int, int, int, int, float, int);
 
+extern "C" void attn_fwd_v3_launch(const void* Q, const void* K,
+                                   const void* V, void* O, float* lse, int B,
+                                   int S, int Hq, int Hkv, float scale,
+                                   bool causal, hipStream_t stream);
+
 extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
                                 void* O, float* lse, int B, int S, int Hq,
                                 int Hkv, float scale, bool causal,
@@ -442,6 +447,17 @@ extern "C" void attn_fwd_launch(const void* Q, const void* K, const void* V,
   // experiments.
   // Swapped-operand kernel is the default (238 vs 207 TF/s measured at
   // the bench shape); SKY_ATTN_FWD_SWAPPED=0 falls back for A/B tests.
+  // v3 (8-wave 32x32 deep-pipelined swapped schedule) is the round-2
+  // default for the flagship shapes; SKY_ATTN_FWD_V3=0 for A/B.
+  static const int use_v3 = [] {
+    const char* e = getenv("SKY_ATTN_FWD_V3");
+    return e ? atoi(e) : 1;
+  }();
+  if (use_v3 && S % 256 == 0) {
+    attn_fwd_v3_launch(Q, K, V, O, lse, B, S, Hq, Hkv, scale, causal,
+                       stream);
+    return;
+  }
   static const int use_swapped = [] {
     const char* e = getenv("SKY_ATTN_FWD_SWAPPED");
     return e ? atoi(e) : 1;

@@ -1,0 +1,337 @@
+// Flash-attention forward v3 — 8-wave 32x32 swapped-operand schedule.
+//
+// The round-2 deep-pipeline forward (guide §B "8-warp 32x32 ladder",
+// techniques T2/T5/T10/T12/T13/T14 composed, not grafted):
+//   - 512-thread blocks: 8 waves, wave w owns q rows [32w, 32w+32) of a
+//     256-row q block; kv tiles of 64.
+//   - Swapped QK^T (S^T = K Q^T): each lane holds 32 kv-scores of ONE q
+//     column in MFMA C-registers, so the online softmax is 31 in-lane
+//     fmax + one lane<->lane+32 exchange — zero LDS softmax state.
+//   - K in XOR-swizzled LDS (16-slot swizzle keyed on row&15, rows are
+//     256 B), staged by global_load_lds with pre-swizzled SOURCE
+//     addresses (linear LDS dest, guide m201 stage pattern).
+//   - V in a [kv/4][d/16][4][16] subtiled LDS layout, consumed as the
+//     PV A-operand by paired ds_read_b64_tr_b16 (hardware transpose
+//     read, guide T10) — no transposed scatter staging.
+//   - P^T -> B-fragments by v_cvt_pk_bf16_f32 + permlane32_swap (T12,
+//     semantics HW-verified: profiles/r01_hw_probe_semantics.txt).
+//   - defer-max rescale threshold (T13, THR=8 in exp2 domain).
+//   - One barrier per kv tile: V commits from prefetch registers, then
+//     barrier (the compiler's vmcnt drain covers the async K loads),
+//     then next tile's loads issue and stay in flight under compute.
+//   - Q is pre-scaled by scale*log2(e) so softmax uses native exp2.
+//
+// Layouts: Q,O [B,S,Hq,128]; K,V [B,S,Hkv,128]; lse [B,Hq,S] fp32
+// (natural-log convention, same as v1/v2 kernels).
+#include "common.h"
+
+#define ATT_D 128
+#define KVB 64
+#define QBW 32
+#define NWV3 8
+#define QBLK3 (QBW * NWV3)  // 256 q rows per block
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+#define MFMA32V3(a, b, c) \
+  __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
+
+// 16-slot XOR swizzle for the K tile (rows are 256 B = 16 chunks): a
+// wave's A-fragment read (32 rows, fixed 16 B chunk) spreads over all
+// 16 slots -> 2 lanes/slot = free (guide §6 G4: 2-way is 1.02x).
+__device__ __forceinline__ int swzK16(int byte_off, int row) {
+  return byte_off ^ ((row & 15) << 4);
+}
+
+// V subtile layout: element index of V[kv][d] inside the
+// [kv/4][d/16][4][16] tile.  tr-read gather addresses land on
+// consecutive 8 B slots per 16-lane group (conflict-free); the staging
+// s16x8 writes stay 16 B contiguous because d moves within one subtile
+// row.
+__device__ __forceinline__ int vsub(int kv, int d) {
+  return (((kv >> 2) << 3) + (d >> 4)) * 64 + ((kv & 3) << 4) + (d & 15);
+}
+
+__device__ __forceinline__ unsigned cvtpk_bf16(float a, float b) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+  return r;
+}
+
+// Async 16B global->LDS copy: per-lane global source, wave-uniform LDS
+// base + lane*16 destination (guide §5: the only supported dest form).
+__device__ __forceinline__ void gload_lds16(const void* g, void* l) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)l, 16, 0, 0);
+}
+
+extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, unsigned short* __restrict__ O,
+    float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, float scale,
+    int causal) {
+  __shared__ unsigned short k_lds[2][KVB * ATT_D];  // 2 x 16 KB
+  __shared__ unsigned short v_lds[2][KVB * ATT_D];  // 2 x 16 KB
+
+  const int qt = gridDim.x - 1 - blockIdx.x;  // heavy blocks first
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * QBLK3;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int col = lane & 31;  // q column owned by this lane
+  const int h = lane >> 5;
+  const int g = lane >> 4;
+  const int lw = lane & 15;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  const int my_q = qbase + QBW * w + col;
+
+  // ---- Q as B-fragments, pre-scaled by scale*log2(e) (exp2 softmax).
+  const float qs = scale * 1.44269504088896340736f;
+  s16x8 q_b[8];
+  {
+    const unsigned short* src = Qb + (long long)my_q * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks) {
+      s16x8 raw = *(const s16x8*)(src + ks * 16 + h * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * qs);
+      q_b[ks] = raw;
+    }
+  }
+
+  f32x16 o_t[4];
+#pragma unroll
+  for (int dn = 0; dn < 4; ++dn)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) o_t[dn][r] = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;
+
+  const int n_tiles = causal ? (qbase + QBLK3) / KVB : S / KVB;
+  // Tiles this wave actually computes (beyond its diagonal: staging +
+  // barriers only).
+  const int w_tiles =
+      causal ? ((qbase + QBW * w + QBW - 1) >> 6) + 1 : n_tiles;
+
+  // ---- staging assignments.
+  // V: thread owns 16B chunks 2*tid, 2*tid+1 of the [64][128] tile.
+  // K: wave w stages rows [8w, 8w+8) via 2 global_load_lds.
+  s16x8 vpre[2];
+  const int vc0 = tid * 2;
+
+  // prologue: issue tile 0 (V loads to registers first, then async K).
+  {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = vc0 + i;
+      vpre[i] = *(const s16x8*)(Vb + (long long)(c >> 4) * kv_rowstride +
+                                (c & 15) * 8);
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int row = 8 * w + 4 * i + (lane >> 4);
+      int chunk = lane & 15;
+      const void* src = (const char*)(Kb + (long long)row * kv_rowstride) +
+                        ((chunk ^ (row & 15)) << 4);
+      gload_lds16(src, (char*)k_lds[0] + (8 * w + 4 * i) * 256);
+    }
+  }
+
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int buf = kt & 1;
+    // 1. commit the prefetched V registers into this tile's buffer.
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int c = vc0 + i;
+      *(s16x8*)((char*)v_lds[buf] + 2 * vsub(c >> 4, (c & 15) * 8)) =
+          vpre[i];
+    }
+    // 2. one barrier per tile; hipcc's pre-barrier waitcnt drains the
+    // in-flight K global_load_lds for this buffer too.
+    __syncthreads();
+    // 3. issue next tile's staging; it stays in flight under compute.
+    if (kt + 1 < n_tiles) {
+      const long long nb = (long long)(kt + 1) * KVB;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int c = vc0 + i;
+        vpre[i] = *(const s16x8*)(
+            Vb + (nb + (c >> 4)) * kv_rowstride + (c & 15) * 8);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int row = 8 * w + 4 * i + (lane >> 4);
+        int chunk = lane & 15;
+        const void* src =
+            (const char*)(Kb + (nb + row) * kv_rowstride) +
+            ((chunk ^ (row & 15)) << 4);
+        gload_lds16(src, (char*)k_lds[buf ^ 1] + (8 * w + 4 * i) * 256);
+      }
+    }
+    if (kt >= w_tiles) continue;  // past this wave's diagonal
+
+    const int kvbase = kt * KVB;
+    // ---- S^T = K Q^T (Q pre-scaled).  2 kv blocks x 8 k-steps,
+    // kv-inner so two accumulator chains interleave.
+    f32x16 st[2];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[n][r] = 0.f;
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        int krow = n * 32 + col;
+        s16x8 kf = *(const s16x8*)(
+            (char*)k_lds[buf] + swzK16(krow * 256 + (ks * 2 + h) * 16, krow));
+        st[n] = MFMA32V3(as_bf16x8(kf), as_bf16x8(q_b[ks]), st[n]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- causal mask (finite big-negative so exp2 underflows to 0;
+    // m_run is already real for every row because tile 0 is unmasked).
+    if (causal && kvbase + KVB - 1 > qbase + QBW * w) {
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = kvbase + n * 32 + (r & 3) + ((r >> 2) << 3) + (h << 2);
+          if (kv > my_q) st[n][r] = -30000.f;
+        }
+    }
+
+    // ---- online softmax on the lane's 32 scores of q column `col`.
+    float pmax = st[0][0];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pmax = fmaxf(pmax, st[n][r]);
+    pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
+    // defer-max (T13): skip the O-rescale while the running max grows
+    // by <= 8 (P bounded by 2^8; f32 accumulators absorb it).
+    if (!__all(pmax - m_run <= 8.0f)) {
+      float m_new = fmaxf(m_run, pmax);
+      float corr = (m_run == -INFINITY) ? 0.f : exp2f(m_run - m_new);
+#pragma unroll
+      for (int dn = 0; dn < 4; ++dn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_t[dn][r] *= corr;
+      l_run *= corr;
+      m_run = m_new;
+    }
+    float rs = 0.f;
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float e = exp2f(st[n][r] - m_run);
+        st[n][r] = e;
+        rs += e;
+      }
+    rs += __shfl_xor(rs, 32, 64);
+    l_run += rs;
+
+    // ---- PV: per k-step, issue the 8 tr reads for all four d-blocks,
+    // pack P under their latency, then 4 independent MFMA chains.
+    const int kv0g = (g >> 1) << 3;
+    const int trd = ((g & 1) << 4) + ((lw & 3) << 2);
+    const int trr = lw >> 2;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      tr4 t0, t1;
+      {
+        int kvr = ks * 16 + kv0g + trr;
+        unsigned a00 = (unsigned)(2 * vsub(kvr, 0 * 32 + trd));
+        unsigned a01 = (unsigned)(2 * vsub(kvr + 4, 0 * 32 + trd));
+        unsigned a10 = (unsigned)(2 * vsub(kvr, 1 * 32 + trd));
+        unsigned a11 = (unsigned)(2 * vsub(kvr + 4, 1 * 32 + trd));
+        unsigned a20 = (unsigned)(2 * vsub(kvr, 2 * 32 + trd));
+        unsigned a21 = (unsigned)(2 * vsub(kvr + 4, 2 * 32 + trd));
+        unsigned a30 = (unsigned)(2 * vsub(kvr, 3 * 32 + trd));
+        unsigned a31 = (unsigned)(2 * vsub(kvr + 4, 3 * 32 + trd));
+        unsigned base = (unsigned)(size_t)((char*)v_lds[buf]);
+        ds_tr4_issue(&t0, base + a00, base + a01, base + a10, base + a11);
+        ds_tr4_issue(&t1, base + a20, base + a21, base + a30, base + a31);
+      }
+      // pack P^T B-fragment while the tr reads are in flight (T12).
+      s16x8 pb;
+      {
+        const int e0 = (ks & 1) << 3;
+        const int n = ks >> 1;
+        unsigned w0a = cvtpk_bf16(st[n][e0 + 0], st[n][e0 + 1]);
+        unsigned w1a = cvtpk_bf16(st[n][e0 + 2], st[n][e0 + 3]);
+        unsigned w0b = cvtpk_bf16(st[n][e0 + 4], st[n][e0 + 5]);
+        unsigned w1b = cvtpk_bf16(st[n][e0 + 6], st[n][e0 + 7]);
+        auto p0 = __builtin_amdgcn_permlane32_swap(w0a, w0b, false, false);
+        auto p1 = __builtin_amdgcn_permlane32_swap(w1a, w1b, false, false);
+        union { unsigned u[4]; s16x8 v; } pk;
+        pk.u[0] = p0[0];
+        pk.u[1] = p1[0];
+        pk.u[2] = p0[1];
+        pk.u[3] = p1[1];
+        pb = pk.v;
+      }
+      lgkm_wait0_bind(&t0);
+      lgkm_wait0_bind(&t1);
+      union { unsigned long long u[2]; s16x8 v; } vf;
+      __builtin_amdgcn_s_setprio(1);
+      vf.u[0] = t0.d[0];
+      vf.u[1] = t0.d[1];
+      o_t[0] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[0]);
+      vf.u[0] = t0.d[2];
+      vf.u[1] = t0.d[3];
+      o_t[1] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[1]);
+      vf.u[0] = t1.d[0];
+      vf.u[1] = t1.d[1];
+      o_t[2] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[2]);
+      vf.u[0] = t1.d[2];
+      vf.u[1] = t1.d[3];
+      o_t[3] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[3]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+  // ---- epilogue: O[q][d] = O^T[d][q] / l; lse = ln2*m' + ln(l).
+  unsigned short* Ob = O + ((long long)b * S * Hq + qh) * ATT_D;
+  float* lse_b = lse_out + ((long long)b * Hq + qh) * S;
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  unsigned short* orow = Ob + (long long)my_q * q_rowstride;
+#pragma unroll
+  for (int dn = 0; dn < 4; ++dn)
+#pragma unroll
+    for (int rq = 0; rq < 4; ++rq) {
+      s16x4 ov;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        ov[r] = (short)f2bf(o_t[dn][rq * 4 + r] * inv_l);
+      *(s16x4*)(orow + dn * 32 + rq * 8 + h * 4) = ov;
+    }
+  if (h == 0)
+    lse_b[my_q] = (l_run > 0.f)
+                      ? 0.69314718055994530942f * m_run + __logf(l_run)
+                      : -INFINITY;
+}
+
+extern "C" void attn_fwd_v3_launch(const void* Q, const void* K,
+                                   const void* V, void* O, float* lse, int B,
+                                   int S, int Hq, int Hkv, float scale,
+                                   bool causal, hipStream_t stream) {
+  dim3 grid(S / QBLK3, B * Hq);
+  hipLaunchKernelGGL(attn_fwd_v3_kernel, grid, dim3(512), 0, stream,
+                     (const unsigned short*)Q, (const unsigned short*)K,
+                     (const unsigned short*)V, (unsigned short*)O, lse, B, S,
+                     Hq, Hkv, scale, causal ? 1 : 0);
+}

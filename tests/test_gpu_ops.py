@@ -316,7 +316,9 @@ def test_attention_shape_fuzz():
     for (B, S, Hq, Hkv, causal) in [
             (1, 64, 8, 8, True), (3, 128, 16, 2, True),
             (2, 256, 32, 8, False), (1, 1024, 8, 2, True),
-            (2, 192, 4, 4, False)]:
+            (2, 192, 4, 4, False),
+            # S % 256 == 0 routes to the v3 8-wave 32x32 kernel
+            (2, 512, 16, 4, True), (1, 256, 8, 8, True)]:
         q = (torch.randn(B, S, Hq, 128, device="cuda") * 0.5).bfloat16()
         k = (torch.randn(B, S, Hkv, 128, device="cuda") * 0.5).bfloat16()
         v = (torch.randn(B, S, Hkv, 128, device="cuda") * 0.5).bfloat16()
