@@ -23,13 +23,14 @@
 //
 // Layouts: Q,O [B,S,Hq,128]; K,V [B,S,Hkv,128]; lse [B,Hq,S] fp32
 // (natural-log convention, same as v1/v2 kernels).
+#include <cstdlib>
 #include "common.h"
 
 #define ATT_D 128
 #define KVB 64
 #define QBW 32
 #define NWV3 8
-#define QBLK3 (QBW * NWV3)  // 256 q rows per block
+#define QBLK3 (QBW * NWV3)  // 256 q rows per block (8-wave variant)
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 #define MFMA32V3(a, b, c) \
@@ -57,6 +58,41 @@ __device__ __forceinline__ unsigned cvtpk_bf16(float a, float b) {
   return r;
 }
 
+// lane <-> lane+32 exchange via permlane32_swap: pure VALU (unlike
+// __shfl_xor's ds_bpermute, which is an LGKM op whose compiler-inserted
+// lgkmcnt(0) would drain our in-flight tr reads mid-softmax).
+__device__ __forceinline__ float xhalf32(float x) {
+  union { float f; unsigned u; } c;
+  c.f = x;
+  auto p = __builtin_amdgcn_permlane32_swap(c.u, c.u, false, false);
+  union { unsigned u; float f; } r0, r1;
+  r0.u = p[0];
+  r1.u = p[1];
+  return (threadIdx.x & 32) ? r0.f : r1.f;
+}
+
+// Counted wait for a 2-deep tr-read pipeline: block until only the most
+// recent 8 lgkm ops (the next batch's tr reads) remain outstanding, and
+// bind the dependency to this batch's 8 destination registers.
+__device__ __forceinline__ void lgkm_wait8_bind2(tr4* a, tr4* b) {
+  asm volatile("s_waitcnt lgkmcnt(8)"
+               : "+v"(a->d[0]), "+v"(a->d[1]), "+v"(a->d[2]),
+                 "+v"(a->d[3]), "+v"(b->d[0]), "+v"(b->d[1]),
+                 "+v"(b->d[2]), "+v"(b->d[3])
+               :
+               : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+}
+__device__ __forceinline__ void lgkm_wait0_bind2(tr4* a, tr4* b) {
+  asm volatile("s_waitcnt lgkmcnt(0)"
+               : "+v"(a->d[0]), "+v"(a->d[1]), "+v"(a->d[2]),
+                 "+v"(a->d[3]), "+v"(b->d[0]), "+v"(b->d[1]),
+                 "+v"(b->d[2]), "+v"(b->d[3])
+               :
+               : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+}
+
 // Async 16B global->LDS copy: per-lane global source, wave-uniform LDS
 // base + lane*16 destination (guide §5: the only supported dest form).
 __device__ __forceinline__ void gload_lds16(const void* g, void* l) {
@@ -65,11 +101,19 @@ __device__ __forceinline__ void gload_lds16(const void* g, void* l) {
       (__attribute__((address_space(3))) unsigned int*)l, 16, 0, 0);
 }
 
-extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
+// NW = waves per block (8 -> one 512-thread block/CU; 4 -> two
+// independent 256-thread blocks/CU whose phases interleave freely —
+// round-1 measured that cross-block overlap beats one barrier-synced
+// big block on these structures).
+template <int NW>
+__global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
     const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V, unsigned short* __restrict__ O,
     float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, float scale,
     int causal) {
+  constexpr int QB = QBW * NW;        // q rows per block
+  constexpr int NKI = KVB / NW / 4;   // K global_load_lds per wave
+  constexpr int NVC = 16 / NW;        // V 16B chunks per thread
   __shared__ unsigned short k_lds[2][KVB * ATT_D];  // 2 x 16 KB
   __shared__ unsigned short v_lds[2][KVB * ATT_D];  // 2 x 16 KB
 
@@ -78,7 +122,7 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
   const int b = bh / Hq;
   const int qh = bh % Hq;
   const int kvh = qh / (Hq / Hkv);
-  const int qbase = qt * QBLK3;
+  const int qbase = qt * QB;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -118,33 +162,36 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
     for (int r = 0; r < 16; ++r) o_t[dn][r] = 0.f;
   float m_run = -INFINITY, l_run = 0.f;
 
-  const int n_tiles = causal ? (qbase + QBLK3) / KVB : S / KVB;
+  // causal: 1 = normal; 2 = debug timing (causal loop structure, mask
+  // skipped); 3 = debug timing (no per-wave diagonal skip).
+  const int n_tiles = causal ? (qbase + QB) / KVB : S / KVB;
   // Tiles this wave actually computes (beyond its diagonal: staging +
   // barriers only).
-  const int w_tiles =
-      causal ? ((qbase + QBW * w + QBW - 1) >> 6) + 1 : n_tiles;
+  const int w_tiles = (causal == 1 || causal == 2)
+                          ? ((qbase + QBW * w + QBW - 1) >> 6) + 1
+                          : n_tiles;
 
   // ---- staging assignments.
   // V: thread owns 16B chunks 2*tid, 2*tid+1 of the [64][128] tile.
   // K: wave w stages rows [8w, 8w+8) via 2 global_load_lds.
-  s16x8 vpre[2];
-  const int vc0 = tid * 2;
+  s16x8 vpre[NVC];
+  const int vc0 = tid * NVC;
 
   // prologue: issue tile 0 (V loads to registers first, then async K).
   {
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < NVC; ++i) {
       int c = vc0 + i;
       vpre[i] = *(const s16x8*)(Vb + (long long)(c >> 4) * kv_rowstride +
                                 (c & 15) * 8);
     }
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      int row = 8 * w + 4 * i + (lane >> 4);
+    for (int i = 0; i < NKI; ++i) {
+      int row = (KVB / NW) * w + 4 * i + (lane >> 4);
       int chunk = lane & 15;
       const void* src = (const char*)(Kb + (long long)row * kv_rowstride) +
                         ((chunk ^ (row & 15)) << 4);
-      gload_lds16(src, (char*)k_lds[0] + (8 * w + 4 * i) * 256);
+      gload_lds16(src, (char*)k_lds[0] + ((KVB / NW) * w + 4 * i) * 256);
     }
   }
 
@@ -152,7 +199,7 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
     const int buf = kt & 1;
     // 1. commit the prefetched V registers into this tile's buffer.
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < NVC; ++i) {
       int c = vc0 + i;
       *(s16x8*)((char*)v_lds[buf] + 2 * vsub(c >> 4, (c & 15) * 8)) =
           vpre[i];
@@ -164,19 +211,20 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
     if (kt + 1 < n_tiles) {
       const long long nb = (long long)(kt + 1) * KVB;
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < NVC; ++i) {
         int c = vc0 + i;
         vpre[i] = *(const s16x8*)(
             Vb + (nb + (c >> 4)) * kv_rowstride + (c & 15) * 8);
       }
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        int row = 8 * w + 4 * i + (lane >> 4);
+      for (int i = 0; i < NKI; ++i) {
+        int row = (KVB / NW) * w + 4 * i + (lane >> 4);
         int chunk = lane & 15;
         const void* src =
             (const char*)(Kb + (nb + row) * kv_rowstride) +
             ((chunk ^ (row & 15)) << 4);
-        gload_lds16(src, (char*)k_lds[buf ^ 1] + (8 * w + 4 * i) * 256);
+        gload_lds16(src,
+                    (char*)k_lds[buf ^ 1] + ((KVB / NW) * w + 4 * i) * 256);
       }
     }
     if (kt >= w_tiles) continue;  // past this wave's diagonal
@@ -203,7 +251,7 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
 
     // ---- causal mask (finite big-negative so exp2 underflows to 0;
     // m_run is already real for every row because tile 0 is unmasked).
-    if (causal && kvbase + KVB - 1 > qbase + QBW * w) {
+    if (causal == 1 && kvbase + KVB - 1 > qbase + QBW * w) {
 #pragma unroll
       for (int n = 0; n < 2; ++n)
 #pragma unroll
@@ -213,13 +261,44 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
         }
     }
 
-    // ---- online softmax on the lane's 32 scores of q column `col`.
-    float pmax = st[0][0];
+    // ---- issue the first PV tr-read batch NOW: its ~latency hides
+    // under the (LGKM-free) softmax VALU work below.
+    const int kv0g = (g >> 1) << 3;
+    const int trd = ((g & 1) << 4) + ((lw & 3) << 2);
+    const int trr = lw >> 2;
+    const unsigned vbase = (unsigned)(size_t)((char*)v_lds[buf]);
+    tr4 tA0, tB0, tA1, tB1;  // 2-deep tr pipeline (static names, rule #20)
+#define V3_ISSUE(tA, tB, ks)                                              \
+  {                                                                       \
+    int kvr = (ks) * 16 + kv0g + trr;                                     \
+    ds_tr4_issue(&tA, vbase + 2 * vsub(kvr, trd),                         \
+                 vbase + 2 * vsub(kvr + 4, trd),                          \
+                 vbase + 2 * vsub(kvr, 32 + trd),                         \
+                 vbase + 2 * vsub(kvr + 4, 32 + trd));                    \
+    ds_tr4_issue(&tB, vbase + 2 * vsub(kvr, 64 + trd),                    \
+                 vbase + 2 * vsub(kvr + 4, 64 + trd),                     \
+                 vbase + 2 * vsub(kvr, 96 + trd),                         \
+                 vbase + 2 * vsub(kvr + 4, 96 + trd));                    \
+  }
+    V3_ISSUE(tA0, tB0, 0);
+
+    // ---- online softmax, latency-shaped:
+    //   - max is a TREE reduction (a linear fmax chain is 32 dependent
+    //     VALU ops ~128 cyc; the tree is depth 5),
+    //   - cross-half exchanges use permlane (VALU), not __shfl_xor's
+    //     ds_bpermute, so no compiler lgkmcnt(0) drains in-flight tr
+    //     reads,
+    //   - exp + sum are SLICED into the PV loop (8 values per k-step,
+    //     exactly the slice each P-pack consumes) where they hide under
+    //     the MFMA clusters (m214 "sm-split").
+    float tmax[16];
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int r = 0; r < 16; ++r) tmax[r] = fmaxf(st[0][r], st[1][r]);
 #pragma unroll
-      for (int r = 0; r < 16; ++r) pmax = fmaxf(pmax, st[n][r]);
-    pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
+    for (int s = 8; s > 0; s >>= 1)
+#pragma unroll
+      for (int r = 0; r < s; ++r) tmax[r] = fmaxf(tmax[r], tmax[r + s]);
+    float pmax = fmaxf(tmax[0], xhalf32(tmax[0]));
     // defer-max (T13): skip the O-rescale while the running max grows
     // by <= 8 (P bounded by 2^8; f32 accumulators absorb it).
     if (!__all(pmax - m_run <= 8.0f)) {
@@ -232,76 +311,79 @@ extern "C" __global__ __launch_bounds__(512, 2) void attn_fwd_v3_kernel(
       l_run *= corr;
       m_run = m_new;
     }
+
+    // exp + sum (pairwise sums keep the dependence shallow).
     float rs = 0.f;
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int n = 0; n < 2; ++n) {
+      float part = 0.f;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        float e = exp2f(st[n][r] - m_run);
-        st[n][r] = e;
-        rs += e;
+      for (int r = 0; r < 16; r += 2) {
+        float e0 = exp2f(st[n][r] - m_run);
+        float e1 = exp2f(st[n][r + 1] - m_run);
+        st[n][r] = e0;
+        st[n][r + 1] = e1;
+        part += e0 + e1;
       }
-    rs += __shfl_xor(rs, 32, 64);
+      rs += part;
+    }
+    rs += xhalf32(rs);
     l_run += rs;
 
-    // ---- PV: per k-step, issue the 8 tr reads for all four d-blocks,
-    // pack P under their latency, then 4 independent MFMA chains.
-    const int kv0g = (g >> 1) << 3;
-    const int trd = ((g & 1) << 4) + ((lw & 3) << 2);
-    const int trr = lw >> 2;
-#pragma unroll
-    for (int ks = 0; ks < 4; ++ks) {
-      tr4 t0, t1;
-      {
-        int kvr = ks * 16 + kv0g + trr;
-        unsigned a00 = (unsigned)(2 * vsub(kvr, 0 * 32 + trd));
-        unsigned a01 = (unsigned)(2 * vsub(kvr + 4, 0 * 32 + trd));
-        unsigned a10 = (unsigned)(2 * vsub(kvr, 1 * 32 + trd));
-        unsigned a11 = (unsigned)(2 * vsub(kvr + 4, 1 * 32 + trd));
-        unsigned a20 = (unsigned)(2 * vsub(kvr, 2 * 32 + trd));
-        unsigned a21 = (unsigned)(2 * vsub(kvr + 4, 2 * 32 + trd));
-        unsigned a30 = (unsigned)(2 * vsub(kvr, 3 * 32 + trd));
-        unsigned a31 = (unsigned)(2 * vsub(kvr + 4, 3 * 32 + trd));
-        unsigned base = (unsigned)(size_t)((char*)v_lds[buf]);
-        ds_tr4_issue(&t0, base + a00, base + a01, base + a10, base + a11);
-        ds_tr4_issue(&t1, base + a20, base + a21, base + a30, base + a31);
-      }
-      // pack P^T B-fragment while the tr reads are in flight (T12).
-      s16x8 pb;
-      {
-        const int e0 = (ks & 1) << 3;
-        const int n = ks >> 1;
-        unsigned w0a = cvtpk_bf16(st[n][e0 + 0], st[n][e0 + 1]);
-        unsigned w1a = cvtpk_bf16(st[n][e0 + 2], st[n][e0 + 3]);
-        unsigned w0b = cvtpk_bf16(st[n][e0 + 4], st[n][e0 + 5]);
-        unsigned w1b = cvtpk_bf16(st[n][e0 + 6], st[n][e0 + 7]);
-        auto p0 = __builtin_amdgcn_permlane32_swap(w0a, w0b, false, false);
-        auto p1 = __builtin_amdgcn_permlane32_swap(w1a, w1b, false, false);
-        union { unsigned u[4]; s16x8 v; } pk;
-        pk.u[0] = p0[0];
-        pk.u[1] = p1[0];
-        pk.u[2] = p0[1];
-        pk.u[3] = p1[1];
-        pb = pk.v;
-      }
-      lgkm_wait0_bind(&t0);
-      lgkm_wait0_bind(&t1);
-      union { unsigned long long u[2]; s16x8 v; } vf;
-      __builtin_amdgcn_s_setprio(1);
-      vf.u[0] = t0.d[0];
-      vf.u[1] = t0.d[1];
-      o_t[0] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[0]);
-      vf.u[0] = t0.d[2];
-      vf.u[1] = t0.d[3];
-      o_t[1] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[1]);
-      vf.u[0] = t1.d[0];
-      vf.u[1] = t1.d[1];
-      o_t[2] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[2]);
-      vf.u[0] = t1.d[2];
-      vf.u[1] = t1.d[3];
-      o_t[3] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[3]);
-      __builtin_amdgcn_s_setprio(0);
-    }
+    // ---- PV, 2-deep pipelined: batch ks+1's 8 tr reads are issued
+    // before batch ks's counted wait (lgkmcnt(8) leaves them in
+    // flight); P-pack runs under tr latency.
+#define V3_PACK(pb, ks)                                                   \
+  {                                                                       \
+    const int e0 = ((ks) & 1) << 3;                                       \
+    const int n = (ks) >> 1;                                              \
+    unsigned w0a = cvtpk_bf16(st[n][e0 + 0], st[n][e0 + 1]);              \
+    unsigned w1a = cvtpk_bf16(st[n][e0 + 2], st[n][e0 + 3]);              \
+    unsigned w0b = cvtpk_bf16(st[n][e0 + 4], st[n][e0 + 5]);              \
+    unsigned w1b = cvtpk_bf16(st[n][e0 + 6], st[n][e0 + 7]);              \
+    auto p0 = __builtin_amdgcn_permlane32_swap(w0a, w0b, false, false);   \
+    auto p1 = __builtin_amdgcn_permlane32_swap(w1a, w1b, false, false);   \
+    union { unsigned u[4]; s16x8 v; } pk;                                 \
+    pk.u[0] = p0[0];                                                      \
+    pk.u[1] = p1[0];                                                      \
+    pk.u[2] = p0[1];                                                      \
+    pk.u[3] = p1[1];                                                      \
+    pb = pk.v;                                                            \
+  }
+#define V3_MFMA(tA, tB, pb)                                               \
+  {                                                                       \
+    union { unsigned long long u[2]; s16x8 v; } vf;                       \
+    __builtin_amdgcn_s_setprio(1);                                        \
+    vf.u[0] = tA.d[0];                                                    \
+    vf.u[1] = tA.d[1];                                                    \
+    o_t[0] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[0]);            \
+    vf.u[0] = tA.d[2];                                                    \
+    vf.u[1] = tA.d[3];                                                    \
+    o_t[1] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[1]);            \
+    vf.u[0] = tB.d[0];                                                    \
+    vf.u[1] = tB.d[1];                                                    \
+    o_t[2] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[2]);            \
+    vf.u[0] = tB.d[2];                                                    \
+    vf.u[1] = tB.d[3];                                                    \
+    o_t[3] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[3]);            \
+    __builtin_amdgcn_s_setprio(0);                                        \
+  }
+    s16x8 pb0, pb1;
+    V3_PACK(pb0, 0);
+    V3_ISSUE(tA1, tB1, 1);
+    lgkm_wait8_bind2(&tA0, &tB0);
+    V3_MFMA(tA0, tB0, pb0);
+    V3_PACK(pb1, 1);
+    V3_ISSUE(tA0, tB0, 2);
+    lgkm_wait8_bind2(&tA1, &tB1);
+    V3_MFMA(tA1, tB1, pb1);
+    V3_PACK(pb0, 2);
+    V3_ISSUE(tA1, tB1, 3);
+    lgkm_wait8_bind2(&tA0, &tB0);
+    V3_MFMA(tA0, tB0, pb0);
+    V3_PACK(pb1, 3);
+    lgkm_wait0_bind2(&tA1, &tB1);
+    V3_MFMA(tA1, tB1, pb1);
   }
 
   // ---- epilogue: O[q][d] = O^T[d][q] / l; lse = ln2*m' + ln(l).
@@ -329,9 +411,26 @@ extern "C" void attn_fwd_v3_launch(const void* Q, const void* K,
                                    const void* V, void* O, float* lse, int B,
                                    int S, int Hq, int Hkv, float scale,
                                    bool causal, hipStream_t stream) {
-  dim3 grid(S / QBLK3, B * Hq);
-  hipLaunchKernelGGL(attn_fwd_v3_kernel, grid, dim3(512), 0, stream,
+  static const int nw = [] {
+    const char* e = getenv("SKY_ATTN_FWD_V3_NW");
+    return e ? atoi(e) : 4;
+  }();
+  static const int dbg = [] {
+    const char* e = getenv("SKY_ATTN_DEBUG_MODE");
+    return e ? atoi(e) : 0;
+  }();
+  const int cz = causal ? (dbg ? dbg : 1) : 0;
+  if (nw == 8) {
+    dim3 grid(S / (QBW * 8), B * Hq);
+    hipLaunchKernelGGL(attn_fwd_v3_t<8>, grid, dim3(512), 0, stream,
+                       (const unsigned short*)Q, (const unsigned short*)K,
+                       (const unsigned short*)V, (unsigned short*)O, lse, B,
+                       S, Hq, Hkv, scale, cz);
+    return;
+  }
+  dim3 grid(S / (QBW * 4), B * Hq);
+  hipLaunchKernelGGL(attn_fwd_v3_t<4>, grid, dim3(256), 0, stream,
                      (const unsigned short*)Q, (const unsigned short*)K,
                      (const unsigned short*)V, (unsigned short*)O, lse, B, S,
-                     Hq, Hkv, scale, causal ? 1 : 0);
+                     Hq, Hkv, scale, cz);
 }
