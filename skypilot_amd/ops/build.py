@@ -26,6 +26,7 @@ HIP_SOURCES = [
     "attention_fwd.hip",
     "attention_fwd_v3.hip",
     "attention_bwd.hip",
+    "attention_bwd_v3.hip",
     "attention_decode.hip",
     "swiglu.hip",
     "skinny_gemm.hip",

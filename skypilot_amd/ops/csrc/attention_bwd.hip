@@ -570,6 +570,13 @@ extern "C" __global__ void attn_bwd_dkv_tr_kernel(
     const unsigned short*, const float*, const float*, unsigned short*,
     unsigned short*, int, int, int, int, float, int);
 
+extern "C" void attn_bwd_v3_launch(const void* Q, const void* K,
+                                   const void* V, const void* dO,
+                                   const float* lse, const float* Dvec,
+                                   void* dQ, void* dK, void* dV, int B,
+                                   int S, int Hq, int Hkv, float scale,
+                                   bool causal, hipStream_t stream);
+
 extern "C" __global__ void attn_bwd_dkv_swapped_kernel(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const unsigned short*, const float*, const float*, unsigned short*,
@@ -590,6 +597,17 @@ extern "C" void attn_bwd_launch(const void* Q, const void* K, const void* V,
   hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3(membound_grid(rows, 4)),
                      dim3(256), 0, stream, (const unsigned short*)dO,
                      (const unsigned short*)O, Dvec, rows, Hq);
+  // v3 (32x32 deep-pipelined dkv/dq) is the round-2 default for the
+  // flagship shapes; SKY_ATTN_BWD_V3=0 for A/B.
+  static const int use_bwd_v3 = [] {
+    const char* e = getenv("SKY_ATTN_BWD_V3");
+    return e ? atoi(e) : 1;
+  }();
+  if (use_bwd_v3 && S % 128 == 0) {
+    attn_bwd_v3_launch(Q, K, V, dO, lse, Dvec, dQ, dK, dV, B, S, Hq, Hkv,
+                       scale, causal, stream);
+    return;
+  }
   // NOTE: attn_bwd_dkv_kernel32 (8-wave 32x32, half the instructions)
   // measured 148 vs 163 TF/s — its 512-thread block leaves 1 block/CU,
   // losing the cross-block barrier/compute overlap that two independent

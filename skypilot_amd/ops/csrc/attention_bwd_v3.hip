@@ -1,0 +1,499 @@
+// Flash-attention backward v3 — 32x32 MFMA deep-pipeline kernels.
+//
+// Round-2 rewrite of dkv/dq on the v3 structure (see attention_fwd_v3):
+// tr reads instead of transposed scatter staging, permlane P/dS
+// redistribution instead of LDS roundtrips, exp2-domain softmax.
+//
+// dkv: computes S UNSWAPPED (S[q][kv] = Q K^T with K as the B operand)
+// so the MFMA C-layout puts each lane's OWN kv column on the lane and q
+// on the registers; P^T / dS^T A-fragments for the dV/dK accumulation
+// then come from the SAME cvt_pk+permlane half-swap as the forward's
+// P^T B-fragments — the round-1 dkv LDS roundtrip (p_lds/ds_lds, two
+// swizzled write+read passes per tile) disappears entirely.  K and V
+// rows live in per-wave registers for the whole kernel (one block per
+// 128 kv rows); Q/dO tiles of 32 rows are staged per iteration in a
+// bank-swizzled subtile layout serving both b128 row-fragments and
+// hardware-transpose column-fragments.
+//
+// dq: a structural clone of the forward (block per 128 q rows, kv tiles
+// of 64 staged by global_load_lds, swapped S^T), with a second MFMA
+// pass for dP^T = V dO^T and the output pass dQ^T = K^T dS^T whose
+// A-fragments are tr reads of the K tile and B-fragments the permlane
+// pack of dS^T.
+//
+// Math (identical to v1, attention_bwd.hip):
+//   P = exp(scale*QK^T - lse); dP = dO V^T; dS = P*(dP - Dvec);
+//   dV += P^T dO; dK += scale * dS^T Q; dQ = scale * dS K.
+#include <cstdlib>
+
+#include "attn_v3.h"
+
+#define ATT_D 128
+
+// Q/dO tile layout for dkv: [32 q][128 d] in vsub subtiles with an
+// extra XOR on the d-chunk keyed by q>>2 — without it, b128 row-reads
+// (32 q rows at one fixed d-chunk) land 8-way on the same bank set.
+__device__ __forceinline__ int vsubz(int q, int d) {
+  return (((q >> 2) << 3) + ((d >> 4) ^ ((q >> 2) & 7))) * 64 +
+         ((q & 3) << 4) + (d & 15);
+}
+
+// ---------------------------------------------------------------------------
+// dkv kernel.  Grid: (S/128, B*Hkv); 256 threads (4 waves); wave w owns
+// kv rows [kvb+32w, kvb+32w+32).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dK, unsigned short* __restrict__ dV, int B,
+    int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ unsigned short q_lds[32 * ATT_D];     // 8 KB, vsubz layout
+  __shared__ unsigned short do_lds[32 * ATT_D];    // 8 KB
+  __shared__ unsigned short k_own[128 * ATT_D];    // 32 KB, block's K rows
+  __shared__ unsigned short v_own[128 * ATT_D];    // 32 KB, block's V rows
+  // (80 KB total: exactly two blocks per CU; lse/Dvec read from L2)
+
+  const int kt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int kvh = bh % Hkv;
+  const int group = Hq / Hkv;
+  const int kvb = kt * 128;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int col = lane & 31;
+  const int h = lane >> 5;
+  const int g = lane >> 4;
+  const int lw = lane & 15;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+
+  const int my_kv = kvb + 32 * w + col;
+
+  // Block's K/V rows staged ONCE into LDS (vsubz layout): keeping them
+  // in per-wave registers (64 VGPRs) spilled.  K is pre-scaled by
+  // scale*log2(e) during staging (it feeds ONLY the S pass; dK
+  // accumulates separately).  V raw.
+  const float ksc = scale * 1.44269504088896340736f;
+  {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int c = tid * 8 + i;  // 2048 16B chunks in a [128][128] tile
+      int kr = c >> 4;
+      int dch = (c & 15) * 8;
+      long long src = (long long)(kvb + kr) * kv_rowstride + dch;
+      s16x8 raw = *(const s16x8*)(Kb + src);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * ksc);
+      *(s16x8*)((char*)k_own + 2 * vsubz(kr, dch)) = raw;
+      *(s16x8*)((char*)v_own + 2 * vsubz(kr, dch)) =
+          *(const s16x8*)(Vb + src);
+    }
+  }
+
+  f32x16 dv_acc[4], dk_acc[4];
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dv_acc[dt][r] = 0.f;
+      dk_acc[dt][r] = 0.f;
+    }
+
+  const int qt0 = causal ? kvb / 32 : 0;
+  const int n_qt = S / 32;
+
+  for (int gi = 0; gi < group; ++gi) {
+    const int qh = kvh * group + gi;
+    const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+    const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+    const float* lse_b = lse + ((long long)b * Hq + qh) * S;
+    const float* dvec_b = Dvec + (long long)b * S * Hq + qh;
+
+    for (int qt = qt0; qt < n_qt; ++qt) {
+      const int qbase = qt * 32;
+      __syncthreads();
+      // ---- stage Q/dO tile (2 x 16B chunks per thread) + lse/Dvec.
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int c = tid * 2 + i;
+        int qr = c >> 4;
+        int dch = (c & 15) * 8;
+        long long src = (long long)(qbase + qr) * q_rowstride + dch;
+        *(s16x8*)((char*)q_lds + 2 * vsubz(qr, dch)) =
+            *(const s16x8*)(Qb + src);
+        *(s16x8*)((char*)do_lds + 2 * vsubz(qr, dch)) =
+            *(const s16x8*)(dOb + src);
+      }
+      __syncthreads();
+      // causal: this wave's kv rows see q tiles >= its diagonal only.
+      if (causal && qbase + 31 < kvb + 32 * w) continue;
+
+      // ---- S[q][kv_own] and dP[q][kv_own]: A = Q/dO row-fragments
+      // from LDS, B = K'/V register fragments.  Two interleaved
+      // accumulator chains.
+      f32x16 st, dpt;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        st[r] = 0.f;
+        dpt[r] = 0.f;
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 8; ++ks) {
+        s16x8 aq = *(const s16x8*)(
+            (char*)q_lds + 2 * vsubz(col, ks * 16 + h * 8));
+        s16x8 kfr = *(const s16x8*)(
+            (char*)k_own + 2 * vsubz(32 * w + col, ks * 16 + h * 8));
+        st = MFMA32V3(as_bf16x8(aq), as_bf16x8(kfr), st);
+        s16x8 ad = *(const s16x8*)(
+            (char*)do_lds + 2 * vsubz(col, ks * 16 + h * 8));
+        s16x8 vfr = *(const s16x8*)(
+            (char*)v_own + 2 * vsubz(32 * w + col, ks * 16 + h * 8));
+        dpt = MFMA32V3(as_bf16x8(ad), as_bf16x8(vfr), dpt);
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- P = exp2(S' - lse'); dS = P * (dP - Dvec).  q is the
+      // register row, kv the lane column; causal mask only on the
+      // diagonal q-tile band.
+      const bool need_mask = causal && qbase < kvb + 128;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int ql = (r & 3) + ((r >> 2) << 3) + (h << 2);
+        float lv = lse_b[qbase + ql];
+        float l2 = (lv == -INFINITY) ? 3.0e37f
+                                     : lv * 1.44269504088896340736f;
+        float p = exp2f(st[r] - l2);
+        if (need_mask && qbase + ql < my_kv) p = 0.f;
+        st[r] = p;
+        dpt[r] = p * (dpt[r] - dvec_b[(long long)(qbase + ql) * Hq]);
+      }
+
+      // ---- dV += P^T dO ; dK += dS^T Q.  A-fragments come straight
+      // from the C-layout registers via the permlane half-swap pack;
+      // B-fragments are tr reads of the dO / Q tiles.
+      s16x8 pa, da;
+      tr4 t0, t1;
+#define DKV_TR_ISSUE(lds_base, ks)                                        \
+  {                                                                       \
+    int q_s = (ks) * 16 + ((g >> 1) << 3) + (lw >> 2);                    \
+    int d_s = ((g & 1) << 4) + ((lw & 3) << 2);                           \
+    unsigned base = (unsigned)(size_t)((char*)(lds_base));                \
+    ds_tr4_issue(&t0, base + 2 * vsubz(q_s, d_s),                         \
+                 base + 2 * vsubz(q_s + 4, d_s),                          \
+                 base + 2 * vsubz(q_s, 32 + d_s),                         \
+                 base + 2 * vsubz(q_s + 4, 32 + d_s));                    \
+    ds_tr4_issue(&t1, base + 2 * vsubz(q_s, 64 + d_s),                    \
+                 base + 2 * vsubz(q_s + 4, 64 + d_s),                     \
+                 base + 2 * vsubz(q_s, 96 + d_s),                         \
+                 base + 2 * vsubz(q_s + 4, 96 + d_s));                    \
+  }
+#define DKV_MFMA(acc, afrag)                                              \
+  {                                                                       \
+    union { unsigned long long u[2]; s16x8 v; } bf;                       \
+    __builtin_amdgcn_s_setprio(1);                                        \
+    bf.u[0] = t0.d[0];                                                    \
+    bf.u[1] = t0.d[1];                                                    \
+    acc[0] = MFMA32V3(as_bf16x8(afrag), as_bf16x8(bf.v), acc[0]);         \
+    bf.u[0] = t0.d[2];                                                    \
+    bf.u[1] = t0.d[3];                                                    \
+    acc[1] = MFMA32V3(as_bf16x8(afrag), as_bf16x8(bf.v), acc[1]);         \
+    bf.u[0] = t1.d[0];                                                    \
+    bf.u[1] = t1.d[1];                                                    \
+    acc[2] = MFMA32V3(as_bf16x8(afrag), as_bf16x8(bf.v), acc[2]);         \
+    bf.u[0] = t1.d[2];                                                    \
+    bf.u[1] = t1.d[3];                                                    \
+    acc[3] = MFMA32V3(as_bf16x8(afrag), as_bf16x8(bf.v), acc[3]);         \
+    __builtin_amdgcn_s_setprio(0);                                        \
+  }
+      // ks = 0: q rows 0..15; ks = 1: q rows 16..31.
+      DKV_TR_ISSUE(do_lds, 0);
+      V3_PACK1(pa, st, 0);
+      lgkm_wait0_bind2(&t0, &t1);
+      DKV_MFMA(dv_acc, pa);
+      DKV_TR_ISSUE(do_lds, 1);
+      V3_PACK1(pa, st, 1);
+      lgkm_wait0_bind2(&t0, &t1);
+      DKV_MFMA(dv_acc, pa);
+      DKV_TR_ISSUE(q_lds, 0);
+      V3_PACK1(da, dpt, 0);
+      lgkm_wait0_bind2(&t0, &t1);
+      DKV_MFMA(dk_acc, da);
+      DKV_TR_ISSUE(q_lds, 1);
+      V3_PACK1(da, dpt, 1);
+      lgkm_wait0_bind2(&t0, &t1);
+      DKV_MFMA(dk_acc, da);
+    }
+    // group loop: reset nothing (dv/dk keep accumulating); barrier
+    // protects the next gi's staging via loop-top __syncthreads.
+  }
+
+  // ---- epilogue: C-layout [kv row][d col]; dK gets the outer scale.
+  unsigned short* dKb = dK + ((long long)b * S * Hkv + kvh) * ATT_D;
+  unsigned short* dVb = dV + ((long long)b * S * Hkv + kvh) * ATT_D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kvrow = kvb + 32 * w + (r & 3) + ((r >> 2) << 3) + (h << 2);
+    unsigned short* krow = dKb + (long long)kvrow * kv_rowstride;
+    unsigned short* vrow = dVb + (long long)kvrow * kv_rowstride;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      krow[dt * 32 + col] = f2bf(dk_acc[dt][r] * scale);
+      vrow[dt * 32 + col] = f2bf(dv_acc[dt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dq kernel.  Grid: (S/128, B*Hq); 256 threads (4 waves); wave w owns q
+// rows [qb+32w, qb+32w+32).  kv tiles of 64 staged by global_load_lds
+// into the forward's linear+swzK16 layout (K is read both as b128
+// row-fragments for S^T and as tr column-fragments for dQ^T = K^T dS^T).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
+    const unsigned short* __restrict__ Q, const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, const unsigned short* __restrict__ dO,
+    const float* __restrict__ lse, const float* __restrict__ Dvec,
+    unsigned short* __restrict__ dQ, int B, int S, int Hq, int Hkv,
+    float scale, int causal) {
+  __shared__ unsigned short k_lds[2][64 * ATT_D];  // 2 x 16 KB
+  __shared__ unsigned short v_lds[2][64 * ATT_D];  // 2 x 16 KB
+
+  const int qt = gridDim.x - 1 - blockIdx.x;  // heavy blocks first
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int qh = bh % Hq;
+  const int kvh = qh / (Hq / Hkv);
+  const int qbase = qt * 128;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  const int col = lane & 31;
+  const int h = lane >> 5;
+  const int g = lane >> 4;
+  const int lw = lane & 15;
+
+  const long long q_rowstride = (long long)Hq * ATT_D;
+  const long long kv_rowstride = (long long)Hkv * ATT_D;
+  const unsigned short* Qb = Q + ((long long)b * S * Hq + qh) * ATT_D;
+  const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
+  const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
+
+  const int my_q = qbase + 32 * w + col;
+
+  // Q as B-fragments pre-scaled by scale*log2(e); dO fragments are
+  // re-read from L2 per kv tile (keeping them resident spilled).
+  const float qs = scale * 1.44269504088896340736f;
+  s16x8 q_b[8];
+  {
+    const unsigned short* src = Qb + (long long)my_q * q_rowstride;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks) {
+      s16x8 raw = *(const s16x8*)(src + ks * 16 + h * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * qs);
+      q_b[ks] = raw;
+    }
+  }
+  const unsigned short* do_src = dOb + (long long)my_q * q_rowstride;
+  float lv = lse[((long long)b * Hq + qh) * S + my_q];
+  const float lse2 =
+      (lv == -INFINITY) ? 3.0e37f : lv * 1.44269504088896340736f;
+  const float dvq = Dvec[((long long)b * S + my_q) * Hq + qh];
+
+  f32x16 dq_acc[4];
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
+
+  const int n_tiles = causal ? (qbase + 128) / 64 : S / 64;
+  const int w_tiles = causal ? ((qbase + 32 * w + 31) >> 6) + 1 : n_tiles;
+
+  // staging: wave w stages rows [16w,16w+16) of both K and V via 4+4
+  // global_load_lds (linear dest, pre-swizzled source).
+#define DQ_STAGE(kvoff, bufi)                                             \
+  {                                                                       \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i) {                       \
+      int row = 16 * w + 4 * i + (lane >> 4);                             \
+      int chunk = lane & 15;                                              \
+      long long srcoff = (kvoff + row) * kv_rowstride;                    \
+      int co = (chunk ^ (row & 15)) << 4;                                 \
+      gload_lds16((const char*)(Kb + srcoff) + co,                        \
+                  (char*)k_lds[bufi] + (16 * w + 4 * i) * 256);           \
+      gload_lds16((const char*)(Vb + srcoff) + co,                        \
+                  (char*)v_lds[bufi] + (16 * w + 4 * i) * 256);           \
+    }                                                                     \
+  }
+  DQ_STAGE((long long)0, 0);
+
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int buf = kt & 1;
+    __syncthreads();  // drains this tile's global_load_lds (vmcnt)
+    if (kt + 1 < n_tiles) DQ_STAGE((long long)(kt + 1) * 64, buf ^ 1);
+    if (kt >= w_tiles) continue;
+
+    const int kvbase = kt * 64;
+    // ---- S^T = K' Q^T and dP^T = V dO^T (4 interleaved chains).
+    f32x16 st[2], dpt[2];
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        st[n][r] = 0.f;
+        dpt[n][r] = 0.f;
+      }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        int krow = n * 32 + col;
+        s16x8 akf = *(const s16x8*)(
+            (char*)k_lds[buf] + swzK16(krow * 256 + (ks * 2 + h) * 16, krow));
+        st[n] = MFMA32V3(as_bf16x8(akf), as_bf16x8(q_b[ks]), st[n]);
+        s16x8 avf = *(const s16x8*)(
+            (char*)v_lds[buf] + swzK16(krow * 256 + (ks * 2 + h) * 16, krow));
+        s16x8 dof = *(const s16x8*)(do_src + ks * 16 + h * 8);
+        dpt[n] = MFMA32V3(as_bf16x8(avf), as_bf16x8(dof), dpt[n]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- P^T = exp2(S' - lse'); dS^T = P^T * (dP^T - Dvec[q]).
+    if (causal && kvbase + 63 > qbase + 32 * w) {
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kv = kvbase + n * 32 + (r & 3) + ((r >> 2) << 3) + (h << 2);
+          if (kv > my_q) st[n][r] = -30000.f;
+        }
+    }
+#pragma unroll
+    for (int n = 0; n < 2; ++n)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = exp2f(st[n][r] - lse2);
+        dpt[n][r] = p * (dpt[n][r] - dvq);
+      }
+
+    // ---- dQ^T += K^T dS^T: A = tr reads of the K tile (column
+    // fragments over the swizzled layout), B = permlane pack of dS^T.
+    tr4 t0, t1;
+    s16x8 db;
+#define DQ_TR_ISSUE(ks)                                                   \
+  {                                                                       \
+    unsigned base = (unsigned)(size_t)((char*)k_lds[buf]);                \
+    int kv_s = (ks) * 16 + ((g >> 1) << 3) + (lw >> 2);                   \
+    int d_s0 = ((g & 1) << 4) + ((lw & 3) << 2);                          \
+    unsigned a0 = (unsigned)(kv_s * 256 +                                 \
+                             ((((d_s0) >> 3) ^ (kv_s & 15)) << 4) +       \
+                             ((d_s0 & 7) << 1));                          \
+    unsigned a1 = (unsigned)((kv_s + 4) * 256 +                           \
+                             ((((d_s0) >> 3) ^ ((kv_s + 4) & 15)) << 4) + \
+                             ((d_s0 & 7) << 1));                          \
+    unsigned a2 = (unsigned)(kv_s * 256 +                                 \
+                             ((((32 + d_s0) >> 3) ^ (kv_s & 15)) << 4) +  \
+                             ((d_s0 & 7) << 1));                          \
+    unsigned a3 = (unsigned)((kv_s + 4) * 256 +                           \
+                             ((((32 + d_s0) >> 3) ^ ((kv_s + 4) & 15))    \
+                              << 4) +                                     \
+                             ((d_s0 & 7) << 1));                          \
+    ds_tr4_issue(&t0, base + a0, base + a1, base + a2, base + a3);        \
+    unsigned a4 = (unsigned)(kv_s * 256 +                                 \
+                             ((((64 + d_s0) >> 3) ^ (kv_s & 15)) << 4) +  \
+                             ((d_s0 & 7) << 1));                          \
+    unsigned a5 = (unsigned)((kv_s + 4) * 256 +                           \
+                             ((((64 + d_s0) >> 3) ^ ((kv_s + 4) & 15))    \
+                              << 4) +                                     \
+                             ((d_s0 & 7) << 1));                          \
+    unsigned a6 = (unsigned)(kv_s * 256 +                                 \
+                             ((((96 + d_s0) >> 3) ^ (kv_s & 15)) << 4) +  \
+                             ((d_s0 & 7) << 1));                          \
+    unsigned a7 = (unsigned)((kv_s + 4) * 256 +                           \
+                             ((((96 + d_s0) >> 3) ^ ((kv_s + 4) & 15))    \
+                              << 4) +                                     \
+                             ((d_s0 & 7) << 1));                          \
+    ds_tr4_issue(&t1, base + a4, base + a5, base + a6, base + a7);        \
+  }
+#define DQ_MFMA(bfrag)                                                    \
+  {                                                                       \
+    union { unsigned long long u[2]; s16x8 v; } af;                       \
+    __builtin_amdgcn_s_setprio(1);                                        \
+    af.u[0] = t0.d[0];                                                    \
+    af.u[1] = t0.d[1];                                                    \
+    dq_acc[0] = MFMA32V3(as_bf16x8(af.v), as_bf16x8(bfrag), dq_acc[0]);   \
+    af.u[0] = t0.d[2];                                                    \
+    af.u[1] = t0.d[3];                                                    \
+    dq_acc[1] = MFMA32V3(as_bf16x8(af.v), as_bf16x8(bfrag), dq_acc[1]);   \
+    af.u[0] = t1.d[0];                                                    \
+    af.u[1] = t1.d[1];                                                    \
+    dq_acc[2] = MFMA32V3(as_bf16x8(af.v), as_bf16x8(bfrag), dq_acc[2]);   \
+    af.u[0] = t1.d[2];                                                    \
+    af.u[1] = t1.d[3];                                                    \
+    dq_acc[3] = MFMA32V3(as_bf16x8(af.v), as_bf16x8(bfrag), dq_acc[3]);   \
+    __builtin_amdgcn_s_setprio(0);                                        \
+  }
+    DQ_TR_ISSUE(0);
+    V3_PACK2(db, dpt, 0);
+    lgkm_wait0_bind2(&t0, &t1);
+    DQ_MFMA(db);
+    DQ_TR_ISSUE(1);
+    V3_PACK2(db, dpt, 1);
+    lgkm_wait0_bind2(&t0, &t1);
+    DQ_MFMA(db);
+    DQ_TR_ISSUE(2);
+    V3_PACK2(db, dpt, 2);
+    lgkm_wait0_bind2(&t0, &t1);
+    DQ_MFMA(db);
+    DQ_TR_ISSUE(3);
+    V3_PACK2(db, dpt, 3);
+    lgkm_wait0_bind2(&t0, &t1);
+    DQ_MFMA(db);
+  }
+
+  // ---- epilogue: dQ[q][d] = scale * dQ^T[d][q].
+  unsigned short* dQb = dQ + ((long long)b * S * Hq + qh) * ATT_D;
+  unsigned short* qrow = dQb + (long long)my_q * q_rowstride;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int rq = 0; rq < 4; ++rq) {
+      s16x4 ov;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        ov[r] = (short)f2bf(dq_acc[dt][rq * 4 + r] * scale);
+      *(s16x4*)(qrow + dt * 32 + rq * 8 + h * 4) = ov;
+    }
+}
+
+extern "C" void attn_bwd_v3_launch(const void* Q, const void* K,
+                                   const void* V, const void* dO,
+                                   const float* lse, const float* Dvec,
+                                   void* dQ, void* dK, void* dV, int B,
+                                   int S, int Hq, int Hkv, float scale,
+                                   bool causal, hipStream_t stream) {
+  dim3 gkv(S / 128, B * Hkv);
+  hipLaunchKernelGGL(attn_bwd_dkv_v3_kernel, gkv, dim3(256), 0, stream,
+                     (const unsigned short*)Q, (const unsigned short*)K,
+                     (const unsigned short*)V, (const unsigned short*)dO,
+                     lse, Dvec, (unsigned short*)dK, (unsigned short*)dV, B,
+                     S, Hq, Hkv, scale, causal ? 1 : 0);
+  dim3 gq(S / 128, B * Hq);
+  hipLaunchKernelGGL(attn_bwd_dq_v3_kernel, gq, dim3(256), 0, stream,
+                     (const unsigned short*)Q, (const unsigned short*)K,
+                     (const unsigned short*)V, (const unsigned short*)dO,
+                     lse, Dvec, (unsigned short*)dQ, B, S, Hq, Hkv, scale,
+                     causal ? 1 : 0);
+}

@@ -32,74 +32,7 @@
 #define NWV3 8
 #define QBLK3 (QBW * NWV3)  // 256 q rows per block (8-wave variant)
 
-typedef __attribute__((ext_vector_type(16))) float f32x16;
-#define MFMA32V3(a, b, c) \
-  __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0)
-
-// 16-slot XOR swizzle for the K tile (rows are 256 B = 16 chunks): a
-// wave's A-fragment read (32 rows, fixed 16 B chunk) spreads over all
-// 16 slots -> 2 lanes/slot = free (guide §6 G4: 2-way is 1.02x).
-__device__ __forceinline__ int swzK16(int byte_off, int row) {
-  return byte_off ^ ((row & 15) << 4);
-}
-
-// V subtile layout: element index of V[kv][d] inside the
-// [kv/4][d/16][4][16] tile.  tr-read gather addresses land on
-// consecutive 8 B slots per 16-lane group (conflict-free); the staging
-// s16x8 writes stay 16 B contiguous because d moves within one subtile
-// row.
-__device__ __forceinline__ int vsub(int kv, int d) {
-  return (((kv >> 2) << 3) + (d >> 4)) * 64 + ((kv & 3) << 4) + (d & 15);
-}
-
-__device__ __forceinline__ unsigned cvtpk_bf16(float a, float b) {
-  unsigned r;
-  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
-  return r;
-}
-
-// lane <-> lane+32 exchange via permlane32_swap: pure VALU (unlike
-// __shfl_xor's ds_bpermute, which is an LGKM op whose compiler-inserted
-// lgkmcnt(0) would drain our in-flight tr reads mid-softmax).
-__device__ __forceinline__ float xhalf32(float x) {
-  union { float f; unsigned u; } c;
-  c.f = x;
-  auto p = __builtin_amdgcn_permlane32_swap(c.u, c.u, false, false);
-  union { unsigned u; float f; } r0, r1;
-  r0.u = p[0];
-  r1.u = p[1];
-  return (threadIdx.x & 32) ? r0.f : r1.f;
-}
-
-// Counted wait for a 2-deep tr-read pipeline: block until only the most
-// recent 8 lgkm ops (the next batch's tr reads) remain outstanding, and
-// bind the dependency to this batch's 8 destination registers.
-__device__ __forceinline__ void lgkm_wait8_bind2(tr4* a, tr4* b) {
-  asm volatile("s_waitcnt lgkmcnt(8)"
-               : "+v"(a->d[0]), "+v"(a->d[1]), "+v"(a->d[2]),
-                 "+v"(a->d[3]), "+v"(b->d[0]), "+v"(b->d[1]),
-                 "+v"(b->d[2]), "+v"(b->d[3])
-               :
-               : "memory");
-  __builtin_amdgcn_sched_barrier(0);
-}
-__device__ __forceinline__ void lgkm_wait0_bind2(tr4* a, tr4* b) {
-  asm volatile("s_waitcnt lgkmcnt(0)"
-               : "+v"(a->d[0]), "+v"(a->d[1]), "+v"(a->d[2]),
-                 "+v"(a->d[3]), "+v"(b->d[0]), "+v"(b->d[1]),
-                 "+v"(b->d[2]), "+v"(b->d[3])
-               :
-               : "memory");
-  __builtin_amdgcn_sched_barrier(0);
-}
-
-// Async 16B global->LDS copy: per-lane global source, wave-uniform LDS
-// base + lane*16 destination (guide §5: the only supported dest form).
-__device__ __forceinline__ void gload_lds16(const void* g, void* l) {
-  __builtin_amdgcn_global_load_lds(
-      (const __attribute__((address_space(1))) unsigned int*)g,
-      (__attribute__((address_space(3))) unsigned int*)l, 16, 0, 0);
-}
+#include "attn_v3.h"
 
 // NW = waves per block (8 -> one 512-thread block/CU; 4 -> two
 // independent 256-thread blocks/CU whose phases interleave freely —
@@ -333,23 +266,6 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
     // ---- PV, 2-deep pipelined: batch ks+1's 8 tr reads are issued
     // before batch ks's counted wait (lgkmcnt(8) leaves them in
     // flight); P-pack runs under tr latency.
-#define V3_PACK(pb, ks)                                                   \
-  {                                                                       \
-    const int e0 = ((ks) & 1) << 3;                                       \
-    const int n = (ks) >> 1;                                              \
-    unsigned w0a = cvtpk_bf16(st[n][e0 + 0], st[n][e0 + 1]);              \
-    unsigned w1a = cvtpk_bf16(st[n][e0 + 2], st[n][e0 + 3]);              \
-    unsigned w0b = cvtpk_bf16(st[n][e0 + 4], st[n][e0 + 5]);              \
-    unsigned w1b = cvtpk_bf16(st[n][e0 + 6], st[n][e0 + 7]);              \
-    auto p0 = __builtin_amdgcn_permlane32_swap(w0a, w0b, false, false);   \
-    auto p1 = __builtin_amdgcn_permlane32_swap(w1a, w1b, false, false);   \
-    union { unsigned u[4]; s16x8 v; } pk;                                 \
-    pk.u[0] = p0[0];                                                      \
-    pk.u[1] = p1[0];                                                      \
-    pk.u[2] = p0[1];                                                      \
-    pk.u[3] = p1[1];                                                      \
-    pb = pk.v;                                                            \
-  }
 #define V3_MFMA(tA, tB, pb)                                               \
   {                                                                       \
     union { unsigned long long u[2]; s16x8 v; } vf;                       \
@@ -369,19 +285,19 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
     __builtin_amdgcn_s_setprio(0);                                        \
   }
     s16x8 pb0, pb1;
-    V3_PACK(pb0, 0);
+    V3_PACK2(pb0, st, 0);
     V3_ISSUE(tA1, tB1, 1);
     lgkm_wait8_bind2(&tA0, &tB0);
     V3_MFMA(tA0, tB0, pb0);
-    V3_PACK(pb1, 1);
+    V3_PACK2(pb1, st, 1);
     V3_ISSUE(tA0, tB0, 2);
     lgkm_wait8_bind2(&tA1, &tB1);
     V3_MFMA(tA1, tB1, pb1);
-    V3_PACK(pb0, 2);
+    V3_PACK2(pb0, st, 2);
     V3_ISSUE(tA1, tB1, 3);
     lgkm_wait8_bind2(&tA0, &tB0);
     V3_MFMA(tA0, tB0, pb0);
-    V3_PACK(pb1, 3);
+    V3_PACK2(pb1, st, 3);
     lgkm_wait0_bind2(&tA1, &tB1);
     V3_MFMA(tA1, tB1, pb1);
   }
