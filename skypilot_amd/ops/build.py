@@ -30,6 +30,7 @@ HIP_SOURCES = [
     "attention_decode.hip",
     "swiglu.hip",
     "skinny_gemm.hip",
+    "wgrad_gemm.hip",
     "decode_fused.hip",
     "mfma_probe.hip",
 ]

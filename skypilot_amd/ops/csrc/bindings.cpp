@@ -44,6 +44,9 @@ void skinny_gemm_launch(const void*, const void*, void*, int, int, int,
                         hipStream_t);
 void skinny_gemm_swiglu_launch(const void*, const void*, void*, int, int,
                                int, hipStream_t);
+void transpose_bf16_launch(const void*, void*, int, int, hipStream_t);
+void gemm_nt_launch(const void*, const void*, void*, int, int, long long,
+                    hipStream_t);
 void adamw_mt_launch(const void*, const void*, const void*, const void*,
                      const void*, long long, float, float, float, float,
                      int, float, hipStream_t);
@@ -226,6 +229,43 @@ torch::Tensor skinny_gemm_swiglu(torch::Tensor GU, torch::Tensor W) {
   return y;
 }
 
+// Transpose-then-NT weight-gradient GEMM (wgrad_gemm.hip).
+torch::Tensor transpose_bf16(torch::Tensor X) {
+  CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
+  int R = X.size(0), C = X.size(1);
+  TORCH_CHECK(R % 64 == 0 && C % 64 == 0, "transpose_bf16: dims % 64");
+  auto Y = torch::empty({C, R}, X.options());
+  transpose_bf16_launch(X.data_ptr(), Y.data_ptr(), R, C, cur_stream());
+  return Y;
+}
+
+torch::Tensor gemm_nt(torch::Tensor A, torch::Tensor B) {
+  // C[i,j] = sum_m A[i,m] * B[j,m]
+  CHECK_GPU(A); CHECK_CONTIG(A); CHECK_BF16(A);
+  CHECK_GPU(B); CHECK_CONTIG(B); CHECK_BF16(B);
+  long long M = A.size(1);
+  int I = A.size(0), J = B.size(0);
+  TORCH_CHECK(B.size(1) == M, "gemm_nt: inner dims mismatch");
+  TORCH_CHECK(M % 64 == 0, "gemm_nt: M % 64 != 0");
+  TORCH_CHECK(I % 128 == 0 && J % 256 == 0, "gemm_nt: I%128/J%256");
+  auto C = torch::empty({I, J}, A.options());
+  gemm_nt_launch(A.data_ptr(), B.data_ptr(), C.data_ptr(), I, J, M,
+                 cur_stream());
+  return C;
+}
+
+torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x) {
+  // dW[i,j] = sum_m dy[m,i] * x[m,j]  (nn.Linear weight grad).
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  long long M = dy.size(0);
+  int I = dy.size(1), J = x.size(1);
+  TORCH_CHECK(x.size(0) == M, "wgrad_tn: token dims mismatch");
+  auto dyT = transpose_bf16(dy);
+  auto xT = transpose_bf16(x);
+  return gemm_nt(dyT, xT);
+}
+
 torch::Tensor skinny_gemm(torch::Tensor X, torch::Tensor W) {
   // Y[N,O] = X[N,I] @ W[O,I]^T (decode GEMV; see skinny_gemm.hip)
   CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
@@ -344,6 +384,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_decode", &attn_decode);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("transpose_bf16", &transpose_bf16);
+  m.def("gemm_nt", &gemm_nt);
+  m.def("wgrad_tn", &wgrad_tn);
   m.def("adamw_step_mt", &adamw_step_mt);
   m.def("skinny_gemm_swiglu", &skinny_gemm_swiglu);
   m.def("rmsnorm_res", &rmsnorm_res);

@@ -357,3 +357,20 @@ def test_skinny_gemm_swiglu_matches_ref():
         x = torch.nn.functional.silu(g) * u
         ref = torch.nn.functional.linear(x, w.float())
         torch.testing.assert_close(y.float(), ref, atol=0.05, rtol=0.05)
+
+
+@pytest.mark.gpu
+def test_wgrad_tn_matches_ref():
+    """Transpose-then-NT wgrad GEMM vs fp32 reference."""
+    C = ops.native()
+    torch.manual_seed(5)
+    for (M, I, J) in [(256, 128, 256), (512, 256, 512), (1024, 384, 256)]:
+        dy = (torch.randn(M, I, device="cuda") * 0.3).bfloat16()
+        x = (torch.randn(M, J, device="cuda") * 0.3).bfloat16()
+        out = C.wgrad_tn(dy, x)
+        ref = dy.t().float() @ x.float()
+        torch.testing.assert_close(out.float(), ref, atol=0.5, rtol=0.02,
+                                   msg=str((M, I, J)))
+        # transpose alone
+        t = C.transpose_bf16(dy)
+        assert torch.equal(t, dy.t().contiguous())
