@@ -12,9 +12,10 @@ from skypilot_amd.exceptions import ClusterNotUpError
 
 class AgentClient:
     def __init__(self, port: int, host: str = "127.0.0.1",
-                 timeout: float = 30.0):
+                 timeout: float = 30.0, token: Optional[str] = None):
         self.base = f"http://{host}:{port}"
-        self._client = httpx.Client(timeout=timeout)
+        headers = {"Authorization": f"Bearer {token}"} if token else {}
+        self._client = httpx.Client(timeout=timeout, headers=headers)
 
     def close(self):
         self._client.close()

@@ -29,8 +29,25 @@ SCHEDULER_INTERVAL = 1.0
 AUTOSTOP_INTERVAL = 30.0
 
 
-def create_app(cluster_dir: str, gpu_ids: list[int]) -> FastAPI:
+def create_app(cluster_dir: str, gpu_ids: list[int],
+               token: str | None = None) -> FastAPI:
     app = FastAPI()
+
+    if token:
+        # Per-cluster shared-secret auth (reference: skylet gRPC is only
+        # reachable through an SSH tunnel — cloud_vm_ray_backend.py:2414;
+        # an HTTP agent on a pod network needs its own gate).
+        from fastapi import Request
+        from fastapi.responses import JSONResponse
+
+        @app.middleware("http")
+        async def _auth(request: Request, call_next):
+            if request.url.path != "/health":
+                auth = request.headers.get("authorization", "")
+                if auth != f"Bearer {token}":
+                    return JSONResponse({"detail": "agent token required"},
+                                        status_code=401)
+            return await call_next(request)
     table = job_lib.JobTable(cluster_dir)
     state = {"autostop_idle_minutes": -1, "autostop_down": False,
              "last_active": time.time(), "autostopping": False,
@@ -211,7 +228,12 @@ def main():
                          "peer agents are reachable at podIP)")
     args = ap.parse_args()
     gpu_ids = [int(g) for g in args.gpu_ids.split(",") if g != ""]
-    app = create_app(args.cluster_dir, gpu_ids)
+    # Shared secret via env (not argv: visible in ps) or cluster file.
+    token = os.environ.get("SKY_AMD_AGENT_TOKEN")
+    if not token:
+        tf = Path(args.cluster_dir) / "agent_token"
+        token = tf.read_text().strip() if tf.exists() else None
+    app = create_app(args.cluster_dir, gpu_ids, token=token)
     # Record readiness for the provisioner.
     Path(args.cluster_dir).mkdir(parents=True, exist_ok=True)
     (Path(args.cluster_dir) / "agent.json").write_text(

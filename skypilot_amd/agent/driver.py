@@ -121,7 +121,8 @@ def run_job(cluster_dir: str, job_id: int) -> int:
         if node_rank > 0 and node_rank <= len(peer_agents):
             from skypilot_amd.agent.client import AgentClient
             host, _, port = peer_agents[node_rank - 1].rpartition(":")
-            peer = AgentClient(int(port), host=host)
+            peer = AgentClient(int(port), host=host,
+                               token=spec.get("agent_token"))
             leaf_env = {k: v for k, v in env.items()
                         if k.startswith(("SKYPILOT_", "MASTER_"))}
             leaf_env.update(envs)

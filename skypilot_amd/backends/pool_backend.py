@@ -28,7 +28,7 @@ class PoolBackend(Backend):
         port = handle.get("agent_port")
         if not port:
             raise ClusterNotUpError("cluster has no agent")
-        return AgentClient(port)
+        return AgentClient(port, token=handle.get("agent_token"))
 
     # ---- provisioning -----------------------------------------------------
     def provision(self, task: Task, cluster_name: str,
@@ -105,6 +105,7 @@ class PoolBackend(Backend):
             "workdir": str(Path(handle["cluster_dir"]) / "workdir"),
             "node_ips": handle.get("node_ips"),
             "peer_agents": handle.get("peer_agents"),
+            "agent_token": handle.get("agent_token"),
             "master_addr": handle.get("master_addr"),
             "task_id": task_id,
             "managed_job_id": managed_job_id,

@@ -192,10 +192,69 @@ def cancel_request(request_id: str) -> bool:
         return r.json().get("cancelled", False)
 
 
+UPLOAD_CHUNK_BYTES = 8 << 20
+
+
+def upload_path(path: str) -> str:
+    """Tar+gzip a local directory (or file) and ship it to the API
+    server in chunks (reference: client/common.py:154-192 chunked
+    upload to /upload).  Returns the upload id for task markers."""
+    import tarfile
+    import tempfile
+    import uuid
+
+    upload_id = uuid.uuid4().hex
+    src = os.path.expanduser(path)
+    with tempfile.TemporaryDirectory() as td:
+        tarball = os.path.join(td, "u.tar.gz")
+        with tarfile.open(tarball, "w:gz") as tf:
+            if os.path.isdir(src):
+                for entry in sorted(os.listdir(src)):
+                    tf.add(os.path.join(src, entry), arcname=entry)
+            else:
+                tf.add(src, arcname=os.path.basename(src))
+        size = os.path.getsize(tarball)
+        total = max(1, (size + UPLOAD_CHUNK_BYTES - 1) // UPLOAD_CHUNK_BYTES)
+        with _client() as c, open(tarball, "rb") as f:
+            for i in range(total):
+                chunk = f.read(UPLOAD_CHUNK_BYTES)
+                r = c.post("/api/upload",
+                           params={"upload_id": upload_id,
+                                   "chunk_index": i,
+                                   "total_chunks": total},
+                           content=chunk)
+                if r.status_code != 200:
+                    raise ApiServerError(
+                        f"upload failed: {r.status_code} {r.text[:200]}")
+    return upload_id
+
+
+def _server_is_remote() -> bool:
+    """True when the API server is not on this host — local paths in the
+    task are then invisible to it and must be uploaded."""
+    if os.environ.get("SKY_AMD_FORCE_UPLOAD") == "1":
+        return True
+    if _TEST_CLIENT is not None:
+        return False
+    from urllib.parse import urlparse
+    host = urlparse(server_url()).hostname or "127.0.0.1"
+    return host not in ("127.0.0.1", "localhost", "::1")
+
+
 def _task_body(task) -> Dict[str, Any]:
-    if isinstance(task, Task):
-        return task.to_yaml_config()
-    return dict(task)
+    body = task.to_yaml_config() if isinstance(task, Task) else dict(task)
+    if _server_is_remote():
+        wd = body.get("workdir")
+        if isinstance(wd, str) and os.path.exists(os.path.expanduser(wd)):
+            body["workdir"] = {"upload": upload_path(wd)}
+        fm = body.get("file_mounts")
+        if fm:
+            body["file_mounts"] = {
+                k: ({"upload": upload_path(v)}
+                    if isinstance(v, str) and not v.startswith(("s3://",))
+                    and os.path.exists(os.path.expanduser(v)) else v)
+                for k, v in fm.items()}
+    return body
 
 
 # ---- public API (mirrors reference sdk surface) ---------------------------

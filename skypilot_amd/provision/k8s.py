@@ -54,7 +54,7 @@ def _kubectl_base(settings: Dict[str, Any]) -> List[str]:
     return cmd
 
 
-def render_pod_manifest(cluster_name: str, acc_count: int,
+def render_pod_manifest(cluster_name: str, token: str, acc_count: int,
                         settings: Dict[str, Any]) -> Dict[str, Any]:
     """Pod spec: one agent container with N AMD GPUs (reference pattern:
     provision/kubernetes/instance.py pod template)."""
@@ -82,7 +82,9 @@ def render_pod_manifest(cluster_name: str, acc_count: int,
                 "workingDir": REMOTE_REPO,
                 "resources": resources,
                 "env": [{"name": "HSA_ENABLE_IPC_MODE_LEGACY",
-                         "value": "0"}],
+                         "value": "0"},
+                        {"name": "SKY_AMD_AGENT_TOKEN",
+                         "value": token}],
             }],
         },
     }
@@ -126,9 +128,12 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator,
     (reference: multi-node pods via Ray in
     sky/provision/kubernetes/instance.py)."""
     settings = k8s_settings()
+    import secrets as _secrets
+    token = (existing_handle or {}).get("agent_token") or _secrets.token_hex(16)
     pods = [_pod_name(cluster_name, i) for i in range(num_nodes)]
     for pod in pods:
-        manifest = render_pod_manifest(cluster_name, acc_count, settings)
+        manifest = render_pod_manifest(cluster_name, token, acc_count,
+                                       settings)
         manifest["metadata"]["name"] = pod
         proc = _run_kubectl(["apply", "-f", "-"], settings,
                             input_text=yaml.safe_dump(manifest))
@@ -139,7 +144,7 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator,
     ips = [_wait_running(pod, settings, deadline) for pod in pods]
 
     local_port = _port_forward(pods[0], settings)
-    AgentClient(local_port).wait_ready(timeout=60)
+    AgentClient(local_port, token=token).wait_ready(timeout=60)
     cdir = global_state.root_dir() / "clusters" / cluster_name
     cdir.mkdir(parents=True, exist_ok=True)
     return {
@@ -156,6 +161,7 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator,
         "agent_port": local_port,
         "master_addr": ips[0] or "127.0.0.1",
         "peer_agents": [f"{ip}:{AGENT_PORT}" for ip in ips[1:]],
+        "agent_token": token,
     }
 
 

@@ -179,6 +179,7 @@ def _run_instances_locked(cluster_name, num_nodes, accelerator, acc_count,
     }
     handle["agent_port"] = _ensure_agent(cdir, gpu_ids,
                                          existing_handle or {})
+    handle["agent_token"] = _agent_token(cdir)
     return handle
 
 
@@ -223,8 +224,21 @@ def _agent_alive(cdir: Path) -> Optional[int]:
     return None
 
 
+def _agent_token(cdir: Path) -> str:
+    """Per-cluster shared secret for the agent HTTP surface."""
+    tf = cdir / "agent_token"
+    if tf.exists():
+        return tf.read_text().strip()
+    import secrets
+    tok = secrets.token_hex(16)
+    tf.touch(mode=0o600)
+    tf.write_text(tok)
+    return tok
+
+
 def _ensure_agent(cdir: Path, gpu_ids: List[int],
                   existing_handle: Dict) -> int:
+    tok = _agent_token(cdir)
     port = _agent_alive(cdir)
     if port is not None:
         return port
@@ -235,9 +249,10 @@ def _ensure_agent(cdir: Path, gpu_ids: List[int],
          "--cluster-dir", str(cdir), "--port", str(port),
          "--gpu-ids", ",".join(str(g) for g in gpu_ids)],
         stdout=log, stderr=subprocess.STDOUT, start_new_session=True,
-        env={**os.environ, "SKY_AMD_HOME": str(global_state.root_dir())})
+        env={**os.environ, "SKY_AMD_HOME": str(global_state.root_dir()),
+             "SKY_AMD_AGENT_TOKEN": tok})
     log.close()
-    AgentClient(port).wait_ready(timeout=30)
+    AgentClient(port, token=tok).wait_ready(timeout=30)
     return port
 
 
@@ -261,7 +276,8 @@ def _kill_agent(handle: Dict[str, Any]) -> None:
     port = handle.get("agent_port")
     if port:
         try:
-            AgentClient(port).cancel_all()
+            AgentClient(port,
+                        token=handle.get("agent_token")).cancel_all()
         except Exception:  # noqa: BLE001
             pass
     meta = Path(cdir) / "agent.json"

@@ -124,6 +124,17 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator,
     runner.run(f"mkdir -p {REMOTE_ROOT}", check=True, timeout=60)
     runner.rsync(str(repo_root) + "/", f"{REMOTE_ROOT}/repo")
     gpu_ids = ",".join(str(i) for i in range(acc_count))
+    # Per-cluster agent token (reference: skylet is only reachable via
+    # the SSH tunnel; the token also protects against other local users
+    # of the remote box).
+    import secrets as _secrets
+    token = _secrets.token_hex(16)
+    runner.run(f"mkdir -p {REMOTE_ROOT}/cluster && umask 077 && "
+               f"[ -s {REMOTE_ROOT}/cluster/agent_token ] || "
+               f"echo {token} > {REMOTE_ROOT}/cluster/agent_token",
+               check=True, timeout=60)
+    token = runner.run(f"cat {REMOTE_ROOT}/cluster/agent_token",
+                       check=True, timeout=60).stdout.strip()
     start_cmd = (
         f"cd {REMOTE_ROOT}/repo && "
         f"nohup python3 -m skypilot_amd.agent.daemon "
@@ -133,7 +144,7 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator,
     runner.run(f"pgrep -f 'skypilot_amd.agent.daemon.*{AGENT_PORT}' "
                f">/dev/null || ({start_cmd})", check=True, timeout=120)
     local_port = _tunnel(head, AGENT_PORT)
-    AgentClient(local_port).wait_ready(timeout=60)
+    AgentClient(local_port, token=token).wait_ready(timeout=60)
 
     cdir = global_state.root_dir() / "clusters" / cluster_name
     cdir.mkdir(parents=True, exist_ok=True)
@@ -148,6 +159,7 @@ def run_instances(cluster_name: str, num_nodes: int, accelerator,
         "head_ip": head["ip"],
         "node_ips": [h["ip"] for h in chosen],
         "agent_port": local_port,
+        "agent_token": token,
     }
 
 

@@ -136,6 +136,41 @@ def create_app(start_workers: bool = True) -> FastAPI:
         _require_admin(request)
         return {"revoked": users.revoke_token(body["name"])}
 
+    # ---- client file upload (reference: sky server /upload,/upload_v2
+    # server.py:1826,1885; chunked client in client/common.py:154-192).
+    @app.post("/api/upload")
+    async def api_upload(request: Request, upload_id: str,
+                         chunk_index: int = 0, total_chunks: int = 1):
+        import re as _re
+        import tarfile
+        _identity(request)
+        if not _re.fullmatch(r"[a-f0-9]{32}", upload_id):
+            raise HTTPException(400, "bad upload_id")
+        if not (0 <= chunk_index < total_chunks <= 10000):
+            raise HTTPException(400, "bad chunk bounds")
+        updir = global_state.root_dir() / "api" / "uploads"
+        updir.mkdir(parents=True, exist_ok=True)
+        part = updir / f".{upload_id}.part{chunk_index}"
+        part.write_bytes(await request.body())
+        parts = [updir / f".{upload_id}.part{i}"
+                 for i in range(total_chunks)]
+        if not all(p.exists() for p in parts):
+            return {"status": "uploading", "chunk_index": chunk_index}
+        tarball = updir / f".{upload_id}.tar.gz"
+        with open(tarball, "wb") as out:
+            for p in parts:
+                out.write(p.read_bytes())
+                p.unlink()
+        dest = updir / upload_id
+        dest.mkdir(parents=True, exist_ok=True)
+        with tarfile.open(tarball, "r:gz") as tf:
+            for m in tf.getmembers():  # path-traversal guard
+                if m.name.startswith("/") or ".." in m.name.split("/"):
+                    raise HTTPException(400, f"unsafe path {m.name!r}")
+            tf.extractall(dest)
+        tarball.unlink()
+        return {"status": "completed", "upload_id": upload_id}
+
     # ---- generic async request plumbing -----------------------------------
     @app.post(API_PREFIX + "/{name}")
     def submit(name: str, request: Request, body: Dict[str, Any] = None):

@@ -26,6 +26,24 @@ KNOWN_TASK_KEYS = {
 }
 
 
+
+def _resolve_upload(value):
+    """Resolve a client-upload marker {"upload": <id>} to the server's
+    extracted blob directory (see server /api/upload)."""
+    if isinstance(value, dict) and "upload" in value:
+        import re as _re
+        uid = str(value["upload"])
+        if not _re.fullmatch(r"[a-f0-9]{32}", uid):
+            raise TaskValidationError(f"bad upload id {uid!r}")
+        from skypilot_amd import global_state
+        path = global_state.root_dir() / "api" / "uploads" / uid
+        if not path.exists():
+            raise TaskValidationError(
+                f"upload {uid} not found on the API server")
+        return str(path)
+    return value
+
+
 class Task:
     def __init__(self,
                  name: Optional[str] = None,
@@ -113,6 +131,15 @@ class Task:
         envs = dict(config.get("envs") or {})
         if env_overrides:
             envs.update(env_overrides)
+        # Client-uploaded blobs (reference: sky server /upload +
+        # client/common.py:154-192): {"upload": <id>} markers in workdir
+        # and file_mounts resolve to the server-side extraction dir.
+        config = dict(config)
+        config["workdir"] = _resolve_upload(config.get("workdir"))
+        if config.get("file_mounts"):
+            config["file_mounts"] = {
+                k: _resolve_upload(v)
+                for k, v in config["file_mounts"].items()}
         # ${VAR} substitution in run/setup from envs (reference behavior is
         # shell-level; we keep envs as env vars, no substitution needed).
         task = cls(

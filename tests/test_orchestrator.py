@@ -284,3 +284,55 @@ def test_git_workdir(sky_env, client, tmp_path):
     assert (wd / "hello.txt").read_text() == "v1"
     assert (wd / ".git").exists()
     sdk.get(sdk.down("git-wd"))
+
+
+def test_client_upload_workdir(client, tmp_path, monkeypatch):
+    """A client not on the server host ships its workdir via chunked
+    /api/upload; the server resolves the marker and the job sees the
+    files (reference: sky server /upload + client/common.py chunked
+    upload).  SKY_AMD_FORCE_UPLOAD simulates the remote-client case."""
+    from skypilot_amd.client import sdk
+    wd = tmp_path / "proj"
+    wd.mkdir()
+    (wd / "data.txt").write_text("uploaded-content")
+    (wd / "sub").mkdir()
+    (wd / "sub" / "x.py").write_text("print('hi')")
+    monkeypatch.setenv("SKY_AMD_FORCE_UPLOAD", "1")
+    # small chunks force the multi-chunk path
+    monkeypatch.setattr(sdk, "UPLOAD_CHUNK_BYTES", 128)
+    rid = sdk.launch({"workdir": str(wd), "run": "cat data.txt"},
+                     "up-c")
+    sdk.get(rid, timeout=60)
+    j = _wait_job_done("up-c", 1)
+    assert j["status"] == "SUCCEEDED"
+    import os as _os
+    home = Path(_os.environ["SKY_AMD_HOME"])
+    cwd = home / "clusters" / "up-c" / "workdir"
+    assert (cwd / "data.txt").read_text() == "uploaded-content"
+    assert (cwd / "sub" / "x.py").exists()
+    sdk.get(sdk.down("up-c"))
+
+
+def test_agent_requires_token(client):
+    """The node agent rejects requests without the per-cluster bearer
+    token (unauthenticated agent endpoint)."""
+    import httpx
+    from skypilot_amd import global_state
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"run": "sleep 0.1", "resources": {"cpus": 1}},
+                       "tok-c"), timeout=60)
+    rec = global_state.get_cluster("tok-c")
+    port = rec["handle"]["agent_port"]
+    token = rec["handle"]["agent_token"]
+    assert token
+    # health open, everything else closed without the token
+    assert httpx.get(f"http://127.0.0.1:{port}/health").status_code == 200
+    r = httpx.post(f"http://127.0.0.1:{port}/jobs/queue",
+                   json={"name": "x", "spec": {"run": "true"}})
+    assert r.status_code == 401
+    assert httpx.get(f"http://127.0.0.1:{port}/jobs").status_code == 401
+    # with the token it works
+    r = httpx.get(f"http://127.0.0.1:{port}/jobs",
+                  headers={"Authorization": f"Bearer {token}"})
+    assert r.status_code == 200
+    sdk.get(sdk.down("tok-c"))
