@@ -384,10 +384,13 @@ def jobs_logs(job_id: int) -> str:
     return _submit("jobs_logs", {"job_id": job_id})
 
 
-def jobs_pool_apply(name: str, template, num_workers: int = 2) -> str:
-    return _submit("jobs_pool_apply", {"name": name,
-                                       "template": _task_body(template),
-                                       "num_workers": num_workers})
+def jobs_pool_apply(name: str, template, num_workers: int = 2,
+                    min_workers: Optional[int] = None,
+                    max_workers: Optional[int] = None) -> str:
+    return _submit("jobs_pool_apply", {
+        "name": name, "template": _task_body(template),
+        "num_workers": num_workers, "min_workers": min_workers,
+        "max_workers": max_workers})
 
 
 def jobs_pool_status(name: Optional[str] = None) -> str:

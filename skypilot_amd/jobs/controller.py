@@ -79,10 +79,18 @@ class JobController:
             # Pool mode (reference: `sky jobs pool`): claim a warm worker
             # and just exec — no provisioning on the job path.
             from skypilot_amd.jobs import pools
+            nudged = False
             while True:
                 cluster = pools.acquire(self.pool, self.job_id)
                 if cluster is not None:
                     break
+                if not nudged:
+                    # queue-length autoscaler (serve/autoscalers.py:1094)
+                    try:
+                        pools.autoscale(self.pool)
+                    except Exception:  # noqa: BLE001
+                        pass
+                    nudged = True
                 me = state.get(self.job_id)
                 if me and me["status"] == state.CANCELLED:
                     raise RuntimeError("cancelled while queued for pool")
@@ -181,7 +189,16 @@ class JobController:
 
 def main():
     job_id = int(sys.argv[1])
-    JobController(job_id).run()
+    try:
+        JobController(job_id).run()
+    finally:
+        # Free our controller slot: start the next PENDING job's
+        # controller (reference: jobs/scheduler.py transitions).
+        try:
+            from skypilot_amd.jobs import scheduler
+            scheduler.maybe_start_controllers()
+        except Exception:  # noqa: BLE001
+            pass
 
 
 if __name__ == "__main__":

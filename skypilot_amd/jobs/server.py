@@ -26,20 +26,13 @@ def launch(task: Dict[str, Any], name: Optional[str] = None
     else:
         Task.from_yaml_config(dict(task))  # validate before persisting
     job_id = state.create(name or task.get("name"), task)
-    pkg_root = os.path.dirname(os.path.dirname(
-        os.path.dirname(os.path.abspath(__file__))))
-    env = dict(os.environ)
-    env["PYTHONPATH"] = pkg_root + (
-        ":" + env["PYTHONPATH"] if env.get("PYTHONPATH") else "")
-    log = open(state.global_state.root_dir() / f"jobs-controller-{job_id}.log",
-               "ab")
-    proc = subprocess.Popen(
-        [sys.executable, "-m", "skypilot_amd.jobs.controller", str(job_id)],
-        stdout=log, stderr=subprocess.STDOUT, start_new_session=True,
-        env=env)
-    log.close()
-    state.update(job_id, controller_pid=proc.pid)
-    return {"job_id": job_id, "controller_pid": proc.pid}
+    # The controller scheduler bounds concurrent controller processes
+    # (reference: sky/jobs/scheduler.py:232); beyond the cap, jobs wait
+    # as PENDING and start when a running controller finishes.
+    from skypilot_amd.jobs import scheduler
+    scheduler.maybe_start_controllers()
+    pid = (state.get(job_id) or {}).get("controller_pid")
+    return {"job_id": job_id, "controller_pid": pid}
 
 
 def queue() -> List[Dict[str, Any]]:

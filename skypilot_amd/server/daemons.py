@@ -76,6 +76,11 @@ def _loop():
         try:
             from skypilot_amd.jobs import state as jobs_state
             jobs_state.reconcile()
+            # drain PENDING controllers freed by finished/crashed ones
+            from skypilot_amd.jobs import scheduler as jobs_scheduler
+            jobs_scheduler.maybe_start_controllers()
+            from skypilot_amd.jobs import pools as jobs_pools
+            jobs_pools.autoscale()
         except Exception:  # noqa: BLE001
             pass
         try:

@@ -197,9 +197,12 @@ def jobs_group_down(name: str) -> Dict[str, Any]:
 
 @register("jobs_pool_apply", LONG)
 def jobs_pool_apply(name: str, template: Dict[str, Any],
-                    num_workers: int = 2):
+                    num_workers: int = 2,
+                    min_workers: Optional[int] = None,
+                    max_workers: Optional[int] = None):
     from skypilot_amd.jobs import pools
-    return pools.apply(name, template, num_workers)
+    return pools.apply(name, template, num_workers,
+                       min_workers=min_workers, max_workers=max_workers)
 
 
 @register("jobs_pool_status", SHORT)

@@ -245,3 +245,35 @@ def test_job_group_concurrent_colocated(client, tmp_path):
     assert "SKYPILOT_JOBGROUP_HOST=127.0.0.1" in ps_env
     assert "SKYPILOT_JOBGROUP_TASKS=ps,worker" in ps_env
     sdk.get(sdk.jobs_group_down("g1"))
+
+
+def test_controller_concurrency_capped(sky_env, client, monkeypatch):
+    """With SKY_AMD_MAX_CONTROLLERS=2, launching 6 managed jobs keeps at
+    most 2 controller processes alive at any time and all 6 complete
+    (reference: sky/jobs/scheduler.py:232 maybe_start_controllers)."""
+    import time
+    from skypilot_amd.client import sdk
+    from skypilot_amd.jobs import scheduler, state
+    monkeypatch.setenv("SKY_AMD_MAX_CONTROLLERS", "2")
+    ids = []
+    for i in range(6):
+        r = sdk.get(sdk.jobs_launch({"run": "sleep 0.4",
+                                     "resources": {"cpus": 1}},
+                                    name=f"cap-{i}"), timeout=60)
+        ids.append(r["job_id"])
+    max_live = 0
+    deadline = time.time() + 180
+    while time.time() < deadline:
+        jobs = {j["job_id"]: j for j in state.list_jobs()}
+        live = sum(1 for j in jobs.values()
+                   if j["status"] not in state.TERMINAL
+                   and scheduler._pid_alive(j.get("controller_pid")))
+        max_live = max(max_live, live)
+        done = sum(1 for i in ids if jobs[i]["status"] in state.TERMINAL)
+        if done == 6:
+            break
+        scheduler.maybe_start_controllers()  # belt-and-braces drain
+        time.sleep(0.3)
+    jobs = {j["job_id"]: j for j in state.list_jobs()}
+    assert all(jobs[i]["status"] == "SUCCEEDED" for i in ids), jobs
+    assert max_live <= 2, max_live
