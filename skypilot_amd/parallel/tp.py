@@ -24,13 +24,25 @@ from skypilot_amd import ops
 from skypilot_amd.models import llama as L
 
 
+# Sentinel: a tp=1 build must not touch collectives even when the
+# process happens to be in a >1-world process group (e.g. a reference
+# model built alongside a TP engine).
+_NO_COMM = "no_comm"
+
+
+def _group_world(group) -> int:
+    if group == _NO_COMM or not dist.is_initialized():
+        return 1
+    return dist.get_world_size(group)
+
+
 class _AllReduceFwd(torch.autograd.Function):
     """All-reduce activations forward; identity backward (used after
     row-parallel GEMMs)."""
 
     @staticmethod
     def forward(ctx, x, group):
-        if dist.is_initialized() and dist.get_world_size(group) > 1:
+        if _group_world(group) > 1:
             x = x.contiguous()
             dist.all_reduce(x, group=group)
         return x
@@ -51,7 +63,7 @@ class _AllReduceBwd(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, g):
-        if dist.is_initialized() and dist.get_world_size(ctx.group) > 1:
+        if _group_world(ctx.group) > 1:
             g = g.contiguous()
             dist.all_reduce(g, group=ctx.group)
         return g, None
@@ -87,9 +99,16 @@ class TPAttention(nn.Module):
         if infer_ctx is None:
             o = ops.attention(q, k, v, self.scale, causal=True)
         elif infer_ctx.mode == "prefill":
-            infer_ctx.cache.write_prefill(self.layer_idx,
-                                          infer_ctx.prefill_slot, k, v,
-                                          infer_ctx.prefill_len)
+            # batched prefill (same contract as models/llama.py:117)
+            slots = (infer_ctx.prefill_slots
+                     if infer_ctx.prefill_slots is not None
+                     else [infer_ctx.prefill_slot])
+            lens = (infer_ctx.prefill_lens
+                    if infer_ctx.prefill_lens is not None
+                    else [infer_ctx.prefill_len])
+            for i, (slot, ln) in enumerate(zip(slots, lens)):
+                infer_ctx.cache.write_prefill(self.layer_idx, slot,
+                                              k[i:i + 1], v[i:i + 1], ln)
             o = ops.attention(q, k, v, self.scale, causal=True)
         else:
             infer_ctx.cache.write_decode(self.layer_idx, infer_ctx.slots,
@@ -112,7 +131,7 @@ class TPMLP(nn.Module):
         self.w_gate_up = nn.Linear(h, 2 * self.m, bias=False)
         self.w_down = nn.Linear(self.m, h, bias=False)
 
-    def forward(self, x):
+    def forward(self, x, infer_ctx=None):
         x = _AllReduceBwd.apply(x, self.group)
         gu = self.w_gate_up(x)
         g, u = gu.split(self.m, dim=-1)
@@ -143,6 +162,8 @@ class TPLlama(L.Llama):
 
     def __init__(self, cfg: L.LlamaConfig, tp: int, group=None):
         nn.Module.__init__(self)
+        if tp == 1:
+            group = _NO_COMM
         self.cfg = cfg
         self.tp = tp
         self.group = group
@@ -233,4 +254,11 @@ def build_tp_model(name: str, tp: int, rank: int, device="cpu",
             blk.mlp_norm.fill_(1.0)
         model.final_norm.fill_(1.0)
         model.lm_head.weight.copy_(draw(model.lm_head.weight.shape))
-    return model.to(dtype)
+    model = model.to(dtype)
+    # Per-shard view of the config for inference plumbing (the KV cache
+    # holds this rank's kv heads only).
+    import dataclasses
+    model.cfg_shard = dataclasses.replace(
+        cfg, num_heads=cfg.num_heads // tp,
+        num_kv_heads=cfg.num_kv_heads // tp)
+    return model

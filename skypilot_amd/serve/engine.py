@@ -18,6 +18,7 @@ from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
 import torch
+import torch.distributed as dist
 
 from skypilot_amd.models.llama import build_model
 from skypilot_amd.serve.kv_cache import KVCache
@@ -60,14 +61,30 @@ class Engine:
     def __init__(self, model_name: str, device: Optional[str] = None,
                  max_seq: int = 4096, max_batch: Optional[int] = None,
                  hbm_budget_gb: Optional[float] = None,
-                 use_graphs: bool = True):
+                 use_graphs: bool = True, model=None, tp_group=None,
+                 tp_rank: int = 0, tp_world: int = 1):
         self.device = torch.device(device or (
             "cuda" if torch.cuda.is_available() else "cpu"))
         dtype = torch.bfloat16
-        self.model = build_model(model_name, device=str(self.device),
-                                 dtype=dtype)
+        # Tensor-parallel serving (BASELINE config 5's serve twin): a
+        # pre-built TP shard is passed in; rank 0 leads (scheduler +
+        # HTTP), ranks > 0 mirror its forwards via broadcast step plans
+        # (see follower_loop).  All ranks hold identical activations
+        # after each block's all-reduce, so token selection needs no
+        # extra communication.
+        self.tp_group = tp_group
+        self.tp_rank = tp_rank
+        # TP participation is EXPLICIT (tp_world > 1): a process may be
+        # part of a distributed group for other reasons (e.g. a non-TP
+        # engine built alongside) and must not broadcast step plans.
+        self.tp_world = tp_world
+        if model is not None:
+            self.model = model
+        else:
+            self.model = build_model(model_name, device=str(self.device),
+                                     dtype=dtype)
         self.model.eval()
-        self.cfg = self.model.cfg
+        self.cfg = getattr(self.model, "cfg_shard", self.model.cfg)
         self.max_seq = min(max_seq, self.cfg.max_seq_len)
         if max_batch is None:
             if hbm_budget_gb is None:
@@ -136,6 +153,8 @@ class Engine:
         if self._thread is not None:
             self._thread.join(timeout=10)
             self._thread = None
+        if self.tp_world > 1 and self.tp_rank == 0:
+            self._tp_bcast({"op": "stop"})
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -146,6 +165,8 @@ class Engine:
         n = len(reqs)
         slots = [self.free_slots.pop() for _ in range(n)]
         lens = [len(r.prompt_ids) for r in reqs]
+        self._tp_bcast({"op": "prefill", "slots": slots, "lens": lens,
+                        "prompts": [r.prompt_ids for r in reqs]})
         pad = (max(lens) + 63) // 64 * 64
         toks = torch.zeros(n, pad, dtype=torch.long, device=self.device)
         for i, r in enumerate(reqs):
@@ -248,9 +269,23 @@ class Engine:
         tok_l = [r.out_ids[-1] for r in reqs]
         if self.use_graphs:
             b = self._bucket(n)
+            all_greedy = all(r.temperature == 0 for r in reqs)
+            chunk = 1
+            if all_greedy and self.pending.empty():
+                rem = min(r.max_tokens - len(r.out_ids) for r in reqs)
+                cap = min(self.max_seq - 2 - l for l in lens)
+                chunk = max(1, min(self.CHUNK, rem, cap))
+            # TP: the plan must go out BEFORE capture — _get_graph's
+            # warmup forwards contain all-reduces that need every rank
+            # participating (the follower captures on receipt).
+            self._tp_bcast({"op": "decode", "graph": True, "bucket": b,
+                            "slots": slots, "lens": lens, "toks": tok_l,
+                            "chunk": chunk})
             try:
                 g, st = self._get_graph(b)
             except Exception as e:  # capture unsupported -> eager forever
+                if self.tp_world > 1:
+                    raise  # followers already committed to the graph plan
                 import sys
                 print(f"[engine] hipGraph capture failed, falling back "
                       f"to eager decode: {e!r}", file=sys.stderr)
@@ -267,12 +302,6 @@ class Engine:
             h[3] = h[1]
             h[4] = h[1] + 1
             h[5] = h[2]
-            all_greedy = all(r.temperature == 0 for r in reqs)
-            chunk = 1
-            if all_greedy and self.pending.empty():
-                rem = min(r.max_tokens - len(r.out_ids) for r in reqs)
-                cap = min(self.max_seq - 2 - l for l in lens)
-                chunk = max(1, min(self.CHUNK, rem, cap))
             st["ctr"].zero_()
             st["stage"].copy_(st["host"], non_blocking=True)
             for _ in range(chunk):
@@ -294,6 +323,9 @@ class Engine:
                 return
             logits = st["out"]  # [b, 1, V]
         else:
+            self._tp_bcast({"op": "decode", "graph": False,
+                            "slots": slots, "lens": lens, "toks": tok_l,
+                            "chunk": 1})
             toks = torch.tensor([[t] for t in tok_l], dtype=torch.long,
                                 device=self.device)
             slots_t = torch.tensor(slots, dtype=torch.long,
@@ -353,6 +385,79 @@ class Engine:
             if req.stream_queue is not None:
                 req.stream_queue.put(None)  # end-of-stream marker
             req.done.set()
+
+    # ---------------------- tensor-parallel plumbing -------------------
+    def _tp_bcast(self, obj) -> None:
+        if self.tp_world > 1 and self.tp_rank == 0:
+            dist.broadcast_object_list([obj], src=0, group=self.tp_group)
+
+    def follower_loop(self) -> None:
+        """TP rank > 0: execute the leader's broadcast step plan.  The
+        follower keeps no scheduler state — every forward's slots/
+        lengths/tokens arrive in the plan, and its KV cache stays
+        consistent because it runs the identical deterministic
+        forwards."""
+        assert self.tp_world > 1 and self.tp_rank > 0
+        while True:
+            lst = [None]
+            dist.broadcast_object_list(lst, src=0, group=self.tp_group)
+            msg = lst[0]
+            if msg["op"] == "stop":
+                return
+            if msg["op"] == "prefill":
+                self._follow_prefill(msg)
+            else:
+                self._follow_decode(msg)
+
+    @torch.no_grad()
+    def _follow_prefill(self, msg) -> None:
+        slots, lens = msg["slots"], msg["lens"]
+        n = len(slots)
+        pad = (max(lens) + 63) // 64 * 64
+        toks = torch.zeros(n, pad, dtype=torch.long, device=self.device)
+        for i, ids in enumerate(msg["prompts"]):
+            toks[i, :lens[i]] = torch.tensor(ids, device=self.device)
+        positions = torch.arange(pad, dtype=torch.int32,
+                                 device=self.device).repeat(n)
+        ctx = InferenceContext(cache=self.cache, mode="prefill",
+                               prefill_slots=slots, prefill_lens=lens)
+        self.model(toks, positions, ctx)
+
+    @torch.no_grad()
+    def _follow_decode(self, msg) -> None:
+        slots, lens, tok_l = msg["slots"], msg["lens"], msg["toks"]
+        n = len(slots)
+        if msg["graph"]:
+            g, st = self._get_graph(msg["bucket"])
+            scr = self._scratch_slot
+            h = st["host"].numpy()
+            h[0, :n] = tok_l
+            h[0, n:] = 0
+            h[1, :n] = lens
+            h[1, n:] = 0
+            h[2, :n] = slots
+            h[2, n:] = scr
+            h[3] = h[1]
+            h[4] = h[1] + 1
+            h[5] = h[2]
+            st["ctr"].zero_()
+            st["stage"].copy_(st["host"], non_blocking=True)
+            for _ in range(msg["chunk"]):
+                g.replay()
+            torch.cuda.synchronize()
+            return
+        toks = torch.tensor([[t] for t in tok_l], dtype=torch.long,
+                            device=self.device)
+        slots_t = torch.tensor(slots, dtype=torch.long, device=self.device)
+        pos_t = torch.tensor(lens, dtype=torch.long, device=self.device)
+        positions = torch.tensor(lens, dtype=torch.int32,
+                                 device=self.device)
+        kv_lens = torch.tensor([l + 1 for l in lens], dtype=torch.int32,
+                               device=self.device)
+        ctx = InferenceContext(cache=self.cache, mode="decode",
+                               slots=slots_t, pos=pos_t, kv_lens=kv_lens,
+                               slot_ids_i32=slots_t.int())
+        self.model(toks, positions, ctx)
 
     def _loop(self):
         while not self._stop.is_set():

@@ -167,7 +167,43 @@ def main():
     ap.add_argument("--max-seq", type=int, default=4096)
     ap.add_argument("--max-batch", type=int, default=None)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree; run under torchrun "
+                         "with one rank per GPU (70B serving: --tp 8)")
     args = ap.parse_args()
+    if args.tp > 1:
+        # TP serving (BASELINE config 5's serve twin): every rank builds
+        # its shard; rank 0 leads (scheduler + HTTP), others follow the
+        # broadcast step plan.  Launched by the gang driver as
+        # torchrun --nproc-per-node TP ... --tp TP.
+        import torch
+        import torch.distributed as dist
+        from skypilot_amd.parallel.tp import build_tp_model
+        rank = int(os.environ.get("RANK", "0"))
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if not dist.is_initialized():
+            dist.init_process_group(backend)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        device = args.device or (
+            f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}"
+            if torch.cuda.is_available() else "cpu")
+        shard = build_tp_model(args.model, tp=args.tp, rank=rank,
+                               device=device)
+        engine = Engine(args.model, device=device, max_seq=args.max_seq,
+                        max_batch=args.max_batch, model=shard,
+                        tp_rank=rank, tp_world=args.tp)
+        if rank > 0:
+            engine.follower_loop()
+            return
+        engine.start()
+        app = create_app(engine, args.model)
+        print(f"serving {args.model} tp={args.tp} on "
+              f"{args.host}:{args.port} (max_batch={engine.max_batch})",
+              flush=True)
+        uvicorn.run(app, host=args.host, port=args.port,
+                    log_level="warning")
+        return
     engine = Engine(args.model, device=args.device, max_seq=args.max_seq,
                     max_batch=args.max_batch)
     engine.start()

@@ -72,3 +72,70 @@ def test_tp2_matches_single_process():
         p.join(timeout=60)
     for rank, res in results:
         assert res == "ok", f"rank {rank}: {res}"
+
+
+def _tp_serve_worker(rank, world, port, q):
+    import os
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from skypilot_amd.parallel.tp import build_tp_model
+        from skypilot_amd.serve.engine import Engine
+        shard = build_tp_model("llama-smoke", tp=world, rank=rank,
+                               device="cpu", seed=3)
+        eng = Engine("llama-smoke", device="cpu", max_seq=256,
+                     max_batch=4, model=shard, tp_rank=rank,
+                     tp_world=world)
+        if rank > 0:
+            eng.follower_loop()  # exits on the leader's stop broadcast
+            q.put((rank, "ok"))
+            return
+        eng.start()
+        out1 = eng.generate([5, 9, 200, 3], max_tokens=6)
+        out2 = eng.generate([7, 7, 1, 42, 77, 11], max_tokens=6)
+        eng.stop()
+        # reference: tp=1 build draws the identical full weights
+        ref_model = build_tp_model("llama-smoke", tp=1, rank=0,
+                                   device="cpu", seed=3)
+        ref = Engine("llama-smoke", device="cpu", max_seq=256,
+                     max_batch=4, model=ref_model)
+        ref.start()
+        r1 = ref.generate([5, 9, 200, 3], max_tokens=6)
+        r2 = ref.generate([7, 7, 1, 42, 77, 11], max_tokens=6)
+        ref.stop()
+        m1 = sum(a == b for a, b in zip(out1, r1))
+        m2 = sum(a == b for a, b in zip(out2, r2))
+        # bf16 all-reduce ordering may flip an occasional argmax
+        assert m1 >= 5 and m2 >= 5, (out1, r1, out2, r2)
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+        q.put((rank, f"{e}\n{traceback.format_exc()}"))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_tp_serving_leader_follower():
+    """TP=2 engine over gloo: rank 0 schedules + samples, rank 1
+    mirrors the broadcast step plan; generations match the tp=1
+    reference (VERDICT r01 #5: TP serving)."""
+    import socket
+    import torch.multiprocessing as mp
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_serve_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, res in results:
+        assert res == "ok", f"rank {rank}: {res}"
