@@ -175,7 +175,8 @@ def run_sky_launched(args) -> int:
     backend = PoolBackend()
     t_launch = time.time()
     job_id, handle = execution.launch(task, cluster, detach_run=True)
-    agent = AgentClient(handle["agent_port"])
+    agent = AgentClient(handle["agent_port"],
+                        token=handle.get("agent_token"))
 
     # Job-start latency: launch() call -> agent marks the job RUNNING.
     started_at = None
@@ -241,6 +242,17 @@ def main() -> int:
         print(f"[bench] sky-launched path failed ({e}); falling back to "
               "direct trainer (config.sky_launched=false)",
               file=sys.stderr)
+        try:
+            # Release the GPU before the direct run: the failed sky
+            # attempt may have left the cluster (and its job) alive.
+            from skypilot_amd import global_state
+            from skypilot_amd.backends.pool_backend import PoolBackend
+            rec = global_state.get_cluster("bench-cluster")
+            if rec:
+                PoolBackend().teardown(rec["handle"], terminate=True)
+                time.sleep(3.0)
+        except Exception:  # noqa: BLE001
+            pass
         return run_direct(args, extra_config={"sky_launched": False})
 
 
