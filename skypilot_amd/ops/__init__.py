@@ -323,6 +323,32 @@ def attn_decode(q, kc, vc, kv_lens, slot_ids, scale):
     return attn_decode_ref(q, kc, vc, kv_lens, slot_ids, scale)
 
 
+# ---- fp8 decode weights (serving) -------------------------
+# Per-output-row e4m3fn quantization: W_row ~= scale_r * fp8(W_row/scale_r).
+# Registered weights reroute decode_linear through the fp8 GEMV (half the
+# streamed bytes on the weight-bandwidth-bound decode step).
+_FP8_WEIGHTS: dict = {}
+
+
+def quantize_fp8_rowwise(w: torch.Tensor):
+    """bf16 [O, I] -> (uint8 e4m3fn [O, I], fp32 scale [O])."""
+    wf = w.float()
+    scale = wf.abs().amax(dim=1).clamp(min=1e-8) / 448.0
+    q = (wf / scale[:, None]).clamp(-448.0, 448.0)
+    q8 = q.to(torch.float8_e4m3fn).view(torch.uint8)
+    return q8.contiguous(), scale.contiguous()
+
+
+def register_fp8_weight(w: torch.Tensor) -> None:
+    """Quantize and register a weight for fp8 decode GEMV routing."""
+    q8, scale = quantize_fp8_rowwise(w)
+    _FP8_WEIGHTS[w.data_ptr()] = (q8, scale)
+
+
+def clear_fp8_weights() -> None:
+    _FP8_WEIGHTS.clear()
+
+
 def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """F.linear for the decode step: y = x @ W^T with x [.., n, in],
     n <= 8.  On GPU this runs the weight-bandwidth-bound skinny GEMV
@@ -342,6 +368,14 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     # projections (O >= 100k) up to n<=4.
     use_native = (n == 1 or wbytes <= 34_000_000
                   or (o >= 100_000 and n <= 4))
+    if (x.is_cuda and x.dtype == torch.bfloat16 and 1 <= n <= 8
+            and i % 1024 == 0 and _FP8_WEIGHTS):
+        ent = _FP8_WEIGHTS.get(weight.data_ptr())
+        if ent is not None:
+            C = _require_native("skinny_gemm_fp8")
+            y = C.skinny_gemm_fp8(x.reshape(n, i).contiguous(), ent[0],
+                                  ent[1])
+            return y.view(*lead, o)
     if (use_native and x.is_cuda and x.dtype == torch.bfloat16
             and 1 <= n <= 8 and i % 512 == 0):
         C = _require_native("skinny_gemm")

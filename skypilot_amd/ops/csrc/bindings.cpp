@@ -45,6 +45,8 @@ void skinny_gemm_launch(const void*, const void*, void*, int, int, int,
 void skinny_gemm_swiglu_launch(const void*, const void*, void*, int, int,
                                int, hipStream_t);
 void transpose_bf16_launch(const void*, void*, int, int, hipStream_t);
+void skinny_gemm_fp8_launch(const void*, const float*, const void*, void*,
+                            int, int, int, hipStream_t);
 void gemm_nt_launch(const void*, const void*, void*, int, int, long long,
                     hipStream_t);
 void adamw_mt_launch(const void*, const void*, const void*, const void*,
@@ -266,6 +268,28 @@ torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x) {
   return gemm_nt(dyT, xT);
 }
 
+torch::Tensor skinny_gemm_fp8(torch::Tensor X, torch::Tensor W8,
+                              torch::Tensor scale) {
+  // Y[N,O] = X[N,I] @ (scale[o] * W8[O,I])^T ; W8 = OCP e4m3fn bytes.
+  CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
+  CHECK_GPU(W8); CHECK_CONTIG(W8);
+  TORCH_CHECK(W8.scalar_type() == at::kByte ||
+              W8.scalar_type() == at::kFloat8_e4m3fn,
+              "W8 must be uint8/float8_e4m3fn");
+  CHECK_GPU(scale); CHECK_CONTIG(scale);
+  TORCH_CHECK(scale.scalar_type() == at::kFloat, "scale must be fp32");
+  int N = X.size(0), I = X.size(1), O = W8.size(0);
+  TORCH_CHECK(W8.size(1) == I, "skinny_gemm_fp8: inner dims mismatch");
+  TORCH_CHECK(scale.numel() == O, "skinny_gemm_fp8: scale size");
+  TORCH_CHECK(N >= 1 && N <= 8, "skinny_gemm_fp8: N must be 1..8");
+  TORCH_CHECK(I % 16 == 0, "skinny_gemm_fp8: I must be 16-aligned");
+  auto y = torch::empty({N, O}, X.options());
+  skinny_gemm_fp8_launch(W8.data_ptr(), scale.data_ptr<float>(),
+                         X.data_ptr(), y.data_ptr(), N, I, O,
+                         cur_stream());
+  return y;
+}
+
 torch::Tensor skinny_gemm(torch::Tensor X, torch::Tensor W) {
   // Y[N,O] = X[N,I] @ W[O,I]^T (decode GEMV; see skinny_gemm.hip)
   CHECK_GPU(X); CHECK_CONTIG(X); CHECK_BF16(X);
@@ -384,6 +408,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_decode", &attn_decode);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("skinny_gemm_fp8", &skinny_gemm_fp8);
   m.def("transpose_bf16", &transpose_bf16);
   m.def("gemm_nt", &gemm_nt);
   m.def("wgrad_tn", &wgrad_tn);

@@ -236,3 +236,170 @@ extern "C" void skinny_gemm_launch(const void* W, const void* X, void* Y,
   }
 #undef CASE
 }
+
+// ---------------------------------------------------------------------------
+// fp8 (OCP e4m3fn) weight GEMV: Y[N,O] = X[N,I] @ (scale[o] * W8[O,I])^T.
+// Decode is weight-BANDWIDTH-bound (header comment): storing W as fp8
+// with one fp32 scale per output row halves the streamed bytes, so the
+// single-stream decode rate doubles at the same 6.5-7.3 TB/s roofline.
+// Hardware fp8->f32 conversion (v_cvt_pk_f32_fp8, 2 elems/op) keeps the
+// kernel bandwidth-bound; X stays bf16 (exact), accumulation fp32.
+// ---------------------------------------------------------------------------
+union fp8x16 {
+  i32x4 i;
+  unsigned int u[4];
+};
+
+__device__ __forceinline__ void dot_fp8_16(unsigned int w4x4[4],
+                                           const unsigned short* xb,
+                                           float* acc) {
+  float xf[16];
+#pragma unroll
+  for (int j = 0; j < 16; ++j) xf[j] = bf2f(xb[j]);
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w4x4[q], false);
+    f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w4x4[q], true);
+    *acc += lo[0] * xf[q * 4 + 0] + lo[1] * xf[q * 4 + 1] +
+            hi[0] * xf[q * 4 + 2] + hi[1] * xf[q * 4 + 3];
+  }
+}
+
+template <int N>
+__global__ __launch_bounds__(256) void skinny_gemm_fp8_kernel(
+    const unsigned char* __restrict__ W8,
+    const float* __restrict__ scale,
+    const unsigned short* __restrict__ X,
+    unsigned short* __restrict__ Y, int I, int O) {
+  int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  int o = blockIdx.x * 4 + wave;
+  if (o >= O) return;
+  const unsigned char* wrow = W8 + (long long)o * I;
+  // 4 independent 16B loads in flight (same pipeline shape as the bf16
+  // kernel: a single loop-carried chain stalls on vmcnt).
+  float a0[N], a1[N], a2[N], a3[N];
+#pragma unroll
+  for (int b = 0; b < N; ++b) a0[b] = a1[b] = a2[b] = a3[b] = 0.f;
+  int i = lane * 16;
+  for (; i + 3 * 1024 + 16 <= I; i += 4 * 1024) {
+    fp8x16 w0, w1, w2, w3;
+    w0.i = __builtin_nontemporal_load((const i32x4*)(wrow + i));
+    w1.i = __builtin_nontemporal_load((const i32x4*)(wrow + i + 1024));
+    w2.i = __builtin_nontemporal_load((const i32x4*)(wrow + i + 2048));
+    w3.i = __builtin_nontemporal_load((const i32x4*)(wrow + i + 3072));
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      const unsigned short* xb = X + (long long)b * I + i;
+      dot_fp8_16(w0.u, xb, &a0[b]);
+      dot_fp8_16(w1.u, xb + 1024, &a1[b]);
+      dot_fp8_16(w2.u, xb + 2048, &a2[b]);
+      dot_fp8_16(w3.u, xb + 3072, &a3[b]);
+    }
+  }
+  for (; i < I; i += 1024) {
+    fp8x16 wv;
+    wv.i = __builtin_nontemporal_load((const i32x4*)(wrow + i));
+#pragma unroll
+    for (int b = 0; b < N; ++b)
+      dot_fp8_16(wv.u, X + (long long)b * I + i, &a0[b]);
+  }
+#pragma unroll
+  for (int b = 0; b < N; ++b) {
+    float r = wave_reduce_sum(a0[b] + a1[b] + a2[b] + a3[b]);
+    if (lane == 0) Y[(long long)b * O + o] = f2bf(r * scale[o]);
+  }
+}
+
+// Multi-row fp8 variant: R rows per wave share one X conversion and
+// keep 2R 16B loads in flight (the R=1 kernel is latency-bound at fp8
+// row sizes — half the bytes of bf16 per row).
+template <int N, int R>
+__global__ __launch_bounds__(256) void skinny_gemm_fp8_mr_kernel(
+    const unsigned char* __restrict__ W8,
+    const float* __restrict__ scale,
+    const unsigned short* __restrict__ X,
+    unsigned short* __restrict__ Y, int I, int O) {
+  int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  int o0 = (blockIdx.x * 4 + wave) * R;
+  if (o0 >= O) return;
+  const unsigned char* wrow = W8 + (long long)o0 * I;
+  float acc[R][N];
+#pragma unroll
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int b = 0; b < N; ++b) acc[r][b] = 0.f;
+  int i = lane * 16;
+  for (; i + 1024 + 16 <= I; i += 2048) {
+    fp8x16 w[R][2];
+#pragma unroll
+    for (int r = 0; r < R; ++r) {
+      w[r][0] = *(fp8x16*)&(((const i32x4*)(wrow + (long long)r * I + i))[0]);
+      w[r][1] = *(fp8x16*)&(((const i32x4*)(wrow + (long long)r * I + i + 1024))[0]);
+    }
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      const unsigned short* xb = X + (long long)b * I + i;
+#pragma unroll
+      for (int r = 0; r < R; ++r) {
+        dot_fp8_16(w[r][0].u, xb, &acc[r][b]);
+        dot_fp8_16(w[r][1].u, xb + 1024, &acc[r][b]);
+      }
+    }
+  }
+  for (; i < I; i += 1024) {
+    fp8x16 w[R];
+#pragma unroll
+    for (int r = 0; r < R; ++r)
+      w[r].i = __builtin_nontemporal_load(
+          (const i32x4*)(wrow + (long long)r * I + i));
+#pragma unroll
+    for (int b = 0; b < N; ++b)
+#pragma unroll
+      for (int r = 0; r < R; ++r)
+        dot_fp8_16(w[r].u, X + (long long)b * I + i, &acc[r][b]);
+  }
+#pragma unroll
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      float v = wave_reduce_sum(acc[r][b]);
+      if (lane == 0 && o0 + r < O)
+        Y[(long long)b * O + o0 + r] = f2bf(v * scale[o0 + r]);
+    }
+}
+
+extern "C" void skinny_gemm_fp8_launch(const void* W8, const float* scale,
+                                       const void* X, void* Y, int N,
+                                       int I, int O, hipStream_t stream) {
+  // SKY_FP8_MR: 0 = single-row kernel, else rows-per-wave (default 2).
+  static const int mr = [] {
+    const char* e = getenv("SKY_FP8_MR");
+    return e ? atoi(e) : 2;
+  }();
+  if (mr >= 2 && O % 2 == 0) {
+    dim3 gridm((O / 2 + 3) / 4), block(256);
+#define F8MR(n)                                                          case n:                                                                  hipLaunchKernelGGL((skinny_gemm_fp8_mr_kernel<n, 2>), gridm,                              block, 0, stream, (const unsigned char*)W8,                            scale, (const unsigned short*)X,                                       (unsigned short*)Y, I, O);                          break;
+    switch (N) {
+      F8MR(1) F8MR(2) F8MR(3) F8MR(4) F8MR(5) F8MR(6) F8MR(7) F8MR(8)
+      default:
+        break;
+    }
+#undef F8MR
+    return;
+  }
+  dim3 grid((O + 3) / 4), block(256);
+#define F8CASE(n)                                                      \
+  case n:                                                              \
+    hipLaunchKernelGGL(skinny_gemm_fp8_kernel<n>, grid, block, 0,      \
+                       stream, (const unsigned char*)W8, scale,        \
+                       (const unsigned short*)X, (unsigned short*)Y,   \
+                       I, O);                                          \
+    break;
+  switch (N) {
+    F8CASE(1) F8CASE(2) F8CASE(3) F8CASE(4)
+    F8CASE(5) F8CASE(6) F8CASE(7) F8CASE(8)
+    default:
+      break;
+  }
+#undef F8CASE
+}

@@ -167,6 +167,9 @@ def main():
     ap.add_argument("--max-seq", type=int, default=4096)
     ap.add_argument("--max-batch", type=int, default=None)
     ap.add_argument("--device", default=None)
+    ap.add_argument("--fp8-decode", action="store_true",
+                    help="quantize weights to rowwise e4m3fn for the "
+                         "decode GEMVs (~2x single-stream decode)")
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree; run under torchrun "
                          "with one rank per GPU (70B serving: --tp 8)")
@@ -193,6 +196,8 @@ def main():
         engine = Engine(args.model, device=device, max_seq=args.max_seq,
                         max_batch=args.max_batch, model=shard,
                         tp_rank=rank, tp_world=args.tp)
+        if args.fp8_decode:
+            engine.enable_fp8_decode()
         if rank > 0:
             engine.follower_loop()
             return
@@ -206,6 +211,9 @@ def main():
         return
     engine = Engine(args.model, device=args.device, max_seq=args.max_seq,
                     max_batch=args.max_batch)
+    if args.fp8_decode:
+        n8 = engine.enable_fp8_decode()
+        print(f"fp8 decode weights: {n8} tensors quantized", flush=True)
     engine.start()
     app = create_app(engine, args.model)
     print(f"serving {args.model} on {args.host}:{args.port} "

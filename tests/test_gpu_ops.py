@@ -374,3 +374,47 @@ def test_wgrad_tn_matches_ref():
         # transpose alone
         t = C.transpose_bf16(dy)
         assert torch.equal(t, dy.t().contiguous())
+
+
+@pytest.mark.gpu
+def test_skinny_gemm_fp8_matches_ref():
+    """fp8 (e4m3fn rowwise) decode GEMV vs fp32 reference: the error
+    must be bounded by the quantization error itself (VERDICT r01 #7
+    fp8 numerics harness)."""
+    C = ops.native()
+    torch.manual_seed(11)
+    for n, i, o in [(1, 4096, 4096), (2, 14336, 4096), (4, 1024, 2048),
+                    (8, 2048, 1024)]:
+        x = (torch.randn(n, i, device="cuda") * 0.5).bfloat16()
+        w = (torch.randn(o, i, device="cuda") * 0.02).bfloat16()
+        q8, scale = ops.quantize_fp8_rowwise(w)
+        y = C.skinny_gemm_fp8(x, q8, scale)
+        # reference through the SAME quantized weights (isolates kernel
+        # error from quantization error)...
+        wq = q8.view(torch.float8_e4m3fn).float() * scale[:, None]
+        ref_q = x.float() @ wq.t()
+        torch.testing.assert_close(y.float(), ref_q, atol=0.02, rtol=0.02)
+        # ...and against the unquantized weights with fp8-scale bounds.
+        ref = x.float() @ w.float().t()
+        err = (y.float() - ref).abs().max()
+        bound = 0.04 * ref.abs().max() + 0.5
+        assert err < bound, (n, i, o, float(err), float(bound))
+
+
+@pytest.mark.gpu
+def test_fp8_decode_linear_routing():
+    """Registered weights reroute ops.decode_linear through the fp8
+    GEMV; unregistered weights keep the bf16 path."""
+    torch.manual_seed(12)
+    x = (torch.randn(1, 4096, device="cuda") * 0.5).bfloat16()
+    w = (torch.randn(2048, 4096, device="cuda") * 0.02).bfloat16()
+    y_bf16 = ops.decode_linear(x, w)
+    ops.register_fp8_weight(w)
+    try:
+        y_fp8 = ops.decode_linear(x, w)
+        # outputs differ slightly (quantization) but agree loosely
+        assert not torch.equal(y_bf16, y_fp8)
+        torch.testing.assert_close(y_fp8.float(), y_bf16.float(),
+                                   atol=0.25, rtol=0.1)
+    finally:
+        ops.clear_fp8_weights()
