@@ -10,7 +10,14 @@ rclone later); the three modes map to:
   MOUNT        — symlink the mount path to the store dir (write-through,
                  survives cluster teardown: this is the managed-jobs
                  checkpoint contract, reference SURVEY.md §2.7)
-  MOUNT_CACHED — same as MOUNT on a local FS (writeback cache is a no-op)
+  MOUNT_CACHED — local store: same as MOUNT (writeback is a no-op on a
+                 local FS).  REMOTE source (s3:// etc): a real rclone
+                 FUSE mount with VFS write-back
+                 (`rclone mount --vfs-cache-mode writes`), matching the
+                 reference's MOUNT_CACHED semantics
+                 (sky/data/mounting_utils.py:698 + storage.py:378 VFS
+                 presets) — writes land locally and flush to the bucket
+                 asynchronously; on recovery the same bucket re-mounts.
 """
 from __future__ import annotations
 
@@ -120,6 +127,51 @@ def delete_storage(name: str) -> bool:
     return existed
 
 
+def _mount_cached_remote(source: str, target: Path,
+                         handle: Dict[str, Any]) -> None:
+    """rclone FUSE mount with VFS write-back for a remote bucket
+    (reference: sky/data/mounting_utils.py:698 rclone mount command,
+    sky/data/storage.py:378 MOUNT_CACHED VFS presets)."""
+    import shutil as _sh
+    import subprocess
+    if _sh.which("rclone") is None:
+        raise TaskValidationError(
+            "MOUNT_CACHED with a remote source requires rclone on PATH")
+    target.mkdir(parents=True, exist_ok=True)
+    if os.path.ismount(target):
+        return
+    cache_dir = Path(handle["cluster_dir"]) / ".rclone-vfs-cache"
+    cache_dir.mkdir(parents=True, exist_ok=True)
+    proc = subprocess.run(
+        ["rclone", "mount", _rclone_target(source), str(target),
+         "--daemon", "--allow-non-empty",
+         "--vfs-cache-mode", "writes",
+         "--cache-dir", str(cache_dir),
+         "--dir-cache-time", "10s",
+         "--vfs-write-back", "1s"],
+        capture_output=True, text=True, timeout=60)
+    if proc.returncode != 0:
+        raise TaskValidationError(
+            f"rclone mount failed: {proc.stderr[:400]}")
+    # record for best-effort unmount at teardown
+    mounts = Path(handle["cluster_dir"]) / "rclone-mounts"
+    with open(mounts, "a") as f:
+        f.write(str(target) + "\n")
+
+
+def unmount_cluster_mounts(cluster_dir: str) -> None:
+    """Best-effort fusermount -u of rclone mounts at teardown."""
+    import subprocess
+    mounts = Path(cluster_dir) / "rclone-mounts"
+    if not mounts.exists():
+        return
+    for line in mounts.read_text().splitlines():
+        if line.strip():
+            subprocess.run(["fusermount", "-u", line.strip()],
+                           capture_output=True, timeout=30)
+    mounts.unlink(missing_ok=True)
+
+
 def _resolve_dst(handle: Dict[str, Any], dst: str) -> Path:
     """Absolute mount paths are real paths (single-node pool shares the
     FS); relative paths land in the cluster workdir."""
@@ -158,6 +210,16 @@ def execute_file_mounts(handle: Dict[str, Any],
                 raise TaskValidationError(
                     f"storage mount for {dst} needs a name")
             mode = str(src.get("mode", MODE_MOUNT)).upper()
+            if mode == MODE_MOUNT_CACHED and _is_remote(src.get("source")):
+                # register the store row (no local pull: the FUSE mount
+                # IS the data path) and mount the bucket directly.
+                get_or_create_store(name, None)
+                with global_state._DB_LOCK, global_state._conn() as c:
+                    c.execute("UPDATE storage SET source=?, store_type=? "
+                              "WHERE name=?",
+                              (src["source"], "s3-mount", name))
+                _mount_cached_remote(src["source"], target, handle)
+                continue
             store = get_or_create_store(name, src.get("source"))
             if mode == MODE_COPY:
                 if target.is_symlink():

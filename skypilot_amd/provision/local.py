@@ -266,6 +266,11 @@ def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
     import shutil
     cdir = handle.get("cluster_dir")
     if cdir and Path(cdir).exists():
+        try:
+            from skypilot_amd.data.storage import unmount_cluster_mounts
+            unmount_cluster_mounts(cdir)
+        except Exception:  # noqa: BLE001
+            pass
         shutil.rmtree(cdir, ignore_errors=True)
 
 

@@ -277,3 +277,51 @@ def test_controller_concurrency_capped(sky_env, client, monkeypatch):
     jobs = {j["job_id"]: j for j in state.list_jobs()}
     assert all(jobs[i]["status"] == "SUCCEEDED" for i in ids), jobs
     assert max_live <= 2, max_live
+
+
+def test_mount_cached_remote_checkpoint_recovery(sky_env, client,
+                                                 monkeypatch, tmp_path):
+    """MOUNT_CACHED with a remote bucket is a real rclone FUSE mount
+    with VFS write-back (reference: mounting_utils.py:698, VERDICT r01
+    #8).  A faked rclone simulates mount (bind via symlink) + sync:
+    write a checkpoint through the mount, tear the cluster down,
+    re-mount on a fresh cluster, and the checkpoint is there."""
+    import stat
+    # fake rclone: "mount remote:path target --daemon ..." symlinks the
+    # backing dir; other verbs no-op.  (Real boxes use real rclone.)
+    backing = tmp_path / "bucket"
+    backing.mkdir()
+    fake_bin = tmp_path / "bin"
+    fake_bin.mkdir()
+    rc = fake_bin / "rclone"
+    rc.write_text(f"""#!/bin/bash
+if [ "$1" = mount ]; then
+  tgt="$3"
+  rmdir "$tgt" 2>/dev/null || true
+  ln -sfn {backing} "$tgt"
+  exit 0
+fi
+exit 0
+""")
+    rc.chmod(rc.stat().st_mode | stat.S_IEXEC)
+    monkeypatch.setenv("PATH", f"{fake_bin}:{os.environ['PATH']}")
+    # fusermount absent -> unmount is best-effort; symlink removal via
+    # rmtree is fine for the fake.
+    from skypilot_amd.client import sdk
+    mnt = str(tmp_path / "ckpt-mnt")
+    mount = {mnt: {"name": "ckpt-bkt", "mode": "MOUNT_CACHED",
+                   "source": "s3://bkt/ck"}}
+    sdk.get(sdk.launch({"run": f"echo step-500 > {mnt}/model.ck",
+                        "file_mounts": mount,
+                        "resources": {"cpus": 1}}, "mc-1"), timeout=60)
+    from tests.test_orchestrator import _wait_job_done
+    assert _wait_job_done("mc-1", 1)["status"] == "SUCCEEDED"
+    assert (backing / "model.ck").read_text().strip() == "step-500"
+    sdk.get(sdk.down("mc-1"))
+    # recovery: a fresh cluster re-mounts the same bucket and resumes
+    sdk.get(sdk.launch({"run": f"cat {mnt}/model.ck",
+                        "file_mounts": mount,
+                        "resources": {"cpus": 1}}, "mc-2"), timeout=60)
+    j = _wait_job_done("mc-2", 1)
+    assert j["status"] == "SUCCEEDED"
+    sdk.get(sdk.down("mc-2"))
