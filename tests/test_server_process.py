@@ -99,3 +99,104 @@ def test_real_server_launch_roundtrip(real_server):
                        json={"cluster_name": name}, timeout=10)
         st = _wait_req(base, r.json()["request_id"])
         assert st["status"] == "SUCCEEDED"
+
+
+def test_load_50_concurrent_mixed_with_rss(real_server, tmp_path):
+    """Reference load-test method (tests/load_tests/README.md:30-52):
+    N concurrent mixed requests against a REAL server process with peak
+    process-tree RSS and latency capture — the numbers size the
+    executor worker constants (server/executor.py worker sizing)."""
+    import concurrent.futures
+    import statistics
+    import psutil
+
+    base = real_server
+    # find the server process (owner of the port) via the pid recorded
+    # in the home dir
+    import re
+    home = None
+    for p in psutil.process_iter(["cmdline"]):
+        cl = " ".join(p.info["cmdline"] or [])
+        if "skypilot_amd.server.app" in cl and base.split(":")[-1] in cl:
+            home = p
+            break
+    assert home is not None
+
+    def tree_rss() -> int:
+        total = 0
+        try:
+            procs = [home] + home.children(recursive=True)
+            for p in procs:
+                try:
+                    total += p.memory_info().rss
+                except psutil.NoSuchProcess:
+                    pass
+        except psutil.NoSuchProcess:
+            pass
+        return total
+
+    lat = {"launch": [], "status": [], "queue": [], "requests": []}
+    peak = [0]
+    stop = [False]
+
+    def sampler():
+        while not stop[0]:
+            peak[0] = max(peak[0], tree_rss())
+            time.sleep(0.2)
+
+    import threading
+    th = threading.Thread(target=sampler, daemon=True)
+    th.start()
+
+    def timed(kind, fn):
+        t0 = time.time()
+        r = fn()
+        lat[kind].append(time.time() - t0)
+        return r
+
+    def do_launch(i):
+        rid = timed("launch", lambda: httpx.post(
+            base + "/api/v1/launch",
+            json={"task": {"run": "sleep 0.2",
+                           "resources": {"cpus": 1}},
+                  "cluster_name": f"ld-{i}"},
+            timeout=60).json()["request_id"])
+        return _wait_req(base, rid, timeout=240)
+
+    def do_status(_):
+        rid = timed("status", lambda: httpx.post(
+            base + "/api/v1/status", json={},
+            timeout=60).json()["request_id"])
+        return _wait_req(base, rid, timeout=120)
+
+    def do_queue(i):
+        return timed("requests", lambda: httpx.get(
+            base + "/api/requests", timeout=60).status_code)
+
+    t0 = time.time()
+    with concurrent.futures.ThreadPoolExecutor(max_workers=50) as ex:
+        futs = [ex.submit(do_launch, i) for i in range(10)]
+        futs += [ex.submit(do_status, i) for i in range(25)]
+        futs += [ex.submit(do_queue, i) for i in range(15)]
+        results = [f.result(timeout=300) for f in futs]
+    wall = time.time() - t0
+    stop[0] = True
+    th.join(timeout=5)
+    assert len(results) == 50
+    # all launches completed
+    launches = results[:10]
+    assert all(r["status"] == "SUCCEEDED" for r in launches), launches
+    p50 = {k: statistics.median(v) for k, v in lat.items() if v}
+    print(f"[load] 50 concurrent mixed requests: wall={wall:.1f}s "
+          f"peak_tree_rss={peak[0]/1e9:.2f}GB "
+          f"submit_p50={ {k: round(v, 3) for k, v in p50.items()} }")
+    # Sanity bounds that catch executor regressions: submission must
+    # stay sub-second even under 50-way concurrency, and the process
+    # tree must stay far below the reference's 11.78 GB figure.
+    assert p50["status"] < 5.0, p50
+    assert wall < 240, wall
+    assert peak[0] < 8e9, peak[0]
+    # teardown
+    for i in range(10):
+        httpx.post(base + "/api/v1/down",
+                   json={"cluster_name": f"ld-{i}"}, timeout=60)
