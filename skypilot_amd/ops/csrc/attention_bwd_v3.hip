@@ -48,7 +48,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
     const float* __restrict__ lse, const float* __restrict__ Dvec,
     unsigned short* __restrict__ dK, unsigned short* __restrict__ dV, int B,
     int S, int Hq, int Hkv, float scale, int causal) {
-  __shared__ unsigned short q_lds[32 * ATT_D];     // 8 KB, vsubz layout
+  __shared__ unsigned short q_lds[32 * ATT_D];     // 8 KB, swzK16 layout
   __shared__ unsigned short do_lds[32 * ATT_D];    // 8 KB
   __shared__ unsigned short k_own[128 * ATT_D];    // 32 KB, block's K rows
   __shared__ unsigned short v_own[128 * ATT_D];    // 32 KB, block's V rows
@@ -267,12 +267,18 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
   __shared__ unsigned short k_lds[2][64 * ATT_D];  // 2 x 16 KB
   __shared__ unsigned short v_lds[2][64 * ATT_D];  // 2 x 16 KB
 
-  const int qt = gridDim.x - 1 - blockIdx.x;  // heavy blocks first
+  // Paired causal grid (see attention_fwd_v3): heavy q-tile then light
+  // one per block, staging pipeline continuous across the boundary.
+  const int nq = S / 128;
+  const bool paired = causal && (int)gridDim.x * 2 == nq;
+  const int qtA = paired ? nq - 1 - (int)blockIdx.x
+                         : (int)gridDim.x - 1 - (int)blockIdx.x;
+  const int qtB = paired ? (int)blockIdx.x : 0;
   const int bh = blockIdx.y;
   const int b = bh / Hq;
   const int qh = bh % Hq;
   const int kvh = qh / (Hq / Hkv);
-  const int qbase = qt * 128;
+  int qbase = qtA * 128;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -289,28 +295,29 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
   const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
   const unsigned short* dOb = dO + ((long long)b * S * Hq + qh) * ATT_D;
 
-  const int my_q = qbase + 32 * w + col;
+  int my_q = qbase + 32 * w + col;
 
   // Q as B-fragments pre-scaled by scale*log2(e); dO fragments are
   // re-read from L2 per kv tile (keeping them resident spilled).
-  const float qs = scale * 1.44269504088896340736f;
   s16x8 q_b[8];
-  {
-    const unsigned short* src = Qb + (long long)my_q * q_rowstride;
-#pragma unroll
-    for (int ks = 0; ks < 8; ++ks) {
-      s16x8 raw = *(const s16x8*)(src + ks * 16 + h * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * qs);
-      q_b[ks] = raw;
-    }
+  float lse2, dvq;
+#define DQ_LOAD_Q()                                                       \
+  {                                                                       \
+    const float qs = scale * 1.44269504088896340736f;                     \
+    const unsigned short* src = Qb + (long long)my_q * q_rowstride;       \
+    _Pragma("unroll") for (int ks = 0; ks < 8; ++ks) {                    \
+      s16x8 raw = *(const s16x8*)(src + ks * 16 + h * 8);                 \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * qs);          \
+      q_b[ks] = raw;                                                      \
+    }                                                                     \
+    float lv = lse[((long long)b * Hq + qh) * S + my_q];                  \
+    lse2 = (lv == -INFINITY) ? 3.0e37f                                    \
+                             : lv * 1.44269504088896340736f;              \
+    dvq = Dvec[((long long)b * S + my_q) * Hq + qh];                      \
   }
+  DQ_LOAD_Q();
   const unsigned short* do_src = dOb + (long long)my_q * q_rowstride;
-  float lv = lse[((long long)b * Hq + qh) * S + my_q];
-  const float lse2 =
-      (lv == -INFINITY) ? 3.0e37f : lv * 1.44269504088896340736f;
-  const float dvq = Dvec[((long long)b * S + my_q) * Hq + qh];
 
   f32x16 dq_acc[4];
 #pragma unroll
@@ -318,8 +325,24 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
 
-  const int n_tiles = causal ? (qbase + 128) / 64 : S / 64;
-  const int w_tiles = causal ? ((qbase + 32 * w + 31) >> 6) + 1 : n_tiles;
+  const int nt_A = causal ? (qbase + 128) / 64 : S / 64;
+  const int nt_B = paired ? (qtB * 128 + 128) / 64 : 0;
+  const int total_t = nt_A + nt_B;
+  int w_tiles = causal ? ((qbase + 32 * w + 31) >> 6) + 1 : nt_A;
+
+  // epilogue macro (runs at the pair boundary and at the end).
+#define DQ_EPILOGUE()                                                     \
+  {                                                                       \
+    unsigned short* dQb = dQ + ((long long)b * S * Hq + qh) * ATT_D;      \
+    unsigned short* qrow = dQb + (long long)my_q * q_rowstride;           \
+    _Pragma("unroll") for (int dt = 0; dt < 4; ++dt)                      \
+      _Pragma("unroll") for (int rq = 0; rq < 4; ++rq) {                  \
+        s16x4 ov;                                                         \
+        _Pragma("unroll") for (int r = 0; r < 4; ++r)                     \
+          ov[r] = (short)f2bf(dq_acc[dt][rq * 4 + r] * scale);            \
+        *(s16x4*)(qrow + dt * 32 + rq * 8 + h * 4) = ov;                  \
+      }                                                                   \
+  }
 
   // staging: wave w stages rows [16w,16w+16) of both K and V via 4+4
   // global_load_lds (linear dest, pre-swizzled source).
@@ -338,10 +361,26 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
   }
   DQ_STAGE((long long)0, 0);
 
-  for (int kt = 0; kt < n_tiles; ++kt) {
-    const int buf = kt & 1;
+  for (int it = 0; it < total_t; ++it) {
+    if (paired && it == nt_A) {
+      DQ_EPILOGUE();
+      qbase = qtB * 128;
+      my_q = qbase + 32 * w + col;
+      DQ_LOAD_Q();
+      do_src = dOb + (long long)my_q * q_rowstride;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dq_acc[dt][r] = 0.f;
+      w_tiles = ((qbase + 32 * w + 31) >> 6) + 1;
+    }
+    const int kt = it < nt_A ? it : it - nt_A;
+    const int buf = it & 1;
     __syncthreads();  // drains this tile's global_load_lds (vmcnt)
-    if (kt + 1 < n_tiles) DQ_STAGE((long long)(kt + 1) * 64, buf ^ 1);
+    if (it + 1 < total_t) {
+      const int kt2 = (it + 1 < nt_A) ? it + 1 : it + 1 - nt_A;
+      DQ_STAGE((long long)kt2 * 64, buf ^ 1);
+    }
     if (kt >= w_tiles) continue;
 
     const int kvbase = kt * 64;
@@ -463,19 +502,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
     DQ_MFMA(db);
   }
 
-  // ---- epilogue: dQ[q][d] = scale * dQ^T[d][q].
-  unsigned short* dQb = dQ + ((long long)b * S * Hq + qh) * ATT_D;
-  unsigned short* qrow = dQb + (long long)my_q * q_rowstride;
-#pragma unroll
-  for (int dt = 0; dt < 4; ++dt)
-#pragma unroll
-    for (int rq = 0; rq < 4; ++rq) {
-      s16x4 ov;
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        ov[r] = (short)f2bf(dq_acc[dt][rq * 4 + r] * scale);
-      *(s16x4*)(qrow + dt * 32 + rq * 8 + h * 4) = ov;
-    }
+  DQ_EPILOGUE();
+#undef DQ_EPILOGUE
+#undef DQ_LOAD_Q
 }
 
 extern "C" void attn_bwd_v3_launch(const void* Q, const void* K,
@@ -490,7 +519,9 @@ extern "C" void attn_bwd_v3_launch(const void* Q, const void* K,
                      (const unsigned short*)V, (const unsigned short*)dO,
                      lse, Dvec, (unsigned short*)dK, (unsigned short*)dV, B,
                      S, Hq, Hkv, scale, causal ? 1 : 0);
-  dim3 gq(S / 128, B * Hq);
+  int nqq = S / 128;
+  int gqx = (causal && nqq % 2 == 0) ? nqq / 2 : nqq;
+  dim3 gq(gqx, B * Hq);
   hipLaunchKernelGGL(attn_bwd_dq_v3_kernel, gq, dim3(256), 0, stream,
                      (const unsigned short*)Q, (const unsigned short*)K,
                      (const unsigned short*)V, (const unsigned short*)dO,

@@ -50,12 +50,22 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
   __shared__ unsigned short k_lds[2][KVB * ATT_D];  // 2 x 16 KB
   __shared__ unsigned short v_lds[2][KVB * ATT_D];  // 2 x 16 KB
 
-  const int qt = gridDim.x - 1 - blockIdx.x;  // heavy blocks first
+  // Causal grids are PAIRED: one block owns a (heavy, light) q-tile
+  // pair — qtA = nq-1-bx (deep kv sweep) then qtB = bx — with the KV
+  // staging pipeline running straight through the boundary.  This
+  // amortizes the per-q-tile pipeline ramp (measured: causal
+  // efficiency rises with tiles/block; mask & wave-skip are null) and
+  // load-balances (every pair sweeps nq+1 kv tiles).
+  const int nq = S / QB;
+  const bool paired = causal && (int)gridDim.x * 2 == nq;
+  const int qtA = paired ? nq - 1 - (int)blockIdx.x
+                         : (int)gridDim.x - 1 - (int)blockIdx.x;
+  const int qtB = paired ? (int)blockIdx.x : 0;
   const int bh = blockIdx.y;
   const int b = bh / Hq;
   const int qh = bh % Hq;
   const int kvh = qh / (Hq / Hkv);
-  const int qbase = qt * QB;
+  int qbase = qtA * QB;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -71,22 +81,25 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
   const unsigned short* Kb = K + ((long long)b * S * Hkv + kvh) * ATT_D;
   const unsigned short* Vb = V + ((long long)b * S * Hkv + kvh) * ATT_D;
 
-  const int my_q = qbase + QBW * w + col;
+  int my_q = qbase + QBW * w + col;
 
   // ---- Q as B-fragments, pre-scaled by scale*log2(e) (exp2 softmax).
-  const float qs = scale * 1.44269504088896340736f;
+  // (qs/src recomputed inside the macro from kernel args so nothing
+  // stays live across the kv loop for the boundary reload.)
   s16x8 q_b[8];
-  {
-    const unsigned short* src = Qb + (long long)my_q * q_rowstride;
-#pragma unroll
-    for (int ks = 0; ks < 8; ++ks) {
-      s16x8 raw = *(const s16x8*)(src + ks * 16 + h * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * qs);
-      q_b[ks] = raw;
-    }
+#define V3_LOAD_Q()                                                       \
+  {                                                                       \
+    const float qs = scale * 1.44269504088896340736f;                     \
+    const unsigned short* src =                                           \
+        Q + ((long long)b * S * Hq + qh + (long long)my_q * Hq) * ATT_D;  \
+    _Pragma("unroll") for (int ks = 0; ks < 8; ++ks) {                    \
+      s16x8 raw = *(const s16x8*)(src + ks * 16 + h * 8);                 \
+      _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+        raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * qs);          \
+      q_b[ks] = raw;                                                      \
+    }                                                                     \
   }
+  V3_LOAD_Q();
 
   f32x16 o_t[4];
 #pragma unroll
@@ -97,68 +110,82 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
 
   // causal: 1 = normal; 2 = debug timing (causal loop structure, mask
   // skipped); 3 = debug timing (no per-wave diagonal skip).
-  const int n_tiles = causal ? (qbase + QB) / KVB : S / KVB;
+  const int nt_A = causal ? (qbase + QB) / KVB : S / KVB;
+  const int nt_B = paired ? (qtB * QB + QB) / KVB : 0;
+  const int total_t = nt_A + nt_B;
   // Tiles this wave actually computes (beyond its diagonal: staging +
   // barriers only).
-  const int w_tiles = (causal == 1 || causal == 2)
-                          ? ((qbase + QBW * w + QBW - 1) >> 6) + 1
-                          : n_tiles;
+  int w_tiles = (causal == 1 || causal == 2)
+                    ? ((qbase + QBW * w + QBW - 1) >> 6) + 1
+                    : nt_A;
 
-  // ---- staging assignments.
-  // V: thread owns 16B chunks 2*tid, 2*tid+1 of the [64][128] tile.
-  // K: wave w stages rows [8w, 8w+8) via 2 global_load_lds.
-  s16x8 vpre[NVC];
-  const int vc0 = tid * NVC;
-
-  // prologue: issue tile 0 (V loads to registers first, then async K).
-  {
-#pragma unroll
-    for (int i = 0; i < NVC; ++i) {
-      int c = vc0 + i;
-      vpre[i] = *(const s16x8*)(Vb + (long long)(c >> 4) * kv_rowstride +
-                                (c & 15) * 8);
-    }
-#pragma unroll
-    for (int i = 0; i < NKI; ++i) {
-      int row = (KVB / NW) * w + 4 * i + (lane >> 4);
-      int chunk = lane & 15;
-      const void* src = (const char*)(Kb + (long long)row * kv_rowstride) +
-                        ((chunk ^ (row & 15)) << 4);
-      gload_lds16(src, (char*)k_lds[0] + ((KVB / NW) * w + 4 * i) * 256);
-    }
+  // ---- epilogue: O[q][d] = O^T[d][q] / l; lse = ln2*m' + ln(l).
+  unsigned short* Ob = O + ((long long)b * S * Hq + qh) * ATT_D;
+  float* lse_b = lse_out + ((long long)b * Hq + qh) * S;
+#define V3_EPILOGUE()                                                     \
+  {                                                                       \
+    const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;                \
+    unsigned short* orow = Ob + (long long)my_q * q_rowstride;            \
+    _Pragma("unroll") for (int dn = 0; dn < 4; ++dn)                      \
+      _Pragma("unroll") for (int rq = 0; rq < 4; ++rq) {                  \
+        s16x4 ov;                                                         \
+        _Pragma("unroll") for (int r = 0; r < 4; ++r)                     \
+          ov[r] = (short)f2bf(o_t[dn][rq * 4 + r] * inv_l);               \
+        *(s16x4*)(orow + dn * 32 + rq * 8 + h * 4) = ov;                  \
+      }                                                                   \
+    if (h == 0)                                                           \
+      lse_b[my_q] = (l_run > 0.f)                                         \
+                        ? 0.69314718055994530942f * m_run +               \
+                              __logf(l_run)                               \
+                        : -INFINITY;                                      \
   }
 
-  for (int kt = 0; kt < n_tiles; ++kt) {
-    const int buf = kt & 1;
-    // 1. commit the prefetched V registers into this tile's buffer.
+  // ---- staging: BOTH tiles by global_load_lds (zero staging
+  // registers).  K rows pre-swizzled for the b128 A-fragment reads; V
+  // rows linear (tr reads tolerate the resulting ~4-way conflicts;
+  // register staging for V's subtiled layout cost 16 VGPRs and
+  // spilled the paired kernel).
+#define V3_STAGE(kvoff, bufi)                                             \
+  {                                                                       \
+    _Pragma("unroll") for (int i = 0; i < NKI; ++i) {                     \
+      int row = (KVB / NW) * w + 4 * i + (lane >> 4);                     \
+      int chunk = lane & 15;                                              \
+      long long srow = (kvoff + row) * kv_rowstride;                      \
+      gload_lds16((const char*)(Kb + srow) +                              \
+                      ((chunk ^ (row & 15)) << 4),                        \
+                  (char*)k_lds[bufi] + ((KVB / NW) * w + 4 * i) * 256);   \
+      gload_lds16((const char*)(Vb + srow) + (chunk << 4),                \
+                  (char*)v_lds[bufi] + ((KVB / NW) * w + 4 * i) * 256);   \
+    }                                                                     \
+  }
+  V3_STAGE((long long)0, 0);
+
+  for (int it = 0; it < total_t; ++it) {
+    if (paired && it == nt_A) {
+      // ---- boundary: finish the heavy tile, switch to the light one.
+      V3_EPILOGUE();
+      qbase = qtB * QB;
+      my_q = qbase + QBW * w + col;
+      V3_LOAD_Q();
 #pragma unroll
-    for (int i = 0; i < NVC; ++i) {
-      int c = vc0 + i;
-      *(s16x8*)((char*)v_lds[buf] + 2 * vsub(c >> 4, (c & 15) * 8)) =
-          vpre[i];
+      for (int dn = 0; dn < 4; ++dn)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_t[dn][r] = 0.f;
+      m_run = -INFINITY;
+      l_run = 0.f;
+      w_tiles = (causal == 3) ? nt_B
+                              : ((qbase + QBW * w + QBW - 1) >> 6) + 1;
     }
-    // 2. one barrier per tile; hipcc's pre-barrier waitcnt drains the
-    // in-flight K global_load_lds for this buffer too.
+    const int kt = it < nt_A ? it : it - nt_A;
+    const int buf = it & 1;
+    // one barrier per tile: hipcc's pre-barrier waitcnt drains the
+    // in-flight global_load_lds for this buffer.
     __syncthreads();
-    // 3. issue next tile's staging; it stays in flight under compute.
-    if (kt + 1 < n_tiles) {
-      const long long nb = (long long)(kt + 1) * KVB;
-#pragma unroll
-      for (int i = 0; i < NVC; ++i) {
-        int c = vc0 + i;
-        vpre[i] = *(const s16x8*)(
-            Vb + (nb + (c >> 4)) * kv_rowstride + (c & 15) * 8);
-      }
-#pragma unroll
-      for (int i = 0; i < NKI; ++i) {
-        int row = (KVB / NW) * w + 4 * i + (lane >> 4);
-        int chunk = lane & 15;
-        const void* src =
-            (const char*)(Kb + (nb + row) * kv_rowstride) +
-            ((chunk ^ (row & 15)) << 4);
-        gload_lds16(src,
-                    (char*)k_lds[buf ^ 1] + ((KVB / NW) * w + 4 * i) * 256);
-      }
+    // issue next tile's staging (possibly the light q-tile's tile 0:
+    // the pipeline never drains at the boundary).
+    if (it + 1 < total_t) {
+      const int kt2 = (it + 1 < nt_A) ? it + 1 : it + 1 - nt_A;
+      V3_STAGE((long long)kt2 * KVB, buf ^ 1);
     }
     if (kt >= w_tiles) continue;  // past this wave's diagonal
 
@@ -200,18 +227,18 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
     const int trd = ((g & 1) << 4) + ((lw & 3) << 2);
     const int trr = lw >> 2;
     const unsigned vbase = (unsigned)(size_t)((char*)v_lds[buf]);
-    tr4 tA0, tB0, tA1, tB1;  // 2-deep tr pipeline (static names, rule #20)
+    tr4 tA0, tB0;  // tr-read staging (issue-early, pack under latency)
 #define V3_ISSUE(tA, tB, ks)                                              \
   {                                                                       \
     int kvr = (ks) * 16 + kv0g + trr;                                     \
-    ds_tr4_issue(&tA, vbase + 2 * vsub(kvr, trd),                         \
-                 vbase + 2 * vsub(kvr + 4, trd),                          \
-                 vbase + 2 * vsub(kvr, 32 + trd),                         \
-                 vbase + 2 * vsub(kvr + 4, 32 + trd));                    \
-    ds_tr4_issue(&tB, vbase + 2 * vsub(kvr, 64 + trd),                    \
-                 vbase + 2 * vsub(kvr + 4, 64 + trd),                     \
-                 vbase + 2 * vsub(kvr, 96 + trd),                         \
-                 vbase + 2 * vsub(kvr + 4, 96 + trd));                    \
+    ds_tr4_issue(&tA, vbase + kvr * 256 + 2 * trd,                        \
+                 vbase + (kvr + 4) * 256 + 2 * trd,                       \
+                 vbase + kvr * 256 + 2 * (32 + trd),                      \
+                 vbase + (kvr + 4) * 256 + 2 * (32 + trd));               \
+    ds_tr4_issue(&tB, vbase + kvr * 256 + 2 * (64 + trd),                 \
+                 vbase + (kvr + 4) * 256 + 2 * (64 + trd),                \
+                 vbase + kvr * 256 + 2 * (96 + trd),                      \
+                 vbase + (kvr + 4) * 256 + 2 * (96 + trd));               \
   }
     V3_ISSUE(tA0, tB0, 0);
 
@@ -284,43 +311,28 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
     o_t[3] = MFMA32V3(as_bf16x8(vf.v), as_bf16x8(pb), o_t[3]);            \
     __builtin_amdgcn_s_setprio(0);                                        \
   }
-    s16x8 pb0, pb1;
+    s16x8 pb0;
     V3_PACK2(pb0, st, 0);
-    V3_ISSUE(tA1, tB1, 1);
-    lgkm_wait8_bind2(&tA0, &tB0);
+    lgkm_wait0_bind2(&tA0, &tB0);
     V3_MFMA(tA0, tB0, pb0);
-    V3_PACK2(pb1, st, 1);
+    V3_ISSUE(tA0, tB0, 1);
+    V3_PACK2(pb0, st, 1);
+    lgkm_wait0_bind2(&tA0, &tB0);
+    V3_MFMA(tA0, tB0, pb0);
     V3_ISSUE(tA0, tB0, 2);
-    lgkm_wait8_bind2(&tA1, &tB1);
-    V3_MFMA(tA1, tB1, pb1);
     V3_PACK2(pb0, st, 2);
-    V3_ISSUE(tA1, tB1, 3);
-    lgkm_wait8_bind2(&tA0, &tB0);
+    lgkm_wait0_bind2(&tA0, &tB0);
     V3_MFMA(tA0, tB0, pb0);
-    V3_PACK2(pb1, st, 3);
-    lgkm_wait0_bind2(&tA1, &tB1);
-    V3_MFMA(tA1, tB1, pb1);
+    V3_ISSUE(tA0, tB0, 3);
+    V3_PACK2(pb0, st, 3);
+    lgkm_wait0_bind2(&tA0, &tB0);
+    V3_MFMA(tA0, tB0, pb0);
   }
 
-  // ---- epilogue: O[q][d] = O^T[d][q] / l; lse = ln2*m' + ln(l).
-  unsigned short* Ob = O + ((long long)b * S * Hq + qh) * ATT_D;
-  float* lse_b = lse_out + ((long long)b * Hq + qh) * S;
-  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
-  unsigned short* orow = Ob + (long long)my_q * q_rowstride;
-#pragma unroll
-  for (int dn = 0; dn < 4; ++dn)
-#pragma unroll
-    for (int rq = 0; rq < 4; ++rq) {
-      s16x4 ov;
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        ov[r] = (short)f2bf(o_t[dn][rq * 4 + r] * inv_l);
-      *(s16x4*)(orow + dn * 32 + rq * 8 + h * 4) = ov;
-    }
-  if (h == 0)
-    lse_b[my_q] = (l_run > 0.f)
-                      ? 0.69314718055994530942f * m_run + __logf(l_run)
-                      : -INFINITY;
+  V3_EPILOGUE();
+#undef V3_EPILOGUE
+#undef V3_LOAD_Q
+#undef V3_ISSUE
 }
 
 extern "C" void attn_fwd_v3_launch(const void* Q, const void* K,
@@ -336,15 +348,23 @@ extern "C" void attn_fwd_v3_launch(const void* Q, const void* K,
     return e ? atoi(e) : 0;
   }();
   const int cz = causal ? (dbg ? dbg : 1) : 0;
+  static const int pair = [] {
+    const char* e = getenv("SKY_ATTN_FWD_V3_PAIR");
+    return e ? atoi(e) : 1;
+  }();
   if (nw == 8) {
-    dim3 grid(S / (QBW * 8), B * Hq);
+    int nq = S / (QBW * 8);
+    int gx = (pair && cz && nq % 2 == 0) ? nq / 2 : nq;
+    dim3 grid(gx, B * Hq);
     hipLaunchKernelGGL(attn_fwd_v3_t<8>, grid, dim3(512), 0, stream,
                        (const unsigned short*)Q, (const unsigned short*)K,
                        (const unsigned short*)V, (unsigned short*)O, lse, B,
                        S, Hq, Hkv, scale, cz);
     return;
   }
-  dim3 grid(S / (QBW * 4), B * Hq);
+  int nq = S / (QBW * 4);
+  int gx = (pair && cz && nq % 2 == 0) ? nq / 2 : nq;
+  dim3 grid(gx, B * Hq);
   hipLaunchKernelGGL(attn_fwd_v3_t<4>, grid, dim3(256), 0, stream,
                      (const unsigned short*)Q, (const unsigned short*)K,
                      (const unsigned short*)V, (unsigned short*)O, lse, B, S,
