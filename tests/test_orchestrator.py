@@ -336,3 +336,29 @@ def test_agent_requires_token(client):
                   headers={"Authorization": f"Bearer {token}"})
     assert r.status_code == 200
     sdk.get(sdk.down("tok-c"))
+
+
+def test_async_sdk(client):
+    """Async SDK twins (reference: sky/client/sdk_async.py): launch,
+    concurrent status polls, get and stream_and_get awaitables."""
+    import asyncio
+    import io
+    from skypilot_amd.client import sdk_async
+
+    async def flow():
+        rid = await sdk_async.launch({"run": "echo async-ok",
+                                      "resources": {"cpus": 1}}, "as-c")
+        res = await sdk_async.get(rid, timeout=90)
+        assert res["job_id"] == 1
+        # concurrent status requests
+        rids = await asyncio.gather(*[sdk_async.status() for _ in range(4)])
+        outs = await asyncio.gather(*[sdk_async.get(r) for r in rids])
+        assert all(any(c["name"] == "as-c" for c in o) for o in outs)
+        buf = io.StringIO()
+        rid2 = await sdk_async.launch({"run": "echo streamed-line",
+                                       "resources": {"cpus": 1}},
+                                      "as-c")
+        await sdk_async.stream_and_get(rid2, out=buf)
+        await sdk_async.get(await sdk_async.down("as-c"))
+
+    asyncio.run(flow())
