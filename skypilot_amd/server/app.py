@@ -175,14 +175,15 @@ def create_app(start_workers: bool = True) -> FastAPI:
     @app.post(API_PREFIX + "/{name}")
     def submit(name: str, request: Request, body: Dict[str, Any] = None):
         ident = _identity(request)
+        ws = request.headers.get("x-skypilot-workspace")
         try:
             users.authorize(ident["role"], name)
+            users.authorize_workspace(ident["user"], ident["role"], ws)
         except Exception as e:
             raise HTTPException(403, str(e))
         try:
             rid = executor.schedule(
-                name, body or {}, user=ident["user"],
-                workspace=request.headers.get("x-skypilot-workspace"))
+                name, body or {}, user=ident["user"], workspace=ws)
         except KeyError:
             raise HTTPException(404, f"unknown request type {name!r}")
         return {"request_id": rid}

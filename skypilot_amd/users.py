@@ -174,6 +174,30 @@ def authorize(role: str, request_name: str) -> None:
     # enforced by check_cluster_owner at execution time.
 
 
+def authorize_workspace(user: str, role: str, workspace: str) -> None:
+    """Workspace-membership RBAC (reference: sky workspaces with
+    `private: true` + allowed_users).  Config:
+
+        workspaces:
+          team-a:
+            private: true
+            allowed_users: [alice, bob]
+
+    Unlisted workspaces (or non-private ones) are open; admins always
+    pass."""
+    if role == "admin" or not workspace:
+        return
+    from skypilot_amd import config as sky_config
+    spec = sky_config.get_nested(["workspaces", workspace], None)
+    if not isinstance(spec, dict) or not spec.get("private"):
+        return
+    allowed = spec.get("allowed_users") or []
+    if user not in allowed:
+        raise PermissionDeniedError(
+            f"workspace {workspace!r} is private; user {user!r} is not "
+            f"in allowed_users")
+
+
 def check_cluster_owner(cluster_name: str) -> None:
     """Mutating per-cluster ops: owner or admin only."""
     rec = global_state.get_cluster(cluster_name)

@@ -235,3 +235,38 @@ def test_workspaces_scope_clusters(sky_env, client, monkeypatch):
     sdk.get(sdk.down("ws-a"))
     monkeypatch.setenv("SKY_AMD_WORKSPACE", "team-b")
     sdk.get(sdk.down("ws-b"))
+
+
+def test_workspace_rbac(sky_env, client, monkeypatch):
+    """Private workspaces admit only allowed_users (reference: sky
+    workspaces private/allowed_users); open workspaces stay open and
+    admins always pass."""
+    import os
+    from pathlib import Path
+    home = Path(os.environ["SKY_AMD_HOME"])
+    home.mkdir(parents=True, exist_ok=True)
+    (home / "config.yaml").write_text(
+        "workspaces:\n  secret:\n    private: true\n"
+        "    allowed_users: [alice]\n")
+    from skypilot_amd import config as sky_config
+    sky_config.load(refresh=True)
+    try:
+        hdr_a = {"X-Skypilot-User": "alice",
+                 "X-Skypilot-Workspace": "secret"}
+        hdr_b = {"X-Skypilot-User": "bob",
+                 "X-Skypilot-Workspace": "secret"}
+        hdr_open = {"X-Skypilot-User": "bob",
+                    "X-Skypilot-Workspace": "open-ws"}
+        assert client.post("/api/v1/status", json={},
+                           headers=hdr_a).status_code == 200
+        r = client.post("/api/v1/status", json={}, headers=hdr_b)
+        assert r.status_code == 403 and "private" in r.text
+        assert client.post("/api/v1/status", json={},
+                           headers=hdr_open).status_code == 200
+        # admin (server identity) passes everywhere
+        assert client.post("/api/v1/status", json={},
+                           headers={"X-Skypilot-Workspace": "secret"}
+                           ).status_code == 200
+    finally:
+        (home / "config.yaml").unlink(missing_ok=True)
+        sky_config.load(refresh=True)
