@@ -18,12 +18,16 @@ def main():
     ap.add_argument("--gen-len", type=int, default=128)
     ap.add_argument("--batches", default="1,8,32,64")
     ap.add_argument("--max-batch", type=int, default=64)
+    ap.add_argument("--fp8", action="store_true",
+                    help="rowwise e4m3fn decode weights")
     args = ap.parse_args()
 
     from skypilot_amd.serve.engine import Engine
     eng = Engine(args.model,
                  device="cuda:0" if torch.cuda.is_available() else "cpu",
                  max_seq=4096, max_batch=args.max_batch)
+    if args.fp8:
+        print(f"fp8 decode weights: {eng.enable_fp8_decode()} tensors")
     eng.start()
     prompt = list(range(2, 2 + args.prompt_len))
     # warmup

@@ -368,7 +368,11 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     # projections (O >= 100k) up to n<=4.
     use_native = (n == 1 or wbytes <= 34_000_000
                   or (o >= 100_000 and n <= 4))
-    if (x.is_cuda and x.dtype == torch.bfloat16 and 1 <= n <= 8
+    # fp8 pays only while the step is weight-bandwidth-bound: at n >= 3
+    # the bf16 multirow GEMV amortizes the weight stream and fp8's
+    # unpack VALU cost dominates (measured: @8 streams 1521 -> 1069
+    # tok/s with fp8 everywhere).
+    if (x.is_cuda and x.dtype == torch.bfloat16 and 1 <= n <= 2
             and i % 1024 == 0 and _FP8_WEIGHTS):
         ent = _FP8_WEIGHTS.get(weight.data_ptr())
         if ent is not None:
