@@ -65,6 +65,43 @@ def _stall_loop():
         prev = now
 
 
+def _gc_controller_logs(max_age_days: float = 7.0,
+                        max_total_mb: float = 512.0) -> int:
+    """Controller-log retention (reference: sky/jobs/log_gc.py): drop
+    jobs-controller-*.log files for TERMINAL jobs once they are old or
+    the total exceeds the budget (oldest first)."""
+    import os
+    import time as _t
+    from skypilot_amd import global_state
+    from skypilot_amd.jobs import state as jobs_state
+    root = global_state.root_dir()
+    terminal = {j["job_id"] for j in jobs_state.list_jobs()
+                if j["status"] in jobs_state.TERMINAL}
+    logs = []
+    for p in root.glob("jobs-controller-*.log"):
+        try:
+            jid = int(p.stem.rsplit("-", 1)[1])
+        except ValueError:
+            continue
+        if jid in terminal:
+            st = p.stat()
+            logs.append((st.st_mtime, st.st_size, p))
+    logs.sort()
+    removed = 0
+    now = _t.time()
+    total = sum(sz for _, sz, _ in logs)
+    for mtime, size, p in logs:
+        if (now - mtime > max_age_days * 86400
+                or total > max_total_mb * 1e6):
+            try:
+                p.unlink()
+                removed += 1
+                total -= size
+            except OSError:
+                pass
+    return removed
+
+
 def _loop():
     while True:
         time.sleep(REFRESH_INTERVAL)
@@ -81,6 +118,7 @@ def _loop():
             jobs_scheduler.maybe_start_controllers()
             from skypilot_amd.jobs import pools as jobs_pools
             jobs_pools.autoscale()
+            _gc_controller_logs()
         except Exception:  # noqa: BLE001
             pass
         try:

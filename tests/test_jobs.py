@@ -325,3 +325,32 @@ exit 0
     j = _wait_job_done("mc-2", 1)
     assert j["status"] == "SUCCEEDED"
     sdk.get(sdk.down("mc-2"))
+
+
+def test_controller_log_gc(sky_env, client):
+    """Old terminal-job controller logs are garbage-collected
+    (reference: sky/jobs/log_gc.py retention)."""
+    import os
+    from skypilot_amd.client import sdk
+    from skypilot_amd.jobs import state
+    from skypilot_amd.server.daemons import _gc_controller_logs
+    r = sdk.get(sdk.jobs_launch({"run": "true",
+                                 "resources": {"cpus": 1}},
+                                name="gc-job"), timeout=60)
+    jid = r["job_id"]
+    deadline = time.time() + 120
+    while time.time() < deadline:
+        j = state.get(jid)
+        if j and j["status"] in state.TERMINAL:
+            break
+        time.sleep(0.3)
+    log = state.global_state.root_dir() / f"jobs-controller-{jid}.log"
+    assert log.exists()
+    # fresh logs stay
+    assert _gc_controller_logs(max_age_days=7) == 0
+    assert log.exists()
+    # age it and collect
+    old = time.time() - 8 * 86400
+    os.utime(log, (old, old))
+    assert _gc_controller_logs(max_age_days=7) >= 1
+    assert not log.exists()
