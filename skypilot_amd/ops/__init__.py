@@ -375,6 +375,9 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     if (x.is_cuda and x.dtype == torch.bfloat16 and 1 <= n <= 2
             and i % 1024 == 0 and _FP8_WEIGHTS):
         ent = _FP8_WEIGHTS.get(weight.data_ptr())
+        # data_ptr can be recycled after a free; require matching shape
+        if ent is not None and tuple(ent[0].shape) != tuple(weight.shape):
+            ent = None
         if ent is not None:
             C = _require_native("skinny_gemm_fp8")
             y = C.skinny_gemm_fp8(x.reshape(n, i).contiguous(), ent[0],

@@ -491,6 +491,19 @@ class Engine:
         self.model(toks, positions, ctx)
 
     def _loop(self):
+        try:
+            self._loop_inner()
+        except Exception:
+            # A fatal engine error must not leave TP followers blocked
+            # in their broadcast receive forever.
+            if self.tp_world > 1 and self.tp_rank == 0:
+                try:
+                    self._tp_bcast({"op": "stop"})
+                except Exception:  # noqa: BLE001
+                    pass
+            raise
+
+    def _loop_inner(self):
         while not self._stop.is_set():
             did = False
             # Admit pending requests while slots are free — batched

@@ -146,12 +146,16 @@ def create_app(start_workers: bool = True) -> FastAPI:
         _identity(request)
         if not _re.fullmatch(r"[a-f0-9]{32}", upload_id):
             raise HTTPException(400, "bad upload_id")
-        if not (0 <= chunk_index < total_chunks <= 10000):
+        if not (0 <= chunk_index < total_chunks <= 4096):
             raise HTTPException(400, "bad chunk bounds")
+        body = await request.body()
+        max_gb = float(os.environ.get("SKY_AMD_MAX_UPLOAD_GB", "10"))
+        if len(body) > 64 << 20 or total_chunks * len(body) > max_gb * 1e9:
+            raise HTTPException(413, "upload too large")
         updir = global_state.root_dir() / "api" / "uploads"
         updir.mkdir(parents=True, exist_ok=True)
         part = updir / f".{upload_id}.part{chunk_index}"
-        part.write_bytes(await request.body())
+        part.write_bytes(body)
         parts = [updir / f".{upload_id}.part{i}"
                  for i in range(total_chunks)]
         if not all(p.exists() for p in parts):
