@@ -127,10 +127,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
         int qr = c >> 4;
         int dch = (c & 15) * 8;
         long long src = (long long)(qbase + qr) * q_rowstride + dch;
-        *(s16x8*)((char*)q_lds + 2 * vsubz(qr, dch)) =
-            *(const s16x8*)(Qb + src);
-        *(s16x8*)((char*)do_lds + 2 * vsubz(qr, dch)) =
-            *(const s16x8*)(dOb + src);
+        int off = qr * 256 + (((dch >> 3) ^ (qr & 15)) << 4);
+        *(s16x8*)((char*)q_lds + off) = *(const s16x8*)(Qb + src);
+        *(s16x8*)((char*)do_lds + off) = *(const s16x8*)(dOb + src);
       }
       __syncthreads();
       // causal: this wave's kv rows see q tiles >= its diagonal only.
@@ -149,12 +148,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
 #pragma unroll
       for (int ks = 0; ks < 8; ++ks) {
         s16x8 aq = *(const s16x8*)(
-            (char*)q_lds + 2 * vsubz(col, ks * 16 + h * 8));
+            (char*)q_lds + swzK16(col * 256 + (ks * 2 + h) * 16, col));
         s16x8 kfr = *(const s16x8*)(
             (char*)k_own + 2 * vsubz(32 * w + col, ks * 16 + h * 8));
         st = MFMA32V3(as_bf16x8(aq), as_bf16x8(kfr), st);
         s16x8 ad = *(const s16x8*)(
-            (char*)do_lds + 2 * vsubz(col, ks * 16 + h * 8));
+            (char*)do_lds + swzK16(col * 256 + (ks * 2 + h) * 16, col));
         s16x8 vfr = *(const s16x8*)(
             (char*)v_own + 2 * vsubz(32 * w + col, ks * 16 + h * 8));
         dpt = MFMA32V3(as_bf16x8(ad), as_bf16x8(vfr), dpt);
@@ -182,19 +181,22 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
       // B-fragments are tr reads of the dO / Q tiles.
       s16x8 pa, da;
       tr4 t0, t1;
+#define DKV_TRA(base, qv, dv)                                             \
+  ((base) + (qv) * 256 + ((((dv) >> 3) ^ ((qv) & 15)) << 4) +             \
+   (((dv) & 7) << 1))
 #define DKV_TR_ISSUE(lds_base, ks)                                        \
   {                                                                       \
     int q_s = (ks) * 16 + ((g >> 1) << 3) + (lw >> 2);                    \
     int d_s = ((g & 1) << 4) + ((lw & 3) << 2);                           \
     unsigned base = (unsigned)(size_t)((char*)(lds_base));                \
-    ds_tr4_issue(&t0, base + 2 * vsubz(q_s, d_s),                         \
-                 base + 2 * vsubz(q_s + 4, d_s),                          \
-                 base + 2 * vsubz(q_s, 32 + d_s),                         \
-                 base + 2 * vsubz(q_s + 4, 32 + d_s));                    \
-    ds_tr4_issue(&t1, base + 2 * vsubz(q_s, 64 + d_s),                    \
-                 base + 2 * vsubz(q_s + 4, 64 + d_s),                     \
-                 base + 2 * vsubz(q_s, 96 + d_s),                         \
-                 base + 2 * vsubz(q_s + 4, 96 + d_s));                    \
+    ds_tr4_issue(&t0, DKV_TRA(base, q_s, d_s),                            \
+                 DKV_TRA(base, q_s + 4, d_s),                             \
+                 DKV_TRA(base, q_s, 32 + d_s),                            \
+                 DKV_TRA(base, q_s + 4, 32 + d_s));                       \
+    ds_tr4_issue(&t1, DKV_TRA(base, q_s, 64 + d_s),                       \
+                 DKV_TRA(base, q_s + 4, 64 + d_s),                        \
+                 DKV_TRA(base, q_s, 96 + d_s),                            \
+                 DKV_TRA(base, q_s + 4, 96 + d_s));                       \
   }
 #define DKV_MFMA(acc, afrag)                                              \
   {                                                                       \
