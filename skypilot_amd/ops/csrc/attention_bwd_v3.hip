@@ -92,9 +92,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         raw[j] = (short)f2bf(bf2f((unsigned short)raw[j]) * ksc);
-      *(s16x8*)((char*)k_own + 2 * vsubz(kr, dch)) = raw;
-      *(s16x8*)((char*)v_own + 2 * vsubz(kr, dch)) =
-          *(const s16x8*)(Vb + src);
+      int koff = kr * 256 + (((dch >> 3) ^ (kr & 15)) << 4);
+      *(s16x8*)((char*)k_own + koff) = raw;
+      *(s16x8*)((char*)v_own + koff) = *(const s16x8*)(Vb + src);
     }
   }
 
@@ -150,12 +150,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_v3_kernel(
         s16x8 aq = *(const s16x8*)(
             (char*)q_lds + swzK16(col * 256 + (ks * 2 + h) * 16, col));
         s16x8 kfr = *(const s16x8*)(
-            (char*)k_own + 2 * vsubz(32 * w + col, ks * 16 + h * 8));
+            (char*)k_own +
+            swzK16((32 * w + col) * 256 + (ks * 2 + h) * 16, 32 * w + col));
         st = MFMA32V3(as_bf16x8(aq), as_bf16x8(kfr), st);
         s16x8 ad = *(const s16x8*)(
             (char*)do_lds + swzK16(col * 256 + (ks * 2 + h) * 16, col));
         s16x8 vfr = *(const s16x8*)(
-            (char*)v_own + 2 * vsubz(32 * w + col, ks * 16 + h * 8));
+            (char*)v_own +
+            swzK16((32 * w + col) * 256 + (ks * 2 + h) * 16, 32 * w + col));
         dpt = MFMA32V3(as_bf16x8(ad), as_bf16x8(vfr), dpt);
       }
       __builtin_amdgcn_s_setprio(0);
