@@ -77,6 +77,37 @@ def render() -> str:
     gpus = [[g.index, g.name, f"{g.memory_gb} GB", g.numa_node,
              gpus_taken.get(g.index, "-")] for g in detect_gpus()]
 
+    # worker pools + storage + cost (reference dashboard: infra tabs)
+    pool_rows = []
+    try:
+        from skypilot_amd.jobs import pools as jobs_pools
+        for prec in jobs_pools.status():
+            busy = sum(1 for w_ in prec["workers"]
+                       if w_["status"] == "BUSY")
+            pool_rows.append([prec["name"], len(prec["workers"]), busy,
+                              ",".join(w_["cluster_name"]
+                                       for w_ in prec["workers"][:4])])
+    except Exception:  # noqa: BLE001
+        pass
+    storage_rows = []
+    try:
+        from skypilot_amd.data import storage as storage_lib
+        for srec in storage_lib.list_storage():
+            storage_rows.append([srec["name"], srec["store_type"],
+                                 srec["source"] or "-"])
+    except Exception:  # noqa: BLE001
+        pass
+    cost_rows = []
+    try:
+        from skypilot_amd import core as _core
+        for crec in _core.cost_report()[:8]:
+            cost_rows.append([crec["name"],
+                              "live" if crec.get("live") else "done",
+                              f"{crec.get('gpu_hours', 0):.2f}",
+                              f"{crec.get('duration_hours', 0):.2f}"])
+    except Exception:  # noqa: BLE001
+        pass
+
     # recent API requests (reference dashboard: requests table)
     reqs = []
     for r in rdb.list_requests(limit=15):
@@ -125,6 +156,11 @@ th {{ background:#222; color:#e8443a; }}
         jobs)}
 {_table("Services", ["name", "status", "ready", "endpoint"], services)}
 {_table("Pool GPUs", ["idx", "model", "HBM", "numa", "used by"], gpus)}
+{_table("Worker pools", ["name", "workers", "busy", "clusters"],
+        pool_rows)}
+{_table("Storage", ["name", "type", "source"], storage_rows)}
+{_table("Cost report (GPU-hours)", ["cluster", "state", "gpu-hours",
+        "hours"], cost_rows)}
 <h2>Request activity</h2>{spark}
 {_table("Recent API requests",
         ["id", "type", "user", "status", "duration", "at"], reqs)}
