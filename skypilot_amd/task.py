@@ -22,7 +22,8 @@ KNOWN_TASK_KEYS = {
     "name", "workdir", "event_callback", "num_nodes", "resources",
     "file_mounts", "service", "pool", "setup", "run", "envs", "secrets",
     "config", "volumes", "volume_mounts", "inputs", "outputs",
-    "file_mounts_mapping", "_metadata",
+    "file_mounts_mapping", "managed_secrets", "api_server_access",
+    "_metadata",
 }
 
 
