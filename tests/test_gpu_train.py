@@ -105,3 +105,71 @@ def test_tp8_70b_rank0_shard_smoke():
     assert peak_gb < 280, peak_gb
     del model, opt
     torch.cuda.empty_cache()
+
+
+def _tp2_gpu_worker(rank, world, port, q):
+    import os
+    os.environ.update({"MASTER_ADDR": "127.0.0.1",
+                       "MASTER_PORT": str(port),
+                       "RANK": str(rank), "WORLD_SIZE": str(world),
+                       "LOCAL_RANK": "0"})  # both ranks share cuda:0
+    import torch
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from skypilot_amd.parallel.tp import build_tp_model
+        from skypilot_amd.serve.engine import Engine
+        # ---- serve path: leader/follower protocol with real kernels
+        # at the TP-shard head shapes (4 q / 2 kv heads).  Graphs off:
+        # gloo all-reduce is not capturable.
+        shard = build_tp_model("llama-smoke", tp=world, rank=rank,
+                               device="cuda:0", seed=5)
+        eng = Engine("llama-smoke", device="cuda:0", max_seq=256,
+                     max_batch=2, model=shard, tp_rank=rank,
+                     tp_world=world, use_graphs=False)
+        if rank > 0:
+            eng.follower_loop()
+        else:
+            eng.start()
+            out = eng.generate([3, 7, 11, 500], max_tokens=6)
+            assert len(out) == 6, out
+            eng.stop()
+        del eng, shard
+        # ---- train path: one TP=2 fwd+bwd+step through the Trainer
+        from skypilot_amd.train.trainer import TrainConfig, Trainer
+        cfg = TrainConfig(model="llama-smoke", micro_batch=1,
+                          seq_len=256, tp=world, device="cuda")
+        tr = Trainer(cfg)
+        loss0 = tr.train_step()
+        loss1 = tr.train_step()
+        assert loss0 == loss0 and loss1 == loss1  # finite
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+        q.put((rank, f"{e}\n{traceback.format_exc()}"))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.gpu
+def test_tp2_serve_and_train_one_gpu():
+    """TP=2 leader/follower serving AND TP=2 training with both ranks
+    sharing cuda:0 over gloo: validates the tensor-parallel code paths
+    on real MI355X kernels at shard shapes without needing 8 GPUs
+    (RCCL forbids two ranks on one device; gloo all-reduces via host)."""
+    import socket
+    import torch.multiprocessing as mp
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp2_gpu_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=600) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, res in results:
+        assert res == "ok", f"rank {rank}: {res}"
