@@ -1,9 +1,12 @@
-"""Optimizer — feasibility + placement over the pool.
+"""Optimizer — feasibility + cost ranking over the pool.
 
 Reference: sky/optimizer.py:109 (DP over DAG with per-cloud catalog
-enumeration).  On a one-pool deployment the search space is the pool
-inventory; the interface (optimize(dag) -> annotated dag) is preserved
-so a multi-pool catalog can slot in later.
+enumeration and a price model).  On a one-pool deployment the search
+space is the pool inventory; `any_of` candidates are ranked by a
+config-driven hourly-cost estimate (pool.prices; spot discounted),
+`ordered` keeps the user's preference order, and provision-time
+failover walks the ranked list.  The interface (optimize(dag) ->
+annotated dag) is preserved so a multi-pool catalog can slot in later.
 """
 from __future__ import annotations
 
@@ -15,10 +18,29 @@ from skypilot_amd.utils.gpu_topology import detect_gpus
 
 class Optimizer:
     @classmethod
+    def estimate_hourly_cost(cls, task, res) -> float:
+        """Config-driven price model (config `pool.prices`: per-GPU and
+        per-CPU $/hr plus a spot discount factor; defaults keep the
+        round-1 cheapest-first heuristic ordering)."""
+        from skypilot_amd import config as sky_config
+        gpu_price = float(sky_config.get_nested(
+            ["pool", "prices", "MI355X"], 2.0))
+        cpu_price = float(sky_config.get_nested(
+            ["pool", "prices", "cpu"], 0.05))
+        spot_mult = float(sky_config.get_nested(
+            ["pool", "prices", "spot_discount"], 0.3))
+        acc = res.accelerator_count * task.num_nodes
+        cost = acc * gpu_price + (res.cpus or 1) * cpu_price
+        if res.use_spot:
+            cost *= spot_mult
+        return cost
+
+    @classmethod
     def optimize(cls, dag: Dag, quiet: bool = True) -> Dag:
         dag = to_dag(dag)
         for task in dag.tasks:
             cands = task.resources.candidates or (task.resources,)
+            ordered = getattr(task.resources, "ordered", False)
             feasible, errs = [], []
             for cand in cands:
                 try:
@@ -28,10 +50,17 @@ class Optimizer:
                     errs.append(str(e))
             if not feasible:
                 raise ResourcesUnavailableError("; ".join(errs))
+            if not ordered and len(feasible) > 1:
+                # rank any_of candidates cheapest-first by the price
+                # model (reference: optimizer cost ranking)
+                feasible.sort(
+                    key=lambda r: cls.estimate_hourly_cost(task, r))
             # keep the feasible candidates (in order) for provision-time
             # failover in execution.py
             task.resources = feasible[0]
             task.resources.candidates = tuple(feasible)
+            task.estimated_hourly_cost = cls.estimate_hourly_cost(
+                task, feasible[0])
         return dag
 
     @staticmethod

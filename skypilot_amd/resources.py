@@ -133,6 +133,9 @@ class Resources:
     labels: Dict[str, str] = field(default_factory=dict)
     autostop: Optional[AutostopConfig] = None
     priority: Optional[int] = None
+    # True when the candidate list came from `ordered` (preference
+    # order is the user's; the optimizer must not re-rank it).
+    ordered: bool = False
     # Failover candidates (reference: any_of/ordered resources,
     # sky/resources.py multi-candidate sets).  candidates[0] is self's
     # config; execution tries each in order on
@@ -174,6 +177,7 @@ class Resources:
                                           r.cpus or 0))
             primary = cands[0]
             primary.candidates = tuple(cands)
+            primary.ordered = "ordered" in cfg
             return primary
         acc, n = parse_accelerators(cfg.get("accelerators"))
         cpus, cpus_min = _parse_plus(cfg.get("cpus"))

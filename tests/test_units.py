@@ -346,3 +346,50 @@ GPU[3]          : (Topology) Numa Node: 1
     t3 = parse_showtopo("======= Link Type between two GPUs =====\n"
                         "       GPU0\nGPU0   0\n")
     assert t3["n_gpus"] == 1 and t3["fully_connected_xgmi"] is False
+
+
+def test_optimizer_cost_ranking(monkeypatch, tmp_path):
+    """any_of candidates rank by the config price model (spot
+    discounted); ordered lists keep the user's order (reference:
+    sky/optimizer.py cost ranking)."""
+    import os
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "8")
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    from skypilot_amd import config as sky_config
+    sky_config.load(refresh=True)
+    from skypilot_amd.optimizer import Optimizer
+    from skypilot_amd.task import Task
+
+    # any_of: 8 on-demand GPUs vs 8 spot GPUs -> spot is cheaper
+    t = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"any_of": [
+            {"accelerators": "MI355X:8"},
+            {"accelerators": "MI355X:8", "use_spot": True},
+        ]}})
+    from skypilot_amd.dag import to_dag
+    dag = Optimizer.optimize(to_dag(t))
+    best = dag.tasks[0].resources
+    assert best.use_spot is True
+    assert dag.tasks[0].estimated_hourly_cost < 8 * 2.0
+    # ordered: user's order wins even when later is cheaper
+    t2 = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"ordered": [
+            {"accelerators": "MI355X:8"},
+            {"accelerators": "MI355X:1", "use_spot": True},
+        ]}})
+    dag2 = Optimizer.optimize(to_dag(t2))
+    assert dag2.tasks[0].resources.accelerator_count == 8
+    assert dag2.tasks[0].resources.use_spot is False
+    # infeasible candidates are pruned before ranking
+    t3 = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"any_of": [
+            {"accelerators": "MI355X:64"},
+            {"accelerators": "MI355X:2"},
+        ]}})
+    dag3 = Optimizer.optimize(to_dag(t3))
+    assert dag3.tasks[0].resources.accelerator_count == 2
