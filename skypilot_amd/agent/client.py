@@ -78,6 +78,38 @@ class AgentClient:
     def is_idle(self) -> bool:
         return bool(self._get("/idle")["idle"])
 
+    # ---- interactive exec sessions (PTY shell over HTTP streaming;
+    # reference: sky websocket SSH proxy / `sky ssh`) ------------------
+    def exec_start(self, cmd=None, env: Optional[dict] = None,
+                   term: str = "xterm-256color") -> str:
+        body: Dict[str, Any] = {"term": term}
+        if cmd:
+            body["cmd"] = cmd
+        if env:
+            body["env"] = env
+        return self._post("/exec/start", body)["sid"]
+
+    def exec_stdin(self, sid: str, data: bytes) -> bool:
+        r = self._client.post(f"{self.base}/exec/{sid}/stdin",
+                              content=data)
+        r.raise_for_status()
+        return bool(r.json().get("ok"))
+
+    def exec_stdout(self, sid: str) -> Iterator[bytes]:
+        with self._client.stream("GET", f"{self.base}/exec/{sid}/stdout",
+                                 timeout=None) as r:
+            for chunk in r.iter_bytes():
+                yield chunk
+
+    def exec_status(self, sid: str) -> Dict[str, Any]:
+        return self._get(f"/exec/{sid}")
+
+    def exec_resize(self, sid: str, rows: int, cols: int) -> None:
+        self._post(f"/exec/{sid}/resize", {"rows": rows, "cols": cols})
+
+    def exec_close(self, sid: str) -> None:
+        self._post(f"/exec/{sid}/close")
+
     def wait_job(self, job_id: int, timeout: float = 3600,
                  poll: float = 0.5) -> Dict[str, Any]:
         from skypilot_amd.agent import job_lib

@@ -20,7 +20,7 @@ import time
 from pathlib import Path
 
 import uvicorn
-from fastapi import FastAPI
+from fastapi import FastAPI, Request
 from fastapi.responses import StreamingResponse
 
 from skypilot_amd.agent import job_lib
@@ -37,7 +37,6 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
         # Per-cluster shared-secret auth (reference: skylet gRPC is only
         # reachable through an SSH tunnel — cloud_vm_ray_backend.py:2414;
         # an HTTP agent on a pod network needs its own gate).
-        from fastapi import Request
         from fastapi.responses import JSONResponse
 
         @app.middleware("http")
@@ -214,6 +213,135 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
     @app.get("/idle")
     def idle():
         return {"idle": table.is_idle()}
+
+    # ---- interactive exec sessions (reference: the websocket SSH proxy
+    # in sky/server/server.py + `sky ssh`).  No websocket stack ships in
+    # this offline image (uvicorn has neither `websockets` nor `wsproto`),
+    # so the same duplex contract is carried over plain HTTP/1.1: stdin
+    # arrives as raw POST bodies, stdout is one long chunked-streaming
+    # GET, resize/close are control POSTs.  The shell runs on a real PTY
+    # in the cluster workdir, under the agent token gate above. ----------
+    import fcntl
+    import pty
+    import secrets
+    import select
+    import signal
+    import struct
+    import termios
+
+    exec_sessions: dict = {}
+
+    def _reap_session(sid: str):
+        s = exec_sessions.pop(sid, None)
+        if not s:
+            return
+        try:
+            os.kill(s["pid"], signal.SIGKILL)
+        except ProcessLookupError:
+            pass
+        try:
+            os.waitpid(s["pid"], os.WNOHANG)
+        except ChildProcessError:
+            pass
+        try:
+            os.close(s["fd"])
+        except OSError:
+            pass
+
+    @app.post("/exec/start")
+    def exec_start(body: dict = None):
+        body = body or {}
+        cmd = body.get("cmd") or ["/bin/bash", "-i"]
+        if isinstance(cmd, str):
+            cmd = ["/bin/bash", "-lc", cmd]
+        cwd = Path(cluster_dir) / "workdir"
+        cwd.mkdir(parents=True, exist_ok=True)
+        env = dict(os.environ)
+        env["TERM"] = body.get("term", "xterm-256color")
+        env.update({str(k): str(v) for k, v in (body.get("env") or {}).items()})
+        pid, fd = pty.fork()
+        if pid == 0:  # child: become the shell on the PTY slave
+            try:
+                os.chdir(cwd)
+                os.execvpe(cmd[0], cmd, env)
+            finally:
+                os._exit(127)
+        sid = secrets.token_hex(8)
+        exec_sessions[sid] = {"fd": fd, "pid": pid, "exit_code": None}
+        state["last_active"] = time.time()
+        return {"sid": sid, "pid": pid}
+
+    @app.post("/exec/{sid}/stdin")
+    async def exec_stdin(sid: str, request: Request):
+        s = exec_sessions.get(sid)
+        if s is None:
+            return {"ok": False, "error": "no such session"}
+        data = await request.body()
+        try:
+            os.write(s["fd"], data)
+        except OSError:
+            return {"ok": False, "error": "session closed"}
+        state["last_active"] = time.time()
+        return {"ok": True, "n": len(data)}
+
+    @app.get("/exec/{sid}/stdout")
+    def exec_stdout(sid: str):
+        s = exec_sessions.get(sid)
+        if s is None:
+            return StreamingResponse(iter(()),
+                                     media_type="application/octet-stream")
+
+        def gen():
+            fd = s["fd"]
+            try:
+                while sid in exec_sessions:
+                    r, _, _ = select.select([fd], [], [], 0.25)
+                    if not r:
+                        continue
+                    try:
+                        chunk = os.read(fd, 65536)
+                    except OSError:  # EIO: shell exited, PTY drained
+                        break
+                    if not chunk:
+                        break
+                    yield chunk
+            finally:
+                try:
+                    done, st = os.waitpid(s["pid"], os.WNOHANG)
+                    if done:
+                        s["exit_code"] = os.waitstatus_to_exitcode(st)
+                except ChildProcessError:
+                    pass
+
+        return StreamingResponse(gen(),
+                                 media_type="application/octet-stream")
+
+    @app.get("/exec/{sid}")
+    def exec_status(sid: str):
+        s = exec_sessions.get(sid)
+        if s is None:
+            return {"alive": False}
+        try:
+            os.kill(s["pid"], 0)
+            alive = True
+        except ProcessLookupError:
+            alive = False
+        return {"alive": alive, "exit_code": s["exit_code"]}
+
+    @app.post("/exec/{sid}/resize")
+    def exec_resize(sid: str, body: dict):
+        s = exec_sessions.get(sid)
+        if s is None:
+            return {"ok": False}
+        winsz = struct.pack("HHHH", int(body.get("rows", 24)),
+                            int(body.get("cols", 80)), 0, 0)
+        fcntl.ioctl(s["fd"], termios.TIOCSWINSZ, winsz)
+        return {"ok": True}
+
+    @app.post("/exec/{sid}/close")
+    def exec_close(sid: str):
+        _reap_session(sid)
+        return {"ok": True}
 
     return app
 

@@ -79,6 +79,17 @@ def launch(entrypoint, cluster, gpus, num_nodes, env, down,
                       result["job_id"])
 
 
+@cli.command("ssh")
+@click.argument("cluster")
+@click.option("--cmd", default=None,
+              help="run this command instead of an interactive shell")
+def ssh_cmd(cluster, cmd):
+    """Open an interactive shell on the cluster head (reference: sky
+    ssh — tunneled through the API server, no direct node access
+    needed)."""
+    raise SystemExit(sdk.ssh_shell(cluster, cmd=cmd))
+
+
 @cli.command("exec")
 @click.argument("cluster")
 @click.argument("entrypoint")

@@ -446,3 +446,120 @@ def storage_list() -> str:
 
 def storage_delete(name: str) -> str:
     return _submit("storage_delete", {"name": name})
+
+
+# interactive shell tunnel (reference: `sky ssh` + the websocket SSH
+# proxy in sky/server/server.py; HTTP-streaming PTY sessions here — see
+# agent/daemon.py exec sessions) --------------------------------------------
+def ssh_start(cluster_name: str, cmd=None, env: Optional[dict] = None) -> str:
+    body: Dict[str, Any] = {}
+    if cmd:
+        body["cmd"] = cmd
+    if env:
+        body["env"] = env
+    with _client() as c:
+        r = c.post(f"/api/v1/ssh/{cluster_name}/start", json=body,
+                   headers=_auth_headers())
+        if r.status_code != 200:
+            raise ApiServerError(f"ssh start: {r.status_code} {r.text[:300]}")
+        return r.json()["sid"]
+
+
+def ssh_stdin(cluster_name: str, sid: str, data: bytes) -> None:
+    with _client() as c:
+        r = c.post(f"/api/v1/ssh/{cluster_name}/{sid}/stdin", content=data,
+                   headers=_auth_headers())
+        if r.status_code != 200:
+            raise ApiServerError(f"ssh stdin: {r.status_code}")
+
+
+def ssh_stdout(cluster_name: str, sid: str):
+    """Yield the session's PTY output until the shell exits."""
+    if _TEST_CLIENT is not None:
+        with _TEST_CLIENT.stream(
+                "GET", f"/api/v1/ssh/{cluster_name}/{sid}/stdout",
+                headers=_auth_headers()) as r:
+            yield from r.iter_bytes()
+        return
+    with httpx.Client(base_url=server_url(), timeout=None,
+                      headers=_auth_headers()) as c:
+        with c.stream("GET",
+                      f"/api/v1/ssh/{cluster_name}/{sid}/stdout") as r:
+            yield from r.iter_bytes()
+
+
+def ssh_resize(cluster_name: str, sid: str, rows: int, cols: int) -> None:
+    with _client() as c:
+        c.post(f"/api/v1/ssh/{cluster_name}/{sid}/resize",
+               json={"rows": rows, "cols": cols}, headers=_auth_headers())
+
+
+def ssh_close(cluster_name: str, sid: str) -> None:
+    with _client() as c:
+        c.post(f"/api/v1/ssh/{cluster_name}/{sid}/close",
+               headers=_auth_headers())
+
+
+def ssh_shell(cluster_name: str, cmd=None) -> int:
+    """Interactive shell on the cluster head (reference: `sky ssh`).
+
+    Puts the local TTY in raw mode and pumps stdin/stdout through the
+    server's tunnel; with a non-TTY stdin, pipes it through and exits
+    when the remote shell does.  Returns the remote exit code when the
+    agent reports one, else 0."""
+    import select as _select
+    import shutil
+    import threading
+
+    sid = ssh_start(cluster_name, cmd=cmd)
+    sz = shutil.get_terminal_size()
+    try:
+        ssh_resize(cluster_name, sid, sz.lines, sz.columns)
+    except Exception:  # noqa: BLE001 — resize is best-effort
+        pass
+    done = threading.Event()
+
+    def pump_out():
+        try:
+            for chunk in ssh_stdout(cluster_name, sid):
+                sys.stdout.buffer.write(chunk)
+                sys.stdout.buffer.flush()
+        finally:
+            done.set()
+
+    t = threading.Thread(target=pump_out, daemon=True)
+    t.start()
+    is_tty = sys.stdin.isatty()
+    old_attrs = None
+    if is_tty:
+        import termios
+        import tty
+        old_attrs = termios.tcgetattr(sys.stdin.fileno())
+        tty.setraw(sys.stdin.fileno())
+    try:
+        fd = sys.stdin.fileno()
+        while not done.is_set():
+            r, _, _ = _select.select([fd], [], [], 0.2)
+            if not r:
+                continue
+            data = os.read(fd, 4096)
+            if not data:  # EOF on piped stdin
+                ssh_stdin(cluster_name, sid, b"exit\n")
+                done.wait(timeout=30)
+                break
+            ssh_stdin(cluster_name, sid, data)
+    finally:
+        if old_attrs is not None:
+            import termios
+            termios.tcsetattr(sys.stdin.fileno(), termios.TCSADRAIN,
+                              old_attrs)
+        try:
+            with _client() as c:
+                st = c.get(f"/api/v1/ssh/{cluster_name}/{sid}/status",
+                           headers=_auth_headers())
+        except Exception:  # noqa: BLE001
+            st = None
+        ssh_close(cluster_name, sid)
+    if st is not None and st.status_code == 200:
+        return int(st.json().get("exit_code") or 0)
+    return 0

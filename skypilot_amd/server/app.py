@@ -297,6 +297,99 @@ def create_app(start_workers: bool = True) -> FastAPI:
                  ("request_id", "name", "status", "created_at",
                   "finished_at", "error")} for r in reqs]
 
+    # ---- interactive shell tunnel (reference: the websocket SSH proxy
+    # in sky/server/server.py + `sky ssh`; carried over HTTP streaming
+    # here — see agent/daemon.py exec sessions).  The server authenticates
+    # the caller, checks cluster ownership, then bridges to the cluster
+    # agent with the per-cluster bearer token the client never sees. ------
+    def _cluster_agent(cluster_name: str, ident: Dict[str, str]):
+        rec = global_state.get_cluster(cluster_name)
+        if rec is None:
+            raise HTTPException(404, f"no cluster {cluster_name!r}")
+        owner = rec.get("user")
+        if ident["role"] != "admin" and owner not in (None, ident["user"]):
+            raise HTTPException(
+                403, f"cluster {cluster_name!r} belongs to {owner!r}")
+        handle = rec.get("handle") or {}
+        port = handle.get("agent_port")
+        if not port:
+            raise HTTPException(409, f"cluster {cluster_name!r} has no "
+                                     "agent (not UP?)")
+        from skypilot_amd.agent.client import AgentClient
+        return AgentClient(port, host=handle.get("agent_host", "127.0.0.1"),
+                           token=handle.get("agent_token"))
+
+    @app.post(API_PREFIX + "/ssh/{cluster_name}/start")
+    def ssh_start(cluster_name: str, request: Request,
+                  body: Dict[str, Any] = None):
+        ident = _identity(request)
+        agent = _cluster_agent(cluster_name, ident)
+        try:
+            body = body or {}
+            sid = agent.exec_start(cmd=body.get("cmd"),
+                                   env=body.get("env"),
+                                   term=body.get("term", "xterm-256color"))
+        finally:
+            agent.close()
+        return {"sid": sid}
+
+    @app.post(API_PREFIX + "/ssh/{cluster_name}/{sid}/stdin")
+    async def ssh_stdin(cluster_name: str, sid: str, request: Request):
+        ident = _identity(request)
+        agent = _cluster_agent(cluster_name, ident)
+        try:
+            data = await request.body()
+            ok = agent.exec_stdin(sid, data)
+        finally:
+            agent.close()
+        return {"ok": ok}
+
+    @app.get(API_PREFIX + "/ssh/{cluster_name}/{sid}/stdout")
+    def ssh_stdout(cluster_name: str, sid: str, request: Request = None):
+        ident = _identity(request)
+        agent = _cluster_agent(cluster_name, ident)
+
+        def gen():
+            try:
+                yield from agent.exec_stdout(sid)
+            finally:
+                agent.close()
+
+        return StreamingResponse(gen(),
+                                 media_type="application/octet-stream")
+
+    @app.get(API_PREFIX + "/ssh/{cluster_name}/{sid}/status")
+    def ssh_status(cluster_name: str, sid: str, request: Request = None):
+        ident = _identity(request)
+        agent = _cluster_agent(cluster_name, ident)
+        try:
+            return agent.exec_status(sid)
+        finally:
+            agent.close()
+
+    @app.post(API_PREFIX + "/ssh/{cluster_name}/{sid}/resize")
+    def ssh_resize(cluster_name: str, sid: str, request: Request,
+                   body: Dict[str, Any] = None):
+        ident = _identity(request)
+        agent = _cluster_agent(cluster_name, ident)
+        try:
+            body = body or {}
+            agent.exec_resize(sid, int(body.get("rows", 24)),
+                              int(body.get("cols", 80)))
+        finally:
+            agent.close()
+        return {"ok": True}
+
+    @app.post(API_PREFIX + "/ssh/{cluster_name}/{sid}/close")
+    def ssh_close(cluster_name: str, sid: str, request: Request):
+        ident = _identity(request)
+        agent = _cluster_agent(cluster_name, ident)
+        try:
+            agent.exec_close(sid)
+        finally:
+            agent.close()
+        return {"ok": True}
+
     # ---- log streaming for cluster jobs (proxied to the node agent) -------
     @app.get(API_PREFIX + "-logs/{cluster_name}")
     def cluster_logs(cluster_name: str, job_id: Optional[int] = None,

