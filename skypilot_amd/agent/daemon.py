@@ -243,10 +243,12 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
             os.waitpid(s["pid"], os.WNOHANG)
         except ChildProcessError:
             pass
-        try:
-            os.close(s["fd"])
-        except OSError:
-            pass
+        for k in ("fd", "wfd"):
+            try:
+                if s.get(k) is not None and (k == "fd" or s[k] != s["fd"]):
+                    os.close(s[k])
+            except OSError:
+                pass
 
     @app.post("/exec/start")
     def exec_start(body: dict = None):
@@ -259,17 +261,31 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
         env = dict(os.environ)
         env["TERM"] = body.get("term", "xterm-256color")
         env.update({str(k): str(v) for k, v in (body.get("env") or {}).items()})
-        pid, fd = pty.fork()
-        if pid == 0:  # child: become the shell on the PTY slave
-            try:
-                os.chdir(cwd)
-                os.execvpe(cmd[0], cmd, env)
-            finally:
-                os._exit(127)
+        try:
+            pid, fd = pty.fork()
+            if pid == 0:  # child: become the shell on the PTY slave
+                try:
+                    os.chdir(cwd)
+                    os.execvpe(cmd[0], cmd, env)
+                finally:
+                    os._exit(127)
+            sess = {"fd": fd, "wfd": fd, "pid": pid, "exit_code": None,
+                    "tty": True}
+        except OSError:
+            # No PTY devices (restricted container without /dev/pts):
+            # fall back to a pipe-backed session — same wire contract,
+            # no terminal semantics (resize is a no-op, no input echo).
+            proc = subprocess.Popen(
+                cmd, cwd=str(cwd), env=env, stdin=subprocess.PIPE,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                bufsize=0, start_new_session=True)
+            sess = {"fd": proc.stdout.fileno(),
+                    "wfd": proc.stdin.fileno(), "pid": proc.pid,
+                    "exit_code": None, "tty": False, "proc": proc}
         sid = secrets.token_hex(8)
-        exec_sessions[sid] = {"fd": fd, "pid": pid, "exit_code": None}
+        exec_sessions[sid] = sess
         state["last_active"] = time.time()
-        return {"sid": sid, "pid": pid}
+        return {"sid": sid, "pid": sess["pid"], "tty": sess["tty"]}
 
     @app.post("/exec/{sid}/stdin")
     async def exec_stdin(sid: str, request: Request):
@@ -278,7 +294,7 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
             return {"ok": False, "error": "no such session"}
         data = await request.body()
         try:
-            os.write(s["fd"], data)
+            os.write(s["wfd"], data)
         except OSError:
             return {"ok": False, "error": "session closed"}
         state["last_active"] = time.time()
@@ -331,7 +347,7 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
     @app.post("/exec/{sid}/resize")
     def exec_resize(sid: str, body: dict):
         s = exec_sessions.get(sid)
-        if s is None:
+        if s is None or not s.get("tty"):
             return {"ok": False}
         winsz = struct.pack("HHHH", int(body.get("rows", 24)),
                             int(body.get("cols", 80)), 0, 0)
