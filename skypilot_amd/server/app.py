@@ -259,6 +259,18 @@ def create_app(start_workers: bool = True) -> FastAPI:
         from skypilot_amd.server.dashboard import render
         return HTMLResponse(render())
 
+    @app.get("/dashboard/cluster/{cluster_name}")
+    def dashboard_cluster(cluster_name: str):
+        from fastapi.responses import HTMLResponse
+        from skypilot_amd.server.dashboard import render_cluster
+        return HTMLResponse(render_cluster(cluster_name))
+
+    @app.get("/dashboard/cluster/{cluster_name}/job/{job_id}")
+    def dashboard_job(cluster_name: str, job_id: int):
+        from fastapi.responses import HTMLResponse
+        from skypilot_amd.server.dashboard import render_job_logs
+        return HTMLResponse(render_job_logs(cluster_name, job_id))
+
     @app.get("/metrics")
     def metrics():
         # Prometheus metrics (reference: sky/server/metrics.py).

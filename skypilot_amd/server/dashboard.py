@@ -9,12 +9,25 @@ import time
 from typing import Any, Dict, List
 
 
+class _Raw(str):
+    """Cell rendered without escaping (pre-escaped HTML, e.g. links)."""
+
+
+def _link(href: str, text: str) -> "_Raw":
+    return _Raw(f'<a href="{html.escape(href)}" style="color:#e8443a">'
+                f"{html.escape(str(text))}</a>")
+
+
+def _cell(c) -> str:
+    return str(c) if isinstance(c, _Raw) else html.escape(str(c))
+
+
 def _table(title: str, headers: List[str], rows: List[List[Any]]) -> str:
     if not rows:
         body = f"<tr><td colspan={len(headers)} class=empty>none</td></tr>"
     else:
         body = "".join(
-            "<tr>" + "".join(f"<td>{html.escape(str(c))}</td>" for c in row)
+            "<tr>" + "".join(f"<td>{_cell(c)}</td>" for c in row)
             + "</tr>" for row in rows)
     head = "".join(f"<th>{h}</th>" for h in headers)
     return (f"<h2>{title}</h2><table><thead><tr>{head}</tr></thead>"
@@ -52,7 +65,8 @@ def render() -> str:
     for c in global_state.list_clusters():
         h = c["handle"]
         clusters.append([
-            c["name"], c["status"],
+            _link(f"/dashboard/cluster/{c['name']}", c["name"]),
+            c["status"],
             f"{h.get('num_nodes', 1)}x{h.get('gpus_per_node', 0)} GPU",
             ",".join(map(str, h.get("gpu_ids", []))) or "-",
             time.strftime("%m-%d %H:%M",
@@ -138,6 +152,30 @@ def render() -> str:
              f"{((h.get('torn_down_at') or 0) - (h.get('launched_at') or 0)) / 60:.0f} min"]
             for h in global_state.list_cluster_history(limit=10)]
 
+    # users + workspaces (reference dashboard: users / workspaces tabs)
+    user_rows = []
+    try:
+        from skypilot_amd import users as users_lib
+        user_rows = [[u["name"], u["role"]]
+                     for u in users_lib.list_users()]
+    except Exception:  # noqa: BLE001
+        pass
+    ws_rows = []
+    try:
+        seen = {}
+        for c in global_state.list_clusters():
+            w = c.get("workspace") or "default"
+            seen[w] = seen.get(w, 0) + 1
+        from skypilot_amd import config as sky_config
+        for wname, wcfg in (sky_config.get_nested(["workspaces"], {})
+                            or {}).items():
+            seen.setdefault(wname, 0)
+        ws_rows = [[w, n, "private" if (sky_config.get_nested(
+            ["workspaces", w, "private"], False)) else "open"]
+            for w, n in sorted(seen.items())]
+    except Exception:  # noqa: BLE001
+        pass
+
     return f"""<!doctype html><html><head><title>skypilot-amd</title>
 <meta http-equiv="refresh" content="5">
 <style>
@@ -166,5 +204,85 @@ th {{ background:#222; color:#e8443a; }}
         ["id", "type", "user", "status", "duration", "at"], reqs)}
 {_table("Cluster history", ["name", "launched", "torn down", "lifetime"],
         hist)}
+{_table("Users", ["name", "role"], user_rows)}
+{_table("Workspaces", ["name", "clusters", "access"], ws_rows)}
 <p style="color:#555">auto-refreshes every 5s · {time.strftime("%H:%M:%S")}
 </p></body></html>"""
+
+
+_STYLE = """
+<style>
+body { font-family: ui-monospace, monospace; margin: 2em; background:#111;
+       color:#ddd; }
+h1 { color:#e8443a; } h2 { color:#ccc; margin-top:1.4em; }
+table { border-collapse: collapse; min-width: 48em; }
+th, td { border:1px solid #333; padding:4px 10px; text-align:left; }
+th { background:#222; color:#e8443a; }
+.empty { color:#666; } a { color:#e8443a; }
+pre { background:#1a1a1a; padding:1em; overflow-x:auto; }
+</style>"""
+
+
+def render_cluster(name: str) -> str:
+    """Per-cluster drill-down: handle, job queue, events, agent health
+    (reference: the dashboard's cluster detail page)."""
+    from skypilot_amd import core, global_state
+    rec = global_state.get_cluster(name)
+    if rec is None:
+        return (f"<!doctype html><html><head>{_STYLE}</head><body>"
+                f"<h1>no cluster {html.escape(name)!s}</h1>"
+                '<p><a href="/dashboard">back</a></p></body></html>')
+    h = rec.get("handle") or {}
+    handle_rows = [[k, h[k]] for k in sorted(h)
+                   if k not in ("agent_token",)]  # never render secrets
+    try:
+        jobs = [[j["job_id"], j.get("name") or "-", j["status"],
+                 ",".join(map(str, (j.get("spec") or {}).get("gpu_ids",
+                                                             []))) or "-",
+                 _link(f"/dashboard/cluster/{name}/job/{j['job_id']}",
+                       "logs")]
+                for j in core.queue(name)]
+    except Exception as e:  # noqa: BLE001 — agent may be down
+        jobs = [["-", "-", f"unreachable: {e}", "-", "-"]]
+    events = [[time.strftime("%m-%d %H:%M:%S",
+                             time.localtime(e.get("ts") or 0)),
+               e.get("event"), e.get("detail") or "-"]
+              for e in core.cluster_events(name)[-25:]]
+    agent = "no agent"
+    if h.get("agent_port"):
+        from skypilot_amd.agent.client import AgentClient
+        c = AgentClient(h["agent_port"], token=h.get("agent_token"),
+                        timeout=3.0)
+        try:
+            agent = "healthy" if c.healthy() else "UNREACHABLE"
+        finally:
+            c.close()
+    return f"""<!doctype html><html><head>
+<title>{html.escape(name)} — skypilot-amd</title>{_STYLE}</head><body>
+<h1>{html.escape(name)} <small style="color:#666">{rec["status"]} ·
+agent {agent}</small></h1>
+<p><a href="/dashboard">← all clusters</a></p>
+{_table("Handle", ["key", "value"], handle_rows)}
+{_table("Job queue", ["id", "name", "status", "gpus", ""], jobs)}
+{_table("Events", ["at", "event", "detail"], events)}
+</body></html>"""
+
+
+def render_job_logs(cluster_name: str, job_id: int, tail: int = 400) -> str:
+    """Job log view (reference: dashboard job logs page)."""
+    from skypilot_amd import core
+    lines: List[str] = []
+    try:
+        for chunk in core.tail_logs(cluster_name, job_id, follow=False):
+            lines.append(chunk.decode(errors="replace"))
+            if sum(len(x) for x in lines) > 1 << 20:
+                break
+    except Exception as e:  # noqa: BLE001
+        lines = [f"(logs unavailable: {e})"]
+    text = "".join(lines)
+    tail_text = "\n".join(text.splitlines()[-tail:])
+    return f"""<!doctype html><html><head>
+<title>job {job_id} — {html.escape(cluster_name)}</title>{_STYLE}</head>
+<body><h1>job {job_id} on {html.escape(cluster_name)}</h1>
+<p><a href="/dashboard/cluster/{html.escape(cluster_name)}">← cluster</a></p>
+<pre>{html.escape(tail_text)}</pre></body></html>"""

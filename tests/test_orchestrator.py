@@ -228,6 +228,36 @@ def test_dashboard_renders(client):
     r = client.get("/dashboard")
     assert r.status_code == 200
     assert "skypilot-amd" in r.text and "Clusters" in r.text
+    assert "Users" in r.text and "Workspaces" in r.text
+
+
+def test_dashboard_cluster_detail_and_job_logs(sky_env, client):
+    """Cluster drill-down page + job log view (reference: dashboard
+    cluster/job detail pages); agent token must never be rendered."""
+    from skypilot_amd.client import sdk
+    sdk.get(sdk.launch({"run": "echo dash-$((8*8))",
+                        "resources": {"cpus": 1}}, "dash-c"), timeout=60)
+    r = client.get("/dashboard/cluster/dash-c")
+    assert r.status_code == 200
+    assert "dash-c" in r.text and "Job queue" in r.text
+    assert "Events" in r.text and "PROVISION" in r.text
+    assert "agent_token" not in r.text
+    # the clusters table links to the detail page
+    main = client.get("/dashboard").text
+    assert "/dashboard/cluster/dash-c" in main
+    # job log page shows the run output
+    import time as _t
+    deadline = _t.time() + 30
+    while _t.time() < deadline:
+        jl = client.get("/dashboard/cluster/dash-c/job/1").text
+        if "dash-64" in jl:
+            break
+        _t.sleep(0.5)
+    assert "dash-64" in jl
+    # unknown cluster renders a not-found page, not a 500
+    r = client.get("/dashboard/cluster/nope")
+    assert r.status_code == 200 and "no cluster" in r.text
+    sdk.get(sdk.down("dash-c"))
 
 
 def test_event_callback_and_priority(client, tmp_path):
