@@ -380,7 +380,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_v3_kernel(
     }
     const int kt = it < nt_A ? it : it - nt_A;
     const int buf = it & 1;
-    __syncthreads();  // drains this tile's global_load_lds (vmcnt)
+    vm_drain();  // hipcc's barrier wait is lgkm-only; drain the
+                 // global_load_lds DMA for this buffer explicitly
+    __syncthreads();
     if (it + 1 < total_t) {
       const int kt2 = (it + 1 < nt_A) ? it + 1 : it + 1 - nt_A;
       DQ_STAGE((long long)kt2 * 64, buf ^ 1);

@@ -178,8 +178,10 @@ __global__ __launch_bounds__(64 * NW, 2) void attn_fwd_v3_t(
     }
     const int kt = it < nt_A ? it : it - nt_A;
     const int buf = it & 1;
-    // one barrier per tile: hipcc's pre-barrier waitcnt drains the
-    // in-flight global_load_lds for this buffer.
+    // one barrier per tile; the explicit vm_drain is REQUIRED: hipcc's
+    // pre-barrier waitcnt is lgkmcnt(0) only and does NOT cover the
+    // in-flight global_load_lds for this buffer (see attn_v3.h).
+    vm_drain();
     __syncthreads();
     // issue next tile's staging (possibly the light q-tile's tile 0:
     // the pipeline never drains at the boundary).

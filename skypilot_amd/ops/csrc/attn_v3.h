@@ -53,6 +53,15 @@ __device__ __forceinline__ void lgkm_wait8_bind2(tr4* a, tr4* b) {
                : "memory");
   __builtin_amdgcn_sched_barrier(0);
 }
+// Drain outstanding global_load_lds before a buffer-handoff barrier.
+// LDS-DMA completion is tracked by VMcnt, and hipcc inserts only
+// `s_waitcnt lgkmcnt(0)` before s_barrier (verified in r02 ISA) — so
+// without this a wave can read staged K/V rows before the DMA lands.
+// Intermittent small-S numerics failures (r02) were exactly this race.
+__device__ __forceinline__ void vm_drain() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+
 __device__ __forceinline__ void lgkm_wait0_bind2(tr4* a, tr4* b) {
   asm volatile("s_waitcnt lgkmcnt(0)"
                : "+v"(a->d[0]), "+v"(a->d[1]), "+v"(a->d[2]),
