@@ -318,7 +318,13 @@ def test_attention_shape_fuzz():
             (2, 256, 32, 8, False), (1, 1024, 8, 2, True),
             (2, 192, 4, 4, False),
             # S % 256 == 0 routes to the v3 8-wave 32x32 kernel
-            (2, 512, 16, 4, True), (1, 256, 8, 8, True)]:
+            (2, 512, 16, 4, True), (1, 256, 8, 8, True),
+            # v3 edge shapes around the r02 gl_lds barrier race:
+            # minimal paired grid (nq=2), single-head, non-causal v3,
+            # odd head ratios, and a deep-S single-block case
+            (1, 256, 1, 1, True), (1, 256, 8, 8, False),
+            (4, 768, 6, 2, True), (1, 2048, 1, 1, True),
+            (3, 512, 5, 1, True)]:
         q = (torch.randn(B, S, Hq, 128, device="cuda") * 0.5).bfloat16()
         k = (torch.randn(B, S, Hkv, 128, device="cuda") * 0.5).bfloat16()
         v = (torch.randn(B, S, Hkv, 128, device="cuda") * 0.5).bfloat16()
