@@ -71,3 +71,32 @@ def test_bench_inner_direct_under_torchrun_env(tmp_path):
     res = json.loads(line)
     # inner path: no sky_launched marker
     assert "sky_launched" not in res["config"]
+
+
+def test_bench_torchrun_world2_cpu(tmp_path):
+    """The driver's SCALE launch shape: torchrun --nproc-per-node 2
+    bench.py --gpus 2.  On CPU this exercises setup_distributed (gloo),
+    the DDP trainer, the MAX-over-ranks timing reduction and the
+    whole-job token aggregation (value counts both ranks)."""
+    env = dict(os.environ)
+    env.update({
+        "SKY_AMD_HOME": str(tmp_path / "home"),
+        "SKY_BENCH_ALLOW_CPU": "1",
+        "PYTHONPATH": str(REPO),
+    })
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29811", str(REPO / "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "0",
+         "--model", "llama-debug", "--micro-batch", "1",
+         "--seq-len", "32"],
+        capture_output=True, text=True, timeout=600, env=env,
+        cwd=str(REPO))
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
+    line = [l for l in out.stdout.splitlines()
+            if l.startswith('{"metric"')][-1]
+    res = json.loads(line)
+    assert res["n_gpus"] == 2
+    assert res["config"]["parallelism"] == "dp2"
+    assert res["config"]["global_batch"] == 2
