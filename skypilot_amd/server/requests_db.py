@@ -52,7 +52,14 @@ def _conn():
     conn = sqlite3.connect(api_dir() / "requests.db", timeout=30)
     try:
         conn.execute("PRAGMA busy_timeout=30000")
-        conn.execute("PRAGMA journal_mode=WAL")
+        try:
+            # WAL is a persistent property of the db file; switching it
+            # needs an exclusive lock, so under heavy concurrent opens
+            # (or teardown races in tests) this can raise "database is
+            # locked" — it is an optimization, never correctness.
+            conn.execute("PRAGMA journal_mode=WAL")
+        except sqlite3.OperationalError:
+            pass
         conn.executescript(_SCHEMA)
         cols = [r[1] for r in conn.execute("PRAGMA table_info(requests)")]
         if "user" not in cols:  # pre-RBAC databases
