@@ -119,6 +119,8 @@ def _loop():
             from skypilot_amd.jobs import pools as jobs_pools
             jobs_pools.autoscale()
             _gc_controller_logs()
+            from skypilot_amd.server import requests_db as _rdb
+            _rdb.gc_requests()
         except Exception:  # noqa: BLE001
             pass
         try:

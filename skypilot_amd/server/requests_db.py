@@ -160,3 +160,38 @@ def mark_cancelled(rid: str) -> Optional[int]:
             "UPDATE requests SET status=?, finished_at=? WHERE request_id=?",
             (CANCELLED, time.time(), rid))
     return row[1]
+
+
+def gc_requests(max_age_days: float = None, keep_latest: int = None
+                ) -> int:
+    """Retention for terminal request rows + their log files
+    (reference: sky server requests GC — the table is otherwise
+    unbounded on a long-lived server).  Keeps every non-terminal row,
+    the newest `keep_latest` terminal rows, and anything younger than
+    `max_age_days`.  Returns rows deleted."""
+    import os as _os
+    if max_age_days is None:
+        max_age_days = float(_os.environ.get(
+            "SKY_AMD_REQUEST_RETENTION_DAYS", "7"))
+    if keep_latest is None:
+        keep_latest = int(_os.environ.get(
+            "SKY_AMD_REQUEST_RETENTION_COUNT", "2000"))
+    cutoff = time.time() - max_age_days * 86400
+    removed = 0
+    with _conn() as c:
+        rows = c.execute(
+            "SELECT request_id, log_path FROM requests WHERE status IN "
+            "('SUCCEEDED','FAILED','CANCELLED') AND created_at < ? "
+            "AND request_id NOT IN (SELECT request_id FROM requests "
+            "WHERE status IN ('SUCCEEDED','FAILED','CANCELLED') "
+            "ORDER BY created_at DESC LIMIT ?)",
+            (cutoff, keep_latest)).fetchall()
+        for rid, log_path in rows:
+            c.execute("DELETE FROM requests WHERE request_id=?", (rid,))
+            removed += 1
+            if log_path:
+                try:
+                    _os.unlink(log_path)
+                except OSError:
+                    pass
+    return removed

@@ -393,3 +393,35 @@ def test_optimizer_cost_ranking(monkeypatch, tmp_path):
         ]}})
     dag3 = Optimizer.optimize(to_dag(t3))
     assert dag3.tasks[0].resources.accelerator_count == 2
+
+
+def test_request_gc_retention(monkeypatch, tmp_path):
+    """Terminal request rows past the retention window are deleted
+    (logs too); non-terminal and recent rows survive (reference: sky
+    server request retention)."""
+    import os
+    import time as _time
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    from skypilot_amd.server import requests_db as rdb
+    old = rdb.create("status", {}, "SHORT")
+    rdb.finish(old, rdb.SUCCEEDED, result={})
+    live = rdb.create("launch", {}, "LONG")  # stays PENDING
+    fresh = rdb.create("status", {}, "SHORT")
+    rdb.finish(fresh, rdb.FAILED, error="x")
+    # age the old row beyond the cutoff
+    with rdb._conn() as c:
+        c.execute("UPDATE requests SET created_at=? WHERE request_id=?",
+                  (_time.time() - 30 * 86400, old))
+    logf = rdb.get(old)["log_path"]
+    open(logf, "w").write("x")
+    removed = rdb.gc_requests(max_age_days=7, keep_latest=1)
+    assert removed == 1
+    assert rdb.get(old) is None
+    assert not os.path.exists(logf)
+    assert rdb.get(live)["status"] == "PENDING"
+    assert rdb.get(fresh) is not None
+    # keep_latest guards even ancient rows
+    with rdb._conn() as c:
+        c.execute("UPDATE requests SET created_at=? WHERE request_id=?",
+                  (_time.time() - 30 * 86400, fresh))
+    assert rdb.gc_requests(max_age_days=7, keep_latest=5) == 0
