@@ -241,3 +241,50 @@ def test_lb_under_replica_churn(client):
     assert results["fail"] / total < 0.3, results
     assert results["ok"] > 0
     sdk.get(sdk.serve_down("churn"), timeout=120)
+
+
+def test_serve_rolling_update(client):
+    """`sky serve update` bumps the service version and the controller
+    replaces stale replicas one at a time while the service stays
+    reachable (reference: sky/serve rolling updates / version
+    tracking)."""
+    import httpx
+    from skypilot_amd.client import sdk
+    task = _service_task(2)
+    task["envs"] = {"SVC_TAG": "v1"}
+    task["run"] = task["run"].replace("\"replica:\"", "\"replica-\" + "
+                                      "os.environ.get(\"SVC_TAG\",\"\") + "
+                                      "\":\"")
+    res = sdk.get(sdk.serve_up(task, "svc-roll"), timeout=120)
+    endpoint = res["endpoint"]
+    _wait_ready("svc-roll", 2)
+    r = httpx.get(endpoint + "/", timeout=10)
+    assert r.text.startswith("replica-v1:")
+    # rolling update to v2
+    task2 = dict(task)
+    task2["envs"] = {"SVC_TAG": "v2"}
+    up2 = sdk.get(sdk.serve_update(task2, "svc-roll"), timeout=60)
+    assert up2["version"] == 2
+    deadline = time.time() + 180
+    ok = False
+    while time.time() < deadline:
+        stats = sdk.get(sdk.serve_status("svc-roll"))
+        reps = stats[0]["replicas"]
+        ready = [x for x in reps if x["status"] == "READY"]
+        # availability: never zero ready replicas during the roll
+        assert len(ready) >= 1, reps
+        if (len(ready) >= 2
+                and all(x.get("version", 1) == 2 for x in ready)):
+            ok = True
+            break
+        time.sleep(1)
+    assert ok, stats
+    # traffic now serves the new version
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        txt = httpx.get(endpoint + "/", timeout=10).text
+        if txt.startswith("replica-v2:"):
+            break
+        time.sleep(0.5)
+    assert txt.startswith("replica-v2:"), txt
+    sdk.get(sdk.serve_down("svc-roll"), timeout=120)
