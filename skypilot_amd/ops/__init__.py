@@ -403,6 +403,17 @@ def rmsnorm_res(x, res, weight, eps: float = 1e-5):
     return x2, rmsnorm(x2, weight, eps)
 
 
+def decode_advance(logits, stage, ring, ctr):
+    """Fused greedy decode advance: per-row argmax over `logits`
+    [b, 1, V] plus the in-graph state bump — ring token store,
+    stage[0] next-token feedback, stage[1]/[3]/[4] increments — in one
+    launch (decode_fused.hip; replaces torch argmax+index_copy+copy_+
+    3 adds; tie-break matches torch.argmax).  GPU-only: the decode
+    graph never runs on CPU."""
+    C = _require_native("decode_advance")
+    C.decode_advance(logits.contiguous(), stage, ring, ctr)
+
+
 def decode_swiglu_down(gu, weight):
     """silu(gate)*up followed by the down projection.  MEASURED
     NEGATIVE as a fused GEMV (skinny_gemm_swiglu kept for reference):

@@ -52,6 +52,8 @@ void gemm_nt_launch(const void*, const void*, void*, int, int, long long,
 void adamw_mt_launch(const void*, const void*, const void*, const void*,
                      const void*, long long, float, float, float, float,
                      int, float, hipStream_t);
+void decode_advance_launch(const void*, long long, void*, long long,
+                           void*, const void*, int, hipStream_t);
 void rmsnorm_res_launch(const void*, const void*, const void*, void*, void*,
                         int, int, float, hipStream_t);
 void rope_kvwrite_launch(const void*, void*, void*, void*, const void*,
@@ -322,6 +324,25 @@ std::vector<torch::Tensor> rmsnorm_res(torch::Tensor x, torch::Tensor res,
   return {x_out, h_out};
 }
 
+void decode_advance(torch::Tensor logits, torch::Tensor stage,
+                    torch::Tensor ring, torch::Tensor ctr) {
+  // fused greedy argmax + in-graph decode state bump (decode_fused.hip)
+  CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+  CHECK_GPU(stage); CHECK_CONTIG(stage);
+  CHECK_GPU(ring); CHECK_CONTIG(ring);
+  TORCH_CHECK(stage.dtype() == torch::kInt32, "decode_advance: stage int32");
+  TORCH_CHECK(ring.dtype() == torch::kInt64, "decode_advance: ring int64");
+  TORCH_CHECK(ctr.dtype() == torch::kInt64, "decode_advance: ctr int64");
+  const long long V = logits.size(-1);
+  const long long b = logits.numel() / V;
+  TORCH_CHECK(stage.size(0) == 6 && stage.size(1) == b,
+              "decode_advance: stage must be [6, b]");
+  TORCH_CHECK(ring.size(1) == b, "decode_advance: ring must be [chunk, b]");
+  decode_advance_launch(logits.data_ptr(), V, stage.data_ptr(), b,
+                        ring.data_ptr(), ctr.data_ptr(),
+                        (int)ring.size(0), cur_stream());
+}
+
 torch::Tensor rope_kvwrite(torch::Tensor qkv, torch::Tensor kc,
                            torch::Tensor vc, torch::Tensor cos_tab,
                            torch::Tensor sin_tab, torch::Tensor positions,
@@ -416,6 +437,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm_swiglu", &skinny_gemm_swiglu);
   m.def("rmsnorm_res", &rmsnorm_res);
   m.def("rope_kvwrite", &rope_kvwrite);
+  m.def("decode_advance", &decode_advance);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);

@@ -174,3 +174,64 @@ extern "C" void rope_kvwrite_launch(const void* qkv, void* q_out, void* kc,
                      (const float*)sin_tab, (const int*)positions,
                      (const int*)slot_ids, n, Hq, Hkv, D, S_max);
 }
+
+// Fused greedy-decode advance: per-row argmax over the logits + the
+// whole in-graph state bump (ring token store, next-token stage,
+// positions/pos/kv_lens increments) in ONE launch.  Replaces torch's
+// reduce_kernel argmax (measured 43 us for [1,128k] bf16 — a 6 GB/s
+// config) + index_copy + copy_ + three adds = ~6 dispatches per decode
+// step (profiles/r02_fp8_decode_kernel_stats.txt).  Tie-break matches
+// torch.argmax: lowest index wins.  The ctr increment stays a separate
+// captured op so every block reads the same ring row (stream order
+// makes that race-free inside the graph).
+extern "C" __global__ __launch_bounds__(256) void decode_advance_kernel(
+    const unsigned short* __restrict__ logits,  // [b, V] bf16
+    long long V, int* __restrict__ stage,       // [6, b] int32
+    long long b, long long* __restrict__ ring,  // [chunk, b] int64
+    const long long* __restrict__ ctr,          // [1]
+    int chunk) {
+  const int i = blockIdx.x;  // batch row
+  const unsigned short* row = logits + (long long)i * V;
+  float best = -INFINITY;
+  int bidx = 0;
+  for (long long j = threadIdx.x; j < V; j += 256) {
+    float v = bf2f(row[j]);
+    if (v > best) {
+      best = v;
+      bidx = (int)j;
+    }
+  }
+  __shared__ float smax[256];
+  __shared__ int sidx[256];
+  smax[threadIdx.x] = best;
+  sidx[threadIdx.x] = bidx;
+  __syncthreads();
+  for (int s = 128; s > 0; s >>= 1) {
+    if (threadIdx.x < s) {
+      float a = smax[threadIdx.x], c = smax[threadIdx.x + s];
+      int ai = sidx[threadIdx.x], ci = sidx[threadIdx.x + s];
+      if (c > a || (c == a && ci < ai)) {
+        smax[threadIdx.x] = c;
+        sidx[threadIdx.x] = ci;
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const int tok = sidx[0];
+    ring[(*ctr % chunk) * b + i] = (long long)tok;
+    stage[0 * b + i] = tok;   // next token fed back into the graph
+    stage[1 * b + i] += 1;    // positions
+    stage[3 * b + i] += 1;    // pos
+    stage[4 * b + i] += 1;    // kv_lens
+  }
+}
+
+extern "C" void decode_advance_launch(const void* logits, long long V,
+                                      void* stage, long long b, void* ring,
+                                      const void* ctr, int chunk,
+                                      hipStream_t stream) {
+  hipLaunchKernelGGL(decode_advance_kernel, dim3((unsigned)b), dim3(256), 0,
+                     stream, (const unsigned short*)logits, V, (int*)stage,
+                     b, (long long*)ring, (const long long*)ctr, chunk);
+}

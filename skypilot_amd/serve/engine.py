@@ -20,6 +20,7 @@ from typing import Dict, List, Optional
 import torch
 import torch.distributed as dist
 
+from skypilot_amd import ops
 from skypilot_amd.models.llama import build_model
 from skypilot_amd.serve.kv_cache import KVCache
 
@@ -272,19 +273,17 @@ class Engine:
         with torch.cuda.graph(g):
             toks, positions, ctx = unpack()
             st["out"] = self.model(toks, positions, ctx)
-            st["argmax"] = st["out"][:, 0].argmax(-1)
-            # Self-advancing state: feeding argmax back into the token
-            # stage and bumping positions/pos/kv_lens INSIDE the graph
-            # lets greedy decode replay CHUNK times with zero host work
-            # in between (the 2.5 ms/token host+sync serial tax measured
-            # at 1 stream).  Tokens land in a ring read once per chunk.
-            st["ring"].index_copy_(0, st["ctr"] % self.CHUNK,
-                                   st["argmax"].unsqueeze(0))
+            # Self-advancing state: the fused advance kernel does the
+            # argmax AND feeds it back into the token stage + bumps
+            # positions/pos/kv_lens + stores the token in the ring, so
+            # greedy decode replays CHUNK times with zero host work in
+            # between (torch's argmax alone measured 43 us/step —
+            # profiles/r02_fp8_decode_kernel_stats.txt).  The ctr
+            # increment stays a separate captured op so the kernel's
+            # ring-row read is ordered before it.
+            ops.decode_advance(st["out"][:, 0], st["stage"], st["ring"],
+                               st["ctr"])
             st["ctr"] += 1
-            st["stage"][0].copy_(st["argmax"].int())
-            st["stage"][1] += 1
-            st["stage"][3] += 1
-            st["stage"][4] += 1
         self._graphs[b] = (g, st)
         self.stats["graph_buckets"] = sorted(self._graphs)
         return self._graphs[b]

@@ -424,3 +424,48 @@ def test_fp8_decode_linear_routing():
                                    atol=0.25, rtol=0.1)
     finally:
         ops.clear_fp8_weights()
+
+
+@pytest.mark.gpu
+def test_decode_advance_fused():
+    """Fused argmax + decode state bump vs the torch reference chain
+    (tie-break must match torch.argmax: lowest index)."""
+    torch.manual_seed(11)
+    for b, V, chunk in [(1, 128256, 8), (4, 4096, 8), (8, 512, 4)]:
+        logits = (torch.randn(b, V, device="cuda") * 2).bfloat16()
+        # force ties on row 0: two equal maxima, lowest index must win
+        logits[0, 7] = 40.0
+        logits[0, V - 3] = 40.0
+        stage = torch.randint(0, 100, (6, b), dtype=torch.int32,
+                              device="cuda")
+        ring = torch.zeros(chunk, b, dtype=torch.long, device="cuda")
+        ctr = torch.tensor([5], dtype=torch.long, device="cuda")
+        ref_stage = stage.clone()
+        ref_ring = ring.clone()
+        am = logits.float().argmax(-1)
+        ref_ring[5 % chunk] = am
+        ref_stage[0] = am.int()
+        for r in (1, 3, 4):
+            ref_stage[r] += 1
+        ops.decode_advance(logits, stage, ring, ctr)
+        torch.cuda.synchronize()
+        assert int(stage[0, 0]) == 7  # tie: lowest index
+        assert torch.equal(stage, ref_stage), (b, V)
+        assert torch.equal(ring, ref_ring), (b, V)
+
+
+@pytest.mark.gpu
+def test_decode_graph_chain_matches_eager():
+    """The graph-captured greedy chain (with the fused advance) must
+    produce the same tokens as step-by-step eager decode."""
+    from skypilot_amd.serve.engine import Engine
+    eng = Engine("llama-smoke", device="cuda:0", max_seq=256, max_batch=2)
+    eng.start()
+    try:
+        prompt = list(range(1, 33))
+        out1 = eng.generate(prompt, max_tokens=24)
+        out2 = eng.generate(prompt, max_tokens=24)
+        assert out1 == out2  # deterministic greedy
+        assert len(out1) == 24
+    finally:
+        eng.stop()
