@@ -95,13 +95,14 @@ class Attention(nn.Module):
                     0).contiguous()
             cache = infer_ctx.cache
             qkv = ops.decode_linear(x.reshape(B, -1), self._wqkv)
-            q = ops.rope_kvwrite(
+            # fused rope + cache-write + attention: one kernel instead
+            # of rope_kvwrite + attn_decode (the current token's k/v is
+            # attended from registers; the cache row is written for
+            # future steps by one block per kv head).
+            o = ops.attn_decode_qkv(
                 qkv, cache.k[self.layer_idx], cache.v[self.layer_idx],
-                cos, sin, positions, infer_ctx.slot_ids_i32, self.n_q,
-                self.n_kv)
-            o = ops.attn_decode(
-                q, cache.k[self.layer_idx], cache.v[self.layer_idx],
-                infer_ctx.kv_lens, infer_ctx.slot_ids_i32, self.scale)
+                cos, sin, positions, infer_ctx.kv_lens,
+                infer_ctx.slot_ids_i32, self.n_q, self.n_kv, self.scale)
             return ops.decode_linear(o.reshape(B, self.n_q * d),
                                      self.wo.weight).view(B, S, -1)
         lin = torch.nn.functional.linear

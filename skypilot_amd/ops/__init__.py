@@ -423,6 +423,24 @@ def decode_swiglu_down(gu, weight):
     return decode_linear(swiglu(gu), weight)
 
 
+def attn_decode_qkv(qkv, kc, vc, cos, sin, positions, kv_lens, slot_ids,
+                    Hq, Hkv, scale):
+    """Fused rope + KV-cache write + decode attention from the RAW
+    packed qkv GEMV output (attention_decode.hip): the current token's
+    roped k / raw v stay in registers for this step's attention (no
+    write->read hazard) while one block per kv head writes the cache
+    row for future steps.  Replaces rope_kvwrite + attn_decode (two
+    launches -> one).  CPU/odd-D fallback composes the two ops."""
+    D = kc.shape[3]
+    if qkv.is_cuda and qkv.dtype == torch.bfloat16 and D == 128:
+        C = _require_native("attn_decode_qkv")
+        return C.attn_decode_qkv(qkv.contiguous(), kc, vc, cos, sin,
+                                 positions.int(), kv_lens.int(),
+                                 slot_ids.int(), Hq, Hkv, scale)
+    q = rope_kvwrite(qkv, kc, vc, cos, sin, positions, slot_ids, Hq, Hkv)
+    return attn_decode(q, kc, vc, kv_lens, slot_ids, scale)
+
+
 def rope_kvwrite(qkv, kc, vc, cos, sin, positions, slot_ids, Hq, Hkv):
     """Packed-qkv rope + KV-cache scatter (decode_fused.hip): applies
     rope to the q and k segments of qkv [n, (Hq+2*Hkv)*D], writes the

@@ -62,6 +62,10 @@ void rope_kvwrite_launch(const void*, void*, void*, void*, const void*,
 void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
                        hipStream_t);
+void attn_decode_qkv_launch(const void*, void*, void*, void*, const void*,
+                            const void*, const int*, const int*,
+                            const int*, int, int, int, int, float,
+                            hipStream_t);
 void attn_decode_launch(const void*, const void*, const void*, void*,
                         const int*, const int*, int, int, int, int, float,
                         hipStream_t);
@@ -201,6 +205,34 @@ torch::Tensor attn_decode(torch::Tensor Q, torch::Tensor Kc,
                      O.data_ptr(), kv_lens.data_ptr<int>(),
                      slot_ids.data_ptr<int>(), B, S_max, Hq,
                      Hkv, (float)scale, cur_stream());
+  return O;
+}
+
+torch::Tensor attn_decode_qkv(torch::Tensor qkv, torch::Tensor Kc,
+                              torch::Tensor Vc, torch::Tensor cos_tab,
+                              torch::Tensor sin_tab,
+                              torch::Tensor positions,
+                              torch::Tensor kv_lens,
+                              torch::Tensor slot_ids, int64_t Hq,
+                              int64_t Hkv, double scale) {
+  // fused rope + cache-write + decode attention (attention_decode.hip)
+  CHECK_GPU(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
+  TORCH_CHECK(positions.scalar_type() == at::kInt);
+  TORCH_CHECK(kv_lens.scalar_type() == at::kInt);
+  TORCH_CHECK(slot_ids.scalar_type() == at::kInt);
+  const int B = (int)qkv.size(0);
+  const int S_max = (int)Kc.size(1);
+  const int D = (int)Kc.size(3);
+  TORCH_CHECK(D == 128, "attn_decode_qkv: head_dim must be 128");
+  TORCH_CHECK(qkv.size(1) == (Hq + 2 * Hkv) * D,
+              "attn_decode_qkv: qkv width mismatch");
+  auto O = torch::empty({B, Hq, (long)D}, qkv.options());
+  attn_decode_qkv_launch(qkv.data_ptr(), Kc.data_ptr(), Vc.data_ptr(),
+                         O.data_ptr(), cos_tab.data_ptr(),
+                         sin_tab.data_ptr(), positions.data_ptr<int>(),
+                         kv_lens.data_ptr<int>(),
+                         slot_ids.data_ptr<int>(), B, S_max, (int)Hq,
+                         (int)Hkv, (float)scale, cur_stream());
   return O;
 }
 
@@ -438,6 +470,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_res", &rmsnorm_res);
   m.def("rope_kvwrite", &rope_kvwrite);
   m.def("decode_advance", &decode_advance);
+  m.def("attn_decode_qkv", &attn_decode_qkv);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);

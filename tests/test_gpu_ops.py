@@ -469,3 +469,40 @@ def test_decode_graph_chain_matches_eager():
         assert len(out1) == 24
     finally:
         eng.stop()
+
+
+@pytest.mark.gpu
+def test_attn_decode_qkv_fused():
+    """Fused rope+cache-write+attention vs the two-kernel chain
+    (rope_kvwrite then attn_decode) on the same random cache state —
+    outputs AND written cache rows must match."""
+    torch.manual_seed(13)
+    B, Hq, Hkv, D, S_max = 5, 8, 2, 128, 512
+    half = D // 2
+    inv_freq = 1.0 / (500000.0 ** (torch.arange(half).float() / half))
+    freqs = torch.outer(torch.arange(S_max).float(), inv_freq)
+    cos = freqs.cos().to(dev())
+    sin = freqs.sin().to(dev())
+    qkv = (torch.randn(B, (Hq + 2 * Hkv) * D) * 0.5).bfloat16().to(dev())
+    kv_lens = torch.tensor([3, 100, 512, 1, 77], dtype=torch.int32,
+                           device=dev())
+    positions = kv_lens - 1
+    slot_ids = torch.tensor([6, 0, 2, 4, 3], dtype=torch.int32,
+                            device=dev())
+    kc0 = (torch.randn(8, S_max, Hkv, D) * 0.5).bfloat16().to(dev())
+    vc0 = (torch.randn(8, S_max, Hkv, D) * 0.5).bfloat16().to(dev())
+    # reference chain
+    kc_ref, vc_ref = kc0.clone(), vc0.clone()
+    q = ops.rope_kvwrite(qkv, kc_ref, vc_ref, cos, sin, positions,
+                         slot_ids, Hq, Hkv)
+    o_ref = ops.attn_decode(q, kc_ref, vc_ref, kv_lens, slot_ids,
+                            128 ** -0.5)
+    # fused
+    kc_f, vc_f = kc0.clone(), vc0.clone()
+    o = ops.attn_decode_qkv(qkv, kc_f, vc_f, cos, sin, positions,
+                            kv_lens, slot_ids, Hq, Hkv, 128 ** -0.5)
+    torch.cuda.synchronize()
+    assert rel_err(o, o_ref) < 2e-2
+    # cache rows written identically (incl. the roped k)
+    assert torch.equal(kc_f, kc_ref)
+    assert torch.equal(vc_f, vc_ref)
