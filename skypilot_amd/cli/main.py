@@ -79,6 +79,20 @@ def launch(entrypoint, cluster, gpus, num_nodes, env, down,
                       result["job_id"])
 
 
+@cli.command("dashboard")
+def dashboard_cmd():
+    """Print (and try to open) the web dashboard URL (reference: sky
+    dashboard)."""
+    from skypilot_amd.server.app import server_url
+    url = server_url() + "/dashboard"
+    click.echo(url)
+    try:
+        import webbrowser
+        webbrowser.open(url)
+    except Exception:  # noqa: BLE001 — headless is fine, URL printed
+        pass
+
+
 @cli.command("ssh")
 @click.argument("cluster")
 @click.option("--cmd", default=None,
