@@ -200,3 +200,32 @@ def test_load_50_concurrent_mixed_with_rss(real_server, tmp_path):
     for i in range(10):
         httpx.post(base + "/api/v1/down",
                    json={"cluster_name": f"ld-{i}"}, timeout=60)
+
+
+def test_ssh_cli_one_shot_over_real_http(real_server, tmp_path):
+    """`sky ssh <cluster> --cmd ...` end-to-end through a REAL server
+    process (streaming stdout over real HTTP, not TestClient): the
+    tunnel must carry the command output and exit 0."""
+    base = real_server
+    # launch a cluster over HTTP
+    rid = httpx.post(base + "/api/v1/launch",
+                     json={"task": {"run": "true",
+                                    "resources": {"cpus": 1}},
+                           "cluster_name": "ssh-real"},
+                     timeout=30).json()["request_id"]
+    st = _wait_req(base, rid)
+    assert st["status"] == "SUCCEEDED", st
+    env = dict(os.environ)
+    env["SKY_AMD_API_SERVER"] = base
+    out = subprocess.run(
+        [sys.executable, "-m", "skypilot_amd.cli", "ssh", "ssh-real",
+         "--cmd", "echo tunnel-$((6*7)); pwd"],
+        capture_output=True, text=True, timeout=60, env=env,
+        stdin=subprocess.DEVNULL)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "tunnel-42" in out.stdout
+    assert "workdir" in out.stdout
+    rid = httpx.post(base + "/api/v1/down",
+                     json={"cluster_name": "ssh-real"},
+                     timeout=30).json()["request_id"]
+    assert _wait_req(base, rid)["status"] == "SUCCEEDED"
