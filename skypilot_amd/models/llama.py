@@ -80,9 +80,13 @@ class Attention(nn.Module):
         self.scale = 1.0 / math.sqrt(d)
         self._wqkv = None  # lazy fused [q|k|v] weight for the decode GEMV
 
-    def forward(self, x, cos, sin, positions, infer_ctx=None):
-        B, S, _ = x.shape
+    def forward(self, x, cos, sin, positions, infer_ctx=None,
+                qkv=None):
         d = self.cfg.head_dim
+        if x is None:  # decode with a precomputed (norm-fused) qkv
+            B, S = qkv.shape[0], 1
+        else:
+            B, S, _ = x.shape
         decode = infer_ctx is not None and infer_ctx.mode == "decode"
         if decode:
             # One packed qkv GEMV, then one fused rope+cache-write
@@ -94,7 +98,8 @@ class Attention(nn.Module):
                     [self.wq.weight, self.wk.weight, self.wv.weight],
                     0).contiguous()
             cache = infer_ctx.cache
-            qkv = ops.decode_linear(x.reshape(B, -1), self._wqkv)
+            if qkv is None:
+                qkv = ops.decode_linear(x.reshape(B, -1), self._wqkv)
             # fused rope + cache-write + attention: one kernel instead
             # of rope_kvwrite + attn_decode (the current token's k/v is
             # attended from registers; the cache row is written for
@@ -144,6 +149,10 @@ class MLP(nn.Module):
         lin = ops.decode_linear if decode else torch.nn.functional.linear
         return lin(ops.swiglu(lin(x, self.w_gate_up.weight)),
                    self.w_down.weight)
+
+    def down_from_gu(self, gu):
+        """Decode path with a precomputed (norm-fused) gate_up GEMV."""
+        return ops.decode_linear(ops.swiglu(gu), self.w_down.weight)
 
     # (decode path is driven from Llama.forward's fused-residual loop)
 
@@ -197,18 +206,39 @@ class Llama(nn.Module):
             # Fused-residual decode loop: every residual add rides in
             # the next rmsnorm_res kernel (one launch instead of two),
             # including the final-norm + lm-head hand-off.
+            # Norm-fused decode loop: each rmsnorm_res rides INSIDE the
+            # following fp8 GEMV (one launch; decode is dispatch-gap
+            # bound) — falls back to rmsnorm_res + GEMV when fp8 is
+            # off.  The attention/MLP modules accept the precomputed
+            # qkv / gate_up projections.
             res = None
             for blk in self.blocks:
-                if res is None:
-                    h = ops.rmsnorm(x, blk.attn_norm, blk.eps)
-                else:
-                    x, h = ops.rmsnorm_res(x, res, blk.attn_norm, blk.eps)
-                a = blk.attn(h, cos, sin, positions, infer_ctx)
-                x, h2 = ops.rmsnorm_res(x, a, blk.mlp_norm, blk.eps)
-                res = blk.mlp(h2, infer_ctx)
-            _, hf = ops.rmsnorm_res(x, res, self.final_norm,
-                                    self.cfg.norm_eps)
-            return ops.decode_linear(hf, self.lm_head.weight)
+                if not hasattr(blk.attn, "_wqkv"):
+                    # TP blocks keep the classic split projections
+                    if res is None:
+                        h = ops.rmsnorm(x, blk.attn_norm, blk.eps)
+                    else:
+                        x, h = ops.rmsnorm_res(x, res, blk.attn_norm,
+                                               blk.eps)
+                    a = blk.attn(h, cos, sin, positions, infer_ctx)
+                    x, h2 = ops.rmsnorm_res(x, a, blk.mlp_norm, blk.eps)
+                    res = blk.mlp(h2, infer_ctx)
+                    continue
+                if blk.attn._wqkv is None:
+                    blk.attn._wqkv = torch.cat(
+                        [blk.attn.wq.weight, blk.attn.wk.weight,
+                         blk.attn.wv.weight], 0).contiguous()
+                x, qkv = ops.decode_norm_linear(
+                    x, res, blk.attn_norm, blk.eps, blk.attn._wqkv)
+                a = blk.attn(None, cos, sin, positions, infer_ctx,
+                             qkv=qkv)
+                x, gu = ops.decode_norm_linear(
+                    x, a, blk.mlp_norm, blk.eps, blk.mlp.w_gate_up.weight)
+                res = blk.mlp.down_from_gu(gu)
+            _, logits = ops.decode_norm_linear(
+                x, res, self.final_norm, self.cfg.norm_eps,
+                self.lm_head.weight)
+            return logits
         for blk in self.blocks:
             x = blk(x, cos, sin, positions, infer_ctx)
         x = ops.rmsnorm(x, self.final_norm, self.cfg.norm_eps)

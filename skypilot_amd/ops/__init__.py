@@ -391,6 +391,39 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, weight)
 
 
+def decode_norm_linear(x, res, norm_w, eps, weight):
+    """Fused rmsnorm(x [+ res]) * norm_w -> fp8 GEMV in ONE kernel
+    (skinny_gemm.hip norm-fused variant) — decode is dispatch-gap
+    bound, so deleting the rmsnorm_res launch saves its kernel time
+    AND the ~5 us graph-replay gap.  Returns (x2, y) with x2 = x+res
+    (x itself when res is None).  Falls back to rmsnorm_res +
+    decode_linear when the fp8 table or shape doesn't apply."""
+    lead = x.shape[:-1]
+    n = 1
+    for d in lead:
+        n *= d
+    i = x.shape[-1]
+    o = weight.shape[0]
+    if (x.is_cuda and x.dtype == torch.bfloat16 and 1 <= n <= 2
+            and i % 1024 == 0 and i <= 4096 and o % 2 == 0
+            and _FP8_WEIGHTS):
+        ent = _FP8_WEIGHTS.get(weight.data_ptr())
+        if ent is not None and tuple(ent[0].shape) != tuple(weight.shape):
+            ent = None
+        if ent is not None:
+            C = _require_native("skinny_gemm_fp8_norm")
+            res2 = (res.reshape(n, i).contiguous() if res is not None
+                    else torch.empty(0, dtype=x.dtype, device=x.device))
+            x2, y = C.skinny_gemm_fp8_norm(
+                x.reshape(n, i).contiguous(), res2, norm_w, eps,
+                ent[0], ent[1], res is not None)
+            return x2.view(*lead, i), y.view(*lead, o)
+    if res is None:
+        return x, decode_linear(rmsnorm(x, norm_w, eps), weight)
+    x2, h = rmsnorm_res(x, res, norm_w, eps)
+    return x2, decode_linear(h, weight)
+
+
 def rmsnorm_res(x, res, weight, eps: float = 1e-5):
     """Fused (x + res, rmsnorm(x + res) * w) for the decode step (one
     kernel instead of add + norm; decode_fused.hip).  No autograd."""

@@ -62,6 +62,9 @@ void rope_kvwrite_launch(const void*, void*, void*, void*, const void*,
 void swiglu_fwd_launch(const void*, void*, long long, int, hipStream_t);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int,
                        hipStream_t);
+void skinny_gemm_fp8_norm_launch(const void*, const float*, const void*,
+                                 const void*, const void*, void*, float,
+                                 void*, int, int, int, hipStream_t);
 void attn_decode_qkv_launch(const void*, void*, void*, void*, const void*,
                             const void*, const int*, const int*,
                             const int*, int, int, int, int, float,
@@ -234,6 +237,30 @@ torch::Tensor attn_decode_qkv(torch::Tensor qkv, torch::Tensor Kc,
                          slot_ids.data_ptr<int>(), B, S_max, (int)Hq,
                          (int)Hkv, (float)scale, cur_stream());
   return O;
+}
+
+std::vector<torch::Tensor> skinny_gemm_fp8_norm(
+    torch::Tensor x, torch::Tensor res, torch::Tensor nw, double eps,
+    torch::Tensor W8, torch::Tensor scale, bool want_xout) {
+  // norm-fused fp8 GEMV (skinny_gemm.hip): y = rmsnorm(x+res)*nw @ W8^T
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  const int N = (int)x.size(0), I = (int)x.size(1);
+  const int O = (int)W8.size(0);
+  TORCH_CHECK(I <= 4096 && I % 1024 == 0,
+              "fp8_norm: I must be <= 4096 and 1024-aligned");
+  TORCH_CHECK(N >= 1 && N <= 2, "fp8_norm: N <= 2");
+  TORCH_CHECK(O % 2 == 0, "fp8_norm: O must be even");
+  const bool has_res = res.defined() && res.numel() > 0;
+  auto y = torch::empty({N, O}, x.options());
+  torch::Tensor xout;
+  if (want_xout) xout = torch::empty_like(x);
+  skinny_gemm_fp8_norm_launch(
+      W8.data_ptr(), scale.data_ptr<float>(), x.data_ptr(),
+      has_res ? res.data_ptr() : nullptr, nw.data_ptr(),
+      want_xout ? xout.data_ptr() : nullptr, (float)eps, y.data_ptr(),
+      N, I, O, cur_stream());
+  if (!want_xout) xout = x;
+  return {xout, y};
 }
 
 void adamw_step_mt(torch::Tensor ptrs, torch::Tensor wd_arr,
@@ -471,6 +498,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_kvwrite", &rope_kvwrite);
   m.def("decode_advance", &decode_advance);
   m.def("attn_decode_qkv", &attn_decode_qkv);
+  m.def("skinny_gemm_fp8_norm", &skinny_gemm_fp8_norm);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("mfma_probe", &mfma_probe);

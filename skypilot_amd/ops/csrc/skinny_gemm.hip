@@ -403,3 +403,120 @@ extern "C" void skinny_gemm_fp8_launch(const void* W8, const float* scale,
   }
 #undef F8CASE
 }
+
+// Norm-fused fp8 GEMV: rmsnorm(x [+ res]) * nw computed ONCE per block
+// into LDS, then the standard multi-row weight-streaming loops read
+// the normed input from LDS.  Deletes the separate rmsnorm_res launch
+// before the qkv / gate_up GEMVs — decode is dispatch-gap-bound (~5 us
+// idle PER KERNEL inside graph replay, profiles/
+// r02_fp8_decode_kernel_stats.txt), so one fewer launch saves kernel
+// time AND gap.  Unlike the silu-in-GEMV negative result (per-output-
+// row recompute), the LDS staging makes the norm a per-BLOCK prologue
+// that all blocks run in parallel (~2 us wall).  Block 0 additionally
+// writes x+res back for the residual stream.  I <= 4096 (h of the
+// norm-fed projections; N <= 2 fp8 rows fit 16 KB LDS).
+template <int N, int R>
+__global__ __launch_bounds__(256) void skinny_gemm_fp8_norm_kernel(
+    const unsigned char* __restrict__ W8, const float* __restrict__ scale,
+    const unsigned short* __restrict__ X,    // [N, I] raw x
+    const unsigned short* __restrict__ RES,  // [N, I] residual or null
+    const unsigned short* __restrict__ NW,   // [I] norm weight
+    unsigned short* __restrict__ XOUT,       // [N, I] x+res or null
+    float eps, unsigned short* __restrict__ Y, int I, int O) {
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  __shared__ unsigned short xn[N][4096];
+  __shared__ float rred[4];
+  // ---- prologue: t = x (+ res); rms over f32 t; xn = t*inv*nw ------
+#pragma unroll
+  for (int b = 0; b < N; ++b) {
+    float ss = 0.f;
+    for (int j = threadIdx.x; j < I; j += 256) {
+      float t = bf2f(X[(long long)b * I + j]);
+      if (RES) t += bf2f(RES[(long long)b * I + j]);
+      xn[b][j] = f2bf(t);
+      ss += t * t;
+      if (XOUT && blockIdx.x == 0)
+        XOUT[(long long)b * I + j] = f2bf(t);
+    }
+    ss = wave_reduce_sum(ss);
+    if (lane == 0) rred[wave] = ss;
+    __syncthreads();
+    float inv = rsqrtf((rred[0] + rred[1] + rred[2] + rred[3]) / I + eps);
+    __syncthreads();  // rred reused next b; xn[b] writes drained
+    for (int j = threadIdx.x; j < I; j += 256) {
+      float t = bf2f(xn[b][j]);
+      xn[b][j] = f2bf(t * inv * bf2f(NW[j]));
+    }
+  }
+  __syncthreads();
+  // ---- standard multi-row fp8 weight stream over LDS-resident xn ---
+  int o0 = (blockIdx.x * 4 + wave) * R;
+  if (o0 >= O) return;
+  const unsigned char* wrow = W8 + (long long)o0 * I;
+  float acc[R][N];
+#pragma unroll
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int b = 0; b < N; ++b) acc[r][b] = 0.f;
+  int i = lane * 16;
+  for (; i + 1024 + 16 <= I; i += 2048) {
+    fp8x16 w[R][2];
+#pragma unroll
+    for (int r = 0; r < R; ++r) {
+      w[r][0] = *(fp8x16*)&(((const i32x4*)(wrow + (long long)r * I + i))[0]);
+      w[r][1] =
+          *(fp8x16*)&(((const i32x4*)(wrow + (long long)r * I + i + 1024))[0]);
+    }
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      const unsigned short* xb = &xn[b][i];
+#pragma unroll
+      for (int r = 0; r < R; ++r) {
+        dot_fp8_16(w[r][0].u, xb, &acc[r][b]);
+        dot_fp8_16(w[r][1].u, xb + 1024, &acc[r][b]);
+      }
+    }
+  }
+  for (; i < I; i += 1024) {
+    fp8x16 w[R];
+#pragma unroll
+    for (int r = 0; r < R; ++r)
+      w[r].i = __builtin_nontemporal_load(
+          (const i32x4*)(wrow + (long long)r * I + i));
+#pragma unroll
+    for (int b = 0; b < N; ++b)
+#pragma unroll
+      for (int r = 0; r < R; ++r) dot_fp8_16(w[r].u, &xn[b][i], &acc[r][b]);
+  }
+#pragma unroll
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int b = 0; b < N; ++b) {
+      float v = wave_reduce_sum(acc[r][b]);
+      if (lane == 0 && o0 + r < O)
+        Y[(long long)b * O + o0 + r] = f2bf(v * scale[o0 + r]);
+    }
+}
+
+extern "C" void skinny_gemm_fp8_norm_launch(
+    const void* W8, const float* scale, const void* X, const void* RES,
+    const void* NW, void* XOUT, float eps, void* Y, int N, int I, int O,
+    hipStream_t stream) {
+  dim3 gridm((O / 2 + 3) / 4), block(256);
+#define F8NORM(n)                                                        \
+  case n:                                                                \
+    hipLaunchKernelGGL((skinny_gemm_fp8_norm_kernel<n, 2>), gridm,       \
+                       block, 0, stream, (const unsigned char*)W8,       \
+                       scale, (const unsigned short*)X,                  \
+                       (const unsigned short*)RES,                       \
+                       (const unsigned short*)NW,                        \
+                       (unsigned short*)XOUT, eps, (unsigned short*)Y,   \
+                       I, O);                                            \
+    break;
+  switch (N) {
+    F8NORM(1) F8NORM(2)
+    default:
+      break;
+  }
+#undef F8NORM
+}

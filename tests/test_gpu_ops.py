@@ -506,3 +506,35 @@ def test_attn_decode_qkv_fused():
     # cache rows written identically (incl. the roped k)
     assert torch.equal(kc_f, kc_ref)
     assert torch.equal(vc_f, vc_ref)
+
+
+@pytest.mark.gpu
+def test_decode_norm_linear_fused():
+    """Norm-fused fp8 GEMV vs the two-kernel chain (rmsnorm_res +
+    fp8 GEMV) on the SAME quantized weights — y within bf16/fp8 noise,
+    x2 (the residual stream) bit-equal."""
+    torch.manual_seed(17)
+    from skypilot_amd.ops import (_FP8_WEIGHTS, decode_norm_linear,
+                                  register_fp8_weight)
+    I, O = 4096, 6144
+    w = (torch.randn(O, I, device="cuda") * 0.02).bfloat16()
+    register_fp8_weight(w)
+    try:
+        nw = torch.randn(I, device="cuda").bfloat16().abs() + 0.5
+        for n, with_res in [(1, True), (2, True), (1, False)]:
+            x = (torch.randn(n, I, device="cuda") * 0.5).bfloat16()
+            res = ((torch.randn(n, I, device="cuda") * 0.5).bfloat16()
+                   if with_res else None)
+            x2, y = decode_norm_linear(x, res, nw, 1e-5, w)
+            # reference chain on the same fp8 table
+            if with_res:
+                x2_ref, h = ops.rmsnorm_res(x, res, nw, 1e-5)
+            else:
+                x2_ref, h = x, ops.rmsnorm(x, nw, 1e-5)
+            y_ref = ops.decode_linear(h, w)
+            torch.cuda.synchronize()
+            assert rel_err(y, y_ref) < 2e-2, (n, with_res)
+            assert torch.allclose(x2.float(), x2_ref.float(),
+                                  atol=1e-2, rtol=1e-2)
+    finally:
+        _FP8_WEIGHTS.pop(w.data_ptr(), None)
