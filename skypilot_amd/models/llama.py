@@ -100,6 +100,7 @@ class Attention(nn.Module):
             cache = infer_ctx.cache
             if qkv is None:
                 qkv = ops.decode_linear(x.reshape(B, -1), self._wqkv)
+            qkv = qkv.reshape(B, -1)  # [B,1,width] from norm-fused GEMV
             # fused rope + cache-write + attention: one kernel instead
             # of rope_kvwrite + attn_decode (the current token's k/v is
             # attended from registers; the cache row is written for
