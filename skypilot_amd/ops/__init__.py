@@ -16,6 +16,7 @@ from __future__ import annotations
 
 import importlib.util
 import math
+import os
 from pathlib import Path
 
 import torch
@@ -404,9 +405,16 @@ def decode_norm_linear(x, res, norm_w, eps, weight):
         n *= d
     i = x.shape[-1]
     o = weight.shape[0]
-    if (x.is_cuda and x.dtype == torch.bfloat16 and 1 <= n <= 2
-            and i % 1024 == 0 and i <= 4096 and o % 2 == 0
-            and _FP8_WEIGHTS):
+    # MEASURED NEGATIVE as the default path (371 -> 246 tok/s @1):
+    # the dot loop's x reads moved from global (4-deep vm pipeline,
+    # latency hidden under the nontemporal weight stream) to flat-LDS
+    # loads, which serialize on lgkmcnt and stall the stream.  The
+    # launch+gap saving (~10 us) never amortized the per-block slowdown.
+    # Kept as a tested lib op behind SKY_FP8_NORM_FUSED=1.
+    fused_on = os.environ.get("SKY_FP8_NORM_FUSED", "0") == "1"
+    if (fused_on and x.is_cuda and x.dtype == torch.bfloat16
+            and 1 <= n <= 2 and i % 1024 == 0 and i <= 4096
+            and o % 2 == 0 and _FP8_WEIGHTS):
         ent = _FP8_WEIGHTS.get(weight.data_ptr())
         if ent is not None and tuple(ent[0].shape) != tuple(weight.shape):
             ent = None
