@@ -253,6 +253,18 @@ def create_app(cluster_dir: str, gpu_ids: list[int],
     @app.post("/exec/start")
     def exec_start(body: dict = None):
         body = body or {}
+        # cap live sessions (abandoned PTYs hold fds until /close or
+        # cluster teardown); reap any whose shell already exited first
+        for sid_ in list(exec_sessions):
+            se = exec_sessions[sid_]
+            try:
+                os.kill(se["pid"], 0)
+            except ProcessLookupError:
+                _reap_session(sid_)
+        if len(exec_sessions) >= 32:
+            from fastapi.responses import JSONResponse
+            return JSONResponse({"detail": "too many exec sessions"},
+                                status_code=429)
         cmd = body.get("cmd") or ["/bin/bash", "-i"]
         if isinstance(cmd, str):
             cmd = ["/bin/bash", "-lc", cmd]
