@@ -256,6 +256,22 @@ def api_status():
     click.echo("healthy" if sdk.api_healthy() else "not running")
 
 
+@api.command("logs")
+@click.option("--tail", "-n", type=int, default=100)
+def api_logs(tail):
+    """Show the local API server log (reference: sky api logs)."""
+    import os
+    path = os.path.expanduser("~/.sky_amd_api.log")
+    if not os.path.exists(path):
+        click.echo("no server log at ~/.sky_amd_api.log "
+                   "(server not auto-started from this client?)")
+        return
+    with open(path, "rb") as f:
+        data = f.read().decode(errors="replace")
+    for line in data.splitlines()[-tail:]:
+        click.echo(line)
+
+
 # ---- jobs -----------------------------------------------------------------
 @cli.group()
 def jobs():
