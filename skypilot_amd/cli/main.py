@@ -61,11 +61,18 @@ def cli():
               help="don't stream logs after submit")
 @click.option("--async", "async_", is_flag=True,
               help="print request id and return")
+@click.option("--dryrun", is_flag=True,
+              help="optimize + print the placement plan, provision nothing")
 def launch(entrypoint, cluster, gpus, num_nodes, env, down,
-           idle_minutes_to_autostop, retry_until_up, detach_run, async_):
+           idle_minutes_to_autostop, retry_until_up, detach_run, async_,
+           dryrun):
     """Launch a task (provision + sync + setup + run)."""
     task = _load_task(entrypoint, env,
                       {"accelerators": gpus, "_num_nodes": num_nodes})
+    if dryrun:
+        plan = sdk.get(sdk.launch(task, cluster, dryrun=True))
+        _print_result(plan["handle"])
+        return
     rid = sdk.launch(task, cluster, down=down, retry_until_up=retry_until_up,
                      idle_minutes_to_autostop=idle_minutes_to_autostop)
     if async_:

@@ -260,12 +260,14 @@ def _task_body(task) -> Dict[str, Any]:
 # ---- public API (mirrors reference sdk surface) ---------------------------
 def launch(task, cluster_name: Optional[str] = None, *, down: bool = False,
            idle_minutes_to_autostop: Optional[int] = None,
-           detach_run: bool = True, retry_until_up: bool = False) -> str:
+           detach_run: bool = True, retry_until_up: bool = False,
+           dryrun: bool = False) -> str:
     return _submit("launch", {
         "task": _task_body(task), "cluster_name": cluster_name,
         "down": down,
         "idle_minutes_to_autostop": idle_minutes_to_autostop,
-        "detach_run": detach_run, "retry_until_up": retry_until_up})
+        "detach_run": detach_run, "retry_until_up": retry_until_up,
+        "dryrun": dryrun})
 
 
 def exec(task, cluster_name: str, *, detach_run: bool = True) -> str:  # noqa: A001

@@ -133,8 +133,23 @@ def launch(task, cluster_name: Optional[str] = None, *,
            detach_run: bool = True, down: bool = False,
            idle_minutes_to_autostop: Optional[int] = None,
            managed_job_id: Optional[int] = None,
-           retry_until_up: bool = False):
-    """reference: sky/execution.py:688 (launch)."""
+           retry_until_up: bool = False, dryrun: bool = False):
+    """reference: sky/execution.py:688 (launch; --dryrun stops after
+    OPTIMIZE and returns the placement plan)."""
+    if dryrun:
+        dag = to_dag(task)
+        Optimizer.optimize(dag)
+        t = dag.tasks[0]
+        return None, {
+            "dryrun": True,
+            "cluster_name": cluster_name,
+            "resources": t.resources.to_yaml_config(),
+            "estimated_hourly_cost": getattr(
+                t, "estimated_hourly_cost", None),
+            "candidates": [c.to_yaml_config()
+                           for c in (t.resources.candidates
+                                     or (t.resources,))],
+        }
     return _execute(task, cluster_name, ALL_STAGES, detach_run, down,
                     idle_minutes_to_autostop, managed_job_id,
                     retry_until_up)

@@ -16,15 +16,15 @@ from skypilot_amd.task import Task
 @register("launch", LONG)
 def launch(task: Dict[str, Any], cluster_name: Optional[str] = None,
            down: bool = False, idle_minutes_to_autostop: Optional[int] = None,
-           detach_run: bool = True,
-           retry_until_up: bool = False) -> Dict[str, Any]:
+           detach_run: bool = True, retry_until_up: bool = False,
+           dryrun: bool = False) -> Dict[str, Any]:
     from skypilot_amd import admin_policy
     task = admin_policy.apply(task, cluster_name, "launch")
     t = Task.from_yaml_config(task)
     job_id, handle = execution.launch(
         t, cluster_name, detach_run=detach_run, down=down,
         idle_minutes_to_autostop=idle_minutes_to_autostop,
-        retry_until_up=retry_until_up)
+        retry_until_up=retry_until_up, dryrun=dryrun)
     return {"job_id": job_id, "cluster_name": cluster_name,
             "handle": handle}
 

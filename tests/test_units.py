@@ -425,3 +425,28 @@ def test_request_gc_retention(monkeypatch, tmp_path):
         c.execute("UPDATE requests SET created_at=? WHERE request_id=?",
                   (_time.time() - 30 * 86400, fresh))
     assert rdb.gc_requests(max_age_days=7, keep_latest=5) == 0
+
+
+def test_launch_dryrun_returns_plan(monkeypatch, tmp_path):
+    """--dryrun optimizes and returns the placement plan without
+    provisioning anything (reference: sky launch --dryrun)."""
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    monkeypatch.setenv("SKY_AMD_FAKE_GPUS", "8")
+    from skypilot_amd.utils import gpu_topology
+    gpu_topology.detect_gpus.cache_clear()
+    from skypilot_amd import execution, global_state
+    from skypilot_amd.task import Task
+    t = Task.from_yaml_config({
+        "run": "true",
+        "resources": {"any_of": [
+            {"accelerators": "MI355X:8"},
+            {"accelerators": "MI355X:2", "use_spot": True}]}})
+    job_id, plan = execution.launch(t, "dry-c", dryrun=True)
+    assert job_id is None
+    assert plan["dryrun"] is True
+    assert plan["resources"]["use_spot"] is True  # cheapest ranked first
+    assert plan["estimated_hourly_cost"] is not None
+    assert len(plan["candidates"]) == 2
+    # nothing was provisioned
+    assert not any(c["name"] == "dry-c"
+                   for c in global_state.list_clusters())
