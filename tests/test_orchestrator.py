@@ -411,11 +411,18 @@ def test_ssh_tunnel_interactive_shell(sky_env, client):
         if b"tun-42" in out:
             break
     assert b"tun-42" in out, out[-500:]
-    # status + close
+    # status, resize (PTY mode only — pipes report ok: False), close
     st = client.get(f"/api/v1/ssh/ssh-c/{sid}/status")
     assert st.status_code == 200
+    rz = client.post(f"/api/v1/ssh/ssh-c/{sid}/resize",
+                     json={"rows": 40, "cols": 120})
+    assert rz.status_code == 200
     assert client.post(f"/api/v1/ssh/ssh-c/{sid}/close"
                        ).json()["ok"]
+    # closed session: stdin reports not-ok, stdout drains empty
+    r = client.post(f"/api/v1/ssh/ssh-c/{sid}/stdin", content=b"late")
+    assert r.json()["ok"] is False
+    assert b"".join(sdk.ssh_stdout("ssh-c", sid)) == b""
     # non-owner denied: cluster owner is the admin/server identity, and
     # 'intruder' is a plain user
     r = client.post("/api/v1/ssh/ssh-c/start", json={},
