@@ -51,6 +51,29 @@ def test_concurrent_requests_batch(engine):
         assert results[i] == _greedy_no_cache(engine.model, p, 6), i
 
 
+def test_mixed_greedy_and_sampled_batch(engine):
+    """Concurrent greedy + temperature>0 requests share decode batches:
+    the greedy rows must still match the no-cache reference while the
+    sampled row completes (exercises the mixed per-row sampling branch
+    in _decode_step)."""
+    results = {}
+
+    def worker(i, p, temp):
+        results[i] = engine.generate(p, max_tokens=5, temperature=temp)
+
+    specs = [([3, 5, 7], 0.0), ([11, 13], 0.9), ([2, 4, 6, 8], 0.0)]
+    threads = [threading.Thread(target=worker, args=(i, p, t))
+               for i, (p, t) in enumerate(specs)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(60)
+    assert len(results) == 3
+    assert results[0] == _greedy_no_cache(engine.model, [3, 5, 7], 5)
+    assert results[2] == _greedy_no_cache(engine.model, [2, 4, 6, 8], 5)
+    assert len(results[1]) == 5  # sampled row completed
+
+
 def test_openai_api_surface(engine):
     from fastapi.testclient import TestClient
     from skypilot_amd.serve.entrypoint import create_app
