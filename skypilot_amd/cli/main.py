@@ -208,8 +208,18 @@ def queue(cluster):
 @click.argument("cluster")
 @click.argument("job_id", type=int, required=False)
 @click.option("--no-follow", is_flag=True)
-def logs(cluster, job_id, no_follow):
+@click.option("--status", "status_only", is_flag=True,
+              help="print the job's status instead of its logs "
+                   "(exit 0 iff SUCCEEDED; reference: sky logs --status)")
+def logs(cluster, job_id, no_follow, status_only):
     """Stream a job's logs."""
+    if status_only:
+        j = sdk.get(sdk.job_status(cluster, job_id if job_id else 1))
+        if j is None:
+            click.echo("no such job")
+            raise SystemExit(1)
+        click.echo(j["status"])
+        raise SystemExit(0 if j["status"] == "SUCCEEDED" else 1)
     sdk.tail_logs(cluster, job_id, follow=not no_follow)
 
 

@@ -88,3 +88,25 @@ def test_cli_users_cost_group_storage(client, tmp_path):
 
     res = r.invoke(cli, ["storage", "ls"])
     assert res.exit_code == 0
+
+
+def test_logs_status_flag(client):
+    """`sky logs --status` prints the job status and exits 0 iff
+    SUCCEEDED (reference: sky logs --status)."""
+    from click.testing import CliRunner
+    from skypilot_amd.cli.main import cli
+    runner = CliRunner()
+    r = runner.invoke(cli, ["launch", "echo ok", "-c", "st-c",
+                            "--detach-run"])
+    assert r.exit_code == 0, r.output
+    import time
+    from skypilot_amd.client import sdk
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        j = sdk.get(sdk.job_status("st-c", 1))
+        if j and j["status"] in ("SUCCEEDED", "FAILED"):
+            break
+        time.sleep(0.3)
+    r = runner.invoke(cli, ["logs", "st-c", "1", "--status"])
+    assert r.exit_code == 0 and "SUCCEEDED" in r.output, r.output
+    runner.invoke(cli, ["down", "st-c", "-y"])
