@@ -182,10 +182,19 @@ def down(clusters, yes):
 
 @cli.command()
 @click.argument("cluster")
-@click.option("--idle-minutes", "-i", type=int, required=True)
+@click.option("--idle-minutes", "-i", type=int, default=None)
 @click.option("--down", is_flag=True)
-def autostop(cluster, idle_minutes, down):
-    """Stop the cluster after N idle minutes (-i)."""
+@click.option("--cancel", is_flag=True,
+              help="cancel a scheduled autostop (reference: sky "
+                   "autostop --cancel)")
+def autostop(cluster, idle_minutes, down, cancel):
+    """Stop the cluster after N idle minutes (-i), or --cancel."""
+    if cancel:
+        sdk.get(sdk.autostop(cluster, -1, False))
+        click.echo(f"Autostop cancelled on {cluster}.")
+        return
+    if idle_minutes is None:
+        raise click.UsageError("pass -i <minutes> or --cancel")
     sdk.get(sdk.autostop(cluster, idle_minutes, down))
     click.echo(f"Autostop set on {cluster}: {idle_minutes}m "
                f"({'down' if down else 'stop'})")

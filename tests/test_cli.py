@@ -110,3 +110,27 @@ def test_logs_status_flag(client):
     r = runner.invoke(cli, ["logs", "st-c", "1", "--status"])
     assert r.exit_code == 0 and "SUCCEEDED" in r.output, r.output
     runner.invoke(cli, ["down", "st-c", "-y"])
+
+
+def test_autostop_cancel(client, tmp_path):
+    """`sky autostop --cancel` clears a scheduled autostop."""
+    from skypilot_amd.cli.main import cli
+    r = CliRunner()
+    y = tmp_path / "a.yaml"
+    y.write_text("run: 'true'\nresources: {cpus: 1}\n")
+    assert r.invoke(cli, ["launch", str(y), "-c", "as-c",
+                          "--detach-run"]).exit_code == 0
+    assert r.invoke(cli, ["autostop", "as-c", "-i", "15"]).exit_code == 0
+    from skypilot_amd.client import sdk
+    recs = sdk.get(sdk.status())
+    h = next(x for x in recs if x["name"] == "as-c")["handle"]
+    from skypilot_amd.agent.client import AgentClient
+    a = AgentClient(h["agent_port"], token=h.get("agent_token"))
+    try:
+        assert a.is_autostopping()["idle_minutes"] == 15
+        assert r.invoke(cli, ["autostop", "as-c", "--cancel"]
+                        ).exit_code == 0
+        assert a.is_autostopping()["idle_minutes"] == -1
+    finally:
+        a.close()
+    r.invoke(cli, ["down", "as-c", "-y"])
