@@ -162,19 +162,38 @@ def start(clusters):
 
 
 @cli.command()
-@click.argument("clusters", nargs=-1, required=True)
-def stop(clusters):
+@click.argument("clusters", nargs=-1)
+@click.option("--all", "-a", "all_", is_flag=True)
+def stop(clusters, all_):
     """Stop a cluster (keep its state dir for sky start)."""
+    if all_:
+        clusters = [r["name"] for r in sdk.get(sdk.status())
+                    if r["status"] == "UP"]
+    if not clusters:
+        click.echo("No clusters." if all_ else
+                   "Pass cluster names or --all.")
+        return
     for cl in clusters:
         sdk.get(sdk.stop(cl))
         click.echo(f"Cluster {cl} stopped.")
 
 
 @cli.command()
-@click.argument("clusters", nargs=-1, required=True)
+@click.argument("clusters", nargs=-1)
+@click.option("--all", "-a", "all_", is_flag=True,
+              help="tear down every cluster in the workspace")
 @click.option("--yes", "-y", is_flag=True)
-def down(clusters, yes):
+def down(clusters, all_, yes):
     """Tear the cluster down and release its GPUs."""
+    if all_:
+        clusters = [r["name"] for r in sdk.get(sdk.status())]
+    if not clusters:
+        click.echo("No clusters." if all_ else
+                   "Pass cluster names or --all.")
+        return
+    if not yes and not click.confirm(
+            f"Tear down {', '.join(clusters)}?"):
+        return
     for cl in clusters:
         sdk.get(sdk.down(cl))
         click.echo(f"Cluster {cl} terminated.")

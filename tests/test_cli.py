@@ -134,3 +134,20 @@ def test_autostop_cancel(client, tmp_path):
     finally:
         a.close()
     r.invoke(cli, ["down", "as-c", "-y"])
+
+
+def test_down_all(client, tmp_path):
+    """`sky down --all -y` tears down every workspace cluster."""
+    from skypilot_amd.cli.main import cli
+    from skypilot_amd.client import sdk
+    r = CliRunner()
+    y = tmp_path / "b.yaml"
+    y.write_text("run: 'true'\nresources: {cpus: 1}\n")
+    for c in ("da-1", "da-2"):
+        assert r.invoke(cli, ["launch", str(y), "-c", c,
+                              "--detach-run"]).exit_code == 0
+    res = r.invoke(cli, ["down", "--all", "-y"])
+    assert res.exit_code == 0, res.output
+    assert "da-1 terminated" in res.output
+    assert "da-2 terminated" in res.output
+    assert not sdk.get(sdk.status())
