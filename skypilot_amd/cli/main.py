@@ -191,7 +191,8 @@ def down(clusters, all_, yes):
         click.echo("No clusters." if all_ else
                    "Pass cluster names or --all.")
         return
-    if not yes and not click.confirm(
+    # only mass teardown prompts; single-name down stays scriptable
+    if all_ and not yes and not click.confirm(
             f"Tear down {', '.join(clusters)}?"):
         return
     for cl in clusters:
