@@ -130,11 +130,14 @@ def exec_cmd(cluster, entrypoint, env, gpus, async_):
 
 
 @cli.command()
+@click.argument("clusters", nargs=-1)
 @click.option("--refresh", "-r", is_flag=True)
 @click.option("--all-workspaces", "-u", is_flag=True)
-def status(refresh, all_workspaces):
-    """Show clusters (active workspace; -u for every workspace)."""
-    records = sdk.get(sdk.status(refresh=refresh,
+def status(clusters, refresh, all_workspaces):
+    """Show clusters (optionally filtered by name; -u for every
+    workspace)."""
+    records = sdk.get(sdk.status(cluster_names=list(clusters) or None,
+                                 refresh=refresh,
                                  all_workspaces=all_workspaces))
     if not records:
         click.echo("No existing clusters.")

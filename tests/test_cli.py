@@ -151,3 +151,17 @@ def test_down_all(client, tmp_path):
     assert "da-1 terminated" in res.output
     assert "da-2 terminated" in res.output
     assert not sdk.get(sdk.status())
+
+
+def test_status_name_filter(client, tmp_path):
+    from skypilot_amd.cli.main import cli
+    r = CliRunner()
+    y = tmp_path / "c.yaml"
+    y.write_text("run: 'true'\nresources: {cpus: 1}\n")
+    for c in ("sf-1", "sf-2"):
+        assert r.invoke(cli, ["launch", str(y), "-c", c,
+                              "--detach-run"]).exit_code == 0
+    res = r.invoke(cli, ["status", "sf-1"])
+    assert res.exit_code == 0, res.output
+    assert "sf-1" in res.output and "sf-2" not in res.output
+    r.invoke(cli, ["down", "--all", "-y"])
