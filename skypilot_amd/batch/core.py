@@ -12,7 +12,7 @@ import json
 import math
 import time
 from pathlib import Path
-from typing import Any, Callable, Dict, Iterable, List, Optional
+from typing import Any, Dict, Iterable, List, Optional
 
 from skypilot_amd import execution, global_state
 from skypilot_amd.backends.pool_backend import PoolBackend

@@ -6,8 +6,6 @@ Run as `python -m skypilot_amd.cli ...` or via the repo's `bin/sky`.
 from __future__ import annotations
 
 import json
-import sys
-from typing import Optional
 
 import click
 

@@ -4,9 +4,6 @@ spawned on the API-server host (reference: utils/controller_utils.py:1422)
 — and recurse into execution.launch for the user's cluster."""
 from __future__ import annotations
 
-import os
-import subprocess
-import sys
 from typing import Any, Dict, List, Optional
 
 from skypilot_amd.jobs import state

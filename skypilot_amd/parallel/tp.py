@@ -13,7 +13,6 @@ SURVEY.md §2.11); this is the bundled-entrypoint strategy layer.
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -16,11 +16,9 @@ Config (~/.sky_amd/config.yaml):
 from __future__ import annotations
 
 import json
-import shlex
 import socket
 import subprocess
 import time
-from pathlib import Path
 from typing import Any, Dict, List, Optional
 
 import yaml

@@ -15,7 +15,6 @@ import signal
 import socket
 import subprocess
 import sys
-import time
 from pathlib import Path
 from typing import Any, Dict, List, Optional
 

@@ -3,7 +3,7 @@ cache sized against 288 GB HBM3E per MI355X (no reference counterpart;
 SkyPilot delegates serving to user images)."""
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 
