@@ -32,10 +32,10 @@ MIN_CLIENT_API_VERSION = 1
 
 
 def create_app(start_workers: bool = True) -> FastAPI:
-    app = FastAPI(title="skypilot-amd API server")
+    from contextlib import asynccontextmanager
 
-    @app.on_event("startup")
-    def _startup():
+    @asynccontextmanager
+    async def _lifespan(app_):
         # the server's own OS identity bootstraps as admin
         users.ensure_user(global_state.current_user())
         if start_workers:
@@ -43,6 +43,9 @@ def create_app(start_workers: bool = True) -> FastAPI:
         # serve/jobs background refreshers (reference: server/daemons.py)
         from skypilot_amd.server import daemons
         daemons.start()
+        yield
+
+    app = FastAPI(title="skypilot-amd API server", lifespan=_lifespan)
 
     @app.get("/health")
     def health():
