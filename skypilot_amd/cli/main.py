@@ -409,9 +409,15 @@ def jobs_group_down_cmd(name):
 @click.argument("entrypoint")
 @click.option("--pool", "-p", "name", required=True)
 @click.option("--workers", type=int, default=2)
-def jobs_pool_apply(entrypoint, name, workers):
+@click.option("--min-workers", type=int, default=None,
+              help="autoscaling floor (queue-length autoscaler)")
+@click.option("--max-workers", type=int, default=None,
+              help="autoscaling ceiling")
+def jobs_pool_apply(entrypoint, name, workers, min_workers, max_workers):
     task = _load_task(entrypoint, (), {})
-    _print_result(sdk.get(sdk.jobs_pool_apply(name, task, workers)))
+    _print_result(sdk.get(sdk.jobs_pool_apply(
+        name, task, workers, min_workers=min_workers,
+        max_workers=max_workers)))
 
 
 @jobs_pool.command("status")
