@@ -457,10 +457,26 @@ def serve_update(entrypoint, service_name):
 
 
 @serve.command("down")
-@click.argument("service_name")
-def serve_down(service_name):
-    sdk.get(sdk.serve_down(service_name))
-    click.echo(f"Service {service_name} torn down.")
+@click.argument("service_name", required=False)
+@click.option("--all", "-a", "all_", is_flag=True,
+              help="tear down every service")
+@click.option("--yes", "-y", is_flag=True)
+def serve_down(service_name, all_, yes):
+    if all_:
+        names = [sv["name"] for sv in sdk.get(sdk.serve_status(None))]
+        if not names:
+            click.echo("No services.")
+            return
+        if not yes and not click.confirm(
+                f"Tear down {', '.join(names)}?"):
+            return
+    elif service_name:
+        names = [service_name]
+    else:
+        raise click.UsageError("pass a service name or --all")
+    for n in names:
+        sdk.get(sdk.serve_down(n))
+        click.echo(f"Service {n} torn down.")
 
 
 @serve.command("logs")
