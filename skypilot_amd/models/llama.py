@@ -39,6 +39,8 @@ class LlamaConfig:
     norm_eps: float = 1e-5
     max_seq_len: int = 8192
     tie_embeddings: bool = False
+    # Qwen2-style families put biases on the q/k/v projections only
+    attn_bias: bool = False
 
 
 CONFIGS = {
@@ -46,6 +48,19 @@ CONFIGS = {
     "llama3-70b": LlamaConfig(
         name="llama3-70b", hidden_size=8192, intermediate_size=28672,
         num_layers=80, num_heads=64, num_kv_heads=8),
+    # Qwen2 dense family (same RMSNorm+RoPE+GQA+SwiGLU skeleton with
+    # q/k/v biases and a 1e6 rope base; reference architecture is
+    # public — weights are random-init here, no network).
+    "qwen2-7b": LlamaConfig(
+        name="qwen2-7b", vocab_size=152064, hidden_size=3584,
+        intermediate_size=18944, num_layers=28, num_heads=28,
+        num_kv_heads=4, rope_theta=1000000.0, norm_eps=1e-6,
+        max_seq_len=32768, attn_bias=True),
+    "qwen2-debug": LlamaConfig(
+        name="qwen2-debug", vocab_size=512, hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=2,
+        num_kv_heads=1, head_dim=128, rope_theta=1000000.0,
+        max_seq_len=512, attn_bias=True),
     # Small configs for tests / smoke.
     "llama-debug": LlamaConfig(
         name="llama-debug", vocab_size=512, hidden_size=256,
@@ -73,12 +88,25 @@ class Attention(nn.Module):
         self.cfg = cfg
         h, d = cfg.hidden_size, cfg.head_dim
         self.n_q, self.n_kv = cfg.num_heads, cfg.num_kv_heads
-        self.wq = nn.Linear(h, self.n_q * d, bias=False)
-        self.wk = nn.Linear(h, self.n_kv * d, bias=False)
-        self.wv = nn.Linear(h, self.n_kv * d, bias=False)
+        self.wq = nn.Linear(h, self.n_q * d, bias=cfg.attn_bias)
+        self.wk = nn.Linear(h, self.n_kv * d, bias=cfg.attn_bias)
+        self.wv = nn.Linear(h, self.n_kv * d, bias=cfg.attn_bias)
         self.wo = nn.Linear(self.n_q * d, h, bias=False)
         self.scale = 1.0 / math.sqrt(d)
         self._wqkv = None  # lazy fused [q|k|v] weight for the decode GEMV
+        self._bqkv = None  # packed [q|k|v] bias (attn_bias families)
+
+    def packed_qkv(self):
+        """Lazily build the fused decode-GEMV weight (and bias)."""
+        if self._wqkv is None:
+            self._wqkv = torch.cat(
+                [self.wq.weight, self.wk.weight, self.wv.weight],
+                0).contiguous()
+            if self.wq.bias is not None:
+                self._bqkv = torch.cat(
+                    [self.wq.bias, self.wk.bias, self.wv.bias],
+                    0).contiguous()
+        return self._wqkv
 
     def forward(self, x, cos, sin, positions, infer_ctx=None,
                 qkv=None):
@@ -93,14 +121,13 @@ class Attention(nn.Module):
             # kernel (decode_fused.hip) — replaces 3 GEMVs + rope x2 +
             # scatter x2.  Decode GEMMs route via ops.decode_linear
             # (skinny GEMV vs hipBLASLt per measured thresholds).
-            if self._wqkv is None:
-                self._wqkv = torch.cat(
-                    [self.wq.weight, self.wk.weight, self.wv.weight],
-                    0).contiguous()
             cache = infer_ctx.cache
             if qkv is None:
-                qkv = ops.decode_linear(x.reshape(B, -1), self._wqkv)
+                qkv = ops.decode_linear(x.reshape(B, -1),
+                                        self.packed_qkv())
             qkv = qkv.reshape(B, -1)  # [B,1,width] from norm-fused GEMV
+            if self._bqkv is not None:
+                qkv = qkv + self._bqkv
             # fused rope + cache-write + attention: one kernel instead
             # of rope_kvwrite + attn_decode (the current token's k/v is
             # attended from registers; the cache row is written for
@@ -112,9 +139,9 @@ class Attention(nn.Module):
             return ops.decode_linear(o.reshape(B, self.n_q * d),
                                      self.wo.weight).view(B, S, -1)
         lin = torch.nn.functional.linear
-        q = lin(x, self.wq.weight).view(B, S, self.n_q, d)
-        k = lin(x, self.wk.weight).view(B, S, self.n_kv, d)
-        v = lin(x, self.wv.weight).view(B, S, self.n_kv, d)
+        q = lin(x, self.wq.weight, self.wq.bias).view(B, S, self.n_q, d)
+        k = lin(x, self.wk.weight, self.wk.bias).view(B, S, self.n_kv, d)
+        v = lin(x, self.wv.weight, self.wv.bias).view(B, S, self.n_kv, d)
         q = ops.rope(q.reshape(B * S, self.n_q, d), cos, sin,
                      positions).view(B, S, self.n_q, d)
         k = ops.rope(k.reshape(B * S, self.n_kv, d), cos, sin,
@@ -225,12 +252,8 @@ class Llama(nn.Module):
                     x, h2 = ops.rmsnorm_res(x, a, blk.mlp_norm, blk.eps)
                     res = blk.mlp(h2, infer_ctx)
                     continue
-                if blk.attn._wqkv is None:
-                    blk.attn._wqkv = torch.cat(
-                        [blk.attn.wq.weight, blk.attn.wk.weight,
-                         blk.attn.wv.weight], 0).contiguous()
                 x, qkv = ops.decode_norm_linear(
-                    x, res, blk.attn_norm, blk.eps, blk.attn._wqkv)
+                    x, res, blk.attn_norm, blk.eps, blk.attn.packed_qkv())
                 a = blk.attn(None, cos, sin, positions, infer_ctx,
                              qkv=qkv)
                 x, gu = ops.decode_norm_linear(
