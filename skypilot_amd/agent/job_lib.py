@@ -158,7 +158,7 @@ class JobTable:
         # (kill-then-mark would race the driver's own FAILED write).
         self.set_status(job_id, CANCELLED)
         pid = j.get("driver_pid")
-        if pid:
+        if pid and _proc_matches(pid, b"skypilot_amd.agent.driver"):
             try:
                 os.killpg(pid, signal.SIGTERM)
             except (ProcessLookupError, PermissionError):
@@ -176,11 +176,33 @@ class JobTable:
             pid = j.get("driver_pid")
             if pid and not _pid_alive(pid):
                 self.set_status(j["job_id"], FAILED_DRIVER, exit_code=-1)
+                tag = (f"SKYPILOT_INTERNAL_JOB_ID={j['job_id']}"
+                       .encode())
                 for npid in j["spec"].get("node_pids", []):
+                    # PID-recycling guard: only kill a recorded pgid if
+                    # the live process still carries THIS job's env tag
+                    # (a long-dead pid number can be reused by an
+                    # unrelated process — killing it blind took out
+                    # innocent processes in CI).
+                    if not _proc_matches(npid, tag):
+                        continue
                     try:
                         os.killpg(npid, signal.SIGTERM)
                     except (ProcessLookupError, PermissionError):
                         pass
+
+
+def _proc_matches(pid: int, needle: bytes) -> bool:
+    """True iff /proc/<pid>'s cmdline or environ contains `needle` —
+    guards every kill-by-recorded-pid against PID recycling."""
+    for f in ("cmdline", "environ"):
+        try:
+            with open(f"/proc/{pid}/{f}", "rb") as fh:
+                if needle in fh.read():
+                    return True
+        except OSError:
+            pass
+    return False
 
 
 def _pid_alive(pid: int) -> bool:

@@ -65,6 +65,38 @@ def _stall_loop():
         prev = now
 
 
+def _reap_orphan_agents(min_age_s: float = 120.0) -> int:
+    """Stop agent daemons whose cluster has NO record (a launch request
+    cancelled between agent spawn and state write leaves one behind).
+    Identity-checked: the pid is only killed if its cmdline still names
+    this exact cluster dir (PID-recycling guard)."""
+    import json as _json
+    import os
+    import signal
+    import time as _t
+    from skypilot_amd import global_state
+    reaped = 0
+    cdir = global_state.root_dir() / "clusters"
+    if not cdir.exists():
+        return 0
+    for meta in cdir.glob("*/agent.json"):
+        name = meta.parent.name
+        try:
+            if global_state.get_cluster(name) is not None:
+                continue
+            if _t.time() - meta.stat().st_mtime < min_age_s:
+                continue  # grace: provision may still be writing state
+            pid = int(_json.loads(meta.read_text())["pid"])
+            with open(f"/proc/{pid}/cmdline", "rb") as fh:
+                if str(meta.parent).encode() not in fh.read():
+                    continue
+            os.kill(pid, signal.SIGTERM)
+            reaped += 1
+        except (OSError, ValueError, KeyError):
+            continue
+    return reaped
+
+
 def _gc_controller_logs(max_age_days: float = 7.0,
                         max_total_mb: float = 512.0) -> int:
     """Controller-log retention (reference: sky/jobs/log_gc.py): drop
@@ -119,6 +151,7 @@ def _loop():
             from skypilot_amd.jobs import pools as jobs_pools
             jobs_pools.autoscale()
             _gc_controller_logs()
+            _reap_orphan_agents()
             from skypilot_amd.server import requests_db as _rdb
             _rdb.gc_requests()
         except Exception:  # noqa: BLE001

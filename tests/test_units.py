@@ -450,3 +450,44 @@ def test_launch_dryrun_returns_plan(monkeypatch, tmp_path):
     # nothing was provisioned
     assert not any(c["name"] == "dry-c"
                    for c in global_state.list_clusters())
+
+
+def test_reap_orphan_agents(monkeypatch, tmp_path):
+    """An agent whose cluster record vanished (launch cancelled between
+    agent spawn and state write) is identity-checked and stopped; live
+    recorded clusters and non-matching pids are untouched."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import time
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    from skypilot_amd.server.daemons import _reap_orphan_agents
+    cdir = tmp_path / "clusters" / "orphan-c"
+    cdir.mkdir(parents=True)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "skypilot_amd.agent.daemon",
+         "--cluster-dir", str(cdir), "--port", "0"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        (cdir / "agent.json").write_text(
+            json.dumps({"port": 0, "pid": proc.pid}))
+        # fresh file is inside the grace window: untouched
+        assert _reap_orphan_agents(min_age_s=120) == 0
+        assert proc.poll() is None
+        # age it out: reaped
+        old = time.time() - 600
+        os.utime(cdir / "agent.json", (old, old))
+        assert _reap_orphan_agents(min_age_s=120) == 1
+        deadline = time.time() + 10
+        while time.time() < deadline and proc.poll() is None:
+            time.sleep(0.2)
+        assert proc.poll() is not None
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+    # stale agent.json with a recycled (non-matching) pid: skipped
+    (cdir / "agent.json").write_text(
+        json.dumps({"port": 0, "pid": os.getpid()}))
+    os.utime(cdir / "agent.json", (old, old))
+    assert _reap_orphan_agents(min_age_s=120) == 0
