@@ -243,7 +243,7 @@ def _ensure_agent(cdir: Path, gpu_ids: List[int],
         return port
     port = _free_port()
     log = open(cdir / "agent.log", "ab")
-    subprocess.Popen(
+    proc = subprocess.Popen(
         [sys.executable, "-m", "skypilot_amd.agent.daemon",
          "--cluster-dir", str(cdir), "--port", str(port),
          "--gpu-ids", ",".join(str(g) for g in gpu_ids)],
@@ -251,6 +251,12 @@ def _ensure_agent(cdir: Path, gpu_ids: List[int],
         env={**os.environ, "SKY_AMD_HOME": str(global_state.root_dir()),
              "SKY_AMD_AGENT_TOKEN": tok})
     log.close()
+    # Record the pid IMMEDIATELY (the daemon re-writes the same file at
+    # startup): if this provision request is cancelled before the agent
+    # finishes booting, teardown/orphan-sweep must still find the pid —
+    # the boot-window gap leaked agent processes in CI.
+    (cdir / "agent.json").write_text(
+        json.dumps({"port": port, "pid": proc.pid}))
     AgentClient(port, token=tok).wait_ready(timeout=30)
     return port
 

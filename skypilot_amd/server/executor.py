@@ -112,7 +112,7 @@ class RequestWorker(threading.Thread):
             if busy >= self.parallelism:
                 time.sleep(0.05)
                 return
-            req = rdb.claim_next(self.queue, os.getpid())
+            req = rdb.claim_next(self.queue, None)
             if req is None:
                 time.sleep(0.05)
                 return
@@ -189,11 +189,15 @@ def schedule(name: str, body: Dict[str, Any],
 
 
 def cancel_request(rid: str) -> bool:
-    pid = rdb.mark_cancelled(rid)
-    if pid:
+    marked, pid = rdb.mark_cancelled(rid)
+    # Only a DEDICATED request-runner pid may be signalled.  Inline
+    # (SHORT) requests run in a thread of this very process and LONG
+    # requests have a claim->spawn window before set_pid records the
+    # runner: killing the recorded pid blindly SIGTERMed the whole
+    # server (observed as intermittent full-test-suite terminations).
+    if marked and pid and pid != os.getpid():
         try:
             os.kill(pid, signal.SIGTERM)
         except (ProcessLookupError, PermissionError):
             pass
-        return True
-    return pid is not None
+    return marked

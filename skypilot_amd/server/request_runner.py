@@ -17,6 +17,9 @@ def main() -> int:
     req = rdb.get(rid)
     if req is None:
         return 1
+    if req["status"] == rdb.CANCELLED:
+        return 0  # cancelled in the claim->spawn window: never execute
+    rdb.set_pid(rid, os.getpid())  # the runner is the killable pid
     with open(req["log_path"], "ab", buffering=0) as logf:
         os.dup2(logf.fileno(), 1)
         os.dup2(logf.fileno(), 2)

@@ -148,18 +148,19 @@ def list_requests(limit: int = 100) -> List[Dict[str, Any]]:
     return out
 
 
-def mark_cancelled(rid: str) -> Optional[int]:
-    """Mark cancelled; returns worker pid if it was running."""
+def mark_cancelled(rid: str):
+    """Mark cancelled.  Returns (marked, worker_pid): worker_pid is the
+    dedicated runner's pid or None (inline request / not yet spawned)."""
     with _conn() as c:
         row = c.execute(
             "SELECT status, worker_pid FROM requests WHERE request_id=?",
             (rid,)).fetchone()
         if row is None or row[0] in TERMINAL:
-            return None
+            return False, None
         c.execute(
             "UPDATE requests SET status=?, finished_at=? WHERE request_id=?",
             (CANCELLED, time.time(), rid))
-    return row[1]
+    return True, row[1]
 
 
 def gc_requests(max_age_days: float = None, keep_latest: int = None
