@@ -261,12 +261,32 @@ def _ensure_agent(cdir: Path, gpu_ids: List[int],
     return port
 
 
+def _drain_jobs(handle: Dict[str, Any]) -> None:
+    """Cancel every non-terminal job straight through the on-disk job
+    table (guarded killpg of the drivers).  Teardown used to kill only
+    the AGENT; job drivers run in their own sessions and survived a
+    cancel->teardown sequence (leaked `sleep`-ing drivers in CI)."""
+    cdir = handle.get("cluster_dir")
+    if not cdir or not Path(cdir).exists():
+        return
+    try:
+        from skypilot_amd.agent import job_lib
+        t = job_lib.JobTable(cdir)
+        for j in t.list():
+            if j["status"] in job_lib.NONTERMINAL:
+                t.cancel(j["job_id"])
+    except Exception:  # noqa: BLE001 — teardown is best-effort
+        pass
+
+
 def stop_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
     """Stop = kill the agent + all jobs; runtime dir and GPU lease kept."""
+    _drain_jobs(handle)
     _kill_agent(handle)
 
 
 def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
+    _drain_jobs(handle)
     _kill_agent(handle)
     import shutil
     cdir = handle.get("cluster_dir")

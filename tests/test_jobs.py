@@ -69,6 +69,8 @@ def test_managed_job_failure_budget_exhausted(client):
 
 
 def test_managed_job_cancel(client):
+    import subprocess
+    import time as _t
     from skypilot_amd.client import sdk
     task = {"name": "mj-sleep", "run": "sleep 600"}
     res = sdk.get(sdk.jobs_launch(task, "mj-sleep"))
@@ -77,6 +79,20 @@ def test_managed_job_cancel(client):
     assert n == 1
     job = _wait_managed(res["job_id"], {"CANCELLED"})
     assert job["status"] == "CANCELLED"
+    # no leaked driver: cancel + teardown must reap the job driver
+    # process (it runs in its own session and used to survive)
+    import os
+    home = os.environ["SKY_AMD_HOME"]
+    deadline = _t.time() + 30
+    while _t.time() < deadline:
+        out = subprocess.run(
+            ["ps", "-eo", "args"], capture_output=True, text=True).stdout
+        leaked = [ln for ln in out.splitlines()
+                  if "skypilot_amd.agent.driver" in ln and home in ln]
+        if not leaked:
+            break
+        _t.sleep(1)
+    assert not leaked, leaked
 
 
 def test_managed_job_preemption_recovery(client, tmp_path):
