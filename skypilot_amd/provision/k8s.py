@@ -177,16 +177,16 @@ def _port_forward(pod: str, settings) -> int:
 
 
 def stop_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
+    # k8s has no "stopped pod": drain jobs, then delete the pods (the
+    # cluster record survives, `sky start` re-provisions fresh pods).
+    # Leaving the pod running made a STOPPED cluster flip back to UP on
+    # the next refresh (query_instances probes the agent).
     port = handle.get("agent_port")
     if port:
         try:
             AgentClient(port).cancel_all()
         except Exception:  # noqa: BLE001
             pass
-
-
-def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
-    stop_instances(cluster_name, handle)
     settings = k8s_settings()
     for pod in handle.get("pods") or ([handle["pod"]]
                                       if handle.get("pod") else []):
@@ -194,6 +194,10 @@ def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
             _run_kubectl(["delete", "pod", pod, "--wait=false"], settings)
         except (OSError, subprocess.TimeoutExpired):
             pass
+
+
+def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
+    stop_instances(cluster_name, handle)
 
 
 def query_instances(cluster_name: str, handle: Dict[str, Any]) -> str:

@@ -182,17 +182,7 @@ def _tunnel(host: Dict[str, Any], remote_port: int) -> int:
     return local_port
 
 
-def stop_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
-    port = handle.get("agent_port")
-    if port:
-        try:
-            AgentClient(port).cancel_all()
-        except Exception:  # noqa: BLE001
-            pass
-
-
-def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
-    stop_instances(cluster_name, handle)
+def _kill_remote_agents(handle: Dict[str, Any]) -> None:
     for h in handle.get("hosts", []):
         try:
             # Kill by the exact pid the agent recorded (never by pattern).
@@ -203,6 +193,24 @@ def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
                 timeout=30)
         except Exception:  # noqa: BLE001
             pass
+
+
+def stop_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
+    # Drain jobs through the live agent (guarded driver killpg), THEN
+    # kill the agent: a stopped cluster with a live agent would flip
+    # back to UP on the next status refresh (query_instances probes the
+    # agent port).
+    port = handle.get("agent_port")
+    if port:
+        try:
+            AgentClient(port).cancel_all()
+        except Exception:  # noqa: BLE001
+            pass
+    _kill_remote_agents(handle)
+
+
+def terminate_instances(cluster_name: str, handle: Dict[str, Any]) -> None:
+    stop_instances(cluster_name, handle)
 
 
 def query_instances(cluster_name: str, handle: Dict[str, Any]) -> str:
