@@ -317,7 +317,8 @@ def test_attention_shape_fuzz():
             (1, 64, 8, 8, True), (3, 128, 16, 2, True),
             (2, 256, 32, 8, False), (1, 1024, 8, 2, True),
             (2, 192, 4, 4, False),
-            # S % 256 == 0 routes to the v3 8-wave 32x32 kernel
+            # S % 256 == 0 routes to the v3 32x32 kernel (4-wave by
+            # default; SKY_ATTN_FWD_V3_NW=8 selects the 8-wave twin)
             (2, 512, 16, 4, True), (1, 256, 8, 8, True),
             # v3 edge shapes around the r02 gl_lds barrier race:
             # minimal paired grid (nq=2), single-head, non-causal v3,
