@@ -57,3 +57,35 @@ def test_example_autostop_recorded(client):
     finally:
         a.close()
     sdk.get(sdk.down("smoke-as"))
+
+
+def test_volume_mount_persists_across_clusters(client, tmp_path):
+    """`file_mounts: {path: {volume: name}}` mounts a persistent volume
+    (reference: sky/volumes); data written by one cluster is visible to
+    the next.  Mount targets live under tmp_path — absolute mount paths
+    are created verbatim, so tests must never use root-level ones."""
+    from skypilot_amd.client import sdk
+    from skypilot_amd.data import volumes
+    volumes.create("smoke-vol", size_gb=1)
+    m1 = str(tmp_path / "vmnt")
+    m2 = str(tmp_path / "vmnt2")
+    try:
+        task1 = {"run": f"echo vol-data-$((3*3)) > {m1}/out.txt",
+                 "file_mounts": {m1: {"volume": "smoke-vol"}},
+                 "resources": {"cpus": 1}}
+        res = sdk.get(sdk.launch(task1, "vol-c1"), timeout=90)
+        j = _wait_job(sdk, "vol-c1", res["job_id"])
+        assert j["status"] == "SUCCEEDED", j
+        sdk.get(sdk.down("vol-c1"))
+        # second cluster sees the data
+        task2 = {"run": f"cat {m2}/out.txt",
+                 "file_mounts": {m2: {"volume": "smoke-vol"}},
+                 "resources": {"cpus": 1}}
+        res = sdk.get(sdk.launch(task2, "vol-c2"), timeout=90)
+        j = _wait_job(sdk, "vol-c2", res["job_id"])
+        assert j["status"] == "SUCCEEDED", j
+        vp = volumes.mount_path("smoke-vol")
+        assert (vp / "out.txt").read_text().strip() == "vol-data-9"
+        sdk.get(sdk.down("vol-c2"))
+    finally:
+        volumes.delete("smoke-vol")
