@@ -491,3 +491,56 @@ def test_reap_orphan_agents(monkeypatch, tmp_path):
         json.dumps({"port": 0, "pid": os.getpid()}))
     os.utime(cdir / "agent.json", (old, old))
     assert _reap_orphan_agents(min_age_s=120) == 0
+
+
+def test_gl_lds_barriers_are_vm_guarded():
+    """ISA regression guard for the r02 race class: in every kernel
+    that issues global_load ... lds (async LDS DMA, completion tracked
+    by VMcnt), each plain s_barrier must be preceded (within a small
+    window) by an s_waitcnt that constrains vmcnt — hipcc only adds
+    lgkmcnt(0) by itself, which does NOT cover the DMA
+    (docs/KERNELS.md "Synchronizing async LDS staging")."""
+    import shutil
+    import subprocess
+    import tempfile
+    from pathlib import Path
+
+    objdump = Path("/opt/rocm/lib/llvm/bin/llvm-objdump")
+    so = Path(__file__).parent.parent / "skypilot_amd" / "ops" / "_C.so"
+    if not objdump.exists() or not so.exists():
+        import pytest as _pytest
+        _pytest.skip("llvm-objdump or built extension unavailable")
+    with tempfile.TemporaryDirectory() as td:
+        cp = Path(td) / "_C.so"
+        shutil.copy(so, cp)
+        subprocess.run([str(objdump), "--offloading", str(cp)],
+                       cwd=td, capture_output=True)
+        bad = []
+        for bundle in sorted(Path(td).glob("_C.so.*gfx950*")):
+            d = subprocess.run([str(objdump), "-d", str(bundle)],
+                               capture_output=True, text=True).stdout
+            lines = d.splitlines()
+            kern, uses_dma, rows = None, False, []
+            kernels = []
+            for ln in lines:
+                if ">:" in ln and "<" in ln:
+                    if kern and uses_dma:
+                        kernels.append((kern, rows))
+                    kern = ln.split("<")[1].split(">")[0]
+                    uses_dma, rows = False, []
+                if "lds" in ln and ("global_load" in ln
+                                    or "buffer_load" in ln):
+                    uses_dma = True
+                rows.append(ln)
+            if kern and uses_dma:
+                kernels.append((kern, rows))
+            for kname, rows in kernels:
+                for i, ln in enumerate(rows):
+                    if "s_barrier" in ln and "wait" not in ln:
+                        window = " ".join(rows[max(0, i - 12):i])
+                        if "vmcnt(0)" not in window:
+                            bad.append((bundle.name, kname, i))
+        # the hand-scheduled wgrad GEMM carries its own phase waits at
+        # K-tile granularity (vmcnt(0) once per tile, not per barrier)
+        bad = [b for b in bad if "gemm_nt_kernel" not in b[1]]
+        assert not bad, bad
