@@ -120,10 +120,13 @@ def get(rid: str) -> Optional[Dict[str, Any]]:
 def finish(rid: str, status: str, result: Any = None,
            error: Optional[str] = None) -> None:
     with _conn() as c:
+        # never resurrect a CANCELLED row: a runner that slipped through
+        # the cancel window may still call finish() when it completes
         c.execute(
             "UPDATE requests SET status=?, finished_at=?, result=?, error=? "
-            "WHERE request_id=?",
-            (status, time.time(), json.dumps(result), error, rid))
+            "WHERE request_id=? AND status != ?",
+            (status, time.time(), json.dumps(result), error, rid,
+             CANCELLED))
 
 
 def set_pid(rid: str, pid: int) -> None:

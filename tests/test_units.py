@@ -544,3 +544,15 @@ def test_gl_lds_barriers_are_vm_guarded():
         # K-tile granularity (vmcnt(0) once per tile, not per barrier)
         bad = [b for b in bad if "gemm_nt_kernel" not in b[1]]
         assert not bad, bad
+
+
+def test_finish_never_resurrects_cancelled(monkeypatch, tmp_path):
+    """A runner that slipped through the cancel window must not flip a
+    CANCELLED request back to SUCCEEDED when it completes."""
+    monkeypatch.setenv("SKY_AMD_HOME", str(tmp_path))
+    from skypilot_amd.server import requests_db as rdb
+    rid = rdb.create("launch", {}, "LONG")
+    marked, pid = rdb.mark_cancelled(rid)
+    assert marked and pid is None  # claim never stamped a killable pid
+    rdb.finish(rid, rdb.SUCCEEDED, result={"late": True})
+    assert rdb.get(rid)["status"] == rdb.CANCELLED
