@@ -256,20 +256,25 @@ def create_app(start_workers: bool = True) -> FastAPI:
                 403, f"request {rid} belongs to {req.get('user')!r}")
         return {"cancelled": executor.cancel_request(rid)}
 
+    # dashboard pages carry the same identity gate as the API: in token
+    # auth mode an anonymous browser must not see cluster metadata/logs
     @app.get("/dashboard")
-    def dashboard():
+    def dashboard(request: Request):
+        _identity(request)
         from fastapi.responses import HTMLResponse
         from skypilot_amd.server.dashboard import render
         return HTMLResponse(render())
 
     @app.get("/dashboard/cluster/{cluster_name}")
-    def dashboard_cluster(cluster_name: str):
+    def dashboard_cluster(cluster_name: str, request: Request):
+        _identity(request)
         from fastapi.responses import HTMLResponse
         from skypilot_amd.server.dashboard import render_cluster
         return HTMLResponse(render_cluster(cluster_name))
 
     @app.get("/dashboard/cluster/{cluster_name}/job/{job_id}")
-    def dashboard_job(cluster_name: str, job_id: int):
+    def dashboard_job(cluster_name: str, job_id: int, request: Request):
+        _identity(request)
         from fastapi.responses import HTMLResponse
         from skypilot_amd.server.dashboard import render_job_logs
         return HTMLResponse(render_job_logs(cluster_name, job_id))

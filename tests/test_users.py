@@ -86,6 +86,11 @@ def test_token_auth_mode_rejects_header_identity(sky_env, client,
     assert client.post("/api/v1/status", json={},
                        headers={"Authorization": f"Bearer {tok}"}
                        ).status_code == 200
+    # the dashboard carries the same gate
+    assert client.get("/dashboard").status_code == 401
+    assert client.get("/dashboard",
+                      headers={"Authorization": f"Bearer {tok}"}
+                      ).status_code == 200
     # trusted-proxy header accepted only when explicitly configured
     hdr = {"X-Auth-Request-Email": "eve@corp"}
     assert client.post("/api/v1/status", json={},
